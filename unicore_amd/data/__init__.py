@@ -1,0 +1,58 @@
+"""Data pipeline: composable wrapper datasets + resumable iterators
+(parity: reference unicore/data/__init__.py:9-33)."""
+
+from .unicore_dataset import UnicoreDataset, EpochListening
+from .base_wrapper_dataset import BaseWrapperDataset
+from .dictionary import Dictionary
+from .lru_cache_dataset import LRUCacheDataset
+from .lmdb_dataset import LMDBDataset
+from .tokenize_dataset import TokenizeDataset, BertTokenizeDataset
+from .mask_tokens_dataset import MaskTokensDataset
+from .pad_dataset import (
+    PadDataset,
+    LeftPadDataset,
+    RightPadDataset,
+    RightPadDataset2D,
+)
+from .sort_dataset import SortDataset, EpochShuffleDataset
+from .nested_dictionary_dataset import NestedDictionaryDataset
+from .misc_datasets import (
+    AppendTokenDataset,
+    FromNumpyDataset,
+    NumelDataset,
+    NumSamplesDataset,
+    PrependTokenDataset,
+    RawArrayDataset,
+    RawLabelDataset,
+    RawNumpyDataset,
+)
+from . import data_utils, iterators
+
+__all__ = [
+    "AppendTokenDataset",
+    "BaseWrapperDataset",
+    "BertTokenizeDataset",
+    "Dictionary",
+    "EpochListening",
+    "EpochShuffleDataset",
+    "FromNumpyDataset",
+    "LeftPadDataset",
+    "LMDBDataset",
+    "LRUCacheDataset",
+    "MaskTokensDataset",
+    "NestedDictionaryDataset",
+    "NumelDataset",
+    "NumSamplesDataset",
+    "PadDataset",
+    "PrependTokenDataset",
+    "RawArrayDataset",
+    "RawLabelDataset",
+    "RawNumpyDataset",
+    "RightPadDataset",
+    "RightPadDataset2D",
+    "SortDataset",
+    "TokenizeDataset",
+    "UnicoreDataset",
+    "data_utils",
+    "iterators",
+]

@@ -1,0 +1,314 @@
+"""FlatDDP — the framework's own backward-overlapped bucketed DDP engine.
+
+Functional replacement for torch-c10d DDP in the reference stack
+(reference unicore/models/distributed_unicore_model.py:37-48), redesigned
+MI355X-first:
+
+- Gradients live in preallocated flat bucket buffers; every ``param.grad`` is
+  a *view* into its bucket, so backward accumulates in place and grad sync is
+  one RCCL all-reduce per bucket with zero gather/scatter copies.
+- All-reduces launch on a dedicated side HIP stream as soon as a bucket's
+  last gradient lands (post-accumulate-grad hooks), overlapping xGMI traffic
+  with the remainder of backward. hipEvents order the comm stream against the
+  compute stream in both directions.
+- Buckets are filled in reverse parameter-registration order (the approximate
+  autograd completion order), sized by ``--bucket-cap-mb``. On xGMI the ring
+  all-reduce is per-link bound (~153 GB/s/link), so larger buckets than
+  NVSwitch defaults amortize launch latency; the default here is 32 MB.
+- Gradients are pre-divided by the world size before reduction (same
+  numerics contract as the reference's engines,
+  reference unicore/distributed/legacy_distributed_data_parallel.py:104-105).
+- Unused parameters need no flag: any bucket whose hooks did not all fire is
+  reduced at ``finish_grad_sync`` time (its untouched views are zero), so
+  partial graphs cannot hang the engine.
+
+On CPU (gloo) the same code runs with ``async_op=True`` handles instead of a
+side stream, which is how the multi-process CPU tests cover the engine.
+"""
+
+import logging
+from contextlib import contextmanager
+from typing import List
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+logger = logging.getLogger(__name__)
+
+
+class _Bucket:
+    __slots__ = [
+        "params",
+        "flat",
+        "views",
+        "pending",
+        "launched",
+        "work",
+        "event",
+    ]
+
+    def __init__(self):
+        self.params: List[nn.Parameter] = []
+        self.flat = None
+        self.views = []
+        self.pending = 0
+        self.launched = False
+        self.work = None
+        self.event = None
+
+
+class FlatDDP(nn.Module):
+    """DistributedDataParallel with flat grad buckets + backward overlap."""
+
+    def __init__(
+        self,
+        module: nn.Module,
+        process_group,
+        bucket_cap_mb: float = 32,
+        world_size: int = None,
+    ):
+        super().__init__()
+        self.module = module
+        self.process_group = process_group
+        self.world_size = (
+            world_size
+            if world_size is not None
+            else (dist.get_world_size(process_group) if dist.is_initialized() else 1)
+        )
+        self.bucket_bytes = int(bucket_cap_mb * 1024 * 1024)
+        self.accumulate_grads = False  # no_sync flag
+
+        self._use_cuda = any(p.is_cuda for p in module.parameters())
+        self._comm_stream = (
+            torch.cuda.Stream() if self._use_cuda and self.world_size > 1 else None
+        )
+
+        self._buckets: List[_Bucket] = []
+        self._param_to_bucket = {}
+        self._hook_handles = []
+        self._build_buckets()
+        self._register_hooks()
+
+    # ------------------------------------------------------------------
+    # bucket construction
+    # ------------------------------------------------------------------
+
+    def _build_buckets(self):
+        params = [p for p in self.module.parameters() if p.requires_grad]
+
+        # Fast path: the mixed-precision optimizer has already flattened
+        # grads (every p.grad is a view into one flat tensor per group).
+        # Bucket over contiguous RANGES of those flats — zero extra memory,
+        # and the all-reduce payloads are the very tensors the optimizer
+        # reads, so there are no gather/scatter copies anywhere.
+        aliased = self._try_alias_flat_grads(params)
+        if not aliased:
+            self._allocate_own_buckets(params)
+
+        if len(self._buckets) > 0:
+            nbytes = sum(b.flat.numel() * b.flat.element_size() for b in self._buckets)
+            logger.info(
+                "FlatDDP: {} params in {} buckets ({:.1f} MB flat grads{})".format(
+                    len(self._param_to_bucket),
+                    len(self._buckets),
+                    nbytes / 1e6,
+                    ", aliased onto optimizer flats" if aliased else "",
+                )
+            )
+
+    def _try_alias_flat_grads(self, params):
+        bases = {}
+        for p in params:
+            g = p.grad
+            if g is None or g._base is None:
+                return False
+            base = g._base
+            bases.setdefault(id(base), (base, []))[1].append(p)
+        for _, (base, plist) in bases.items():
+            plist.sort(key=lambda p: p.grad.storage_offset())
+            cur = None
+            cur_start = cur_end = 0
+            for p in plist:
+                start = p.grad.storage_offset()
+                end = start + p.numel()
+                psize = p.numel() * p.element_size()
+                if (
+                    cur is None
+                    or (cur_end - cur_start) * p.element_size() + psize
+                    > self.bucket_bytes
+                ):
+                    if cur is not None:
+                        cur.flat = base[cur_start:cur_end]
+                    cur = _Bucket()
+                    self._buckets.append(cur)
+                    cur_start = start
+                cur.params.append(p)
+                cur.views.append(p.grad)
+                self._param_to_bucket[p] = cur
+                cur_end = end
+            if cur is not None:
+                cur.flat = base[cur_start:cur_end]
+        return True
+
+    def _allocate_own_buckets(self, params):
+        # reverse registration order approximates backward completion order
+        params = list(reversed(params))
+
+        cur = None
+        cur_key = None
+        cur_bytes = 0
+        for p in params:
+            key = (p.device, p.dtype)
+            psize = p.numel() * p.element_size()
+            if (
+                cur is None
+                or key != cur_key
+                or (cur_bytes + psize > self.bucket_bytes and cur_bytes > 0)
+            ):
+                cur = _Bucket()
+                self._buckets.append(cur)
+                cur_key = key
+                cur_bytes = 0
+            cur.params.append(p)
+            cur_bytes += psize
+
+        for b in self._buckets:
+            total = sum(p.numel() for p in b.params)
+            device, dtype = b.params[0].device, b.params[0].dtype
+            b.flat = torch.zeros(total, dtype=dtype, device=device)
+            offset = 0
+            for p in b.params:
+                view = b.flat[offset : offset + p.numel()].view_as(p)
+                b.views.append(view)
+                p.grad = view
+                offset += p.numel()
+                self._param_to_bucket[p] = b
+
+    def _register_hooks(self):
+        for p in self._param_to_bucket:
+            handle = p.register_post_accumulate_grad_hook(self._make_hook())
+            self._hook_handles.append(handle)
+
+    def detach_hooks(self):
+        """Remove all grad hooks (call before re-wrapping the module)."""
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
+
+    def _make_hook(self):
+        def hook(param):
+            if self.accumulate_grads or self.world_size == 1:
+                return
+            bucket = self._param_to_bucket[param]
+            # re-attach the grad view if something replaced it (e.g. a
+            # zero_grad(set_to_none=True) outside our control)
+            if param.grad is not None and param.grad.data_ptr() != self._view_ptr(
+                bucket, param
+            ):
+                self._restore_view(bucket, param)
+            bucket.pending -= 1
+            if bucket.pending == 0 and not bucket.launched:
+                self._launch_reduce(bucket)
+
+        return hook
+
+    @staticmethod
+    def _view_ptr(bucket, param):
+        idx = bucket.params.index(param)
+        return bucket.views[idx].data_ptr()
+
+    @staticmethod
+    def _restore_view(bucket, param):
+        idx = bucket.params.index(param)
+        bucket.views[idx].copy_(param.grad)
+        param.grad = bucket.views[idx]
+
+    # ------------------------------------------------------------------
+    # reduction
+    # ------------------------------------------------------------------
+
+    def _launch_reduce(self, bucket: _Bucket):
+        bucket.launched = True
+        if self.world_size == 1:
+            return
+        if self._comm_stream is not None:
+            # compute stream -> comm stream ordering: the bucket's grads must
+            # all have landed before the all-reduce reads them
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.cuda.stream(self._comm_stream):
+                ev.wait(self._comm_stream)
+                bucket.flat.div_(self.world_size)
+                dist.all_reduce(bucket.flat, group=self.process_group)
+                bucket.event = torch.cuda.Event()
+                bucket.event.record(self._comm_stream)
+        else:
+            bucket.flat.div_(self.world_size)
+            bucket.work = dist.all_reduce(
+                bucket.flat, group=self.process_group, async_op=True
+            )
+
+    def prepare_for_backward(self):
+        """Arm the buckets for the next backward pass."""
+        for b in self._buckets:
+            b.pending = len(b.params)
+            b.launched = False
+            b.work = None
+            b.event = None
+            # re-pin any grads that were detached from their views
+            for p, v in zip(b.params, b.views):
+                if p.grad is None or p.grad.data_ptr() != v.data_ptr():
+                    p.grad = v
+
+    def finish_grad_sync(self):
+        """Block the compute stream until every bucket's all-reduce is done.
+
+        Buckets whose hooks did not all fire (unused parameters in this
+        forward) are reduced here — their stale views are zero after
+        zero_grad, so the reduction is still correct.
+        """
+        if self.accumulate_grads or self.world_size == 1:
+            return
+        for b in self._buckets:
+            if not b.launched:
+                self._launch_reduce(b)
+        for b in self._buckets:
+            if b.event is not None:
+                b.event.wait(torch.cuda.current_stream())
+            if b.work is not None:
+                b.work.wait()
+
+    def zero_grad_buffers(self):
+        """Zero the flat grad buffers (one memset per bucket)."""
+        for b in self._buckets:
+            b.flat.zero_()
+            for p, v in zip(b.params, b.views):
+                p.grad = v
+
+    # ------------------------------------------------------------------
+    # nn.Module plumbing
+    # ------------------------------------------------------------------
+
+    @contextmanager
+    def no_sync(self):
+        """Context manager to disable gradient sync (grad accumulation)."""
+        old = self.accumulate_grads
+        self.accumulate_grads = True
+        yield
+        self.accumulate_grads = old
+
+    def forward(self, *args, **kwargs):
+        if not self.accumulate_grads:
+            self.prepare_for_backward()
+        return self.module(*args, **kwargs)
+
+    def all_reduce_grads(self):
+        """Explicit post-backward sync entry point (Trainer calls this)."""
+        self.finish_grad_sync()
+
+    def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, *args, **kwargs):
+        return self.module.load_state_dict(*args, **kwargs)

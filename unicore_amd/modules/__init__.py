@@ -1,0 +1,14 @@
+"""isort:skip_file"""
+
+from .layer_norm import LayerNorm
+from .rms_norm import RMSNorm
+from .softmax_dropout import softmax_dropout
+from .multihead_attention import SelfMultiheadAttention, CrossMultiheadAttention
+from .transformer_encoder_layer import TransformerEncoderLayer
+from .transformer_encoder import (
+    TransformerEncoder,
+    init_bert_params,
+    relative_position_bucket,
+)
+from .transformer_decoder_layer import TransformerDecoderLayer
+from .transformer_decoder import TransformerDecoder

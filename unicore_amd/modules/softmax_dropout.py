@@ -1,0 +1,207 @@
+"""Fused softmax(+attn-mask,+pair-bias)+dropout.
+
+Functional parity with the reference's softmax_dropout op
+(reference unicore/modules/softmax_dropout.py:100-144 and
+csrc/softmax_dropout/*): softmax over the last dim of a ``(..., q, k)``
+tensor with optional additive attention mask and pair bias (both broadcast),
+fused with dropout whose mask is stored as a per-lane bitfield.
+
+The GPU path is a wave64 CDNA4 kernel (rows register-resident, 64-lane
+shuffle reductions, in-kernel Philox, vectorized 16B/lane loads) for
+k <= 4096 and an LDS-staged block kernel beyond. The eager path below is the
+numerics oracle used on CPU and in parity tests.
+
+Broadcast contract (shim + kernel): for each of mask/bias, the non-1
+broadcast dims (right-aligned against the input batch dims) must form one
+contiguous block. The kernel addresses the source row as
+``((row_batch / outer_div) % src_nb) * src_q + (qi % src_q)`` with
+``outer_div`` = product of dims inner to the block — no broadcast is ever
+materialized. Other patterns (non-contiguous blocks) are added eagerly
+before the kernel.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+from torch import Tensor
+
+
+def _broadcast_descr(src_batch_shape, batch_dims):
+    """Return (src_nb, outer_div) for a right-aligned broadcast of
+    ``src_batch_shape`` against ``batch_dims``; None if the non-1 dims do
+    not form one contiguous block matching the input dims."""
+    m = len(batch_dims)
+    if len(src_batch_shape) > m:
+        extra = src_batch_shape[: len(src_batch_shape) - m]
+        for e in extra:
+            if e != 1:
+                return None
+        src_batch_shape = src_batch_shape[len(src_batch_shape) - m :]
+    padded = [1] * (m - len(src_batch_shape)) + list(src_batch_shape)
+    non1 = [i for i, e in enumerate(padded) if e != 1]
+    if len(non1) == 0:
+        return 1, 1
+    i, j = non1[0], non1[-1]
+    for t in range(i, j + 1):
+        if padded[t] != batch_dims[t]:
+            return None
+    src_nb = 1
+    for t in range(i, j + 1):
+        src_nb *= batch_dims[t]
+    outer_div = 1
+    for t in range(j + 1, m):
+        outer_div *= batch_dims[t]
+    return src_nb, outer_div
+
+
+class SoftmaxDropoutFast(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx,
+        is_training,
+        inputs,
+        mask,
+        mask_outer_div,
+        bias,
+        bias_outer_div,
+        dropout_prob,
+    ):
+        """inputs: (n_batch, q, k) contiguous; mask (mb, mq, k) or None;
+        bias (bb, bq, k) or None."""
+        from unicore_amd import ops
+
+        dropout_results, dropout_mask, softmax_results = ops.softmax_dropout_fwd(
+            is_training,
+            inputs,
+            mask,
+            mask_outer_div,
+            bias,
+            bias_outer_div,
+            dropout_prob,
+        )
+        ctx.dropout_prob = dropout_prob
+        ctx.has_bias = bias is not None
+        if bias is not None:
+            ctx.bias_shape = bias.shape
+            ctx.bias_outer_div = bias_outer_div
+        ctx.save_for_backward(softmax_results, dropout_mask)
+        return dropout_results
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        from unicore_amd import ops
+
+        softmax_results, dropout_mask = ctx.saved_tensors
+        grad_output = grad_output.contiguous()
+        # in-place on grad_output (same contract as the reference backward,
+        # reference csrc/softmax_dropout/softmax_dropout_kernel.cu:277-278)
+        grad_input = ops.softmax_dropout_bwd(
+            grad_output, softmax_results, dropout_mask, ctx.dropout_prob
+        )
+        grad_bias = None
+        if ctx.has_bias:
+            n_batch, q, k = softmax_results.shape
+            bb, bq, _ = ctx.bias_shape
+            od = ctx.bias_outer_div
+            a = n_batch // (bb * od)
+            g = grad_input.view(a, bb, od, q, k).sum(dim=(0, 2))
+            if bq == 1 and q != 1:
+                g = g.sum(dim=1, keepdim=True)
+            grad_bias = g
+        return None, grad_input, None, None, grad_bias, None, None
+
+
+def _eager_softmax_dropout(inputs, dropout_prob, is_training, mask, bias):
+    x = inputs
+    if mask is not None:
+        x = x + mask
+    if bias is not None:
+        x = x + bias
+    x = F.softmax(x, dim=-1)
+    if is_training and dropout_prob > 0:
+        x = F.dropout(x, p=dropout_prob)
+    return x
+
+
+def softmax_dropout(
+    inputs: Tensor,
+    dropout_prob: float,
+    is_training: bool = True,
+    mask: Optional[Tensor] = None,
+    bias: Optional[Tensor] = None,
+    inplace: bool = True,
+) -> Tensor:
+    """softmax(inputs + mask + bias) over the last dim, with fused dropout.
+
+    Shapes: inputs (..., q, k); mask/bias broadcastable to inputs.
+    """
+    input_shape = inputs.shape
+    use_kernel = False
+    if inputs.is_cuda:
+        from unicore_amd import ops
+
+        # the kernel path is mandatory on GPU: ops.* raises if the extension
+        # is missing (unless UNICORE_AMD_ALLOW_EAGER=1)
+        use_kernel = ops.gpu_kernels_available() or not ops.allow_eager_on_gpu()
+
+    if not use_kernel:
+        return _eager_softmax_dropout(inputs, dropout_prob, is_training, mask, bias)
+
+    inputs = inputs.contiguous()
+    if not inplace:
+        inputs = inputs.clone()
+    k = input_shape[-1]
+    q = input_shape[-2]
+    batch_dims = tuple(input_shape[:-2])
+    n_batch = 1
+    for d in batch_dims:
+        n_batch *= d
+    inputs_3d = inputs.view(n_batch, q, k)
+
+    mask_k = bias_k = None
+    mask_od = bias_od = 1
+    if mask is not None:
+        mask_k, mask_od, inputs_3d = _prep_additive(mask, inputs_3d, batch_dims, q, k)
+    if bias is not None:
+        bias_k, bias_od, inputs_3d = _prep_additive(bias, inputs_3d, batch_dims, q, k)
+
+    if k <= 4096:
+        out = SoftmaxDropoutFast.apply(
+            is_training, inputs_3d, mask_k, mask_od, bias_k, bias_od, dropout_prob
+        )
+    else:
+        # very wide rows: fused softmax (p=0) via the LDS block kernel,
+        # separate torch dropout (reference routes the same way,
+        # unicore/modules/softmax_dropout.py:131-138)
+        out = SoftmaxDropoutFast.apply(
+            is_training, inputs_3d, mask_k, mask_od, bias_k, bias_od, 0.0
+        )
+        if is_training and dropout_prob > 0:
+            out = F.dropout(out, p=dropout_prob)
+    return out.view(input_shape)
+
+
+def _prep_additive(src, inputs_3d, batch_dims, q, k):
+    """Normalize mask/bias for the kernel.
+
+    Returns (kernel_tensor (src_nb, sq, k) or None, outer_div, inputs_3d).
+    When the broadcast pattern is unsupported the add is merged eagerly and
+    (None, 1, merged_inputs) is returned.
+    """
+    n_batch = inputs_3d.shape[0]
+    ok = src.shape[-1] == k and src.dim() >= 1
+    sq = src.shape[-2] if src.dim() >= 2 else 1
+    if ok and sq not in (1, q):
+        ok = False
+    descr = _broadcast_descr(tuple(src.shape[:-2]), batch_dims) if ok else None
+    if descr is None:
+        merged = (
+            inputs_3d.view(*batch_dims, q, k) + src
+        ).view(n_batch, q, k)
+        return None, 1, merged
+    src_nb, outer_div = descr
+    src_c = src.reshape(src_nb, sq, k)
+    if not src_c.is_contiguous():
+        src_c = src_c.contiguous()
+    return src_c, outer_div, inputs_3d

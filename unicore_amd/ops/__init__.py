@@ -1,0 +1,139 @@
+"""Python surface over the gfx950 HIP kernel extension.
+
+The extension (``unicore_amd._kernels``) is built in-tree by setup.py with
+``PYTORCH_ROCM_ARCH=gfx950`` and provides the fused hot ops:
+softmax(+bias,+mask)+dropout fwd/bwd, LayerNorm/RMSNorm fwd/bwd, fused AdamW,
+multi-tensor L2 norm, and fp32->bf16 stochastic rounding
+(CDNA4-native equivalents of the reference's 8 CUDA extensions,
+reference csrc/ + setup.py:141-387).
+
+Policy: on a CUDA/ROCm device the HIP path is mandatory — if the extension
+failed to import, any op called with GPU tensors raises loudly (no silent
+eager fallback on GPU). On CPU the modules use their eager reference path,
+which also serves as the numerics oracle in tests.
+"""
+
+import logging
+import os
+
+import torch
+
+logger = logging.getLogger(__name__)
+
+_kernels = None
+_import_error = None
+
+try:
+    from unicore_amd import _kernels  # type: ignore[attr-defined]
+except ImportError as e:  # extension not built
+    _kernels = None
+    _import_error = e
+
+
+def has_kernels() -> bool:
+    return _kernels is not None
+
+
+def require_kernels():
+    """Raise if the HIP extension is unavailable.
+
+    Called from every op entry point when tensors live on the GPU: running
+    eager fallbacks silently on an MI355X would defeat the point of the
+    framework, so we fail loudly instead.
+    """
+    if _kernels is None:
+        raise RuntimeError(
+            "unicore_amd._kernels HIP extension is not available "
+            f"(import error: {_import_error}). Build it in-tree with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+            "Set UNICORE_AMD_ALLOW_EAGER=1 only for debugging."
+        )
+
+
+def allow_eager_on_gpu() -> bool:
+    return os.environ.get("UNICORE_AMD_ALLOW_EAGER", "0") == "1"
+
+
+def gpu_kernels_available() -> bool:
+    """True if ops on CUDA tensors should take the HIP kernel path."""
+    return _kernels is not None
+
+
+# ---------------------------------------------------------------------------
+# Thin wrappers (keep all tensor-shape munging in the module-level shims)
+# ---------------------------------------------------------------------------
+
+
+def softmax_dropout_fwd(
+    is_training, inputs, mask, mask_outer_div, bias, bias_outer_div, dropout_prob
+):
+    """inputs: (n_batch, q, k) contiguous; mask (mb, mq, k) / bias (bb, bq, k)
+    addressed as ((b / outer_div) % nb, qi % sq, :).
+    Returns (out, dropout_mask, softmax_out)."""
+    require_kernels()
+    return _kernels.softmax_dropout_forward(
+        is_training,
+        inputs,
+        mask,
+        int(mask_outer_div),
+        bias,
+        int(bias_outer_div),
+        float(dropout_prob),
+    )
+
+
+def softmax_dropout_bwd(grad_output, softmax_results, dropout_mask, dropout_prob):
+    require_kernels()
+    return _kernels.softmax_dropout_backward(
+        grad_output, softmax_results, dropout_mask, float(dropout_prob)
+    )
+
+
+def layernorm_fwd(input, gamma, beta, eps):
+    require_kernels()
+    return _kernels.layernorm_forward(input, gamma, beta, float(eps))
+
+
+def layernorm_bwd(grad_out, input, mean, invvar, gamma):
+    require_kernels()
+    return _kernels.layernorm_backward(grad_out, input, mean, invvar, gamma)
+
+
+def rmsnorm_fwd(input, gamma, eps):
+    require_kernels()
+    return _kernels.rmsnorm_forward(input, gamma, float(eps))
+
+
+def rmsnorm_bwd(grad_out, input, invvar, gamma):
+    require_kernels()
+    return _kernels.rmsnorm_backward(grad_out, input, invvar, gamma)
+
+
+def fused_adam(
+    p, m, v, g, lr, beta1, beta2, eps, grad_scale, step, bias_correction, weight_decay
+):
+    require_kernels()
+    _kernels.adam(
+        p,
+        m,
+        v,
+        g,
+        float(lr),
+        float(beta1),
+        float(beta2),
+        float(eps),
+        float(grad_scale),
+        int(step),
+        bool(bias_correction),
+        float(weight_decay),
+    )
+
+
+def fused_l2norm(tensors, chunk_size=2048 * 64):
+    require_kernels()
+    return _kernels.multi_tensor_l2norm(int(chunk_size), tensors)
+
+
+def fused_fp32_to_bf16_sr(src_fp32, dst_bf16):
+    require_kernels()
+    _kernels.fp32_to_bf16_sr(src_fp32, dst_bf16)

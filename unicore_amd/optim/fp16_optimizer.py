@@ -1,0 +1,417 @@
+"""Mixed-precision (fp16/bf16) optimizer with flattened params + fp32 master.
+
+Behavioral parity with the reference FP16/BF16 optimizer
+(reference unicore/optim/fp16_optimizer.py): per-(decay-group, dtype)
+flattened low-precision params padded to even numel, a flat fp32 master copy
+per group, grad sync fp16->fp32, a deferred ``_multiply_factor`` that fuses
+unscale+clip into the fused-Adam kernel's scale argument, dynamic loss
+scaling for fp16 (disabled for bf16), optional stochastic-rounding writeback
+fp32->bf16, an ``--allreduce-fp32-grad`` mode that all-reduces the fp32 flat
+grads instead of the low-precision ones, and per-sample gradient clipping.
+
+MI355X-first detail: the flattened low-precision params ALSO get a single
+flat grad tensor per (group, dtype), with every ``param.grad`` a view into
+it. Our FlatDDP engine detects these shared flat grads and buckets directly
+over ranges of them, so backward-overlap all-reduce, grad accumulation and
+the fp16->fp32 sync all operate on a handful of large contiguous tensors —
+no per-parameter copies anywhere in the step.
+"""
+
+import logging
+from typing import List
+
+import torch
+
+from unicore_amd import utils
+
+from .dynamic_loss_scaler import DynamicLossScaler
+from .unicore_optimizer import UnicoreOptimizer
+
+logger = logging.getLogger(__name__)
+
+
+def pad_numel(numel, multiple=2):
+    """Pad to even numel so bf16/fp16 segments stay 4-byte aligned."""
+    return (numel + multiple - 1) // multiple * multiple
+
+
+def separate_decay_params(args, named_params):
+    """Split (name, param) pairs into decay / no-decay groups.
+
+    Exemption rule (same contract as reference fp16_optimizer.py:16-43):
+    ``.bias`` suffix, 1-D params, or any substring from
+    ``--no-weight-decay-names``.
+    """
+    if args.weight_decay <= 0:
+        return [{"params": [p for _, p in named_params if p.requires_grad]}]
+
+    no_wd_names = set(
+        filter(None, getattr(args, "no_weight_decay_names", "").split(","))
+    )
+
+    def exempt(name, p):
+        return (
+            name.endswith(".bias")
+            or p.ndim == 1
+            or any(nd in name for nd in no_wd_names)
+        )
+
+    decay, no_decay = [], []
+    for name, p in named_params:
+        if not p.requires_grad:
+            continue
+        (no_decay if exempt(name, p) else decay).append(p)
+    groups = []
+    if decay:
+        groups.append({"params": decay})
+    if no_decay:
+        groups.append({"params": no_decay, "weight_decay": 0.0})
+    return groups
+
+
+@torch.no_grad()
+def flatten_fp32_master(params, set_to_param=False):
+    """One flat fp32 tensor over *params* with the same (dtype-group, pad)
+    layout as ``_flatten_group``'s fp32 master; optionally re-views each
+    param into it (EMA fast path)."""
+    by_dtype = {}
+    order = []
+    for p in params:
+        if p.dtype not in by_dtype:
+            by_dtype[p.dtype] = []
+            order.append(p.dtype)
+        by_dtype[p.dtype].append(p)
+    total = sum(pad_numel(p.numel()) for p in params)
+    flat = torch.zeros(total, dtype=torch.float32, device=params[0].device)
+    off = 0
+    for dtype in order:
+        for p in by_dtype[dtype]:
+            n = p.numel()
+            flat[off : off + n].copy_(p.data.view(-1).float())
+            if set_to_param:
+                p.data = flat.data[off : off + n].view(*p.shape)
+                p.grad = None
+            off += pad_numel(n)
+    flat = torch.nn.Parameter(flat)
+    if torch.cuda.is_available():
+        torch.cuda.empty_cache()
+    return flat
+
+
+class _FlatGroup:
+    """One decay group: low-precision flats (per dtype) + one fp32 master."""
+
+    __slots__ = ["lp_flats", "lp_members", "fp32_flat", "weight_decay"]
+
+    def __init__(self):
+        self.lp_flats = []  # list of Parameter (one per dtype), each with .grad
+        self.lp_members = []  # list of (param, fp32_offset, numel)
+        self.fp32_flat = None  # Parameter with .grad
+        self.weight_decay = None
+
+
+@torch.no_grad()
+def _flatten_group(params) -> _FlatGroup:
+    g = _FlatGroup()
+    # group by dtype, preserving order
+    by_dtype = {}
+    order = []
+    for p in params:
+        if p.dtype not in by_dtype:
+            by_dtype[p.dtype] = []
+            order.append(p.dtype)
+        by_dtype[p.dtype].append(p)
+
+    total = sum(pad_numel(p.numel()) for p in params)
+    device = params[0].device
+    fp32 = torch.zeros(total, dtype=torch.float32, device=device)
+
+    fp32_offset = 0
+    for dtype in order:
+        dps = by_dtype[dtype]
+        dtotal = sum(pad_numel(p.numel()) for p in dps)
+        flat = torch.zeros(dtotal, dtype=dtype, device=device)
+        flat_grad = torch.zeros(dtotal, dtype=dtype, device=device)
+        off = 0
+        for p in dps:
+            n = p.numel()
+            flat[off : off + n].copy_(p.data.view(-1))
+            fp32[fp32_offset : fp32_offset + n].copy_(p.data.view(-1).float())
+            p.data = flat[off : off + n].view(*p.shape)
+            p.grad = flat_grad[off : off + n].view(*p.shape)
+            g.lp_members.append((p, fp32_offset, n))
+            off += pad_numel(n)
+            fp32_offset += pad_numel(n)
+        flat_param = torch.nn.Parameter(flat)
+        flat_param.grad = flat_grad
+        g.lp_flats.append(flat_param)
+
+    g.fp32_flat = torch.nn.Parameter(fp32)
+    g.fp32_flat.grad = torch.zeros_like(fp32)
+    if torch.cuda.is_available():
+        torch.cuda.empty_cache()
+    return g
+
+
+class FP16Optimizer(UnicoreOptimizer):
+    """Wraps an fp32 optimizer over flat master weights; the model keeps
+    flattened fp16/bf16 params."""
+
+    def __init__(self, args, groups: List[_FlatGroup], fp32_optimizer, scaler):
+        super().__init__(args)
+        self.groups = groups
+        self.fp32_optimizer = fp32_optimizer
+        self.scaler = scaler
+        self._multiply_factor = (
+            1.0 / float(scaler.loss_scale) if scaler is not None else 1.0
+        )
+        self._needs_sync = False
+        self.bf16_sr = getattr(args, "bf16_sr", False)
+        self._per_sample_pending = False
+
+    @classmethod
+    def build_optimizer(cls, args, named_params, **kwargs):
+        """named_params: iterable of (name, param) from the (cast) model."""
+        from unicore_amd import optim
+
+        param_groups = separate_decay_params(args, list(named_params))
+        groups = []
+        fp32_param_groups = []
+        for pg in param_groups:
+            g = _flatten_group(pg["params"])
+            g.weight_decay = pg.get("weight_decay", None)
+            groups.append(g)
+            fg = {"params": [g.fp32_flat]}
+            if "weight_decay" in pg:
+                fg["weight_decay"] = pg["weight_decay"]
+            fp32_param_groups.append(fg)
+
+        fp32_optimizer = optim.build_raw_optimizer(args, fp32_param_groups)
+
+        if getattr(args, "bf16", False):
+            # bf16 has fp32's exponent range: no loss scaling
+            scaler = None
+        else:
+            if args.fp16_scale_window is None:
+                if len(args.update_freq) > 1:
+                    raise ValueError(
+                        "--fp16-scale-window must be given explicitly when using a "
+                        "custom --update-freq schedule"
+                    )
+                data_parallel_size = int(
+                    args.distributed_world_size / args.model_parallel_size
+                    if getattr(args, "model_parallel_size", 1) > 1
+                    else args.distributed_world_size
+                )
+                scale_window = int(
+                    2**14 / data_parallel_size / args.update_freq[0]
+                )
+            else:
+                scale_window = args.fp16_scale_window
+            scaler = DynamicLossScaler(
+                init_scale=args.fp16_init_scale,
+                scale_window=scale_window,
+                tolerance=args.fp16_scale_tolerance,
+                threshold=getattr(args, "threshold_loss_scale", None),
+                min_loss_scale=args.min_loss_scale,
+            )
+        return cls(args, groups, fp32_optimizer, scaler)
+
+    # ------------------------------------------------------------------
+    # views used by EMA / DDP
+    # ------------------------------------------------------------------
+
+    @property
+    def fp32_params(self):
+        """Flat fp32 master params, one per decay group (EMA fast path)."""
+        return [g.fp32_flat for g in self.groups]
+
+    def fp32_view_of(self, p):
+        """The fp32 master segment backing low-precision param *p*."""
+        for g in self.groups:
+            for q, off, n in g.lp_members:
+                if q is p:
+                    return g.fp32_flat.data[off : off + n].view(*p.shape)
+        raise KeyError("param not managed by this optimizer")
+
+    # ------------------------------------------------------------------
+    # UnicoreOptimizer interface
+    # ------------------------------------------------------------------
+
+    @property
+    def optimizer(self):
+        return self.fp32_optimizer.optimizer
+
+    @optimizer.setter
+    def optimizer(self, optimizer):
+        self.fp32_optimizer.optimizer = optimizer
+
+    @property
+    def optimizer_config(self):
+        return self.fp32_optimizer.optimizer_config
+
+    @property
+    def params(self):
+        # expose the low-precision flats (what backward writes into)
+        for g in self.groups:
+            for f in g.lp_flats:
+                yield f
+
+    def state_dict(self):
+        state_dict = self.fp32_optimizer.state_dict()
+        if self.scaler is not None:
+            state_dict["loss_scale"] = self.scaler.loss_scale
+        return state_dict
+
+    def load_state_dict(self, state_dict, optimizer_overrides=None):
+        if "loss_scale" in state_dict and self.scaler is not None:
+            self.scaler.loss_scale = state_dict["loss_scale"]
+        self.fp32_optimizer.load_state_dict(state_dict, optimizer_overrides)
+        # master weights may have been restored: push them into the model
+        self._sync_fp32_params_to_lp()
+
+    def backward(self, loss):
+        if self.scaler is not None:
+            loss = self.scaler.scale(loss)
+        loss.backward()
+        self._needs_sync = True
+
+    @torch.no_grad()
+    def _sync_lp_grads_to_fp32(self):
+        if not self._needs_sync:
+            return
+        for g in self.groups:
+            off = 0
+            for f in g.lp_flats:
+                n = f.grad.numel()
+                g.fp32_flat.grad[off : off + n].copy_(f.grad)
+                off += n
+        self._needs_sync = False
+
+    @torch.no_grad()
+    def _accumulate_lp_grads_to_fp32(self, mul):
+        """Per-sample clipping path: fp32_grad += mul * lp_grad; zero lp."""
+        for g in self.groups:
+            off = 0
+            for f in g.lp_flats:
+                n = f.grad.numel()
+                g.fp32_flat.grad[off : off + n].add_(f.grad.float(), alpha=float(mul))
+                f.grad.zero_()
+                off += n
+        self._needs_sync = False
+        self._per_sample_pending = True
+
+    @torch.no_grad()
+    def _sync_fp32_params_to_lp(self):
+        for g in self.groups:
+            off = 0
+            for f in g.lp_flats:
+                n = f.numel()
+                master = g.fp32_flat.data[off : off + n]
+                if self.bf16_sr and f.dtype == torch.bfloat16:
+                    utils.fp32_to_bf16_sr(master, f.data)
+                else:
+                    f.data.copy_(master)
+                off += n
+
+    def multiply_grads(self, c):
+        """Deferred: folded into _multiply_factor (applied at step/clip)."""
+        if self._needs_sync or not self._per_sample_pending:
+            self._multiply_factor *= c
+        else:
+            # grads were already accumulated into fp32 (per-sample path)
+            self.fp32_optimizer.multiply_grads(c)
+
+    def per_sample_clip_grad_norm(self, max_norm, aggregate_norm_fn=None):
+        """Clip the current (single-sample) lp grads, then accumulate them
+        into the fp32 grads and clear the lp grads."""
+        if max_norm <= 0.0:
+            return 0.0
+        lp_flat_params = [f for g in self.groups for f in g.lp_flats]
+        grad_norm = self._multiply_factor * utils.clip_grad_norm_(
+            lp_flat_params, 0, aggregate_norm_fn
+        )
+        if grad_norm > max_norm > 0.0:
+            clip_coef = max_norm / (grad_norm + 1e-6)
+        else:
+            clip_coef = 1.0
+        self._accumulate_lp_grads_to_fp32(mul=clip_coef * self._multiply_factor)
+        return grad_norm
+
+    def clip_grad_norm(self, max_norm, aggregate_norm_fn=None):
+        """Clips gradient norm and updates dynamic loss scaler."""
+        self._sync_lp_grads_to_fp32()
+        grad_norm = self._multiply_factor * utils.clip_grad_norm_(
+            self.fp32_params_with_grads(), 0, aggregate_norm_fn
+        )
+
+        if self.scaler is not None:
+            if grad_norm > max_norm > 0.0:
+                self._multiply_factor *= max_norm / grad_norm
+            self.scaler.check_overflow(grad_norm)
+        elif max_norm > 0.0:
+            clip_coef = (max_norm / (grad_norm + 1e-6)).clamp_(max=1)
+            self._multiply_factor *= clip_coef
+
+        return grad_norm
+
+    def fp32_params_with_grads(self):
+        return [g.fp32_flat for g in self.groups]
+
+    def all_reduce_grads(self, module):
+        if getattr(self.args, "allreduce_fp32_grad", False):
+            # reduce the fp32 master grads over RCCL instead of the
+            # low-precision ones (reference fp16_optimizer.py:381-388)
+            import torch.distributed as dist
+
+            from unicore_amd.distributed import utils as dist_utils
+
+            self._sync_lp_grads_to_fp32()
+            if dist.is_initialized():
+                ws = dist_utils.get_data_parallel_world_size()
+                group = dist_utils.get_data_parallel_group()
+                for g in self.groups:
+                    g.fp32_flat.grad.div_(ws)
+                    dist.all_reduce(g.fp32_flat.grad, group=group)
+        else:
+            super().all_reduce_grads(module)
+
+    def step(self, closure=None, groups=None):
+        """Performs a single optimization step."""
+        self._sync_lp_grads_to_fp32()
+        if self.fp32_optimizer.supports_step_with_scale:
+            self.fp32_optimizer.step(closure, scale=(1.0 / self._multiply_factor))
+        else:
+            self._unscale_grads()
+            self.fp32_optimizer.step(closure)
+
+        if self.scaler is not None:
+            self.scaler.update()
+
+        self._sync_fp32_params_to_lp()
+        self._per_sample_pending = False
+
+    def _unscale_grads(self):
+        self._sync_lp_grads_to_fp32()
+        if torch.is_tensor(self._multiply_factor) or self._multiply_factor != 1.0:
+            self.fp32_optimizer.multiply_grads(self._multiply_factor)
+            self._multiply_factor = 1.0
+
+    def zero_grad(self):
+        """Clears the gradients of all optimized parameters."""
+        for g in self.groups:
+            for f in g.lp_flats:
+                f.grad.zero_()
+            g.fp32_flat.grad.zero_()
+        if self.scaler is not None:
+            self._multiply_factor = 1.0 / float(self.scaler.loss_scale)
+        else:
+            self._multiply_factor = 1.0
+        self._needs_sync = False
+        self._per_sample_pending = False
+
+    def get_lr(self):
+        return self.fp32_optimizer.get_lr()
+
+    def set_lr(self, lr):
+        self.fp32_optimizer.set_lr(lr)

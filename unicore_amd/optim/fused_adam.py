@@ -1,0 +1,109 @@
+"""FusedAdam: python wrapper over the gfx950 fused AdamW kernel.
+
+Parity: reference unicore/optim/fused_adam.py:20-143. The kernel fuses grad
+unscaling (``g / grad_scale``), both moment updates, bias correction and the
+decoupled weight-decay parameter update into a single grid-stride pass per
+flat tensor; m/v state is kept in fp32 while p/g may be fp16/bf16/fp32
+(reference csrc/adam/adam_kernel.cu:16-46). With the flattened fp16/bf16
+optimizer there are only ~2 flat tensors per step, so kernel-launch count is
+negligible.
+"""
+
+import math
+
+import torch
+
+
+class FusedAdam(torch.optim.Optimizer):
+    """AdamW with fused HIP kernel.
+
+    Compared to the eager path, ``step`` supports a fused grad scale:
+    ``step(scale=s)`` divides grads by s inside the kernel without a separate
+    sweep over memory.
+    """
+
+    def __init__(
+        self,
+        params,
+        lr=1e-3,
+        bias_correction=True,
+        betas=(0.9, 0.999),
+        eps=1e-8,
+        weight_decay=0.0,
+        amsgrad=False,
+    ):
+        if amsgrad:
+            raise RuntimeError("FusedAdam does not support the AMSGrad variant.")
+        defaults = {
+            "lr": lr,
+            "bias_correction": bias_correction,
+            "betas": betas,
+            "eps": eps,
+            "weight_decay": weight_decay,
+        }
+        super().__init__(params, defaults)
+
+    @property
+    def supports_memory_efficient_fp16(self):
+        return True
+
+    @property
+    def supports_flat_params(self):
+        return True
+
+    @property
+    def supports_step_with_scale(self):
+        return True
+
+    def step(self, closure=None, scale=1.0):
+        """Performs a single optimization step.
+
+        Args:
+            closure (callable, optional): A closure that reevaluates the model
+                and returns the loss.
+            scale (float, optional): factor to divide gradient tensor values
+                by before applying to weights (fused into the kernel).
+        """
+        from unicore_amd import ops
+
+        loss = None
+        if closure is not None:
+            loss = closure()
+
+        for group in self.param_groups:
+            bias_correction = group.get("bias_correction", True)
+
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                grad = p.grad.data
+                if grad.is_sparse:
+                    raise RuntimeError("FusedAdam does not support sparse gradients")
+
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    # moments in fp32 regardless of param dtype
+                    state["exp_avg"] = torch.zeros_like(p.data, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(
+                        p.data, dtype=torch.float32
+                    )
+
+                state["step"] += 1
+                beta1, beta2 = group["betas"]
+                ops.fused_adam(
+                    p.data,
+                    state["exp_avg"],
+                    state["exp_avg_sq"],
+                    grad,
+                    group["lr"],
+                    beta1,
+                    beta2,
+                    group["eps"],
+                    scale,
+                    state["step"],
+                    bias_correction,
+                    group["weight_decay"],
+                )
+
+        return loss

@@ -1,0 +1,40 @@
+"""SGD wrapper (parity: reference unicore/optim/sgd.py:13)."""
+
+import torch.optim
+
+from . import register_optimizer
+from .unicore_optimizer import UnicoreOptimizer
+
+
+@register_optimizer("sgd")
+class SGD(UnicoreOptimizer):
+    def __init__(self, args, params):
+        super().__init__(args)
+        self._optimizer = torch.optim.SGD(params, **self.optimizer_config)
+
+    @classmethod
+    def add_args(cls, parser):
+        """Add optimizer-specific arguments to the parser."""
+        parser.add_argument(
+            "--momentum", default=0.0, type=float, metavar="M", help="momentum factor"
+        )
+        parser.add_argument(
+            "--weight-decay",
+            "--wd",
+            default=0.0,
+            type=float,
+            metavar="WD",
+            help="weight decay",
+        )
+
+    @property
+    def optimizer_config(self):
+        return {
+            "lr": self.args.lr[0] if isinstance(self.args.lr, list) else self.args.lr,
+            "momentum": self.args.momentum,
+            "weight_decay": self.args.weight_decay,
+        }
+
+    @property
+    def supports_flat_params(self):
+        return True

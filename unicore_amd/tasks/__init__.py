@@ -1,0 +1,65 @@
+"""Task registry (parity: reference unicore/tasks/__init__.py:16-61)."""
+
+import argparse
+import importlib
+import os
+
+from .unicore_task import UnicoreTask
+
+# register dataclass
+TASK_REGISTRY = {}
+TASK_CLASS_NAMES = set()
+
+
+def setup_task(args, **kwargs):
+    return TASK_REGISTRY[args.task].setup_task(args, **kwargs)
+
+
+def register_task(name):
+    """
+    New tasks can be added with the :func:`register_task` decorator::
+
+        @register_task('classification')
+        class ClassificationTask(UnicoreTask):
+            (...)
+
+    Args:
+        name (str): the name of the task
+    """
+
+    def register_task_cls(cls):
+        if name in TASK_REGISTRY:
+            raise ValueError("Cannot register duplicate task ({})".format(name))
+        if not issubclass(cls, UnicoreTask):
+            raise ValueError(
+                "Task ({}: {}) must extend UnicoreTask".format(name, cls.__name__)
+            )
+        if cls.__name__ in TASK_CLASS_NAMES:
+            raise ValueError(
+                "Cannot register task with duplicate class name ({})".format(
+                    cls.__name__
+                )
+            )
+        TASK_REGISTRY[name] = cls
+        TASK_CLASS_NAMES.add(cls.__name__)
+        return cls
+
+    return register_task_cls
+
+
+def get_task(name):
+    return TASK_REGISTRY[name]
+
+
+# automatically import any Python files in the tasks/ directory
+tasks_dir = os.path.dirname(__file__)
+for file in sorted(os.listdir(tasks_dir)):
+    path = os.path.join(tasks_dir, file)
+    if (
+        not file.startswith("_")
+        and not file.startswith(".")
+        and (file.endswith(".py") or os.path.isdir(path))
+    ):
+        task_name = file[: file.find(".py")] if file.endswith(".py") else file
+        if task_name != "unicore_task":
+            importlib.import_module("unicore_amd.tasks." + task_name)
