@@ -1,0 +1,49 @@
+// pybind11 bindings for the unicore_amd gfx950 kernel extension.
+// One module (`unicore_amd._kernels`) covers the same API surface as the
+// reference's 8 separate CUDA extensions (reference setup.py:141-387).
+#include <torch/extension.h>
+
+#include <optional>
+#include <vector>
+
+std::vector<at::Tensor> softmax_dropout_forward(bool is_training, at::Tensor input,
+                                                std::optional<at::Tensor> mask,
+                                                int64_t mask_outer_div,
+                                                std::optional<at::Tensor> bias,
+                                                int64_t bias_outer_div,
+                                                double dropout_prob);
+at::Tensor softmax_dropout_backward(at::Tensor grad_output,
+                                    at::Tensor softmax_results,
+                                    at::Tensor dropout_mask, double dropout_prob);
+std::vector<at::Tensor> layernorm_forward(at::Tensor input, at::Tensor gamma,
+                                          at::Tensor beta, double eps);
+std::vector<at::Tensor> layernorm_backward(at::Tensor grad_out, at::Tensor input,
+                                           at::Tensor mean, at::Tensor invvar,
+                                           at::Tensor gamma);
+std::vector<at::Tensor> rmsnorm_forward(at::Tensor input, at::Tensor gamma,
+                                        double eps);
+std::vector<at::Tensor> rmsnorm_backward(at::Tensor grad_out, at::Tensor input,
+                                         at::Tensor invvar, at::Tensor gamma);
+void fused_adam(at::Tensor p, at::Tensor m, at::Tensor v, at::Tensor g, double lr,
+                double beta1, double beta2, double eps, double grad_scale,
+                int64_t step, bool bias_correction, double weight_decay);
+at::Tensor multi_tensor_l2norm(int64_t chunk_size, std::vector<at::Tensor> tensors);
+void fp32_to_bf16_sr(at::Tensor src, at::Tensor dst);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("softmax_dropout_forward", &softmax_dropout_forward,
+        "fused softmax(+mask,+bias)+dropout forward (gfx950)");
+  m.def("softmax_dropout_backward", &softmax_dropout_backward,
+        "fused softmax+dropout backward, in-place on grad");
+  m.def("layernorm_forward", &layernorm_forward, "fused LayerNorm forward");
+  m.def("layernorm_backward", &layernorm_backward,
+        "fused LayerNorm backward -> (dx, dgamma, dbeta)");
+  m.def("rmsnorm_forward", &rmsnorm_forward, "fused RMSNorm forward");
+  m.def("rmsnorm_backward", &rmsnorm_backward,
+        "fused RMSNorm backward -> (dx, dgamma)");
+  m.def("adam", &fused_adam, "fused AdamW step (in-place)");
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm,
+        "global L2 norm over a tensor list");
+  m.def("fp32_to_bf16_sr", &fp32_to_bf16_sr,
+        "stochastic-rounding fp32 -> bf16 copy");
+}

@@ -1,0 +1,155 @@
+// Common device helpers for the unicore_amd gfx950 (CDNA4) kernels.
+//
+// Design notes (MI355X):
+//  - wavefront = 64 lanes; all cross-lane reductions are 6-step __shfl_xor
+//    trees over the full wave.
+//  - every memory-bound kernel loads 16 B/lane (8 bf16/fp16 elems or two
+//    float4 for fp32) — scalar 2-byte loads are ~2x slower on this chip.
+//  - Philox4x32-10 is implemented here directly (no cuRAND/hipRAND device
+//    lib); seed/offset come from PyTorch's CUDA generator so the
+//    seed-determinism contract of the reference framework
+//    (unicore/utils.torch_seed) is preserved at the API level.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+#define UNICORE_WAVE 64
+
+// ---------------------------------------------------------------------------
+// dtype conversion traits
+// ---------------------------------------------------------------------------
+template <typename T>
+struct Cvt;
+
+template <>
+struct Cvt<float> {
+  static __device__ __forceinline__ float to_f(float x) { return x; }
+  static __device__ __forceinline__ float from_f(float x) { return x; }
+};
+
+template <>
+struct Cvt<__half> {
+  static __device__ __forceinline__ float to_f(__half x) { return __half2float(x); }
+  static __device__ __forceinline__ __half from_f(float x) { return __float2half(x); }
+};
+
+template <>
+struct Cvt<__hip_bfloat16> {
+  static __device__ __forceinline__ float to_f(__hip_bfloat16 x) {
+    return __bfloat162float(x);
+  }
+  static __device__ __forceinline__ __hip_bfloat16 from_f(float x) {
+    return __float2bfloat16(x);
+  }
+};
+
+// ---------------------------------------------------------------------------
+// vectorized 8-element load/store (16 B/lane for 2-byte dtypes, 32 B for f32)
+// pointers must be 16-byte aligned (guaranteed when the row length is a
+// multiple of 8 elements and the base comes from the caching allocator).
+// ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ void load8(const T* p, float (&f)[8]) {
+  if constexpr (sizeof(T) == 2) {
+    union {
+      uint4 u;
+      T t[8];
+    } U;
+    U.u = *reinterpret_cast<const uint4*>(p);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = Cvt<T>::to_f(U.t[j]);
+  } else {
+    const float4 a = reinterpret_cast<const float4*>(p)[0];
+    const float4 b = reinterpret_cast<const float4*>(p)[1];
+    f[0] = a.x; f[1] = a.y; f[2] = a.z; f[3] = a.w;
+    f[4] = b.x; f[5] = b.y; f[6] = b.z; f[7] = b.w;
+  }
+}
+
+template <typename T>
+__device__ __forceinline__ void store8(T* p, const float (&f)[8]) {
+  if constexpr (sizeof(T) == 2) {
+    union {
+      uint4 u;
+      T t[8];
+    } U;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) U.t[j] = Cvt<T>::from_f(f[j]);
+    *reinterpret_cast<uint4*>(p) = U.u;
+  } else {
+    float4 a, b;
+    a.x = f[0]; a.y = f[1]; a.z = f[2]; a.w = f[3];
+    b.x = f[4]; b.y = f[5]; b.z = f[6]; b.w = f[7];
+    reinterpret_cast<float4*>(p)[0] = a;
+    reinterpret_cast<float4*>(p)[1] = b;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wave64 reductions
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// Philox4x32-10 counter-based RNG (own implementation; not bitwise-identical
+// to cuRAND but honours PyTorch's (seed, subsequence, offset) contract).
+// ---------------------------------------------------------------------------
+struct Philox4 {
+  uint2 key;
+  uint4 ctr;
+
+  __device__ Philox4(uint64_t seed, uint64_t subseq, uint64_t offset) {
+    key.x = (uint32_t)seed;
+    key.y = (uint32_t)(seed >> 32);
+    ctr.x = (uint32_t)offset;
+    ctr.y = (uint32_t)(offset >> 32);
+    ctr.z = (uint32_t)subseq;
+    ctr.w = (uint32_t)(subseq >> 32);
+  }
+
+  static __device__ __forceinline__ uint32_t mulhilo(uint32_t a, uint32_t b,
+                                                     uint32_t* hi) {
+    const uint64_t p = (uint64_t)a * (uint64_t)b;
+    *hi = (uint32_t)(p >> 32);
+    return (uint32_t)p;
+  }
+
+  // one 4x32 draw; advances the counter for the next call
+  __device__ __forceinline__ uint4 next() {
+    uint4 c = ctr;
+    uint2 k = key;
+#pragma unroll
+    for (int i = 0; i < 10; ++i) {
+      uint32_t hi0, hi1;
+      const uint32_t lo0 = mulhilo(0xD2511F53u, c.x, &hi0);
+      const uint32_t lo1 = mulhilo(0xCD9E8D57u, c.z, &hi1);
+      c.x = hi1 ^ c.y ^ k.x;
+      c.y = lo1;
+      c.z = hi0 ^ c.w ^ k.y;
+      c.w = lo0;
+      k.x += 0x9E3779B9u;
+      k.y += 0xBB67AE85u;
+    }
+    if (++ctr.x == 0) ++ctr.y;
+    return c;
+  }
+};
+
+// grid sizing: cap at 8 blocks per CU (256 CUs) and grid-stride the rest
+static inline int unicore_grid(int64_t want, int cap = 2048) {
+  if (want < 1) return 1;
+  return (int)(want < cap ? want : cap);
+}

@@ -1,0 +1,435 @@
+// Fused softmax(+attention-mask,+pair-bias)+dropout for gfx950 (CDNA4).
+//
+// Functional counterpart of the reference's softmax_dropout extension
+// (reference csrc/softmax_dropout/*), re-derived for 64-lane wavefronts:
+//
+//  * vector path (k % 8 == 0, k <= 4096): one wave64 per row, the whole row
+//    register-resident as NV x 8 floats per lane (16 B/lane loads), two
+//    6-step __shfl_xor reductions (max, sum).  Dropout is fused: in-kernel
+//    Philox4x32-10 keyed by PyTorch's generator (seed, subsequence =
+//    row*64+lane, offset), and the keep-mask is stored as a bitfield —
+//    one uint8 per 8-element vector, written by the owning lane.
+//  * block path (any k, p == 0 on this path — the Python shim routes
+//    wide/odd rows with dropout through torch dropout, reference
+//    unicore/modules/softmax_dropout.py:131-138): 256-thread block per row,
+//    3-pass (max / exp+sum / normalize), no LDS row staging needed.
+//
+// mask/bias broadcast contract (see unicore_amd/modules/softmax_dropout.py):
+// source row = ((b / outer_div) % src_nb) * src_q + (q % src_q).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGeneratorImpl.h>
+
+#include <optional>
+#include <vector>
+
+namespace {
+
+template <typename T, int NV, bool DROP>
+__global__ void softmax_fwd_vec_kernel(
+    T* __restrict__ out, uint8_t* __restrict__ dmask, T* __restrict__ x,
+    const T* __restrict__ amask, int64_t am_nb, int am_q, int64_t am_od,
+    const T* __restrict__ bias, int64_t bs_nb, int bs_q, int64_t bs_od,
+    int64_t n_rows, int q_len, int k, float pinv, uint32_t pthresh,
+    uint64_t seed, uint64_t rng_offset) {
+  const int lane = threadIdx.x;
+  const int wid = threadIdx.y;
+  const int mrow_bytes = k / 8;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.y + wid; row < n_rows;
+       row += (int64_t)gridDim.x * blockDim.y) {
+    T* xrow = x + row * (int64_t)k;
+    const int64_t b = row / q_len;
+    const int qi = (int)(row - b * q_len);
+    const T* mrow = amask ? amask + (((b / am_od) % am_nb) * am_q + (qi % am_q)) * (int64_t)k
+                          : nullptr;
+    const T* brow = bias ? bias + (((b / bs_od) % bs_nb) * bs_q + (qi % bs_q)) * (int64_t)k
+                         : nullptr;
+
+    float vals[NV][8];
+    float mx = -INFINITY;
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        load8(xrow + e0, vals[i]);
+        if (mrow) {
+          float t[8];
+          load8(mrow + e0, t);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[i][j] += t[j];
+        }
+        if (brow) {
+          float t[8];
+          load8(brow + e0, t);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[i][j] += t[j];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) mx = fmaxf(mx, vals[i][j]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vals[i][j] = -INFINITY;
+      }
+    }
+    mx = wave_max(mx);
+    float sum = 0.f;
+#pragma unroll
+    for (int i = 0; i < NV; ++i)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        vals[i][j] = __expf(vals[i][j] - mx);
+        sum += vals[i][j];
+      }
+    sum = wave_sum(sum);
+    const float inv = 1.0f / sum;
+
+    Philox4 ph(seed, (uint64_t)row * 64 + lane, rng_offset);
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        float y[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) y[j] = vals[i][j] * inv;
+        store8(xrow + e0, y);  // pre-dropout softmax, in-place over input
+        if constexpr (DROP) {
+          const uint4 r0 = ph.next();
+          const uint4 r1 = ph.next();
+          const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+          uint8_t bits = 0;
+          float o[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const bool keep = rr[j] >= pthresh;
+            bits |= (uint8_t)(keep ? 1u : 0u) << j;
+            o[j] = keep ? y[j] * pinv : 0.f;
+          }
+          dmask[row * (int64_t)mrow_bytes + lane + i * 64] = bits;
+          store8(out + row * (int64_t)k + e0, o);
+        }
+      }
+    }
+  }
+}
+
+template <typename T, int NV, bool DROP>
+__global__ void softmax_bwd_vec_kernel(T* __restrict__ g, const T* __restrict__ y,
+                                       const uint8_t* __restrict__ dmask,
+                                       int64_t n_rows, int k, float pinv) {
+  const int lane = threadIdx.x;
+  const int wid = threadIdx.y;
+  const int mrow_bytes = k / 8;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.y + wid; row < n_rows;
+       row += (int64_t)gridDim.x * blockDim.y) {
+    T* grow = g + row * (int64_t)k;
+    const T* yrow = y + row * (int64_t)k;
+    float tv[NV][8], yv[NV][8];
+    float s = 0.f;
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        load8(grow + e0, tv[i]);
+        load8(yrow + e0, yv[i]);
+        if constexpr (DROP) {
+          const uint8_t bits = dmask[row * (int64_t)mrow_bytes + lane + i * 64];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tv[i][j] = (bits >> j) & 1 ? tv[i][j] * pinv : 0.f;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s += tv[i][j] * yv[i][j];
+      }
+    }
+    s = wave_sum(s);
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        float dx[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dx[j] = yv[i][j] * (tv[i][j] - s);
+        store8(grow + e0, dx);
+      }
+    }
+  }
+}
+
+__device__ __forceinline__ float block_red_max(float v, float* red) {
+  v = wave_max(v);
+  const int wid = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) red[wid] = v;
+  __syncthreads();
+  v = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  __syncthreads();
+  return v;
+}
+
+__device__ __forceinline__ float block_red_sum(float v, float* red) {
+  v = wave_sum(v);
+  const int wid = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) red[wid] = v;
+  __syncthreads();
+  v = red[0] + red[1] + red[2] + red[3];
+  __syncthreads();
+  return v;
+}
+
+// generic-width row softmax (no dropout on this path), 256 threads per row
+template <typename T>
+__global__ void softmax_fwd_block_kernel(T* __restrict__ x,
+                                         const T* __restrict__ amask, int64_t am_nb,
+                                         int am_q, int64_t am_od,
+                                         const T* __restrict__ bias, int64_t bs_nb,
+                                         int bs_q, int64_t bs_od, int64_t n_rows,
+                                         int q_len, int k) {
+  __shared__ float red[4];
+  const int tid = threadIdx.x;
+  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    T* xrow = x + row * (int64_t)k;
+    const int64_t b = row / q_len;
+    const int qi = (int)(row - b * q_len);
+    const T* mrow = amask ? amask + (((b / am_od) % am_nb) * am_q + (qi % am_q)) * (int64_t)k
+                          : nullptr;
+    const T* brow = bias ? bias + (((b / bs_od) % bs_nb) * bs_q + (qi % bs_q)) * (int64_t)k
+                         : nullptr;
+    float mx = -INFINITY;
+    for (int e = tid; e < k; e += 256) {
+      float v = Cvt<T>::to_f(xrow[e]);
+      if (mrow) v += Cvt<T>::to_f(mrow[e]);
+      if (brow) v += Cvt<T>::to_f(brow[e]);
+      mx = fmaxf(mx, v);
+    }
+    mx = block_red_max(mx, red);
+    float sum = 0.f;
+    for (int e = tid; e < k; e += 256) {
+      float v = Cvt<T>::to_f(xrow[e]);
+      if (mrow) v += Cvt<T>::to_f(mrow[e]);
+      if (brow) v += Cvt<T>::to_f(brow[e]);
+      const float t = __expf(v - mx);
+      xrow[e] = Cvt<T>::from_f(t);
+      sum += t;
+    }
+    sum = block_red_sum(sum, red);
+    const float inv = 1.0f / sum;
+    for (int e = tid; e < k; e += 256)
+      xrow[e] = Cvt<T>::from_f(Cvt<T>::to_f(xrow[e]) * inv);
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void softmax_bwd_block_kernel(T* __restrict__ g, const T* __restrict__ y,
+                                         int64_t n_rows, int k) {
+  __shared__ float red[4];
+  const int tid = threadIdx.x;
+  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    T* grow = g + row * (int64_t)k;
+    const T* yrow = y + row * (int64_t)k;
+    float s = 0.f;
+    for (int e = tid; e < k; e += 256)
+      s += Cvt<T>::to_f(grow[e]) * Cvt<T>::to_f(yrow[e]);
+    s = block_red_sum(s, red);
+    for (int e = tid; e < k; e += 256) {
+      const float yv = Cvt<T>::to_f(yrow[e]);
+      grow[e] = Cvt<T>::from_f(yv * (Cvt<T>::to_f(grow[e]) - s));
+    }
+    __syncthreads();
+  }
+}
+
+#define DISPATCH_FTYPES(st, NAME, ...)                               \
+  switch (st) {                                                      \
+    case at::ScalarType::Float: {                                    \
+      using scalar_t = float;                                        \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::Half: {                                     \
+      using scalar_t = __half;                                       \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::BFloat16: {                                 \
+      using scalar_t = __hip_bfloat16;                               \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    default:                                                         \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", st);          \
+  }
+
+struct SrcDesc {
+  const void* ptr = nullptr;
+  int64_t nb = 1;
+  int q = 1;
+  int64_t od = 1;
+};
+
+SrcDesc describe_src(const std::optional<at::Tensor>& t, int64_t outer_div, int k,
+                     const char* what) {
+  SrcDesc d;
+  if (t.has_value() && t->defined()) {
+    TORCH_CHECK(t->is_cuda() && t->is_contiguous(), what, " must be contiguous CUDA");
+    TORCH_CHECK(t->dim() == 3 && t->size(2) == k, what, " must be (nb, q, k)");
+    d.ptr = t->data_ptr();
+    d.nb = t->size(0);
+    d.q = (int)t->size(1);
+    d.od = outer_div > 0 ? outer_div : 1;
+  }
+  return d;
+}
+
+}  // namespace
+
+std::vector<at::Tensor> softmax_dropout_forward(
+    bool is_training, at::Tensor input, std::optional<at::Tensor> mask,
+    int64_t mask_outer_div, std::optional<at::Tensor> bias, int64_t bias_outer_div,
+    double dropout_prob) {
+  TORCH_CHECK(input.is_cuda() && input.is_contiguous(),
+              "softmax_dropout: input must be contiguous CUDA");
+  TORCH_CHECK(input.dim() == 3, "softmax_dropout: input must be (n, q, k)");
+  const int64_t n_batch = input.size(0);
+  const int q_len = (int)input.size(1);
+  const int k = (int)input.size(2);
+  const int64_t n_rows = n_batch * q_len;
+  const bool drop = is_training && dropout_prob > 0.0;
+  const bool vec_ok = (k % 8 == 0) && k <= 4096;
+  TORCH_CHECK(!drop || vec_ok,
+              "softmax_dropout: fused dropout requires k % 8 == 0 and k <= 4096 "
+              "(the Python shim routes other shapes through torch dropout)");
+
+  const SrcDesc m = describe_src(mask, mask_outer_div, k, "mask");
+  const SrcDesc bsrc = describe_src(bias, bias_outer_div, k, "bias");
+  if (m.ptr) TORCH_CHECK(mask->scalar_type() == input.scalar_type(), "mask dtype mismatch");
+  if (bsrc.ptr) TORCH_CHECK(bias->scalar_type() == input.scalar_type(), "bias dtype mismatch");
+
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  at::Tensor out = input;  // p == 0: softmax written in place, out aliases it
+  at::Tensor dmask;
+  float pinv = 1.f;
+  uint32_t pthresh = 0;
+  uint64_t seed = 0, rng_offset = 0;
+  if (drop) {
+    out = at::empty_like(input);
+    dmask = at::empty({n_batch, q_len, k / 8},
+                      input.options().dtype(at::kByte));
+    const double p = std::min(dropout_prob, 0.999999);
+    pinv = (float)(1.0 / (1.0 - p));
+    pthresh = (uint32_t)std::min<double>(p * 4294967296.0, 4294967295.0);
+    auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+        std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+    const int nv = (k + 511) / 512;
+    at::PhiloxCudaState state;
+    {
+      std::lock_guard<std::mutex> lock(gen->mutex_);
+      state = gen->philox_cuda_state(2 * nv);
+    }
+    seed = state.seed_.val;
+    rng_offset = state.offset_.val;
+  } else {
+    dmask = at::empty({0}, input.options().dtype(at::kByte));
+  }
+
+  if (vec_ok) {
+    const dim3 block(64, 4);
+    const dim3 grid(unicore_grid((n_rows + 3) / 4));
+    DISPATCH_FTYPES(input.scalar_type(), "softmax_dropout_forward", {
+      auto launch = [&](auto nv_tag, auto drop_tag) {
+        constexpr int NV = decltype(nv_tag)::value;
+        constexpr bool DROP = decltype(drop_tag)::value;
+        softmax_fwd_vec_kernel<scalar_t, NV, DROP><<<grid, block, 0, stream>>>(
+            reinterpret_cast<scalar_t*>(out.data_ptr()),
+            drop ? dmask.data_ptr<uint8_t>() : nullptr,
+            reinterpret_cast<scalar_t*>(input.data_ptr()),
+            reinterpret_cast<const scalar_t*>(m.ptr), m.nb, m.q, m.od,
+            reinterpret_cast<const scalar_t*>(bsrc.ptr), bsrc.nb, bsrc.q, bsrc.od,
+            n_rows, q_len, k, pinv, pthresh, seed, rng_offset);
+      };
+      auto pick_nv = [&](auto drop_tag) {
+        if (k <= 512)
+          launch(std::integral_constant<int, 1>{}, drop_tag);
+        else if (k <= 1024)
+          launch(std::integral_constant<int, 2>{}, drop_tag);
+        else if (k <= 2048)
+          launch(std::integral_constant<int, 4>{}, drop_tag);
+        else
+          launch(std::integral_constant<int, 8>{}, drop_tag);
+      };
+      if (drop)
+        pick_nv(std::true_type{});
+      else
+        pick_nv(std::false_type{});
+    });
+  } else {
+    const dim3 block(256);
+    const dim3 grid(unicore_grid(n_rows));
+    DISPATCH_FTYPES(input.scalar_type(), "softmax_dropout_forward", {
+      softmax_fwd_block_kernel<scalar_t><<<grid, block, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(input.data_ptr()),
+          reinterpret_cast<const scalar_t*>(m.ptr), m.nb, m.q, m.od,
+          reinterpret_cast<const scalar_t*>(bsrc.ptr), bsrc.nb, bsrc.q, bsrc.od,
+          n_rows, q_len, k);
+    });
+  }
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  // (out, dropout_mask, softmax_results); softmax is in-place over the input
+  return {out, dmask, input};
+}
+
+at::Tensor softmax_dropout_backward(at::Tensor grad_output, at::Tensor softmax_results,
+                                    at::Tensor dropout_mask, double dropout_prob) {
+  TORCH_CHECK(grad_output.is_cuda() && grad_output.is_contiguous(),
+              "softmax_dropout_backward: grad must be contiguous CUDA");
+  TORCH_CHECK(grad_output.sizes() == softmax_results.sizes(), "shape mismatch");
+  TORCH_CHECK(grad_output.dim() == 3, "grad must be (n, q, k)");
+  const int64_t n_rows = grad_output.size(0) * grad_output.size(1);
+  const int k = (int)grad_output.size(2);
+  const bool drop = dropout_mask.defined() && dropout_mask.numel() > 0;
+  const bool vec_ok = (k % 8 == 0) && k <= 4096;
+  TORCH_CHECK(!drop || vec_ok, "softmax_dropout_backward: bad dropout shape");
+  const float pinv = drop ? (float)(1.0 / (1.0 - std::min(dropout_prob, 0.999999))) : 1.f;
+
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (vec_ok) {
+    const dim3 block(64, 4);
+    const dim3 grid(unicore_grid((n_rows + 3) / 4));
+    DISPATCH_FTYPES(grad_output.scalar_type(), "softmax_dropout_backward", {
+      auto launch = [&](auto nv_tag, auto drop_tag) {
+        constexpr int NV = decltype(nv_tag)::value;
+        constexpr bool DROP = decltype(drop_tag)::value;
+        softmax_bwd_vec_kernel<scalar_t, NV, DROP><<<grid, block, 0, stream>>>(
+            reinterpret_cast<scalar_t*>(grad_output.data_ptr()),
+            reinterpret_cast<const scalar_t*>(softmax_results.data_ptr()),
+            drop ? dropout_mask.data_ptr<uint8_t>() : nullptr, n_rows, k, pinv);
+      };
+      auto pick_nv = [&](auto drop_tag) {
+        if (k <= 512)
+          launch(std::integral_constant<int, 1>{}, drop_tag);
+        else if (k <= 1024)
+          launch(std::integral_constant<int, 2>{}, drop_tag);
+        else if (k <= 2048)
+          launch(std::integral_constant<int, 4>{}, drop_tag);
+        else
+          launch(std::integral_constant<int, 8>{}, drop_tag);
+      };
+      if (drop)
+        pick_nv(std::true_type{});
+      else
+        pick_nv(std::false_type{});
+    });
+  } else {
+    const dim3 block(256);
+    const dim3 grid(unicore_grid(n_rows));
+    DISPATCH_FTYPES(grad_output.scalar_type(), "softmax_dropout_backward", {
+      softmax_bwd_block_kernel<scalar_t><<<grid, block, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(grad_output.data_ptr()),
+          reinterpret_cast<const scalar_t*>(softmax_results.data_ptr()), n_rows, k);
+    });
+  }
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return grad_output;
+}
