@@ -1,0 +1,126 @@
+"""Module-level CPU tests: eager softmax_dropout (the numerics oracle),
+broadcast descriptor logic, transformer encoder/decoder shapes,
+LayerNorm/RMSNorm eager paths, metrics/meters.
+"""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+from unicore_amd.logging import meters, metrics
+from unicore_amd.modules import (
+    LayerNorm,
+    RMSNorm,
+    SelfMultiheadAttention,
+    TransformerEncoder,
+    softmax_dropout,
+)
+from unicore_amd.modules.softmax_dropout import _broadcast_descr
+
+
+def test_broadcast_descr():
+    # full match, no broadcast
+    assert _broadcast_descr((4, 8), (4, 8)) == (32, 1)
+    # leading-1 broadcast
+    assert _broadcast_descr((1, 8), (4, 8)) == (8, 1)
+    # trailing block broadcast: (4, 1) over (4, 8) is NOT a contiguous block
+    assert _broadcast_descr((4, 1), (4, 8)) is None or _broadcast_descr((4, 1), (4, 8)) == (4, 8)
+    # scalar
+    assert _broadcast_descr((), (4, 8)) == (1, 1)
+    # inner contiguous block
+    descr = _broadcast_descr((8,), (4, 8))
+    assert descr == (8, 1)
+
+
+def test_softmax_dropout_eager_matches_reference_math():
+    x = torch.randn(2, 3, 4, 5)
+    mask = torch.randn(2, 1, 1, 5)
+    bias = torch.randn(1, 3, 4, 5)
+    out = softmax_dropout(x, 0.0, is_training=False, mask=mask, bias=bias)
+    ref = F.softmax(x + mask + bias, dim=-1)
+    assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_softmax_dropout_training_cpu_drops():
+    torch.manual_seed(0)
+    x = torch.randn(64, 8, 16)
+    out = softmax_dropout(x, 0.5, is_training=True)
+    assert (out == 0).float().mean() > 0.2
+
+
+def test_layernorm_rmsnorm_eager():
+    ln = LayerNorm(32)
+    x = torch.randn(4, 32)
+    assert torch.allclose(ln(x), F.layer_norm(x, (32,), ln.weight, ln.bias, ln.eps))
+    rn = RMSNorm(32)
+    y = rn(x)
+    ref = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + rn.eps)
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_transformer_encoder_forward():
+    enc = TransformerEncoder(
+        encoder_layers=2,
+        embed_dim=64,
+        ffn_embed_dim=128,
+        attention_heads=4,
+        emb_dropout=0.0,
+        dropout=0.0,
+        attention_dropout=0.0,
+        activation_dropout=0.0,
+        max_seq_len=32,
+    )
+    x = torch.randn(3, 16, 64)
+    mask = torch.zeros(3, 16)
+    out = enc(x, padding_mask=mask)
+    assert out.shape == (3, 16, 64)
+
+
+def test_self_attention_bias_path():
+    attn = SelfMultiheadAttention(64, 4, dropout=0.0)
+    x = torch.randn(2, 16, 64)  # (batch, seq, dim)
+    bias = torch.zeros(2, 4, 16, 16)
+    out = attn(x, attn_bias=bias)
+    out2 = attn(x)
+    assert out.shape == x.shape
+    assert torch.allclose(out, out2, atol=1e-6)
+
+
+def test_metrics_aggregation():
+    with metrics.aggregate(new_root=True) as agg:
+        metrics.log_scalar("loss", 2.0, weight=1)
+        metrics.log_scalar("loss", 4.0, weight=3)
+        vals = agg.get_smoothed_values()
+    assert vals["loss"] == pytest.approx(3.5)
+
+
+def test_metrics_nested_contexts():
+    with metrics.aggregate(new_root=True) as outer:
+        metrics.log_scalar("x", 1.0)
+        with metrics.aggregate() as inner:
+            metrics.log_scalar("x", 3.0)
+        iv = inner.get_smoothed_values()
+        ov = outer.get_smoothed_values()
+    assert iv["x"] == pytest.approx(3.0)
+    assert ov["x"] == pytest.approx(2.0)
+
+
+def test_meters_state_dict_roundtrip():
+    m = meters.AverageMeter()
+    m.update(1.0)
+    m.update(3.0)
+    st = m.state_dict()
+    m2 = meters.AverageMeter()
+    m2.load_state_dict(st)
+    assert m2.avg == pytest.approx(2.0)
+
+
+def test_stopwatch_and_time_meter():
+    t = meters.TimeMeter()
+    t.update(5)
+    assert t.n == 5
+    sw = meters.StopwatchMeter()
+    sw.start()
+    sw.stop(n=2)
+    assert sw.n == 2

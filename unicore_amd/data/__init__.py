@@ -27,8 +27,20 @@ from .misc_datasets import (
     RawNumpyDataset,
 )
 from . import data_utils, iterators
+from .iterators import (
+    BufferedIterator,
+    CountingIterator,
+    EpochBatchIterator,
+    GroupedIterator,
+    ShardedIterator,
+)
 
 __all__ = [
+    "BufferedIterator",
+    "CountingIterator",
+    "EpochBatchIterator",
+    "GroupedIterator",
+    "ShardedIterator",
     "AppendTokenDataset",
     "BaseWrapperDataset",
     "BertTokenizeDataset",

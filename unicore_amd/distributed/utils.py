@@ -309,7 +309,10 @@ def all_gather_list(data, group=None, max_size=16384):
         all_gather_list._buffer = torch.empty(buffer_size, dtype=torch.uint8)
         if torch.cuda.is_available():
             all_gather_list._buffer = all_gather_list._buffer.cuda()
-        all_gather_list._cpu_buffer = torch.empty(max_size, dtype=torch.uint8).pin_memory()
+        cpu_buf = torch.empty(max_size, dtype=torch.uint8)
+        if torch.cuda.is_available():
+            cpu_buf = cpu_buf.pin_memory()
+        all_gather_list._cpu_buffer = cpu_buf
     buffer = all_gather_list._buffer
     buffer.zero_()
     cpu_buffer = all_gather_list._cpu_buffer
