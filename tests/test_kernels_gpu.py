@@ -1,0 +1,288 @@
+"""GPU kernel parity tests (MI355X): every HIP kernel vs a plain PyTorch
+fp32 eager reference, following the reference's test pattern
+(reference tests/test_softmax.py: dims x dtypes, fwd+bwd, 1e-3 tolerance)
+but with wider coverage — block-kernel paths, odd widths, dropout mask
+consistency, determinism under torch.manual_seed.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available(), reason="needs an MI355X"
+)
+
+TOL = {
+    torch.float32: 1e-5,
+    torch.float16: 1e-3,
+    torch.bfloat16: 8e-3,
+}
+
+DTYPES = [torch.float32, torch.float16, torch.bfloat16]
+DIMS = [64, 128, 256, 512, 1024, 1536, 2048, 4096, 5000]  # 5000 -> block path
+
+
+def _kernels():
+    from unicore_amd import ops
+
+    assert ops.has_kernels(), "HIP extension must be built/loaded on a GPU box"
+    return ops
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("k", DIMS)
+def test_softmax_fwd_bwd_parity(dtype, k):
+    _kernels()
+    from unicore_amd.modules import softmax_dropout
+
+    torch.manual_seed(0)
+    q = 16
+    x = torch.randn(4, 3, q, k, device="cuda", dtype=dtype)
+    mask = torch.randn(4, 1, 1, k, device="cuda", dtype=dtype)
+    bias = torch.randn(1, 3, q, k, device="cuda", dtype=dtype)
+
+    xk = x.clone().requires_grad_(True)
+    bk = bias.clone().requires_grad_(True)
+    out = softmax_dropout(xk, 0.0, is_training=True, mask=mask, bias=bk,
+                          inplace=False)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    xr = x.detach().float().clone().requires_grad_(True)
+    br = bias.detach().float().clone().requires_grad_(True)
+    ref = F.softmax(xr + mask.float() + br, dim=-1)
+    ref.backward(gout.float())
+
+    tol = TOL[dtype]
+    assert (out.float() - ref).abs().max().item() < tol
+    assert (xk.grad.float() - xr.grad).abs().max().item() < tol * 4
+    assert (bk.grad.float() - br.grad).abs().max().item() < tol * 16
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_softmax_broadcast_variants(dtype):
+    """5-D pair-bias shapes (reference tests/test_softmax.py:81,127)."""
+    _kernels()
+    from unicore_amd.modules import softmax_dropout
+
+    torch.manual_seed(1)
+    B, G, H, q, k = 2, 3, 4, 8, 128
+    x = torch.randn(B, G, H, q, k, device="cuda", dtype=dtype)
+    bias = torch.randn(1, 1, H, q, k, device="cuda", dtype=dtype)
+    mask = torch.randn(B, 1, 1, 1, k, device="cuda", dtype=dtype)
+    out = softmax_dropout(x, 0.0, is_training=True, mask=mask, bias=bias,
+                          inplace=False)
+    ref = F.softmax(x.float() + mask.float() + bias.float(), dim=-1)
+    assert (out.float() - ref).abs().max().item() < TOL[dtype]
+
+
+@requires_gpu
+def test_softmax_dropout_mask_statistics_and_determinism():
+    _kernels()
+    from unicore_amd.modules import softmax_dropout
+
+    x = torch.randn(64, 32, 512, device="cuda", dtype=torch.bfloat16)
+    p = 0.3
+    torch.manual_seed(123)
+    out1 = softmax_dropout(x, p, is_training=True, inplace=False)
+    torch.manual_seed(123)
+    out2 = softmax_dropout(x, p, is_training=True, inplace=False)
+    assert torch.equal(out1, out2), "dropout must be seed-deterministic"
+    zfrac = (out1 == 0).float().mean().item()
+    assert abs(zfrac - p) < 0.02, zfrac
+    # kept values are scaled by 1/(1-p)
+    ref = F.softmax(x.float(), dim=-1)
+    kept = out1 != 0
+    ratio = (out1.float()[kept] / ref[kept]).mean().item()
+    assert abs(ratio - 1.0 / (1.0 - p)) < 0.02
+
+    # different seed -> different mask
+    torch.manual_seed(124)
+    out3 = softmax_dropout(x, p, is_training=True, inplace=False)
+    assert not torch.equal(out1, out3)
+
+
+@requires_gpu
+def test_softmax_dropout_backward_with_mask():
+    """Backward through the fused dropout must equal eager math computed
+    with the SAME mask (recovered from the forward output)."""
+    _kernels()
+    from unicore_amd.modules import softmax_dropout
+
+    torch.manual_seed(5)
+    p = 0.25
+    x = torch.randn(8, 16, 256, device="cuda", dtype=torch.float32,
+                    requires_grad=True)
+    out = softmax_dropout(x, p, is_training=True, inplace=False)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    y = F.softmax(x.detach(), dim=-1)
+    keep = (out.detach() != 0).float() / (1.0 - p)
+    xr = x.detach().clone().requires_grad_(True)
+    yr = F.softmax(xr, dim=-1)
+    (yr * keep).backward(gout)
+    assert (x.grad - xr.grad).abs().max().item() < 1e-4
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("n2", [64, 768, 770, 1024, 3072, 4096, 8192])
+def test_layernorm_parity(dtype, n2):
+    _kernels()
+    from unicore_amd.modules import LayerNorm
+
+    torch.manual_seed(0)
+    ln = LayerNorm(n2).cuda().to(dtype)
+    with torch.no_grad():
+        ln.weight.normal_(1.0, 0.1)
+        ln.bias.normal_(0.0, 0.1)
+    x = torch.randn(512, n2, device="cuda", dtype=dtype, requires_grad=True)
+    out = ln(x)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    xr = x.detach().float().clone().requires_grad_(True)
+    wr = ln.weight.detach().float().clone().requires_grad_(True)
+    br = ln.bias.detach().float().clone().requires_grad_(True)
+    ref = F.layer_norm(xr, (n2,), wr, br, ln.eps)
+    ref.backward(gout.float())
+
+    tol = TOL[dtype]
+    assert (out.float() - ref).abs().max().item() < tol * 10
+    assert (x.grad.float() - xr.grad).abs().max().item() < tol * 20
+    rel = (ln.weight.grad.float() - wr.grad).abs().max() / (wr.grad.abs().max() + 1)
+    assert rel.item() < tol * 20
+    relb = (ln.bias.grad.float() - br.grad).abs().max() / (br.grad.abs().max() + 1)
+    assert relb.item() < tol * 20
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("n2", [64, 768, 770, 4096, 8192])
+def test_rmsnorm_parity(dtype, n2):
+    _kernels()
+    from unicore_amd.modules import RMSNorm
+    from unicore_amd.modules.rms_norm import _eager_rms_norm
+
+    torch.manual_seed(0)
+    rn = RMSNorm(n2).cuda().to(dtype)
+    with torch.no_grad():
+        rn.weight.normal_(1.0, 0.1)
+    x = torch.randn(512, n2, device="cuda", dtype=dtype, requires_grad=True)
+    out = rn(x)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    xr = x.detach().float().clone().requires_grad_(True)
+    wr = rn.weight.detach().float().clone().requires_grad_(True)
+    var = xr.pow(2).mean(-1, keepdim=True)
+    ref = wr * (xr * torch.rsqrt(var + rn.eps))
+    ref.backward(gout.float())
+
+    tol = TOL[dtype]
+    assert (out.float() - ref).abs().max().item() < tol * 10
+    assert (x.grad.float() - xr.grad).abs().max().item() < tol * 20
+    rel = (rn.weight.grad.float() - wr.grad).abs().max() / (wr.grad.abs().max() + 1)
+    assert rel.item() < tol * 20
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+def test_fused_adam_parity(dtype):
+    ops = _kernels()
+    torch.manual_seed(0)
+    n = 4099  # odd size exercises the scalar tail
+    p = torch.randn(n, device="cuda", dtype=dtype)
+    g = torch.randn(n, device="cuda", dtype=dtype)
+    m = torch.randn(n, device="cuda").abs()
+    v = torch.randn(n, device="cuda").abs()
+    p2, g2, m2, v2 = (t.clone().float() for t in (p, g, m, v))
+
+    lr, b1, b2, eps, wd, step, scale = 1e-2, 0.9, 0.98, 1e-6, 0.01, 3, 2.0
+    ops.fused_adam(p, m, v, g, lr, b1, b2, eps, scale, step, True, wd)
+
+    # eager oracle (matches unicore_amd/optim/adam.py semantics + fused scale)
+    import math
+
+    grad = g2 / scale
+    m2.mul_(b1).add_(grad, alpha=1 - b1)
+    v2.mul_(b2).addcmul_(grad, grad, value=1 - b2)
+    denom = v2.sqrt().add_(eps)
+    bc1 = 1 - b1**step
+    bc2 = 1 - b2**step
+    step_size = lr * math.sqrt(bc2) / bc1
+    p2.add_(p2, alpha=-wd * lr)
+    p2.addcdiv_(m2, denom, value=-step_size)
+
+    tol = TOL[dtype]
+    assert (m.float() - m2).abs().max().item() < 1e-5
+    assert (v.float() - v2).abs().max().item() < 1e-5
+    assert (p.float() - p2.to(dtype).float()).abs().max().item() < tol * 4
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+def test_multi_tensor_l2norm(dtype):
+    ops = _kernels()
+    torch.manual_seed(0)
+    tensors = [
+        torch.randn(n, device="cuda", dtype=dtype)
+        for n in (17, 1024, 100000, 3)
+    ]
+    got = ops.fused_l2norm(tensors).item()
+    want = torch.sqrt(sum(t.float().pow(2).sum() for t in tensors)).item()
+    assert abs(got - want) / want < 1e-3
+
+
+@requires_gpu
+def test_fp32_to_bf16_stochastic_rounding():
+    ops = _kernels()
+    torch.manual_seed(0)
+    src = torch.randn(100000, device="cuda") * 0.01
+    dst = torch.empty_like(src, dtype=torch.bfloat16)
+    ops.fused_fp32_to_bf16_sr(src, dst)
+    # every output is the bf16 floor or ceil of the input
+    down = src.bfloat16()
+    eq_floor = dst == down
+    diff = (dst.float() - src).abs()
+    assert eq_floor.float().mean().item() > 0.3
+    # within one bf16 ulp of the source
+    assert (diff <= (src.abs() * 2**-7 + 1e-30)).all()
+    # unbiased in expectation: mean of many roundings approaches the source
+    reps = torch.zeros_like(src)
+    n_rep = 16
+    for i in range(n_rep):
+        torch.manual_seed(1000 + i)
+        d = torch.empty_like(dst)
+        ops.fused_fp32_to_bf16_sr(src, d)
+        reps += d.float()
+    reps /= n_rep
+    bias_sr = (reps - src).mean().abs().item()
+    bias_trunc = (src.bfloat16().float() - src).mean().abs().item()
+    assert bias_sr < 5e-6 or bias_sr < bias_trunc
+
+    # determinism under the same torch seed
+    torch.manual_seed(7)
+    a = torch.empty_like(dst)
+    ops.fused_fp32_to_bf16_sr(src, a)
+    torch.manual_seed(7)
+    b = torch.empty_like(dst)
+    ops.fused_fp32_to_bf16_sr(src, b)
+    assert torch.equal(a, b)
+
+
+@requires_gpu
+def test_gpu_ops_fail_loudly_without_extension(monkeypatch):
+    """On a GPU box, a missing extension must raise, not fall back."""
+    from unicore_amd import ops
+
+    monkeypatch.setattr(ops, "_kernels", None)
+    monkeypatch.setattr(ops, "_import_error", ImportError("simulated"))
+    with pytest.raises(RuntimeError, match="HIP extension"):
+        ops.require_kernels()

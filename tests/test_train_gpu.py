@@ -1,0 +1,128 @@
+"""GPU end-to-end training tests (1 MI355X): bf16 trainer steps through the
+fused HIP path, loss decreases, fp16 loss-scaler path works, and the HIP
+extension is genuinely loaded (no silent eager fallback).
+"""
+
+import os
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available(), reason="needs an MI355X"
+)
+
+
+def _build_trainer(extra_argv=(), seed=1):
+    from unicore_amd import options, tasks
+    from unicore_amd.trainer import Trainer
+
+    argv = [
+        "--task", "bert_synthetic",
+        "--arch", "bert_base",
+        "--loss", "masked_lm",
+        "--optimizer", "adam",
+        "--adam-betas", "(0.9, 0.98)",
+        "--adam-eps", "1e-6",
+        "--clip-norm", "1.0",
+        "--lr-scheduler", "polynomial_decay",
+        "--total-num-update", "1000",
+        "--lr", "3e-4",
+        "--batch-size", "8",
+        "--dataset-size", "64",
+        "--tokens-per-sample", "128",
+        "--vocab-size", "2048",
+        "--encoder-layers", "4",
+        "--encoder-embed-dim", "256",
+        "--encoder-ffn-embed-dim", "512",
+        "--encoder-attention-heads", "4",
+        "--log-format", "none",
+        "--num-workers", "0",
+        "--seed", str(seed),
+    ] + list(extra_argv)
+    parser = options.get_training_parser()
+    args = options.parse_args_and_arch(parser, input_args=argv)
+    args.distributed_world_size = 1
+    args.distributed_rank = 0
+    args.device_id = 0
+    torch.manual_seed(seed)
+    np.random.seed(seed)
+    task = tasks.setup_task(args)
+    task.load_dataset("train")
+    model = task.build_model(args)
+    loss = task.build_loss(args)
+    trainer = Trainer(args, task, model, loss)
+    epoch_itr = trainer.get_train_iterator(epoch=1)
+    trainer.init_total_train_steps(epoch_itr)
+    return trainer, epoch_itr
+
+
+def _run_steps(trainer, epoch_itr, n):
+    itr = epoch_itr.next_epoch_itr(shuffle=False)
+    losses = []
+    batches = list(itr)
+    for i in range(n):
+        logs = trainer.train_step([batches[i % len(batches)]])
+        if logs:
+            losses.append(float(logs[0].get("loss", float("nan"))))
+    torch.cuda.synchronize()
+    return losses
+
+
+@requires_gpu
+def test_extension_is_loaded_and_used():
+    from unicore_amd import ops
+
+    assert ops.has_kernels()
+    # the in-tree .so must be the one loaded
+    import unicore_amd._kernels as K
+
+    assert os.path.dirname(os.path.abspath(K.__file__)).endswith("unicore_amd")
+
+
+@requires_gpu
+def test_bf16_training_loss_decreases():
+    trainer, epoch_itr = _build_trainer(["--bf16"])
+    losses = _run_steps(trainer, epoch_itr, 12)
+    assert len(losses) == 12
+    assert all(np.isfinite(losses))
+    assert np.mean(losses[-3:]) < np.mean(losses[:3]), losses
+
+
+@requires_gpu
+def test_fp16_training_with_loss_scaler():
+    trainer, epoch_itr = _build_trainer(["--fp16"])
+    losses = _run_steps(trainer, epoch_itr, 8)
+    assert all(np.isfinite(losses))
+    assert np.mean(losses[-2:]) < np.mean(losses[:2]) + 0.5
+
+
+@requires_gpu
+def test_bf16_sr_training():
+    trainer, epoch_itr = _build_trainer(["--bf16", "--bf16-sr"])
+    losses = _run_steps(trainer, epoch_itr, 6)
+    assert all(np.isfinite(losses))
+
+
+@requires_gpu
+def test_bf16_seed_determinism():
+    t1, e1 = _build_trainer(["--bf16"], seed=11)
+    l1 = _run_steps(t1, e1, 4)
+    t2, e2 = _build_trainer(["--bf16"], seed=11)
+    l2 = _run_steps(t2, e2, 4)
+    assert l1 == l2, (l1, l2)
+
+
+@requires_gpu
+def test_valid_step_gpu():
+    trainer, epoch_itr = _build_trainer(["--bf16"])
+    _run_steps(trainer, epoch_itr, 2)
+    trainer.task.load_dataset("valid")
+    vitr = trainer.get_valid_iterator("valid").next_epoch_itr(shuffle=False)
+    sample = next(vitr)
+    logs = trainer.valid_step(sample)
+    assert logs and np.isfinite(float(logs[0]["loss"]))
