@@ -50,7 +50,9 @@ def test_softmax_fwd_bwd_parity(dtype, k):
     out = softmax_dropout(xk, 0.0, is_training=True, mask=mask, bias=bk,
                           inplace=False)
     gout = torch.randn_like(out)
-    out.backward(gout)
+    # NOTE: the fused backward is in-place on the incoming grad (reference
+    # contract) -> hand the kernel a clone so `gout` stays pristine
+    out.backward(gout.clone())
 
     xr = x.detach().float().clone().requires_grad_(True)
     br = bias.detach().float().clone().requires_grad_(True)
@@ -120,7 +122,7 @@ def test_softmax_dropout_backward_with_mask():
                     requires_grad=True)
     out = softmax_dropout(x, p, is_training=True, inplace=False)
     gout = torch.randn_like(out)
-    out.backward(gout)
+    out.backward(gout.clone())
 
     y = F.softmax(x.detach(), dim=-1)
     keep = (out.detach() != 0).float() / (1.0 - p)

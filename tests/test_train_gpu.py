@@ -68,7 +68,7 @@ def _run_steps(trainer, epoch_itr, n):
     for i in range(n):
         logs = trainer.train_step([batches[i % len(batches)]])
         if logs:
-            losses.append(float(logs[0].get("loss", float("nan"))))
+            losses.append(float(logs.get("loss", float("nan"))))
     torch.cuda.synchronize()
     return losses
 
@@ -125,4 +125,4 @@ def test_valid_step_gpu():
     vitr = trainer.get_valid_iterator("valid").next_epoch_itr(shuffle=False)
     sample = next(vitr)
     logs = trainer.valid_step(sample)
-    assert logs and np.isfinite(float(logs[0]["loss"]))
+    assert logs and np.isfinite(float(logs["loss"]))
