@@ -29,6 +29,10 @@ void fused_adam(at::Tensor p, at::Tensor m, at::Tensor v, at::Tensor g, double l
                 int64_t step, bool bias_correction, double weight_decay);
 at::Tensor multi_tensor_l2norm(int64_t chunk_size, std::vector<at::Tensor> tensors);
 void fp32_to_bf16_sr(at::Tensor src, at::Tensor dst);
+std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv, int64_t num_heads,
+                                          double scale);
+at::Tensor qkv_split_backward(at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                              int64_t B, int64_t num_heads, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
@@ -46,4 +50,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "global L2 norm over a tensor list");
   m.def("fp32_to_bf16_sr", &fp32_to_bf16_sr,
         "stochastic-rounding fp32 -> bf16 copy");
+  m.def("qkv_split_forward", &qkv_split_forward,
+        "fused QKV head-split + q-scale -> (q, k, v) each (B*H, L, D)");
+  m.def("qkv_split_backward", &qkv_split_backward,
+        "fused QKV head-split backward -> dqkv (B, L, 3E)");
 }

@@ -26,6 +26,7 @@ HIP_SOURCES = [
     "csrc/adam.hip",
     "csrc/multi_tensor.hip",
     "csrc/rounding.hip",
+    "csrc/qkv.hip",
 ]
 
 setup(

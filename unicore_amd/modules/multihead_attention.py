@@ -15,6 +15,32 @@ from torch import Tensor, nn
 from .softmax_dropout import softmax_dropout
 
 
+class _QKVSplit(torch.autograd.Function):
+    """Fused head-split: qkv (B, L, 3E) -> q,k,v each (B*H, L, D), with the
+    q-scaling folded in (one HIP permute-copy each way instead of the
+    chunk + 3x transpose-contiguous + scale chain and its backward cat)."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, scale):
+        from unicore_amd import ops
+
+        q, k, v = ops.qkv_split_fwd(qkv, num_heads, scale)
+        ctx.num_heads = num_heads
+        ctx.scale = scale
+        ctx.bsz = qkv.shape[0]
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        from unicore_amd import ops
+
+        dqkv = ops.qkv_split_bwd(
+            dq.contiguous(), dk.contiguous(), dv.contiguous(),
+            ctx.bsz, ctx.num_heads, ctx.scale,
+        )
+        return dqkv, None, None
+
+
 class SelfMultiheadAttention(nn.Module):
     def __init__(
         self,
@@ -48,23 +74,29 @@ class SelfMultiheadAttention(nn.Module):
         bsz, tgt_len, embed_dim = query.size()
         assert embed_dim == self.embed_dim
 
-        q, k, v = self.in_proj(query).chunk(3, dim=-1)
+        qkv = self.in_proj(query)
+        use_fused_split = False
+        if qkv.is_cuda and self.head_dim % 8 == 0:
+            from unicore_amd import ops
 
-        q = (
-            q.view(bsz, tgt_len, self.num_heads, self.head_dim)
-            .transpose(1, 2)
-            .contiguous()
-            .view(bsz * self.num_heads, -1, self.head_dim)
-            * self.scaling
-        )
-        if k is not None:
+            use_fused_split = ops.gpu_kernels_available()
+        if use_fused_split:
+            q, k, v = _QKVSplit.apply(qkv.contiguous(), self.num_heads, self.scaling)
+        else:
+            q, k, v = qkv.chunk(3, dim=-1)
+            q = (
+                q.view(bsz, tgt_len, self.num_heads, self.head_dim)
+                .transpose(1, 2)
+                .contiguous()
+                .view(bsz * self.num_heads, -1, self.head_dim)
+                * self.scaling
+            )
             k = (
                 k.view(bsz, -1, self.num_heads, self.head_dim)
                 .transpose(1, 2)
                 .contiguous()
                 .view(bsz * self.num_heads, -1, self.head_dim)
             )
-        if v is not None:
             v = (
                 v.view(bsz, -1, self.num_heads, self.head_dim)
                 .transpose(1, 2)

@@ -137,3 +137,13 @@ def fused_l2norm(tensors, chunk_size=2048 * 64):
 def fused_fp32_to_bf16_sr(src_fp32, dst_bf16):
     require_kernels()
     _kernels.fp32_to_bf16_sr(src_fp32, dst_bf16)
+
+
+def qkv_split_fwd(qkv, num_heads, scale):
+    require_kernels()
+    return _kernels.qkv_split_forward(qkv, int(num_heads), float(scale))
+
+
+def qkv_split_bwd(dq, dk, dv, bsz, num_heads, scale):
+    require_kernels()
+    return _kernels.qkv_split_backward(dq, dk, dv, int(bsz), int(num_heads), float(scale))
