@@ -13,6 +13,7 @@ from .pad_dataset import (
     LeftPadDataset,
     RightPadDataset,
     RightPadDataset2D,
+    RightPadDatasetCoord,
 )
 from .sort_dataset import SortDataset, EpochShuffleDataset
 from .nested_dictionary_dataset import NestedDictionaryDataset
@@ -62,6 +63,7 @@ __all__ = [
     "RawNumpyDataset",
     "RightPadDataset",
     "RightPadDataset2D",
+    "RightPadDatasetCoord",
     "SortDataset",
     "TokenizeDataset",
     "UnicoreDataset",

@@ -1,0 +1,48 @@
+"""Masked-MSA cross entropy for the Evoformer stress config — masked_lm
+semantics over (B, S, L) token grids."""
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from unicore_amd import metrics
+from unicore_amd.losses import UnicoreLoss, register_loss
+
+
+@register_loss("masked_msa")
+class MaskedMSALoss(UnicoreLoss):
+    def __init__(self, task):
+        super().__init__(task)
+        self.padding_idx = task.dictionary.pad()
+
+    def forward(self, model, sample, reduce=True):
+        logits = model(**sample["net_input"])  # (B, S, L, V)
+        target = sample["target"]
+        masked = target.ne(self.padding_idx)
+        sample_size = masked.int().sum()
+        masked = torch.where(masked.any(), masked, masked.new([True]))
+        loss = F.nll_loss(
+            F.log_softmax(logits[masked], dim=-1, dtype=torch.float32),
+            target[masked],
+            ignore_index=self.padding_idx,
+            reduction="sum",
+        )
+        logging_output = {
+            "loss": loss.data,
+            "bsz": target.size(0),
+            "sample_size": sample_size,
+        }
+        return loss, sample_size, logging_output
+
+    @staticmethod
+    def reduce_metrics(logging_outputs, split="valid") -> None:
+        loss_sum = sum(log.get("loss", 0) for log in logging_outputs)
+        sample_size = sum(log.get("sample_size", 0) for log in logging_outputs)
+        metrics.log_scalar(
+            "loss", loss_sum / sample_size / math.log(2), sample_size, round=3
+        )
+
+    @staticmethod
+    def logging_outputs_can_be_summed(is_train) -> bool:
+        return True

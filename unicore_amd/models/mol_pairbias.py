@@ -1,0 +1,158 @@
+"""Uni-Mol-style 3D molecular transformer with gaussian pair-bias attention.
+
+BASELINE.json stress config 4: every attention layer carries an additive
+pair bias (B, H, L, L) derived from interatomic distances through a gaussian
+basis — exercising the fused softmax_dropout bias path and its broadcast
+bias-gradient reduction — and the trunk normalization is RMSNorm
+(exercising the fused RMSNorm fwd/bwd kernels).
+
+This is an original compact model in the spirit of Uni-Mol (which is built
+ON the reference framework, not part of it); it demonstrates the framework's
+model-extension API: model registry + add_args + arch registry
+(reference unicore/models/__init__.py:17-102 pattern).
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from unicore_amd import utils
+from unicore_amd.models import (
+    BaseUnicoreModel,
+    register_model,
+    register_model_architecture,
+)
+from unicore_amd.modules import RMSNorm, SelfMultiheadAttention, init_bert_params
+
+
+class GaussianPairBias(nn.Module):
+    """Pairwise distances -> K gaussian basis functions -> per-head bias."""
+
+    def __init__(self, n_kernels, n_heads, cutoff=10.0):
+        super().__init__()
+        self.n_kernels = n_kernels
+        means = torch.linspace(0.0, cutoff, n_kernels)
+        self.means = nn.Parameter(means)
+        self.stds = nn.Parameter(torch.full((n_kernels,), cutoff / n_kernels))
+        self.out = nn.Linear(n_kernels, n_heads)
+
+    def forward(self, coords, padding_mask=None):
+        # coords: (B, L, 3)
+        dist = torch.cdist(coords.float(), coords.float())  # (B, L, L)
+        x = dist.unsqueeze(-1) - self.means.float().view(1, 1, 1, -1)
+        std = self.stds.float().abs() + 1e-3
+        g = torch.exp(-0.5 * (x / std) ** 2)  # (B, L, L, K)
+        g = g.to(self.out.weight.dtype)
+        bias = self.out(g)  # (B, L, L, H)
+        bias = bias.permute(0, 3, 1, 2).contiguous()  # (B, H, L, L)
+        if padding_mask is not None:
+            # additive -inf on padded keys
+            fill = torch.finfo(bias.dtype).min
+            bias = bias.masked_fill(
+                padding_mask.view(padding_mask.size(0), 1, 1, -1).to(torch.bool),
+                fill,
+            )
+        return bias
+
+
+class PairBiasEncoderLayer(nn.Module):
+    """Pre-RMSNorm transformer layer whose attention takes a pair bias."""
+
+    def __init__(self, embed_dim, ffn_dim, heads, dropout, attention_dropout,
+                 activation_dropout):
+        super().__init__()
+        self.attn = SelfMultiheadAttention(embed_dim, heads,
+                                           dropout=attention_dropout)
+        self.attn_norm = RMSNorm(embed_dim)
+        self.fc1 = nn.Linear(embed_dim, ffn_dim)
+        self.fc2 = nn.Linear(ffn_dim, embed_dim)
+        self.ffn_norm = RMSNorm(embed_dim)
+        self.dropout = dropout
+        self.activation_dropout = activation_dropout
+
+    def forward(self, x, bias, padding_mask):
+        residual = x
+        x = self.attn_norm(x)
+        x = self.attn(x, attn_bias=bias)
+        x = F.dropout(x, p=self.dropout, training=self.training)
+        x = residual + x
+        residual = x
+        x = self.ffn_norm(x)
+        x = F.gelu(self.fc1(x))
+        x = F.dropout(x, p=self.activation_dropout, training=self.training)
+        x = self.fc2(x)
+        x = F.dropout(x, p=self.dropout, training=self.training)
+        return residual + x
+
+
+@register_model("mol_pairbias")
+class MolPairBiasModel(BaseUnicoreModel):
+    @staticmethod
+    def add_args(parser):
+        parser.add_argument("--encoder-layers", type=int, metavar="N")
+        parser.add_argument("--encoder-embed-dim", type=int, metavar="N")
+        parser.add_argument("--encoder-ffn-embed-dim", type=int, metavar="N")
+        parser.add_argument("--encoder-attention-heads", type=int, metavar="N")
+        parser.add_argument("--gaussian-kernels", type=int, metavar="N")
+        parser.add_argument("--dropout", type=float)
+        parser.add_argument("--attention-dropout", type=float)
+        parser.add_argument("--activation-dropout", type=float)
+        parser.add_argument("--max-seq-len", type=int)
+
+    def __init__(self, args, dictionary):
+        super().__init__()
+        base_mol_architecture(args)
+        self.args = args
+        self.padding_idx = dictionary.pad()
+        E = args.encoder_embed_dim
+        self.embed_tokens = nn.Embedding(len(dictionary), E, self.padding_idx)
+        self.pair_bias = GaussianPairBias(
+            args.gaussian_kernels, args.encoder_attention_heads
+        )
+        self.layers = nn.ModuleList(
+            [
+                PairBiasEncoderLayer(
+                    E,
+                    args.encoder_ffn_embed_dim,
+                    args.encoder_attention_heads,
+                    args.dropout,
+                    args.attention_dropout,
+                    args.activation_dropout,
+                )
+                for _ in range(args.encoder_layers)
+            ]
+        )
+        self.final_norm = RMSNorm(E)
+        self.lm_head = nn.Linear(E, len(dictionary), bias=False)
+        self.lm_head.weight = self.embed_tokens.weight  # tied
+        # per-atom 3D coordinate delta head (SE(3)-style denoising target)
+        self.coord_head = nn.Linear(E, 3)
+        self.apply(init_bert_params)
+
+    @classmethod
+    def build_model(cls, args, task):
+        return cls(args, task.dictionary)
+
+    def forward(self, src_tokens, src_coord, **kwargs):
+        padding_mask = src_tokens.eq(self.padding_idx)
+        x = self.embed_tokens(src_tokens)
+        bias = self.pair_bias(src_coord, padding_mask)
+        for layer in self.layers:
+            x = layer(x, bias, padding_mask)
+        x = self.final_norm(x)
+        logits = self.lm_head(x)
+        coord_delta = self.coord_head(x.float())
+        return logits, coord_delta
+
+
+@register_model_architecture("mol_pairbias", "mol_pairbias")
+def base_mol_architecture(args):
+    args.encoder_layers = getattr(args, "encoder_layers", 8)
+    args.encoder_embed_dim = getattr(args, "encoder_embed_dim", 512)
+    args.encoder_ffn_embed_dim = getattr(args, "encoder_ffn_embed_dim", 2048)
+    args.encoder_attention_heads = getattr(args, "encoder_attention_heads", 8)
+    args.gaussian_kernels = getattr(args, "gaussian_kernels", 128)
+    args.dropout = getattr(args, "dropout", 0.1)
+    args.attention_dropout = getattr(args, "attention_dropout", 0.1)
+    args.activation_dropout = getattr(args, "activation_dropout", 0.0)
+    args.max_seq_len = getattr(args, "max_seq_len", 512)

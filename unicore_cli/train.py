@@ -43,6 +43,10 @@ def main(args) -> None:
     ), "Must specify batch size either with --batch-size"
 
     metrics.reset()
+    # clear any best-metric state from a previous in-process run; a resumed
+    # run restores it from the checkpoint's extra_state instead
+    if hasattr(checkpoint_utils.save_checkpoint, "best"):
+        del checkpoint_utils.save_checkpoint.best
 
     np_seed = args.seed
     torch.manual_seed(args.seed)
