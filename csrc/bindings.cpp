@@ -33,6 +33,10 @@ std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv, int64_t num_heads,
                                           double scale);
 at::Tensor qkv_split_backward(at::Tensor dq, at::Tensor dk, at::Tensor dv,
                               int64_t B, int64_t num_heads, double scale);
+std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
+                                             bool is_training);
+at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
+                                 double p);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
@@ -54,4 +58,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused QKV head-split + q-scale -> (q, k, v) each (B*H, L, D)");
   m.def("qkv_split_backward", &qkv_split_backward,
         "fused QKV head-split backward -> dqkv (B, L, 3E)");
+  m.def("gelu_dropout_forward", &gelu_dropout_forward,
+        "fused exact-GELU + bitfield dropout forward");
+  m.def("gelu_dropout_backward", &gelu_dropout_backward,
+        "fused GELU + dropout backward (recomputes gelu grad)");
 }

@@ -1,0 +1,177 @@
+// Fused GELU(+bias-free)+dropout for the FFN inner activation on gfx950.
+//
+// Replaces the torch gelu -> dropout kernel pair (each a full read+write of
+// the (B, L, 4E) inner tensor) with one pass each way.  Exact (erf) GELU to
+// match torch's default; dropout keep-mask stored as a bitfield (one uint8
+// per 8-element vector, same contract as the softmax_dropout kernel);
+// backward recomputes gelu'(x) from the saved input.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGeneratorImpl.h>
+
+#include <vector>
+
+namespace {
+
+__device__ __forceinline__ float gelu_fwd(float x) {
+  return 0.5f * x * (1.0f + erff(x * 0.70710678118654752f));
+}
+
+__device__ __forceinline__ float gelu_grad(float x) {
+  const float cdf = 0.5f * (1.0f + erff(x * 0.70710678118654752f));
+  const float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
+  return cdf + x * pdf;
+}
+
+template <typename T, bool DROP>
+__global__ void gelu_dropout_fwd_kernel(T* __restrict__ out,
+                                        uint8_t* __restrict__ dmask,
+                                        const T* __restrict__ x, int64_t n8,
+                                        float pinv, uint32_t pthresh,
+                                        uint64_t seed, uint64_t offset) {
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  Philox4 ph(seed, (uint64_t)tid, offset);
+  for (int64_t i = tid; i < n8; i += stride) {
+    float f[8];
+    load8(x + i * 8, f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = gelu_fwd(f[j]);
+    if constexpr (DROP) {
+      const uint4 r0 = ph.next();
+      const uint4 r1 = ph.next();
+      const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+      uint8_t bits = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const bool keep = rr[j] >= pthresh;
+        bits |= (uint8_t)(keep ? 1u : 0u) << j;
+        f[j] = keep ? f[j] * pinv : 0.f;
+      }
+      dmask[i] = bits;
+    }
+    store8(out + i * 8, f);
+  }
+}
+
+template <typename T, bool DROP>
+__global__ void gelu_dropout_bwd_kernel(T* __restrict__ dx, const T* __restrict__ g,
+                                        const T* __restrict__ x,
+                                        const uint8_t* __restrict__ dmask,
+                                        int64_t n8, float pinv) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    float gv[8], xv[8];
+    load8(g + i * 8, gv);
+    load8(x + i * 8, xv);
+    uint8_t bits = 0xFF;
+    if constexpr (DROP) bits = dmask[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float t = gv[j];
+      if constexpr (DROP) t = (bits >> j) & 1 ? t * pinv : 0.f;
+      gv[j] = t * gelu_grad(xv[j]);
+    }
+    store8(dx + i * 8, gv);
+  }
+}
+
+#define DISPATCH_FTYPES(st, NAME, ...)                               \
+  switch (st) {                                                      \
+    case at::ScalarType::Float: {                                    \
+      using scalar_t = float;                                        \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::Half: {                                     \
+      using scalar_t = __half;                                       \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::BFloat16: {                                 \
+      using scalar_t = __hip_bfloat16;                               \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    default:                                                         \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", st);          \
+  }
+
+}  // namespace
+
+std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
+                                             bool is_training) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "gelu_dropout: contiguous CUDA");
+  TORCH_CHECK(x.numel() % 8 == 0, "gelu_dropout: numel % 8 == 0");
+  const int64_t n8 = x.numel() / 8;
+  const bool drop = is_training && p > 0.0;
+  auto out = at::empty_like(x);
+  at::Tensor dmask;
+  float pinv = 1.f;
+  uint32_t pthresh = 0;
+  uint64_t seed = 0, offset = 0;
+  if (drop) {
+    dmask = at::empty({n8}, x.options().dtype(at::kByte));
+    const double pc = std::min(p, 0.999999);
+    pinv = (float)(1.0 / (1.0 - pc));
+    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+        std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+    at::PhiloxCudaState state;
+    {
+      std::lock_guard<std::mutex> lock(gen->mutex_);
+      state = gen->philox_cuda_state(4 + n8 / (2048LL * 256) * 2);
+    }
+    seed = state.seed_.val;
+    offset = state.offset_.val;
+  } else {
+    dmask = at::empty({0}, x.options().dtype(at::kByte));
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n8 + 255) / 256);
+  DISPATCH_FTYPES(x.scalar_type(), "gelu_dropout_forward", {
+    if (drop)
+      gelu_dropout_fwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()), dmask.data_ptr<uint8_t>(),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()), n8, pinv, pthresh, seed,
+          offset);
+    else
+      gelu_dropout_fwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()), nullptr,
+          reinterpret_cast<const scalar_t*>(x.data_ptr()), n8, pinv, pthresh, seed,
+          offset);
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return {out, dmask};
+}
+
+at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
+                                 double p) {
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() && x.is_contiguous(),
+              "gelu_dropout_backward: contiguous CUDA");
+  const int64_t n8 = x.numel() / 8;
+  const bool drop = dmask.defined() && dmask.numel() > 0;
+  const float pinv =
+      drop ? (float)(1.0 / (1.0 - std::min(p, 0.999999))) : 1.f;
+  auto dx = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n8 + 255) / 256);
+  DISPATCH_FTYPES(x.scalar_type(), "gelu_dropout_backward", {
+    if (drop)
+      gelu_dropout_bwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(dx.data_ptr()),
+          reinterpret_cast<const scalar_t*>(grad.data_ptr()),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()),
+          dmask.data_ptr<uint8_t>(), n8, pinv);
+    else
+      gelu_dropout_bwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(dx.data_ptr()),
+          reinterpret_cast<const scalar_t*>(grad.data_ptr()),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()), nullptr, n8, pinv);
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return dx;
+}

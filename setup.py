@@ -27,6 +27,7 @@ HIP_SOURCES = [
     "csrc/multi_tensor.hip",
     "csrc/rounding.hip",
     "csrc/qkv.hip",
+    "csrc/gelu_dropout.hip",
 ]
 
 setup(

@@ -353,3 +353,33 @@ def test_attention_module_fused_vs_eager_split():
         os.environ.pop("UNICORE_AMD_ALLOW_EAGER", None)
     assert torch.allclose(out_fused, out_eager, atol=1e-5)
     assert torch.allclose(g_fused, x.grad, atol=1e-5)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("p", [0.0, 0.3])
+def test_gelu_dropout_parity(dtype, p):
+    _kernels()
+    from unicore_amd.modules import gelu_dropout
+
+    torch.manual_seed(3)
+    x = torch.randn(4096, 256, device="cuda", dtype=dtype, requires_grad=True)
+    out = gelu_dropout(x, p, is_training=True)
+    gout = torch.randn_like(out)
+    out.backward(gout.clone())
+
+    xr = x.detach().float().clone().requires_grad_(True)
+    ref = F.gelu(xr)
+    if p > 0:
+        keep = (out.detach() != 0).float() / (1.0 - p)
+        ref = ref * keep
+    ref.backward(gout.float())
+
+    tol = TOL[dtype]
+    if p == 0:
+        assert (out.float() - ref).abs().max().item() < tol * 4
+    assert (x.grad.float() - xr.grad).abs().max().item() < tol * 8
+
+    if p > 0:
+        zfrac = (out == 0).float().mean().item()
+        assert abs(zfrac - p) < 0.05  # gelu(x)==0 only at x==0

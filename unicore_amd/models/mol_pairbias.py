@@ -141,7 +141,7 @@ class MolPairBiasModel(BaseUnicoreModel):
             x = layer(x, bias, padding_mask)
         x = self.final_norm(x)
         logits = self.lm_head(x)
-        coord_delta = self.coord_head(x.float())
+        coord_delta = self.coord_head(x).float()
         return logits, coord_delta
 
 

@@ -11,6 +11,7 @@ from torch import Tensor, nn
 
 from unicore_amd import utils
 
+from .gelu_dropout import gelu_dropout
 from .layer_norm import LayerNorm
 from .multihead_attention import SelfMultiheadAttention
 
@@ -42,6 +43,7 @@ class TransformerEncoderLayer(nn.Module):
         self.dropout = dropout
         self.activation_dropout = activation_dropout
         self.activation_fn = utils.get_activation_fn(activation_fn)
+        self._fuse_gelu = activation_fn == "gelu" 
 
         self.self_attn = SelfMultiheadAttention(
             self.embed_dim,
@@ -86,8 +88,11 @@ class TransformerEncoderLayer(nn.Module):
         if not self.post_ln:
             x = self.final_layer_norm(x)
         x = self.fc1(x)
-        x = self.activation_fn(x)
-        x = F.dropout(x, p=self.activation_dropout, training=self.training)
+        if self._fuse_gelu and x.is_cuda:
+            x = gelu_dropout(x, self.activation_dropout, self.training)
+        else:
+            x = self.activation_fn(x)
+            x = F.dropout(x, p=self.activation_dropout, training=self.training)
         x = self.fc2(x)
         x = F.dropout(x, p=self.dropout, training=self.training)
         x = residual + x

@@ -147,3 +147,13 @@ def qkv_split_fwd(qkv, num_heads, scale):
 def qkv_split_bwd(dq, dk, dv, bsz, num_heads, scale):
     require_kernels()
     return _kernels.qkv_split_backward(dq, dk, dv, int(bsz), int(num_heads), float(scale))
+
+
+def gelu_dropout_fwd(x, p, is_training):
+    require_kernels()
+    return _kernels.gelu_dropout_forward(x, float(p), bool(is_training))
+
+
+def gelu_dropout_bwd(grad, x, dmask, p):
+    require_kernels()
+    return _kernels.gelu_dropout_backward(grad, x, dmask, float(p))
