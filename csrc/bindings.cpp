@@ -37,6 +37,7 @@ std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
                                              bool is_training);
 at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
                                  double p);
+at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
@@ -62,4 +63,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused exact-GELU + bitfield dropout forward");
   m.def("gelu_dropout_backward", &gelu_dropout_backward,
         "fused GELU + dropout backward (recomputes gelu grad)");
+  m.def("mfma_gemm_16x16x32", &mfma_gemm_16x16x32,
+        "one-wave bf16 MFMA probe (fragment-layout unit test)");
 }

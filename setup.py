@@ -28,6 +28,7 @@ HIP_SOURCES = [
     "csrc/rounding.hip",
     "csrc/qkv.hip",
     "csrc/gelu_dropout.hip",
+    "csrc/mfma_probe.hip",
 ]
 
 setup(

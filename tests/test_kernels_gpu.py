@@ -383,3 +383,17 @@ def test_gelu_dropout_parity(dtype, p):
     if p > 0:
         zfrac = (out == 0).float().mean().item()
         assert abs(zfrac - p) < 0.05  # gelu(x)==0 only at x==0
+
+
+@requires_gpu
+def test_mfma_fragment_layout_probe():
+    """Asymmetric-input check of the 16x16x32 bf16 MFMA lane mappings the
+    flash-attention kernels assume (transpose-detecting)."""
+    import unicore_amd._kernels as K
+
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    B = (torch.randn(32, 16) * 0.5).bfloat16().cuda()
+    D = K.mfma_gemm_16x16x32(A, B)
+    ref = A.float() @ B.float()
+    assert (D - ref).abs().max().item() < 2e-2, (D - ref).abs().max()
