@@ -38,6 +38,12 @@ std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
 at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
                                  double p);
 at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
+std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
+                                           std::optional<at::Tensor> bias,
+                                           int64_t bias_outer_div,
+                                           std::optional<at::Tensor> mask,
+                                           int64_t mask_outer_div,
+                                           double dropout_p, bool is_training);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
@@ -65,4 +71,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused GELU + dropout backward (recomputes gelu grad)");
   m.def("mfma_gemm_16x16x32", &mfma_gemm_16x16x32,
         "one-wave bf16 MFMA probe (fragment-layout unit test)");
+  m.def("flash_attn_forward", &flash_attn_forward,
+        "flash attention forward (bf16, D=64) -> (o, lse, seed)");
 }

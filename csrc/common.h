@@ -148,6 +148,25 @@ struct Philox4 {
   }
 };
 
+// bf16 <-> bits helpers (avoid relying on __bfloat16_as_ushort availability)
+__device__ __forceinline__ uint16_t f32_to_bf16_bits(float x) {
+  union {
+    __hip_bfloat16 b;
+    uint16_t u;
+  } U;
+  U.b = __float2bfloat16(x);
+  return U.u;
+}
+
+__device__ __forceinline__ float bf16_bits_to_f32(uint16_t u) {
+  union {
+    __hip_bfloat16 b;
+    uint16_t u16;
+  } U;
+  U.u16 = u;
+  return __bfloat162float(U.b);
+}
+
 // grid sizing: cap at 8 blocks per CU (256 CUs) and grid-stride the rest
 static inline int unicore_grid(int64_t want, int cap = 2048) {
   if (want < 1) return 1;

@@ -1,0 +1,347 @@
+// Flash-style fused attention for gfx950 (CDNA4 MFMA), bf16, head_dim 64.
+//
+// Beyond-reference optimization: the reference materializes the O(L^2)
+// score matrix (bmm -> fused softmax -> bmm, reference
+// unicore/modules/multihead_attention.py:83-105); this kernel computes
+// O = dropout(softmax(Q K^T + bias + mask)) V tile-by-tile with online
+// softmax, never touching HBM with the score matrix.
+//
+// Geometry: one 256-thread block (4 wave64) per 64 query rows of one
+// (batch*head); each wave owns 16 rows.  S tiles are built from
+// v_mfma_f32_16x16x32_bf16 (fragment mappings verified by
+// csrc/mfma_probe.hip / tests/test_kernels_gpu.py):
+//   A (16x32): lane l -> row l&15,  k (l>>4)*8+[0..8)
+//   B (32x16): lane l -> col l&15,  k (l>>4)*8+[0..8)
+//   C (16x16): lane l -> col l&15,  rows (l>>4)*4+[0..4)
+// P is redistributed C-layout -> A-layout through a per-wave LDS tile.
+// Dropout keep bits are a pure function of (seed, bh*L+q, kv):
+//   philox(seed, bh*L+q, kv/4) component kv%4 — recomputable by the
+// backward kernels with no stored mask.
+//
+// Per-row LSE (= m + log l) is saved for the backward recomputation.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGeneratorImpl.h>
+
+#include <optional>
+#include <vector>
+
+namespace {
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+
+constexpr int BM = 64;   // query rows per block
+constexpr int BN = 64;   // kv rows per tile
+constexpr int HD = 64;   // head dim
+
+__device__ __forceinline__ bf16x8 load_frag(const uint16_t* p) {
+  union {
+    uint4 u;
+    bf16x8 v;
+  } U;
+  U.u = *reinterpret_cast<const uint4*>(p);
+  return U.v;
+}
+
+// reduce a per-lane value across the 16 lanes that share a C-tile row
+__device__ __forceinline__ float rowgroup_max(float v) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+__device__ __forceinline__ float rowgroup_sum(float v) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// keep-decision for dropout: element (row_global = bh*L + q, col kv)
+// philox(seed, row_global, kv/4) component kv%4, keep iff r >= pthresh.
+template <bool DROP>
+__device__ __forceinline__ void keep_bits8(uint64_t seed, uint64_t subseq,
+                                           int kv0, uint32_t pthresh,
+                                           bool (&keep)[8]) {
+  if constexpr (!DROP) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) keep[j] = true;
+    return;
+  }
+  Philox4 ph(seed, subseq, (uint64_t)(kv0 >> 2));
+  const uint4 r0 = ph.next();
+  const uint4 r1 = ph.next();
+  const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+#pragma unroll
+  for (int j = 0; j < 8; ++j) keep[j] = rr[j] >= pthresh;
+}
+
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(256) void flash_fwd_kernel(
+    uint16_t* __restrict__ out, float* __restrict__ lse,
+    const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
+    const uint16_t* __restrict__ vp,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  // grid: (L/BM, B*H)
+  const int qt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;   // wave 0..3
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;           // fragment k-group / row group
+  const int lr = lane & 15;           // fragment row (A) / col (B, C)
+
+  const int q0 = qt * BM + wid * 16;  // this wave's first q row
+  const int64_t qbase = (bh * L + q0) * HD;
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_p[4][16][BN];
+
+  // Q A-fragments (row lr, d = lg*8 + ks*32 + [0..8))
+  bf16x8 aq[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+    aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+
+  f32x4 o_acc[4] = {};
+  float m_i[4], l_i[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_i[r] = -INFINITY;
+    l_i[r] = 0.f;
+  }
+
+  // bias/mask source rows for this wave's 4 C rows (rows lg*4 + r)
+  // bias element addressing (same contract as softmax_dropout):
+  //   src_row = ((bh / od) % nb) * src_q + (q % src_q)
+  const uint16_t* bias_rows[4];
+  const uint16_t* mask_row = nullptr;
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + lg * 4 + r;
+      bias_rows[r] =
+          bias + (((bh / bias_od) % bias_nb) * bias_q + (q % bias_q)) * (int64_t)L;
+    }
+  }
+  if (HAS_MASK) {
+    // mask is (nb, 1, L): one row per (bh / od) batch, shared by all q
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+  }
+
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    // ---- S = Q K^T (per wave: 16 x 64 as 4 C tiles) -------------------
+    f32x4 s[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      f32x4 acc = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        // B frag: col = kv0 + cb*16 + lr, k = d = ks*32 + lg*8 + [0..8)
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
+      }
+      s[cb] = acc;
+    }
+    // ---- add bias / mask ---------------------------------------------
+    if (HAS_BIAS || HAS_MASK) {
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float add = 0.f;
+          if (HAS_BIAS)
+            add += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
+          if (HAS_MASK)
+            add += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+          s[cb][r] += add;
+        }
+      }
+    }
+    // ---- online softmax ----------------------------------------------
+    float mnew[4], alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
+      mx = rowgroup_max(mx);
+      mnew[r] = fmaxf(m_i[r], mx);
+      alpha[r] = __expf(m_i[r] - mnew[r]);
+      m_i[r] = mnew[r];
+    }
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        s[cb][r] = __expf(s[cb][r] - mnew[r]);
+        psum[r] += s[cb][r];
+      }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      l_i[r] = l_i[r] * alpha[r] + rowgroup_sum(psum[r]);
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) o_acc[cb][r] *= alpha[r];
+    }
+    // ---- redistribute P (C layout) -> A layout via wave-local LDS ----
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_p[wid][lg * 4 + r][cb * 16 + lr] =
+            f32_to_bf16_bits(s[cb][r]);
+    // wave-local LDS: the compiler's lgkmcnt waits order write->read within
+    // the wave; no cross-wave sharing of lds_p[wid].
+    // ---- O += dropout(P) V -------------------------------------------
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      // A frag of P: row = lr, cols kv = ks2*32 + lg*8 + [0..8)
+      bf16x8 ap = load_frag(&lds_p[wid][lr][ks2 * 32 + lg * 8]);
+      if constexpr (DROP) {
+        bool keep[8];
+        keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
+                         kv0 + ks2 * 32 + lg * 8, pthresh, keep);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float pv = bf16_bits_to_f32((uint16_t)(unsigned short)ap[j]);
+          pv = keep[j] ? pv * pinv : 0.f;
+          ap[j] = (short)f32_to_bf16_bits(pv);
+        }
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        // B frag of V: col d = cb*16 + lr, k = kv = kv0 + ks2*32 + lg*8 + j
+        bf16x8 bv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bv[j] = (short)vp[(bh * L + kv0 + ks2 * 32 + lg * 8 + j) * (int64_t)HD +
+                            cb * 16 + lr];
+        o_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, o_acc[cb], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: normalize, store O (bf16) and LSE (fp32) -------------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float inv = 1.0f / l_i[r];
+    const int q = q0 + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+      out[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(o_acc[cb][r] * inv);
+    if (lr == 0) lse[bh * L + q] = m_i[r] + __logf(l_i[r]);
+  }
+}
+
+struct SrcDesc {
+  const void* ptr = nullptr;
+  int64_t nb = 1;
+  int q = 1;
+  int64_t od = 1;
+};
+
+SrcDesc describe(const std::optional<at::Tensor>& t, int64_t outer_div, int L,
+                 const char* what) {
+  SrcDesc d;
+  if (t.has_value() && t->defined()) {
+    TORCH_CHECK(t->is_cuda() && t->is_contiguous() && t->dim() == 3 &&
+                    t->size(2) == L && t->scalar_type() == at::kBFloat16,
+                what, " must be contiguous bf16 (nb, q, L)");
+    d.ptr = t->data_ptr();
+    d.nb = t->size(0);
+    d.q = (int)t->size(1);
+    d.od = outer_div > 0 ? outer_div : 1;
+  }
+  return d;
+}
+
+}  // namespace
+
+// q, k, v: (B*H, L, 64) contiguous bf16 (q pre-scaled).  Returns (o, lse).
+std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
+                                           std::optional<at::Tensor> bias,
+                                           int64_t bias_outer_div,
+                                           std::optional<at::Tensor> mask,
+                                           int64_t mask_outer_div,
+                                           double dropout_p, bool is_training) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+                  v.is_contiguous(),
+              "flash_attn: q/k/v must be contiguous CUDA");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash_attn: bf16 only");
+  TORCH_CHECK(q.dim() == 3 && q.size(2) == HD, "flash_attn: (BH, L, 64) only");
+  const int64_t BH = q.size(0);
+  const int L = (int)q.size(1);
+  TORCH_CHECK(L % BN == 0, "flash_attn: L must be a multiple of 64");
+  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes(), "shape mismatch");
+
+  const SrcDesc bd = describe(bias, bias_outer_div, L, "bias");
+  const SrcDesc md = describe(mask, mask_outer_div, L, "mask");
+  if (md.ptr) TORCH_CHECK(md.q == 1, "flash_attn: mask must broadcast over q");
+
+  const bool drop = is_training && dropout_p > 0.0;
+  float pinv = 1.f;
+  uint32_t pthresh = 0;
+  uint64_t seed = 0;
+  if (drop) {
+    const double pc = std::min(dropout_p, 0.999999);
+    pinv = (float)(1.0 / (1.0 - pc));
+    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+        std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+    at::PhiloxCudaState state;
+    {
+      std::lock_guard<std::mutex> lock(gen->mutex_);
+      // counters consumed per (row) subsequence: L/4 starting at 0; the
+      // backward recomputes the same stream, so advance by the full row.
+      state = gen->philox_cuda_state((L + 3) / 4 + 4);
+    }
+    seed = state.seed_.val;
+    // NOTE: offset is intentionally NOT used — keep bits must be identical
+    // in forward and backward, and both derive them from (seed, row, kv).
+    // The generator advance still makes the NEXT op see fresh randomness.
+    seed += state.offset_.val * 0x9E3779B97F4A7C15ull;
+  }
+
+  auto o = at::empty_like(q);
+  auto lse_t = at::empty({BH, (int64_t)L}, q.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const dim3 grid(L / BM, BH);
+
+  auto launch = [&](auto has_bias, auto has_mask, auto dropt) {
+    flash_fwd_kernel<decltype(has_bias)::value, decltype(has_mask)::value,
+                     decltype(dropt)::value><<<grid, 256, 0, stream>>>(
+        reinterpret_cast<uint16_t*>(o.data_ptr()), lse_t.data_ptr<float>(),
+        reinterpret_cast<const uint16_t*>(q.data_ptr()),
+        reinterpret_cast<const uint16_t*>(k.data_ptr()),
+        reinterpret_cast<const uint16_t*>(v.data_ptr()),
+        reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
+        reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L, pinv,
+        pthresh, seed);
+  };
+  auto pick = [&](auto has_bias, auto has_mask) {
+    if (drop)
+      launch(has_bias, has_mask, std::true_type{});
+    else
+      launch(has_bias, has_mask, std::false_type{});
+  };
+  if (bd.ptr && md.ptr)
+    pick(std::true_type{}, std::true_type{});
+  else if (bd.ptr)
+    pick(std::true_type{}, std::false_type{});
+  else if (md.ptr)
+    pick(std::false_type{}, std::true_type{});
+  else
+    pick(std::false_type{}, std::false_type{});
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return {o, lse_t,
+          at::scalar_tensor((int64_t)seed, q.options().dtype(at::kLong))};
+}

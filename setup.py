@@ -29,6 +29,7 @@ HIP_SOURCES = [
     "csrc/qkv.hip",
     "csrc/gelu_dropout.hip",
     "csrc/mfma_probe.hip",
+    "csrc/flash_attn.hip",
 ]
 
 setup(

@@ -157,3 +157,11 @@ def gelu_dropout_fwd(x, p, is_training):
 def gelu_dropout_bwd(grad, x, dmask, p):
     require_kernels()
     return _kernels.gelu_dropout_backward(grad, x, dmask, float(p))
+
+
+def flash_attn_fwd(q, k, v, bias, bias_outer_div, mask, mask_outer_div, p, is_training):
+    require_kernels()
+    return _kernels.flash_attn_forward(
+        q, k, v, bias, int(bias_outer_div), mask, int(mask_outer_div),
+        float(p), bool(is_training)
+    )
