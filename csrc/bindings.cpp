@@ -44,6 +44,11 @@ std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tenso
                                            std::optional<at::Tensor> mask,
                                            int64_t mask_outer_div,
                                            double dropout_p, bool is_training);
+std::vector<at::Tensor> flash_attn_backward(
+    at::Tensor d_out, at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+    at::Tensor lse, std::optional<at::Tensor> bias, int64_t bias_outer_div,
+    bool bias_needs_grad, std::optional<at::Tensor> mask, int64_t mask_outer_div,
+    double dropout_p, bool dropped, int64_t seed_in);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
@@ -73,4 +78,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "one-wave bf16 MFMA probe (fragment-layout unit test)");
   m.def("flash_attn_forward", &flash_attn_forward,
         "flash attention forward (bf16, D=64) -> (o, lse, seed)");
+  m.def("flash_attn_backward", &flash_attn_backward,
+        "flash attention backward -> (dq, dk, dv[, dbias])");
 }

@@ -345,3 +345,527 @@ std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tenso
   return {o, lse_t,
           at::scalar_tensor((int64_t)seed, q.options().dtype(at::kLong))};
 }
+
+// ===========================================================================
+// backward
+// ===========================================================================
+//
+// Standard flash recomputation: P = exp(S - lse); dP = dO V^T;
+// dS = P o (dropmask*pinv*dP - Di),  Di = rowsum(dO o O);
+// dQ = dS K;  dK = dS^T Q;  dV = (dropmask*pinv*P)^T dO;
+// dBias = sum over broadcast batches of dS (separate grouped pass).
+// Dropout keep bits are recomputed from (seed, bh*L+q, kv) — no stored mask.
+
+namespace {
+
+// Di = rowsum(dO o O), fp32; one 8-lane group per row
+__global__ void flash_dot_do_o_kernel(float* __restrict__ di,
+                                      const uint16_t* __restrict__ dop,
+                                      const uint16_t* __restrict__ op,
+                                      int64_t n_rows) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int sub = lane >> 3;       // 8 rows per wave
+  const int el = lane & 7;         // 8 elems x 8 bytes each
+  for (int64_t row = ((int64_t)blockIdx.x * 4 + wid) * 8 + sub;
+       row < n_rows; row += (int64_t)gridDim.x * 32) {
+    float a[8], b[8];
+    load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row * HD + el * 8, a);
+    load8(reinterpret_cast<const __hip_bfloat16*>(op) + row * HD + el * 8, b);
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += a[j] * b[j];
+#pragma unroll
+    for (int off = 1; off < 8; off <<= 1) s += __shfl_xor(s, off, 64);
+    if (el == 0) di[row] = s;
+  }
+}
+
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
+    uint16_t* __restrict__ dq, const uint16_t* __restrict__ dop,
+    const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
+    const uint16_t* __restrict__ vp, const float* __restrict__ lse,
+    const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int qt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int q0 = qt * BM + wid * 16;
+  const int64_t qbase = (bh * L + q0) * HD;
+
+  // [wave][0]=P tile, [wave][1]=dS tile (C-layout write, A-layout read)
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+
+  bf16x8 aq[2], ado[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    ado[ks] = load_frag(dop + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+  }
+  float lse_r[4], di_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    lse_r[r] = lse[bh * L + q0 + lg * 4 + r];
+    di_r[r] = di[bh * L + q0 + lg * 4 + r];
+  }
+  const float di_row = di[bh * L + q0 + lr];   // Di for this lane's A row
+  const uint16_t* bias_rows[4];
+  const uint16_t* mask_row = nullptr;
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + lg * 4 + r;
+      bias_rows[r] =
+          bias + (((bh / bias_od) % bias_nb) * bias_q + (q % bias_q)) * (int64_t)L;
+    }
+  }
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+
+  f32x4 dq_acc[4] = {};
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    f32x4 s[4], dp[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      f32x4 acc = {}, accd = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
+        const bf16x8 bvt = load_frag(
+            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bvt, accd, 0, 0, 0);
+      }
+      s[cb] = acc;
+      dp[cb] = accd;
+    }
+    // P = exp(S + bias + mask - lse)
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s[cb][r];
+        if (HAS_BIAS)
+          sv += __bfloat162float(
+              reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
+        if (HAS_MASK)
+          sv += __bfloat162float(
+              reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+        s[cb][r] = __expf(sv - lse_r[r]);
+      }
+    }
+    // redistribute P and dP to A layout
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        lds_t[wid][0][lg * 4 + r][cb * 16 + lr] = f32_to_bf16_bits(s[cb][r]);
+        lds_t[wid][1][lg * 4 + r][cb * 16 + lr] = f32_to_bf16_bits(dp[cb][r]);
+      }
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      const bf16x8 pa = load_frag(&lds_t[wid][0][lr][ks2 * 32 + lg * 8]);
+      const bf16x8 dpa = load_frag(&lds_t[wid][1][lr][ks2 * 32 + lg * 8]);
+      bool keep[8];
+      keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
+                       kv0 + ks2 * 32 + lg * 8, pthresh, keep);
+      bf16x8 dsa;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float pv = bf16_bits_to_f32((uint16_t)(unsigned short)pa[j]);
+        float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
+        if (DROP) dpv = keep[j] ? dpv * pinv : 0.f;
+        dsa[j] = (short)f32_to_bf16_bits(pv * (dpv - di_row));
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        bf16x8 bkf;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bkf[j] = (short)kp[(bh * L + kv0 + ks2 * 32 + lg * 8 + j) * (int64_t)HD +
+                             cb * 16 + lr];
+        dq_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bkf, dq_acc[cb], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q = q0 + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+      dq[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dq_acc[cb][r]);
+  }
+}
+
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
+    uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
+    const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
+    const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int kt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int kv0w = kt * BN + wid * 16;   // this wave's first kv row
+  const int64_t kvbase = (bh * L + kv0w) * HD;
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+
+  bf16x8 ak[2], av[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    ak[ks] = load_frag(kp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    av[ks] = load_frag(vp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+  }
+  const uint16_t* mask_row = nullptr;
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+  float maskv[4];   // additive mask for this wave's kv rows (constant over q)
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    maskv[r] = HAS_MASK
+                   ? __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+                         mask_row)[kv0w + lg * 4 + r])
+                   : 0.f;
+
+  f32x4 dk_acc[4] = {}, dv_acc[4] = {};
+  const int n_tiles = L / BM;
+  for (int tq = 0; tq < n_tiles; ++tq) {
+    const int q0 = tq * BM;
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      f32x4 acc = {}, accd = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bq =
+            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq, acc, 0, 0, 0);
+        const bf16x8 bdo =
+            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo, accd, 0, 0, 0);
+      }
+      st[cq] = acc;
+      dpt[cq] = accd;
+    }
+    // P^T = exp(S^T + bias + mask - lse[q]); dS^T = P^T o (drop(dP^T) - Di[q])
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      const float lse_c = lse[bh * L + qcol];
+      const float di_c = di[bh * L + qcol];
+      const uint16_t* brow =
+          HAS_BIAS ? bias + (((bh / bias_od) % bias_nb) * bias_q +
+                             (qcol % bias_q)) * (int64_t)L
+                   : nullptr;
+      bool keep[4] = {true, true, true, true};
+      if (DROP) {
+        // kv rows of this lane: kv0w + lg*4 + r, all in one philox counter
+        Philox4 ph(seed, (uint64_t)(bh * L + qcol),
+                   (uint64_t)((kv0w + lg * 4) >> 2));
+        const uint4 rr = ph.next();
+        const uint32_t rv[4] = {rr.x, rr.y, rr.z, rr.w};
+#pragma unroll
+        for (int r = 0; r < 4; ++r) keep[r] = rv[r] >= pthresh;
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = st[cq][r];
+        if (HAS_BIAS)
+          sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+              brow)[kv0w + lg * 4 + r]);
+        sv += maskv[r];
+        const float pv = __expf(sv - lse_c);
+        float dpv = dpt[cq][r];
+        if (DROP) dpv = keep[r] ? dpv * pinv : 0.f;
+        st[cq][r] = DROP ? (keep[r] ? pv * pinv : 0.f) : pv;  // dropped P^T for dV
+        dpt[cq][r] = pv * (dpv - di_c);                       // dS^T
+      }
+    }
+    // redistribute P^T(dropped) and dS^T to A layout
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        lds_t[wid][0][lg * 4 + r][cq * 16 + lr] = f32_to_bf16_bits(st[cq][r]);
+        lds_t[wid][1][lg * 4 + r][cq * 16 + lr] = f32_to_bf16_bits(dpt[cq][r]);
+      }
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      const bf16x8 pta = load_frag(&lds_t[wid][0][lr][ks2 * 32 + lg * 8]);
+      const bf16x8 dsta = load_frag(&lds_t[wid][1][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        bf16x8 bdo, bqf;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int64_t qrow = bh * L + q0 + ks2 * 32 + lg * 8 + j;
+          bdo[j] = (short)dop[qrow * HD + cb * 16 + lr];
+          bqf[j] = (short)qp[qrow * HD + cb * 16 + lr];
+        }
+        dv_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta, bdo, dv_acc[cb], 0, 0, 0);
+        dk_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsta, bqf, dk_acc[cb], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kv = kv0w + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      dk[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dk_acc[cb][r]);
+      dv[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dv_acc[cb][r]);
+    }
+  }
+}
+
+// grouped dBias pass: dBias[src] = sum over the bh's mapping to src of dS.
+// grid: (L/BM qtile, nb src rows, BG groups); fp32 partials [BG][nb][L][L].
+template <bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(256) void flash_bwd_dbias_kernel(
+    float* __restrict__ part, const uint16_t* __restrict__ dop,
+    const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
+    const uint16_t* __restrict__ vp, const float* __restrict__ lse,
+    const float* __restrict__ di, const uint16_t* __restrict__ bias,
+    int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, int64_t BH, int bg_count, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int qt = blockIdx.x;
+  const int src = blockIdx.y;      // bias source row block (0..nb)
+  const int bg = blockIdx.z;       // batch group
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int q0 = qt * BM + wid * 16;
+
+  const uint16_t* bias_rows[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q = q0 + lg * 4 + r;
+    bias_rows[r] = bias + ((int64_t)src * bias_q + (q % bias_q)) * (int64_t)L;
+  }
+
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    f32x4 db_acc[4] = {};
+    // loop the batch-heads whose bias source is `src`, strided by group
+    for (int64_t bh = 0; bh < BH; ++bh) {
+      if ((bh / bias_od) % bias_nb != src) continue;
+      // partition the broadcast batches by their OUTER index so the
+      // groups are even (bh and src share low-order structure)
+      if ((bh / (bias_od * bias_nb)) % (int64_t)bg_count != bg) continue;
+      const int64_t qbase = (bh * L + q0) * HD;
+      bf16x8 aq[2], ado[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+        ado[ks] = load_frag(dop + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+      }
+      const uint16_t* mask_row =
+          HAS_MASK ? mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L
+                   : nullptr;
+      f32x4 s[4], dp[4];
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        f32x4 acc = {}, accd = {};
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const bf16x8 bk = load_frag(
+              kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
+          const bf16x8 bvt = load_frag(
+              vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+          accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bvt, accd, 0, 0, 0);
+        }
+        s[cb] = acc;
+        dp[cb] = accd;
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q = q0 + lg * 4 + r;
+          float sv = s[cb][r];
+          sv += __bfloat162float(
+              reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
+          if (HAS_MASK)
+            sv += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+          const float pv = __expf(sv - lse[bh * L + q]);
+          float dpv = dp[cb][r];
+          if (DROP) {
+            Philox4 ph(seed, (uint64_t)(bh * L + q), (uint64_t)(kv >> 2));
+            const uint4 rr = ph.next();
+            const uint32_t rv[4] = {rr.x, rr.y, rr.z, rr.w};
+            dpv = rv[kv & 3] >= pthresh ? dpv * pinv : 0.f;
+          }
+          db_acc[cb][r] += pv * (dpv - di[bh * L + q]);
+        }
+      }
+    }
+    // write this (qtile, kvtile) slice of the partial
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q = q0 + lg * 4 + r;
+        part[(((int64_t)bg * bias_nb + src) * L + q) * L + kv] = db_acc[cb][r];
+      }
+    }
+  }
+}
+
+__global__ void flash_dbias_reduce_kernel(uint16_t* __restrict__ dbias,
+                                          const float* __restrict__ part,
+                                          int bg_count, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int g = 0; g < bg_count; ++g) s += part[(int64_t)g * n + i];
+    dbias[i] = f32_to_bf16_bits(s);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> flash_attn_backward(
+    at::Tensor d_out, at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+    at::Tensor lse, std::optional<at::Tensor> bias, int64_t bias_outer_div,
+    bool bias_needs_grad, std::optional<at::Tensor> mask, int64_t mask_outer_div,
+    double dropout_p, bool dropped, int64_t seed_in) {
+  TORCH_CHECK(d_out.is_cuda() && d_out.is_contiguous() && q.is_contiguous() &&
+                  k.is_contiguous() && v.is_contiguous() && o.is_contiguous(),
+              "flash_attn_backward: tensors must be contiguous CUDA");
+  const int64_t BH = q.size(0);
+  const int L = (int)q.size(1);
+  const SrcDesc bd = describe(bias, bias_outer_div, L, "bias");
+  const SrcDesc md = describe(mask, mask_outer_div, L, "mask");
+
+  const bool drop = dropped && dropout_p > 0.0;
+  float pinv = 1.f;
+  uint32_t pthresh = 0;
+  if (drop) {
+    const double pc = std::min(dropout_p, 0.999999);
+    pinv = (float)(1.0 / (1.0 - pc));
+    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+  }
+  const uint64_t seed = (uint64_t)seed_in;
+
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto di = at::empty({BH, (int64_t)L}, q.options().dtype(at::kFloat));
+  const int64_t n_rows = BH * L;
+  flash_dot_do_o_kernel<<<unicore_grid((n_rows + 31) / 32), 256, 0, stream>>>(
+      di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+      reinterpret_cast<const uint16_t*>(o.data_ptr()), n_rows);
+
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  const dim3 grid(L / BM, BH);
+
+  auto launch_all = [&](auto hb, auto hm, auto dr) {
+    constexpr bool HB = decltype(hb)::value;
+    constexpr bool HM = decltype(hm)::value;
+    constexpr bool DR = decltype(dr)::value;
+    flash_bwd_dq_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
+        reinterpret_cast<uint16_t*>(dq.data_ptr()),
+        reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+        reinterpret_cast<const uint16_t*>(q.data_ptr()),
+        reinterpret_cast<const uint16_t*>(k.data_ptr()),
+        reinterpret_cast<const uint16_t*>(v.data_ptr()), lse.data_ptr<float>(),
+        di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb,
+        bd.q, bd.od, reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q,
+        md.od, L, pinv, pthresh, seed);
+    flash_bwd_dkv_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
+        reinterpret_cast<uint16_t*>(dk.data_ptr()),
+        reinterpret_cast<uint16_t*>(dv.data_ptr()),
+        reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+        reinterpret_cast<const uint16_t*>(q.data_ptr()),
+        reinterpret_cast<const uint16_t*>(k.data_ptr()),
+        reinterpret_cast<const uint16_t*>(v.data_ptr()), lse.data_ptr<float>(),
+        di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb,
+        bd.q, bd.od, reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q,
+        md.od, L, pinv, pthresh, seed);
+  };
+  auto pick = [&](auto hb, auto hm) {
+    if (drop)
+      launch_all(hb, hm, std::true_type{});
+    else
+      launch_all(hb, hm, std::false_type{});
+  };
+  if (bd.ptr && md.ptr)
+    pick(std::true_type{}, std::true_type{});
+  else if (bd.ptr)
+    pick(std::true_type{}, std::false_type{});
+  else if (md.ptr)
+    pick(std::false_type{}, std::true_type{});
+  else
+    pick(std::false_type{}, std::false_type{});
+
+  at::Tensor dbias;
+  if (bd.ptr && bias_needs_grad) {
+    TORCH_CHECK(bd.od == 1,
+                "flash_attn: bias grad supports outer_div == 1 broadcasts");
+    const int64_t group = std::max<int64_t>(BH / bd.nb, 1);
+    const int bg_count = (int)std::min<int64_t>(group, 8);
+    auto part = at::empty({bg_count, bd.nb, (int64_t)L, (int64_t)L},
+                          q.options().dtype(at::kFloat));
+    const dim3 bgrid(L / BM, bd.nb, bg_count);
+    auto launch_db = [&](auto hm, auto dr) {
+      flash_bwd_dbias_kernel<decltype(hm)::value, decltype(dr)::value>
+          <<<bgrid, 256, 0, stream>>>(
+              part.data_ptr<float>(),
+              reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+              reinterpret_cast<const uint16_t*>(q.data_ptr()),
+              reinterpret_cast<const uint16_t*>(k.data_ptr()),
+              reinterpret_cast<const uint16_t*>(v.data_ptr()),
+              lse.data_ptr<float>(), di.data_ptr<float>(),
+              reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
+              reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L,
+              BH, bg_count, pinv, pthresh, seed);
+    };
+    if (md.ptr) {
+      if (drop) launch_db(std::true_type{}, std::true_type{});
+      else launch_db(std::true_type{}, std::false_type{});
+    } else {
+      if (drop) launch_db(std::false_type{}, std::true_type{});
+      else launch_db(std::false_type{}, std::false_type{});
+    }
+    dbias = at::empty({bd.nb, (int64_t)bd.q, (int64_t)L}, q.options());
+    TORCH_CHECK(bd.q == L, "flash_attn: bias grad requires bias_q == L");
+    const int64_t n = (int64_t)bd.nb * L * L;
+    flash_dbias_reduce_kernel<<<unicore_grid((n + 255) / 256), 256, 0, stream>>>(
+        reinterpret_cast<uint16_t*>(dbias.data_ptr()), part.data_ptr<float>(),
+        bg_count, n);
+  }
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  if (dbias.defined()) return {dq, dk, dv, dbias};
+  return {dq, dk, dv};
+}

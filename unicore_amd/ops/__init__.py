@@ -165,3 +165,12 @@ def flash_attn_fwd(q, k, v, bias, bias_outer_div, mask, mask_outer_div, p, is_tr
         q, k, v, bias, int(bias_outer_div), mask, int(mask_outer_div),
         float(p), bool(is_training)
     )
+
+
+def flash_attn_bwd(d_out, q, k, v, o, lse, bias, bias_outer_div, bias_needs_grad,
+                   mask, mask_outer_div, p, dropped, seed):
+    require_kernels()
+    return _kernels.flash_attn_backward(
+        d_out, q, k, v, o, lse, bias, int(bias_outer_div), bool(bias_needs_grad),
+        mask, int(mask_outer_div), float(p), bool(dropped), int(seed)
+    )
