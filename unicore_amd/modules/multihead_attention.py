@@ -12,7 +12,69 @@ from typing import Optional
 import torch
 from torch import Tensor, nn
 
-from .softmax_dropout import softmax_dropout
+from .softmax_dropout import _broadcast_descr, softmax_dropout
+
+
+class _FlashAttn(torch.autograd.Function):
+    """Flash attention (bf16, head_dim 64): O = dropout(softmax(QK^T +
+    bias + mask)) V computed tile-wise by the MFMA kernel — the score
+    matrix never touches HBM.  Dropout keep bits are recomputed from the
+    philox seed in backward (no stored mask)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, bias_od, mask, mask_od, p, training):
+        from unicore_amd import ops
+
+        o, lse, seed = ops.flash_attn_fwd(
+            q, k, v, bias, bias_od, mask, mask_od, p, training
+        )
+        ctx.save_for_backward(q, k, v, o, lse,
+                              bias if bias is not None else q.new_empty(0),
+                              mask if mask is not None else q.new_empty(0))
+        ctx.bias_od = bias_od
+        ctx.mask_od = mask_od
+        ctx.p = p
+        ctx.dropped = training and p > 0
+        ctx.seed = int(seed)
+        return o
+
+    @staticmethod
+    def backward(ctx, d_out):
+        from unicore_amd import ops
+
+        q, k, v, o, lse, bias, mask = ctx.saved_tensors
+        bias_t = bias if bias.numel() else None
+        mask_t = mask if mask.numel() else None
+        want_dbias = bias_t is not None and ctx.needs_input_grad[3]
+        grads = ops.flash_attn_bwd(
+            d_out.contiguous(), q, k, v, o, lse, bias_t, ctx.bias_od,
+            want_dbias, mask_t, ctx.mask_od, ctx.p, ctx.dropped, ctx.seed,
+        )
+        dq, dk, dv = grads[0], grads[1], grads[2]
+        dbias = grads[3] if len(grads) > 3 else None
+        return dq, dk, dv, dbias, None, None, None, None, None
+
+
+def _prep_flash_src(src, batch_dims, k_len, allowed_sq):
+    """Normalize an additive bias/mask to the flash kernel's
+    (src_nb, src_q, k) + outer_div form; (None, 1, False) if the broadcast
+    pattern or shape is unsupported (caller falls back to the materialized
+    path)."""
+    if src is None:
+        return None, 1, True
+    if src.shape[-1] != k_len or src.dim() < 2:
+        return None, 1, False
+    sq = src.shape[-2]
+    if sq not in allowed_sq:
+        return None, 1, False
+    descr = _broadcast_descr(tuple(src.shape[:-2]), batch_dims)
+    if descr is None:
+        return None, 1, False
+    src_nb, outer_div = descr
+    t = src.reshape(src_nb, sq, k_len)
+    if not t.is_contiguous():
+        t = t.contiguous()
+    return t, outer_div, True
 
 
 class _QKVSplit(torch.autograd.Function):
@@ -107,10 +169,6 @@ class SelfMultiheadAttention(nn.Module):
         assert k is not None
         src_len = k.size(1)
 
-        attn_weights = torch.bmm(q, k.transpose(1, 2))
-
-        assert list(attn_weights.size()) == [bsz * self.num_heads, tgt_len, src_len]
-
         mask = None
         if key_padding_mask is not None and key_padding_mask.dim() == 0:
             key_padding_mask = None
@@ -119,27 +177,59 @@ class SelfMultiheadAttention(nn.Module):
             # broadcast over heads + query positions by the fused kernel
             assert key_padding_mask.size(0) == bsz
             assert key_padding_mask.size(-1) == src_len
-            mask = key_padding_mask.view(bsz, 1, 1, src_len).to(attn_weights.dtype)
+            mask = key_padding_mask.view(bsz, 1, 1, src_len).to(q.dtype)
 
-        attn_weights = attn_weights.view(bsz, self.num_heads, tgt_len, src_len)
-        if not return_attn:
-            attn = softmax_dropout(
-                attn_weights,
-                self.dropout,
-                self.training,
-                mask=mask,
-                bias=attn_bias,
+        # flash path: bf16, head_dim 64, L % 64 == 0, kernel-expressible
+        # bias/mask broadcasts — the L x L score matrix never hits HBM
+        o = None
+        if (
+            use_fused_split
+            and not return_attn
+            and q.dtype == torch.bfloat16
+            and self.head_dim == 64
+            and tgt_len == src_len
+            and tgt_len % 64 == 0
+        ):
+            batch_dims = (bsz, self.num_heads)
+            bias_allowed = (
+                {tgt_len}
+                if attn_bias is not None and attn_bias.requires_grad
+                else {1, tgt_len}
             )
-        else:
-            attn_weights = attn_weights + (mask if mask is not None else 0)
-            if attn_bias is not None:
-                attn_weights = attn_weights + attn_bias
-            attn = softmax_dropout(
-                attn_weights, self.dropout, self.training, inplace=False
+            bias_k, bias_od, ok_b = _prep_flash_src(
+                attn_bias, batch_dims, src_len, bias_allowed
             )
+            mask_k, mask_od, ok_m = _prep_flash_src(mask, batch_dims, src_len, {1})
+            if ok_b and ok_m:
+                o = _FlashAttn.apply(
+                    q, k, v, bias_k, bias_od, mask_k, mask_od,
+                    self.dropout, self.training,
+                )
 
-        attn = attn.view(bsz * self.num_heads, tgt_len, src_len)
-        o = torch.bmm(attn, v)
+        if o is None:
+            attn_weights = torch.bmm(q, k.transpose(1, 2))
+            assert list(attn_weights.size()) == [
+                bsz * self.num_heads, tgt_len, src_len,
+            ]
+            attn_weights = attn_weights.view(bsz, self.num_heads, tgt_len, src_len)
+            if not return_attn:
+                attn = softmax_dropout(
+                    attn_weights,
+                    self.dropout,
+                    self.training,
+                    mask=mask,
+                    bias=attn_bias,
+                )
+            else:
+                attn_weights = attn_weights + (mask if mask is not None else 0)
+                if attn_bias is not None:
+                    attn_weights = attn_weights + attn_bias
+                attn = softmax_dropout(
+                    attn_weights, self.dropout, self.training, inplace=False
+                )
+
+            attn = attn.view(bsz * self.num_heads, tgt_len, src_len)
+            o = torch.bmm(attn, v)
         assert list(o.size()) == [bsz * self.num_heads, tgt_len, self.head_dim]
 
         o = (
