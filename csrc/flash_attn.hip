@@ -98,6 +98,10 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   const int64_t qbase = (bh * L + q0) * HD;
 
   __shared__ __attribute__((aligned(16))) uint16_t lds_p[4][16][BN];
+  // V tile transposed [d][kv], kv XOR-swizzled in 8-element blocks keyed on
+  // d&7 so the PV B-fragment reads (different d rows, same kv range) are
+  // bank-conflict-free (guide T2 pattern)
+  __shared__ __attribute__((aligned(16))) uint16_t lds_vt[HD][BN];
 
   // Q A-fragments (row lr, d = lg*8 + ks*32 + [0..8))
   bf16x8 aq[2];
@@ -131,9 +135,31 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
   }
 
+  // staging role: thread handles V row kv = tid/4, d block (tid%4)*16
+  const int st_kv = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
+
   const int n_tiles = L / BN;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
+    // ---- stage V^T tile ----------------------------------------------
+    {
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(vp) +
+                (bh * L + kv0 + st_kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(vp) +
+                (bh * L + kv0 + st_kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_vt[d0][st_kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_vt[d1][st_kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
     // ---- S = Q K^T (per wave: 16 x 64 as 4 C tiles) -------------------
     f32x4 s[4];
 #pragma unroll
@@ -217,16 +243,16 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       }
 #pragma unroll
       for (int cb = 0; cb < 4; ++cb) {
-        // B frag of V: col d = cb*16 + lr, k = kv = kv0 + ks2*32 + lg*8 + j
-        bf16x8 bv;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          bv[j] = (short)vp[(bh * L + kv0 + ks2 * 32 + lg * 8 + j) * (int64_t)HD +
-                            cb * 16 + lr];
+        // B frag of V from the staged transpose: row d = cb*16+lr,
+        // kv block (ks2*32 + lg*8) ^ swizzle — one 16 B LDS read
+        const int d = cb * 16 + lr;
+        const bf16x8 bv =
+            load_frag(&lds_vt[d][(ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
         o_acc[cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, o_acc[cb], 0, 0, 0);
       }
     }
+    __syncthreads();
   }
 
   // ---- epilogue: normalize, store O (bf16) and LSE (fp32) -------------
@@ -383,7 +409,8 @@ __global__ void flash_dot_do_o_kernel(float* __restrict__ di,
 
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
-    uint16_t* __restrict__ dq, const uint16_t* __restrict__ dop,
+    uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    const uint16_t* __restrict__ dop,
     const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
     const uint16_t* __restrict__ vp, const float* __restrict__ lse,
     const float* __restrict__ di,
@@ -401,6 +428,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
 
   // [wave][0]=P tile, [wave][1]=dS tile (C-layout write, A-layout read)
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+  // K tile transposed [d][kv] (swizzled) for the dQ = dS K product
+  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][BN];
+  const int st_kv = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
 
   bf16x8 aq[2], ado[2];
 #pragma unroll
@@ -432,6 +463,23 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const int n_tiles = L / BN;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
+    {
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv0 + st_kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv0 + st_kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_kt[d0][st_kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_kt[d1][st_kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
     f32x4 s[4], dp[4];
 #pragma unroll
     for (int cb = 0; cb < 4; ++cb) {
@@ -487,17 +535,27 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
         if (DROP) dpv = keep[j] ? dpv * pinv : 0.f;
         dsa[j] = (short)f32_to_bf16_bits(pv * (dpv - di_row));
       }
+      if (ds_out != nullptr) {
+        // materialize dS for the bias gradient (contiguous 16 B store;
+        // the broadcast-batch sum happens as one torch reduction)
+        union {
+          bf16x8 v;
+          uint4 u;
+        } U;
+        U.v = dsa;
+        *reinterpret_cast<uint4*>(ds_out + (bh * L + q0 + lr) * (int64_t)L +
+                                  kv0 + ks2 * 32 + lg * 8) = U.u;
+      }
 #pragma unroll
       for (int cb = 0; cb < 4; ++cb) {
-        bf16x8 bkf;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          bkf[j] = (short)kp[(bh * L + kv0 + ks2 * 32 + lg * 8 + j) * (int64_t)HD +
-                             cb * 16 + lr];
+        const int d = cb * 16 + lr;
+        const bf16x8 bkf =
+            load_frag(&lds_kt[d][(ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
         dq_acc[cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bkf, dq_acc[cb], 0, 0, 0);
       }
     }
+    __syncthreads();
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -528,6 +586,11 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const int64_t kvbase = (bh * L + kv0w) * HD;
 
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+  // dO and Q tiles transposed [d][q] (swizzled) for the dV / dK products
+  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][BM];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_qt[HD][BM];
+  const int st_q = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
 
   bf16x8 ak[2], av[2];
 #pragma unroll
@@ -550,6 +613,29 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const int n_tiles = L / BM;
   for (int tq = 0; tq < n_tiles; ++tq) {
     const int q0 = tq * BM;
+    {
+      float f0[8], f1[8];
+      const int64_t row = (bh * L + q0 + st_q) * (int64_t)HD;
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_dot[d0][st_q ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_dot[d1][st_q ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_qt[d0][st_q ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_qt[d1][st_q ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
@@ -615,19 +701,17 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
       const bf16x8 dsta = load_frag(&lds_t[wid][1][lr][ks2 * 32 + lg * 8]);
 #pragma unroll
       for (int cb = 0; cb < 4; ++cb) {
-        bf16x8 bdo, bqf;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int64_t qrow = bh * L + q0 + ks2 * 32 + lg * 8 + j;
-          bdo[j] = (short)dop[qrow * HD + cb * 16 + lr];
-          bqf[j] = (short)qp[qrow * HD + cb * 16 + lr];
-        }
+        const int d = cb * 16 + lr;
+        const int qx = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);
+        const bf16x8 bdo = load_frag(&lds_dot[d][qx]);
+        const bf16x8 bqf = load_frag(&lds_qt[d][qx]);
         dv_acc[cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta, bdo, dv_acc[cb], 0, 0, 0);
         dk_acc[cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsta, bqf, dk_acc[cb], 0, 0, 0);
       }
     }
+    __syncthreads();
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -639,117 +723,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
       dv[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
           f32_to_bf16_bits(dv_acc[cb][r]);
     }
-  }
-}
-
-// grouped dBias pass: dBias[src] = sum over the bh's mapping to src of dS.
-// grid: (L/BM qtile, nb src rows, BG groups); fp32 partials [BG][nb][L][L].
-template <bool HAS_MASK, bool DROP>
-__global__ __launch_bounds__(256) void flash_bwd_dbias_kernel(
-    float* __restrict__ part, const uint16_t* __restrict__ dop,
-    const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
-    const uint16_t* __restrict__ vp, const float* __restrict__ lse,
-    const float* __restrict__ di, const uint16_t* __restrict__ bias,
-    int64_t bias_nb, int bias_q, int64_t bias_od,
-    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
-    int L, int64_t BH, int bg_count, float pinv, uint32_t pthresh, uint64_t seed) {
-  const int qt = blockIdx.x;
-  const int src = blockIdx.y;      // bias source row block (0..nb)
-  const int bg = blockIdx.z;       // batch group
-  const int wid = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int lg = lane >> 4;
-  const int lr = lane & 15;
-  const int q0 = qt * BM + wid * 16;
-
-  const uint16_t* bias_rows[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int q = q0 + lg * 4 + r;
-    bias_rows[r] = bias + ((int64_t)src * bias_q + (q % bias_q)) * (int64_t)L;
-  }
-
-  const int n_tiles = L / BN;
-  for (int t = 0; t < n_tiles; ++t) {
-    const int kv0 = t * BN;
-    f32x4 db_acc[4] = {};
-    // loop the batch-heads whose bias source is `src`, strided by group
-    for (int64_t bh = 0; bh < BH; ++bh) {
-      if ((bh / bias_od) % bias_nb != src) continue;
-      // partition the broadcast batches by their OUTER index so the
-      // groups are even (bh and src share low-order structure)
-      if ((bh / (bias_od * bias_nb)) % (int64_t)bg_count != bg) continue;
-      const int64_t qbase = (bh * L + q0) * HD;
-      bf16x8 aq[2], ado[2];
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
-        ado[ks] = load_frag(dop + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
-      }
-      const uint16_t* mask_row =
-          HAS_MASK ? mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L
-                   : nullptr;
-      f32x4 s[4], dp[4];
-#pragma unroll
-      for (int cb = 0; cb < 4; ++cb) {
-        f32x4 acc = {}, accd = {};
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          const bf16x8 bk = load_frag(
-              kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
-          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
-          const bf16x8 bvt = load_frag(
-              vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
-          accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bvt, accd, 0, 0, 0);
-        }
-        s[cb] = acc;
-        dp[cb] = accd;
-      }
-#pragma unroll
-      for (int cb = 0; cb < 4; ++cb) {
-        const int kv = kv0 + cb * 16 + lr;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int q = q0 + lg * 4 + r;
-          float sv = s[cb][r];
-          sv += __bfloat162float(
-              reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
-          if (HAS_MASK)
-            sv += __bfloat162float(
-                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
-          const float pv = __expf(sv - lse[bh * L + q]);
-          float dpv = dp[cb][r];
-          if (DROP) {
-            Philox4 ph(seed, (uint64_t)(bh * L + q), (uint64_t)(kv >> 2));
-            const uint4 rr = ph.next();
-            const uint32_t rv[4] = {rr.x, rr.y, rr.z, rr.w};
-            dpv = rv[kv & 3] >= pthresh ? dpv * pinv : 0.f;
-          }
-          db_acc[cb][r] += pv * (dpv - di[bh * L + q]);
-        }
-      }
-    }
-    // write this (qtile, kvtile) slice of the partial
-#pragma unroll
-    for (int cb = 0; cb < 4; ++cb) {
-      const int kv = kv0 + cb * 16 + lr;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q = q0 + lg * 4 + r;
-        part[(((int64_t)bg * bias_nb + src) * L + q) * L + kv] = db_acc[cb][r];
-      }
-    }
-  }
-}
-
-__global__ void flash_dbias_reduce_kernel(uint16_t* __restrict__ dbias,
-                                          const float* __restrict__ part,
-                                          int bg_count, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    float s = 0.f;
-    for (int g = 0; g < bg_count; ++g) s += part[(int64_t)g * n + i];
-    dbias[i] = f32_to_bf16_bits(s);
   }
 }
 
@@ -788,6 +761,11 @@ std::vector<at::Tensor> flash_attn_backward(
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
+  at::Tensor ds;
+  if (bd.ptr && bias_needs_grad) {
+    TORCH_CHECK(bd.q == L, "flash_attn: bias grad requires bias_q == L");
+    ds = at::empty({BH, (int64_t)L, (int64_t)L}, q.options());
+  }
   const dim3 grid(L / BM, BH);
 
   auto launch_all = [&](auto hb, auto hm, auto dr) {
@@ -796,6 +774,7 @@ std::vector<at::Tensor> flash_attn_backward(
     constexpr bool DR = decltype(dr)::value;
     flash_bwd_dq_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
         reinterpret_cast<uint16_t*>(dq.data_ptr()),
+        ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
         reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
         reinterpret_cast<const uint16_t*>(q.data_ptr()),
         reinterpret_cast<const uint16_t*>(k.data_ptr()),
@@ -829,43 +808,7 @@ std::vector<at::Tensor> flash_attn_backward(
   else
     pick(std::false_type{}, std::false_type{});
 
-  at::Tensor dbias;
-  if (bd.ptr && bias_needs_grad) {
-    TORCH_CHECK(bd.od == 1,
-                "flash_attn: bias grad supports outer_div == 1 broadcasts");
-    const int64_t group = std::max<int64_t>(BH / bd.nb, 1);
-    const int bg_count = (int)std::min<int64_t>(group, 8);
-    auto part = at::empty({bg_count, bd.nb, (int64_t)L, (int64_t)L},
-                          q.options().dtype(at::kFloat));
-    const dim3 bgrid(L / BM, bd.nb, bg_count);
-    auto launch_db = [&](auto hm, auto dr) {
-      flash_bwd_dbias_kernel<decltype(hm)::value, decltype(dr)::value>
-          <<<bgrid, 256, 0, stream>>>(
-              part.data_ptr<float>(),
-              reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
-              reinterpret_cast<const uint16_t*>(q.data_ptr()),
-              reinterpret_cast<const uint16_t*>(k.data_ptr()),
-              reinterpret_cast<const uint16_t*>(v.data_ptr()),
-              lse.data_ptr<float>(), di.data_ptr<float>(),
-              reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
-              reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L,
-              BH, bg_count, pinv, pthresh, seed);
-    };
-    if (md.ptr) {
-      if (drop) launch_db(std::true_type{}, std::true_type{});
-      else launch_db(std::true_type{}, std::false_type{});
-    } else {
-      if (drop) launch_db(std::false_type{}, std::true_type{});
-      else launch_db(std::false_type{}, std::false_type{});
-    }
-    dbias = at::empty({bd.nb, (int64_t)bd.q, (int64_t)L}, q.options());
-    TORCH_CHECK(bd.q == L, "flash_attn: bias grad requires bias_q == L");
-    const int64_t n = (int64_t)bd.nb * L * L;
-    flash_dbias_reduce_kernel<<<unicore_grid((n + 255) / 256), 256, 0, stream>>>(
-        reinterpret_cast<uint16_t*>(dbias.data_ptr()), part.data_ptr<float>(),
-        bg_count, n);
-  }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
-  if (dbias.defined()) return {dq, dk, dv, dbias};
+  if (ds.defined()) return {dq, dk, dv, ds};
   return {dq, dk, dv};
 }

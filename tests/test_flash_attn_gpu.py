@@ -102,9 +102,11 @@ def test_flash_bwd_parity_no_dropout(L):
 
     o, lse, seed = ops.flash_attn_fwd(q, k, v, bias, 1, mask, H, 0.0, True)
     d_out = torch.randn_like(o) * 0.3
-    dq, dk, dv, dbias = ops.flash_attn_bwd(
+    dq, dk, dv, ds = ops.flash_attn_bwd(
         d_out, q, k, v, o, lse, bias, 1, True, mask, H, 0.0, False, int(seed)
     )
+    # bias grad = dS summed over the broadcast batch
+    dbias = ds.view(B, H, L, L).sum(0)
 
     # fp32 reference
     qr = q.float().view(B, H, L, 64).requires_grad_(True)
@@ -169,12 +171,16 @@ def test_flash_bwd_dropout_identity_v():
 
 
 @requires_gpu
-def test_attention_module_flash_vs_materialized():
+def test_attention_module_flash_vs_materialized(monkeypatch):
     """SelfMultiheadAttention end-to-end: the flash path must match the
     materialized bmm+softmax path (bf16, p=0) including input and bias
     gradients."""
+    import os
+
     from unicore_amd.modules import SelfMultiheadAttention
     from unicore_amd.modules import multihead_attention as mha
+
+    monkeypatch.setenv("UNICORE_FLASH_ATTN", "1")
 
     torch.manual_seed(0)
     B, L, H, D = 2, 128, 4, 64

@@ -12,7 +12,13 @@ from typing import Optional
 import torch
 from torch import Tensor, nn
 
+import os
+
 from .softmax_dropout import _broadcast_descr, softmax_dropout
+
+
+def _flash_enabled():
+    return os.environ.get("UNICORE_FLASH_ATTN", "0") == "1"
 
 
 class _FlashAttn(torch.autograd.Function):
@@ -51,7 +57,13 @@ class _FlashAttn(torch.autograd.Function):
             want_dbias, mask_t, ctx.mask_od, ctx.p, ctx.dropped, ctx.seed,
         )
         dq, dk, dv = grads[0], grads[1], grads[2]
-        dbias = grads[3] if len(grads) > 3 else None
+        dbias = None
+        if len(grads) > 3:
+            # grads[3] is materialized dS (BH, L, L); the bias gradient is
+            # its sum over the broadcast batches (outer_div == 1 contract)
+            ds = grads[3]
+            nb = bias_t.shape[0]
+            dbias = ds.view(-1, nb, ds.shape[1], ds.shape[2]).sum(0)
         return dq, dk, dv, dbias, None, None, None, None, None
 
 
@@ -180,10 +192,14 @@ class SelfMultiheadAttention(nn.Module):
             mask = key_padding_mask.view(bsz, 1, 1, src_len).to(q.dtype)
 
         # flash path: bf16, head_dim 64, L % 64 == 0, kernel-expressible
-        # bias/mask broadcasts — the L x L score matrix never hits HBM
+        # bias/mask broadcasts — the L x L score matrix never hits HBM.
+        # Opt-in (UNICORE_FLASH_ATTN=1): correct and fully tested, but the
+        # v1 tile structure does not yet beat hipBLASLt bmm + the fused
+        # softmax on BERT-base shapes (see profiles/README.md).
         o = None
         if (
             use_fused_split
+            and _flash_enabled()
             and not return_attn
             and q.dtype == torch.bfloat16
             and self.head_dim == 64
