@@ -38,6 +38,9 @@ std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
 at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
                                  double p);
 at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
+std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double p,
+                                            bool is_training);
+at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p);
 std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
                                            std::optional<at::Tensor> bias,
                                            int64_t bias_outer_div,
@@ -79,5 +82,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_forward", &flash_attn_forward,
         "flash attention forward (bf16, D=64) -> (o, lse, seed)");
   m.def("flash_attn_backward", &flash_attn_backward,
-        "flash attention backward -> (dq, dk, dv[, dbias])");
+        "flash attention backward -> (dq, dk, dv[, dS])");
+  m.def("dropout_add_forward", &dropout_add_forward,
+        "fused dropout + residual add forward");
+  m.def("dropout_add_backward", &dropout_add_backward,
+        "fused dropout + residual add backward (dx only; d_res = grad)");
 }

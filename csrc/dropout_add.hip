@@ -1,0 +1,157 @@
+// Fused dropout(x) + residual add for gfx950: out = residual + keep*pinv*x.
+// Replaces the torch dropout kernel + add kernel pair (saves one full
+// read+write of the hidden tensor per site; two sites per transformer
+// layer).  Bitfield keep-mask, Philox keyed by PyTorch's generator.
+// Backward: dx = g * keep * pinv (one pass); d_residual = g (pass-through,
+// no kernel).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGeneratorImpl.h>
+
+#include <vector>
+
+namespace {
+
+template <typename T, bool DROP>
+__global__ void dropout_add_fwd_kernel(T* __restrict__ out,
+                                       uint8_t* __restrict__ dmask,
+                                       const T* __restrict__ x,
+                                       const T* __restrict__ res, int64_t n8,
+                                       float pinv, uint32_t pthresh,
+                                       uint64_t seed, uint64_t offset) {
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  Philox4 ph(seed, (uint64_t)tid, offset);
+  for (int64_t i = tid; i < n8; i += stride) {
+    float fx[8], fr[8];
+    load8(x + i * 8, fx);
+    load8(res + i * 8, fr);
+    if constexpr (DROP) {
+      const uint4 r0 = ph.next();
+      const uint4 r1 = ph.next();
+      const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+      uint8_t bits = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const bool keep = rr[j] >= pthresh;
+        bits |= (uint8_t)(keep ? 1u : 0u) << j;
+        fx[j] = fr[j] + (keep ? fx[j] * pinv : 0.f);
+      }
+      dmask[i] = bits;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) fx[j] += fr[j];
+    }
+    store8(out + i * 8, fx);
+  }
+}
+
+template <typename T>
+__global__ void dropout_add_bwd_kernel(T* __restrict__ dx, const T* __restrict__ g,
+                                       const uint8_t* __restrict__ dmask,
+                                       int64_t n8, float pinv) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    float f[8];
+    load8(g + i * 8, f);
+    const uint8_t bits = dmask[i];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = (bits >> j) & 1 ? f[j] * pinv : 0.f;
+    store8(dx + i * 8, f);
+  }
+}
+
+#define DISPATCH_FTYPES(st, NAME, ...)                               \
+  switch (st) {                                                      \
+    case at::ScalarType::Float: {                                    \
+      using scalar_t = float;                                        \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::Half: {                                     \
+      using scalar_t = __half;                                       \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::BFloat16: {                                 \
+      using scalar_t = __hip_bfloat16;                               \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    default:                                                         \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", st);          \
+  }
+
+}  // namespace
+
+std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double p,
+                                            bool is_training) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && res.is_contiguous(),
+              "dropout_add: contiguous CUDA");
+  TORCH_CHECK(x.sizes() == res.sizes() && x.scalar_type() == res.scalar_type(),
+              "dropout_add: x/res mismatch");
+  TORCH_CHECK(x.numel() % 8 == 0, "dropout_add: numel % 8 == 0");
+  const int64_t n8 = x.numel() / 8;
+  const bool drop = is_training && p > 0.0;
+  auto out = at::empty_like(x);
+  at::Tensor dmask;
+  float pinv = 1.f;
+  uint32_t pthresh = 0;
+  uint64_t seed = 0, offset = 0;
+  if (drop) {
+    dmask = at::empty({n8}, x.options().dtype(at::kByte));
+    const double pc = std::min(p, 0.999999);
+    pinv = (float)(1.0 / (1.0 - pc));
+    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+        std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+    at::PhiloxCudaState state;
+    {
+      std::lock_guard<std::mutex> lock(gen->mutex_);
+      state = gen->philox_cuda_state(4 + n8 / (2048LL * 256) * 2);
+    }
+    seed = state.seed_.val;
+    offset = state.offset_.val;
+  } else {
+    dmask = at::empty({0}, x.options().dtype(at::kByte));
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n8 + 255) / 256);
+  DISPATCH_FTYPES(x.scalar_type(), "dropout_add_forward", {
+    if (drop)
+      dropout_add_fwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()), dmask.data_ptr<uint8_t>(),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()),
+          reinterpret_cast<const scalar_t*>(res.data_ptr()), n8, pinv, pthresh,
+          seed, offset);
+    else
+      dropout_add_fwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()), nullptr,
+          reinterpret_cast<const scalar_t*>(x.data_ptr()),
+          reinterpret_cast<const scalar_t*>(res.data_ptr()), n8, pinv, pthresh,
+          seed, offset);
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return {out, dmask};
+}
+
+at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p) {
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous(), "dropout_add_backward");
+  const int64_t n8 = grad.numel() / 8;
+  TORCH_CHECK(dmask.numel() == n8, "dropout_add_backward: mask mismatch");
+  const float pinv = (float)(1.0 / (1.0 - std::min(p, 0.999999)));
+  auto dx = at::empty_like(grad);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n8 + 255) / 256);
+  DISPATCH_FTYPES(grad.scalar_type(), "dropout_add_backward", {
+    dropout_add_bwd_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+        reinterpret_cast<scalar_t*>(dx.data_ptr()),
+        reinterpret_cast<const scalar_t*>(grad.data_ptr()),
+        dmask.data_ptr<uint8_t>(), n8, pinv);
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return dx;
+}

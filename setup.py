@@ -30,6 +30,7 @@ HIP_SOURCES = [
     "csrc/gelu_dropout.hip",
     "csrc/mfma_probe.hip",
     "csrc/flash_attn.hip",
+    "csrc/dropout_add.hip",
 ]
 
 setup(

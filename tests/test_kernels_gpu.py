@@ -397,3 +397,34 @@ def test_mfma_fragment_layout_probe():
     D = K.mfma_gemm_16x16x32(A, B)
     ref = A.float() @ B.float()
     assert (D - ref).abs().max().item() < 2e-2, (D - ref).abs().max()
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("p", [0.0, 0.2])
+def test_dropout_add_parity(dtype, p):
+    ops = _kernels()
+    from unicore_amd.modules import dropout_add
+
+    torch.manual_seed(6)
+    x = torch.randn(2048, 512, device="cuda", dtype=dtype, requires_grad=True)
+    res = torch.randn_like(x, requires_grad=True)
+    out = dropout_add(x, res, p, is_training=True)
+    gout = torch.randn_like(out)
+    out.backward(gout.clone())
+
+    if p == 0:
+        ref = x.detach() + res.detach()
+        assert torch.allclose(out.float(), ref.float(), atol=1e-5)
+        assert torch.allclose(x.grad.float(), gout.float(), atol=1e-6)
+    else:
+        # recover the keep mask from the output
+        keep = (out.detach() - res.detach()) != 0
+        zfrac = 1 - keep.float().mean().item()
+        assert abs(zfrac - p) < 0.02
+        ref = res.detach().float() + keep.float() * x.detach().float() / (1 - p)
+        tol = TOL[dtype]
+        assert (out.float() - ref).abs().max().item() < tol * 8
+        gref = gout.float() * keep.float() / (1 - p)
+        assert (x.grad.float() - gref).abs().max().item() < tol * 8
+    assert torch.allclose(res.grad.float(), gout.float(), atol=1e-6)

@@ -11,6 +11,7 @@ from torch import Tensor, nn
 
 from unicore_amd import utils
 
+from .dropout_add import dropout_add
 from .gelu_dropout import gelu_dropout
 from .layer_norm import LayerNorm
 from .multihead_attention import SelfMultiheadAttention
@@ -79,8 +80,7 @@ class TransformerEncoderLayer(nn.Module):
         )
         if return_attn:
             x, attn_weights, attn_probs = x
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        x = residual + x
+        x = dropout_add(x, residual, self.dropout, self.training)
         if self.post_ln:
             x = self.self_attn_layer_norm(x)
 
@@ -94,8 +94,7 @@ class TransformerEncoderLayer(nn.Module):
             x = self.activation_fn(x)
             x = F.dropout(x, p=self.activation_dropout, training=self.training)
         x = self.fc2(x)
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        x = residual + x
+        x = dropout_add(x, residual, self.dropout, self.training)
         if self.post_ln:
             x = self.final_layer_norm(x)
         if not return_attn:

@@ -174,3 +174,13 @@ def flash_attn_bwd(d_out, q, k, v, o, lse, bias, bias_outer_div, bias_needs_grad
         d_out, q, k, v, o, lse, bias, int(bias_outer_div), bool(bias_needs_grad),
         mask, int(mask_outer_div), float(p), bool(dropped), int(seed)
     )
+
+
+def dropout_add_fwd(x, res, p, is_training):
+    require_kernels()
+    return _kernels.dropout_add_forward(x, res, float(p), bool(is_training))
+
+
+def dropout_add_bwd(grad, dmask, p):
+    require_kernels()
+    return _kernels.dropout_add_backward(grad, dmask, float(p))
