@@ -148,6 +148,27 @@ struct Philox4 {
   }
 };
 
+// 8 dropout keep-decisions from ONE philox draw (16-bit threshold
+// resolution — plenty for dropout probabilities).  Definition shared by
+// every dropout producer AND the flash-attention backward recompute:
+//   keep(subseq, idx) = u16 #(idx%8) of philox(seed, subseq, idx/8)
+__device__ __forceinline__ void keep16x8(uint64_t seed, uint64_t subseq,
+                                         int idx0 /* multiple of 8 */,
+                                         uint32_t pthresh16, bool (&keep)[8]) {
+  Philox4 ph(seed, subseq, (uint64_t)(idx0 >> 3));
+  const uint4 r = ph.next();
+  const uint32_t rr[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    keep[j] = ((rr[j >> 1] >> ((j & 1) * 16)) & 0xFFFFu) >= pthresh16;
+}
+
+static inline uint32_t keep16_threshold(double p) {
+  const double pc = p < 0.999999 ? p : 0.999999;
+  const double t = pc * 65536.0;
+  return t > 65535.0 ? 65535u : (uint32_t)t;
+}
+
 // bf16 <-> bits helpers (avoid relying on __bfloat16_as_ushort availability)
 __device__ __forceinline__ uint16_t f32_to_bf16_bits(float x) {
   union {

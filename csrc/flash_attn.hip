@@ -59,8 +59,8 @@ __device__ __forceinline__ float rowgroup_sum(float v) {
   return v;
 }
 
-// keep-decision for dropout: element (row_global = bh*L + q, col kv)
-// philox(seed, row_global, kv/4) component kv%4, keep iff r >= pthresh.
+// keep-decision for dropout: element (row_global = bh*L + q, col kv) via
+// the shared 16-bit keep definition: keep16x8(seed, row_global, kv_block).
 template <bool DROP>
 __device__ __forceinline__ void keep_bits8(uint64_t seed, uint64_t subseq,
                                            int kv0, uint32_t pthresh,
@@ -70,12 +70,7 @@ __device__ __forceinline__ void keep_bits8(uint64_t seed, uint64_t subseq,
     for (int j = 0; j < 8; ++j) keep[j] = true;
     return;
   }
-  Philox4 ph(seed, subseq, (uint64_t)(kv0 >> 2));
-  const uint4 r0 = ph.next();
-  const uint4 r1 = ph.next();
-  const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
-#pragma unroll
-  for (int j = 0; j < 8; ++j) keep[j] = rr[j] >= pthresh;
+  keep16x8(seed, subseq, kv0, pthresh, keep);
 }
 
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
@@ -320,7 +315,7 @@ std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tenso
   if (drop) {
     const double pc = std::min(dropout_p, 0.999999);
     pinv = (float)(1.0 / (1.0 - pc));
-    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    pthresh = keep16_threshold(pc);
     auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
         std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
     at::PhiloxCudaState state;
@@ -665,13 +660,13 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
                    : nullptr;
       bool keep[4] = {true, true, true, true};
       if (DROP) {
-        // kv rows of this lane: kv0w + lg*4 + r, all in one philox counter
-        Philox4 ph(seed, (uint64_t)(bh * L + qcol),
-                   (uint64_t)((kv0w + lg * 4) >> 2));
-        const uint4 rr = ph.next();
-        const uint32_t rv[4] = {rr.x, rr.y, rr.z, rr.w};
+        // this lane's kv rows kv0w + lg*4 + [0..4) live in one 8-block
+        bool k8[8];
+        const int blk = (kv0w + lg * 4) & ~7;
+        keep16x8(seed, (uint64_t)(bh * L + qcol), blk, pthresh, k8);
+        const int off = (kv0w + lg * 4) - blk;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) keep[r] = rv[r] >= pthresh;
+        for (int r = 0; r < 4; ++r) keep[r] = k8[off + r];
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -747,7 +742,7 @@ std::vector<at::Tensor> flash_attn_backward(
   if (drop) {
     const double pc = std::min(dropout_p, 0.999999);
     pinv = (float)(1.0 / (1.0 - pc));
-    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    pthresh = keep16_threshold(pc);
   }
   const uint64_t seed = (uint64_t)seed_in;
 

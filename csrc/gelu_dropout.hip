@@ -33,22 +33,20 @@ __global__ void gelu_dropout_fwd_kernel(T* __restrict__ out,
                                         uint64_t seed, uint64_t offset) {
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  Philox4 ph(seed, (uint64_t)tid, offset);
   for (int64_t i = tid; i < n8; i += stride) {
     float f[8];
     load8(x + i * 8, f);
 #pragma unroll
     for (int j = 0; j < 8; ++j) f[j] = gelu_fwd(f[j]);
     if constexpr (DROP) {
-      const uint4 r0 = ph.next();
-      const uint4 r1 = ph.next();
-      const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+      bool keep[8];
+      keep16x8(seed + offset * 0x9E3779B97F4A7C15ull, (uint64_t)i, 0, pthresh,
+               keep);
       uint8_t bits = 0;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const bool keep = rr[j] >= pthresh;
-        bits |= (uint8_t)(keep ? 1u : 0u) << j;
-        f[j] = keep ? f[j] * pinv : 0.f;
+        bits |= (uint8_t)(keep[j] ? 1u : 0u) << j;
+        f[j] = keep[j] ? f[j] * pinv : 0.f;
       }
       dmask[i] = bits;
     }
@@ -117,7 +115,7 @@ std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
     dmask = at::empty({n8}, x.options().dtype(at::kByte));
     const double pc = std::min(p, 0.999999);
     pinv = (float)(1.0 / (1.0 - pc));
-    pthresh = (uint32_t)std::min<double>(pc * 4294967296.0, 4294967295.0);
+    pthresh = keep16_threshold(pc);
     auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
         std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
     at::PhiloxCudaState state;

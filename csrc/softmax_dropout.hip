@@ -85,7 +85,6 @@ __global__ void softmax_fwd_vec_kernel(
     sum = wave_sum(sum);
     const float inv = 1.0f / sum;
 
-    Philox4 ph(seed, (uint64_t)row * 64 + lane, rng_offset);
 #pragma unroll
     for (int i = 0; i < NV; ++i) {
       const int e0 = (lane + i * 64) * 8;
@@ -95,16 +94,15 @@ __global__ void softmax_fwd_vec_kernel(
         for (int j = 0; j < 8; ++j) y[j] = vals[i][j] * inv;
         store8(xrow + e0, y);  // pre-dropout softmax, in-place over input
         if constexpr (DROP) {
-          const uint4 r0 = ph.next();
-          const uint4 r1 = ph.next();
-          const uint32_t rr[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+          bool keep[8];
+          keep16x8(seed, (uint64_t)row * 64 + lane, (lane + i * 64) * 8,
+                   pthresh, keep);
           uint8_t bits = 0;
           float o[8];
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            const bool keep = rr[j] >= pthresh;
-            bits |= (uint8_t)(keep ? 1u : 0u) << j;
-            o[j] = keep ? y[j] * pinv : 0.f;
+            bits |= (uint8_t)(keep[j] ? 1u : 0u) << j;
+            o[j] = keep[j] ? y[j] * pinv : 0.f;
           }
           dmask[row * (int64_t)mrow_bytes + lane + i * 64] = bits;
           store8(out + row * (int64_t)k + e0, o);
@@ -319,7 +317,7 @@ std::vector<at::Tensor> softmax_dropout_forward(
                       input.options().dtype(at::kByte));
     const double p = std::min(dropout_prob, 0.999999);
     pinv = (float)(1.0 / (1.0 - p));
-    pthresh = (uint32_t)std::min<double>(p * 4294967296.0, 4294967295.0);
+    pthresh = keep16_threshold(p);
     auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
         std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
     const int nv = (k + 511) / 512;
@@ -328,8 +326,10 @@ std::vector<at::Tensor> softmax_dropout_forward(
       std::lock_guard<std::mutex> lock(gen->mutex_);
       state = gen->philox_cuda_state(2 * nv);
     }
-    seed = state.seed_.val;
-    rng_offset = state.offset_.val;
+    // fold the generator offset into the key so consecutive calls draw
+    // independent streams while fwd remains seed-deterministic
+    seed = state.seed_.val + state.offset_.val * 0x9E3779B97F4A7C15ull;
+    rng_offset = 0;
   } else {
     dmask = at::empty({0}, input.options().dtype(at::kByte));
   }
