@@ -34,7 +34,7 @@ def _reference_grads(batch):
     return [p.grad.clone() for p in m.parameters()]
 
 
-def _ddp_worker(rank, world, port, engine, out_dir):
+def _ddp_worker(rank, world, port, engine, out_dir, bucket_cap=0.0001):
     _init(rank, world, port)
     from unicore_amd.distributed import FlatDDP, LegacyDDP
 
@@ -44,7 +44,9 @@ def _ddp_worker(rank, world, port, engine, out_dir):
 
     m = _model()
     if engine == "flat":
-        ddp = FlatDDP(m, process_group=dist.group.WORLD, bucket_cap_mb=0.0001)
+        # small cap -> one param per bucket; large cap -> multi-param buckets
+        # (regression: identity lookup inside a mixed-shape bucket)
+        ddp = FlatDDP(m, process_group=dist.group.WORLD, bucket_cap_mb=bucket_cap)
     else:
         ddp = LegacyDDP(m, process_group=dist.group.WORLD, buffer_size=2**10)
 
@@ -64,13 +66,16 @@ def _ddp_worker(rank, world, port, engine, out_dir):
     dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("engine", ["flat", "legacy"])
-def test_ddp_grad_parity(engine, tmp_path):
-    port = 29531 if engine == "flat" else 29532
+@pytest.mark.parametrize("engine,bucket_cap", [
+    ("flat", 0.0001), ("flat", 32.0), ("legacy", 0.0001),
+])
+def test_ddp_grad_parity(engine, bucket_cap, tmp_path):
+    port = 29531 + int(bucket_cap > 0.001) * 7 + (engine == "legacy")
     world = 2
     ctx = mp.get_context("spawn")
     procs = [
-        ctx.Process(target=_ddp_worker, args=(r, world, port, engine, str(tmp_path)))
+        ctx.Process(target=_ddp_worker,
+                    args=(r, world, port, engine, str(tmp_path), bucket_cap))
         for r in range(world)
     ]
     for p in procs:

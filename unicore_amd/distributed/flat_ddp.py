@@ -214,13 +214,21 @@ class FlatDDP(nn.Module):
         return hook
 
     @staticmethod
-    def _view_ptr(bucket, param):
-        idx = bucket.params.index(param)
-        return bucket.views[idx].data_ptr()
+    def _param_index(bucket, param):
+        # identity scan — list.index would invoke tensor __eq__ on
+        # non-identical params (elementwise broadcast error)
+        for i, q in enumerate(bucket.params):
+            if q is param:
+                return i
+        raise KeyError("param not in bucket")
 
-    @staticmethod
-    def _restore_view(bucket, param):
-        idx = bucket.params.index(param)
+    @classmethod
+    def _view_ptr(cls, bucket, param):
+        return bucket.views[cls._param_index(bucket, param)].data_ptr()
+
+    @classmethod
+    def _restore_view(cls, bucket, param):
+        idx = cls._param_index(bucket, param)
         bucket.views[idx].copy_(param.grad)
         param.grad = bucket.views[idx]
 
