@@ -27,129 +27,163 @@
 
 namespace {
 
-template <typename T, int NV, bool DROP>
+template <typename T, int NV, int ROWS, bool DROP>
 __global__ void softmax_fwd_vec_kernel(
     T* __restrict__ out, uint8_t* __restrict__ dmask, T* __restrict__ x,
     const T* __restrict__ amask, int64_t am_nb, int am_q, int64_t am_od,
     const T* __restrict__ bias, int64_t bs_nb, int bs_q, int64_t bs_od,
     int64_t n_rows, int q_len, int k, float pinv, uint32_t pthresh,
     uint64_t seed, uint64_t rng_offset) {
+  // ROWS rows per wave per iteration: the row loads issue back-to-back so
+  // several 16 B/lane requests are in flight (one row alone leaves the
+  // memory system starved at high occupancy)
   const int lane = threadIdx.x;
   const int wid = threadIdx.y;
   const int mrow_bytes = k / 8;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.y + wid; row < n_rows;
-       row += (int64_t)gridDim.x * blockDim.y) {
-    T* xrow = x + row * (int64_t)k;
-    const int64_t b = row / q_len;
-    const int qi = (int)(row - b * q_len);
-    const T* mrow = amask ? amask + (((b / am_od) % am_nb) * am_q + (qi % am_q)) * (int64_t)k
-                          : nullptr;
-    const T* brow = bias ? bias + (((b / bs_od) % bs_nb) * bs_q + (qi % bs_q)) * (int64_t)k
-                         : nullptr;
-
-    float vals[NV][8];
-    float mx = -INFINITY;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.y * ROWS;
+  for (int64_t row0 = ((int64_t)blockIdx.x * blockDim.y + wid) * ROWS;
+       row0 < n_rows; row0 += stride) {
+    float vals[ROWS][NV][8];
+    float mx[ROWS], inv[ROWS];
 #pragma unroll
-    for (int i = 0; i < NV; ++i) {
-      const int e0 = (lane + i * 64) * 8;
-      if (e0 < k) {
-        load8(xrow + e0, vals[i]);
-        if (mrow) {
-          float t[8];
-          load8(mrow + e0, t);
+    for (int r = 0; r < ROWS; ++r) {
+      const int64_t row = row0 + r;
+      if (row >= n_rows) break;
+      T* xrow = x + row * (int64_t)k;
+      const int64_t b = row / q_len;
+      const int qi = (int)(row - b * q_len);
+      const T* mrow =
+          amask ? amask + (((b / am_od) % am_nb) * am_q + (qi % am_q)) * (int64_t)k
+                : nullptr;
+      const T* brow =
+          bias ? bias + (((b / bs_od) % bs_nb) * bs_q + (qi % bs_q)) * (int64_t)k
+               : nullptr;
+      float m = -INFINITY;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) vals[i][j] += t[j];
-        }
-        if (brow) {
-          float t[8];
-          load8(brow + e0, t);
+      for (int i = 0; i < NV; ++i) {
+        const int e0 = (lane + i * 64) * 8;
+        if (e0 < k) {
+          load8(xrow + e0, vals[r][i]);
+          if (mrow) {
+            float t[8];
+            load8(mrow + e0, t);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) vals[i][j] += t[j];
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j) mx = fmaxf(mx, vals[i][j]);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vals[i][j] = -INFINITY;
-      }
-    }
-    mx = wave_max(mx);
-    float sum = 0.f;
-#pragma unroll
-    for (int i = 0; i < NV; ++i)
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        vals[i][j] = __expf(vals[i][j] - mx);
-        sum += vals[i][j];
-      }
-    sum = wave_sum(sum);
-    const float inv = 1.0f / sum;
-
-#pragma unroll
-    for (int i = 0; i < NV; ++i) {
-      const int e0 = (lane + i * 64) * 8;
-      if (e0 < k) {
-        float y[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) y[j] = vals[i][j] * inv;
-        store8(xrow + e0, y);  // pre-dropout softmax, in-place over input
-        if constexpr (DROP) {
-          bool keep[8];
-          keep16x8(seed, (uint64_t)row * 64 + lane, (lane + i * 64) * 8,
-                   pthresh, keep);
-          uint8_t bits = 0;
-          float o[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            bits |= (uint8_t)(keep[j] ? 1u : 0u) << j;
-            o[j] = keep[j] ? y[j] * pinv : 0.f;
+            for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
           }
-          dmask[row * (int64_t)mrow_bytes + lane + i * 64] = bits;
-          store8(out + row * (int64_t)k + e0, o);
+          if (brow) {
+            float t[8];
+            load8(brow + e0, t);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j) m = fmaxf(m, vals[r][i][j]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[r][i][j] = -INFINITY;
+        }
+      }
+      mx[r] = wave_max(m);
+    }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      if (row0 + r >= n_rows) break;
+      float sum = 0.f;
+#pragma unroll
+      for (int i = 0; i < NV; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          vals[r][i][j] = __expf(vals[r][i][j] - mx[r]);
+          sum += vals[r][i][j];
+        }
+      inv[r] = 1.0f / wave_sum(sum);
+    }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      const int64_t row = row0 + r;
+      if (row >= n_rows) break;
+      T* xrow = x + row * (int64_t)k;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        const int e0 = (lane + i * 64) * 8;
+        if (e0 < k) {
+          float y[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) y[j] = vals[r][i][j] * inv[r];
+          store8(xrow + e0, y);  // pre-dropout softmax, in-place over input
+          if constexpr (DROP) {
+            bool keep[8];
+            keep16x8(seed, (uint64_t)row * 64 + lane, (lane + i * 64) * 8,
+                     pthresh, keep);
+            uint8_t bits = 0;
+            float o[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              bits |= (uint8_t)(keep[j] ? 1u : 0u) << j;
+              o[j] = keep[j] ? y[j] * pinv : 0.f;
+            }
+            dmask[row * (int64_t)mrow_bytes + lane + i * 64] = bits;
+            store8(out + row * (int64_t)k + e0, o);
+          }
         }
       }
     }
   }
 }
 
-template <typename T, int NV, bool DROP>
+template <typename T, int NV, int ROWS, bool DROP>
 __global__ void softmax_bwd_vec_kernel(T* __restrict__ g, const T* __restrict__ y,
                                        const uint8_t* __restrict__ dmask,
                                        int64_t n_rows, int k, float pinv) {
   const int lane = threadIdx.x;
   const int wid = threadIdx.y;
   const int mrow_bytes = k / 8;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.y + wid; row < n_rows;
-       row += (int64_t)gridDim.x * blockDim.y) {
-    T* grow = g + row * (int64_t)k;
-    const T* yrow = y + row * (int64_t)k;
-    float tv[NV][8], yv[NV][8];
-    float s = 0.f;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.y * ROWS;
+  for (int64_t row0 = ((int64_t)blockIdx.x * blockDim.y + wid) * ROWS;
+       row0 < n_rows; row0 += stride) {
+    float tv[ROWS][NV][8], yv[ROWS][NV][8];
+    float sr[ROWS];
 #pragma unroll
-    for (int i = 0; i < NV; ++i) {
-      const int e0 = (lane + i * 64) * 8;
-      if (e0 < k) {
-        load8(grow + e0, tv[i]);
-        load8(yrow + e0, yv[i]);
-        if constexpr (DROP) {
-          const uint8_t bits = dmask[row * (int64_t)mrow_bytes + lane + i * 64];
+    for (int r = 0; r < ROWS; ++r) {
+      const int64_t row = row0 + r;
+      if (row >= n_rows) break;
+      T* grow = g + row * (int64_t)k;
+      const T* yrow = y + row * (int64_t)k;
+      float ssum = 0.f;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        const int e0 = (lane + i * 64) * 8;
+        if (e0 < k) {
+          load8(grow + e0, tv[r][i]);
+          load8(yrow + e0, yv[r][i]);
+          if constexpr (DROP) {
+            const uint8_t bits =
+                dmask[row * (int64_t)mrow_bytes + lane + i * 64];
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              tv[r][i][j] = (bits >> j) & 1 ? tv[r][i][j] * pinv : 0.f;
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j) ssum += tv[r][i][j] * yv[r][i][j];
+        }
+      }
+      sr[r] = wave_sum(ssum);
+    }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      const int64_t row = row0 + r;
+      if (row >= n_rows) break;
+      T* grow = g + row * (int64_t)k;
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        const int e0 = (lane + i * 64) * 8;
+        if (e0 < k) {
+          float dx[8];
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            tv[i][j] = (bits >> j) & 1 ? tv[i][j] * pinv : 0.f;
+            dx[j] = yv[r][i][j] * (tv[r][i][j] - sr[r]);
+          store8(grow + e0, dx);
         }
-#pragma unroll
-        for (int j = 0; j < 8; ++j) s += tv[i][j] * yv[i][j];
-      }
-    }
-    s = wave_sum(s);
-#pragma unroll
-    for (int i = 0; i < NV; ++i) {
-      const int e0 = (lane + i * 64) * 8;
-      if (e0 < k) {
-        float dx[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) dx[j] = yv[i][j] * (tv[i][j] - s);
-        store8(grow + e0, dx);
       }
     }
   }
@@ -336,18 +370,21 @@ std::vector<at::Tensor> softmax_dropout_forward(
 
   if (vec_ok) {
     const dim3 block(64, 4);
-    const dim3 grid(unicore_grid((n_rows + 3) / 4));
+    const int rows_mult = k <= 512 ? 16 : (k <= 1024 ? 8 : 4);
+    const dim3 grid(unicore_grid((n_rows + rows_mult - 1) / rows_mult));
     DISPATCH_FTYPES(input.scalar_type(), "softmax_dropout_forward", {
       auto launch = [&](auto nv_tag, auto drop_tag) {
         constexpr int NV = decltype(nv_tag)::value;
         constexpr bool DROP = decltype(drop_tag)::value;
-        softmax_fwd_vec_kernel<scalar_t, NV, DROP><<<grid, block, 0, stream>>>(
-            reinterpret_cast<scalar_t*>(out.data_ptr()),
-            drop ? dmask.data_ptr<uint8_t>() : nullptr,
-            reinterpret_cast<scalar_t*>(input.data_ptr()),
-            reinterpret_cast<const scalar_t*>(m.ptr), m.nb, m.q, m.od,
-            reinterpret_cast<const scalar_t*>(bsrc.ptr), bsrc.nb, bsrc.q, bsrc.od,
-            n_rows, q_len, k, pinv, pthresh, seed, rng_offset);
+        constexpr int ROWS = NV == 1 ? 4 : (NV == 2 ? 2 : 1);
+        softmax_fwd_vec_kernel<scalar_t, NV, ROWS, DROP>
+            <<<grid, block, 0, stream>>>(
+                reinterpret_cast<scalar_t*>(out.data_ptr()),
+                drop ? dmask.data_ptr<uint8_t>() : nullptr,
+                reinterpret_cast<scalar_t*>(input.data_ptr()),
+                reinterpret_cast<const scalar_t*>(m.ptr), m.nb, m.q, m.od,
+                reinterpret_cast<const scalar_t*>(bsrc.ptr), bsrc.nb, bsrc.q,
+                bsrc.od, n_rows, q_len, k, pinv, pthresh, seed, rng_offset);
       };
       auto pick_nv = [&](auto drop_tag) {
         if (k <= 512)
@@ -396,15 +433,19 @@ at::Tensor softmax_dropout_backward(at::Tensor grad_output, at::Tensor softmax_r
   auto stream = at::cuda::getCurrentCUDAStream();
   if (vec_ok) {
     const dim3 block(64, 4);
-    const dim3 grid(unicore_grid((n_rows + 3) / 4));
+    const int rows_mult = k <= 512 ? 16 : (k <= 1024 ? 8 : 4);
+    const dim3 grid(unicore_grid((n_rows + rows_mult - 1) / rows_mult));
     DISPATCH_FTYPES(grad_output.scalar_type(), "softmax_dropout_backward", {
       auto launch = [&](auto nv_tag, auto drop_tag) {
         constexpr int NV = decltype(nv_tag)::value;
         constexpr bool DROP = decltype(drop_tag)::value;
-        softmax_bwd_vec_kernel<scalar_t, NV, DROP><<<grid, block, 0, stream>>>(
-            reinterpret_cast<scalar_t*>(grad_output.data_ptr()),
-            reinterpret_cast<const scalar_t*>(softmax_results.data_ptr()),
-            drop ? dropout_mask.data_ptr<uint8_t>() : nullptr, n_rows, k, pinv);
+        constexpr int ROWS = NV == 1 ? 4 : (NV == 2 ? 2 : 1);
+        softmax_bwd_vec_kernel<scalar_t, NV, ROWS, DROP>
+            <<<grid, block, 0, stream>>>(
+                reinterpret_cast<scalar_t*>(grad_output.data_ptr()),
+                reinterpret_cast<const scalar_t*>(softmax_results.data_ptr()),
+                drop ? dropout_mask.data_ptr<uint8_t>() : nullptr, n_rows, k,
+                pinv);
       };
       auto pick_nv = [&](auto drop_tag) {
         if (k <= 512)
