@@ -45,63 +45,73 @@ __global__ void softmax_fwd_vec_kernel(
        row0 < n_rows; row0 += stride) {
     float vals[ROWS][NV][8];
     float mx[ROWS], inv[ROWS];
+    // phase 1: issue every row's loads (keeps several 16 B requests in
+    // flight) and fold mask/bias + the per-lane max; cross-lane reductions
+    // happen in phase 2 so no shuffle chain sits between two rows' loads
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) {
       const int64_t row = row0 + r;
-      if (row >= n_rows) break;
-      T* xrow = x + row * (int64_t)k;
-      const int64_t b = row / q_len;
-      const int qi = (int)(row - b * q_len);
-      const T* mrow =
-          amask ? amask + (((b / am_od) % am_nb) * am_q + (qi % am_q)) * (int64_t)k
-                : nullptr;
-      const T* brow =
-          bias ? bias + (((b / bs_od) % bs_nb) * bs_q + (qi % bs_q)) * (int64_t)k
-               : nullptr;
       float m = -INFINITY;
+      if (row < n_rows) {
+        T* xrow = x + row * (int64_t)k;
+        const int64_t b = row / q_len;
+        const int qi = (int)(row - b * q_len);
+        const T* mrow = amask ? amask + (((b / am_od) % am_nb) * am_q +
+                                         (qi % am_q)) * (int64_t)k
+                              : nullptr;
+        const T* brow = bias ? bias + (((b / bs_od) % bs_nb) * bs_q +
+                                       (qi % bs_q)) * (int64_t)k
+                             : nullptr;
 #pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        const int e0 = (lane + i * 64) * 8;
-        if (e0 < k) {
-          load8(xrow + e0, vals[r][i]);
-          if (mrow) {
-            float t[8];
-            load8(mrow + e0, t);
+        for (int i = 0; i < NV; ++i) {
+          const int e0 = (lane + i * 64) * 8;
+          if (e0 < k) {
+            load8(xrow + e0, vals[r][i]);
+            if (mrow) {
+              float t[8];
+              load8(mrow + e0, t);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
+              for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
+            }
+            if (brow) {
+              float t[8];
+              load8(brow + e0, t);
+#pragma unroll
+              for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
+            }
+#pragma unroll
+            for (int j = 0; j < 8; ++j) m = fmaxf(m, vals[r][i][j]);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) vals[r][i][j] = -INFINITY;
           }
-          if (brow) {
-            float t[8];
-            load8(brow + e0, t);
-#pragma unroll
-            for (int j = 0; j < 8; ++j) vals[r][i][j] += t[j];
-          }
-#pragma unroll
-          for (int j = 0; j < 8; ++j) m = fmaxf(m, vals[r][i][j]);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) vals[r][i][j] = -INFINITY;
         }
       }
-      mx[r] = wave_max(m);
+      mx[r] = m;
     }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) mx[r] = wave_max(mx[r]);
+    float psum[ROWS];
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) {
-      if (row0 + r >= n_rows) break;
       float sum = 0.f;
+      if (row0 + r < n_rows) {
 #pragma unroll
-      for (int i = 0; i < NV; ++i)
+        for (int i = 0; i < NV; ++i)
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          vals[r][i][j] = __expf(vals[r][i][j] - mx[r]);
-          sum += vals[r][i][j];
-        }
-      inv[r] = 1.0f / wave_sum(sum);
+          for (int j = 0; j < 8; ++j) {
+            vals[r][i][j] = __expf(vals[r][i][j] - mx[r]);
+            sum += vals[r][i][j];
+          }
+      }
+      psum[r] = sum;
     }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) inv[r] = 1.0f / wave_sum(psum[r]);
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) {
       const int64_t row = row0 + r;
-      if (row >= n_rows) break;
+      if (row >= n_rows) continue;
       T* xrow = x + row * (int64_t)k;
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
@@ -146,33 +156,36 @@ __global__ void softmax_bwd_vec_kernel(T* __restrict__ g, const T* __restrict__ 
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) {
       const int64_t row = row0 + r;
-      if (row >= n_rows) break;
-      T* grow = g + row * (int64_t)k;
-      const T* yrow = y + row * (int64_t)k;
       float ssum = 0.f;
+      if (row < n_rows) {
+        T* grow = g + row * (int64_t)k;
+        const T* yrow = y + row * (int64_t)k;
 #pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        const int e0 = (lane + i * 64) * 8;
-        if (e0 < k) {
-          load8(grow + e0, tv[r][i]);
-          load8(yrow + e0, yv[r][i]);
-          if constexpr (DROP) {
-            const uint8_t bits =
-                dmask[row * (int64_t)mrow_bytes + lane + i * 64];
+        for (int i = 0; i < NV; ++i) {
+          const int e0 = (lane + i * 64) * 8;
+          if (e0 < k) {
+            load8(grow + e0, tv[r][i]);
+            load8(yrow + e0, yv[r][i]);
+            if constexpr (DROP) {
+              const uint8_t bits =
+                  dmask[row * (int64_t)mrow_bytes + lane + i * 64];
 #pragma unroll
-            for (int j = 0; j < 8; ++j)
-              tv[r][i][j] = (bits >> j) & 1 ? tv[r][i][j] * pinv : 0.f;
+              for (int j = 0; j < 8; ++j)
+                tv[r][i][j] = (bits >> j) & 1 ? tv[r][i][j] * pinv : 0.f;
+            }
+#pragma unroll
+            for (int j = 0; j < 8; ++j) ssum += tv[r][i][j] * yv[r][i][j];
           }
-#pragma unroll
-          for (int j = 0; j < 8; ++j) ssum += tv[r][i][j] * yv[r][i][j];
         }
       }
-      sr[r] = wave_sum(ssum);
+      sr[r] = ssum;
     }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) sr[r] = wave_sum(sr[r]);
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) {
       const int64_t row = row0 + r;
-      if (row >= n_rows) break;
+      if (row >= n_rows) continue;
       T* grow = g + row * (int64_t)k;
 #pragma unroll
       for (int i = 0; i < NV; ++i) {
