@@ -41,6 +41,8 @@ at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
 std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double p,
                                             bool is_training);
 at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p);
+at::Tensor embedding_backward(at::Tensor grad, at::Tensor indices,
+                              int64_t num_embeddings, int64_t padding_idx);
 std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
                                            std::optional<at::Tensor> bias,
                                            int64_t bias_outer_div,
@@ -87,4 +89,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused dropout + residual add forward");
   m.def("dropout_add_backward", &dropout_add_backward,
         "fused dropout + residual add backward (dx only; d_res = grad)");
+  m.def("embedding_backward", &embedding_backward,
+        "atomic-scatter embedding gradient (fp32 accumulate)");
 }

@@ -1,0 +1,122 @@
+// Embedding backward (dense-gradient scatter) for gfx950.
+//
+// torch's sort-based embedding backward (sum_and_scatter) costs ~0.6 ms per
+// step on BERT-base; this kernel instead accumulates the token-gradient rows
+// into an fp32 buffer with atomics (contention per vocab row is
+// tokens/vocab, ~1.6 on the BERT bench) and converts to the param dtype.
+// NOTE: atomic accumulation order is non-deterministic; callers fall back to
+// the torch path when torch.use_deterministic_algorithms is on.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+namespace {
+
+template <typename T>
+__global__ void embedding_bwd_scatter_kernel(float* __restrict__ acc,
+                                             const int64_t* __restrict__ idx,
+                                             const T* __restrict__ grad,
+                                             int64_t n_tokens, int dim,
+                                             int64_t padding_idx) {
+  // one 16-lane group per token row chunk: lane handles 4 elems
+  const int64_t t0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / 16;
+  const int sub = threadIdx.x & 15;
+  const int64_t stride = ((int64_t)gridDim.x * blockDim.x) / 16;
+  for (int64_t t = t0; t < n_tokens; t += stride) {
+    const int64_t v = idx[t];
+    if (v == padding_idx) continue;
+    float* dst = acc + v * dim;
+    const T* src = grad + t * dim;
+    for (int e = sub * 4; e + 3 < dim; e += 64) {
+      float f[4];
+      if constexpr (sizeof(T) == 2) {
+        union {
+          uint2 u;
+          T tt[4];
+        } U;
+        U.u = *reinterpret_cast<const uint2*>(src + e);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) f[j] = Cvt<T>::to_f(U.tt[j]);
+      } else {
+        const float4 g4 = *reinterpret_cast<const float4*>(src + e);
+        f[0] = g4.x; f[1] = g4.y; f[2] = g4.z; f[3] = g4.w;
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) atomicAdd(dst + e + j, f[j]);
+    }
+    // ragged tail (dim % 4 != 0)
+    if (sub == 15) {
+      for (int e = dim & ~3; e < dim; ++e)
+        atomicAdd(dst + e, Cvt<T>::to_f(src[e]));
+    }
+  }
+}
+
+template <typename T>
+__global__ void f32_to_t_kernel(T* __restrict__ out, const float* __restrict__ in,
+                                int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4; i0 < n;
+       i0 += stride) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      if (i0 + j < n) out[i0 + j] = Cvt<T>::from_f(in[i0 + j]);
+  }
+}
+
+#define DISPATCH_FTYPES(st, NAME, ...)                               \
+  switch (st) {                                                      \
+    case at::ScalarType::Float: {                                    \
+      using scalar_t = float;                                        \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::Half: {                                     \
+      using scalar_t = __half;                                       \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::BFloat16: {                                 \
+      using scalar_t = __hip_bfloat16;                               \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    default:                                                         \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", st);          \
+  }
+
+}  // namespace
+
+at::Tensor embedding_backward(at::Tensor grad, at::Tensor indices,
+                              int64_t num_embeddings, int64_t padding_idx) {
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous(), "embedding_backward");
+  TORCH_CHECK(indices.scalar_type() == at::kLong && indices.is_contiguous(),
+              "embedding_backward: int64 indices");
+  const int dim = (int)grad.size(-1);
+  const int64_t n_tokens = grad.numel() / dim;
+  TORCH_CHECK(indices.numel() == n_tokens, "embedding_backward: size mismatch");
+  auto acc =
+      at::zeros({num_embeddings, (int64_t)dim}, grad.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n_tokens * 16 + 255) / 256);
+  DISPATCH_FTYPES(grad.scalar_type(), "embedding_backward", {
+    embedding_bwd_scatter_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+        acc.data_ptr<float>(), indices.data_ptr<int64_t>(),
+        reinterpret_cast<const scalar_t*>(grad.data_ptr()), n_tokens, dim,
+        padding_idx);
+  });
+  if (grad.scalar_type() == at::kFloat) {
+    C10_CUDA_KERNEL_LAUNCH_CHECK();
+    return acc;
+  }
+  auto out = at::empty({num_embeddings, (int64_t)dim}, grad.options());
+  const int64_t n = out.numel();
+  DISPATCH_FTYPES(grad.scalar_type(), "embedding_backward_cast", {
+    f32_to_t_kernel<scalar_t><<<unicore_grid((n / 4 + 255) / 256), 256, 0,
+                                stream>>>(
+        reinterpret_cast<scalar_t*>(out.data_ptr()), acc.data_ptr<float>(), n);
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return out;
+}

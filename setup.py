@@ -31,6 +31,7 @@ HIP_SOURCES = [
     "csrc/mfma_probe.hip",
     "csrc/flash_attn.hip",
     "csrc/dropout_add.hip",
+    "csrc/embedding.hip",
 ]
 
 setup(

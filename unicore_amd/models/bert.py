@@ -10,6 +10,8 @@ import logging
 
 import torch
 import torch.nn as nn
+
+from unicore_amd.modules.embedding import Embedding
 import torch.nn.functional as F
 
 from unicore_amd import utils
@@ -98,10 +100,10 @@ class BertModel(BaseUnicoreModel):
         base_architecture(args)
         self.args = args
         self.padding_idx = dictionary.pad()
-        self.embed_tokens = nn.Embedding(
+        self.embed_tokens = Embedding(
             len(dictionary), args.encoder_embed_dim, self.padding_idx
         )
-        self.embed_positions = nn.Embedding(args.max_seq_len, args.encoder_embed_dim)
+        self.embed_positions = Embedding(args.max_seq_len, args.encoder_embed_dim)
         self.sentence_encoder = TransformerEncoder(
             encoder_layers=args.encoder_layers,
             embed_dim=args.encoder_embed_dim,

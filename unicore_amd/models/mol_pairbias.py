@@ -23,6 +23,7 @@ from unicore_amd.models import (
     register_model_architecture,
 )
 from unicore_amd.modules import RMSNorm, SelfMultiheadAttention, init_bert_params
+from unicore_amd.modules.embedding import Embedding
 
 
 class GaussianPairBias(nn.Module):
@@ -105,7 +106,7 @@ class MolPairBiasModel(BaseUnicoreModel):
         self.args = args
         self.padding_idx = dictionary.pad()
         E = args.encoder_embed_dim
-        self.embed_tokens = nn.Embedding(len(dictionary), E, self.padding_idx)
+        self.embed_tokens = Embedding(len(dictionary), E, self.padding_idx)
         self.pair_bias = GaussianPairBias(
             args.gaussian_kernels, args.encoder_attention_heads
         )

@@ -184,3 +184,11 @@ def dropout_add_fwd(x, res, p, is_training):
 def dropout_add_bwd(grad, dmask, p):
     require_kernels()
     return _kernels.dropout_add_backward(grad, dmask, float(p))
+
+
+def embedding_bwd(grad, indices, num_embeddings, padding_idx):
+    require_kernels()
+    return _kernels.embedding_backward(
+        grad, indices, int(num_embeddings),
+        int(padding_idx) if padding_idx is not None else -1,
+    )
