@@ -82,4 +82,43 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--long-seq", action="store_true")
+    a = ap.parse_args()
+    if a.long_seq:
+        long_seq()
+    else:
+        main()
+
+
+def long_seq():
+    """Long-sequence attention: flash vs materialized as L grows.  The
+    materialized path's L x L scores explode quadratically; flash stays
+    linear in memory — sequence lengths beyond the reference's reach."""
+    from unicore_amd import ops
+    from unicore_amd.modules import softmax_dropout
+
+    H, D = 8, 64
+    for L, B in ((2048, 8), (4096, 4), (8192, 2), (16384, 1)):
+        BH = B * H
+        q = torch.randn(BH, L, D, device="cuda", dtype=torch.bfloat16) * 0.2
+        k, v = torch.randn_like(q), torch.randn_like(q)
+
+        def fl():
+            return ops.flash_attn_fwd(q, k, v, None, 1, None, 1, 0.0, True)
+
+        fl_ms = timeit(fl, iters=10, warmup=3)
+        try:
+            def mat():
+                s = torch.bmm(q, k.transpose(1, 2))
+                attn = softmax_dropout(s.view(B, H, L, L), 0.0, True)
+                return torch.bmm(attn.view(BH, L, L), v)
+
+            mat_ms = timeit(mat, iters=10, warmup=3)
+            mat_s = f"{mat_ms:8.2f} ms"
+        except torch.cuda.OutOfMemoryError:
+            mat_s = "     OOM"
+        torch.cuda.empty_cache()
+        print(f"L={L:6d} B={B}: flash {fl_ms:8.2f} ms   materialized {mat_s}")

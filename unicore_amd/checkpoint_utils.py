@@ -13,7 +13,6 @@ import logging
 import os
 import re
 import shutil
-import time
 import traceback
 
 import torch

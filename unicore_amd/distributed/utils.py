@@ -13,12 +13,10 @@ import logging
 import os
 import pickle
 import random
-import signal
 import socket
 import struct
 import subprocess
 import warnings
-from argparse import Namespace
 from collections import OrderedDict
 from datetime import timedelta
 from typing import Any, Dict, List, Mapping, Optional

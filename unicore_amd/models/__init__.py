@@ -1,7 +1,6 @@
 """Model + architecture registries (parity: reference
 unicore/models/__init__.py:17-102)."""
 
-import argparse
 import importlib
 import os
 

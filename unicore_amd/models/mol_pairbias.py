@@ -16,7 +16,6 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from unicore_amd import utils
 from unicore_amd.models import (
     BaseUnicoreModel,
     register_model,

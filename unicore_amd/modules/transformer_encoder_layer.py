@@ -5,7 +5,6 @@ Functional parity with reference unicore/modules/transformer_encoder_layer.py:56
 
 from typing import Optional
 
-import torch
 import torch.nn.functional as F
 from torch import Tensor, nn
 

@@ -16,8 +16,6 @@ which also serves as the numerics oracle in tests.
 import logging
 import os
 
-import torch
-
 logger = logging.getLogger(__name__)
 
 _kernels = None

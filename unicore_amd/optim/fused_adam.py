@@ -9,8 +9,6 @@ optimizer there are only ~2 flat tensors per step, so kernel-launch count is
 negligible.
 """
 
-import math
-
 import torch
 
 

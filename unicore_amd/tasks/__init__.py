@@ -1,6 +1,5 @@
 """Task registry (parity: reference unicore/tasks/__init__.py:16-61)."""
 
-import argparse
 import importlib
 import os
 

@@ -3,7 +3,7 @@
 import logging
 import os
 from argparse import Namespace
-from typing import Any, Callable, Dict, List
+from typing import Any, Callable, Dict
 
 import torch
 

@@ -11,9 +11,8 @@ import contextlib
 import importlib
 import os
 import sys
-import warnings
 from functools import partial
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, List
 
 import torch
 import torch.nn.functional as F

@@ -9,7 +9,6 @@ import logging
 import math
 import os
 import sys
-import time
 from multiprocessing.pool import ThreadPool
 
 import torch
