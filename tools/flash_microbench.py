@@ -81,16 +81,6 @@ def main():
     print(f"flash peak:        {torch.cuda.max_memory_allocated()/2**30:.2f} GiB")
 
 
-if __name__ == "__main__":
-    import argparse
-
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--long-seq", action="store_true")
-    a = ap.parse_args()
-    if a.long_seq:
-        long_seq()
-    else:
-        main()
 
 
 def long_seq():
@@ -122,3 +112,15 @@ def long_seq():
             mat_s = "     OOM"
         torch.cuda.empty_cache()
         print(f"L={L:6d} B={B}: flash {fl_ms:8.2f} ms   materialized {mat_s}")
+
+
+if __name__ == "__main__":
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--long-seq", action="store_true")
+    a = ap.parse_args()
+    if a.long_seq:
+        long_seq()
+    else:
+        main()
