@@ -111,7 +111,34 @@ def long_seq():
         except torch.cuda.OutOfMemoryError:
             mat_s = "     OOM"
         torch.cuda.empty_cache()
-        print(f"L={L:6d} B={B}: flash {fl_ms:8.2f} ms   materialized {mat_s}")
+        # fwd+bwd via the module-level autograd functions
+        from unicore_amd.modules.multihead_attention import _FlashAttn
+
+        def fl_fb():
+            qq = q.detach().requires_grad_(True)
+            kk = k.detach().requires_grad_(True)
+            vv = v.detach().requires_grad_(True)
+            o = _FlashAttn.apply(qq, kk, vv, None, 1, None, 1, 0.1, True)
+            o.backward(torch.ones_like(o))
+
+        fl_fb_ms = timeit(fl_fb, iters=5, warmup=2)
+        try:
+            def mat_fb():
+                qq = q.detach().requires_grad_(True)
+                kk = k.detach().requires_grad_(True)
+                vv = v.detach().requires_grad_(True)
+                s2 = torch.bmm(qq, kk.transpose(1, 2))
+                attn = softmax_dropout(s2.view(B, H, L, L), 0.1, True)
+                o = torch.bmm(attn.view(BH, L, L), vv)
+                o.backward(torch.ones_like(o))
+
+            mat_fb_ms = timeit(mat_fb, iters=5, warmup=2)
+            mat_fb_s = f"{mat_fb_ms:8.2f} ms"
+        except torch.cuda.OutOfMemoryError:
+            mat_fb_s = "     OOM"
+        torch.cuda.empty_cache()
+        print(f"L={L:6d} B={B}: fwd flash {fl_ms:8.2f} / mat {mat_s} | "
+              f"f+b flash {fl_fb_ms:8.2f} / mat {mat_fb_s}")
 
 
 if __name__ == "__main__":

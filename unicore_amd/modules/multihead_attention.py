@@ -199,10 +199,13 @@ class SelfMultiheadAttention(nn.Module):
         o = None
         if (
             use_fused_split
-            # opt-in for training; always on under no_grad (forward-only
-            # flash is ~30% faster than the materialized chain and never
-            # materializes the L x L score matrix)
-            and (_flash_enabled() or not torch.is_grad_enabled())
+            # opt-in for training at short L; always on under no_grad
+            # (forward-only flash is ~1.4-3.8x the materialized chain) and
+            # for long sequences, where the measured fwd+bwd crossover sits
+            # at L ~= 2048-4096 and the L x L score matrix stops fitting
+            # (tools/flash_microbench.py --long-seq)
+            and (_flash_enabled() or not torch.is_grad_enabled()
+                 or tgt_len >= 4096)
             and not return_attn
             and q.dtype == torch.bfloat16
             and self.head_dim == 64
