@@ -73,6 +73,172 @@ __device__ __forceinline__ void keep_bits8(uint64_t seed, uint64_t subseq,
   keep16x8(seed, subseq, kv0, pthresh, keep);
 }
 
+// V^T-resident forward (L <= 512): whole V^T staged once, kv loop runs
+// barrier-free (same PMC-driven rationale as the dq K-resident variant).
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
+__global__ __launch_bounds__(256) void flash_fwd_vres_kernel(
+    uint16_t* __restrict__ out, float* __restrict__ lse,
+    const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
+    const uint16_t* __restrict__ vp,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int qt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int q0 = qt * BM + wid * 16;
+  const int64_t qbase = (bh * L + q0) * HD;
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_p[4][16][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_vt[HD][LMAX];
+
+  {
+    const int st_kv0 = (int)threadIdx.x >> 2;
+    const int st_d0 = ((int)threadIdx.x & 3) * 16;
+    for (int c = 0; c < L / 64; ++c) {
+      const int kv = st_kv0 + c * 64;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(vp) +
+                (bh * L + kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(vp) +
+                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_vt[d0][kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_vt[d1][kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+  }
+  __syncthreads();
+
+  bf16x8 aq[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+    aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+
+  f32x4 o_acc[4] = {};
+  float m_i[4], l_i[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_i[r] = -INFINITY;
+    l_i[r] = 0.f;
+  }
+  const uint16_t* bias_rows[4];
+  const uint16_t* mask_row = nullptr;
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + lg * 4 + r;
+      bias_rows[r] =
+          bias + (((bh / bias_od) % bias_nb) * bias_q + (q % bias_q)) * (int64_t)L;
+    }
+  }
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    f32x4 s[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      f32x4 acc = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
+      }
+      s[cb] = acc;
+    }
+    if (HAS_BIAS || HAS_MASK) {
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float add = 0.f;
+          if (HAS_BIAS)
+            add += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
+          if (HAS_MASK)
+            add += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+          s[cb][r] += add;
+        }
+      }
+    }
+    float mnew[4], alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
+      mx = rowgroup_max(mx);
+      mnew[r] = fmaxf(m_i[r], mx);
+      alpha[r] = __expf(m_i[r] - mnew[r]);
+      m_i[r] = mnew[r];
+    }
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        s[cb][r] = __expf(s[cb][r] - mnew[r]);
+        psum[r] += s[cb][r];
+      }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      l_i[r] = l_i[r] * alpha[r] + rowgroup_sum(psum[r]);
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) o_acc[cb][r] *= alpha[r];
+    }
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_p[wid][lg * 4 + r][cb * 16 + lr] = f32_to_bf16_bits(s[cb][r]);
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      bf16x8 ap = load_frag(&lds_p[wid][lr][ks2 * 32 + lg * 8]);
+      if constexpr (DROP) {
+        bool keep[8];
+        keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
+                         kv0 + ks2 * 32 + lg * 8, pthresh, keep);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float pv = bf16_bits_to_f32((uint16_t)(unsigned short)ap[j]);
+          pv = keep[j] ? pv * pinv : 0.f;
+          ap[j] = (short)f32_to_bf16_bits(pv);
+        }
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int d = cb * 16 + lr;
+        const bf16x8 bv = load_frag(
+            &lds_vt[d][(kv0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
+        o_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, o_acc[cb], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float inv = 1.0f / l_i[r];
+    const int q = q0 + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+      out[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(o_acc[cb][r] * inv);
+    if (lr == 0) lse[bh * L + q] = m_i[r] + __logf(l_i[r]);
+  }
+}
+
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
 __global__ __launch_bounds__(256) void flash_fwd_kernel(
     uint16_t* __restrict__ out, float* __restrict__ lse,
@@ -338,15 +504,27 @@ std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tenso
   const dim3 grid(L / BM, BH);
 
   auto launch = [&](auto has_bias, auto has_mask, auto dropt) {
-    flash_fwd_kernel<decltype(has_bias)::value, decltype(has_mask)::value,
-                     decltype(dropt)::value><<<grid, 256, 0, stream>>>(
-        reinterpret_cast<uint16_t*>(o.data_ptr()), lse_t.data_ptr<float>(),
-        reinterpret_cast<const uint16_t*>(q.data_ptr()),
-        reinterpret_cast<const uint16_t*>(k.data_ptr()),
-        reinterpret_cast<const uint16_t*>(v.data_ptr()),
-        reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
-        reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L, pinv,
-        pthresh, seed);
+    constexpr bool HB = decltype(has_bias)::value;
+    constexpr bool HM = decltype(has_mask)::value;
+    constexpr bool DR = decltype(dropt)::value;
+    if (L <= 512)
+      flash_fwd_vres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<uint16_t*>(o.data_ptr()), lse_t.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(q.data_ptr()),
+          reinterpret_cast<const uint16_t*>(k.data_ptr()),
+          reinterpret_cast<const uint16_t*>(v.data_ptr()),
+          reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
+          reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L,
+          pinv, pthresh, seed);
+    else
+      flash_fwd_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<uint16_t*>(o.data_ptr()), lse_t.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(q.data_ptr()),
+          reinterpret_cast<const uint16_t*>(k.data_ptr()),
+          reinterpret_cast<const uint16_t*>(v.data_ptr()),
+          reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
+          reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L,
+          pinv, pthresh, seed);
   };
   auto pick = [&](auto has_bias, auto has_mask) {
     if (drop)
@@ -399,6 +577,167 @@ __global__ void flash_dot_do_o_kernel(float* __restrict__ di,
 #pragma unroll
     for (int off = 1; off < 8; off <<= 1) s += __shfl_xor(s, off, 64);
     if (el == 0) di[row] = s;
+  }
+}
+
+// K^T-resident dq variant (L <= 1024): the whole K^T tile is staged once
+// (L*HD bf16 <= 128 KB LDS) so the kv loop runs with NO per-tile barriers —
+// the per-tile barrier+staging serialization is what keeps the tiled dq
+// kernel at ~2%% MFMA utilization (PMC evidence in profiles/).
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
+__global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
+    uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
+    const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int qt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int q0 = qt * BM + wid * 16;
+  const int64_t qbase = (bh * L + q0) * HD;
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][LMAX];
+
+  // stage the WHOLE K^T once: thread covers kv rows tid/4 + 64*c
+  {
+    const int st_kv0 = (int)threadIdx.x >> 2;
+    const int st_d0 = ((int)threadIdx.x & 3) * 16;
+    for (int c = 0; c < L / 64; ++c) {
+      const int kv = st_kv0 + c * 64;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_kt[d0][kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_kt[d1][kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+  }
+  __syncthreads();
+
+  bf16x8 aq[2], ado[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    aq[ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    ado[ks] = load_frag(dop + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+  }
+  float lse_r[4], di_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    lse_r[r] = lse[bh * L + q0 + lg * 4 + r];
+    di_r[r] = di[bh * L + q0 + lg * 4 + r];
+  }
+  const float di_row = di[bh * L + q0 + lr];
+  const uint16_t* bias_rows[4];
+  const uint16_t* mask_row = nullptr;
+  if (HAS_BIAS) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + lg * 4 + r;
+      bias_rows[r] =
+          bias + (((bh / bias_od) % bias_nb) * bias_q + (q % bias_q)) * (int64_t)L;
+    }
+  }
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+
+  f32x4 dq_acc[4] = {};
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    f32x4 s[4], dp[4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      f32x4 acc = {}, accd = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk, acc, 0, 0, 0);
+        const bf16x8 bvt = load_frag(
+            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[ks], bvt, accd, 0, 0, 0);
+      }
+      s[cb] = acc;
+      dp[cb] = accd;
+    }
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = s[cb][r];
+        if (HAS_BIAS)
+          sv += __bfloat162float(
+              reinterpret_cast<const __hip_bfloat16*>(bias_rows[r])[kv]);
+        if (HAS_MASK)
+          sv += __bfloat162float(
+              reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+        s[cb][r] = __expf(sv - lse_r[r]);
+      }
+    }
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        lds_t[wid][0][lg * 4 + r][cb * 16 + lr] = f32_to_bf16_bits(s[cb][r]);
+        lds_t[wid][1][lg * 4 + r][cb * 16 + lr] = f32_to_bf16_bits(dp[cb][r]);
+      }
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      const bf16x8 pa = load_frag(&lds_t[wid][0][lr][ks2 * 32 + lg * 8]);
+      const bf16x8 dpa = load_frag(&lds_t[wid][1][lr][ks2 * 32 + lg * 8]);
+      bool keep[8];
+      keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
+                       kv0 + ks2 * 32 + lg * 8, pthresh, keep);
+      bf16x8 dsa;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float pv = bf16_bits_to_f32((uint16_t)(unsigned short)pa[j]);
+        float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
+        if (DROP) dpv = keep[j] ? dpv * pinv : 0.f;
+        dsa[j] = (short)f32_to_bf16_bits(pv * (dpv - di_row));
+      }
+      if (ds_out != nullptr) {
+        union {
+          bf16x8 v;
+          uint4 u;
+        } U;
+        U.v = dsa;
+        *reinterpret_cast<uint4*>(ds_out + (bh * L + q0 + lr) * (int64_t)L +
+                                  kv0 + ks2 * 32 + lg * 8) = U.u;
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int d = cb * 16 + lr;
+        const bf16x8 bkf = load_frag(
+            &lds_kt[d][(kv0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
+        dq_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bkf, dq_acc[cb], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q = q0 + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+      dq[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dq_acc[cb][r]);
   }
 }
 
@@ -559,6 +898,186 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
     for (int cb = 0; cb < 4; ++cb)
       dq[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
           f32_to_bf16_bits(dq_acc[cb][r]);
+  }
+}
+
+// Q^T/dO^T-hybrid dkv variant (L <= 512): dO^T staged fully (the q loop
+// runs without per-tile dO staging barriers); the P/dS redistribute buffer
+// is shared sequentially (wave-local ordering), keeping the block at
+// <= 80 KB LDS for 2 blocks/CU.
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
+__global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
+    uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
+    const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
+    const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int kt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int kv0w = kt * BN + wid * 16;
+  const int64_t kvbase = (bh * L + kv0w) * HD;
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][16][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][LMAX];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_qt[HD][BM];
+  const int st_q0 = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
+
+  // dO^T resident (the dominant reuse); Q^T staged per q-tile below
+  {
+    for (int c = 0; c < L / 64; ++c) {
+      const int qq = st_q0 + c * 64;
+      const int64_t row = (bh * L + qq) * (int64_t)HD;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_dot[d0][qq ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_dot[d1][qq ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+  }
+  __syncthreads();
+
+  bf16x8 ak[2], av[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    ak[ks] = load_frag(kp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    av[ks] = load_frag(vp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+  }
+  const uint16_t* mask_row = nullptr;
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+  float maskv[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    maskv[r] = HAS_MASK
+                   ? __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+                         mask_row)[kv0w + lg * 4 + r])
+                   : 0.f;
+
+  f32x4 dk_acc[4] = {}, dv_acc[4] = {};
+  const int n_tiles = L / BM;
+  for (int tq = 0; tq < n_tiles; ++tq) {
+    const int q0 = tq * BM;
+    // stage this q-tile's Q^T (8 KB)
+    {
+      const int qq = q0 + st_q0;
+      const int64_t row = (bh * L + qq) * (int64_t)HD;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_qt[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_qt[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      f32x4 acc = {}, accd = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bq =
+            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq, acc, 0, 0, 0);
+        const bf16x8 bdo =
+            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo, accd, 0, 0, 0);
+      }
+      st[cq] = acc;
+      dpt[cq] = accd;
+    }
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      const float lse_c = lse[bh * L + qcol];
+      const float di_c = di[bh * L + qcol];
+      const uint16_t* brow =
+          HAS_BIAS ? bias + (((bh / bias_od) % bias_nb) * bias_q +
+                             (qcol % bias_q)) * (int64_t)L
+                   : nullptr;
+      bool keep[4] = {true, true, true, true};
+      if (DROP) {
+        bool k8[8];
+        const int blk = (kv0w + lg * 4) & ~7;
+        keep16x8(seed, (uint64_t)(bh * L + qcol), blk, pthresh, k8);
+        const int off = (kv0w + lg * 4) - blk;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) keep[r] = k8[off + r];
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = st[cq][r];
+        if (HAS_BIAS)
+          sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+              brow)[kv0w + lg * 4 + r]);
+        sv += maskv[r];
+        const float pv = __expf(sv - lse_c);
+        float dpv = dpt[cq][r];
+        if (DROP) dpv = keep[r] ? dpv * pinv : 0.f;
+        st[cq][r] = DROP ? (keep[r] ? pv * pinv : 0.f) : pv;
+        dpt[cq][r] = pv * (dpv - di_c);
+      }
+    }
+    // redistribute P^T first (shared buffer, wave-local ordering), read
+    // both A-fragments into registers, then reuse the buffer for dS^T
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_t[wid][lg * 4 + r][cq * 16 + lr] = f32_to_bf16_bits(st[cq][r]);
+    bf16x8 pta[2];
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2)
+      pta[ks2] = load_frag(&lds_t[wid][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_t[wid][lg * 4 + r][cq * 16 + lr] = f32_to_bf16_bits(dpt[cq][r]);
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      const bf16x8 dsta = load_frag(&lds_t[wid][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int d = cb * 16 + lr;
+        const int qx_dot = (q0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3);
+        const int qx_q = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);  // tile-local
+        const bf16x8 bdo = load_frag(&lds_dot[d][qx_dot]);
+        const bf16x8 bqf = load_frag(&lds_qt[d][qx_q]);
+        dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta[ks2], bdo,
+                                                            dv_acc[cb], 0, 0, 0);
+        dk_acc[cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsta, bqf, dk_acc[cb], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kv = kv0w + lg * 4 + r;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      dk[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dk_acc[cb][r]);
+      dv[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+          f32_to_bf16_bits(dv_acc[cb][r]);
+    }
   }
 }
 
@@ -767,7 +1286,19 @@ std::vector<at::Tensor> flash_attn_backward(
     constexpr bool HB = decltype(hb)::value;
     constexpr bool HM = decltype(hm)::value;
     constexpr bool DR = decltype(dr)::value;
-    flash_bwd_dq_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
+    if (L <= 512)
+      flash_bwd_dq_kres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<uint16_t*>(dq.data_ptr()),
+          ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
+          reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+          reinterpret_cast<const uint16_t*>(q.data_ptr()),
+          reinterpret_cast<const uint16_t*>(k.data_ptr()),
+          reinterpret_cast<const uint16_t*>(v.data_ptr()), lse.data_ptr<float>(),
+          di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb,
+          bd.q, bd.od, reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q,
+          md.od, L, pinv, pthresh, seed);
+    else
+      flash_bwd_dq_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
         reinterpret_cast<uint16_t*>(dq.data_ptr()),
         ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
         reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
@@ -777,7 +1308,19 @@ std::vector<at::Tensor> flash_attn_backward(
         di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb,
         bd.q, bd.od, reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q,
         md.od, L, pinv, pthresh, seed);
-    flash_bwd_dkv_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
+    if (L <= 512)
+      flash_bwd_dkv_qres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<uint16_t*>(dk.data_ptr()),
+          reinterpret_cast<uint16_t*>(dv.data_ptr()),
+          reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+          reinterpret_cast<const uint16_t*>(q.data_ptr()),
+          reinterpret_cast<const uint16_t*>(k.data_ptr()),
+          reinterpret_cast<const uint16_t*>(v.data_ptr()), lse.data_ptr<float>(),
+          di.data_ptr<float>(), reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb,
+          bd.q, bd.od, reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q,
+          md.od, L, pinv, pthresh, seed);
+    else
+      flash_bwd_dkv_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
         reinterpret_cast<uint16_t*>(dk.data_ptr()),
         reinterpret_cast<uint16_t*>(dv.data_ptr()),
         reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
