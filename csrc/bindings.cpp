@@ -43,6 +43,11 @@ std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double
 at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p);
 at::Tensor embedding_backward(at::Tensor grad, at::Tensor indices,
                               int64_t num_embeddings, int64_t padding_idx);
+std::vector<at::Tensor> cross_entropy_forward(at::Tensor logits, at::Tensor target,
+                                              int64_t ignore_index);
+at::Tensor cross_entropy_backward(at::Tensor logits, at::Tensor target,
+                                  at::Tensor lse, at::Tensor grad_scale,
+                                  int64_t ignore_index);
 std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
                                            std::optional<at::Tensor> bias,
                                            int64_t bias_outer_div,
@@ -91,4 +96,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused dropout + residual add backward (dx only; d_res = grad)");
   m.def("embedding_backward", &embedding_backward,
         "atomic-scatter embedding gradient (fp32 accumulate)");
+  m.def("cross_entropy_forward", &cross_entropy_forward,
+        "fused online-logsumexp token cross entropy -> (loss, lse)");
+  m.def("cross_entropy_backward", &cross_entropy_backward,
+        "cross entropy backward (softmax - onehot, no materialized fp32)");
 }

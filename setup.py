@@ -32,6 +32,7 @@ HIP_SOURCES = [
     "csrc/flash_attn.hip",
     "csrc/dropout_add.hip",
     "csrc/embedding.hip",
+    "csrc/cross_entropy.hip",
 ]
 
 setup(

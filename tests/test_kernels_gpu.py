@@ -468,3 +468,30 @@ def test_embedding_module_matches_torch():
     rout.backward(g)
     assert torch.equal(out, rout)
     assert (emb.weight.grad - ref.weight.grad).abs().max().item() < 1e-4
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("V", [30523, 4096])
+def test_fused_cross_entropy_parity(dtype, V):
+    _kernels()
+    from unicore_amd.modules.cross_entropy import fused_nll_loss
+
+    torch.manual_seed(2)
+    N = 777
+    logits = (torch.randn(N, V, device="cuda", dtype=dtype) * 2).requires_grad_(True)
+    target = torch.randint(0, V, (N,), device="cuda")
+    target[::11] = 1  # ignore_index hits
+    loss = fused_nll_loss(logits, target, ignore_index=1)
+    loss.backward(torch.tensor(0.7, device="cuda"))
+
+    lr = logits.detach().float().clone().requires_grad_(True)
+    ref = F.nll_loss(
+        F.log_softmax(lr, dim=-1, dtype=torch.float32), target,
+        ignore_index=1, reduction="sum",
+    )
+    ref.backward(torch.tensor(0.7, device="cuda"))
+    rel = abs(loss.item() - ref.item()) / (abs(ref.item()) + 1e-6)
+    assert rel < 1e-3, (loss.item(), ref.item())
+    gd = (logits.grad.float() - lr.grad).abs().max().item()
+    assert gd < (1e-5 if dtype == torch.float32 else 5e-3), gd

@@ -7,6 +7,7 @@ import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss
+from unicore_amd.modules.cross_entropy import fused_nll_loss
 
 
 @register_loss("masked_lm")
@@ -35,12 +36,7 @@ class MaskedLMLoss(UnicoreLoss):
         target = sample["target"]
         if masked_tokens is not None:
             target = target[masked_tokens]
-        loss = F.nll_loss(
-            F.log_softmax(logits, dim=-1, dtype=torch.float32),
-            target,
-            ignore_index=self.padding_idx,
-            reduction="sum",
-        )
+        loss = fused_nll_loss(logits, target, ignore_index=self.padding_idx)
         logging_output = {
             "loss": loss.data,
             "bsz": sample["target"].size(0),

@@ -8,6 +8,7 @@ import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss
+from unicore_amd.modules.cross_entropy import fused_nll_loss
 
 
 @register_loss("masked_msa")
@@ -22,12 +23,7 @@ class MaskedMSALoss(UnicoreLoss):
         masked = target.ne(self.padding_idx)
         sample_size = masked.int().sum()
         masked = torch.where(masked.any(), masked, masked.new([True]))
-        loss = F.nll_loss(
-            F.log_softmax(logits[masked], dim=-1, dtype=torch.float32),
-            target[masked],
-            ignore_index=self.padding_idx,
-            reduction="sum",
-        )
+        loss = fused_nll_loss(logits[masked], target[masked], ignore_index=self.padding_idx)
         logging_output = {
             "loss": loss.data,
             "bsz": target.size(0),

@@ -10,6 +10,7 @@ import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss
+from unicore_amd.modules.cross_entropy import fused_nll_loss
 
 
 @register_loss("mol_pretrain")
@@ -25,12 +26,7 @@ class MolPretrainLoss(UnicoreLoss):
         masked = target.ne(self.padding_idx)
         sample_size = masked.int().sum()
         masked = torch.where(masked.any(), masked, masked.new([True]))
-        token_loss = F.nll_loss(
-            F.log_softmax(logits[masked], dim=-1, dtype=torch.float32),
-            target[masked],
-            ignore_index=self.padding_idx,
-            reduction="sum",
-        )
+        token_loss = fused_nll_loss(logits[masked], target[masked], ignore_index=self.padding_idx)
         # coordinate denoising: predict the delta back to clean positions
         coord_target = sample["coord_target"].float()
         atom_mask = sample["net_input"]["src_tokens"].ne(self.padding_idx)

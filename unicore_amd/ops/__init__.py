@@ -190,3 +190,15 @@ def embedding_bwd(grad, indices, num_embeddings, padding_idx):
         grad, indices, int(num_embeddings),
         int(padding_idx) if padding_idx is not None else -1,
     )
+
+
+def cross_entropy_fwd(logits, target, ignore_index):
+    require_kernels()
+    return _kernels.cross_entropy_forward(logits, target, int(ignore_index))
+
+
+def cross_entropy_bwd(logits, target, lse, grad_scale, ignore_index):
+    require_kernels()
+    return _kernels.cross_entropy_backward(
+        logits, target, lse, grad_scale, int(ignore_index)
+    )
