@@ -20,9 +20,12 @@ __device__ __forceinline__ void lse_combine(float& m, float& s, float m2,
   if (m2 > m) {
     s = s * __expf(m - m2) + s2;
     m = m2;
-  } else {
+  } else if (m2 != -INFINITY) {
     s = s + s2 * __expf(m2 - m);
   }
+  // m2 == -INFINITY and m2 <= m: empty state, nothing to add (guards the
+  // exp(-inf - -inf) = NaN case when V < blockDim and some lanes saw no
+  // elements)
 }
 
 template <typename T, bool VEC8>

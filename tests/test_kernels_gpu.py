@@ -472,7 +472,7 @@ def test_embedding_module_matches_torch():
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", DTYPES)
-@pytest.mark.parametrize("V", [30523, 4096])
+@pytest.mark.parametrize("V", [31, 30523, 4096])
 def test_fused_cross_entropy_parity(dtype, V):
     _kernels()
     from unicore_amd.modules.cross_entropy import fused_nll_loss
