@@ -161,3 +161,24 @@ def test_bf16_checkpoint_resume_continuity(tmp_path):
         assert losses[:3] == losses_full[:3]
         for a, b in zip(losses[3:], losses_full[3:]):
             assert abs(a - b) < 2e-2, (losses, losses_full)
+
+
+@requires_gpu
+def test_fp16_overflow_backoff_and_recovery():
+    """With an absurd initial loss scale the first steps overflow; the
+    scaler must halve repeatedly (skipping updates) until training
+    proceeds with finite grads (reference dynamic_loss_scaler contract)."""
+    trainer, epoch_itr = _build_trainer(
+        ["--fp16", "--fp16-init-scale", str(2**28)], seed=5
+    )
+    init_scale = trainer.optimizer.scaler.loss_scale
+    itr = epoch_itr.next_epoch_itr(shuffle=False)
+    batches = list(itr)
+    outputs = []
+    for i in range(10):
+        outputs.append(trainer.train_step([batches[i % len(batches)]]))
+    torch.cuda.synchronize()
+    assert trainer.optimizer.scaler.loss_scale < init_scale
+    # at least one overflow skip (None) and at least one successful update
+    assert any(o is None for o in outputs)
+    assert trainer.get_num_updates() > 0
