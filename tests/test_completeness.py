@@ -185,3 +185,33 @@ def test_fp16_optimizer_allreduce_fp32_flag_cpu():
     fopt.clip_grad_norm(1.0)
     fopt.step()
     assert all(torch.isfinite(p.detach().float()).all() for p in model.parameters())
+
+
+def test_reset_flags_on_resume(tmp_path, monkeypatch):
+    """--reset-optimizer / --reset-lr-scheduler / --reset-meters /
+    --reset-dataloader start those components fresh while keeping weights."""
+    from unicore_cli import train as train_cli
+
+    base = [
+        "--task", "bert_synthetic", "--arch", "bert_base",
+        "--loss", "masked_lm", "--optimizer", "adam",
+        "--lr-scheduler", "fixed", "--lr", "1e-4",
+        "--batch-size", "4", "--dataset-size", "16",
+        "--tokens-per-sample", "16", "--vocab-size", "64",
+        "--encoder-layers", "1", "--encoder-embed-dim", "32",
+        "--encoder-ffn-embed-dim", "64", "--encoder-attention-heads", "2",
+        "--log-format", "none", "--cpu", "--num-workers", "0",
+    ]
+    d = str(tmp_path / "rr")
+    monkeypatch.setattr(sys, "argv", ["t"] + base + [
+        "--save-dir", d, "--max-update", "4"])
+    train_cli.cli_main()
+
+    monkeypatch.setattr(sys, "argv", ["t"] + base + [
+        "--save-dir", d, "--max-update", "2",
+        "--reset-optimizer", "--reset-lr-scheduler", "--reset-meters",
+        "--reset-dataloader"])
+    train_cli.cli_main()
+    st = torch.load(os.path.join(d, "checkpoint_last.pt"), weights_only=False)
+    # optimizer history restarts counting
+    assert st["optimizer_history"][-1]["num_updates"] == 2

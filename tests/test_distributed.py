@@ -122,6 +122,18 @@ def _collectives_worker(rank, world, port, out_dir):
     gathered = dutils.all_gather_list({"rank": rank, "x": [rank] * 3}, group=group)
     assert [g["rank"] for g in gathered] == [0, 1]
 
+    # all_to_all on a 1-D tensor
+    t = torch.arange(4, dtype=torch.float32) + rank * 10
+    out_t = dutils.all_to_all(t, group=group)
+    expect = torch.tensor(
+        [0.0, 1.0, 10.0, 11.0] if rank == 0 else [2.0, 3.0, 12.0, 13.0]
+    )
+    assert torch.equal(out_t, expect), (rank, out_t)
+
+    # all_gather returning a stacked tensor
+    g = dutils.all_gather(torch.full((2,), float(rank)), group, return_tensor=True)
+    assert g.shape[0] == 2
+
     # broadcast_object with tensors inside
     obj = {"t": torch.full((3,), float(rank)), "s": f"from{rank}"} if rank == 0 else None
     got = dutils.broadcast_object(obj, src_rank=0, group=group)
