@@ -77,6 +77,9 @@ class BertSyntheticTask(UnicoreTask):
         self.dictionary = dictionary
         self.seed = args.seed
         self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
+        # vocab to a multiple of 64: the lm-head GEMM and fused cross
+        # entropy want an even, vectorizable inner dimension
+        dictionary.pad_to_multiple_(64)
 
     @classmethod
     def setup_task(cls, args, **kwargs):
@@ -89,8 +92,9 @@ class BertSyntheticTask(UnicoreTask):
         raw = SyntheticTokensDataset(
             size=self.args.dataset_size,
             seq_len=seq_len,
-            # sample below the [MASK] symbol (always the last added)
-            vocab_size=len(self.dictionary) - 1,
+            # sample strictly below the [MASK] symbol (filler symbols used
+            # to pad the vocab to a multiple of 64 sit above it)
+            vocab_size=self.mask_idx,
             seed=self.seed + (0 if split == "train" else 1),
             fixed_len=not getattr(self.args, "variable_seq_len", False),
         )

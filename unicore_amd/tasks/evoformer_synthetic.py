@@ -44,7 +44,7 @@ class SyntheticMSADataset(UnicoreDataset):
     def __getitem__(self, index):
         with data_utils.numpy_seed(self.seed, self.epoch, index):
             toks = np.random.randint(
-                5, len(self.dictionary) - 1, size=(self.n_seq, self.seq_len)
+                5, self.mask_idx, size=(self.n_seq, self.seq_len)
             )
             mask = np.random.rand(self.n_seq, self.seq_len) < self.mask_prob
         target = np.full_like(toks, self.dictionary.pad())
@@ -92,6 +92,7 @@ class EvoformerSyntheticTask(UnicoreTask):
         self.dictionary = dictionary
         self.seed = args.seed
         self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
+        dictionary.pad_to_multiple_(64)
 
     @classmethod
     def setup_task(cls, args, **kwargs):

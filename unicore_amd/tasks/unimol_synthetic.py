@@ -56,7 +56,7 @@ class SyntheticMolDataset(UnicoreDataset):
     def __getitem__(self, index):
         with data_utils.numpy_seed(self.seed, self.epoch, index):
             n = self.n_atoms
-            toks = np.random.randint(5, len(self.dictionary) - 1, size=n)
+            toks = np.random.randint(5, self.mask_idx, size=n)
             coord = np.random.randn(n, 3).astype(np.float32) * 2.0
             noise = (np.random.randn(n, 3).astype(np.float32) * self.noise_std)
             mask = np.random.rand(n) < self.mask_prob
@@ -104,6 +104,7 @@ class UniMolSyntheticTask(UnicoreTask):
         self.dictionary = dictionary
         self.seed = args.seed
         self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
+        dictionary.pad_to_multiple_(64)
 
     @classmethod
     def setup_task(cls, args, **kwargs):
