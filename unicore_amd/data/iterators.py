@@ -384,6 +384,12 @@ class EpochBatchIterator(EpochBatchIterating):
         if self.buffer_size > 0:
             itr = BufferedIterator(self.buffer_size, itr)
 
+        # Stage pinned batches onto the GPU one step ahead on a dedicated
+        # copy stream.  Placed INSIDE the CountingIterator so resumable
+        # iteration counts consumed batches, not prefetched ones.
+        if torch.cuda.is_available():
+            itr = CudaPrefetcher(itr)
+
         # Wrap with CountingIterator
         itr = CountingIterator(itr, start=offset)
         return itr

@@ -186,11 +186,6 @@ def train(args, trainer, task, epoch_itr, ckp_copy_thread):
         if epoch_itr.epoch <= len(args.update_freq)
         else args.update_freq[-1]
     )
-    if torch.cuda.is_available() and not args.cpu:
-        # stage pinned batches onto the GPU one step ahead on a dedicated
-        # copy stream (hipMemcpyAsync overlapped with compute); the compute
-        # stream waits on the copy event, not the host
-        itr = iterators.CudaPrefetcher(itr)
     itr = iterators.GroupedIterator(itr, update_freq)
     progress = progress_bar.progress_bar(
         itr,
