@@ -741,6 +741,199 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
   }
 }
 
+// 32-row-per-wave K-resident dq (L % 128 == 0, L <= 1024): each wave owns
+// TWO 16-row M-tiles, so every K/V B-fragment load feeds two MFMAs and the
+// per-tile scalar overheads (lse/di/bias addressing, philox) amortize.
+// P/dS redistribute shares one LDS buffer sequentially (wave-local order).
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
+__global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
+    uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
+    const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int qt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int q0 = qt * 128 + wid * 32;   // wave's first q row (2 M-tiles)
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][LMAX];
+
+  {
+    const int st_kv0 = (int)threadIdx.x >> 2;
+    const int st_d0 = ((int)threadIdx.x & 3) * 16;
+    for (int c = 0; c < L / 64; ++c) {
+      const int kv = st_kv0 + c * 64;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_kt[d0][kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_kt[d1][kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+  }
+  __syncthreads();
+
+  bf16x8 aq[2][2], ado[2][2];
+  float lse_r[2][4], di_r[2][4], di_row[2];
+  const uint16_t* bias_rows[2][4];
+  const uint16_t* mask_row = nullptr;
+#pragma unroll
+  for (int mtile = 0; mtile < 2; ++mtile) {
+    const int64_t qbase = (bh * L + q0 + mtile * 16) * (int64_t)HD;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      aq[mtile][ks] = load_frag(qp + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+      ado[mtile][ks] = load_frag(dop + qbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + mtile * 16 + lg * 4 + r;
+      lse_r[mtile][r] = lse[bh * L + q];
+      di_r[mtile][r] = di[bh * L + q];
+      if (HAS_BIAS)
+        bias_rows[mtile][r] =
+            bias +
+            (((bh / bias_od) % bias_nb) * bias_q + (q % bias_q)) * (int64_t)L;
+    }
+    di_row[mtile] = di[bh * L + q0 + mtile * 16 + lr];
+  }
+  if (HAS_MASK)
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+
+  f32x4 dq_acc[2][4] = {};
+  const int n_tiles = L / BN;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * BN;
+    f32x4 s[2][4], dp[2][4];
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb) {
+      f32x4 acc0 = {}, acc1 = {}, accd0 = {}, accd1 = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        const bf16x8 bvt = load_frag(
+            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][ks], bk, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][ks], bk, acc1, 0, 0, 0);
+        accd0 =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[0][ks], bvt, accd0, 0, 0, 0);
+        accd1 =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[1][ks], bvt, accd1, 0, 0, 0);
+      }
+      s[0][cb] = acc0;
+      s[1][cb] = acc1;
+      dp[0][cb] = accd0;
+      dp[1][cb] = accd1;
+    }
+#pragma unroll
+    for (int mtile = 0; mtile < 2; ++mtile)
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int kv = kv0 + cb * 16 + lr;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float sv = s[mtile][cb][r];
+          if (HAS_BIAS)
+            sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+                bias_rows[mtile][r])[kv]);
+          if (HAS_MASK)
+            sv += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
+          s[mtile][cb][r] = __expf(sv - lse_r[mtile][r]);
+        }
+      }
+    // redistribute P (both M-tiles), read A-fragments, then reuse for dS
+#pragma unroll
+    for (int mtile = 0; mtile < 2; ++mtile)
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          lds_t[wid][mtile][lg * 4 + r][cb * 16 + lr] =
+              f32_to_bf16_bits(s[mtile][cb][r]);
+    bf16x8 pa[2][2];
+#pragma unroll
+    for (int mtile = 0; mtile < 2; ++mtile)
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2)
+        pa[mtile][ks2] = load_frag(&lds_t[wid][mtile][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+    for (int mtile = 0; mtile < 2; ++mtile)
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          lds_t[wid][mtile][lg * 4 + r][cb * 16 + lr] =
+              f32_to_bf16_bits(dp[mtile][cb][r]);
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      bool keep[2][8];
+      bf16x8 dsa[2];
+#pragma unroll
+      for (int mtile = 0; mtile < 2; ++mtile) {
+        const bf16x8 dpa =
+            load_frag(&lds_t[wid][mtile][lr][ks2 * 32 + lg * 8]);
+        keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + mtile * 16 + lr),
+                         kv0 + ks2 * 32 + lg * 8, pthresh, keep[mtile]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float pv =
+              bf16_bits_to_f32((uint16_t)(unsigned short)pa[mtile][ks2][j]);
+          float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
+          if (DROP) dpv = keep[mtile][j] ? dpv * pinv : 0.f;
+          dsa[mtile][j] = (short)f32_to_bf16_bits(pv * (dpv - di_row[mtile]));
+        }
+        if (ds_out != nullptr) {
+          union {
+            bf16x8 v;
+            uint4 u;
+          } U;
+          U.v = dsa[mtile];
+          *reinterpret_cast<uint4*>(
+              ds_out + (bh * L + q0 + mtile * 16 + lr) * (int64_t)L + kv0 +
+              ks2 * 32 + lg * 8) = U.u;
+        }
+      }
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int d = cb * 16 + lr;
+        const bf16x8 bkf = load_frag(
+            &lds_kt[d][(kv0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
+        dq_acc[0][cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[0], bkf, dq_acc[0][cb], 0, 0, 0);
+        dq_acc[1][cb] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[1], bkf, dq_acc[1][cb], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int mtile = 0; mtile < 2; ++mtile)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int q = q0 + mtile * 16 + lg * 4 + r;
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb)
+        dq[(bh * L + q) * (int64_t)HD + cb * 16 + lr] =
+            f32_to_bf16_bits(dq_acc[mtile][cb][r]);
+    }
+}
+
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
@@ -905,6 +1098,214 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
 // runs without per-tile dO staging barriers); the P/dS redistribute buffer
 // is shared sequentially (wave-local ordering), keeping the block at
 // <= 80 KB LDS for 2 blocks/CU.
+// Q^T/dO^T-hybrid dkv variant (L <= 512): dO^T staged fully (the q loop
+// runs without per-tile dO staging barriers); the P/dS redistribute buffer
+// is shared sequentially (wave-local ordering), keeping the block at
+// <= 80 KB LDS for 2 blocks/CU.
+// 32-kv-row-per-wave dkv (L % 128 == 0, L <= 512): two kv M-tiles per
+// wave — every Q/dO B-fragment and the staged lds_qt/lds_dot reads feed
+// two MFMAs; per-q-tile staging barriers amortize over 2x the math.
+template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
+__global__ __launch_bounds__(256) void flash_bwd_dkv_q32_kernel(
+    uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
+    const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
+    const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
+    const float* __restrict__ lse, const float* __restrict__ di,
+    const uint16_t* __restrict__ bias, int64_t bias_nb, int bias_q, int64_t bias_od,
+    const uint16_t* __restrict__ mask, int64_t mask_nb, int mask_q, int64_t mask_od,
+    int L, float pinv, uint32_t pthresh, uint64_t seed) {
+  const int kt = blockIdx.x;
+  const int64_t bh = blockIdx.y;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lg = lane >> 4;
+  const int lr = lane & 15;
+  const int kv0w = kt * 128 + wid * 32;   // wave's first kv row (2 M-tiles)
+
+  __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][LMAX];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_qt[HD][BM];
+  const int st_q0 = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
+
+  {
+    for (int c = 0; c < L / 64; ++c) {
+      const int qq = st_q0 + c * 64;
+      const int64_t row = (bh * L + qq) * (int64_t)HD;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_dot[d0][qq ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_dot[d1][qq ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+  }
+  __syncthreads();
+
+  bf16x8 ak[2][2], av[2][2];
+  float maskv[2][4];
+#pragma unroll
+  for (int mt = 0; mt < 2; ++mt) {
+    const int64_t kvbase = (bh * L + kv0w + mt * 16) * (int64_t)HD;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      ak[mt][ks] = load_frag(kp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+      av[mt][ks] = load_frag(vp + kvbase + (int64_t)lr * HD + ks * 32 + lg * 8);
+    }
+  }
+  const uint16_t* mask_row = nullptr;
+  if (HAS_MASK) {
+    mask_row = mask + (((bh / mask_od) % mask_nb) * mask_q) * (int64_t)L;
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        maskv[mt][r] = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+            mask_row)[kv0w + mt * 16 + lg * 4 + r]);
+  } else {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) maskv[mt][r] = 0.f;
+  }
+
+  f32x4 dk_acc[2][4] = {}, dv_acc[2][4] = {};
+  const int n_tiles = L / BM;
+  for (int tq = 0; tq < n_tiles; ++tq) {
+    const int q0 = tq * BM;
+    {
+      const int qq = q0 + st_q0;
+      const int64_t row = (bh * L + qq) * (int64_t)HD;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_qt[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_qt[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
+    f32x4 st[2][4], dpt[2][4];
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      f32x4 a0 = {}, a1 = {}, d0 = {}, d1 = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 bq =
+            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        const bf16x8 bdo =
+            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        a0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[0][ks], bq, a0, 0, 0, 0);
+        a1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[1][ks], bq, a1, 0, 0, 0);
+        d0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[0][ks], bdo, d0, 0, 0, 0);
+        d1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[1][ks], bdo, d1, 0, 0, 0);
+      }
+      st[0][cq] = a0;
+      st[1][cq] = a1;
+      dpt[0][cq] = d0;
+      dpt[1][cq] = d1;
+    }
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
+      const float lse_c = lse[bh * L + qcol];
+      const float di_c = di[bh * L + qcol];
+      const uint16_t* brow =
+          HAS_BIAS ? bias + (((bh / bias_od) % bias_nb) * bias_q +
+                             (qcol % bias_q)) * (int64_t)L
+                   : nullptr;
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt) {
+        bool k8[8];
+        const int kvrow0 = kv0w + mt * 16 + lg * 4;
+        keep_bits8<DROP>(seed, (uint64_t)(bh * L + qcol), kvrow0 & ~7, pthresh,
+                         k8);
+        const int koff = kvrow0 & 7;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float sv = st[mt][cq][r];
+          if (HAS_BIAS)
+            sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+                brow)[kv0w + mt * 16 + lg * 4 + r]);
+          sv += maskv[mt][r];
+          const float pv = __expf(sv - lse_c);
+          float dpv = dpt[mt][cq][r];
+          const bool kp_ = !DROP || k8[koff + r];
+          if (DROP) dpv = kp_ ? dpv * pinv : 0.f;
+          st[mt][cq][r] = DROP ? (kp_ ? pv * pinv : 0.f) : pv;
+          dpt[mt][cq][r] = pv * (dpv - di_c);
+        }
+      }
+    }
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          lds_t[wid][mt][lg * 4 + r][cq * 16 + lr] = f32_to_bf16_bits(st[mt][cq][r]);
+    bf16x8 pta[2][2];
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2)
+        pta[mt][ks2] = load_frag(&lds_t[wid][mt][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          lds_t[wid][mt][lg * 4 + r][cq * 16 + lr] =
+              f32_to_bf16_bits(dpt[mt][cq][r]);
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      bf16x8 dsta[2];
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+        dsta[mt] = load_frag(&lds_t[wid][mt][lr][ks2 * 32 + lg * 8]);
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        const int d = cb * 16 + lr;
+        const int qx_dot = (q0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3);
+        const int qx_q = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);
+        const bf16x8 bdo = load_frag(&lds_dot[d][qx_dot]);
+        const bf16x8 bqf = load_frag(&lds_qt[d][qx_q]);
+        dv_acc[0][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta[0][ks2], bdo,
+                                                               dv_acc[0][cb], 0, 0, 0);
+        dv_acc[1][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta[1][ks2], bdo,
+                                                               dv_acc[1][cb], 0, 0, 0);
+        dk_acc[0][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsta[0], bqf,
+                                                               dk_acc[0][cb], 0, 0, 0);
+        dk_acc[1][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsta[1], bqf,
+                                                               dk_acc[1][cb], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int kv = kv0w + mt * 16 + lg * 4 + r;
+#pragma unroll
+      for (int cb = 0; cb < 4; ++cb) {
+        dk[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+            f32_to_bf16_bits(dk_acc[mt][cb][r]);
+        dv[(bh * L + kv) * (int64_t)HD + cb * 16 + lr] =
+            f32_to_bf16_bits(dv_acc[mt][cb][r]);
+      }
+    }
+}
+
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
 __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
@@ -1286,7 +1687,21 @@ std::vector<at::Tensor> flash_attn_backward(
     constexpr bool HB = decltype(hb)::value;
     constexpr bool HM = decltype(hm)::value;
     constexpr bool DR = decltype(dr)::value;
-    if (L <= 512)
+    if (L <= 512 && L % 128 == 0)
+      flash_bwd_dq_k32_kernel<HB, HM, DR, 512>
+          <<<dim3(L / 128, BH), 256, 0, stream>>>(
+              reinterpret_cast<uint16_t*>(dq.data_ptr()),
+              ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr())
+                           : nullptr,
+              reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
+              reinterpret_cast<const uint16_t*>(q.data_ptr()),
+              reinterpret_cast<const uint16_t*>(k.data_ptr()),
+              reinterpret_cast<const uint16_t*>(v.data_ptr()),
+              lse.data_ptr<float>(), di.data_ptr<float>(),
+              reinterpret_cast<const uint16_t*>(bd.ptr), bd.nb, bd.q, bd.od,
+              reinterpret_cast<const uint16_t*>(md.ptr), md.nb, md.q, md.od, L,
+              pinv, pthresh, seed);
+    else if (L <= 512)
       flash_bwd_dq_kres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
           reinterpret_cast<uint16_t*>(dq.data_ptr()),
           ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
