@@ -37,12 +37,14 @@ class GaussianPairBias(nn.Module):
         self.out = nn.Linear(n_kernels, n_heads)
 
     def forward(self, coords, padding_mask=None):
-        # coords: (B, L, 3)
-        dist = torch.cdist(coords.float(), coords.float())  # (B, L, L)
-        x = dist.unsqueeze(-1) - self.means.float().view(1, 1, 1, -1)
-        std = self.stds.float().abs() + 1e-3
+        # coords: (B, L, 3); distances in fp32 for stability, the gaussian
+        # basis in the model dtype (the (B, L, L, K) tensor is the biggest
+        # intermediate of the model — fp32 here doubled its HBM traffic)
+        dtype = self.out.weight.dtype
+        dist = torch.cdist(coords.float(), coords.float()).to(dtype)
+        x = dist.unsqueeze(-1) - self.means.to(dtype).view(1, 1, 1, -1)
+        std = (self.stds.abs() + 1e-3).to(dtype)
         g = torch.exp(-0.5 * (x / std) ** 2)  # (B, L, L, K)
-        g = g.to(self.out.weight.dtype)
         bias = self.out(g)  # (B, L, L, H)
         bias = bias.permute(0, 3, 1, 2).contiguous()  # (B, H, L, L)
         if padding_mask is not None:
