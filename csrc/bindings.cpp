@@ -48,6 +48,11 @@ std::vector<at::Tensor> cross_entropy_forward(at::Tensor logits, at::Tensor targ
 at::Tensor cross_entropy_backward(at::Tensor logits, at::Tensor target,
                                   at::Tensor lse, at::Tensor grad_scale,
                                   int64_t ignore_index);
+at::Tensor gaussian_basis_forward(at::Tensor coords, at::Tensor means,
+                                  at::Tensor stds, at::ScalarType out_dtype);
+std::vector<at::Tensor> gaussian_basis_backward(at::Tensor dg, at::Tensor coords,
+                                                at::Tensor means, at::Tensor stds);
+bool gaussian_basis_supported(int64_t K);
 std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
                                            std::optional<at::Tensor> bias,
                                            int64_t bias_outer_div,
@@ -98,6 +103,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "atomic-scatter embedding gradient (fp32 accumulate)");
   m.def("cross_entropy_forward", &cross_entropy_forward,
         "fused online-logsumexp token cross entropy -> (loss, lse)");
+  m.def("gaussian_basis_forward", &gaussian_basis_forward,
+        "fused gaussian pair-basis from coords");
+  m.def("gaussian_basis_backward", &gaussian_basis_backward,
+        "gaussian pair-basis backward (d_coords, d_means, d_stds)");
+  m.def("gaussian_basis_supported", &gaussian_basis_supported,
+        "kernel supports this K");
   m.def("cross_entropy_backward", &cross_entropy_backward,
         "cross entropy backward (softmax - onehot, no materialized fp32)");
 }

@@ -33,6 +33,7 @@ HIP_SOURCES = [
     "csrc/dropout_add.hip",
     "csrc/embedding.hip",
     "csrc/cross_entropy.hip",
+    "csrc/gaussian.hip",
 ]
 
 setup(

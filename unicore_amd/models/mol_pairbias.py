@@ -21,7 +21,12 @@ from unicore_amd.models import (
     register_model,
     register_model_architecture,
 )
-from unicore_amd.modules import RMSNorm, SelfMultiheadAttention, init_bert_params
+from unicore_amd.modules import (
+    RMSNorm,
+    SelfMultiheadAttention,
+    gaussian_basis,
+    init_bert_params,
+)
 from unicore_amd.modules.embedding import Embedding
 
 
@@ -37,14 +42,15 @@ class GaussianPairBias(nn.Module):
         self.out = nn.Linear(n_kernels, n_heads)
 
     def forward(self, coords, padding_mask=None):
-        # coords: (B, L, 3); distances in fp32 for stability, the gaussian
-        # basis in the model dtype (the (B, L, L, K) tensor is the biggest
-        # intermediate of the model — fp32 here doubled its HBM traffic)
+        # coords: (B, L, 3); gaussian math in fp32 (fused HIP kernel on
+        # GPU — one (B, L, L, K) write instead of cdist + five
+        # elementwise passes), cast once to the model dtype
         dtype = self.out.weight.dtype
-        dist = torch.cdist(coords.float(), coords.float()).to(dtype)
-        x = dist.unsqueeze(-1) - self.means.to(dtype).view(1, 1, 1, -1)
-        std = (self.stds.abs() + 1e-3).to(dtype)
-        g = torch.exp(-0.5 * (x / std) ** 2)  # (B, L, L, K)
+        # .float() keeps the basis math fp32 even when the trainer casts
+        # the module to bf16 (grads flow back through the cast)
+        g = gaussian_basis(
+            coords.float(), self.means.float(), self.stds.float(), dtype
+        )
         bias = self.out(g)  # (B, L, L, H)
         bias = bias.permute(0, 3, 1, 2).contiguous()  # (B, H, L, L)
         if padding_mask is not None:

@@ -5,6 +5,7 @@ from .rms_norm import RMSNorm
 from .softmax_dropout import softmax_dropout
 from .gelu_dropout import gelu_dropout
 from .dropout_add import dropout_add
+from .gaussian import gaussian_basis
 from .embedding import Embedding
 from .multihead_attention import SelfMultiheadAttention, CrossMultiheadAttention
 from .transformer_encoder_layer import TransformerEncoderLayer

@@ -202,3 +202,20 @@ def cross_entropy_bwd(logits, target, lse, grad_scale, ignore_index):
     return _kernels.cross_entropy_backward(
         logits, target, lse, grad_scale, int(ignore_index)
     )
+
+
+def gaussian_basis_fwd(coords, means, stds, out_dtype):
+    """(B, L, 3) fp32 coords -> (B, L, L, K) gaussian basis in out_dtype."""
+    require_kernels()
+    return _kernels.gaussian_basis_forward(coords, means, stds, out_dtype)
+
+
+def gaussian_basis_bwd(dg, coords, means, stds):
+    require_kernels()
+    return _kernels.gaussian_basis_backward(dg, coords, means, stds)
+
+
+def gaussian_basis_supported(n_kernels) -> bool:
+    return _kernels is not None and _kernels.gaussian_basis_supported(
+        int(n_kernels)
+    )
