@@ -539,10 +539,14 @@ def test_gaussian_basis_parity(dtype, B, L, K):
     out.backward(g)
     ref.backward(g.clone())
     n_pairs = B * L * L
-    assert (c1.grad - c2.grad).abs().max().item() < tol * 10
-    # mean/std grads are sums over B*L*L pairs; scale tolerance
-    assert (m1.grad - m2.grad).abs().max().item() < tol * n_pairs * 0.05 + 1e-4
-    assert (s1.grad - s2.grad).abs().max().item() < tol * n_pairs * 0.05 + 1e-4
+    # grads are long reductions (2L*K terms for coords, B*L*L for
+    # mean/std): compare relative to the grad magnitude, not absolutely
+    c_scale = c2.grad.abs().max().item() + 1e-3
+    assert (c1.grad - c2.grad).abs().max().item() / c_scale < tol * 10
+    m_scale = m2.grad.abs().max().item() + 1e-3
+    s_scale = s2.grad.abs().max().item() + 1e-3
+    assert (m1.grad - m2.grad).abs().max().item() / m_scale < tol * 100
+    assert (s1.grad - s2.grad).abs().max().item() / s_scale < tol * 100
 
 
 @requires_gpu

@@ -9,6 +9,8 @@ the tiny (B, L, 3) coords in backward rather than saved.  Eager fallback
 (also the numerics oracle) computes the same fp32 math then casts once.
 """
 
+import os
+
 import torch
 
 
@@ -53,6 +55,8 @@ def gaussian_basis(coords, means, stds, out_dtype=None):
     """coords (B, L, 3) fp32; means/stds (K,) fp32 parameters."""
     if out_dtype is None:
         out_dtype = coords.dtype
+    if os.environ.get("UNICORE_GAUSSIAN_EAGER", "0") == "1":  # A/B benchmarking
+        return _eager_gaussian_basis(coords, means, stds, out_dtype)
     if coords.is_cuda and _shape_ok(means.numel()):
         from unicore_amd import ops
 
