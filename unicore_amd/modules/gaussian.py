@@ -35,7 +35,13 @@ class _GaussianBasis(torch.autograd.Function):
 
 
 def _eager_gaussian_basis(coords, means, stds, out_dtype):
-    dist = torch.cdist(coords, coords)  # fp32, zero-subgradient diagonal
+    # exact pairwise distances (cdist's matmul path loses ~1e-3 at fp32);
+    # the +eye/-eye shift keeps the diagonal at zero with a zero gradient,
+    # matching both cdist's subgradient and the HIP kernel
+    diff = coords.unsqueeze(2) - coords.unsqueeze(1)  # (B, L, L, 3)
+    ssq = diff.pow(2).sum(-1)
+    eye = torch.eye(coords.size(1), device=coords.device, dtype=ssq.dtype)
+    dist = (ssq + eye).sqrt() - eye
     x = dist.unsqueeze(-1) - means.view(1, 1, 1, -1)
     inv = 1.0 / (stds.abs() + 1e-3)
     return torch.exp(-0.5 * (x * inv.view(1, 1, 1, -1)) ** 2).to(out_dtype)
