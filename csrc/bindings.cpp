@@ -53,6 +53,15 @@ at::Tensor gaussian_basis_forward(at::Tensor coords, at::Tensor means,
 std::vector<at::Tensor> gaussian_basis_backward(at::Tensor dg, at::Tensor coords,
                                                 at::Tensor means, at::Tensor stds);
 bool gaussian_basis_supported(int64_t K);
+at::Tensor gaussian_pair_bias_forward(at::Tensor coords, at::Tensor means,
+                                      at::Tensor stds, at::Tensor W,
+                                      at::Tensor bvec,
+                                      std::optional<at::Tensor> pad,
+                                      double fill, at::ScalarType out_dtype);
+std::vector<at::Tensor> gaussian_pair_bias_backward(
+    at::Tensor dbias, at::Tensor coords, at::Tensor means, at::Tensor stds,
+    at::Tensor W, std::optional<at::Tensor> pad);
+bool gaussian_pair_bias_supported(int64_t K, int64_t H);
 std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tensor v,
                                            std::optional<at::Tensor> bias,
                                            int64_t bias_outer_div,
@@ -109,6 +118,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gaussian pair-basis backward (d_coords, d_means, d_stds)");
   m.def("gaussian_basis_supported", &gaussian_basis_supported,
         "kernel supports this K");
+  m.def("gaussian_pair_bias_forward", &gaussian_pair_bias_forward,
+        "fully-fused gaussian pair bias: coords -> (B,H,L,L)");
+  m.def("gaussian_pair_bias_backward", &gaussian_pair_bias_backward,
+        "fused pair-bias backward (d_coords, d_means, d_stds, dW, db)");
+  m.def("gaussian_pair_bias_supported", &gaussian_pair_bias_supported,
+        "fused pair-bias kernel supports this K/H");
   m.def("cross_entropy_backward", &cross_entropy_backward,
         "cross entropy backward (softmax - onehot, no materialized fp32)");
 }

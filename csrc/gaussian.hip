@@ -202,17 +202,27 @@ __global__ void gaussian_bwd_dd_kernel(const OT* __restrict__ dg,
   }
 }
 
-// partials (nb, 2K) -> d_means (K), d_stds (K); one thread per column.
-__global__ void gaussian_fold_kernel(const float* __restrict__ partials,
-                                     float* __restrict__ d_means,
-                                     float* __restrict__ d_stds, int nb,
-                                     int K) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= 2 * K) return;
+// partials (nb, C) -> out (C): one block per column, threads stride the
+// rows, deterministic wave+LDS tree (nb can be the full 2048-block grid;
+// a single serial block here cost 479 us at the mol shape).
+__global__ void col_fold_kernel(const float* __restrict__ partials,
+                                float* __restrict__ out, int nb, int C) {
+  __shared__ float s_red[4];
+  const int c = blockIdx.x;
+  if (c >= C) return;
   float v = 0.f;
-  for (int r = 0; r < nb; ++r) v += partials[(int64_t)r * 2 * K + c];
-  if (c < K) d_means[c] = v;
-  else d_stds[c - K] = v;
+  for (int r = threadIdx.x; r < nb; r += blockDim.x)
+    v += partials[(int64_t)r * C + c];
+  v = wave_sum(v);
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  if (lane == 0) s_red[wave] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float r = 0.f;
+    const int nwaves = blockDim.x / 64;
+    for (int w = 0; w < nwaves; ++w) r += s_red[w];
+    out[c] = r;
+  }
 }
 
 // dd -> d_coords.  One block per (b,i):
@@ -271,6 +281,227 @@ bool gaussian_k_supported(int64_t K) {
   return (tpp & (tpp - 1)) == 0 && tpp <= 64;  // power of two, <= one wave
 }
 
+// ---------------------------------------------------------------------------
+// Fully-fused gaussian pair bias: coords -> (B, H, L, L) attention bias.
+// Folds the K->H output Linear, the (B,L,L,H)->(B,H,L,L) permute and the
+// padding-key masked_fill into the basis kernel, so the (B,L,L,K) feature
+// tensor never exists and the pathological skinny wgrad GEMM
+// ((B*L*L, K)^T @ (B*L*L, H), 2.1 ms via hipBLASLt at the mol shape) is
+// replaced by in-kernel register accumulation + the deterministic fold.
+// ---------------------------------------------------------------------------
+
+// lane layout: tpp = K/8 consecutive lanes per pair, lane q covers
+// k = q*8..q*8+7; the H head dots reduce over the pair's lanes with shfl.
+template <typename OT, int H>
+__global__ void gaussian_pair_bias_fwd_kernel(
+    const float* __restrict__ coords, const float* __restrict__ means,
+    const float* __restrict__ stds, const float* __restrict__ W,  // (H, K)
+    const float* __restrict__ bvec,                               // (H)
+    const uint8_t* __restrict__ pad,                              // (B, L) | null
+    OT* __restrict__ out, float fill, int64_t n_pairs, int L, int K) {
+  extern __shared__ float smem[];
+  float* s_mean = smem;          // [K]
+  float* s_inv = smem + K;       // [K]
+  float* s_w = smem + 2 * K;     // [H*K]
+  for (int k = threadIdx.x; k < K; k += blockDim.x) {
+    s_mean[k] = means[k];
+    s_inv[k] = 1.0f / (fabsf(stds[k]) + 1e-3f);
+  }
+  for (int k = threadIdx.x; k < H * K; k += blockDim.x) s_w[k] = W[k];
+  __syncthreads();
+
+  const int tpp = K / 8;
+  const int64_t total = n_pairs * tpp;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < total;
+       base += stride) {
+    const int64_t idx = base + threadIdx.x;
+    const bool active = idx < total;
+    const int64_t cidx = active ? idx : total - 1;
+    const int64_t pair = cidx / tpp;
+    const int k0 = (int)(cidx % tpp) * 8;
+    const int64_t b = pair / ((int64_t)L * L);
+    const int64_t rem = pair - b * (int64_t)L * L;
+    const int i = (int)(rem / L), j = (int)(rem % L);
+    float dx, dy, dz;
+    const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
+    float g[8];
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+      const float e = (dist - s_mean[k0 + t]) * s_inv[k0 + t];
+      g[t] = __expf(-0.5f * e * e);
+    }
+    float acc[H];
+#pragma unroll
+    for (int h = 0; h < H; ++h) {
+      float a = 0.f;
+#pragma unroll
+      for (int t = 0; t < 8; ++t) a += g[t] * s_w[h * K + k0 + t];
+      acc[h] = a;
+    }
+    // reduce each head over the pair's tpp aligned lanes
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      if (off < tpp) {
+#pragma unroll
+        for (int h = 0; h < H; ++h) acc[h] += __shfl_xor(acc[h], off, 64);
+      }
+    }
+    const int lane_in_pair = (int)(cidx % tpp);
+    if (active && lane_in_pair < H && lane_in_pair < tpp) {
+      const int h = lane_in_pair;
+      float v = acc[h] + bvec[h];
+      if (pad != nullptr && pad[b * L + j]) v = fill;
+      out[((b * H + h) * (int64_t)L + i) * L + j] = Cvt<OT>::from_f(v);
+    }
+    // tpp < H: lane 0 stores the remaining heads
+    if (active && tpp < H && lane_in_pair == 0) {
+      for (int h = tpp; h < H; ++h) {
+        float v = acc[h] + bvec[h];
+        if (pad != nullptr && pad[b * L + j]) v = fill;
+        out[((b * H + h) * (int64_t)L + i) * L + j] = Cvt<OT>::from_f(v);
+      }
+    }
+  }
+}
+
+// dbias (B,H,L,L) -> dd (B,L,L) + per-block partials of
+// [dW (H*K) | d_mean (K) | d_std (K) | db (H)], C = H*K + 2K + H.
+template <typename OT, int H>
+__global__ void gaussian_pair_bias_bwd_kernel(
+    const OT* __restrict__ dbias, const float* __restrict__ coords,
+    const float* __restrict__ means, const float* __restrict__ stds,
+    const float* __restrict__ W, const uint8_t* __restrict__ pad,
+    float* __restrict__ dd, float* __restrict__ partials, int64_t n_pairs,
+    int L, int K) {
+  const int C = H * K + 2 * K + H;
+  extern __shared__ float smem[];
+  float* s_mean = smem;               // [K]
+  float* s_inv = smem + K;            // [K]
+  float* s_dinv_ds = smem + 2 * K;    // [K]
+  float* s_w = smem + 3 * K;          // [H*K]
+  float* s_red = smem + 3 * K + H * K;  // [nwaves][C]
+  for (int k = threadIdx.x; k < K; k += blockDim.x) {
+    const float s = stds[k];
+    const float inv = 1.0f / (fabsf(s) + 1e-3f);
+    s_mean[k] = means[k];
+    s_inv[k] = inv;
+    const float sgn = (s > 0.f) ? 1.f : ((s < 0.f) ? -1.f : 0.f);
+    s_dinv_ds[k] = -sgn * inv * inv;
+  }
+  for (int k = threadIdx.x; k < H * K; k += blockDim.x) s_w[k] = W[k];
+  __syncthreads();
+
+  const int tpp = K / 8;
+  const int64_t total = n_pairs * tpp;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int k0 = (int)(((int64_t)blockIdx.x * blockDim.x + threadIdx.x) % tpp) * 8;
+
+  float acc_dm[8], acc_ds[8], acc_dw[H][8], acc_db[H];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) acc_dm[t] = acc_ds[t] = 0.f;
+#pragma unroll
+  for (int h = 0; h < H; ++h) {
+    acc_db[h] = 0.f;
+#pragma unroll
+    for (int t = 0; t < 8; ++t) acc_dw[h][t] = 0.f;
+  }
+
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < total;
+       base += stride) {
+    const int64_t idx = base + threadIdx.x;
+    const bool active = idx < total;
+    const int64_t pair = (active ? idx : total - 1) / tpp;
+    const int64_t b = pair / ((int64_t)L * L);
+    const int64_t rem = pair - b * (int64_t)L * L;
+    const int i = (int)(rem / L), j = (int)(rem % L);
+    float dx, dy, dz;
+    const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
+
+    float db[H];
+    const bool masked = (pad != nullptr && pad[b * L + j]);
+    if (active && !masked) {
+#pragma unroll
+      for (int h = 0; h < H; ++h)
+        db[h] = Cvt<OT>::to_f(
+            dbias[((b * H + h) * (int64_t)L + i) * L + j]);
+    } else {
+#pragma unroll
+      for (int h = 0; h < H; ++h) db[h] = 0.f;
+    }
+    if ((int)(idx % tpp) == 0) {
+#pragma unroll
+      for (int h = 0; h < H; ++h) acc_db[h] += db[h];
+    }
+
+    float d_dist = 0.f;
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+      const float m = s_mean[k0 + t], inv = s_inv[k0 + t];
+      const float e = (dist - m) * inv;
+      const float g = __expf(-0.5f * e * e);
+      float dg = 0.f;
+#pragma unroll
+      for (int h = 0; h < H; ++h) {
+        dg += db[h] * s_w[h * K + k0 + t];
+        acc_dw[h][t] += db[h] * g;
+      }
+      const float d_e = dg * g * (-e);
+      d_dist += d_e * inv;
+      acc_dm[t] -= d_e * inv;
+      acc_ds[t] += d_e * (dist - m) * s_dinv_ds[k0 + t];
+    }
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      if (off < tpp) d_dist += __shfl_xor(d_dist, off, 64);
+    }
+    if (active && (int)(idx % tpp) == 0) dd[pair] = d_dist;
+  }
+
+  // deterministic wave reduction (same-k lanes tpp apart; db over all lanes)
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+#pragma unroll
+  for (int t = 0; t < 8; ++t) {
+    for (int off = tpp; off < 64; off <<= 1) {
+      acc_dm[t] += __shfl_xor(acc_dm[t], off, 64);
+      acc_ds[t] += __shfl_xor(acc_ds[t], off, 64);
+#pragma unroll
+      for (int h = 0; h < H; ++h)
+        acc_dw[h][t] += __shfl_xor(acc_dw[h][t], off, 64);
+    }
+  }
+#pragma unroll
+  for (int h = 0; h < H; ++h) {
+    for (int off = 1; off < 64; off <<= 1)
+      acc_db[h] += __shfl_xor(acc_db[h], off, 64);
+  }
+  float* my = s_red + wave * C;
+  if (lane < tpp) {
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+#pragma unroll
+      for (int h = 0; h < H; ++h) my[h * K + k0 + t] = acc_dw[h][t];
+      my[H * K + k0 + t] = acc_dm[t];
+      my[H * K + K + k0 + t] = acc_ds[t];
+    }
+  }
+  if (lane == 0) {
+#pragma unroll
+    for (int h = 0; h < H; ++h) my[H * K + 2 * K + h] = acc_db[h];
+  }
+  __syncthreads();
+  const int nwaves = blockDim.x / 64;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float v = 0.f;
+    for (int w = 0; w < nwaves; ++w) v += s_red[w * C + c];
+    partials[(int64_t)blockIdx.x * C + c] = v;
+  }
+}
+
+bool gaussian_h_supported(int64_t H) {
+  return H == 4 || H == 8 || H == 16;
+}
+
 }  // namespace
 
 torch::Tensor gaussian_basis_forward(torch::Tensor coords,
@@ -315,8 +546,6 @@ std::vector<torch::Tensor> gaussian_basis_backward(torch::Tensor dg,
   const int64_t n_pairs = B * L * L;
   auto dd = torch::empty({B, L, L}, coords.options());
   auto d_coords = torch::empty_like(cc);
-  auto d_means = torch::empty_like(mc);
-  auto d_stds = torch::empty_like(sc);
   const int block = 256;
   const int grid = unicore_grid((n_pairs * (K / 8) + block - 1) / block);
   auto partials = torch::empty({grid, 2 * K}, coords.options());
@@ -329,15 +558,144 @@ std::vector<torch::Tensor> gaussian_basis_backward(torch::Tensor dg,
         dd.data_ptr<float>(), partials.data_ptr<float>(), n_pairs, (int)L,
         (int)K);
   });
-  const int fold_grid = (int)((2 * K + block - 1) / block);
-  gaussian_fold_kernel<<<fold_grid, block, 0, stream>>>(
-      partials.data_ptr<float>(), d_means.data_ptr<float>(),
-      d_stds.data_ptr<float>(), grid, (int)K);
+  auto fold = torch::empty({2 * K}, coords.options());
+  col_fold_kernel<<<2 * (int)K, block, 0, stream>>>(
+      partials.data_ptr<float>(), fold.data_ptr<float>(), grid, 2 * (int)K);
   const int64_t n_rows = B * L;
   gaussian_dcoords_kernel<<<unicore_grid(n_rows), block, 0, stream>>>(
       dd.data_ptr<float>(), cc.data_ptr<float>(), d_coords.data_ptr<float>(),
       n_rows, (int)L);
-  return {d_coords, d_means, d_stds};
+  return {d_coords, fold.narrow(0, 0, K), fold.narrow(0, K, K)};
 }
 
 bool gaussian_basis_supported(int64_t K) { return gaussian_k_supported(K); }
+
+static size_t pair_bias_bwd_lds(int64_t K, int64_t H) {
+  const int64_t C = H * K + 2 * K + H;
+  return (size_t)(3 * K + H * K + 4 * C) * sizeof(float);
+}
+
+bool gaussian_pair_bias_supported(int64_t K, int64_t H) {
+  return gaussian_k_supported(K) && gaussian_h_supported(H) &&
+         pair_bias_bwd_lds(K, H) <= 64 * 1024;
+}
+
+#define DISPATCH_H(H, ...)                         \
+  switch (H) {                                     \
+    case 4: {                                      \
+      constexpr int kH = 4;                        \
+      __VA_ARGS__;                                 \
+      break;                                       \
+    }                                              \
+    case 8: {                                      \
+      constexpr int kH = 8;                        \
+      __VA_ARGS__;                                 \
+      break;                                       \
+    }                                              \
+    case 16: {                                     \
+      constexpr int kH = 16;                       \
+      __VA_ARGS__;                                 \
+      break;                                       \
+    }                                              \
+    default:                                       \
+      TORCH_CHECK(false, "gaussian_pair_bias: unsupported H ", H); \
+  }
+
+torch::Tensor gaussian_pair_bias_forward(torch::Tensor coords,
+                                         torch::Tensor means,
+                                         torch::Tensor stds, torch::Tensor W,
+                                         torch::Tensor bvec,
+                                         std::optional<torch::Tensor> pad,
+                                         double fill,
+                                         at::ScalarType out_dtype) {
+  TORCH_CHECK(coords.is_cuda() && coords.dim() == 3 && coords.size(2) == 3,
+              "gaussian_pair_bias: coords must be CUDA (B, L, 3)");
+  const int64_t B = coords.size(0), L = coords.size(1), K = means.numel();
+  const int64_t H = bvec.numel();
+  TORCH_CHECK(W.dim() == 2 && W.size(0) == H && W.size(1) == K,
+              "gaussian_pair_bias: W must be (H, K)");
+  TORCH_CHECK(gaussian_pair_bias_supported(K, H),
+              "gaussian_pair_bias: unsupported K/H ", K, "/", H);
+  auto cc = coords.contiguous();
+  auto mc = means.contiguous().to(at::kFloat);
+  auto sc = stds.contiguous().to(at::kFloat);
+  auto wc = W.contiguous().to(at::kFloat);
+  auto bc = bvec.contiguous().to(at::kFloat);
+  const uint8_t* pad_ptr = nullptr;
+  torch::Tensor padc;
+  if (pad.has_value()) {
+    padc = pad->contiguous().to(at::kBool);
+    TORCH_CHECK(padc.numel() == B * L, "gaussian_pair_bias: pad must be (B, L)");
+    pad_ptr = (const uint8_t*)padc.data_ptr<bool>();
+  }
+  auto out = torch::empty({B, H, L, L}, coords.options().dtype(out_dtype));
+  const int64_t n_pairs = B * L * L;
+  const int block = 256;
+  const int grid = unicore_grid((n_pairs * (K / 8) + block - 1) / block);
+  const size_t lds = (size_t)(2 * K + H * K) * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  DISPATCH_OUT_FTYPES(out_dtype, "gaussian_pair_bias_fwd", {
+    DISPATCH_H(H, {
+      gaussian_pair_bias_fwd_kernel<scalar_t, kH><<<grid, block, lds, stream>>>(
+          cc.data_ptr<float>(), mc.data_ptr<float>(), sc.data_ptr<float>(),
+          wc.data_ptr<float>(), bc.data_ptr<float>(), pad_ptr,
+          reinterpret_cast<scalar_t*>(out.data_ptr()), (float)fill, n_pairs,
+          (int)L, (int)K);
+    });
+  });
+  return out;
+}
+
+std::vector<torch::Tensor> gaussian_pair_bias_backward(
+    torch::Tensor dbias, torch::Tensor coords, torch::Tensor means,
+    torch::Tensor stds, torch::Tensor W,
+    std::optional<torch::Tensor> pad) {
+  const int64_t B = coords.size(0), L = coords.size(1), K = means.numel();
+  const int64_t H = W.size(0);
+  TORCH_CHECK(dbias.is_cuda() && dbias.dim() == 4 && dbias.size(1) == H &&
+                  dbias.size(2) == L && dbias.size(3) == L,
+              "gaussian_pair_bias: bad grad shape");
+  auto gc = dbias.contiguous();
+  auto cc = coords.contiguous();
+  auto mc = means.contiguous().to(at::kFloat);
+  auto sc = stds.contiguous().to(at::kFloat);
+  auto wc = W.contiguous().to(at::kFloat);
+  const uint8_t* pad_ptr = nullptr;
+  torch::Tensor padc;
+  if (pad.has_value()) {
+    padc = pad->contiguous().to(at::kBool);
+    pad_ptr = (const uint8_t*)padc.data_ptr<bool>();
+  }
+  const int64_t n_pairs = B * L * L;
+  const int64_t C = H * K + 2 * K + H;
+  auto fopts = coords.options().dtype(at::kFloat);
+  auto dd = torch::empty({B, L, L}, fopts);
+  auto d_coords = torch::empty({B, L, 3}, fopts);
+  const int block = 256;
+  const int grid = unicore_grid((n_pairs * (K / 8) + block - 1) / block);
+  auto partials = torch::empty({grid, C}, fopts);
+  auto fold = torch::empty({C}, fopts);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const size_t lds = pair_bias_bwd_lds(K, H);
+  DISPATCH_OUT_FTYPES(gc.scalar_type(), "gaussian_pair_bias_bwd", {
+    DISPATCH_H(H, {
+      gaussian_pair_bias_bwd_kernel<scalar_t, kH><<<grid, block, lds, stream>>>(
+          reinterpret_cast<const scalar_t*>(gc.data_ptr()),
+          cc.data_ptr<float>(), mc.data_ptr<float>(), sc.data_ptr<float>(),
+          wc.data_ptr<float>(), pad_ptr, dd.data_ptr<float>(),
+          partials.data_ptr<float>(), n_pairs, (int)L, (int)K);
+    });
+  });
+  col_fold_kernel<<<(int)C, block, 0, stream>>>(
+      partials.data_ptr<float>(), fold.data_ptr<float>(), grid, (int)C);
+  const int64_t n_rows = B * L;
+  gaussian_dcoords_kernel<<<unicore_grid(n_rows), block, 0, stream>>>(
+      dd.data_ptr<float>(), cc.data_ptr<float>(), d_coords.data_ptr<float>(),
+      n_rows, (int)L);
+  auto d_w = fold.narrow(0, 0, H * K).view({H, K});
+  auto d_means = fold.narrow(0, H * K, K);
+  auto d_stds = fold.narrow(0, H * K + K, K);
+  auto d_b = fold.narrow(0, H * K + 2 * K, H);
+  return {d_coords, d_means, d_stds, d_w, d_b};
+}
+

@@ -27,6 +27,10 @@ from unicore_amd.modules import (
     gaussian_basis,
     init_bert_params,
 )
+from unicore_amd.modules.gaussian import (
+    _GaussianPairBias,
+    gaussian_pair_bias_fused_ok,
+)
 from unicore_amd.modules.embedding import Embedding
 
 
@@ -42,12 +46,30 @@ class GaussianPairBias(nn.Module):
         self.out = nn.Linear(n_kernels, n_heads)
 
     def forward(self, coords, padding_mask=None):
-        # coords: (B, L, 3); gaussian math in fp32 (fused HIP kernel on
-        # GPU — one (B, L, L, K) write instead of cdist + five
-        # elementwise passes), cast once to the model dtype
+        # coords: (B, L, 3); gaussian math in fp32, cast once to the model
+        # dtype.  On GPU the whole module is two HIP kernels (fwd/bwd):
+        # basis + K->H Linear + permute-to-head-major + padding masked_fill
+        # fused, so the (B, L, L, K) feature tensor never exists.
         dtype = self.out.weight.dtype
-        # .float() keeps the basis math fp32 even when the trainer casts
-        # the module to bf16 (grads flow back through the cast)
+        fill = torch.finfo(dtype).min
+        if coords.is_cuda and gaussian_pair_bias_fused_ok(
+            self.n_kernels, self.out.out_features
+        ):
+            pad = None
+            if padding_mask is not None:
+                pad = padding_mask.to(torch.bool).contiguous()
+            # .float() keeps the parameter math fp32 under a bf16-cast
+            # module (grads flow back through the casts)
+            return _GaussianPairBias.apply(
+                coords.float(),
+                self.means.float(),
+                self.stds.float(),
+                self.out.weight,
+                self.out.bias,
+                pad,
+                fill,
+                dtype,
+            )
         g = gaussian_basis(
             coords.float(), self.means.float(), self.stds.float(), dtype
         )
@@ -55,7 +77,6 @@ class GaussianPairBias(nn.Module):
         bias = bias.permute(0, 3, 1, 2).contiguous()  # (B, H, L, L)
         if padding_mask is not None:
             # additive -inf on padded keys
-            fill = torch.finfo(bias.dtype).min
             bias = bias.masked_fill(
                 padding_mask.view(padding_mask.size(0), 1, 1, -1).to(torch.bool),
                 fill,

@@ -69,3 +69,49 @@ def gaussian_basis(coords, means, stds, out_dtype=None):
         if ops.gpu_kernels_available() or not ops.allow_eager_on_gpu():
             return _GaussianBasis.apply(coords, means, stds, out_dtype)
     return _eager_gaussian_basis(coords, means, stds, out_dtype)
+
+
+class _GaussianPairBias(torch.autograd.Function):
+    """Fully-fused pair bias: coords -> (B, H, L, L), with the K->H Linear,
+    the permute to head-major and the padding-key masked_fill folded in."""
+
+    @staticmethod
+    def forward(ctx, coords, means, stds, weight, bias, pad, fill, out_dtype):
+        from unicore_amd import ops
+
+        coords = coords.contiguous()
+        ctx.save_for_backward(coords, means, stds, weight)
+        ctx.pad = pad
+        out = ops.gaussian_pair_bias_fwd(
+            coords, means, stds, weight, bias, pad, fill, out_dtype
+        )
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        from unicore_amd import ops
+
+        coords, means, stds, weight = ctx.saved_tensors
+        d_coords, d_means, d_stds, d_w, d_b = ops.gaussian_pair_bias_bwd(
+            grad.contiguous(), coords, means, stds, weight, ctx.pad
+        )
+        return (
+            d_coords,
+            d_means.to(means.dtype),
+            d_stds.to(stds.dtype),
+            d_w.to(weight.dtype),
+            d_b.to(weight.dtype),
+            None,
+            None,
+            None,
+        )
+
+
+def gaussian_pair_bias_fused_ok(n_kernels, n_heads):
+    if os.environ.get("UNICORE_GAUSSIAN_EAGER", "0") == "1":  # A/B benchmarking
+        return False
+    from unicore_amd import ops
+
+    return ops.has_kernels() and ops.gaussian_pair_bias_supported(
+        n_kernels, n_heads
+    )

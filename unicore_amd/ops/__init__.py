@@ -219,3 +219,24 @@ def gaussian_basis_supported(n_kernels) -> bool:
     return _kernels is not None and _kernels.gaussian_basis_supported(
         int(n_kernels)
     )
+
+
+def gaussian_pair_bias_fwd(coords, means, stds, weight, bias, pad, fill, out_dtype):
+    """coords (B,L,3) fp32 -> (B,H,L,L) attention bias, Linear+permute+mask fused."""
+    require_kernels()
+    return _kernels.gaussian_pair_bias_forward(
+        coords, means, stds, weight, bias, pad, float(fill), out_dtype
+    )
+
+
+def gaussian_pair_bias_bwd(dbias, coords, means, stds, weight, pad):
+    require_kernels()
+    return _kernels.gaussian_pair_bias_backward(
+        dbias, coords, means, stds, weight, pad
+    )
+
+
+def gaussian_pair_bias_supported(n_kernels, n_heads) -> bool:
+    return _kernels is not None and _kernels.gaussian_pair_bias_supported(
+        int(n_kernels), int(n_heads)
+    )
