@@ -290,8 +290,10 @@ bool gaussian_k_supported(int64_t K) {
 // replaced by in-kernel register accumulation + the deterministic fold.
 // ---------------------------------------------------------------------------
 
-// lane layout: tpp = K/8 consecutive lanes per pair, lane q covers
-// k = q*8..q*8+7; the H head dots reduce over the pair's lanes with shfl.
+// One thread per pair: the K-loop is serial, so every LDS read is a
+// wave-wide broadcast (conflict-free), there is no cross-lane reduction,
+// and the H stores of a wave hit 64 consecutive j per head plane
+// (coalesced).  ~50 VGPRs -> full occupancy.
 template <typename OT, int H>
 __global__ void gaussian_pair_bias_fwd_kernel(
     const float* __restrict__ coords, const float* __restrict__ means,
@@ -310,70 +312,103 @@ __global__ void gaussian_pair_bias_fwd_kernel(
   for (int k = threadIdx.x; k < H * K; k += blockDim.x) s_w[k] = W[k];
   __syncthreads();
 
-  const int tpp = K / 8;
-  const int64_t total = n_pairs * tpp;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < total;
-       base += stride) {
-    const int64_t idx = base + threadIdx.x;
-    const bool active = idx < total;
-    const int64_t cidx = active ? idx : total - 1;
-    const int64_t pair = cidx / tpp;
-    const int k0 = (int)(cidx % tpp) * 8;
+  for (int64_t pair = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       pair < n_pairs; pair += stride) {
     const int64_t b = pair / ((int64_t)L * L);
     const int64_t rem = pair - b * (int64_t)L * L;
     const int i = (int)(rem / L), j = (int)(rem % L);
     float dx, dy, dz;
     const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
-    float g[8];
-#pragma unroll
-    for (int t = 0; t < 8; ++t) {
-      const float e = (dist - s_mean[k0 + t]) * s_inv[k0 + t];
-      g[t] = __expf(-0.5f * e * e);
-    }
     float acc[H];
 #pragma unroll
+    for (int h = 0; h < H; ++h) acc[h] = bvec[h];
+    for (int k8 = 0; k8 < K; k8 += 8) {
+      float g[8];
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const float e = (dist - s_mean[k8 + t]) * s_inv[k8 + t];
+        g[t] = __expf(-0.5f * e * e);
+      }
+#pragma unroll
+      for (int h = 0; h < H; ++h) {
+        float a = 0.f;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) a += g[t] * s_w[h * K + k8 + t];
+        acc[h] += a;
+      }
+    }
+    const bool masked = (pad != nullptr && pad[b * L + j]);
+#pragma unroll
     for (int h = 0; h < H; ++h) {
-      float a = 0.f;
-#pragma unroll
-      for (int t = 0; t < 8; ++t) a += g[t] * s_w[h * K + k0 + t];
-      acc[h] = a;
-    }
-    // reduce each head over the pair's tpp aligned lanes
-#pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-      if (off < tpp) {
-#pragma unroll
-        for (int h = 0; h < H; ++h) acc[h] += __shfl_xor(acc[h], off, 64);
-      }
-    }
-    const int lane_in_pair = (int)(cidx % tpp);
-    if (active && lane_in_pair < H && lane_in_pair < tpp) {
-      const int h = lane_in_pair;
-      float v = acc[h] + bvec[h];
-      if (pad != nullptr && pad[b * L + j]) v = fill;
+      const float v = masked ? fill : acc[h];
       out[((b * H + h) * (int64_t)L + i) * L + j] = Cvt<OT>::from_f(v);
-    }
-    // tpp < H: lane 0 stores the remaining heads
-    if (active && tpp < H && lane_in_pair == 0) {
-      for (int h = tpp; h < H; ++h) {
-        float v = acc[h] + bvec[h];
-        if (pad != nullptr && pad[b * L + j]) v = fill;
-        out[((b * H + h) * (int64_t)L + i) * L + j] = Cvt<OT>::from_f(v);
-      }
     }
   }
 }
 
-// dbias (B,H,L,L) -> dd (B,L,L) + per-block partials of
-// [dW (H*K) | d_mean (K) | d_std (K) | db (H)], C = H*K + 2K + H.
+// Backward is split so neither kernel spills registers (a merged
+// version spilled 173 VGPRs -> 544 B of per-thread scratch and ran 10x
+// slow):
+//   A: dd (distance grads), one thread per pair, broadcast LDS reads
+//   B: dW/d_mean/d_std/db per-block partials, 4-k-slice lanes so the dW
+//      accumulator is 32 registers instead of 64.
 template <typename OT, int H>
-__global__ void gaussian_pair_bias_bwd_kernel(
+__global__ void gaussian_pair_bias_bwd_dd_kernel(
     const OT* __restrict__ dbias, const float* __restrict__ coords,
     const float* __restrict__ means, const float* __restrict__ stds,
     const float* __restrict__ W, const uint8_t* __restrict__ pad,
-    float* __restrict__ dd, float* __restrict__ partials, int64_t n_pairs,
-    int L, int K) {
+    float* __restrict__ dd, int64_t n_pairs, int L, int K) {
+  extern __shared__ float smem[];
+  float* s_mean = smem;       // [K]
+  float* s_inv = smem + K;    // [K]
+  float* s_w = smem + 2 * K;  // [H*K]
+  for (int k = threadIdx.x; k < K; k += blockDim.x) {
+    s_mean[k] = means[k];
+    s_inv[k] = 1.0f / (fabsf(stds[k]) + 1e-3f);
+  }
+  for (int k = threadIdx.x; k < H * K; k += blockDim.x) s_w[k] = W[k];
+  __syncthreads();
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t pair = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       pair < n_pairs; pair += stride) {
+    const int64_t b = pair / ((int64_t)L * L);
+    const int64_t rem = pair - b * (int64_t)L * L;
+    const int i = (int)(rem / L), j = (int)(rem % L);
+    if (pad != nullptr && pad[b * L + j]) {
+      dd[pair] = 0.f;
+      continue;
+    }
+    float db[H];
+#pragma unroll
+    for (int h = 0; h < H; ++h)
+      db[h] = Cvt<OT>::to_f(dbias[((b * H + h) * (int64_t)L + i) * L + j]);
+    float dx, dy, dz;
+    const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
+    float d_dist = 0.f;
+    for (int k8 = 0; k8 < K; k8 += 8) {
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const float m = s_mean[k8 + t], inv = s_inv[k8 + t];
+        const float e = (dist - m) * inv;
+        const float g = __expf(-0.5f * e * e);
+        float dg = 0.f;
+#pragma unroll
+        for (int h = 0; h < H; ++h) dg += db[h] * s_w[h * K + k8 + t];
+        d_dist += dg * g * (-e) * inv;
+      }
+    }
+    dd[pair] = d_dist;
+  }
+}
+
+template <typename OT, int H>
+__global__ void gaussian_pair_bias_bwd_part_kernel(
+    const OT* __restrict__ dbias, const float* __restrict__ coords,
+    const float* __restrict__ means, const float* __restrict__ stds,
+    const float* __restrict__ W, const uint8_t* __restrict__ pad,
+    float* __restrict__ partials, int64_t n_pairs, int L, int K) {
   const int C = H * K + 2 * K + H;
   extern __shared__ float smem[];
   float* s_mean = smem;               // [K]
@@ -392,51 +427,40 @@ __global__ void gaussian_pair_bias_bwd_kernel(
   for (int k = threadIdx.x; k < H * K; k += blockDim.x) s_w[k] = W[k];
   __syncthreads();
 
-  const int tpp = K / 8;
+  const int tpp = K / 4;  // 4-k slice keeps acc_dw at 32 registers
   const int64_t total = n_pairs * tpp;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  const int k0 = (int)(((int64_t)blockIdx.x * blockDim.x + threadIdx.x) % tpp) * 8;
+  const int k0 = (int)(((int64_t)blockIdx.x * blockDim.x + threadIdx.x) % tpp) * 4;
 
-  float acc_dm[8], acc_ds[8], acc_dw[H][8], acc_db[H];
+  float acc_dm[4], acc_ds[4], acc_dw[H][4], acc_db[H];
 #pragma unroll
-  for (int t = 0; t < 8; ++t) acc_dm[t] = acc_ds[t] = 0.f;
+  for (int t = 0; t < 4; ++t) acc_dm[t] = acc_ds[t] = 0.f;
 #pragma unroll
   for (int h = 0; h < H; ++h) {
     acc_db[h] = 0.f;
 #pragma unroll
-    for (int t = 0; t < 8; ++t) acc_dw[h][t] = 0.f;
+    for (int t = 0; t < 4; ++t) acc_dw[h][t] = 0.f;
   }
 
-  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < total;
-       base += stride) {
-    const int64_t idx = base + threadIdx.x;
-    const bool active = idx < total;
-    const int64_t pair = (active ? idx : total - 1) / tpp;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const int64_t pair = idx / tpp;
     const int64_t b = pair / ((int64_t)L * L);
     const int64_t rem = pair - b * (int64_t)L * L;
     const int i = (int)(rem / L), j = (int)(rem % L);
-    float dx, dy, dz;
-    const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
-
+    if (pad != nullptr && pad[b * L + j]) continue;
     float db[H];
-    const bool masked = (pad != nullptr && pad[b * L + j]);
-    if (active && !masked) {
 #pragma unroll
-      for (int h = 0; h < H; ++h)
-        db[h] = Cvt<OT>::to_f(
-            dbias[((b * H + h) * (int64_t)L + i) * L + j]);
-    } else {
-#pragma unroll
-      for (int h = 0; h < H; ++h) db[h] = 0.f;
-    }
+    for (int h = 0; h < H; ++h)
+      db[h] = Cvt<OT>::to_f(dbias[((b * H + h) * (int64_t)L + i) * L + j]);
     if ((int)(idx % tpp) == 0) {
 #pragma unroll
       for (int h = 0; h < H; ++h) acc_db[h] += db[h];
     }
-
-    float d_dist = 0.f;
+    float dx, dy, dz;
+    const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
+    for (int t = 0; t < 4; ++t) {
       const float m = s_mean[k0 + t], inv = s_inv[k0 + t];
       const float e = (dist - m) * inv;
       const float g = __expf(-0.5f * e * e);
@@ -447,21 +471,15 @@ __global__ void gaussian_pair_bias_bwd_kernel(
         acc_dw[h][t] += db[h] * g;
       }
       const float d_e = dg * g * (-e);
-      d_dist += d_e * inv;
       acc_dm[t] -= d_e * inv;
       acc_ds[t] += d_e * (dist - m) * s_dinv_ds[k0 + t];
     }
-#pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-      if (off < tpp) d_dist += __shfl_xor(d_dist, off, 64);
-    }
-    if (active && (int)(idx % tpp) == 0) dd[pair] = d_dist;
   }
 
   // deterministic wave reduction (same-k lanes tpp apart; db over all lanes)
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
 #pragma unroll
-  for (int t = 0; t < 8; ++t) {
+  for (int t = 0; t < 4; ++t) {
     for (int off = tpp; off < 64; off <<= 1) {
       acc_dm[t] += __shfl_xor(acc_dm[t], off, 64);
       acc_ds[t] += __shfl_xor(acc_ds[t], off, 64);
@@ -476,9 +494,9 @@ __global__ void gaussian_pair_bias_bwd_kernel(
       acc_db[h] += __shfl_xor(acc_db[h], off, 64);
   }
   float* my = s_red + wave * C;
-  if (lane < tpp) {
+  if (lane < tpp && lane * 4 < K) {
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
+    for (int t = 0; t < 4; ++t) {
 #pragma unroll
       for (int h = 0; h < H; ++h) my[h * K + k0 + t] = acc_dw[h][t];
       my[H * K + k0 + t] = acc_dm[t];
@@ -499,7 +517,9 @@ __global__ void gaussian_pair_bias_bwd_kernel(
 }
 
 bool gaussian_h_supported(int64_t H) {
-  return H == 4 || H == 8 || H == 16;
+  // H=16 would spill ~90 VGPRs in the partials kernel; those shapes take
+  // the (still fused-basis) fallback path instead
+  return H == 4 || H == 8;
 }
 
 }  // namespace
@@ -576,8 +596,12 @@ static size_t pair_bias_bwd_lds(int64_t K, int64_t H) {
 }
 
 bool gaussian_pair_bias_supported(int64_t K, int64_t H) {
-  return gaussian_k_supported(K) && gaussian_h_supported(H) &&
-         pair_bias_bwd_lds(K, H) <= 64 * 1024;
+  // K%8 for the 8-wide fwd chunks; K/4 a power of two <= 64 for the
+  // partials kernel's lane slicing
+  if (K < 8 || K % 8 != 0) return false;
+  const int64_t tpp = K / 4;
+  if ((tpp & (tpp - 1)) != 0 || tpp > 64) return false;
+  return gaussian_h_supported(H) && pair_bias_bwd_lds(K, H) <= 64 * 1024;
 }
 
 #define DISPATCH_H(H, ...)                         \
@@ -589,11 +613,6 @@ bool gaussian_pair_bias_supported(int64_t K, int64_t H) {
     }                                              \
     case 8: {                                      \
       constexpr int kH = 8;                        \
-      __VA_ARGS__;                                 \
-      break;                                       \
-    }                                              \
-    case 16: {                                     \
-      constexpr int kH = 16;                       \
       __VA_ARGS__;                                 \
       break;                                       \
     }                                              \
@@ -672,18 +691,27 @@ std::vector<torch::Tensor> gaussian_pair_bias_backward(
   auto dd = torch::empty({B, L, L}, fopts);
   auto d_coords = torch::empty({B, L, 3}, fopts);
   const int block = 256;
-  const int grid = unicore_grid((n_pairs * (K / 8) + block - 1) / block);
+  const int grid_dd = unicore_grid((n_pairs + block - 1) / block);
+  const int grid = unicore_grid((n_pairs * (K / 4) + block - 1) / block);
   auto partials = torch::empty({grid, C}, fopts);
   auto fold = torch::empty({C}, fopts);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const size_t lds_dd = (size_t)(2 * K + H * K) * sizeof(float);
   const size_t lds = pair_bias_bwd_lds(K, H);
   DISPATCH_OUT_FTYPES(gc.scalar_type(), "gaussian_pair_bias_bwd", {
     DISPATCH_H(H, {
-      gaussian_pair_bias_bwd_kernel<scalar_t, kH><<<grid, block, lds, stream>>>(
-          reinterpret_cast<const scalar_t*>(gc.data_ptr()),
-          cc.data_ptr<float>(), mc.data_ptr<float>(), sc.data_ptr<float>(),
-          wc.data_ptr<float>(), pad_ptr, dd.data_ptr<float>(),
-          partials.data_ptr<float>(), n_pairs, (int)L, (int)K);
+      gaussian_pair_bias_bwd_dd_kernel<scalar_t, kH>
+          <<<grid_dd, block, lds_dd, stream>>>(
+              reinterpret_cast<const scalar_t*>(gc.data_ptr()),
+              cc.data_ptr<float>(), mc.data_ptr<float>(), sc.data_ptr<float>(),
+              wc.data_ptr<float>(), pad_ptr, dd.data_ptr<float>(), n_pairs,
+              (int)L, (int)K);
+      gaussian_pair_bias_bwd_part_kernel<scalar_t, kH>
+          <<<grid, block, lds, stream>>>(
+              reinterpret_cast<const scalar_t*>(gc.data_ptr()),
+              cc.data_ptr<float>(), mc.data_ptr<float>(), sc.data_ptr<float>(),
+              wc.data_ptr<float>(), pad_ptr, partials.data_ptr<float>(),
+              n_pairs, (int)L, (int)K);
     });
   });
   col_fold_kernel<<<(int)C, block, 0, stream>>>(

@@ -613,7 +613,11 @@ def test_gaussian_pair_bias_fused_parity(dtype, B, L, K, H, use_pad):
     ref = GaussianPairBias(n_kernels=K, n_heads=H).cuda()
     ref.load_state_dict(mod.state_dict())
     if dtype == torch.bfloat16:
+        # ref must share mod's bf16-rounded parameters: a bf16-cast module
+        # loses ~0.4% on means/stds (std ~0.08 -> inv ~13, so a rounded
+        # mean shifts e by up to 0.5) and BOTH paths inherit that
         mod = mod.bfloat16()
+        ref = ref.bfloat16()
     coords = torch.randn(B, L, 3, device="cuda") * 3
     pad = None
     if use_pad:
@@ -642,7 +646,7 @@ def test_gaussian_pair_bias_fused_parity(dtype, B, L, K, H, use_pad):
 
     g = torch.randn(B, H, L, L, device="cuda")
     out.float().backward(gradient=g * valid)
-    rout.backward(gradient=(g * valid).float())
+    rout.float().backward(gradient=g * valid)
     pairs = [("means", mod.means, ref.means), ("stds", mod.stds, ref.stds),
              ("W", mod.out.weight, ref.out.weight), ("b", mod.out.bias, ref.out.bias)]
     for name, p, q in pairs:
