@@ -442,21 +442,45 @@ __global__ void gaussian_pair_bias_bwd_part_kernel(
     for (int t = 0; t < 4; ++t) acc_dw[h][t] = 0.f;
   }
 
-  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < total; idx += stride) {
-    const int64_t pair = idx / tpp;
+  // uniform loop: the shfl broadcast of db below needs whole-wave
+  // participation (pair groups are tpp-aligned within the wave)
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < total;
+       base += stride) {
+    const int64_t idx = base + threadIdx.x;
+    const bool active = idx < total;
+    const int64_t pair = (active ? idx : total - 1) / tpp;
     const int64_t b = pair / ((int64_t)L * L);
     const int64_t rem = pair - b * (int64_t)L * L;
     const int i = (int)(rem / L), j = (int)(rem % L);
-    if (pad != nullptr && pad[b * L + j]) continue;
+    const bool masked =
+        !active || (pad != nullptr && pad[b * L + j]);
+    const int lane_in_pair = (int)((threadIdx.x % 64) % tpp);
     float db[H];
+    if (tpp >= H) {
+      // lane h of each pair group loads head h once, then broadcasts
+      float mine = 0.f;
+      if (!masked && lane_in_pair < H)
+        mine = Cvt<OT>::to_f(
+            dbias[((b * H + lane_in_pair) * (int64_t)L + i) * L + j]);
+      const int base_lane = (int)(threadIdx.x % 64) - lane_in_pair;
 #pragma unroll
-    for (int h = 0; h < H; ++h)
-      db[h] = Cvt<OT>::to_f(dbias[((b * H + h) * (int64_t)L + i) * L + j]);
-    if ((int)(idx % tpp) == 0) {
+      for (int h = 0; h < H; ++h) db[h] = __shfl(mine, base_lane + h, 64);
+      if (masked) {
+#pragma unroll
+        for (int h = 0; h < H; ++h) db[h] = 0.f;
+      }
+    } else {
+#pragma unroll
+      for (int h = 0; h < H; ++h)
+        db[h] = masked ? 0.f
+                       : Cvt<OT>::to_f(
+                             dbias[((b * H + h) * (int64_t)L + i) * L + j]);
+    }
+    if (active && (int)(idx % tpp) == 0) {
 #pragma unroll
       for (int h = 0; h < H; ++h) acc_db[h] += db[h];
     }
+    if (masked) continue;
     float dx, dy, dz;
     const float dist = pair_dist(coords, b, i, j, L, dx, dy, dz);
 #pragma unroll
