@@ -819,51 +819,22 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
   const int n_tiles = L / BN;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
-    // hoist the tile's global reads (K/V fragments + bias/mask scalars) so
-    // a single latency wait covers them all — interleaving them with the
-    // MFMAs that consume them left the kernel at ~8% MFMA utilization
-    bf16x8 bkf[4][2], bvf[4][2];
-#pragma unroll
-    for (int cb = 0; cb < 4; ++cb)
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const int64_t row = (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD;
-        bkf[cb][ks] = load_frag(kp + row + ks * 32 + lg * 8);
-        bvf[cb][ks] = load_frag(vp + row + ks * 32 + lg * 8);
-      }
-    float biasv[2][4][4];
-    float maskv[4];
-    if (HAS_BIAS) {
-#pragma unroll
-      for (int mtile = 0; mtile < 2; ++mtile)
-#pragma unroll
-        for (int cb = 0; cb < 4; ++cb)
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            biasv[mtile][cb][r] =
-                __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-                    bias_rows[mtile][r])[kv0 + cb * 16 + lr]);
-    }
-    if (HAS_MASK) {
-#pragma unroll
-      for (int cb = 0; cb < 4; ++cb)
-        maskv[cb] = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-            mask_row)[kv0 + cb * 16 + lr]);
-    }
     f32x4 s[2][4], dp[2][4];
 #pragma unroll
     for (int cb = 0; cb < 4; ++cb) {
       f32x4 acc0 = {}, acc1 = {}, accd0 = {}, accd1 = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][ks], bkf[cb][ks],
-                                                       acc0, 0, 0, 0);
-        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][ks], bkf[cb][ks],
-                                                       acc1, 0, 0, 0);
-        accd0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[0][ks], bvf[cb][ks],
-                                                        accd0, 0, 0, 0);
-        accd1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[1][ks], bvf[cb][ks],
-                                                        accd1, 0, 0, 0);
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        const bf16x8 bvt = load_frag(
+            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][ks], bk, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][ks], bk, acc1, 0, 0, 0);
+        accd0 =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[0][ks], bvt, accd0, 0, 0, 0);
+        accd1 =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[1][ks], bvt, accd1, 0, 0, 0);
       }
       s[0][cb] = acc0;
       s[1][cb] = acc1;
@@ -874,11 +845,16 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     for (int mtile = 0; mtile < 2; ++mtile)
 #pragma unroll
       for (int cb = 0; cb < 4; ++cb) {
+        const int kv = kv0 + cb * 16 + lr;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float sv = s[mtile][cb][r];
-          if (HAS_BIAS) sv += biasv[mtile][cb][r];
-          if (HAS_MASK) sv += maskv[cb];
+          if (HAS_BIAS)
+            sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+                bias_rows[mtile][r])[kv]);
+          if (HAS_MASK)
+            sv += __bfloat162float(
+                reinterpret_cast<const __hip_bfloat16*>(mask_row)[kv]);
           s[mtile][cb][r] = __expf(sv - lse_r[mtile][r]);
         }
       }
@@ -1410,41 +1386,19 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
       }
     }
     __syncthreads();
-    // hoist the tile's global reads (Q/dO fragments + lse/di/bias
-    // scalars) ahead of the MFMAs so one latency wait covers them all
-    bf16x8 bqf[4][2], bdof[4][2];
-    float lse_t[4], di_t[4], biasv[4][4];
-#pragma unroll
-    for (int cq = 0; cq < 4; ++cq) {
-      const int qcol = q0 + cq * 16 + lr;
-      const int64_t row = (bh * L + qcol) * (int64_t)HD;
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        bqf[cq][ks] = load_frag(qp + row + ks * 32 + lg * 8);
-        bdof[cq][ks] = load_frag(dop + row + ks * 32 + lg * 8);
-      }
-      lse_t[cq] = lse[bh * L + qcol];
-      di_t[cq] = di[bh * L + qcol];
-      if (HAS_BIAS) {
-        const uint16_t* brow_h =
-            bias + (((bh / bias_od) % bias_nb) * bias_q + (qcol % bias_q)) *
-                       (int64_t)L;
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          biasv[cq][r] = __bfloat162float(
-              reinterpret_cast<const __hip_bfloat16*>(brow_h)[kv0w + lg * 4 + r]);
-      }
-    }
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
       f32x4 acc = {}, accd = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bqf[cq][ks], acc,
-                                                      0, 0, 0);
-        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdof[cq][ks],
-                                                       accd, 0, 0, 0);
+        const bf16x8 bq =
+            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq, acc, 0, 0, 0);
+        const bf16x8 bdo =
+            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo, accd, 0, 0, 0);
       }
       st[cq] = acc;
       dpt[cq] = accd;
@@ -1452,8 +1406,12 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
       const int qcol = q0 + cq * 16 + lr;
-      const float lse_c = lse_t[cq];
-      const float di_c = di_t[cq];
+      const float lse_c = lse[bh * L + qcol];
+      const float di_c = di[bh * L + qcol];
+      const uint16_t* brow =
+          HAS_BIAS ? bias + (((bh / bias_od) % bias_nb) * bias_q +
+                             (qcol % bias_q)) * (int64_t)L
+                   : nullptr;
       bool keep[4] = {true, true, true, true};
       if (DROP) {
         bool k8[8];
@@ -1466,7 +1424,9 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float sv = st[cq][r];
-        if (HAS_BIAS) sv += biasv[cq][r];
+        if (HAS_BIAS)
+          sv += __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
+              brow)[kv0w + lg * 4 + r]);
         sv += maskv[r];
         const float pv = __expf(sv - lse_c);
         float dpv = dpt[cq][r];
