@@ -762,31 +762,14 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
   const int lr = lane & 15;
   const int q0 = qt * 128 + wid * 32;   // wave's first q row (2 M-tiles)
 
+  // K^T is staged per kv-tile (8 KB) instead of full-L resident (64 KB):
+  // the resident version capped the block at 80 KB LDS -> 2 blocks/CU ->
+  // 2 waves/SIMD, and PMC showed 41% of wave cycles parked on waits —
+  // occupancy, not staging traffic, was the binding constraint.
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
-  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][LMAX];
-
-  {
-    const int st_kv0 = (int)threadIdx.x >> 2;
-    const int st_d0 = ((int)threadIdx.x & 3) * 16;
-    for (int c = 0; c < L / 64; ++c) {
-      const int kv = st_kv0 + c * 64;
-      float f0[8], f1[8];
-      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
-                (bh * L + kv) * (int64_t)HD + st_d0,
-            f0);
-      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
-                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
-            f1);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d0 = st_d0 + j;
-        const int d1 = st_d0 + 8 + j;
-        lds_kt[d0][kv ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
-        lds_kt[d1][kv ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
-      }
-    }
-  }
-  __syncthreads();
+  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][BN];
+  const int st_kv0 = (int)threadIdx.x >> 2;
+  const int st_d0 = ((int)threadIdx.x & 3) * 16;
 
   bf16x8 aq[2][2], ado[2][2];
   float lse_r[2][4], di_r[2][4], di_row[2];
@@ -819,6 +802,25 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
   const int n_tiles = L / BN;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
+    // cooperative stage of this tile's K^T (transposed + XOR-swizzled)
+    {
+      const int kv = kv0 + st_kv0;
+      float f0[8], f1[8];
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0,
+            f0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
+            f1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = st_d0 + j;
+        const int d1 = st_d0 + 8 + j;
+        lds_kt[d0][st_kv0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
+        lds_kt[d1][st_kv0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+      }
+    }
+    __syncthreads();
     f32x4 s[2][4], dp[2][4];
 #pragma unroll
     for (int cb = 0; cb < 4; ++cb) {
@@ -914,13 +916,14 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       for (int cb = 0; cb < 4; ++cb) {
         const int d = cb * 16 + lr;
         const bf16x8 bkf = load_frag(
-            &lds_kt[d][(kv0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
+            &lds_kt[d][(ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
         dq_acc[0][cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[0], bkf, dq_acc[0][cb], 0, 0, 0);
         dq_acc[1][cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[1], bkf, dq_acc[1][cb], 0, 0, 0);
       }
     }
+    __syncthreads();  // lds_kt is re-staged next iteration
   }
 #pragma unroll
   for (int mtile = 0; mtile < 2; ++mtile)
@@ -1307,7 +1310,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_q32_kernel(
 }
 
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
-__global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
+__global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
@@ -1324,30 +1327,15 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
   const int kv0w = kt * BN + wid * 16;
   const int64_t kvbase = (bh * L + kv0w) * HD;
 
+  // Both transposed operands (Q^T, dO^T) are staged per q-tile (8 KB
+  // each) rather than keeping dO^T L-resident (64 KB): the resident
+  // version held the block at 80 KB LDS -> 2 blocks/CU, and PMC showed
+  // the kernel parked on waits — occupancy was the binding constraint.
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][16][BN];
-  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][LMAX];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][BM];
   __shared__ __attribute__((aligned(16))) uint16_t lds_qt[HD][BM];
   const int st_q0 = (int)threadIdx.x >> 2;
   const int st_d0 = ((int)threadIdx.x & 3) * 16;
-
-  // dO^T resident (the dominant reuse); Q^T staged per q-tile below
-  {
-    for (int c = 0; c < L / 64; ++c) {
-      const int qq = st_q0 + c * 64;
-      const int64_t row = (bh * L + qq) * (int64_t)HD;
-      float f0[8], f1[8];
-      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, f0);
-      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, f1);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d0 = st_d0 + j;
-        const int d1 = st_d0 + 8 + j;
-        lds_dot[d0][qq ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
-        lds_dot[d1][qq ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
-      }
-    }
-  }
-  __syncthreads();
 
   bf16x8 ak[2], av[2];
 #pragma unroll
@@ -1370,19 +1358,23 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
   const int n_tiles = L / BM;
   for (int tq = 0; tq < n_tiles; ++tq) {
     const int q0 = tq * BM;
-    // stage this q-tile's Q^T (8 KB)
+    // stage this q-tile's Q^T and dO^T (8 KB each)
     {
       const int qq = q0 + st_q0;
       const int64_t row = (bh * L + qq) * (int64_t)HD;
-      float f0[8], f1[8];
+      float f0[8], f1[8], g0[8], g1[8];
       load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
       load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, g0);
+      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, g1);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int d0 = st_d0 + j;
         const int d1 = st_d0 + 8 + j;
         lds_qt[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
         lds_qt[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
+        lds_dot[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(g0[j]);
+        lds_dot[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(g1[j]);
       }
     }
     __syncthreads();
@@ -1457,10 +1449,9 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_qres_kernel(
 #pragma unroll
       for (int cb = 0; cb < 4; ++cb) {
         const int d = cb * 16 + lr;
-        const int qx_dot = (q0 + ks2 * 32 + lg * 8) ^ ((d & 7) << 3);
-        const int qx_q = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);  // tile-local
-        const bf16x8 bdo = load_frag(&lds_dot[d][qx_dot]);
-        const bf16x8 bqf = load_frag(&lds_qt[d][qx_q]);
+        const int qx = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);  // tile-local
+        const bf16x8 bdo = load_frag(&lds_dot[d][qx]);
+        const bf16x8 bqf = load_frag(&lds_qt[d][qx]);
         dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta[ks2], bdo,
                                                             dv_acc[cb], 0, 0, 0);
         dk_acc[cb] =
