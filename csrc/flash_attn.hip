@@ -767,7 +767,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
   // 2 waves/SIMD, and PMC showed 41% of wave cycles parked on waits —
   // occupancy, not staging traffic, was the binding constraint.
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][2][16][BN];
-  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[HD][BN];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_kt[2][HD][BN];
   const int st_kv0 = (int)threadIdx.x >> 2;
   const int st_d0 = ((int)threadIdx.x & 3) * 16;
 
@@ -800,27 +800,31 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
 
   f32x4 dq_acc[2][4] = {};
   const int n_tiles = L / BN;
+  // ping-pong K^T staging: stage tile t+1 into the other buffer while
+  // computing tile t, one barrier per iteration
+  auto stage_kt = [&](int tile) {
+    const int kv = tile * BN + st_kv0;
+    uint16_t* buf = &lds_kt[tile & 1][0][0];
+    float f0[8], f1[8];
+    load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+              (bh * L + kv) * (int64_t)HD + st_d0,
+          f0);
+    load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
+              (bh * L + kv) * (int64_t)HD + st_d0 + 8,
+          f1);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d0 = st_d0 + j;
+      const int d1 = st_d0 + 8 + j;
+      buf[d0 * BN + (st_kv0 ^ ((d0 & 7) << 3))] = f32_to_bf16_bits(f0[j]);
+      buf[d1 * BN + (st_kv0 ^ ((d1 & 7) << 3))] = f32_to_bf16_bits(f1[j]);
+    }
+  };
+  stage_kt(0);
+  __syncthreads();
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
-    // cooperative stage of this tile's K^T (transposed + XOR-swizzled)
-    {
-      const int kv = kv0 + st_kv0;
-      float f0[8], f1[8];
-      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
-                (bh * L + kv) * (int64_t)HD + st_d0,
-            f0);
-      load8(reinterpret_cast<const __hip_bfloat16*>(kp) +
-                (bh * L + kv) * (int64_t)HD + st_d0 + 8,
-            f1);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d0 = st_d0 + j;
-        const int d1 = st_d0 + 8 + j;
-        lds_kt[d0][st_kv0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
-        lds_kt[d1][st_kv0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
-      }
-    }
-    __syncthreads();
+    if (t + 1 < n_tiles) stage_kt(t + 1);
     f32x4 s[2][4], dp[2][4];
 #pragma unroll
     for (int cb = 0; cb < 4; ++cb) {
@@ -916,14 +920,16 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       for (int cb = 0; cb < 4; ++cb) {
         const int d = cb * 16 + lr;
         const bf16x8 bkf = load_frag(
-            &lds_kt[d][(ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
+            &lds_kt[t & 1][d][(ks2 * 32 + lg * 8) ^ ((d & 7) << 3)]);
         dq_acc[0][cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[0], bkf, dq_acc[0][cb], 0, 0, 0);
         dq_acc[1][cb] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa[1], bkf, dq_acc[1][cb], 0, 0, 0);
       }
     }
-    __syncthreads();  // lds_kt is re-staged next iteration
+    // publish tile t+1's staging; reads of buffer t&1 are also done, so
+    // its restage at t+2 (after the next barrier) is safe
+    __syncthreads();
   }
 #pragma unroll
   for (int mtile = 0; mtile < 2; ++mtile)
@@ -1332,8 +1338,8 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
   // version held the block at 80 KB LDS -> 2 blocks/CU, and PMC showed
   // the kernel parked on waits — occupancy was the binding constraint.
   __shared__ __attribute__((aligned(16))) uint16_t lds_t[4][16][BN];
-  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[HD][BM];
-  __shared__ __attribute__((aligned(16))) uint16_t lds_qt[HD][BM];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_dot[2][HD][BM];
+  __shared__ __attribute__((aligned(16))) uint16_t lds_qt[2][HD][BM];
   const int st_q0 = (int)threadIdx.x >> 2;
   const int st_d0 = ((int)threadIdx.x & 3) * 16;
 
@@ -1356,28 +1362,33 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
 
   f32x4 dk_acc[4] = {}, dv_acc[4] = {};
   const int n_tiles = L / BM;
+  // ping-pong staging of Q^T/dO^T: tile tq+1 stages into the other
+  // buffer while tile tq computes, one barrier per iteration
+  auto stage_q = [&](int tile) {
+    const int qq = tile * BM + st_q0;
+    const int64_t row = (bh * L + qq) * (int64_t)HD;
+    uint16_t* bq = &lds_qt[tile & 1][0][0];
+    uint16_t* bd = &lds_dot[tile & 1][0][0];
+    float f0[8], f1[8], g0[8], g1[8];
+    load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
+    load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
+    load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, g0);
+    load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, g1);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d0 = st_d0 + j;
+      const int d1 = st_d0 + 8 + j;
+      bq[d0 * BM + (st_q0 ^ ((d0 & 7) << 3))] = f32_to_bf16_bits(f0[j]);
+      bq[d1 * BM + (st_q0 ^ ((d1 & 7) << 3))] = f32_to_bf16_bits(f1[j]);
+      bd[d0 * BM + (st_q0 ^ ((d0 & 7) << 3))] = f32_to_bf16_bits(g0[j]);
+      bd[d1 * BM + (st_q0 ^ ((d1 & 7) << 3))] = f32_to_bf16_bits(g1[j]);
+    }
+  };
+  stage_q(0);
+  __syncthreads();
   for (int tq = 0; tq < n_tiles; ++tq) {
     const int q0 = tq * BM;
-    // stage this q-tile's Q^T and dO^T (8 KB each)
-    {
-      const int qq = q0 + st_q0;
-      const int64_t row = (bh * L + qq) * (int64_t)HD;
-      float f0[8], f1[8], g0[8], g1[8];
-      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0, f0);
-      load8(reinterpret_cast<const __hip_bfloat16*>(qp) + row + st_d0 + 8, f1);
-      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0, g0);
-      load8(reinterpret_cast<const __hip_bfloat16*>(dop) + row + st_d0 + 8, g1);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d0 = st_d0 + j;
-        const int d1 = st_d0 + 8 + j;
-        lds_qt[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(f0[j]);
-        lds_qt[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(f1[j]);
-        lds_dot[d0][st_q0 ^ ((d0 & 7) << 3)] = f32_to_bf16_bits(g0[j]);
-        lds_dot[d1][st_q0 ^ ((d1 & 7) << 3)] = f32_to_bf16_bits(g1[j]);
-      }
-    }
-    __syncthreads();
+    if (tq + 1 < n_tiles) stage_q(tq + 1);
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
@@ -1450,8 +1461,8 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
       for (int cb = 0; cb < 4; ++cb) {
         const int d = cb * 16 + lr;
         const int qx = (ks2 * 32 + lg * 8) ^ ((d & 7) << 3);  // tile-local
-        const bf16x8 bdo = load_frag(&lds_dot[d][qx]);
-        const bf16x8 bqf = load_frag(&lds_qt[d][qx]);
+        const bf16x8 bdo = load_frag(&lds_dot[tq & 1][d][qx]);
+        const bf16x8 bqf = load_frag(&lds_qt[tq & 1][d][qx]);
         dv_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pta[ks2], bdo,
                                                             dv_acc[cb], 0, 0, 0);
         dk_acc[cb] =
