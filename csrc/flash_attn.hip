@@ -826,6 +826,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     const int kv0 = t * BN;
     if (t + 1 < n_tiles) stage_kt(t + 1);
     f32x4 s[2][4], dp[2][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int cb = 0; cb < 4; ++cb) {
       f32x4 acc0 = {}, acc1 = {}, accd0 = {}, accd1 = {};
@@ -847,6 +848,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       dp[0][cb] = accd0;
       dp[1][cb] = accd1;
     }
+    __builtin_amdgcn_s_setprio(0);
 #pragma unroll
     for (int mtile = 0; mtile < 2; ++mtile)
 #pragma unroll
