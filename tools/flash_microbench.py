@@ -91,7 +91,7 @@ def long_seq():
     from unicore_amd.modules import softmax_dropout
 
     H, D = 8, 64
-    for L, B in ((2048, 8), (4096, 4), (8192, 2), (16384, 1)):
+    for L, B in ((1024, 16), (2048, 8), (4096, 4), (8192, 2), (16384, 1)):
         BH = B * H
         q = torch.randn(BH, L, D, device="cuda", dtype=torch.bfloat16) * 0.2
         k, v = torch.randn_like(q), torch.randn_like(q)
