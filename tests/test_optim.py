@@ -241,3 +241,55 @@ def test_fp16_flatten_roundtrip():
     for p in model.parameters():
         master = fopt.fp32_view_of(p)
         assert torch.equal(p.detach(), master.bfloat16())
+
+
+def test_lazy_grad_collection_matches_view_path():
+    """The single-process lazy path (autograd-assigned grads batch-copied
+    into the fp32 masters) must match the flat-view path step for step,
+    including grad accumulation and a parameter with no grad."""
+    import argparse
+    import copy
+
+    import torch
+
+    from unicore_amd.optim import FP16Optimizer
+
+    def make(seed):
+        torch.manual_seed(seed)
+        m = torch.nn.Sequential(
+            torch.nn.Linear(16, 32), torch.nn.Tanh(), torch.nn.Linear(32, 16),
+            torch.nn.Linear(16, 4),  # untouched in step 2 (no grad)
+        ).bfloat16()
+        return m
+
+    args = argparse.Namespace(
+        optimizer="adam", lr=[1e-2], adam_betas="(0.9, 0.98)", adam_eps=1e-8,
+        weight_decay=0.01, bf16=True, bf16_sr=False, fp16=False,
+        allreduce_fp32_grad=False, fp16_no_flatten_grads=False,
+        min_loss_scale=1e-4, fp16_scale_window=None, fp16_scale_tolerance=0.0,
+        fp16_init_scale=4, threshold_loss_scale=None, per_sample_clip_norm=0.0,
+        distributed_world_size=1, update_freq=[1],
+    )
+
+    ma, mb = make(7), make(7)
+    oa = FP16Optimizer.build_optimizer(args, list(ma.named_parameters()))
+    ob = FP16Optimizer.build_optimizer(args, list(mb.named_parameters()))
+    ob.enable_lazy_grad_collection()
+
+    x = torch.randn(8, 16).bfloat16()
+    for step in range(3):
+        for m, o in ((ma, oa), (mb, ob)):
+            o.zero_grad()
+            # grad accumulation: two backwards per step
+            for micro in range(2):
+                h = m[1](m[0](x + step * 0.1 + micro * 0.01))
+                if step == 2:
+                    loss = m[2](h).float().pow(2).mean()  # m[3] gets no grad
+                else:
+                    loss = m[3](m[2](h)).float().pow(2).mean()
+                o.backward(loss)
+            o.multiply_grads(0.5)
+            o.clip_grad_norm(1.0)
+            o.step()
+        for (na, pa), (nb, pb) in zip(ma.named_parameters(), mb.named_parameters()):
+            assert torch.equal(pa, pb), (step, na)

@@ -101,7 +101,8 @@ def flatten_fp32_master(params, set_to_param=False):
 class _FlatGroup:
     """One decay group: low-precision flats (per dtype) + one fp32 master."""
 
-    __slots__ = ["lp_flats", "lp_members", "fp32_flat", "weight_decay"]
+    __slots__ = ["lp_flats", "lp_members", "fp32_flat", "weight_decay",
+                 "lazy_pairs"]
 
     def __init__(self):
         self.lp_flats = []  # list of Parameter (one per dtype), each with .grad
@@ -168,6 +169,7 @@ class FP16Optimizer(UnicoreOptimizer):
         self._needs_sync = False
         self.bf16_sr = getattr(args, "bf16_sr", False)
         self._per_sample_pending = False
+        self._lazy = False
 
     @classmethod
     def build_optimizer(cls, args, named_params, **kwargs):
@@ -277,8 +279,46 @@ class FP16Optimizer(UnicoreOptimizer):
         self._needs_sync = True
 
     @torch.no_grad()
+    def enable_lazy_grad_collection(self):
+        """Single-process fast path: detach ``param.grad`` from the flat
+        views so autograd ASSIGNS each gradient once (instead of issuing
+        one small accumulate ``add_`` kernel per parameter per backward —
+        ~190 launches / 1.3 ms per step on BERT-base), then batch-copy
+        them straight into the fp32 master grads with one
+        ``_foreach_copy_`` per group.  The trainer enables this only when
+        no DDP engine overlaps all-reduce with backward (world size 1)
+        and per-sample clipping is off; the flat-view path stays the
+        default everywhere else."""
+        self._lazy = True
+        for g in self.groups:
+            g.fp32_flat.grad.zero_()  # padding stays zero from here on
+            pairs = []
+            for p, off, n in g.lp_members:
+                p.grad = None
+                pairs.append((p, g.fp32_flat.grad[off : off + n].view(*p.shape)))
+            g.lazy_pairs = pairs
+
+    @torch.no_grad()
+    def _collect_grads(self):
+        for g in self.groups:
+            dsts, srcs = [], []
+            for p, dst in g.lazy_pairs:
+                if p.grad is None:
+                    dst.zero_()
+                else:
+                    dsts.append(dst)
+                    srcs.append(p.grad)
+                    p.grad = None
+            if dsts:
+                torch._foreach_copy_(dsts, srcs)
+        self._needs_sync = False
+
+    @torch.no_grad()
     def _sync_lp_grads_to_fp32(self):
         if not self._needs_sync:
+            return
+        if self._lazy:
+            self._collect_grads()
             return
         for g in self.groups:
             off = 0
@@ -325,6 +365,7 @@ class FP16Optimizer(UnicoreOptimizer):
     def per_sample_clip_grad_norm(self, max_norm, aggregate_norm_fn=None):
         """Clip the current (single-sample) lp grads, then accumulate them
         into the fp32 grads and clear the lp grads."""
+        assert not self._lazy, "per-sample clipping requires the flat-view grad path"
         if max_norm <= 0.0:
             return 0.0
         lp_flat_params = [f for g in self.groups for f in g.lp_flats]
@@ -399,10 +440,17 @@ class FP16Optimizer(UnicoreOptimizer):
 
     def zero_grad(self):
         """Clears the gradients of all optimized parameters."""
-        for g in self.groups:
-            for f in g.lp_flats:
-                f.grad.zero_()
-            g.fp32_flat.grad.zero_()
+        if self._lazy:
+            # every fp32 region is either overwritten or zeroed at the
+            # next _collect_grads, so only drop the assigned tensors
+            for g in self.groups:
+                for p, _dst in g.lazy_pairs:
+                    p.grad = None
+        else:
+            for g in self.groups:
+                for f in g.lp_flats:
+                    f.grad.zero_()
+                g.fp32_flat.grad.zero_()
         if self.scaler is not None:
             self._multiply_factor = 1.0 / float(self.scaler.loss_scale)
         else:

@@ -244,6 +244,15 @@ class Trainer(object):
                     "please switch to FP32 which is likely to be faster"
                 )
             self._optimizer = optim.FP16Optimizer.build_optimizer(self.args, params)
+            if (
+                self.data_parallel_world_size == 1
+                and self.args.per_sample_clip_norm <= 0
+                and hasattr(self._optimizer, "enable_lazy_grad_collection")
+            ):
+                # no DDP engine hooks the flat grads: let autograd assign
+                # gradients and batch-copy them at sync time instead of
+                # one accumulate kernel per parameter per backward
+                self._optimizer.enable_lazy_grad_collection()
         else:
             if self.cuda and torch.cuda.get_device_capability(0)[0] >= 7:
                 logger.info("NOTE: your device may support faster training with --fp16")
