@@ -105,6 +105,17 @@ class FlatDDP(nn.Module):
         aliased = self._try_alias_flat_grads(params)
         if not aliased:
             self._allocate_own_buckets(params)
+        # Lazy grad collection (aliased flats only): detach param.grad so
+        # autograd ASSIGNS each gradient once instead of issuing one small
+        # accumulate add_ kernel per parameter per backward; each bucket
+        # batch-copies the assigned tensors into its flat range right
+        # before the all-reduce. The optimizer reads the flats (not
+        # param.grad) in the aliased configuration, so nothing downstream
+        # sees the difference.
+        self.lazy = aliased and self.world_size > 1
+        if self.lazy:
+            for p in params:
+                p.grad = None
 
         if len(self._buckets) > 0:
             nbytes = sum(b.flat.numel() * b.flat.element_size() for b in self._buckets)
@@ -203,8 +214,10 @@ class FlatDDP(nn.Module):
             bucket = self._param_to_bucket[param]
             # re-attach the grad view if something replaced it (e.g. a
             # zero_grad(set_to_none=True) outside our control)
-            if param.grad is not None and param.grad.data_ptr() != self._view_ptr(
-                bucket, param
+            if (
+                not self.lazy
+                and param.grad is not None
+                and param.grad.data_ptr() != self._view_ptr(bucket, param)
             ):
                 self._restore_view(bucket, param)
             bucket.pending -= 1
@@ -240,6 +253,20 @@ class FlatDDP(nn.Module):
         bucket.launched = True
         if self.world_size == 1:
             return
+        if self.lazy:
+            # land the autograd-assigned grads in the flat range (one
+            # batched copy on the compute stream, ahead of the event that
+            # orders the comm stream); unused params contribute zeros
+            dsts, srcs = [], []
+            for p, v in zip(bucket.params, bucket.views):
+                if p.grad is None:
+                    v.zero_()
+                else:
+                    dsts.append(v)
+                    srcs.append(p.grad)
+                    p.grad = None
+            if dsts:
+                torch._foreach_copy_(dsts, srcs)
         if self._comm_stream is not None:
             # compute stream -> comm stream ordering: the bucket's grads must
             # all have landed before the all-reduce reads them
@@ -264,6 +291,10 @@ class FlatDDP(nn.Module):
             b.launched = False
             b.work = None
             b.event = None
+            if self.lazy:
+                # grads stay autograd-assigned (and may hold accumulated
+                # no_sync micro-step sums); the bucket copy lands them
+                continue
             # re-pin any grads that were detached from their views
             for p, v in zip(b.params, b.views):
                 if p.grad is None or p.grad.data_ptr() != v.data_ptr():
@@ -292,7 +323,7 @@ class FlatDDP(nn.Module):
         for b in self._buckets:
             b.flat.zero_()
             for p, v in zip(b.params, b.views):
-                p.grad = v
+                p.grad = None if self.lazy else v
 
     # ------------------------------------------------------------------
     # nn.Module plumbing
