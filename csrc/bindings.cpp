@@ -48,6 +48,12 @@ std::vector<at::Tensor> cross_entropy_forward(at::Tensor logits, at::Tensor targ
 at::Tensor cross_entropy_backward(at::Tensor logits, at::Tensor target,
                                   at::Tensor lse, at::Tensor grad_scale,
                                   int64_t ignore_index);
+bool softmax_dropout_backward_bias_supported(int64_t n_batch, int64_t q,
+                                             int64_t k, int64_t bb, int64_t bq,
+                                             int64_t od);
+std::vector<at::Tensor> softmax_dropout_backward_bias(
+    at::Tensor grad_output, at::Tensor softmax_results, at::Tensor dropout_mask,
+    double dropout_prob, int64_t bb, int64_t bq, int64_t od);
 at::Tensor gaussian_basis_forward(at::Tensor coords, at::Tensor means,
                                   at::Tensor stds, at::ScalarType out_dtype);
 std::vector<at::Tensor> gaussian_basis_backward(at::Tensor dg, at::Tensor coords,
@@ -77,6 +83,11 @@ std::vector<at::Tensor> flash_attn_backward(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_dropout_forward", &softmax_dropout_forward,
         "fused softmax(+mask,+bias)+dropout forward (gfx950)");
+  m.def("softmax_dropout_backward_bias", &softmax_dropout_backward_bias,
+        "fused softmax backward + broadcast bias gradient");
+  m.def("softmax_dropout_backward_bias_supported",
+        &softmax_dropout_backward_bias_supported,
+        "fused bias-grad backward supports this shape");
   m.def("softmax_dropout_backward", &softmax_dropout_backward,
         "fused softmax+dropout backward, in-place on grad");
   m.def("layernorm_forward", &layernorm_forward, "fused LayerNorm forward");

@@ -202,6 +202,94 @@ __global__ void softmax_bwd_vec_kernel(T* __restrict__ g, const T* __restrict__ 
   }
 }
 
+// Backward with the broadcast-bias gradient fused in: one block (4 waves)
+// per bias row; the waves split the rows that reduce onto it (for BERT's
+// (1,H,L,L) rel-pos bias that is the batch dimension), each wave computes
+// those rows' grad_input in place (identical math to the vec kernel) while
+// accumulating the bias-row sum in registers; a deterministic LDS fold
+// writes the fp32 bias grad.  Replaces the eager `.sum(dim=(0,2))` that
+// re-read the full (B,H,L,L) grad tensor (604 MB/layer at BERT-base).
+template <typename T, int NV, bool DROP>
+__global__ void softmax_bwd_biasgrad_kernel(
+    T* __restrict__ g, const T* __restrict__ y,
+    const uint8_t* __restrict__ dmask, float* __restrict__ dbias,
+    int k, float pinv, int q, int bb, int bq, int od, int a) {
+  extern __shared__ float s_red[];  // [4][k]
+  const int64_t brow = blockIdx.x;  // 0 .. bb*bq-1
+  const int jb = (int)(brow / bq);
+  const int qb = (int)(brow % bq);
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int mrow_bytes = k / 8;
+  const int qrep = q / bq;
+  const int64_t group = (int64_t)a * od * qrep;
+
+  float acc[NV][8];
+#pragma unroll
+  for (int i = 0; i < NV; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[i][j] = 0.f;
+
+  for (int64_t gidx = wid; gidx < group; gidx += 4) {
+    const int64_t t = gidx / ((int64_t)od * qrep);
+    const int64_t rem = gidx - t * od * qrep;
+    const int e = (int)(rem / qrep);
+    const int m = (int)(rem - (int64_t)e * qrep);
+    const int64_t rb = ((int64_t)t * bb + jb) * od + e;
+    const int64_t row = rb * q + qb + (int64_t)m * bq;
+
+    T* grow = g + row * (int64_t)k;
+    const T* yrow = y + row * (int64_t)k;
+    float tv[NV][8], yv[NV][8];
+    float ssum = 0.f;
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        load8(grow + e0, tv[i]);
+        load8(yrow + e0, yv[i]);
+        if constexpr (DROP) {
+          const uint8_t bits = dmask[row * (int64_t)mrow_bytes + lane + i * 64];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tv[i][j] = (bits >> j) & 1 ? tv[i][j] * pinv : 0.f;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ssum += tv[i][j] * yv[i][j];
+      }
+    }
+    ssum = wave_sum(ssum);
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int e0 = (lane + i * 64) * 8;
+      if (e0 < k) {
+        float dx[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          dx[j] = yv[i][j] * (tv[i][j] - ssum);
+          acc[i][j] += dx[j];
+        }
+        store8(grow + e0, dx);
+      }
+    }
+  }
+
+  float* my = s_red + (int64_t)wid * k;
+#pragma unroll
+  for (int i = 0; i < NV; ++i) {
+    const int e0 = (lane + i * 64) * 8;
+    if (e0 < k) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) my[e0 + j] = acc[i][j];
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < k; c += blockDim.x) {
+    dbias[brow * (int64_t)k + c] =
+        s_red[c] + s_red[k + c] + s_red[2 * k + c] + s_red[3 * k + c];
+  }
+}
+
 __device__ __forceinline__ float block_red_max(float v, float* red) {
   v = wave_max(v);
   const int wid = threadIdx.x / 64;
@@ -486,4 +574,63 @@ at::Tensor softmax_dropout_backward(at::Tensor grad_output, at::Tensor softmax_r
   }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return grad_output;
+}
+
+bool softmax_dropout_backward_bias_supported(int64_t n_batch, int64_t q,
+                                             int64_t k, int64_t bb, int64_t bq,
+                                             int64_t od) {
+  if (k % 8 != 0 || k > 2048) return false;          // LDS fold <= 32 KB
+  if (bq <= 0 || q % bq != 0) return false;
+  if (bb <= 0 || od <= 0 || n_batch % (bb * od) != 0) return false;
+  const int64_t group = (n_batch / (bb * od)) * od * (q / bq);
+  return group > 1 && bb * bq >= 512;                // enough blocks to fill
+}
+
+std::vector<at::Tensor> softmax_dropout_backward_bias(
+    at::Tensor grad_output, at::Tensor softmax_results, at::Tensor dropout_mask,
+    double dropout_prob, int64_t bb, int64_t bq, int64_t od) {
+  TORCH_CHECK(grad_output.is_cuda() && grad_output.is_contiguous(),
+              "softmax_dropout_backward_bias: grad must be contiguous CUDA");
+  TORCH_CHECK(grad_output.dim() == 3, "grad must be (n, q, k)");
+  const int64_t n_batch = grad_output.size(0);
+  const int q = (int)grad_output.size(1);
+  const int k = (int)grad_output.size(2);
+  TORCH_CHECK(
+      softmax_dropout_backward_bias_supported(n_batch, q, k, bb, bq, od),
+      "softmax_dropout_backward_bias: unsupported shape");
+  const int a = (int)(n_batch / (bb * od));
+  const bool drop = dropout_mask.defined() && dropout_mask.numel() > 0;
+  const float pinv =
+      drop ? (float)(1.0 / (1.0 - std::min(dropout_prob, 0.999999))) : 1.f;
+  auto dbias = torch::empty({bb * bq, k},
+                            grad_output.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const size_t lds = 4 * (size_t)k * sizeof(float);
+  DISPATCH_FTYPES(grad_output.scalar_type(), "softmax_dropout_backward_bias", {
+    auto launch = [&](auto nv_tag, auto drop_tag) {
+      constexpr int NV = decltype(nv_tag)::value;
+      constexpr bool DROP = decltype(drop_tag)::value;
+      softmax_bwd_biasgrad_kernel<scalar_t, NV, DROP>
+          <<<(int)(bb * bq), 256, lds, stream>>>(
+              reinterpret_cast<scalar_t*>(grad_output.data_ptr()),
+              reinterpret_cast<const scalar_t*>(softmax_results.data_ptr()),
+              drop ? dropout_mask.data_ptr<uint8_t>() : nullptr,
+              dbias.data_ptr<float>(), k, pinv, q, (int)bb, (int)bq, (int)od,
+              a);
+    };
+    auto pick_nv = [&](auto drop_tag) {
+      if (k <= 512)
+        launch(std::integral_constant<int, 1>{}, drop_tag);
+      else if (k <= 1024)
+        launch(std::integral_constant<int, 2>{}, drop_tag);
+      else
+        launch(std::integral_constant<int, 4>{}, drop_tag);
+    };
+    if (drop)
+      pick_nv(std::true_type{});
+    else
+      pick_nv(std::false_type{});
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return {grad_output, dbias};
 }

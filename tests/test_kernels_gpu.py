@@ -694,3 +694,59 @@ def test_gaussian_pair_bias_deterministic():
     c = ops.gaussian_pair_bias_bwd(dbias, coords, means, stds, W, None)
     for x, y in zip(a, c):
         assert torch.equal(x, y)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("p", [0.0, 0.3])
+def test_softmax_bwd_fused_bias_grad(dtype, p):
+    """BERT-shaped broadcast bias ((1,H,q,k), bb*bq >= 512): backward must
+    take the fused bias-grad kernel and match the eager fp32 chain."""
+    from unicore_amd import ops
+    from unicore_amd.modules import softmax_dropout
+
+    B, H, q, k = 8, 8, 64, 256
+    assert ops.softmax_dropout_bwd_bias_supported(B * H, q, k, H, q, 1)
+    torch.manual_seed(11)
+    x = torch.randn(B, H, q, k, device="cuda", dtype=dtype)
+    bias = torch.randn(1, H, q, k, device="cuda", dtype=dtype,
+                       requires_grad=True)
+    xk = x.clone().requires_grad_(True)
+    torch.manual_seed(123)
+    out = softmax_dropout(xk, p, is_training=True, bias=bias, inplace=False)
+    gout = torch.randn_like(out)
+    out.backward(gout.clone())
+
+    # eager reference with the same dropout mask (recovered from out)
+    xr = x.detach().float().clone().requires_grad_(True)
+    br = bias.detach().float().clone().requires_grad_(True)
+    yr = F.softmax(xr + br, dim=-1)
+    if p > 0:
+        keep = (out.detach() != 0).float() / (1.0 - p)
+        (yr * keep).backward(gout.float())
+    else:
+        yr.backward(gout.float())
+
+    tol = TOL[dtype]
+    assert (xk.grad.float() - xr.grad).abs().max().item() < tol * 10
+    bscale = br.grad.abs().max().item() + 1e-3
+    assert (bias.grad.float() - br.grad).abs().max().item() / bscale < tol * 10
+
+
+@requires_gpu
+def test_softmax_bwd_fused_bias_grad_deterministic():
+    from unicore_amd import ops
+
+    torch.manual_seed(12)
+    B, H, q, k = 8, 8, 64, 256
+    g = torch.randn(B * H, q, k, device="cuda", dtype=torch.bfloat16)
+    y = torch.softmax(torch.randn_like(g).float(), -1).to(torch.bfloat16)
+    d1 = ops.softmax_dropout_bwd_bias(
+        g.clone(), y, torch.empty(0, dtype=torch.uint8, device="cuda"),
+        0.0, H, q, 1,
+    )[1]
+    d2 = ops.softmax_dropout_bwd_bias(
+        g.clone(), y, torch.empty(0, dtype=torch.uint8, device="cuda"),
+        0.0, H, q, 1,
+    )[1]
+    assert torch.equal(d1, d2)

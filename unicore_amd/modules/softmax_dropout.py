@@ -85,6 +85,7 @@ class SoftmaxDropoutFast(torch.autograd.Function):
         if bias is not None:
             ctx.bias_shape = bias.shape
             ctx.bias_outer_div = bias_outer_div
+            ctx.bias_dtype = bias.dtype
         ctx.save_for_backward(softmax_results, dropout_mask)
         return dropout_results
 
@@ -94,13 +95,30 @@ class SoftmaxDropoutFast(torch.autograd.Function):
 
         softmax_results, dropout_mask = ctx.saved_tensors
         grad_output = grad_output.contiguous()
+        grad_bias = None
+        want_bias_grad = ctx.has_bias and ctx.needs_input_grad[4]
+        if want_bias_grad:
+            n_batch, q, k = softmax_results.shape
+            bb, bq, _ = ctx.bias_shape
+            od = ctx.bias_outer_div
+            if softmax_results.is_cuda and ops.softmax_dropout_bwd_bias_supported(
+                n_batch, q, k, bb, bq, od
+            ):
+                # fused: the backward kernel accumulates the bias grad
+                # while it writes grad_input, instead of re-reading the
+                # whole grad tensor for the eager .sum
+                grad_input, dbias = ops.softmax_dropout_bwd_bias(
+                    grad_output, softmax_results, dropout_mask,
+                    ctx.dropout_prob, bb, bq, od,
+                )
+                grad_bias = dbias.view(bb, bq, k).to(ctx.bias_dtype)
+                return None, grad_input, None, None, grad_bias, None, None
         # in-place on grad_output (same contract as the reference backward,
         # reference csrc/softmax_dropout/softmax_dropout_kernel.cu:277-278)
         grad_input = ops.softmax_dropout_bwd(
             grad_output, softmax_results, dropout_mask, ctx.dropout_prob
         )
-        grad_bias = None
-        if ctx.has_bias:
+        if want_bias_grad:
             n_batch, q, k = softmax_results.shape
             bb, bq, _ = ctx.bias_shape
             od = ctx.bias_outer_div

@@ -240,3 +240,18 @@ def gaussian_pair_bias_supported(n_kernels, n_heads) -> bool:
     return _kernels is not None and _kernels.gaussian_pair_bias_supported(
         int(n_kernels), int(n_heads)
     )
+
+
+def softmax_dropout_bwd_bias(grad, softmax_out, dmask, p, bb, bq, od):
+    """In-place softmax backward that also returns the broadcast bias grad
+    (fp32, (bb*bq, k)) without re-reading the grad tensor."""
+    require_kernels()
+    return _kernels.softmax_dropout_backward_bias(
+        grad, softmax_out, dmask, float(p), int(bb), int(bq), int(od)
+    )
+
+
+def softmax_dropout_bwd_bias_supported(n_batch, q, k, bb, bq, od) -> bool:
+    return _kernels is not None and _kernels.softmax_dropout_backward_bias_supported(
+        int(n_batch), int(q), int(k), int(bb), int(bq), int(od)
+    )
