@@ -29,18 +29,25 @@ void fused_adam(at::Tensor p, at::Tensor m, at::Tensor v, at::Tensor g, double l
                 int64_t step, bool bias_correction, double weight_decay);
 at::Tensor multi_tensor_l2norm(int64_t chunk_size, std::vector<at::Tensor> tensors);
 void fp32_to_bf16_sr(at::Tensor src, at::Tensor dst);
-std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv, int64_t num_heads,
-                                          double scale);
-at::Tensor qkv_split_backward(at::Tensor dq, at::Tensor dk, at::Tensor dv,
-                              int64_t B, int64_t num_heads, double scale);
-std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
-                                             bool is_training);
-at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
-                                 double p);
+std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv,
+                                          std::optional<at::Tensor> bias,
+                                          int64_t num_heads, double scale);
+std::vector<at::Tensor> qkv_split_backward(at::Tensor dq, at::Tensor dk,
+                                           at::Tensor dv, int64_t B,
+                                           int64_t num_heads, double scale,
+                                           bool bias_grad);
+std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x,
+                                             std::optional<at::Tensor> bias,
+                                             double p, bool is_training);
+std::vector<at::Tensor> gelu_dropout_backward(at::Tensor grad, at::Tensor x,
+                                              std::optional<at::Tensor> bias,
+                                              at::Tensor dmask, double p);
 at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
-std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double p,
-                                            bool is_training);
-at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p);
+std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res,
+                                            std::optional<at::Tensor> bias,
+                                            double p, bool is_training);
+std::vector<at::Tensor> dropout_add_backward(at::Tensor grad, at::Tensor dmask,
+                                             double p, int64_t bias_dim);
 at::Tensor embedding_backward(at::Tensor grad, at::Tensor indices,
                               int64_t num_embeddings, int64_t padding_idx);
 std::vector<at::Tensor> cross_entropy_forward(at::Tensor logits, at::Tensor target,

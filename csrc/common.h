@@ -193,3 +193,75 @@ static inline int unicore_grid(int64_t want, int cap = 2048) {
   if (want < 1) return 1;
   return (int)(want < cap ? want : cap);
 }
+
+// ---------------------------------------------------------------------------
+// Deterministic per-block column sums for flat elementwise kernels over a
+// (rows, C) tensor.  The host picks a grid with stride*8 % C == 0 so each
+// thread owns 8 fixed columns; the block fold serializes the (at most
+// ceil(2048/C)) contributor bands so no two threads ever add to the same
+// LDS slot in the same round.
+// ---------------------------------------------------------------------------
+static inline long long unicore_gcd_ll(long long a, long long b) {
+  while (b) {
+    long long t = a % b;
+    a = b;
+    b = t;
+  }
+  return a;
+}
+
+static inline bool colsum_supported(int64_t C) {
+  if (C <= 0 || C % 8 != 0 || C > 4096) return false;
+  return (C / unicore_gcd_ll(2048, C)) <= 256;
+}
+
+static inline int colsum_grid(int64_t n8, int64_t C) {
+  const int64_t mult = C / unicore_gcd_ll(2048, C);
+  const int64_t want = (n8 + 255) / 256;
+  int64_t blocks = ((want + mult - 1) / mult) * mult;
+  int64_t cap = (2048 / mult) * mult;
+  if (cap < mult) cap = mult;
+  if (blocks > cap) blocks = cap;
+  return (int)blocks;
+}
+
+// acc[8] holds this thread's partial sums for columns c0..c0+7 (c0 < 0 =>
+// thread had no elements); s_col is C floats of LDS.
+__device__ __forceinline__ void colsum_block_fold(const float (&acc)[8], int c0,
+                                                  int C, float* s_col,
+                                                  float* partials_row) {
+  for (int c = (int)threadIdx.x; c < C; c += (int)blockDim.x) s_col[c] = 0.f;
+  __syncthreads();
+  const int band = ((int)threadIdx.x * 8) / C;
+  const int nbands = ((int)blockDim.x * 8 + C - 1) / C;
+  for (int m = 0; m < nbands; ++m) {
+    if (band == m && c0 >= 0) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s_col[c0 + j] += acc[j];
+    }
+    __syncthreads();
+  }
+  for (int c = (int)threadIdx.x; c < C; c += (int)blockDim.x)
+    partials_row[c] = s_col[c];
+}
+
+// partials (nb, C) -> out (C); one block per column, deterministic.
+static __global__ void unicore_col_fold_kernel(const float* __restrict__ partials,
+                                               float* __restrict__ out, int nb,
+                                               int C) {
+  __shared__ float s_red[4];
+  const int c = blockIdx.x;
+  if (c >= C) return;
+  float v = 0.f;
+  for (int r = (int)threadIdx.x; r < nb; r += (int)blockDim.x)
+    v += partials[(int64_t)r * C + c];
+  v = wave_sum(v);
+  const int wave = (int)threadIdx.x / 64, lane = (int)threadIdx.x % 64;
+  if (lane == 0) s_red[wave] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float r = 0.f;
+    for (int w = 0; w < (int)blockDim.x / 64; ++w) r += s_red[w];
+    out[c] = r;
+  }
+}

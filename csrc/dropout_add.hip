@@ -10,15 +10,18 @@
 #include <ATen/cuda/CUDAContext.h>
 #include <ATen/cuda/CUDAGeneratorImpl.h>
 
+#include <optional>
 #include <vector>
 
 namespace {
 
-template <typename T, bool DROP>
+template <typename T, bool DROP, bool HAS_BIAS>
 __global__ void dropout_add_fwd_kernel(T* __restrict__ out,
                                        uint8_t* __restrict__ dmask,
                                        const T* __restrict__ x,
-                                       const T* __restrict__ res, int64_t n8,
+                                       const T* __restrict__ res,
+                                       const T* __restrict__ bias, int C,
+                                       int64_t n8,
                                        float pinv, uint32_t pthresh,
                                        uint64_t seed, uint64_t offset) {
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -27,6 +30,12 @@ __global__ void dropout_add_fwd_kernel(T* __restrict__ out,
     float fx[8], fr[8];
     load8(x + i * 8, fx);
     load8(res + i * 8, fr);
+    if constexpr (HAS_BIAS) {
+      float fb[8];
+      load8(bias + (int)((i * 8) % C), fb);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) fx[j] += fb[j];
+    }
     if constexpr (DROP) {
       bool keep[8];
       keep16x8(seed + offset * 0x9E3779B97F4A7C15ull, (uint64_t)i, 0, pthresh,
@@ -46,19 +55,36 @@ __global__ void dropout_add_fwd_kernel(T* __restrict__ out,
   }
 }
 
-template <typename T>
+template <typename T, bool DROP, bool BGRAD>
 __global__ void dropout_add_bwd_kernel(T* __restrict__ dx, const T* __restrict__ g,
                                        const uint8_t* __restrict__ dmask,
+                                       float* __restrict__ partials, int C,
                                        int64_t n8, float pinv) {
+  extern __shared__ float s_col[];
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
-       i += stride) {
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  int c0 = -1;
+  for (int64_t i = tid; i < n8; i += stride) {
+    if (BGRAD && c0 < 0) c0 = (int)((i * 8) % C);
     float f[8];
     load8(g + i * 8, f);
-    const uint8_t bits = dmask[i];
+    if constexpr (DROP) {
+      const uint8_t bits = dmask[i];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) f[j] = (bits >> j) & 1 ? f[j] * pinv : 0.f;
-    store8(dx + i * 8, f);
+      for (int j = 0; j < 8; ++j) f[j] = (bits >> j) & 1 ? f[j] * pinv : 0.f;
+      store8(dx + i * 8, f);
+    }
+    if constexpr (BGRAD) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += f[j];
+    }
+  }
+  if constexpr (BGRAD) {
+    colsum_block_fold(acc, c0, C, s_col,
+                      partials + (int64_t)blockIdx.x * C);
   }
 }
 
@@ -85,8 +111,9 @@ __global__ void dropout_add_bwd_kernel(T* __restrict__ dx, const T* __restrict__
 
 }  // namespace
 
-std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double p,
-                                            bool is_training) {
+std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res,
+                                            std::optional<at::Tensor> bias,
+                                            double p, bool is_training) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && res.is_contiguous(),
               "dropout_add: contiguous CUDA");
   TORCH_CHECK(x.sizes() == res.sizes() && x.scalar_type() == res.scalar_type(),
@@ -116,40 +143,83 @@ std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res, double
   } else {
     dmask = at::empty({0}, x.options().dtype(at::kByte));
   }
+  const bool has_bias = bias.has_value();
+  at::Tensor bc;
+  int C = 0;
+  if (has_bias) {
+    bc = bias->contiguous();
+    C = (int)bc.numel();
+    TORCH_CHECK(C > 0 && C % 8 == 0 && x.size(-1) == C &&
+                    bc.scalar_type() == x.scalar_type(),
+                "dropout_add: bad bias");
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   const int grid = unicore_grid((n8 + 255) / 256);
   DISPATCH_FTYPES(x.scalar_type(), "dropout_add_forward", {
-    if (drop)
-      dropout_add_fwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
-          reinterpret_cast<scalar_t*>(out.data_ptr()), dmask.data_ptr<uint8_t>(),
+    auto launch = [&](auto drop_tag, auto bias_tag) {
+      constexpr bool DROP = decltype(drop_tag)::value;
+      constexpr bool HB = decltype(bias_tag)::value;
+      dropout_add_fwd_kernel<scalar_t, DROP, HB><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()),
+          DROP ? dmask.data_ptr<uint8_t>() : nullptr,
           reinterpret_cast<const scalar_t*>(x.data_ptr()),
-          reinterpret_cast<const scalar_t*>(res.data_ptr()), n8, pinv, pthresh,
-          seed, offset);
-    else
-      dropout_add_fwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
-          reinterpret_cast<scalar_t*>(out.data_ptr()), nullptr,
-          reinterpret_cast<const scalar_t*>(x.data_ptr()),
-          reinterpret_cast<const scalar_t*>(res.data_ptr()), n8, pinv, pthresh,
-          seed, offset);
+          reinterpret_cast<const scalar_t*>(res.data_ptr()),
+          HB ? reinterpret_cast<const scalar_t*>(bc.data_ptr()) : nullptr, C,
+          n8, pinv, pthresh, seed, offset);
+    };
+    if (drop) {
+      if (has_bias) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else {
+      if (has_bias) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    }
   });
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {out, dmask};
 }
 
-at::Tensor dropout_add_backward(at::Tensor grad, at::Tensor dmask, double p) {
+std::vector<at::Tensor> dropout_add_backward(at::Tensor grad, at::Tensor dmask,
+                                             double p, int64_t bias_dim) {
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous(), "dropout_add_backward");
   const int64_t n8 = grad.numel() / 8;
-  TORCH_CHECK(dmask.numel() == n8, "dropout_add_backward: mask mismatch");
+  const bool drop = dmask.numel() > 0;
+  TORCH_CHECK(!drop || dmask.numel() == n8, "dropout_add_backward: mask mismatch");
+  const bool bgrad = bias_dim > 0;
+  const int C = (int)bias_dim;
+  TORCH_CHECK(!bgrad || colsum_supported(C), "dropout_add_backward: bad bias dim");
   const float pinv = (float)(1.0 / (1.0 - std::min(p, 0.999999)));
-  auto dx = at::empty_like(grad);
+  // !drop: dx == grad, the caller aliases it; only the colsum runs
+  auto dx = drop ? at::empty_like(grad) : grad;
+  auto dbias = at::empty({bgrad ? (int64_t)C : 0},
+                         grad.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = unicore_grid((n8 + 255) / 256);
+  const int grid = bgrad ? colsum_grid(n8, C) : unicore_grid((n8 + 255) / 256);
+  at::Tensor partials;
+  if (bgrad) partials = at::empty({grid, (int64_t)C}, dbias.options());
+  const size_t lds = bgrad ? (size_t)C * sizeof(float) : 0;
   DISPATCH_FTYPES(grad.scalar_type(), "dropout_add_backward", {
-    dropout_add_bwd_kernel<scalar_t><<<grid, 256, 0, stream>>>(
-        reinterpret_cast<scalar_t*>(dx.data_ptr()),
-        reinterpret_cast<const scalar_t*>(grad.data_ptr()),
-        dmask.data_ptr<uint8_t>(), n8, pinv);
+    auto launch = [&](auto drop_tag, auto bg_tag) {
+      constexpr bool DROP = decltype(drop_tag)::value;
+      constexpr bool BG = decltype(bg_tag)::value;
+      if (!DROP && !BG) return;  // nothing to do
+      dropout_add_bwd_kernel<scalar_t, DROP, BG><<<grid, 256, lds, stream>>>(
+          reinterpret_cast<scalar_t*>(dx.data_ptr()),
+          reinterpret_cast<const scalar_t*>(grad.data_ptr()),
+          DROP ? dmask.data_ptr<uint8_t>() : nullptr,
+          BG ? partials.data_ptr<float>() : nullptr, C, n8, pinv);
+    };
+    if (drop) {
+      if (bgrad) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else if (bgrad) {
+      launch(std::false_type{}, std::true_type{});
+    }
   });
+  if (bgrad) {
+    unicore_col_fold_kernel<<<C, 256, 0, stream>>>(
+        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, C);
+  }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
-  return dx;
+  return {dx, dbias};
 }

@@ -11,6 +11,7 @@
 #include <ATen/cuda/CUDAContext.h>
 #include <ATen/cuda/CUDAGeneratorImpl.h>
 
+#include <optional>
 #include <vector>
 
 namespace {
@@ -25,10 +26,12 @@ __device__ __forceinline__ float gelu_grad(float x) {
   return cdf + x * pdf;
 }
 
-template <typename T, bool DROP>
+template <typename T, bool DROP, bool HAS_BIAS>
 __global__ void gelu_dropout_fwd_kernel(T* __restrict__ out,
                                         uint8_t* __restrict__ dmask,
-                                        const T* __restrict__ x, int64_t n8,
+                                        const T* __restrict__ x,
+                                        const T* __restrict__ bias, int C,
+                                        int64_t n8,
                                         float pinv, uint32_t pthresh,
                                         uint64_t seed, uint64_t offset) {
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -36,6 +39,12 @@ __global__ void gelu_dropout_fwd_kernel(T* __restrict__ out,
   for (int64_t i = tid; i < n8; i += stride) {
     float f[8];
     load8(x + i * 8, f);
+    if constexpr (HAS_BIAS) {
+      float fb[8];
+      load8(bias + (int)((i * 8) % C), fb);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] += fb[j];
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) f[j] = gelu_fwd(f[j]);
     if constexpr (DROP) {
@@ -54,17 +63,32 @@ __global__ void gelu_dropout_fwd_kernel(T* __restrict__ out,
   }
 }
 
-template <typename T, bool DROP>
+template <typename T, bool DROP, bool BGRAD>
 __global__ void gelu_dropout_bwd_kernel(T* __restrict__ dx, const T* __restrict__ g,
                                         const T* __restrict__ x,
+                                        const T* __restrict__ bias,
                                         const uint8_t* __restrict__ dmask,
+                                        float* __restrict__ partials, int C,
                                         int64_t n8, float pinv) {
+  extern __shared__ float s_col[];
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
-       i += stride) {
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  int c0 = -1;
+  for (int64_t i = tid; i < n8; i += stride) {
     float gv[8], xv[8];
     load8(g + i * 8, gv);
     load8(x + i * 8, xv);
+    if constexpr (BGRAD) {
+      // the saved x is the bias-free Linear output: rebuild gelu's input
+      if (c0 < 0) c0 = (int)((i * 8) % C);
+      float fb[8];
+      load8(bias + c0, fb);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xv[j] += fb[j];
+    }
     uint8_t bits = 0xFF;
     if constexpr (DROP) bits = dmask[i];
 #pragma unroll
@@ -72,8 +96,13 @@ __global__ void gelu_dropout_bwd_kernel(T* __restrict__ dx, const T* __restrict_
       float t = gv[j];
       if constexpr (DROP) t = (bits >> j) & 1 ? t * pinv : 0.f;
       gv[j] = t * gelu_grad(xv[j]);
+      if constexpr (BGRAD) acc[j] += gv[j];
     }
     store8(dx + i * 8, gv);
+  }
+  if constexpr (BGRAD) {
+    colsum_block_fold(acc, c0, C, s_col,
+                      partials + (int64_t)blockIdx.x * C);
   }
 }
 
@@ -100,8 +129,9 @@ __global__ void gelu_dropout_bwd_kernel(T* __restrict__ dx, const T* __restrict_
 
 }  // namespace
 
-std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
-                                             bool is_training) {
+std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x,
+                                             std::optional<at::Tensor> bias,
+                                             double p, bool is_training) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "gelu_dropout: contiguous CUDA");
   TORCH_CHECK(x.numel() % 8 == 0, "gelu_dropout: numel % 8 == 0");
   const int64_t n8 = x.numel() / 8;
@@ -128,48 +158,90 @@ std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x, double p,
   } else {
     dmask = at::empty({0}, x.options().dtype(at::kByte));
   }
+  const bool has_bias = bias.has_value();
+  at::Tensor bc;
+  int C = 0;
+  if (has_bias) {
+    bc = bias->contiguous();
+    C = (int)bc.numel();
+    TORCH_CHECK(C > 0 && C % 8 == 0 && x.size(-1) == C &&
+                    bc.scalar_type() == x.scalar_type(),
+                "gelu_dropout: bad bias");
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   const int grid = unicore_grid((n8 + 255) / 256);
   DISPATCH_FTYPES(x.scalar_type(), "gelu_dropout_forward", {
-    if (drop)
-      gelu_dropout_fwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
-          reinterpret_cast<scalar_t*>(out.data_ptr()), dmask.data_ptr<uint8_t>(),
-          reinterpret_cast<const scalar_t*>(x.data_ptr()), n8, pinv, pthresh, seed,
-          offset);
-    else
-      gelu_dropout_fwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
-          reinterpret_cast<scalar_t*>(out.data_ptr()), nullptr,
-          reinterpret_cast<const scalar_t*>(x.data_ptr()), n8, pinv, pthresh, seed,
-          offset);
+    auto launch = [&](auto drop_tag, auto bias_tag) {
+      constexpr bool DROP = decltype(drop_tag)::value;
+      constexpr bool HB = decltype(bias_tag)::value;
+      gelu_dropout_fwd_kernel<scalar_t, DROP, HB><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()),
+          DROP ? dmask.data_ptr<uint8_t>() : nullptr,
+          reinterpret_cast<const scalar_t*>(x.data_ptr()),
+          HB ? reinterpret_cast<const scalar_t*>(bc.data_ptr()) : nullptr, C,
+          n8, pinv, pthresh, seed, offset);
+    };
+    if (drop) {
+      if (has_bias) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else {
+      if (has_bias) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    }
   });
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {out, dmask};
 }
 
-at::Tensor gelu_dropout_backward(at::Tensor grad, at::Tensor x, at::Tensor dmask,
-                                 double p) {
+std::vector<at::Tensor> gelu_dropout_backward(at::Tensor grad, at::Tensor x,
+                                              std::optional<at::Tensor> bias,
+                                              at::Tensor dmask, double p) {
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() && x.is_contiguous(),
               "gelu_dropout_backward: contiguous CUDA");
   const int64_t n8 = x.numel() / 8;
   const bool drop = dmask.defined() && dmask.numel() > 0;
+  const bool bgrad = bias.has_value();
+  at::Tensor bc;
+  int C = 0;
+  if (bgrad) {
+    bc = bias->contiguous();
+    C = (int)bc.numel();
+    TORCH_CHECK(colsum_supported(C), "gelu_dropout_backward: bad bias dim");
+  }
   const float pinv =
       drop ? (float)(1.0 / (1.0 - std::min(p, 0.999999))) : 1.f;
   auto dx = at::empty_like(x);
+  auto dbias = at::empty({bgrad ? (int64_t)C : 0},
+                         grad.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = unicore_grid((n8 + 255) / 256);
+  const int grid = bgrad ? colsum_grid(n8, C) : unicore_grid((n8 + 255) / 256);
+  at::Tensor partials;
+  if (bgrad) partials = at::empty({grid, (int64_t)C}, dbias.options());
+  const size_t lds = bgrad ? (size_t)C * sizeof(float) : 0;
   DISPATCH_FTYPES(x.scalar_type(), "gelu_dropout_backward", {
-    if (drop)
-      gelu_dropout_bwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+    auto launch = [&](auto drop_tag, auto bg_tag) {
+      constexpr bool DROP = decltype(drop_tag)::value;
+      constexpr bool BG = decltype(bg_tag)::value;
+      gelu_dropout_bwd_kernel<scalar_t, DROP, BG><<<grid, 256, lds, stream>>>(
           reinterpret_cast<scalar_t*>(dx.data_ptr()),
           reinterpret_cast<const scalar_t*>(grad.data_ptr()),
           reinterpret_cast<const scalar_t*>(x.data_ptr()),
-          dmask.data_ptr<uint8_t>(), n8, pinv);
-    else
-      gelu_dropout_bwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
-          reinterpret_cast<scalar_t*>(dx.data_ptr()),
-          reinterpret_cast<const scalar_t*>(grad.data_ptr()),
-          reinterpret_cast<const scalar_t*>(x.data_ptr()), nullptr, n8, pinv);
+          BG ? reinterpret_cast<const scalar_t*>(bc.data_ptr()) : nullptr,
+          DROP ? dmask.data_ptr<uint8_t>() : nullptr,
+          BG ? partials.data_ptr<float>() : nullptr, C, n8, pinv);
+    };
+    if (drop) {
+      if (bgrad) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else {
+      if (bgrad) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    }
   });
+  if (bgrad) {
+    unicore_col_fold_kernel<<<C, 256, 0, stream>>>(
+        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, C);
+  }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
-  return dx;
+  return {dx, dbias};
 }

@@ -12,16 +12,19 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <optional>
 #include <vector>
 
 namespace {
 
-template <typename T>
+template <typename T, bool HAS_BIAS>
 __global__ void qkv_split_fwd_kernel(const T* __restrict__ qkv, T* __restrict__ q,
                                      T* __restrict__ k, T* __restrict__ v,
+                                     const T* __restrict__ bias,
                                      int64_t n8, int L, int H, int D8,
                                      float scale) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int C = 3 * H * D8 * 8;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
        i += stride) {
     int64_t tmp = i;
@@ -36,6 +39,12 @@ __global__ void qkv_split_fwd_kernel(const T* __restrict__ qkv, T* __restrict__ 
 
     float f[8];
     load8(qkv + i * 8, f);
+    if constexpr (HAS_BIAS) {
+      float fb[8];
+      load8(bias + (int)((i * 8) % C), fb);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] += fb[j];
+    }
     T* out = t == 0 ? q : (t == 1 ? k : v);
     if (t == 0) {
 #pragma unroll
@@ -46,12 +55,19 @@ __global__ void qkv_split_fwd_kernel(const T* __restrict__ qkv, T* __restrict__ 
   }
 }
 
-template <typename T>
+template <typename T, bool BGRAD>
 __global__ void qkv_split_bwd_kernel(T* __restrict__ dqkv, const T* __restrict__ dq,
                                      const T* __restrict__ dk,
-                                     const T* __restrict__ dv, int64_t n8, int L,
-                                     int H, int D8, float scale) {
+                                     const T* __restrict__ dv,
+                                     float* __restrict__ partials, int64_t n8,
+                                     int L, int H, int D8, float scale) {
+  extern __shared__ float s_col[];
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int C = 3 * H * D8 * 8;
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  int c0 = -1;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
        i += stride) {
     int64_t tmp = i;
@@ -72,7 +88,16 @@ __global__ void qkv_split_bwd_kernel(T* __restrict__ dqkv, const T* __restrict__
 #pragma unroll
       for (int j = 0; j < 8; ++j) f[j] *= scale;
     }
+    if constexpr (BGRAD) {
+      if (c0 < 0) c0 = (int)((i * 8) % C);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += f[j];
+    }
     store8(dqkv + i * 8, f);
+  }
+  if constexpr (BGRAD) {
+    colsum_block_fold(acc, c0, C, s_col,
+                      partials + (int64_t)blockIdx.x * C);
   }
 }
 
@@ -99,8 +124,9 @@ __global__ void qkv_split_bwd_kernel(T* __restrict__ dqkv, const T* __restrict__
 
 }  // namespace
 
-std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv, int64_t num_heads,
-                                          double scale) {
+std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv,
+                                          std::optional<at::Tensor> bias,
+                                          int64_t num_heads, double scale) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3,
               "qkv_split: expected contiguous (B, L, 3E)");
   const int64_t B = qkv.size(0);
@@ -118,19 +144,38 @@ std::vector<at::Tensor> qkv_split_forward(at::Tensor qkv, int64_t num_heads,
   const int64_t n8 = qkv.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
   const int grid = unicore_grid((n8 + 255) / 256);
+  const bool has_bias = bias.has_value();
+  at::Tensor bc;
+  if (has_bias) {
+    bc = bias->contiguous();
+    TORCH_CHECK(bc.numel() == E3 && bc.scalar_type() == qkv.scalar_type(),
+                "qkv_split: bad bias");
+  }
   DISPATCH_FTYPES(qkv.scalar_type(), "qkv_split_forward", {
-    qkv_split_fwd_kernel<scalar_t><<<grid, 256, 0, stream>>>(
-        reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
-        reinterpret_cast<scalar_t*>(q.data_ptr()),
-        reinterpret_cast<scalar_t*>(k.data_ptr()),
-        reinterpret_cast<scalar_t*>(v.data_ptr()), n8, L, H, D8, (float)scale);
+    if (has_bias)
+      qkv_split_fwd_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+          reinterpret_cast<scalar_t*>(q.data_ptr()),
+          reinterpret_cast<scalar_t*>(k.data_ptr()),
+          reinterpret_cast<scalar_t*>(v.data_ptr()),
+          reinterpret_cast<const scalar_t*>(bc.data_ptr()), n8, L, H, D8,
+          (float)scale);
+    else
+      qkv_split_fwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<const scalar_t*>(qkv.data_ptr()),
+          reinterpret_cast<scalar_t*>(q.data_ptr()),
+          reinterpret_cast<scalar_t*>(k.data_ptr()),
+          reinterpret_cast<scalar_t*>(v.data_ptr()), nullptr, n8, L, H, D8,
+          (float)scale);
   });
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {q, k, v};
 }
 
-at::Tensor qkv_split_backward(at::Tensor dq, at::Tensor dk, at::Tensor dv,
-                              int64_t B, int64_t num_heads, double scale) {
+std::vector<at::Tensor> qkv_split_backward(at::Tensor dq, at::Tensor dk,
+                                           at::Tensor dv, int64_t B,
+                                           int64_t num_heads, double scale,
+                                           bool bias_grad) {
   TORCH_CHECK(dq.is_cuda() && dq.is_contiguous() && dk.is_contiguous() &&
                   dv.is_contiguous(),
               "qkv_split_backward: grads must be contiguous CUDA");
@@ -138,18 +183,36 @@ at::Tensor qkv_split_backward(at::Tensor dq, at::Tensor dk, at::Tensor dv,
   const int L = (int)dq.size(1);
   const int D = (int)dq.size(2);
   TORCH_CHECK(D % 8 == 0, "qkv_split_backward: head_dim % 8");
-  auto dqkv = at::empty({B, L, 3LL * H * D}, dq.options());
+  const int64_t C = 3LL * H * D;
+  const bool bgrad = bias_grad && colsum_supported(C);
+  auto dqkv = at::empty({B, L, C}, dq.options());
+  auto dbias = at::empty({bgrad ? C : 0}, dq.options().dtype(at::kFloat));
   const int64_t n8 = dqkv.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = unicore_grid((n8 + 255) / 256);
+  const int grid = bgrad ? colsum_grid(n8, C) : unicore_grid((n8 + 255) / 256);
+  at::Tensor partials;
+  if (bgrad) partials = at::empty({grid, C}, dbias.options());
+  const size_t lds = bgrad ? (size_t)C * sizeof(float) : 0;
   DISPATCH_FTYPES(dq.scalar_type(), "qkv_split_backward", {
-    qkv_split_bwd_kernel<scalar_t><<<grid, 256, 0, stream>>>(
-        reinterpret_cast<scalar_t*>(dqkv.data_ptr()),
-        reinterpret_cast<const scalar_t*>(dq.data_ptr()),
-        reinterpret_cast<const scalar_t*>(dk.data_ptr()),
-        reinterpret_cast<const scalar_t*>(dv.data_ptr()), n8, L, H, D / 8,
-        (float)scale);
+    if (bgrad)
+      qkv_split_bwd_kernel<scalar_t, true><<<grid, 256, lds, stream>>>(
+          reinterpret_cast<scalar_t*>(dqkv.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dq.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dk.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dv.data_ptr()),
+          partials.data_ptr<float>(), n8, L, H, D / 8, (float)scale);
+    else
+      qkv_split_bwd_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(dqkv.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dq.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dk.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dv.data_ptr()), nullptr, n8, L, H,
+          D / 8, (float)scale);
   });
+  if (bgrad) {
+    unicore_col_fold_kernel<<<(int)C, 256, 0, stream>>>(
+        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, (int)C);
+  }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
-  return dqkv;
+  return {dqkv, dbias};
 }

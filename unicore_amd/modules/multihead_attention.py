@@ -10,6 +10,7 @@ bias and key-padding mask into softmax.
 from typing import Optional
 
 import torch
+import torch.nn.functional as F
 from torch import Tensor, nn
 
 import os
@@ -95,24 +96,27 @@ class _QKVSplit(torch.autograd.Function):
     chunk + 3x transpose-contiguous + scale chain and its backward cat)."""
 
     @staticmethod
-    def forward(ctx, qkv, num_heads, scale):
+    def forward(ctx, qkv, bias, num_heads, scale):
         from unicore_amd import ops
 
-        q, k, v = ops.qkv_split_fwd(qkv, num_heads, scale)
+        q, k, v = ops.qkv_split_fwd(qkv, num_heads, scale, bias)
         ctx.num_heads = num_heads
         ctx.scale = scale
         ctx.bsz = qkv.shape[0]
+        ctx.bias_dtype = bias.dtype if bias is not None else None
         return q, k, v
 
     @staticmethod
     def backward(ctx, dq, dk, dv):
         from unicore_amd import ops
 
-        dqkv = ops.qkv_split_bwd(
+        dqkv, db = ops.qkv_split_bwd(
             dq.contiguous(), dk.contiguous(), dv.contiguous(),
             ctx.bsz, ctx.num_heads, ctx.scale,
+            bias_grad=ctx.bias_dtype is not None,
         )
-        return dqkv, None, None
+        dbias = db.to(ctx.bias_dtype) if ctx.bias_dtype is not None else None
+        return dqkv, dbias, None, None
 
 
 class SelfMultiheadAttention(nn.Module):
@@ -144,19 +148,34 @@ class SelfMultiheadAttention(nn.Module):
         key_padding_mask: Optional[Tensor] = None,
         attn_bias: Optional[Tensor] = None,
         return_attn: bool = False,
+        skip_out_bias: bool = False,
     ) -> Tensor:
+        """``skip_out_bias``: run out_proj without its bias — the caller
+        folds it into the following fused dropout+residual op (which then
+        also produces the bias gradient)."""
         bsz, tgt_len, embed_dim = query.size()
         assert embed_dim == self.embed_dim
 
-        qkv = self.in_proj(query)
         use_fused_split = False
-        if qkv.is_cuda and self.head_dim % 8 == 0:
+        if query.is_cuda and self.head_dim % 8 == 0:
             from unicore_amd import ops
 
             use_fused_split = ops.gpu_kernels_available()
         if use_fused_split:
-            q, k, v = _QKVSplit.apply(qkv.contiguous(), self.num_heads, self.scaling)
+            # bias-free GEMM: the split kernel adds the bias (free) and its
+            # backward emits the bias grad as a deterministic column sum
+            in_bias = self.in_proj.bias
+            fold_bias = in_bias is not None and ops.colsum_supported(
+                in_bias.numel()
+            )
+            qkv = F.linear(query, self.in_proj.weight,
+                           None if fold_bias else in_bias)
+            q, k, v = _QKVSplit.apply(
+                qkv.contiguous(), in_bias if fold_bias else None,
+                self.num_heads, self.scaling,
+            )
         else:
+            qkv = self.in_proj(query)
             q, k, v = qkv.chunk(3, dim=-1)
             q = (
                 q.view(bsz, tgt_len, self.num_heads, self.head_dim)
@@ -260,7 +279,10 @@ class SelfMultiheadAttention(nn.Module):
             .contiguous()
             .view(bsz, tgt_len, embed_dim)
         )
-        o = self.out_proj(o)
+        if skip_out_bias:
+            o = F.linear(o, self.out_proj.weight)
+        else:
+            o = self.out_proj(o)
         if not return_attn:
             return o
         else:

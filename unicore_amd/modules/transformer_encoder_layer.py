@@ -68,6 +68,25 @@ class TransformerEncoderLayer(nn.Module):
         LayerNorm is applied either before or after the self-attention/ffn
         modules similar to the original Transformer implementation.
         """
+        # On GPU with the kernel extension, the Linears run bias-free and
+        # each bias rides the following fused op (add is free there, and
+        # the backward emits the bias grad as a deterministic column sum
+        # instead of an eager activation-sized .sum re-read per Linear).
+        fold_bias = False
+        if x.is_cuda:
+            from unicore_amd import ops
+
+            fold_bias = (
+                ops.gpu_kernels_available()
+                and self.self_attn.out_proj.bias is not None
+                and ops.colsum_supported(self.self_attn.out_proj.bias.numel())
+                and self.fc1.bias is not None
+                and ops.colsum_supported(self.fc1.bias.numel())
+                and self.fc2.bias is not None
+                and ops.colsum_supported(self.fc2.bias.numel())
+                and self._fuse_gelu
+            )
+
         residual = x
         if not self.post_ln:
             x = self.self_attn_layer_norm(x)
@@ -76,24 +95,37 @@ class TransformerEncoderLayer(nn.Module):
             key_padding_mask=padding_mask,
             attn_bias=attn_bias,
             return_attn=return_attn,
+            skip_out_bias=fold_bias,
         )
         if return_attn:
             x, attn_weights, attn_probs = x
-        x = dropout_add(x, residual, self.dropout, self.training)
+        x = dropout_add(
+            x, residual, self.dropout, self.training,
+            bias=self.self_attn.out_proj.bias if fold_bias else None,
+        )
         if self.post_ln:
             x = self.self_attn_layer_norm(x)
 
         residual = x
         if not self.post_ln:
             x = self.final_layer_norm(x)
-        x = self.fc1(x)
-        if self._fuse_gelu and x.is_cuda:
-            x = gelu_dropout(x, self.activation_dropout, self.training)
+        if fold_bias:
+            x = F.linear(x, self.fc1.weight)
+            x = gelu_dropout(x, self.activation_dropout, self.training,
+                             bias=self.fc1.bias)
+            x = F.linear(x, self.fc2.weight)
+            x = dropout_add(x, residual, self.dropout, self.training,
+                            bias=self.fc2.bias)
         else:
-            x = self.activation_fn(x)
-            x = F.dropout(x, p=self.activation_dropout, training=self.training)
-        x = self.fc2(x)
-        x = dropout_add(x, residual, self.dropout, self.training)
+            x = self.fc1(x)
+            if self._fuse_gelu and x.is_cuda:
+                x = gelu_dropout(x, self.activation_dropout, self.training)
+            else:
+                x = self.activation_fn(x)
+                x = F.dropout(x, p=self.activation_dropout,
+                              training=self.training)
+            x = self.fc2(x)
+            x = dropout_add(x, residual, self.dropout, self.training)
         if self.post_ln:
             x = self.final_layer_norm(x)
         if not return_attn:

@@ -137,24 +137,28 @@ def fused_fp32_to_bf16_sr(src_fp32, dst_bf16):
     _kernels.fp32_to_bf16_sr(src_fp32, dst_bf16)
 
 
-def qkv_split_fwd(qkv, num_heads, scale):
+def qkv_split_fwd(qkv, num_heads, scale, bias=None):
     require_kernels()
-    return _kernels.qkv_split_forward(qkv, int(num_heads), float(scale))
+    return _kernels.qkv_split_forward(qkv, bias, int(num_heads), float(scale))
 
 
-def qkv_split_bwd(dq, dk, dv, bsz, num_heads, scale):
+def qkv_split_bwd(dq, dk, dv, bsz, num_heads, scale, bias_grad=False):
     require_kernels()
-    return _kernels.qkv_split_backward(dq, dk, dv, int(bsz), int(num_heads), float(scale))
+    return _kernels.qkv_split_backward(
+        dq, dk, dv, int(bsz), int(num_heads), float(scale), bool(bias_grad)
+    )
 
 
-def gelu_dropout_fwd(x, p, is_training):
+def gelu_dropout_fwd(x, p, is_training, bias=None):
     require_kernels()
-    return _kernels.gelu_dropout_forward(x, float(p), bool(is_training))
+    return _kernels.gelu_dropout_forward(x, bias, float(p), bool(is_training))
 
 
-def gelu_dropout_bwd(grad, x, dmask, p):
+def gelu_dropout_bwd(grad, x, dmask, p, bias=None):
+    # bias given: also return the fp32 column-sum bias grad (x is then the
+    # bias-free Linear output; the kernel re-adds bias for the gelu grad)
     require_kernels()
-    return _kernels.gelu_dropout_backward(grad, x, dmask, float(p))
+    return _kernels.gelu_dropout_backward(grad, x, bias, dmask, float(p))
 
 
 def flash_attn_fwd(q, k, v, bias, bias_outer_div, mask, mask_outer_div, p, is_training):
@@ -174,14 +178,23 @@ def flash_attn_bwd(d_out, q, k, v, o, lse, bias, bias_outer_div, bias_needs_grad
     )
 
 
-def dropout_add_fwd(x, res, p, is_training):
+def dropout_add_fwd(x, res, p, is_training, bias=None):
     require_kernels()
-    return _kernels.dropout_add_forward(x, res, float(p), bool(is_training))
+    return _kernels.dropout_add_forward(x, res, bias, float(p), bool(is_training))
 
 
-def dropout_add_bwd(grad, dmask, p):
+def dropout_add_bwd(grad, dmask, p, bias_dim=0):
+    # bias_dim > 0: also return the fp32 column-sum bias grad
     require_kernels()
-    return _kernels.dropout_add_backward(grad, dmask, float(p))
+    return _kernels.dropout_add_backward(grad, dmask, float(p), int(bias_dim))
+
+
+def colsum_supported(C) -> bool:
+    import math
+
+    if C <= 0 or C % 8 != 0 or C > 4096:
+        return False
+    return (C // math.gcd(2048, C)) <= 256
 
 
 def embedding_bwd(grad, indices, num_embeddings, padding_idx):
