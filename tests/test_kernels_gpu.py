@@ -309,7 +309,7 @@ def test_qkv_split_parity(dtype):
     assert torch.allclose(q.float(), refq.float(), atol=1e-6)
 
     dq, dk, dv = (torch.randn_like(q) for _ in range(3))
-    dqkv = ops.qkv_split_bwd(dq, dk, dv, B, H, scale)
+    dqkv, _ = ops.qkv_split_bwd(dq, dk, dv, B, H, scale)
     ref_dqkv = torch.cat(
         [
             (dq * scale).view(B, H, L, D).permute(0, 2, 1, 3).reshape(B, L, H * D),
@@ -762,35 +762,39 @@ def test_softmax_bwd_fused_bias_grad_deterministic():
 @pytest.mark.parametrize("p", [0.0, 0.3])
 @pytest.mark.parametrize("C", [768, 3072])
 def test_dropout_add_bias_parity(dtype, p, C):
-    from unicore_amd.modules import dropout_add
+    from unicore_amd import ops
 
     torch.manual_seed(21)
     x = torch.randn(64, C, device="cuda", dtype=dtype)
     res = torch.randn_like(x)
     b = torch.randn(C, device="cuda", dtype=dtype)
-    xk = x.clone().requires_grad_(True)
-    bk = b.clone().requires_grad_(True)
     torch.manual_seed(77)
-    out = dropout_add(xk, res, p, True, bias=bk)
-    g = torch.randn_like(out)
-    out.backward(g.clone())
-
-    keep = None
+    out, dmask = ops.dropout_add_fwd(x, res, p, True, b)
     if p > 0:
-        # recover the kernel's mask: kept entries satisfy out != res
-        keep = (out != res).to(torch.float32) / (1.0 - p)
+        # bit j of byte i covers element i*8+j
+        keep = torch.zeros(64 * C, device="cuda")
+        flat_bits = dmask.to(torch.int32)
+        for j in range(8):
+            keep[j::8] = ((flat_bits >> j) & 1).float()
+        keep = keep.view(64, C) / (1.0 - p)
+    else:
+        keep = None
+
     xr = x.detach().float().clone().requires_grad_(True)
     br = b.detach().float().clone().requires_grad_(True)
     y = xr + br
-    y = y * keep if keep is not None else y
-    (y + res.float()).backward(g.float())
+    if keep is not None:
+        y = y * keep
+    refo = y + res.float()
+    g = torch.randn(64, C, device="cuda", dtype=dtype)
+    refo.backward(g.float())
+
     tol = TOL[dtype]
-    assert (out.float() - (
-        ((x + b).float() * (keep if keep is not None else 1)) + res.float()
-    )).abs().max().item() < tol * 4
-    assert (xk.grad.float() - xr.grad).abs().max().item() < tol * 4
+    assert (out.float() - refo.detach()).abs().max().item() < tol * 4
+    dx, db = ops.dropout_add_bwd(g.contiguous(), dmask, p, C)
+    assert (dx.float() - xr.grad).abs().max().item() < tol * 4
     bs = br.grad.abs().max().item() + 1e-3
-    assert (bk.grad.float() - br.grad).abs().max().item() / bs < tol * 10
+    assert (db - br.grad).abs().max().item() / bs < tol * 10
 
 
 @requires_gpu
