@@ -5,6 +5,8 @@ Functional parity with reference unicore/modules/transformer_encoder_layer.py:56
 
 from typing import Optional
 
+import os
+
 import torch.nn.functional as F
 from torch import Tensor, nn
 
@@ -73,7 +75,7 @@ class TransformerEncoderLayer(nn.Module):
         # the backward emits the bias grad as a deterministic column sum
         # instead of an eager activation-sized .sum re-read per Linear).
         fold_bias = False
-        if x.is_cuda:
+        if x.is_cuda and os.environ.get("UNICORE_FOLD_BIAS", "1") == "1":
             from unicore_amd import ops
 
             fold_bias = (
