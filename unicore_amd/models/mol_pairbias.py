@@ -12,6 +12,8 @@ model-extension API: model registry + add_args + arch registry
 (reference unicore/models/__init__.py:17-102 pattern).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -24,7 +26,9 @@ from unicore_amd.models import (
 from unicore_amd.modules import (
     RMSNorm,
     SelfMultiheadAttention,
+    dropout_add,
     gaussian_basis,
+    gelu_dropout,
     init_bert_params,
 )
 from unicore_amd.modules.gaussian import (
@@ -100,18 +104,41 @@ class PairBiasEncoderLayer(nn.Module):
         self.activation_dropout = activation_dropout
 
     def forward(self, x, bias, padding_mask):
+        # same fused-op + bias-folding structure as TransformerEncoderLayer
+        fold = False
+        if x.is_cuda and os.environ.get("UNICORE_FOLD_BIAS", "1") == "1":
+            from unicore_amd import ops
+
+            fold = (
+                ops.gpu_kernels_available()
+                and self.attn.out_proj.bias is not None
+                and ops.colsum_supported(self.attn.out_proj.bias.numel())
+                and self.fc1.bias is not None
+                and ops.colsum_supported(self.fc1.bias.numel())
+                and self.fc2.bias is not None
+                and ops.colsum_supported(self.fc2.bias.numel())
+            )
         residual = x
         x = self.attn_norm(x)
-        x = self.attn(x, attn_bias=bias)
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        x = residual + x
+        x = self.attn(x, attn_bias=bias, skip_out_bias=fold)
+        x = dropout_add(
+            x, residual, self.dropout, self.training,
+            bias=self.attn.out_proj.bias if fold else None,
+        )
         residual = x
         x = self.ffn_norm(x)
-        x = F.gelu(self.fc1(x))
-        x = F.dropout(x, p=self.activation_dropout, training=self.training)
-        x = self.fc2(x)
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        return residual + x
+        if fold:
+            x = F.linear(x, self.fc1.weight)
+            x = gelu_dropout(x, self.activation_dropout, self.training,
+                             bias=self.fc1.bias)
+            x = F.linear(x, self.fc2.weight)
+            x = dropout_add(x, residual, self.dropout, self.training,
+                            bias=self.fc2.bias)
+        else:
+            x = gelu_dropout(self.fc1(x), self.activation_dropout, self.training)
+            x = self.fc2(x)
+            x = dropout_add(x, residual, self.dropout, self.training)
+        return x
 
 
 @register_model("mol_pairbias")
