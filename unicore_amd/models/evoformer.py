@@ -10,6 +10,8 @@ framework's fused LayerNorm.  Intended to run bf16 + stochastic rounding
 with grad accumulation (exercising the rounding and multi-tensor kernels).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -19,8 +21,25 @@ from unicore_amd.models import (
     register_model,
     register_model_architecture,
 )
-from unicore_amd.modules import LayerNorm, softmax_dropout
+from unicore_amd.modules import (
+    LayerNorm,
+    dropout_add,
+    gelu_dropout,
+    softmax_dropout,
+)
 from unicore_amd.modules.embedding import Embedding
+
+
+def _fold_ok(*biases):
+    """GPU bias folding into the fused elementwise ops (see
+    transformer_encoder_layer.py): all biases present and colsum-able."""
+    if os.environ.get("UNICORE_FOLD_BIAS", "1") != "1":
+        return False
+    from unicore_amd import ops
+
+    if not ops.gpu_kernels_available():
+        return False
+    return all(b is not None and ops.colsum_supported(b.numel()) for b in biases)
 
 
 class MSARowAttentionWithPairBias(nn.Module):
@@ -37,7 +56,7 @@ class MSARowAttentionWithPairBias(nn.Module):
         self.dropout = dropout
         self.scaling = self.head_dim**-0.5
 
-    def forward(self, msa, pair):
+    def forward(self, msa, pair, skip_out_bias=False):
         B, S, L, D = msa.shape
         H, Dh = self.heads, self.head_dim
         x = self.norm(msa)
@@ -64,6 +83,8 @@ class MSARowAttentionWithPairBias(nn.Module):
             .reshape(B, S, L, D)
         )
         g = torch.sigmoid(self.gate(x))
+        if skip_out_bias:
+            return F.linear(o * g, self.out.weight)
         return self.out(o * g)
 
 
@@ -79,7 +100,7 @@ class MSAColumnAttention(nn.Module):
         self.dropout = dropout
         self.scaling = self.head_dim**-0.5
 
-    def forward(self, msa):
+    def forward(self, msa, skip_out_bias=False):
         B, S, L, D = msa.shape
         H, Dh = self.heads, self.head_dim
         x = self.norm(msa)
@@ -103,6 +124,8 @@ class MSAColumnAttention(nn.Module):
             .reshape(B, S, L, D)
         )
         g = torch.sigmoid(self.gate(x))
+        if skip_out_bias:
+            return F.linear(o * g, self.out.weight)
         return self.out(o * g)
 
 
@@ -113,8 +136,17 @@ class Transition(nn.Module):
         self.fc1 = nn.Linear(d, d * mult)
         self.fc2 = nn.Linear(d * mult, d)
 
-    def forward(self, x):
-        return self.fc2(F.gelu(self.fc1(self.norm(x))))
+    def forward(self, x, residual=None):
+        y = self.norm(x)
+        if residual is not None and _fold_ok(self.fc1.bias, self.fc2.bias) \
+                and y.is_cuda:
+            y = F.linear(y, self.fc1.weight)
+            y = gelu_dropout(y, 0.0, self.training, bias=self.fc1.bias)
+            y = F.linear(y, self.fc2.weight)
+            return dropout_add(y, residual, 0.0, self.training,
+                               bias=self.fc2.bias)
+        y = self.fc2(F.gelu(self.fc1(y)))
+        return y if residual is None else residual + y
 
 
 class OuterProductMean(nn.Module):
@@ -126,12 +158,14 @@ class OuterProductMean(nn.Module):
         self.out = nn.Linear(c * c, d_pair)
         self.c = c
 
-    def forward(self, msa):
+    def forward(self, msa, skip_out_bias=False):
         x = self.norm(msa)
         a = self.a(x)  # (B, S, L, c)
         b = self.b(x)
         o = torch.einsum("bsic,bsjd->bijcd", a.float(), b.float()) / msa.shape[1]
         o = o.reshape(*o.shape[:3], self.c * self.c).to(msa.dtype)
+        if skip_out_bias:
+            return F.linear(o, self.out.weight)
         return self.out(o)
 
 
@@ -172,13 +206,35 @@ class EvoformerBlock(nn.Module):
         self.pair_transition = Transition(d_pair)
 
     def forward(self, msa, pair):
-        msa = msa + self.row_attn(msa, pair)
-        msa = msa + self.col_attn(msa)
-        msa = msa + self.msa_transition(msa)
-        pair = pair + self.opm(msa)
+        # residual joins go through the fused dropout_add (p=0) with the
+        # producing Linear's bias folded in where the module tail allows
+        # (the triangle-mult tails are gated, so their bias cannot move)
+        fold = msa.is_cuda and _fold_ok(
+            self.row_attn.out.bias, self.col_attn.out.bias, self.opm.out.bias
+        )
+        if fold:
+            msa = dropout_add(
+                self.row_attn(msa, pair, skip_out_bias=True), msa, 0.0,
+                self.training, bias=self.row_attn.out.bias,
+            )
+            msa = dropout_add(
+                self.col_attn(msa, skip_out_bias=True), msa, 0.0,
+                self.training, bias=self.col_attn.out.bias,
+            )
+        else:
+            msa = msa + self.row_attn(msa, pair)
+            msa = msa + self.col_attn(msa)
+        msa = self.msa_transition(msa, residual=msa)
+        if fold:
+            pair = dropout_add(
+                self.opm(msa, skip_out_bias=True), pair, 0.0,
+                self.training, bias=self.opm.out.bias,
+            )
+        else:
+            pair = pair + self.opm(msa)
         pair = pair + self.tri_out(pair)
         pair = pair + self.tri_in(pair)
-        pair = pair + self.pair_transition(pair)
+        pair = self.pair_transition(pair, residual=pair)
         return msa, pair
 
 
