@@ -36,6 +36,8 @@ std::vector<at::Tensor> qkv_split_backward(at::Tensor dq, at::Tensor dk,
                                            at::Tensor dv, int64_t B,
                                            int64_t num_heads, double scale,
                                            bool bias_grad);
+at::Tensor attn_merge(at::Tensor x, int64_t B, int64_t num_heads,
+                      bool inverse);
 std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x,
                                              std::optional<at::Tensor> bias,
                                              double p, bool is_training);
@@ -112,6 +114,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused QKV head-split + q-scale -> (q, k, v) each (B*H, L, D)");
   m.def("qkv_split_backward", &qkv_split_backward,
         "fused QKV head-split backward -> dqkv (B, L, 3E)");
+  m.def("attn_merge", &attn_merge,
+        "(B*H, L, D) <-> (B, L, H*D) vectorized permute-copy");
   m.def("gelu_dropout_forward", &gelu_dropout_forward,
         "fused exact-GELU + bitfield dropout forward");
   m.def("gelu_dropout_backward", &gelu_dropout_backward,

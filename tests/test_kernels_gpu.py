@@ -902,3 +902,18 @@ def test_encoder_layer_bias_fold_parity():
             continue
         s = q.grad.abs().max().item() + 1e-4
         assert (p.grad - q.grad).abs().max().item() / s < 5e-3, n
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_attn_merge_roundtrip(dtype):
+    from unicore_amd import ops
+
+    torch.manual_seed(31)
+    B, H, L, D = 5, 12, 33, 64
+    x = torch.randn(B * H, L, D, device="cuda", dtype=dtype)
+    merged = ops.attn_merge(x, B, H)
+    ref = x.view(B, H, L, D).transpose(1, 2).contiguous().view(B, L, H * D)
+    assert torch.equal(merged, ref)
+    back = ops.attn_merge(merged, B, H, inverse=True)
+    assert torch.equal(back, x)

@@ -119,6 +119,30 @@ class _QKVSplit(torch.autograd.Function):
         return dqkv, dbias, None, None
 
 
+class _AttnMerge(torch.autograd.Function):
+    """(B*H, L, D) -> (B, L, H*D) with 16 B vectors on both sides (torch's
+    strided copy walks the permuted side with 2-byte scalars)."""
+
+    @staticmethod
+    def forward(ctx, x, bsz, num_heads):
+        from unicore_amd import ops
+
+        ctx.bsz = bsz
+        ctx.num_heads = num_heads
+        return ops.attn_merge(x.contiguous(), bsz, num_heads)
+
+    @staticmethod
+    def backward(ctx, grad):
+        from unicore_amd import ops
+
+        return (
+            ops.attn_merge(grad.contiguous(), ctx.bsz, ctx.num_heads,
+                           inverse=True),
+            None,
+            None,
+        )
+
+
 class SelfMultiheadAttention(nn.Module):
     def __init__(
         self,
@@ -273,12 +297,15 @@ class SelfMultiheadAttention(nn.Module):
             o = torch.bmm(attn, v)
         assert list(o.size()) == [bsz * self.num_heads, tgt_len, self.head_dim]
 
-        o = (
-            o.view(bsz, self.num_heads, tgt_len, self.head_dim)
-            .transpose(1, 2)
-            .contiguous()
-            .view(bsz, tgt_len, embed_dim)
-        )
+        if use_fused_split:
+            o = _AttnMerge.apply(o, bsz, self.num_heads)
+        else:
+            o = (
+                o.view(bsz, self.num_heads, tgt_len, self.head_dim)
+                .transpose(1, 2)
+                .contiguous()
+                .view(bsz, tgt_len, embed_dim)
+            )
         if skip_out_bias:
             o = F.linear(o, self.out_proj.weight)
         else:

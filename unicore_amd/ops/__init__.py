@@ -268,3 +268,9 @@ def softmax_dropout_bwd_bias_supported(n_batch, q, k, bb, bq, od) -> bool:
     return _kernels is not None and _kernels.softmax_dropout_backward_bias_supported(
         int(n_batch), int(q), int(k), int(bb), int(bq), int(od)
     )
+
+
+def attn_merge(x, bsz, num_heads, inverse=False):
+    # (B*H, L, D) -> (B, L, H*D); inverse maps back. 16B on both sides.
+    require_kernels()
+    return _kernels.attn_merge(x, int(bsz), int(num_heads), bool(inverse))
