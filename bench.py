@@ -33,7 +33,7 @@ def parse_bench_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=126,
+    p.add_argument("--batch-size", type=int, default=127,
                    help="per-GPU batch size (weak scaling: fixed per GPU)")
     p.add_argument("--seq-len", type=int, default=512)
     p.add_argument("--model", type=str, default="bert_base",
