@@ -253,3 +253,40 @@ def test_flat_ddp_lazy_aliased_bf16_parity(tmp_path):
     for n in ref:
         d = (ref[n] - r0[n]).abs().max().item()
         assert d < 0.05, (n, d)
+
+
+def test_trainer_two_rank_bf16_cpu(tmp_path):
+    """Full trainer at world_size 2 with bf16 flats over gloo — the exact
+    engine combination (FP16Optimizer flats + FlatDDP aliased buckets +
+    lazy DDP grad collection) the multi-GPU scaling run uses, on CPU."""
+    import subprocess
+    import sys
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29742", "-m", "unicore_cli.train",
+         "--task", "bert_synthetic", "--arch", "bert_base",
+         "--encoder-layers", "2", "--encoder-embed-dim", "64",
+         "--encoder-ffn-embed-dim", "128", "--encoder-attention-heads", "4",
+         "--loss", "masked_lm", "--optimizer", "adam",
+         "--lr-scheduler", "polynomial_decay", "--lr", "1e-4",
+         "--warmup-updates", "2", "--total-num-update", "100",
+         "--max-update", "3", "--dataset-size", "16", "--batch-size", "2",
+         "--tokens-per-sample", "64", "--max-seq-len", "66",
+         "--vocab-size", "100", "--bf16", "--cpu",
+         "--ddp-backend", "c10d", "--log-interval", "1",
+         "--log-format", "simple", "--num-workers", "0",
+         "--save-interval-updates", "2", "--save-dir", str(tmp_path)],
+        cwd=repo, capture_output=True, text=True, timeout=900,
+    )
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    assert (tmp_path / "checkpoint_last.pt").exists()
+    import torch
+
+    ck = torch.load(tmp_path / "checkpoint_last.pt", map_location="cpu",
+                    weights_only=False)
+    for k, v in ck["model"].items():
+        assert torch.isfinite(v.float()).all(), k

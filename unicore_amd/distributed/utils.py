@@ -113,8 +113,15 @@ def distributed_init(args):
                 args.distributed_init_method,
             )
         )
+        backend = args.distributed_backend
+        if backend == "nccl" and (
+            getattr(args, "cpu", False) or not torch.cuda.is_available()
+        ):
+            # RCCL needs a GPU; CPU multi-process runs (tests, debugging)
+            # fall back to gloo automatically
+            backend = "gloo"
         dist.init_process_group(
-            backend=args.distributed_backend,
+            backend=backend,
             init_method=args.distributed_init_method,
             world_size=args.distributed_world_size,
             rank=args.distributed_rank,
