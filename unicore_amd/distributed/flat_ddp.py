@@ -319,11 +319,19 @@ class FlatDDP(nn.Module):
                 b.work.wait()
 
     def zero_grad_buffers(self):
-        """Zero the flat grad buffers (one memset per bucket)."""
+        """Reset grad state for a fresh step.  Lazy mode only drops the
+        autograd-assigned tensors (every view is overwritten or zeroed at
+        the next bucket launch, and the aliased flats belong to the
+        optimizer's zero_grad); otherwise zero the flats and re-pin."""
+        if self.lazy:
+            for b in self._buckets:
+                for p in b.params:
+                    p.grad = None
+            return
         for b in self._buckets:
             b.flat.zero_()
             for p, v in zip(b.params, b.views):
-                p.grad = None if self.lazy else v
+                p.grad = v
 
     # ------------------------------------------------------------------
     # nn.Module plumbing

@@ -856,6 +856,13 @@ class Trainer(object):
 
     def zero_grad(self):
         self.optimizer.zero_grad()
+        # lazy-DDP hygiene: an aborted backward (OOM retry) can leave
+        # autograd-assigned grads on the params; the next backward would
+        # accumulate into them. The flat-view engines re-pin at forward
+        # time; the lazy engine drops the assigned tensors here instead.
+        m = self._wrapped_model
+        if m is not None and getattr(m, "lazy", False):
+            m.zero_grad_buffers()
 
     def lr_step_begin_epoch(self, epoch):
         """Adjust the learning rate at the beginning of the epoch."""
