@@ -36,6 +36,13 @@ std::vector<at::Tensor> qkv_split_backward(at::Tensor dq, at::Tensor dk,
                                            at::Tensor dv, int64_t B,
                                            int64_t num_heads, double scale,
                                            bool bias_grad);
+at::Tensor gated_mul_forward(at::Tensor x, at::Tensor g,
+                             std::optional<at::Tensor> bx,
+                             std::optional<at::Tensor> bg);
+std::vector<at::Tensor> gated_mul_backward(at::Tensor grad, at::Tensor x,
+                                           at::Tensor g,
+                                           std::optional<at::Tensor> bx,
+                                           std::optional<at::Tensor> bg);
 at::Tensor attn_merge(at::Tensor x, int64_t B, int64_t num_heads,
                       bool inverse);
 std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x,
@@ -114,6 +121,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused QKV head-split + q-scale -> (q, k, v) each (B*H, L, D)");
   m.def("qkv_split_backward", &qkv_split_backward,
         "fused QKV head-split backward -> dqkv (B, L, 3E)");
+  m.def("gated_mul_forward", &gated_mul_forward,
+        "fused (x+bx)*sigmoid(g+bg)");
+  m.def("gated_mul_backward", &gated_mul_backward,
+        "gated-mul backward -> (dx, dg[, bias colsums])");
   m.def("attn_merge", &attn_merge,
         "(B*H, L, D) <-> (B, L, H*D) vectorized permute-copy");
   m.def("gelu_dropout_forward", &gelu_dropout_forward,

@@ -34,6 +34,7 @@ HIP_SOURCES = [
     "csrc/embedding.hip",
     "csrc/cross_entropy.hip",
     "csrc/gaussian.hip",
+    "csrc/gated.hip",
 ]
 
 setup(

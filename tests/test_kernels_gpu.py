@@ -917,3 +917,39 @@ def test_attn_merge_roundtrip(dtype):
     assert torch.equal(merged, ref)
     back = ops.attn_merge(merged, B, H, inverse=True)
     assert torch.equal(back, x)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("with_bias", [False, True])
+def test_gated_mul_parity(dtype, with_bias):
+    from unicore_amd.modules.gated_mul import _GatedMul
+
+    torch.manual_seed(41)
+    C = 128
+    x = torch.randn(64, C, device="cuda", dtype=dtype)
+    g = torch.randn(64, C, device="cuda", dtype=dtype)
+    bx = torch.randn(C, device="cuda", dtype=dtype) if with_bias else None
+    bg = torch.randn(C, device="cuda", dtype=dtype) if with_bias else None
+
+    xs = [t.clone().requires_grad_(True) if t is not None else None
+          for t in (x, g, bx, bg)]
+    out = _GatedMul.apply(*xs)
+    go = torch.randn_like(out)
+    out.backward(go.clone())
+
+    rs = [t.detach().float().clone().requires_grad_(True) if t is not None
+          else None for t in (x, g, bx, bg)]
+    xr, gr, bxr, bgr = rs
+    yr = (xr + (bxr if bxr is not None else 0)) * torch.sigmoid(
+        gr + (bgr if bgr is not None else 0))
+    yr.backward(go.float())
+
+    tol = TOL[dtype]
+    assert (out.float() - yr.detach()).abs().max().item() < tol * 4
+    assert (xs[0].grad.float() - xr.grad).abs().max().item() < tol * 4
+    assert (xs[1].grad.float() - gr.grad).abs().max().item() < tol * 4
+    if with_bias:
+        for got, ref in ((xs[2].grad, bxr.grad), (xs[3].grad, bgr.grad)):
+            s = ref.abs().max().item() + 1e-3
+            assert (got.float() - ref).abs().max().item() / s < tol * 10

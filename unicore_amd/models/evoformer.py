@@ -24,6 +24,7 @@ from unicore_amd.models import (
 from unicore_amd.modules import (
     LayerNorm,
     dropout_add,
+    gated_mul,
     gelu_dropout,
     softmax_dropout,
 )
@@ -82,10 +83,14 @@ class MSARowAttentionWithPairBias(nn.Module):
             .permute(0, 2, 3, 1, 4)
             .reshape(B, S, L, D)
         )
-        g = torch.sigmoid(self.gate(x))
+        if _fold_ok(self.gate.bias):
+            og = gated_mul(o, F.linear(x, self.gate.weight),
+                           torch.zeros_like(self.gate.bias), self.gate.bias)
+        else:
+            og = o * torch.sigmoid(self.gate(x))
         if skip_out_bias:
-            return F.linear(o * g, self.out.weight)
-        return self.out(o * g)
+            return F.linear(og, self.out.weight)
+        return self.out(og)
 
 
 class MSAColumnAttention(nn.Module):
@@ -123,10 +128,14 @@ class MSAColumnAttention(nn.Module):
             .permute(0, 3, 1, 2, 4)
             .reshape(B, S, L, D)
         )
-        g = torch.sigmoid(self.gate(x))
+        if _fold_ok(self.gate.bias):
+            og = gated_mul(o, F.linear(x, self.gate.weight),
+                           torch.zeros_like(self.gate.bias), self.gate.bias)
+        else:
+            og = o * torch.sigmoid(self.gate(x))
         if skip_out_bias:
-            return F.linear(o * g, self.out.weight)
-        return self.out(o * g)
+            return F.linear(og, self.out.weight)
+        return self.out(og)
 
 
 class Transition(nn.Module):
@@ -184,12 +193,25 @@ class TriangleMultiplication(nn.Module):
 
     def forward(self, pair):
         p = self.norm(pair)
-        a = self.a_proj(p) * torch.sigmoid(self.a_gate(p))  # (B, I, J, c)
-        b = self.b_proj(p) * torch.sigmoid(self.b_gate(p))
+        if _fold_ok(self.a_proj.bias, self.a_gate.bias, self.b_proj.bias,
+                    self.b_gate.bias, self.out.bias, self.gate.bias):
+            a = gated_mul(F.linear(p, self.a_proj.weight),
+                          F.linear(p, self.a_gate.weight),
+                          self.a_proj.bias, self.a_gate.bias)
+            b = gated_mul(F.linear(p, self.b_proj.weight),
+                          F.linear(p, self.b_gate.weight),
+                          self.b_proj.bias, self.b_gate.bias)
+        else:
+            a = self.a_proj(p) * torch.sigmoid(self.a_gate(p))  # (B, I, J, c)
+            b = self.b_proj(p) * torch.sigmoid(self.b_gate(p))
         if self.outgoing:
             o = torch.einsum("bikc,bjkc->bijc", a, b)
         else:
             o = torch.einsum("bkic,bkjc->bijc", a, b)
+        if _fold_ok(self.out.bias, self.gate.bias):
+            return gated_mul(F.linear(self.out_norm(o), self.out.weight),
+                             F.linear(p, self.gate.weight),
+                             self.out.bias, self.gate.bias)
         o = self.out(self.out_norm(o))
         return o * torch.sigmoid(self.gate(p))
 

@@ -6,6 +6,7 @@ from .softmax_dropout import softmax_dropout
 from .gelu_dropout import gelu_dropout
 from .dropout_add import dropout_add
 from .gaussian import gaussian_basis
+from .gated_mul import gated_mul
 from .embedding import Embedding
 from .multihead_attention import SelfMultiheadAttention, CrossMultiheadAttention
 from .transformer_encoder_layer import TransformerEncoderLayer

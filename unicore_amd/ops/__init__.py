@@ -274,3 +274,14 @@ def attn_merge(x, bsz, num_heads, inverse=False):
     # (B*H, L, D) -> (B, L, H*D); inverse maps back. 16B on both sides.
     require_kernels()
     return _kernels.attn_merge(x, int(bsz), int(num_heads), bool(inverse))
+
+
+def gated_mul_fwd(x, g, bias_x=None, bias_g=None):
+    # out = (x + bias_x) * sigmoid(g + bias_g)
+    require_kernels()
+    return _kernels.gated_mul_forward(x, g, bias_x, bias_g)
+
+
+def gated_mul_bwd(grad, x, g, bias_x=None, bias_g=None):
+    require_kernels()
+    return _kernels.gated_mul_backward(grad, x, g, bias_x, bias_g)
