@@ -163,3 +163,28 @@ def test_ema_training(tmp_path, monkeypatch):
         os.path.join(save_dir, "checkpoint_last.pt"), weights_only=False
     )
     assert "ema" in state and state["ema"] is not None
+
+
+def test_evoformer_recycling_and_activation_checkpoint(tmp_path):
+    """Uni-Fold-style recycling (no-grad passes + one grad pass) and
+    per-block activation checkpointing both train on CPU."""
+    from unicore_cli.train import cli_main
+    import sys
+
+    argv = [
+        "train.py", "--task", "evoformer_synthetic", "--arch", "evoformer",
+        "--loss", "masked_msa", "--optimizer", "adam",
+        "--lr-scheduler", "fixed", "--lr", "1e-4",
+        "--max-update", "2", "--dataset-size", "8", "--batch-size", "2",
+        "--msa-depth", "8", "--residues", "16", "--evo-layers", "2",
+        "--msa-dim", "32", "--pair-dim", "16", "--evo-heads", "4",
+        "--recycle-iters", "1", "--activation-checkpoint",
+        "--cpu", "--log-interval", "1", "--log-format", "simple",
+        "--no-save", "--save-dir", str(tmp_path), "--num-workers", "0",
+    ]
+    old = sys.argv
+    sys.argv = argv
+    try:
+        cli_main()
+    finally:
+        sys.argv = old

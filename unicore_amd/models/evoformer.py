@@ -21,6 +21,7 @@ from unicore_amd.models import (
     register_model,
     register_model_architecture,
 )
+from unicore_amd import utils
 from unicore_amd.modules import (
     LayerNorm,
     dropout_add,
@@ -270,6 +271,13 @@ class EvoformerModel(BaseUnicoreModel):
         parser.add_argument("--evo-heads", type=int, metavar="N")
         parser.add_argument("--dropout", type=float)
         parser.add_argument("--max-rel-pos", type=int)
+        parser.add_argument("--recycle-iters", type=int,
+                            help="AlphaFold-style recycling: run the block "
+                                 "stack N+1 times, gradients only through "
+                                 "the last pass")
+        parser.add_argument("--activation-checkpoint", action="store_true",
+                            help="recompute each Evoformer block in "
+                                 "backward (utils.checkpoint_sequential)")
 
     def __init__(self, args, dictionary):
         super().__init__()
@@ -290,10 +298,25 @@ class EvoformerModel(BaseUnicoreModel):
         )
         self.final_norm = LayerNorm(args.msa_dim)
         self.lm_head = nn.Linear(args.msa_dim, len(dictionary))
+        self.recycle_iters = args.recycle_iters
+        self.activation_checkpoint = args.activation_checkpoint
 
     @classmethod
     def build_model(cls, args, task):
         return cls(args, task.dictionary)
+
+    def _trunk(self, msa, pair):
+        if self.activation_checkpoint and self.training:
+            # recompute each block in backward; the msa/pair pair threads
+            # through as the tuple state (utils.checkpoint_sequential)
+            fns = [
+                (lambda b: (lambda m, p: b(m, p)))(blk) for blk in self.blocks
+            ]
+            msa, pair = utils.checkpoint_sequential(fns, (msa, pair))
+        else:
+            for blk in self.blocks:
+                msa, pair = blk(msa, pair)
+        return msa, pair
 
     def forward(self, src_tokens, **kwargs):
         # src_tokens: (B, S, L)
@@ -304,8 +327,12 @@ class EvoformerModel(BaseUnicoreModel):
         pair = self.rel_pos_embed(rel + self.max_rel)  # (L, L, Dp)
         pair = pair.unsqueeze(0).expand(B, -1, -1, -1).contiguous()
         pair = pair.to(msa.dtype)
-        for blk in self.blocks:
-            msa, pair = blk(msa, pair)
+        # AlphaFold-style recycling: extra no-grad passes refine the
+        # representations; only the final pass builds the autograd graph
+        for _ in range(self.recycle_iters):
+            with torch.no_grad():
+                msa, pair = self._trunk(msa, pair)
+        msa, pair = self._trunk(msa, pair)
         return self.lm_head(self.final_norm(msa))
 
 
@@ -317,3 +344,5 @@ def evoformer_base_architecture(args):
     args.evo_heads = getattr(args, "evo_heads", 8)
     args.dropout = getattr(args, "dropout", 0.1)
     args.max_rel_pos = getattr(args, "max_rel_pos", 32)
+    args.recycle_iters = getattr(args, "recycle_iters", None) or 0
+    args.activation_checkpoint = getattr(args, "activation_checkpoint", False)
