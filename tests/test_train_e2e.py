@@ -188,3 +188,37 @@ def test_evoformer_recycling_and_activation_checkpoint(tmp_path):
         cli_main()
     finally:
         sys.argv = old
+
+
+def test_infer_demo_roundtrip(tmp_path):
+    """Train 2 updates, save, then run the inference demo on the saved
+    checkpoint (CPU)."""
+    import subprocess
+    import sys
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "unicore_cli.train",
+         "--task", "bert_synthetic", "--arch", "bert_base",
+         "--encoder-layers", "2", "--encoder-embed-dim", "64",
+         "--encoder-ffn-embed-dim", "128", "--encoder-attention-heads", "4",
+         "--loss", "masked_lm", "--optimizer", "adam",
+         "--lr-scheduler", "fixed", "--lr", "1e-4",
+         "--max-update", "2", "--dataset-size", "8", "--batch-size", "2",
+         "--tokens-per-sample", "64", "--max-seq-len", "66",
+         "--vocab-size", "100", "--cpu", "--log-format", "simple",
+         "--num-workers", "0", "--save-dir", str(tmp_path)],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    ck = tmp_path / "checkpoint_last.pt"
+    assert ck.exists()
+    r = subprocess.run(
+        [sys.executable, "examples/bert/infer_demo.py", "--checkpoint",
+         str(ck), "--cpu", "--batch-size", "2", "--seq-len", "64",
+         "--iters", "2"],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "tokens/s" in r.stdout
