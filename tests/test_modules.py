@@ -124,3 +124,47 @@ def test_stopwatch_and_time_meter():
     sw.start()
     sw.stop(n=2)
     assert sw.n == 2
+
+
+def test_bert_post_ln_and_classification_head():
+    """Model variants: post-LN trains a step on CPU; classification heads
+    register and run (the fine-tuning surface of the reference BERT)."""
+    import argparse
+
+    import torch
+
+    from unicore_amd.models.bert import BertModel, base_architecture
+
+    args = argparse.Namespace()
+    base_architecture(args)
+    args.encoder_layers = 2
+    args.encoder_embed_dim = 32
+    args.encoder_ffn_embed_dim = 64
+    args.encoder_attention_heads = 4
+    args.max_seq_len = 64
+    args.post_ln = True
+
+    class D:
+        def __len__(self):
+            return 50
+
+        def pad(self):
+            return 1
+
+    class T:
+        dictionary = D()
+
+    m = BertModel.build_model(args, T())
+    toks = torch.randint(2, 49, (2, 16))
+    logits = m(toks)
+    out = logits[0] if isinstance(logits, tuple) else logits
+    out.float().sum().backward()
+    assert all(
+        p.grad is None or torch.isfinite(p.grad.float()).all()
+        for p in m.parameters()
+    )
+
+    m.register_classification_head("sent", num_classes=3)
+    feats = m(toks, features_only=True, classification_head_name="sent")
+    x = feats[0] if isinstance(feats, tuple) else feats
+    assert x.shape[-1] == 3
