@@ -587,6 +587,16 @@ class CudaPrefetcher(object):
             self._next_event.record(self.copy_stream)
 
     def __iter__(self):
+        from ..utils import apply_to_sample
+
+        def _claim(t):
+            # the tensors were allocated on the copy stream; mark them as
+            # used by the consumer stream so the caching allocator does not
+            # hand their memory to a LATER prefetch copy while this step's
+            # kernels are still reading it (classic cross-stream free race)
+            t.record_stream(torch.cuda.current_stream())
+            return t
+
         self._itr = iter(self.iterable)
         self._preload()
         while self._next is not None:
@@ -596,4 +606,5 @@ class CudaPrefetcher(object):
             # consumer stream waits for the copy to land; tensors stay alive
             # via the yielded reference
             event.wait(torch.cuda.current_stream())
+            batch = apply_to_sample(_claim, batch)
             yield batch
