@@ -45,6 +45,8 @@ std::vector<at::Tensor> gated_mul_backward(at::Tensor grad, at::Tensor x,
                                            std::optional<at::Tensor> bg);
 at::Tensor attn_merge(at::Tensor x, int64_t B, int64_t num_heads,
                       bool inverse);
+at::Tensor msa_arrange(at::Tensor x, int64_t B, int64_t S, int64_t L,
+                       int64_t H, bool col, bool inverse);
 std::vector<at::Tensor> gelu_dropout_forward(at::Tensor x,
                                              std::optional<at::Tensor> bias,
                                              double p, bool is_training);
@@ -125,6 +127,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused (x+bx)*sigmoid(g+bg)");
   m.def("gated_mul_backward", &gated_mul_backward,
         "gated-mul backward -> (dx, dg[, bias colsums])");
+  m.def("msa_arrange", &msa_arrange,
+        "(B,S,L,H*D) <-> row/col head-major MSA layouts, 16 B both sides");
   m.def("attn_merge", &attn_merge,
         "(B*H, L, D) <-> (B, L, H*D) vectorized permute-copy");
   m.def("gelu_dropout_forward", &gelu_dropout_forward,

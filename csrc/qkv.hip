@@ -291,3 +291,119 @@ at::Tensor attn_merge(at::Tensor x, int64_t B, int64_t num_heads,
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return out;
 }
+
+namespace {
+
+// Evoformer MSA arranges: (B, S, L, H, D) <-> head-major batched layouts.
+//   ROW mode: (B*H*S, L, D)  (attention over L for each msa row)
+//   COL mode: (B*L*H, S, D)  (attention over S for each column)
+// torch runs each as a 5-D permute through its strided copy (2-byte
+// scalars on one side, ~27k small launches per evoformer step); decoding
+// from the OUTPUT index keeps both sides on 16 B vectors.
+template <typename T, bool COL, bool INVERSE>
+__global__ void msa_arrange_kernel(T* __restrict__ out, const T* __restrict__ in,
+                                   int64_t n8, int S, int L, int H, int D8,
+                                   int64_t in_rstride8) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    // decode the NATURAL (B,S,L,H,D8) coordinates from whichever side is
+    // the kernel's output, then compose the other side's flat index
+    int64_t b;
+    int s, l, h, d8;
+    int64_t tmp = i;
+    d8 = (int)(tmp % D8);
+    tmp /= D8;
+    if (INVERSE) {
+      // output is (B,S,L,H,D8)
+      h = (int)(tmp % H);
+      tmp /= H;
+      l = (int)(tmp % L);
+      tmp /= L;
+      s = (int)(tmp % S);
+      b = tmp / S;
+    } else if (COL) {
+      // output is (B,L,H,S,D8)
+      s = (int)(tmp % S);
+      tmp /= S;
+      h = (int)(tmp % H);
+      tmp /= H;
+      l = (int)(tmp % L);
+      b = tmp / L;
+    } else {
+      // output is (B,H,S,L,D8)
+      l = (int)(tmp % L);
+      tmp /= L;
+      s = (int)(tmp % S);
+      tmp /= S;
+      h = (int)(tmp % H);
+      b = tmp / H;
+    }
+    int64_t src;
+    if (!INVERSE) {
+      // input rows are ((b,s,l)) with an arbitrary row stride (chunk()
+      // views of the fused qkv projection stay copy-free)
+      src = ((b * S + s) * (int64_t)L + l) * in_rstride8 + h * D8 + d8;
+    } else if (COL) {
+      // input is (B,L,H,S,D8)
+      src = ((((b * L + l) * (int64_t)H + h) * S + s) * D8 + d8);
+    } else {
+      // input is (B,H,S,L,D8)
+      src = ((((b * H + h) * (int64_t)S + s) * L + l) * D8 + d8);
+    }
+    float f[8];
+    load8(in + src * 8, f);
+    store8(out + i * 8, f);
+  }
+}
+
+}  // namespace
+
+at::Tensor msa_arrange(at::Tensor x, int64_t B, int64_t S, int64_t L,
+                       int64_t H, bool col, bool inverse) {
+  TORCH_CHECK(x.is_cuda(), "msa_arrange: CUDA tensor");
+  const int64_t total = B * S * L * H;
+  TORCH_CHECK(x.numel() % total == 0, "msa_arrange: bad shape");
+  const int64_t D = x.numel() / total;
+  TORCH_CHECK(D % 8 == 0, "msa_arrange: head_dim % 8");
+  int64_t rstride8 = H * D / 8;
+  if (!inverse) {
+    // accept a last-dim-contiguous 4-D view (B,S,L,C) with any row stride
+    TORCH_CHECK(x.dim() == 4 && x.size(3) == H * D && x.stride(3) == 1 &&
+                    x.stride(2) % 8 == 0 &&
+                    x.stride(1) == L * x.stride(2) &&
+                    x.stride(0) == S * x.stride(1),
+                "msa_arrange: need (B,S,L,C) with contiguous C");
+    rstride8 = x.stride(2) / 8;
+  } else {
+    TORCH_CHECK(x.is_contiguous(), "msa_arrange: inverse needs contiguous");
+  }
+  at::Tensor out;
+  if (inverse)
+    out = at::empty({B, S, L, H * D}, x.options());
+  else if (col)
+    out = at::empty({B * L * H, S, D}, x.options());
+  else
+    out = at::empty({B * H, S, L, D}, x.options());
+  const int64_t n8 = x.numel() / 8;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = unicore_grid((n8 + 255) / 256);
+  DISPATCH_FTYPES(x.scalar_type(), "msa_arrange", {
+    auto launch = [&](auto col_tag, auto inv_tag) {
+      msa_arrange_kernel<scalar_t, decltype(col_tag)::value,
+                         decltype(inv_tag)::value><<<grid, 256, 0, stream>>>(
+          reinterpret_cast<scalar_t*>(out.data_ptr()),
+          reinterpret_cast<const scalar_t*>(x.data_ptr()), n8, (int)S, (int)L,
+          (int)H, (int)(D / 8), rstride8);
+    };
+    if (col) {
+      if (inverse) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else {
+      if (inverse) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    }
+  });
+  C10_CUDA_KERNEL_LAUNCH_CHECK();
+  return out;
+}

@@ -953,3 +953,50 @@ def test_gated_mul_parity(dtype, with_bias):
         for got, ref in ((xs[2].grad, bxr.grad), (xs[3].grad, bgr.grad)):
             s = ref.abs().max().item() + 1e-3
             assert (got.float() - ref).abs().max().item() / s < tol * 10
+
+
+@requires_gpu
+@pytest.mark.parametrize("col", [False, True])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_msa_arrange_parity(col, dtype):
+    """Row/col MSA arranges (incl. chunk()-view input) and their merges
+    match the permute/reshape chains exactly, fwd and bwd."""
+    from unicore_amd.modules.msa_arrange import msa_arrange, msa_merge
+
+    torch.manual_seed(51)
+    B, S, L, H, D = 2, 6, 10, 4, 16
+    E = H * D
+    qkv = torch.randn(B, S, L, 3 * E, device="cuda", dtype=dtype,
+                      requires_grad=True)
+    q = qkv.chunk(3, dim=-1)[1]  # strided view input
+    out = msa_arrange(q, H, col)
+    if col:
+        ref = (q.view(B, S, L, H, D).permute(0, 2, 3, 1, 4)
+               .reshape(B * L * H, S, D))
+    else:
+        ref = (q.view(B, S, L, H, D).permute(0, 3, 1, 2, 4)
+               .reshape(B * H * S, L, D))
+    assert torch.equal(out, ref)
+
+    g = torch.randn_like(out)
+    out.backward(g)
+    grad_fused = qkv.grad.clone()
+    qkv.grad = None
+    ref.backward(g)
+    assert torch.equal(grad_fused, qkv.grad)
+
+    x = torch.randn_like(out).requires_grad_(True)
+    merged = msa_merge(x, B, S, L, H, col)
+    if col:
+        mref = (x.view(B, L, H, S, D).permute(0, 3, 1, 2, 4)
+                .reshape(B, S, L, E))
+    else:
+        mref = (x.view(B, H, S, L, D).permute(0, 2, 3, 1, 4)
+                .reshape(B, S, L, E))
+    assert torch.equal(merged, mref)
+    gm = torch.randn_like(merged)
+    merged.backward(gm)
+    gf = x.grad.clone()
+    x.grad = None
+    mref.backward(gm)
+    assert torch.equal(gf, x.grad)

@@ -22,6 +22,7 @@ from unicore_amd.models import (
     register_model_architecture,
 )
 from unicore_amd import utils
+from unicore_amd.modules.msa_arrange import msa_arrange, msa_merge
 from unicore_amd.modules import (
     LayerNorm,
     dropout_add,
@@ -67,9 +68,7 @@ class MSARowAttentionWithPairBias(nn.Module):
         # so the (B, H, 1, L, L) pair bias is a contiguous-block broadcast
         # for the fused softmax kernel
         def arrange(t):
-            return t.view(B, S, L, H, Dh).permute(0, 3, 1, 2, 4).reshape(
-                B * H * S, L, Dh
-            )
+            return msa_arrange(t, H, col=False)
 
         q = arrange(q) * self.scaling
         k = arrange(k)
@@ -79,11 +78,7 @@ class MSARowAttentionWithPairBias(nn.Module):
         bias = bias.permute(0, 3, 1, 2).unsqueeze(2)  # (B, H, 1, L, L)
         attn = softmax_dropout(scores, self.dropout, self.training, bias=bias)
         o = torch.bmm(attn.view(B * H * S, L, L), v)
-        o = (
-            o.view(B, H, S, L, Dh)
-            .permute(0, 2, 3, 1, 4)
-            .reshape(B, S, L, D)
-        )
+        o = msa_merge(o, B, S, L, H, col=False)
         if _fold_ok(self.gate.bias):
             og = gated_mul(o, F.linear(x, self.gate.weight),
                            torch.zeros_like(self.gate.bias), self.gate.bias)
@@ -114,9 +109,7 @@ class MSAColumnAttention(nn.Module):
 
         # attention over the S dimension for each column l
         def arrange(t):
-            return t.view(B, S, L, H, Dh).permute(0, 2, 3, 1, 4).reshape(
-                B * L * H, S, Dh
-            )
+            return msa_arrange(t, H, col=True)
 
         q = arrange(q) * self.scaling
         k = arrange(k)
@@ -124,11 +117,7 @@ class MSAColumnAttention(nn.Module):
         scores = torch.bmm(q, k.transpose(1, 2)).view(B * L * H, S, S)
         attn = softmax_dropout(scores, self.dropout, self.training)
         o = torch.bmm(attn, v)
-        o = (
-            o.view(B, L, H, S, Dh)
-            .permute(0, 3, 1, 2, 4)
-            .reshape(B, S, L, D)
-        )
+        o = msa_merge(o, B, S, L, H, col=True)
         if _fold_ok(self.gate.bias):
             og = gated_mul(o, F.linear(x, self.gate.weight),
                            torch.zeros_like(self.gate.bias), self.gate.bias)

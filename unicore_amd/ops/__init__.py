@@ -285,3 +285,11 @@ def gated_mul_fwd(x, g, bias_x=None, bias_g=None):
 def gated_mul_bwd(grad, x, g, bias_x=None, bias_g=None):
     require_kernels()
     return _kernels.gated_mul_backward(grad, x, g, bias_x, bias_g)
+
+
+def msa_arrange(x, B, S, L, heads, col, inverse=False):
+    # (B,S,L,E) <-> head-major MSA layouts; see modules/msa_arrange.py
+    require_kernels()
+    return _kernels.msa_arrange(
+        x, int(B), int(S), int(L), int(heads), bool(col), bool(inverse)
+    )
