@@ -36,6 +36,7 @@ class _MsaMerge(torch.autograd.Function):
         from unicore_amd import ops
 
         ctx.dims = (B, S, L, heads, col)
+        ctx.in_shape = x.shape
         return ops.msa_arrange(x.contiguous(), B, S, L, heads, col,
                                inverse=True)
 
@@ -45,7 +46,7 @@ class _MsaMerge(torch.autograd.Function):
 
         B, S, L, H, col = ctx.dims
         g = ops.msa_arrange(grad.contiguous(), B, S, L, H, col)
-        return g, None, None, None, None, None
+        return g.reshape(ctx.in_shape), None, None, None, None, None
 
 
 def _fused_ok(x, heads):
