@@ -168,3 +168,42 @@ def test_bert_post_ln_and_classification_head():
     feats = m(toks, features_only=True, classification_head_name="sent")
     x = feats[0] if isinstance(feats, tuple) else feats
     assert x.shape[-1] == 3
+
+
+def test_fused_op_eager_fallbacks_with_bias():
+    """CPU eager fallbacks of the bias-carrying fused ops match explicit
+    torch math (these paths also serve as the GPU numerics oracles)."""
+    import torch
+
+    from unicore_amd.modules import dropout_add, gated_mul, gelu_dropout
+
+    torch.manual_seed(3)
+    x = torch.randn(10, 16)
+    res = torch.randn(10, 16)
+    b = torch.randn(16)
+    out = dropout_add(x, res, 0.0, True, bias=b)
+    assert torch.allclose(out, res + x + b)
+
+    g = torch.randn(10, 16)
+    out = gelu_dropout(x, 0.0, True, bias=b)
+    assert torch.allclose(out, torch.nn.functional.gelu(x + b))
+
+    bg = torch.randn(16)
+    out = gated_mul(x, g, b, bg)
+    assert torch.allclose(out, (x + b) * torch.sigmoid(g + bg))
+    out = gated_mul(x, g)
+    assert torch.allclose(out, x * torch.sigmoid(g))
+
+
+def test_msa_arrange_eager_fallback():
+    import torch
+
+    from unicore_amd.modules.msa_arrange import msa_arrange, msa_merge
+
+    torch.manual_seed(4)
+    B, S, L, H, D = 2, 3, 5, 4, 8
+    x = torch.randn(B, S, L, H * D)
+    for col in (False, True):
+        a = msa_arrange(x, H, col)
+        back = msa_merge(a, B, S, L, H, col)
+        assert torch.allclose(back, x)
