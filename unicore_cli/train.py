@@ -46,6 +46,9 @@ def main(args) -> None:
     # run restores it from the checkpoint's extra_state instead
     if hasattr(checkpoint_utils.save_checkpoint, "best"):
         del checkpoint_utils.save_checkpoint.best
+    if hasattr(should_stop_early, "best"):
+        del should_stop_early.best
+    should_stop_early.num_runs = 0
 
     np_seed = args.seed
     torch.manual_seed(args.seed)
