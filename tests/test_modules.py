@@ -207,3 +207,37 @@ def test_msa_arrange_eager_fallback():
         a = msa_arrange(x, H, col)
         back = msa_merge(a, B, S, L, H, col)
         assert torch.allclose(back, x)
+
+
+def test_broadcast_descr_matches_torch_semantics():
+    """Property test: wherever _broadcast_descr claims a (src_nb, outer_div)
+    addressing, the kernel's source-row formula must agree with torch's
+    broadcast expansion for every batch row."""
+    import itertools
+
+    import torch
+
+    from unicore_amd.modules.softmax_dropout import _broadcast_descr
+
+    torch.manual_seed(9)
+    q, k = 2, 4
+    checked = 0
+    for batch_dims in [(2,), (2, 3), (2, 3, 4)]:
+        m = len(batch_dims)
+        # every 0/1 pattern of which batch dims the source keeps
+        for keep in itertools.product([False, True], repeat=m):
+            src_batch = tuple(b if kp else 1 for b, kp in zip(batch_dims, keep))
+            descr = _broadcast_descr(src_batch, list(batch_dims))
+            if descr is None:
+                continue
+            src_nb, outer_div = descr
+            src = torch.randn(*src_batch, q, k)
+            full = src.expand(*batch_dims, q, k).reshape(-1, q, k)
+            flat_src = src.reshape(-1, q, k)
+            n_batch = full.shape[0]
+            for b in range(n_batch):
+                idx = (b // outer_div) % src_nb
+                assert torch.equal(full[b], flat_src[idx]), (
+                    src_batch, batch_dims, b, idx)
+            checked += 1
+    assert checked >= 10  # the contiguous-block patterns all verified
