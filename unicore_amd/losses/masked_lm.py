@@ -3,7 +3,6 @@
 import math
 
 import torch
-import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss

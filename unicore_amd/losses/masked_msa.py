@@ -4,7 +4,6 @@ semantics over (B, S, L) token grids."""
 import math
 
 import torch
-import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss

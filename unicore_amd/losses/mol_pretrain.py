@@ -6,7 +6,6 @@ coordinate deltas (the Uni-Mol pretraining recipe shape)."""
 import math
 
 import torch
-import torch.nn.functional as F
 
 from unicore_amd import metrics
 from unicore_amd.losses import UnicoreLoss, register_loss

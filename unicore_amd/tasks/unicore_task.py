@@ -7,7 +7,6 @@ from typing import Any, Callable, Dict
 
 import torch
 
-from unicore_amd import utils
 from unicore_amd.data import UnicoreDataset, data_utils, iterators
 from unicore_amd.logging import metrics
 
