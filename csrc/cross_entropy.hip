@@ -85,8 +85,10 @@ __global__ void ce_fwd_kernel(float* __restrict__ loss, float* __restrict__ lse,
     const float row_lse = bm + __logf(bs);
     if (tid == 0) {
       lse[row] = row_lse;
+      // out-of-range targets are treated as ignore_index rather than read
+      // out of bounds (corrupt data should not fault the GPU)
       const int64_t t = target[row];
-      loss[row] = (t == ignore_index)
+      loss[row] = (t == ignore_index || t < 0 || t >= V)
                       ? 0.f
                       : row_lse - Cvt<T>::to_f(x[t]);
     }
@@ -107,7 +109,8 @@ __global__ void ce_bwd_kernel(T* __restrict__ dx, const T* __restrict__ logits,
     T* d = dx + row * V;
     const float l = lse[row];
     const int64_t t = target[row];
-    const float gr = t == ignore_index ? 0.f : g;
+    // same out-of-range policy as forward: contribute no gradient
+    const float gr = (t == ignore_index || t < 0 || t >= V) ? 0.f : g;
     int64_t e = tid * 8;
     if constexpr (VEC8) {
       for (; e + 7 < V; e += 256 * 8) {

@@ -13,7 +13,7 @@
 
 namespace {
 
-template <typename T>
+template <typename T, bool VEC>
 __global__ void embedding_bwd_scatter_kernel(float* __restrict__ acc,
                                              const int64_t* __restrict__ idx,
                                              const T* __restrict__ grad,
@@ -30,7 +30,9 @@ __global__ void embedding_bwd_scatter_kernel(float* __restrict__ acc,
     const T* src = grad + t * dim;
     for (int e = sub * 4; e + 3 < dim; e += 64) {
       float f[4];
-      if constexpr (sizeof(T) == 2) {
+      // VEC requires the row base 8B-aligned: host only enables it when
+      // dim % 4 == 0 (otherwise t*dim can break alignment for 2B dtypes)
+      if constexpr (VEC && sizeof(T) == 2) {
         union {
           uint2 u;
           T tt[4];
@@ -38,9 +40,12 @@ __global__ void embedding_bwd_scatter_kernel(float* __restrict__ acc,
         U.u = *reinterpret_cast<const uint2*>(src + e);
 #pragma unroll
         for (int j = 0; j < 4; ++j) f[j] = Cvt<T>::to_f(U.tt[j]);
-      } else {
+      } else if constexpr (VEC) {
         const float4 g4 = *reinterpret_cast<const float4*>(src + e);
         f[0] = g4.x; f[1] = g4.y; f[2] = g4.z; f[3] = g4.w;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) f[j] = Cvt<T>::to_f(src[e + j]);
       }
 #pragma unroll
       for (int j = 0; j < 4; ++j) atomicAdd(dst + e + j, f[j]);
@@ -101,10 +106,16 @@ at::Tensor embedding_backward(at::Tensor grad, at::Tensor indices,
   auto stream = at::cuda::getCurrentCUDAStream();
   const int grid = unicore_grid((n_tokens * 16 + 255) / 256);
   DISPATCH_FTYPES(grad.scalar_type(), "embedding_backward", {
-    embedding_bwd_scatter_kernel<scalar_t><<<grid, 256, 0, stream>>>(
-        acc.data_ptr<float>(), indices.data_ptr<int64_t>(),
-        reinterpret_cast<const scalar_t*>(grad.data_ptr()), n_tokens, dim,
-        padding_idx);
+    if (dim % 4 == 0)
+      embedding_bwd_scatter_kernel<scalar_t, true><<<grid, 256, 0, stream>>>(
+          acc.data_ptr<float>(), indices.data_ptr<int64_t>(),
+          reinterpret_cast<const scalar_t*>(grad.data_ptr()), n_tokens, dim,
+          padding_idx);
+    else
+      embedding_bwd_scatter_kernel<scalar_t, false><<<grid, 256, 0, stream>>>(
+          acc.data_ptr<float>(), indices.data_ptr<int64_t>(),
+          reinterpret_cast<const scalar_t*>(grad.data_ptr()), n_tokens, dim,
+          padding_idx);
   });
   if (grad.scalar_type() == at::kFloat) {
     C10_CUDA_KERNEL_LAUNCH_CHECK();

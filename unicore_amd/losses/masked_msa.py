@@ -34,8 +34,10 @@ class MaskedMSALoss(UnicoreLoss):
     def reduce_metrics(logging_outputs, split="valid") -> None:
         loss_sum = sum(log.get("loss", 0) for log in logging_outputs)
         sample_size = sum(log.get("sample_size", 0) for log in logging_outputs)
+        # sample_size can be 0 when no tokens were masked in the whole batch
+        denom = sample_size if sample_size > 0 else 1
         metrics.log_scalar(
-            "loss", loss_sum / sample_size / math.log(2), sample_size, round=3
+            "loss", loss_sum / denom / math.log(2), sample_size, round=3
         )
 
     @staticmethod

@@ -48,13 +48,15 @@ class MolPretrainLoss(UnicoreLoss):
         token_sum = sum(log.get("token_loss", 0) for log in logging_outputs)
         coord_sum = sum(log.get("coord_loss", 0) for log in logging_outputs)
         sample_size = sum(log.get("sample_size", 0) for log in logging_outputs)
+        # sample_size can be 0 when no tokens were masked in the whole batch
+        denom = sample_size if sample_size > 0 else 1
         metrics.log_scalar(
-            "loss", loss_sum / sample_size / math.log(2), sample_size, round=3
+            "loss", loss_sum / denom / math.log(2), sample_size, round=3
         )
         metrics.log_scalar(
-            "token_loss", token_sum / sample_size / math.log(2), sample_size, round=3
+            "token_loss", token_sum / denom / math.log(2), sample_size, round=3
         )
-        metrics.log_scalar("coord_loss", coord_sum / sample_size, sample_size, round=3)
+        metrics.log_scalar("coord_loss", coord_sum / denom, sample_size, round=3)
 
     @staticmethod
     def logging_outputs_can_be_summed(is_train) -> bool:

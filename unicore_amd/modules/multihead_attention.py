@@ -60,11 +60,13 @@ class _FlashAttn(torch.autograd.Function):
         dq, dk, dv = grads[0], grads[1], grads[2]
         dbias = None
         if len(grads) > 3:
-            # grads[3] is materialized dS (BH, L, L); the bias gradient is
-            # its sum over the broadcast batches (outer_div == 1 contract)
+            # grads[3] is materialized dS (BH, L, L); batch rows decompose
+            # as (outer, src_nb, outer_div) per the broadcast descriptor, so
+            # the bias gradient sums over both the outer and inner-repeat axes
             ds = grads[3]
             nb = bias_t.shape[0]
-            dbias = ds.view(-1, nb, ds.shape[1], ds.shape[2]).sum(0)
+            od = ctx.bias_od
+            dbias = ds.view(-1, nb, od, ds.shape[1], ds.shape[2]).sum((0, 2))
         return dq, dk, dv, dbias, None, None, None, None, None
 
 
