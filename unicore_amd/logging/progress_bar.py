@@ -1,9 +1,15 @@
-"""Progress-bar abstraction over training iterators.
+"""Progress reporting for training/validation iterators.
 
-Parity with reference unicore/logging/progress_bar.py (json:138, noop:189,
-simple:208, tqdm:243, tensorboard wrapper:302, wandb:313-327): a factory
-``progress_bar()`` returning a wrapper that logs intermediate and end-of-epoch
-stats to the chosen sink; tensorboard/wandb drive on rank 0 only.
+Capability parity with the reference bars (unicore/logging/progress_bar.py:
+json:138, noop:189, simple:208, tqdm:243, tensorboard wrapper:302,
+wandb:313-327): a ``progress_bar()`` factory returns an iterable wrapper with
+``log()`` (periodic, mid-epoch) and ``print()`` (end-of-epoch) hooks; the
+tensorboard/wandb sink wraps any of the text bars and is only built on the
+rank that asks for it.
+
+Structure here differs from the reference: the position-tracking iteration
+shared by the json and simple bars lives in one ``_CountingBar`` mixin, and
+the format name -> class mapping is a registry dict.
 """
 
 import json
@@ -22,6 +28,201 @@ from .meters import AverageMeter, StopwatchMeter, TimeMeter
 logger = logging.getLogger(__name__)
 
 
+@contextmanager
+def rename_logger(logger, new_name):
+    """Temporarily relabel a logger (used to tag train/valid lines)."""
+    saved = logger.name
+    if new_name is not None:
+        logger.name = new_name
+    yield logger
+    logger.name = saved
+
+
+def format_stat(stat):
+    """Render one stat value (number / meter / tensor) for display."""
+    if isinstance(stat, Number):
+        return f"{stat:g}"
+    if isinstance(stat, AverageMeter):
+        return f"{stat.avg:.3f}"
+    if isinstance(stat, TimeMeter):
+        return f"{round(stat.avg):g}"
+    if isinstance(stat, StopwatchMeter):
+        return f"{round(stat.sum):g}"
+    if torch.is_tensor(stat):
+        return stat.tolist()
+    return stat
+
+
+class BaseProgressBar:
+    """Iterable wrapper with log/print hooks; subclasses pick the sink."""
+
+    def __init__(self, iterable, epoch=None, prefix=None):  # noqa: D107
+        self.iterable = iterable
+        self.n = getattr(iterable, "n", 0)  # resume offset within the epoch
+        self.epoch = epoch
+        parts = []
+        if epoch is not None:
+            parts.append(f"epoch {epoch:03d}")
+        if prefix is not None:
+            parts.append(prefix)
+        self.prefix = " | ".join(parts)
+
+    def __len__(self):
+        return len(self.iterable)
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        return False
+
+    def __iter__(self):  # subclasses drive the wrapped iterable
+        raise NotImplementedError
+
+    def log(self, stats, tag=None, step=None) -> None:
+        """Periodic mid-epoch stats."""
+        raise NotImplementedError
+
+    def print(self, stats, tag=None, step=None) -> None:
+        """End-of-epoch stats."""
+        raise NotImplementedError
+
+    def _format_stats(self, stats):
+        return OrderedDict((k, str(format_stat(v))) for k, v in stats.items())
+
+    @staticmethod
+    def _str_commas(stats):
+        return ", ".join(f"{k}={v.strip()}" for k, v in stats.items())
+
+    @staticmethod
+    def _str_pipes(stats):
+        return " | ".join(f"{k} {v.strip()}" for k, v in stats.items())
+
+
+class _CountingBar(BaseProgressBar):
+    """Shared iteration bookkeeping: tracks position + total and rate-limits
+    log() to every ``log_interval`` steps."""
+
+    def __init__(self, iterable, epoch=None, prefix=None, log_interval=1000):
+        super().__init__(iterable, epoch=epoch, prefix=prefix)
+        self.log_interval = log_interval
+        self.i = self.size = None
+
+    def __iter__(self):
+        self.size = len(self.iterable)
+        position = self.n
+        for obj in self.iterable:
+            self.i = position
+            position += 1
+            yield obj
+
+    def _should_log(self, step):
+        step = (step if step else self.i) or 0
+        return (
+            step > 0
+            and self.log_interval is not None
+            and step % self.log_interval == 0
+        )
+
+
+class JsonProgressBar(_CountingBar):
+    """One JSON object per log line (machine-readable)."""
+
+    def log(self, stats, tag=None, step=None) -> None:
+        if not self._should_log(step):
+            return
+        update = None
+        if self.epoch is not None:
+            update = self.epoch - 1 + (self.i + 1) / float(self.size)
+        payload = self._jsonify(stats, epoch=self.epoch, update=update)
+        with rename_logger(logger, tag) as sink:
+            sink.info(json.dumps(payload))
+
+    def print(self, stats, tag=None, step=None) -> None:
+        if tag is not None:
+            stats = OrderedDict((f"{tag}_{k}", v) for k, v in stats.items())
+        self.stats = stats
+        payload = self._jsonify(stats, epoch=self.epoch)
+        with rename_logger(logger, tag) as sink:
+            sink.info(json.dumps(payload))
+
+    @staticmethod
+    def _jsonify(stats, epoch=None, update=None):
+        out = OrderedDict()
+        if epoch is not None:
+            out["epoch"] = epoch
+        if update is not None:
+            out["update"] = round(update, 3)
+        for k, v in stats.items():
+            out[k] = format_stat(v)
+        return out
+
+
+class NoopProgressBar(BaseProgressBar):
+    """Swallows everything (non-master ranks)."""
+
+    def __iter__(self):
+        return iter(self.iterable)
+
+    def log(self, stats, tag=None, step=None) -> None:
+        pass
+
+    def print(self, stats, tag=None, step=None) -> None:
+        pass
+
+
+class SimpleProgressBar(_CountingBar):
+    """Plain log lines for non-TTY environments."""
+
+    def log(self, stats, tag=None, step=None) -> None:
+        if not self._should_log(step):
+            return
+        body = self._str_commas(self._format_stats(stats))
+        with rename_logger(logger, tag) as sink:
+            sink.info(f"{self.prefix}:  {self.i + 1:5d} / {self.size:d} {body}")
+
+    def print(self, stats, tag=None, step=None) -> None:
+        body = self._str_pipes(self._format_stats(stats))
+        with rename_logger(logger, tag) as sink:
+            sink.info(f"{self.prefix} | {body}")
+
+
+class TqdmProgressBar(BaseProgressBar):
+    """Interactive tqdm bar with a postfix stats line."""
+
+    def __init__(self, iterable, epoch=None, prefix=None):
+        super().__init__(iterable, epoch=epoch, prefix=prefix)
+        try:
+            from tqdm import tqdm
+        except ImportError:
+            # degrade gracefully where tqdm is unavailable
+            self.tqdm = None
+            self._fallback = SimpleProgressBar(iterable, epoch, prefix, 100)
+        else:
+            self.tqdm = tqdm(iterable, self.prefix, leave=False, disable=False)
+
+    def __iter__(self):
+        return iter(self._fallback if self.tqdm is None else self.tqdm)
+
+    def log(self, stats, tag=None, step=None) -> None:
+        if self.tqdm is None:
+            return self._fallback.log(stats, tag, step)
+        self.tqdm.set_postfix(self._format_stats(stats), refresh=False)
+
+    def print(self, stats, tag=None, step=None) -> None:
+        body = self._str_pipes(self._format_stats(stats))
+        with rename_logger(logger, tag) as sink:
+            sink.info(f"{self.prefix} | {body}")
+
+
+_BAR_KINDS = {
+    "json": lambda it, ep, pre, ival: JsonProgressBar(it, ep, pre, ival),
+    "none": lambda it, ep, pre, ival: NoopProgressBar(it, ep, pre),
+    "simple": lambda it, ep, pre, ival: SimpleProgressBar(it, ep, pre, ival),
+    "tqdm": lambda it, ep, pre, ival: TqdmProgressBar(it, ep, pre),
+}
+
+
 def progress_bar(
     iterator,
     log_format: Optional[str] = None,
@@ -33,234 +234,26 @@ def progress_bar(
     wandb_project: Optional[str] = None,
     args=None,
 ):
-    if log_format is None:
-        log_format = default_log_format
-    if log_format == "tqdm" and not sys.stderr.isatty():
-        log_format = "simple"
-
-    if log_format == "json":
-        bar = JsonProgressBar(iterator, epoch, prefix, log_interval)
-    elif log_format == "none":
-        bar = NoopProgressBar(iterator, epoch, prefix)
-    elif log_format == "simple":
-        bar = SimpleProgressBar(iterator, epoch, prefix, log_interval)
-    elif log_format == "tqdm":
-        bar = TqdmProgressBar(iterator, epoch, prefix)
-    else:
-        raise ValueError("Unknown log format: {}".format(log_format))
-
+    """Build the configured bar, optionally wrapped by the TB/wandb sink."""
+    kind = log_format if log_format is not None else default_log_format
+    if kind == "tqdm" and not sys.stderr.isatty():
+        kind = "simple"  # tqdm redraws are noise in captured logs
+    try:
+        bar = _BAR_KINDS[kind](iterator, epoch, prefix, log_interval)
+    except KeyError:
+        raise ValueError(f"Unknown log format: {kind}")
     if tensorboard_logdir:
         bar = TensorboardProgressBarWrapper(
             bar, tensorboard_logdir, wandb_project, args
         )
-
     return bar
 
 
-def format_stat(stat):
-    if isinstance(stat, Number):
-        stat = "{:g}".format(stat)
-    elif isinstance(stat, AverageMeter):
-        stat = "{:.3f}".format(stat.avg)
-    elif isinstance(stat, TimeMeter):
-        stat = "{:g}".format(round(stat.avg))
-    elif isinstance(stat, StopwatchMeter):
-        stat = "{:g}".format(round(stat.sum))
-    elif torch.is_tensor(stat):
-        stat = stat.tolist()
-    return stat
-
-
-class BaseProgressBar(object):
-    """Abstract class for progress bars."""
-
-    def __init__(self, iterable, epoch=None, prefix=None):
-        self.iterable = iterable
-        self.n = getattr(iterable, "n", 0)
-        self.epoch = epoch
-        self.prefix = ""
-        if epoch is not None:
-            self.prefix += "epoch {:03d}".format(epoch)
-        if prefix is not None:
-            self.prefix += (" | " if self.prefix != "" else "") + prefix
-
-    def __len__(self):
-        return len(self.iterable)
-
-    def __enter__(self):
-        return self
-
-    def __exit__(self, *exc):
-        return False
-
-    def __iter__(self):
-        raise NotImplementedError
-
-    def log(self, stats, tag=None, step=None):
-        """Log intermediate stats according to log_interval."""
-        raise NotImplementedError
-
-    def print(self, stats, tag=None, step=None):
-        """Print end-of-epoch stats."""
-        raise NotImplementedError
-
-    def _str_commas(self, stats):
-        return ", ".join(key + "=" + stats[key].strip() for key in stats.keys())
-
-    def _str_pipes(self, stats):
-        return " | ".join(key + " " + stats[key].strip() for key in stats.keys())
-
-    def _format_stats(self, stats):
-        postfix = OrderedDict(stats)
-        # Preprocess stats according to datatype
-        for key in postfix.keys():
-            postfix[key] = str(format_stat(postfix[key]))
-        return postfix
-
-
-@contextmanager
-def rename_logger(logger, new_name):
-    old_name = logger.name
-    if new_name is not None:
-        logger.name = new_name
-    yield logger
-    logger.name = old_name
-
-
-class JsonProgressBar(BaseProgressBar):
-    """Log output in JSON format."""
-
-    def __init__(self, iterable, epoch=None, prefix=None, log_interval=1000):
-        super().__init__(iterable, epoch, prefix)
-        self.log_interval = log_interval
-        self.i = None
-        self.size = None
-
-    def __iter__(self):
-        self.size = len(self.iterable)
-        for i, obj in enumerate(self.iterable, start=self.n):
-            self.i = i
-            yield obj
-
-    def log(self, stats, tag=None, step=None):
-        step = step or self.i or 0
-        if step > 0 and self.log_interval is not None and step % self.log_interval == 0:
-            update = (
-                self.epoch - 1 + (self.i + 1) / float(self.size)
-                if self.epoch is not None
-                else None
-            )
-            stats = self._format_stats(stats, epoch=self.epoch, update=update)
-            with rename_logger(logger, tag):
-                logger.info(json.dumps(stats))
-
-    def print(self, stats, tag=None, step=None):
-        self.stats = stats
-        if tag is not None:
-            self.stats = OrderedDict(
-                [(tag + "_" + k, v) for k, v in self.stats.items()]
-            )
-        stats = self._format_stats(self.stats, epoch=self.epoch)
-        with rename_logger(logger, tag):
-            logger.info(json.dumps(stats))
-
-    def _format_stats(self, stats, epoch=None, update=None):
-        postfix = OrderedDict()
-        if epoch is not None:
-            postfix["epoch"] = epoch
-        if update is not None:
-            postfix["update"] = round(update, 3)
-        for key in stats.keys():
-            postfix[key] = format_stat(stats[key])
-        return postfix
-
-
-class NoopProgressBar(BaseProgressBar):
-    """No logging."""
-
-    def __iter__(self):
-        for obj in self.iterable:
-            yield obj
-
-    def log(self, stats, tag=None, step=None):
-        pass
-
-    def print(self, stats, tag=None, step=None):
-        pass
-
-
-class SimpleProgressBar(BaseProgressBar):
-    """A minimal logger for non-TTY environments."""
-
-    def __init__(self, iterable, epoch=None, prefix=None, log_interval=1000):
-        super().__init__(iterable, epoch, prefix)
-        self.log_interval = log_interval
-        self.i = None
-        self.size = None
-
-    def __iter__(self):
-        self.size = len(self.iterable)
-        for i, obj in enumerate(self.iterable, start=self.n):
-            self.i = i
-            yield obj
-
-    def log(self, stats, tag=None, step=None):
-        step = step or self.i or 0
-        if step > 0 and self.log_interval is not None and step % self.log_interval == 0:
-            stats = self._format_stats(stats)
-            postfix = self._str_commas(stats)
-            with rename_logger(logger, tag):
-                logger.info(
-                    "{}:  {:5d} / {:d} {}".format(
-                        self.prefix, self.i + 1, self.size, postfix
-                    )
-                )
-
-    def print(self, stats, tag=None, step=None):
-        postfix = self._str_pipes(self._format_stats(stats))
-        with rename_logger(logger, tag):
-            logger.info("{} | {}".format(self.prefix, postfix))
-
-
-class TqdmProgressBar(BaseProgressBar):
-    """Log to tqdm."""
-
-    def __init__(self, iterable, epoch=None, prefix=None):
-        super().__init__(iterable, epoch, prefix)
-        try:
-            from tqdm import tqdm
-
-            self.tqdm = tqdm(
-                iterable,
-                self.prefix,
-                leave=False,
-                disable=False,
-            )
-        except ImportError:
-            # tqdm not installed: degrade to simple logging
-            self.tqdm = None
-            self._simple = SimpleProgressBar(iterable, epoch, prefix, 100)
-
-    def __iter__(self):
-        if self.tqdm is None:
-            return iter(self._simple)
-        return iter(self.tqdm)
-
-    def log(self, stats, tag=None, step=None):
-        if self.tqdm is None:
-            return self._simple.log(stats, tag, step)
-        self.tqdm.set_postfix(self._format_stats(stats), refresh=False)
-
-    def print(self, stats, tag=None, step=None):
-        postfix = self._str_pipes(self._format_stats(stats))
-        with rename_logger(logger, tag):
-            logger.info("{} | {}".format(self.prefix, postfix))
-
-
 class TensorboardProgressBarWrapper(BaseProgressBar):
-    """Log to tensorboard (and optionally wandb); wraps another bar."""
+    """Mirrors numeric stats into tensorboard (and wandb) around an inner bar."""
 
-    def __init__(self, wrapped_bar, tensorboard_logdir, wandb_project=None, args=None):
+    def __init__(self, wrapped_bar, tensorboard_logdir, wandb_project=None,
+                 args=None):
         self.wrapped_bar = wrapped_bar
         self.tensorboard_logdir = tensorboard_logdir
         self._writers = {}
@@ -269,33 +262,33 @@ class TensorboardProgressBarWrapper(BaseProgressBar):
             try:
                 import wandb
 
-                wandb.init(project=wandb_project, config=vars(args) if args else None)
+                wandb.init(project=wandb_project,
+                           config=vars(args) if args else None)
                 self.wandb = wandb
             except ImportError:
                 logger.warning("wandb not found; skipping wandb logging")
-        try:
-            from torch.utils.tensorboard import SummaryWriter
+        self.SummaryWriter = self._locate_summary_writer()
 
-            self.SummaryWriter = SummaryWriter
-        except ImportError:
+    @staticmethod
+    def _locate_summary_writer():
+        for modname in ("torch.utils.tensorboard", "tensorboardX"):
             try:
-                from tensorboardX import SummaryWriter
-
-                self.SummaryWriter = SummaryWriter
+                mod = __import__(modname, fromlist=["SummaryWriter"])
+                return mod.SummaryWriter
             except ImportError:
-                logger.warning(
-                    "tensorboard not found; please install with: pip install tensorboard"
-                )
-                self.SummaryWriter = None
+                continue
+        logger.warning(
+            "tensorboard not found; please install with: pip install tensorboard"
+        )
+        return None
 
     def _writer(self, key):
         if self.SummaryWriter is None:
             return None
         if key not in self._writers:
-            self._writers[key] = self.SummaryWriter(
-                os.path.join(self.tensorboard_logdir, key)
-            )
-            self._writers[key].add_text("sys.argv", " ".join(sys.argv))
+            w = self.SummaryWriter(os.path.join(self.tensorboard_logdir, key))
+            w.add_text("sys.argv", " ".join(sys.argv))
+            self._writers[key] = w
         return self._writers[key]
 
     def __len__(self):
@@ -304,31 +297,30 @@ class TensorboardProgressBarWrapper(BaseProgressBar):
     def __iter__(self):
         return iter(self.wrapped_bar)
 
-    def log(self, stats, tag=None, step=None):
-        self._log_to_tensorboard(stats, tag, step)
+    def log(self, stats, tag=None, step=None) -> None:
+        self._emit(stats, tag, step)
         self.wrapped_bar.log(stats, tag=tag, step=step)
 
-    def print(self, stats, tag=None, step=None):
-        self._log_to_tensorboard(stats, tag, step)
+    def print(self, stats, tag=None, step=None) -> None:
+        self._emit(stats, tag, step)
         self.wrapped_bar.print(stats, tag=tag, step=step)
 
-    def _log_to_tensorboard(self, stats, tag=None, step=None):
+    def _emit(self, stats, tag=None, step=None):
         writer = self._writer(tag or "")
         if writer is None and self.wandb is None:
             return
         if step is None:
-            step = stats["num_updates"] if "num_updates" in stats else 0
+            step = stats.get("num_updates", 0) if hasattr(stats, "get") else 0
         for key in stats.keys() - {"num_updates"}:
-            if isinstance(stats[key], AverageMeter):
-                val = stats[key].val
-            elif isinstance(stats[key], Number):
-                val = stats[key]
-            else:
+            value = stats[key]
+            if isinstance(value, AverageMeter):
+                value = value.val
+            elif not isinstance(value, Number):
                 continue
             if writer is not None:
-                writer.add_scalar(key, val, step)
+                writer.add_scalar(key, value, step)
             if self.wandb is not None:
-                prefix = (tag + "/") if tag else ""
-                self.wandb.log({prefix + key: val}, step=step)
+                ns = f"{tag}/" if tag else ""
+                self.wandb.log({ns + key: value}, step=step)
         if writer is not None:
             writer.flush()
