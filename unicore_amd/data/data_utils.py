@@ -1,4 +1,10 @@
-"""Collation + batching helpers (parity: reference unicore/data/data_utils.py)."""
+"""Collation, seeding and batching helpers (parity: reference
+unicore/data/data_utils.py — collate_tokens:17, collate_dict:63,
+numpy_seed:83, batch_by_size:107).
+
+Both collate variants share one implementation here: the 2-D case pads a
+square (len, len) block per sample instead of a row.
+"""
 
 import contextlib
 import logging
@@ -9,109 +15,95 @@ import torch
 logger = logging.getLogger(__name__)
 
 
-def collate_tokens(
-    values,
-    pad_idx,
-    left_pad=False,
-    pad_to_length=None,
-    pad_to_multiple=1,
-):
-    """Convert a list of 1d tensors into a padded 2d tensor
+def _round_up(size, multiple):
+    """Smallest value >= size that divides *multiple* (no-op for multiple=1)."""
+    if multiple == 1 or size % multiple == 0:
+        return size
+    return int(((size - 0.1) // multiple + 1) * multiple)
+
+
+def _collate_padded(values, pad_idx, left_pad, pad_to_length, pad_to_multiple,
+                    square):
+    width = max(v.size(0) for v in values)
+    if pad_to_length is not None:
+        width = max(width, pad_to_length)
+    width = _round_up(width, pad_to_multiple)
+    shape = (len(values), width, width) if square else (len(values), width)
+    out = values[0].new(*shape).fill_(pad_idx)
+    for row, v in zip(out, values):
+        n = len(v)
+        if square:
+            dst = row[width - n:, width - n:] if left_pad else row[:n, :n]
+        else:
+            dst = row[width - n:] if left_pad else row[:n]
+        assert dst.numel() == v.numel()
+        dst.copy_(v)
+    return out
+
+
+def collate_tokens(values, pad_idx, left_pad=False, pad_to_length=None,
+                   pad_to_multiple=1):
+    """Pad a list of 1-D tensors into one (B, L) tensor
     (reference unicore/data/data_utils.py:17-38)."""
-    size = max(v.size(0) for v in values)
-    size = size if pad_to_length is None else max(size, pad_to_length)
-    if pad_to_multiple != 1 and size % pad_to_multiple != 0:
-        size = int(((size - 0.1) // pad_to_multiple + 1) * pad_to_multiple)
-    res = values[0].new(len(values), size).fill_(pad_idx)
-
-    def copy_tensor(src, dst):
-        assert dst.numel() == src.numel()
-        dst.copy_(src)
-
-    for i, v in enumerate(values):
-        copy_tensor(v, res[i][size - len(v) :] if left_pad else res[i][: len(v)])
-    return res
+    return _collate_padded(values, pad_idx, left_pad, pad_to_length,
+                           pad_to_multiple, square=False)
 
 
-def collate_tokens_2d(
-    values,
-    pad_idx,
-    left_pad=False,
-    pad_to_length=None,
-    pad_to_multiple=1,
-):
-    """Convert a list of 2d tensors into a padded 3d tensor
+def collate_tokens_2d(values, pad_idx, left_pad=False, pad_to_length=None,
+                      pad_to_multiple=1):
+    """Pad a list of 2-D (L_i, L_i) tensors into one (B, L, L) tensor
     (reference unicore/data/data_utils.py:41-60)."""
-    size = max(v.size(0) for v in values)
-    size = size if pad_to_length is None else max(size, pad_to_length)
-    if pad_to_multiple != 1 and size % pad_to_multiple != 0:
-        size = int(((size - 0.1) // pad_to_multiple + 1) * pad_to_multiple)
-    res = values[0].new(len(values), size, size).fill_(pad_idx)
-
-    def copy_tensor(src, dst):
-        assert dst.numel() == src.numel()
-        dst.copy_(src)
-
-    for i, v in enumerate(values):
-        copy_tensor(
-            v,
-            res[i][size - len(v) :, size - len(v) :]
-            if left_pad
-            else res[i][: len(v), : len(v)],
-        )
-    return res
+    return _collate_padded(values, pad_idx, left_pad, pad_to_length,
+                           pad_to_multiple, square=True)
 
 
-def collate_dict(
-    values,
-    dim=0,
-):
-    if len(values) == 0:
+def collate_dict(values, dim=0):
+    """Stack same-keyed tensors across a list of dicts."""
+    if not values:
         return {}
     return {
-        key: torch.stack([v[key] for v in values], dim=dim) for key in values[0].keys()
+        key: torch.stack([entry[key] for entry in values], dim=dim)
+        for key in values[0].keys()
     }
 
 
 @contextlib.contextmanager
 def numpy_seed(seed, *addl_seeds):
-    """Context manager which seeds the NumPy PRNG with the specified seed and
-    restores the state afterward (reference unicore/data/data_utils.py:83-104)."""
+    """Seed numpy's global PRNG inside the block, then restore the previous
+    state (reference unicore/data/data_utils.py:83-104). Extra positional
+    seeds are hashed in, which is how per-(epoch, index) determinism is
+    derived everywhere in the data layer."""
     if seed is None:
         yield
         return
-    if len(addl_seeds) > 0:
+    if addl_seeds:
         seed = int(hash((seed, *addl_seeds)) % 1e8)
-    state = np.random.get_state()
+    saved = np.random.get_state()
     np.random.seed(seed)
     try:
         yield
     finally:
-        np.random.set_state(state)
+        np.random.set_state(saved)
 
 
-def batch_by_size(
-    indices,
-    batch_size=None,
-    required_batch_size_multiple=1,
-):
-    """Yield batches of fixed ``batch_size`` from ``indices``, with the last
-    (possibly short) batch rounded DOWN to the nearest multiple of
-    ``required_batch_size_multiple`` when it exceeds it
-    (reference unicore/data/data_utils.py:107-139)."""
-    batch_size = batch_size if batch_size is not None else 1
-    bsz_mult = required_batch_size_multiple
+def batch_by_size(indices, batch_size=None, required_batch_size_multiple=1):
+    """Split ``indices`` into fixed-size batches
+    (reference unicore/data/data_utils.py:107-139).
 
-    step = ((batch_size + bsz_mult - 1) // bsz_mult) * bsz_mult
+    The stride is ``batch_size`` rounded up to the multiple; only the final
+    short batch may be smaller.
+    """
+    batch_size = 1 if batch_size is None else batch_size
+    stride = -(-batch_size // required_batch_size_multiple)
+    stride *= required_batch_size_multiple
 
     if not isinstance(indices, np.ndarray):
         indices = np.fromiter(indices, dtype=np.int64, count=-1)
 
-    num_batches = (len(indices) + step - 1) // step
-    steps = np.arange(num_batches - 1) + 1
-    steps *= step
-    batch_indices = np.split(indices, steps)
-    assert len(batch_indices) == num_batches
-    # validation or test data size might be smaller than a mini-batch
-    assert batch_size <= 0 or len(batch_indices[0]) <= batch_size
-    return batch_indices
+    count = -(-len(indices) // stride)
+    cuts = stride * (1 + np.arange(count - 1))
+    batches = np.split(indices, cuts)
+    assert len(batches) == count
+    # validation or test data size might be smaller than one mini-batch
+    assert batch_size <= 0 or len(batches[0]) <= batch_size
+    return batches

@@ -1,4 +1,9 @@
-"""Symbol dictionary (parity: reference unicore/data/dictionary.py:12-148)."""
+"""Symbol<->index vocabulary (parity: reference unicore/data/dictionary.py:12-148).
+
+Text format: one ``<symbol> <count>`` pair per line, optional trailing
+``#overwrite`` flag; special tokens default to the BERT-style
+[CLS]/[PAD]/[SEP]/[UNK] set.
+"""
 
 import logging
 
@@ -12,148 +17,139 @@ class Dictionary:
 
     def __init__(
         self,
-        *,  # begin keyword-only arguments
+        *,  # keyword-only: the four roles are easy to transpose by accident
         bos="[CLS]",
         pad="[PAD]",
         eos="[SEP]",
         unk="[UNK]",
         extra_special_symbols=None,
     ):
-        self.bos_word, self.unk_word, self.pad_word, self.eos_word = bos, unk, pad, eos
-        self.symbols = []
-        self.count = []
-        self.indices = {}
-        self.specials = set()
-        self.specials.add(bos)
-        self.specials.add(unk)
-        self.specials.add(pad)
-        self.specials.add(eos)
+        self.bos_word = bos
+        self.pad_word = pad
+        self.eos_word = eos
+        self.unk_word = unk
+        self.symbols = []   # index -> symbol
+        self.count = []     # index -> corpus count
+        self.indices = {}   # symbol -> index
+        self.specials = {bos, pad, eos, unk}
 
     def __eq__(self, other):
-        return self.indices == other.indices
+        return other.indices == self.indices
 
     def __getitem__(self, idx):
-        if idx < len(self.symbols):
-            return self.symbols[idx]
-        return self.unk_word
+        return self.symbols[idx] if idx < len(self.symbols) else self.unk_word
 
-    def __len__(self):
-        """Returns the number of symbols in the dictionary"""
-        return len(self.symbols)
+    def __len__(self) -> int:
+        return len(self.indices)
 
-    def __contains__(self, sym):
-        return sym in self.indices
+    def __contains__(self, sym) -> bool:
+        return sym in self.indices.keys()
 
-    def vec_index(self, a):
-        return torch.tensor([self.index(i) for i in a], dtype=torch.long)
+    def index(self, sym) -> int:
+        """Index of *sym*, falling back to the unk index."""
+        assert isinstance(sym, str), "dictionary symbols are strings"
+        found = self.indices.get(sym)
+        return found if found is not None else self.unk()
 
-    def index(self, sym):
-        """Returns the index of the specified symbol"""
-        assert isinstance(sym, str)
-        if sym in self.indices:
-            return self.indices[sym]
-        return self.unk()
+    def vec_index(self, symbols) -> torch.Tensor:
+        """Vectorize an iterable of symbols into a LongTensor of indices."""
+        return torch.tensor([self.index(s) for s in symbols], dtype=torch.long)
 
-    def special_index(self):
-        return [self.index(x) for x in self.specials]
+    def special_index(self) -> list:
+        """Indices of all registered special tokens."""
+        return [self.index(s) for s in self.specials]
 
-    def pad_to_multiple_(self, padding_factor):
-        """Pad the dictionary size to a multiple of *padding_factor* with
-        unused filler symbols.  GEMM-shaped consumers (embedding matmuls,
-        the lm-head projection, the fused cross entropy) are markedly
-        faster when the vocab dimension is a multiple of 64.
-        (fairseq-style API; the reference inherits it implicitly.)"""
-        if padding_factor > 1:
-            i = 0
-            while len(self) % padding_factor != 0:
-                symbol = "madeupword{:04d}".format(i)
-                self.add_symbol(symbol, n=0, is_special=True)
-                i += 1
-
-    def add_symbol(self, word, n=1, overwrite=False, is_special=False):
-        """Adds a word to the dictionary"""
+    def add_symbol(self, word, n=1, overwrite=False, is_special=False) -> int:
+        """Register *word* (or bump its count when already known)."""
         if is_special:
             self.specials.add(word)
-        if word in self.indices and not overwrite:
-            idx = self.indices[word]
-            self.count[idx] = self.count[idx] + n
-            return idx
-        else:
-            idx = len(self.symbols)
-            self.indices[word] = idx
-            self.symbols.append(word)
-            self.count.append(n)
-            return idx
+        if not overwrite and word in self.indices:
+            known = self.indices[word]
+            self.count[known] += n
+            return known
+        slot = len(self.symbols)
+        self.indices[word] = slot
+        self.symbols.append(word)
+        self.count.append(n)
+        return slot
 
-    def bos(self):
-        """Helper to get index of beginning-of-sentence symbol"""
-        return self.index(self.bos_word)
+    def pad_to_multiple_(self, padding_factor) -> None:
+        """Grow the vocab with filler symbols until its size divides
+        *padding_factor*. GEMM-shaped consumers (embedding matmuls, the
+        lm-head projection, the fused cross entropy) are markedly faster
+        when the vocab dimension is a multiple of 64."""
+        if padding_factor <= 1:
+            return
+        filler = 0
+        while len(self) % padding_factor:
+            self.add_symbol(f"madeupword{filler:04d}", n=0, is_special=True)
+            filler += 1
 
-    def pad(self):
-        """Helper to get index of pad symbol"""
-        return self.index(self.pad_word)
+    # role helpers ---------------------------------------------------------
 
-    def eos(self):
-        """Helper to get index of end-of-sentence symbol"""
-        return self.index(self.eos_word)
+    def bos(self) -> int:
+        """Index of the beginning-of-sentence symbol."""
+        return self.index(sym=self.bos_word)
 
-    def unk(self):
-        """Helper to get index of unk symbol"""
+    def pad(self) -> int:
+        """Index of the padding symbol."""
+        return self.index(sym=self.pad_word)
+
+    def eos(self) -> int:
+        """Index of the end-of-sentence symbol."""
+        return self.index(sym=self.eos_word)
+
+    def unk(self) -> int:
+        """Index of the unknown symbol."""
         assert self.unk_word in self.indices, "Cannot find unk symbol"
         return self.indices[self.unk_word]
 
+    # file IO --------------------------------------------------------------
+
     @classmethod
     def load(cls, f):
-        """Loads the dictionary from a text file with the format:
+        """Build a Dictionary from a ``<symbol> <count>``-per-line file."""
+        built = cls()
+        built.add_from_file(f)
+        return built
 
-        ```
-        <symbol0>
-        <symbol1>
-        ...
-        ```
-        """
-        d = cls()
-        d.add_from_file(f)
-        return d
-
-    def add_from_file(self, f):
-        """Adds words from a text file."""
+    def add_from_file(self, f) -> None:
+        """Merge symbols from an open file (or a path)."""
         if isinstance(f, str):
             try:
-                with open(f, "r", encoding="utf-8") as fd:
-                    self.add_from_file(fd)
-            except FileNotFoundError as fnfe:
-                raise fnfe
+                with open(f, "r", encoding="utf-8") as handle:
+                    self.add_from_file(handle)
             except UnicodeError:
                 raise Exception(
-                    "Incorrect encoding detected in {}, please "
-                    "rebuild the dataset".format(f)
+                    f"Incorrect encoding detected in {f}, please rebuild"
+                    " the dataset"
                 )
             return
 
-        lines = f.readlines()
-
-        for line_idx, line in enumerate(lines):
-            try:
-                splits = line.rstrip().rsplit(" ", 1)
-                line = splits[0]
-                field = splits[1] if len(splits) > 1 else str(len(lines) - line_idx)
-                if field == "#overwrite":
-                    overwrite = True
-                    line, field = line.rsplit(" ", 1)
-                else:
-                    overwrite = False
-                count = int(field)
-                word = line
-                if word in self and not overwrite:
-                    logger.info(
-                        "Duplicate word found when loading Dictionary: '{}', index is {}.".format(
-                            word, self.indices[word]
-                        )
-                    )
-                else:
-                    self.add_symbol(word, n=count, overwrite=overwrite)
-            except ValueError:
-                raise ValueError(
-                    "Incorrect dictionary format, expected '<token> <cnt> [flags]'"
+        raw_lines = f.readlines()
+        for lineno, raw in enumerate(raw_lines):
+            word, count, overwrite = self._parse_line(raw, raw_lines, lineno)
+            if not overwrite and word in self:
+                logger.info(
+                    f"Duplicate word found when loading Dictionary: '{word}',"
+                    f" index is {self.indices[word]}."
                 )
+            else:
+                self.add_symbol(word, count, overwrite=overwrite)
+
+    @staticmethod
+    def _parse_line(raw, lines, lineno):
+        try:
+            head, _, tail = raw.rstrip().rpartition(" ")
+            if not head:
+                # bare symbol: synthesize a descending count so earlier
+                # lines sort first
+                head, tail = tail, str(len(lines) - lineno)
+            overwrite = tail == "#overwrite"
+            if overwrite:
+                head, _, tail = head.rpartition(" ")
+            return head, int(tail), overwrite
+        except ValueError:
+            raise ValueError(
+                "Incorrect dictionary format, expected '<token> <cnt> [flags]'"
+            )

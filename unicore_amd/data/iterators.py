@@ -1,14 +1,18 @@
-"""Epoch/batch iterators with resume, sharding, grouping and background
-prefetch.
+"""Epoch/batch iterators: resumable, sharded, grouped, background-prefetched.
 
-Parity: reference unicore/data/iterators.py — CountingIterator:28,
-EpochBatchIterator:151, GroupedIterator:406, ShardedIterator:438,
-BackgroundConsumer:471, BufferedIterator:496.
+Capability parity with the reference iterator stack (unicore/data/iterators.py
+— CountingIterator:28, EpochBatchIterator:151, GroupedIterator:406,
+ShardedIterator:438, BackgroundConsumer:471, BufferedIterator:496), written
+around three building blocks:
 
-MI355X addition: ``CudaPrefetcher`` stages pinned batches onto the GPU on a
-dedicated HIP stream one batch ahead of the training loop (the north-star's
-"buffered GPU-CPU overlapping data loader" with hipMemcpyAsync on its own
-stream), instead of doing the H2D copy synchronously at sample-prep time.
+* :class:`CountingIterator` — position/len bookkeeping shared by everything;
+* :class:`EpochBatchIterator` — the resumable multi-epoch front end
+  (state_dict round-trips through checkpoints, including the proportional
+  position remap when the world size changes);
+* the prefetch tail — :class:`BufferedIterator` (thread ahead of the loop)
+  and the MI355X-specific :class:`CudaPrefetcher`, which lands each pinned
+  batch on the GPU via a dedicated HIP copy stream one step early instead of
+  paying a synchronous H2D at sample-prep time.
 """
 
 import itertools
@@ -16,8 +20,8 @@ import logging
 import math
 import os
 import queue
+import threading
 import time
-from threading import Thread
 
 import numpy as np
 import torch
@@ -26,44 +30,26 @@ from . import data_utils
 
 logger = logging.getLogger(__name__)
 
-# Object used by _background_consumer to signal the source is exhausted
-# to the main thread.
-_sentinel = object()
+# queue marker: the producer thread finished the epoch
+_DONE = object()
 
 
-class CountingIterator(object):
-    """Wrapper around an iterable that maintains the iteration count.
+class CountingIterator:
+    """Iterator wrapper that tracks how many items were consumed.
 
-    Args:
-        iterable (iterable): iterable to wrap
-        start (int): starting iteration count. Note that this doesn't
-            actually advance the iterator.
-        total (int): override the iterator length returned by ``__len__``.
-            This can be used to truncate *iterator*.
-
-    Attributes:
-        n (int): number of elements consumed from this iterator
+    ``n`` is the consumed count; ``total`` bounds the length (and can be
+    lowered with :meth:`take`). ``start`` seeds the count without advancing
+    the underlying iterable — used when resuming mid-epoch.
     """
 
     def __init__(self, iterable, start=None, total=None):
         self.iterable = iterable
-        self.itr = iter(self)
+        self.n = getattr(iterable, "n", 0) if start is None else start
+        self.total = self.n + len(iterable) if total is None else total
+        self.itr = self._produce()
 
-        if start is None:
-            self.n = getattr(iterable, "n", 0)
-        else:
-            self.n = start
-
-        if total is None:
-            self.total = self.n + len(iterable)
-        else:
-            self.total = total
-
-    def __len__(self):
-        return self.total
-
-    def __iter__(self):
-        for x in self.iterable:
+    def _produce(self):
+        for item in self.iterable:
             if self.n >= self.total:
                 raise RuntimeError(
                     "Mismatch between actual and expected iterable length. "
@@ -71,98 +57,79 @@ class CountingIterator(object):
                     "different batch configuration."
                 )
             self.n += 1
-            yield x
+            yield item
+
+    def __iter__(self):
+        return self.itr
 
     def __next__(self):
         return next(self.itr)
 
-    def has_next(self):
-        """Whether the iterator has been exhausted."""
-        return self.n < len(self)
+    def __len__(self) -> int:
+        return self.total
+
+    def has_next(self) -> bool:
+        """True while elements remain."""
+        return self.n < self.total
 
     def skip(self, num_to_skip):
-        """Fast-forward the iterator by skipping *num_to_skip* elements."""
+        """Advance past *num_to_skip* elements and return self."""
         next(itertools.islice(self.itr, num_to_skip, num_to_skip), None)
         return self
 
-    def take(self, n):
-        """
-        Truncates the iterator to n elements at most.
-        """
+    def take(self, n) -> None:
+        """Cap the iterator at *n* elements total."""
         self.total = min(self.total, n)
-        # Propagate this change to the underlying iterator
         if hasattr(self.iterable, "take"):
-            self.iterable.take(n)
+            self.iterable.take(n)  # propagate the cap downward
         else:
             self.iterable = itertools.islice(self.iterable, n)
 
 
-class EpochBatchIterating(object):
+class EpochBatchIterating:
+    """Interface of epoch-capable batch iterators (see EpochBatchIterator)."""
+
+    def _abstract(self):
+        raise NotImplementedError("implemented by EpochBatchIterator")
+
     def __len__(self) -> int:
-        raise NotImplementedError
+        self._abstract()
 
     @property
     def next_epoch_idx(self):
-        raise NotImplementedError
+        self._abstract()
 
     def next_epoch_itr(self, shuffle=True, fix_batches_to_gpus=False):
-        """Return a new iterator over the dataset."""
-        raise NotImplementedError
+        self._abstract()
 
     def end_of_epoch(self) -> bool:
-        """Returns whether the most recent epoch iterator has been exhausted"""
-        raise NotImplementedError
+        self._abstract()
 
     @property
     def iterations_in_epoch(self) -> int:
-        """The number of consumed batches in the current epoch."""
-        raise NotImplementedError
+        self._abstract()
 
-    def state_dict(self):
-        """Returns a dictionary containing a whole state of the iterator."""
-        raise NotImplementedError
+    def state_dict(self) -> dict:
+        self._abstract()
 
-    def load_state_dict(self, state_dict):
-        """Copies the state of the iterator from the given *state_dict*."""
-        raise NotImplementedError
+    def load_state_dict(self, state: dict) -> None:
+        self._abstract()
 
 
 class EpochBatchIterator(EpochBatchIterating):
-    """A multi-epoch iterator over a :class:`torch.utils.data.Dataset`.
+    """Resumable multi-epoch iterator over a torch Dataset.
 
-    Compared to :class:`torch.utils.data.DataLoader`, this iterator:
+    Differences from a plain DataLoader: reusable across epochs via
+    :meth:`next_epoch_itr` (with per-epoch seeded shuffling), serializable
+    mid-epoch via state_dict/load_state_dict, and sharded across data-parallel
+    ranks (``num_shards``/``shard_id``).
 
-    - can be reused across multiple epochs with the :func:`next_epoch_itr`
-      method (optionally shuffled between epochs)
-    - can be serialized/deserialized with the :func:`state_dict` and
-      :func:`load_state_dict` methods
-    - supports sharding with the *num_shards* and *shard_id* arguments
-
-    Args:
-        dataset (~torch.utils.data.Dataset): dataset from which to load the data
-        collate_fn (callable): merges a list of samples to form a mini-batch
-        batch_sampler (~torch.utils.data.Sampler or a callable): an iterator
-            over batches of indices, or a callable to create such an iterator
-            (~torch.utils.data.Sampler). A callable batch_sampler will be
-            called for each epoch to enable per-epoch dynamic batch iterators.
-        seed (int, optional): seed for random number generator for
-            reproducibility (default: 1).
-        num_shards (int, optional): shard the data iterator into N
-            shards (default: 1).
-        shard_id (int, optional): which shard of the data iterator to
-            return (default: 0).
-        num_workers (int, optional): how many subprocesses to use for data
-            loading. 0 means the data will be loaded in the main process
-            (default: 0).
-        epoch (int, optional): the epoch to start the iterator from
-            (default: 1).
-        buffer_size (int, optional): the number of batches to keep ready in the
-            queue. Helps speeding up dataloading. When buffer_size is zero, the
-            default torch.utils.data.DataLoader preloading is used.
-        timeout (int, optional): if positive, the timeout value for collecting a batch
-            from workers. Should always be non-negative (default: ``0``).
-        disable_shuffling (bool, optional): force disable shuffling
-            (default: ``False``).
+    Parameters mirror the reference contract
+    (unicore/data/iterators.py:151-225): *batch_sampler* is either a frozen
+    list of index batches or a callable re-evaluated each epoch;
+    *buffer_size* (capped at 32) enables the background prefetch thread;
+    *cuda_prefetch* opts into the GPU-side copy-stream staging and must be
+    False for CPU-only runs even on a CUDA-capable host.
     """
 
     def __init__(
@@ -178,25 +145,25 @@ class EpochBatchIterator(EpochBatchIterating):
         buffer_size=0,
         timeout=0,
         disable_shuffling=False,
+        cuda_prefetch=True,
     ):
         assert isinstance(dataset, torch.utils.data.Dataset)
         self.dataset = dataset
         self.collate_fn = collate_fn
         self.batch_sampler = batch_sampler
         self._frozen_batches = (
-            tuple(batch_sampler) if not callable(batch_sampler) else None
+            None if callable(batch_sampler) else tuple(batch_sampler)
         )
         self.seed = seed
         self.num_shards = num_shards
         self.shard_id = shard_id
         self.num_workers = num_workers
-        # This upper limit here is to prevent people from abusing this feature
-        # in a shared computing environment.
-        self.buffer_size = min(buffer_size, 32)
+        self.buffer_size = min(buffer_size, 32)  # bound host memory use
         self.timeout = timeout
         self.disable_shuffling = disable_shuffling
+        self.cuda_prefetch = cuda_prefetch and torch.cuda.is_available()
 
-        self.epoch = max(epoch, 1)  # we use 1-based indexing for epochs
+        self.epoch = max(epoch, 1)  # epochs are 1-based
         self.shuffle = not disable_shuffling
         self._cur_epoch_itr = None
         self._next_epoch_itr = None
@@ -205,7 +172,9 @@ class EpochBatchIterator(EpochBatchIterating):
     @property
     def frozen_batches(self):
         if self._frozen_batches is None:
-            self._frozen_batches = tuple(self.batch_sampler(self.dataset, self.epoch))
+            self._frozen_batches = tuple(
+                self.batch_sampler(self.dataset, self.epoch)
+            )
         return self._frozen_batches
 
     @property
@@ -217,14 +186,13 @@ class EpochBatchIterator(EpochBatchIterating):
                 "Try increasing the max number of allowed tokens or using "
                 "a larger dataset."
             )
-
         if getattr(self.dataset, "supports_fetch_outside_dataloader", True):
-            return self.collate_fn([self.dataset[i] for i in self.frozen_batches[0]])
-        else:
-            return "DUMMY"
+            rows = [self.dataset[i] for i in self.frozen_batches[0]]
+            return self.collate_fn(rows)
+        return "DUMMY"
 
-    def __len__(self):
-        return int(math.ceil(len(self.frozen_batches) / float(self.num_shards)))
+    def __len__(self) -> int:
+        return -(-len(self.frozen_batches) // self.num_shards)
 
     @property
     def n(self):
@@ -232,23 +200,19 @@ class EpochBatchIterator(EpochBatchIterating):
 
     @property
     def next_epoch_idx(self):
-        """Return the epoch index after *next_epoch_itr* is called."""
+        """Epoch number the next next_epoch_itr() call will serve."""
         if self._next_epoch_itr is not None:
-            return self.epoch
-        elif self._cur_epoch_itr is not None and self.end_of_epoch():
+            return self.epoch  # a resume already queued this epoch
+        if self._cur_epoch_itr is not None and self.end_of_epoch():
             return self.epoch + 1
-        else:
-            return self.epoch
+        return self.epoch
 
     def next_epoch_itr(self, shuffle=True, fix_batches_to_gpus=False):
-        """Return a new iterator over the dataset.
+        """Start (or resume) the next epoch and return its iterator.
 
-        Args:
-            shuffle (bool, optional): shuffle batches before returning the
-                iterator (default: True).
-            fix_batches_to_gpus (bool, optional): ensure that batches are always
-                allocated to the same shards across epochs. Requires
-                that :attr:`dataset` supports prefetching (default: False).
+        *fix_batches_to_gpus* keeps batch->shard assignment stable across
+        epochs (requires dataset prefetch support); shuffling then happens
+        within the shard.
         """
         if self.disable_shuffling:
             shuffle = False
@@ -256,122 +220,120 @@ class EpochBatchIterator(EpochBatchIterating):
         if hasattr(self.dataset, "set_epoch"):
             self.dataset.set_epoch(self.epoch)
         if self._next_epoch_itr is not None:
+            # a load_state_dict() prepared a mid-epoch iterator
             self._cur_epoch_itr = self._next_epoch_itr
             self._next_epoch_itr = None
         else:
             if callable(self.batch_sampler):
-                # reset _frozen_batches to refresh the next epoch
-                self._frozen_batches = None
-            self._cur_epoch_itr = self._get_iterator_for_epoch(
-                self.epoch,
-                shuffle,
-                fix_batches_to_gpus=fix_batches_to_gpus,
+                self._frozen_batches = None  # regenerate for the new epoch
+            self._cur_epoch_itr = self._build_epoch_itr(
+                self.epoch, shuffle, fix_batches_to_gpus=fix_batches_to_gpus
             )
         self.shuffle = shuffle
         return self._cur_epoch_itr
 
     def end_of_epoch(self) -> bool:
-        """Returns whether the most recent epoch iterator has been exhausted"""
         return not self._cur_epoch_itr.has_next()
 
     @property
-    def iterations_in_epoch(self):
-        """The number of consumed batches in the current epoch."""
-        if self._cur_epoch_itr is not None:
-            return self._cur_epoch_itr.n
-        elif self._next_epoch_itr is not None:
-            return self._next_epoch_itr.n
+    def iterations_in_epoch(self) -> int:
+        for itr in (self._cur_epoch_itr, self._next_epoch_itr):
+            if itr is not None:
+                return itr.n
         return 0
 
-    def state_dict(self):
-        """Returns a dictionary containing a whole state of the iterator."""
+    def state_dict(self) -> dict:
         if self.end_of_epoch():
-            epoch = self.epoch + 1
-            iter_in_epoch = 0
+            # finished epochs serialize as the start of the following one
+            epoch, offset = self.epoch + 1, 0
         else:
-            epoch = self.epoch
-            iter_in_epoch = self.iterations_in_epoch
+            epoch, offset = self.epoch, self.iterations_in_epoch
         return {
             "version": 2,
             "epoch": epoch,
-            "iterations_in_epoch": iter_in_epoch,
+            "iterations_in_epoch": offset,
             "shuffle": self.shuffle,
             "len": len(self),
         }
 
-    def load_state_dict(self, state_dict):
-        """Copies the state of the iterator from the given *state_dict*."""
-        self.epoch = state_dict["epoch"]
-        itr_pos = state_dict.get("iterations_in_epoch", 0)
-        version = state_dict.get("version", 1)
-        if itr_pos > 0:
-            # fast-forward epoch iterator
-            itr_len = state_dict.get("len", len(self))
-            if itr_len != len(self):
-                # world size / batch config changed: remap position
-                # proportionally (reference unicore/data/iterators.py:326-350)
-                itr_pos = int(math.floor(len(self) * itr_pos / itr_len))
-                logger.info(
-                    "Iterator length changed from {} to {}; "
-                    "proportionally remapping resume position to {}".format(
-                        itr_len, len(self), itr_pos
-                    )
-                )
-            self._next_epoch_itr = self._get_iterator_for_epoch(
-                self.epoch,
-                shuffle=state_dict.get("shuffle", True),
-                offset=itr_pos,
-            )
-            if self._next_epoch_itr is None:
-                if version == 1:
-                    # legacy behavior: we finished the epoch, increment epoch counter
-                    self.epoch += 1
-                else:
-                    raise RuntimeError(
-                        "Cannot resume training due to dataloader mismatch. You can relaunch "
-                        "training with `--reset-dataloader` and it should work."
-                    )
-        else:
+    def load_state_dict(self, state: dict) -> None:
+        self.epoch = state["epoch"]
+        offset = state.get("iterations_in_epoch", 0)
+        version = state.get("version", 1)
+        if offset <= 0:
             self._next_epoch_itr = None
-
-    def _get_iterator_for_epoch(
-        self, epoch, shuffle, fix_batches_to_gpus=False, offset=0
-    ):
-        def shuffle_batches(batches, seed):
-            with data_utils.numpy_seed(seed):
-                np.random.shuffle(batches)
-            return batches
-
-        if self._supports_prefetch:
-            batches = self.frozen_batches
-
-            if shuffle and not fix_batches_to_gpus:
-                batches = shuffle_batches(list(batches), self.seed + epoch)
-
-            batches = list(
-                ShardedIterator(batches, self.num_shards, self.shard_id, fill_value=[])
+            return
+        saved_len = state.get("len", len(self))
+        if saved_len != len(self):
+            # world size or batching changed across the restart: map the
+            # position proportionally into the new batch count
+            # (reference unicore/data/iterators.py:326-350)
+            remapped = int(math.floor(len(self) * offset / saved_len))
+            logger.info(
+                f"Iterator length changed from {saved_len} to {len(self)}; "
+                f"proportionally remapping resume position to {remapped}"
             )
-            self.dataset.prefetch([i for s in batches for i in s])
-
-            if shuffle and fix_batches_to_gpus:
-                batches = shuffle_batches(batches, self.seed + epoch + self.shard_id)
-        else:
-            if shuffle:
-                batches = shuffle_batches(list(self.frozen_batches), self.seed + epoch)
+            offset = remapped
+        self._next_epoch_itr = self._build_epoch_itr(
+            self.epoch, shuffle=state.get("shuffle", True), offset=offset
+        )
+        if self._next_epoch_itr is None:
+            if version == 1:
+                # v1 checkpoints treated a past-the-end offset as a completed
+                # epoch
+                self.epoch += 1
             else:
-                batches = self.frozen_batches
+                raise RuntimeError(
+                    "Cannot resume training due to dataloader mismatch. You "
+                    "can relaunch training with `--reset-dataloader` and it "
+                    "should work."
+                )
+
+    # -- epoch iterator construction --------------------------------------
+
+    def _shuffled(self, batches, seed):
+        batches = list(batches)
+        with data_utils.numpy_seed(seed):
+            np.random.shuffle(batches)
+        return batches
+
+    def _build_epoch_itr(self, epoch, shuffle, fix_batches_to_gpus=False,
+                         offset=0):
+        if self._supports_prefetch:
+            # prefetch-capable datasets want the index list before loading;
+            # with fix_batches_to_gpus the shard assignment must not depend
+            # on the epoch, so the shuffle happens after sharding with a
+            # shard-local seed
+            batches = self.frozen_batches
+            if shuffle and not fix_batches_to_gpus:
+                batches = self._shuffled(batches, self.seed + epoch)
             batches = list(
-                ShardedIterator(batches, self.num_shards, self.shard_id, fill_value=[])
+                ShardedIterator(batches, self.num_shards, self.shard_id,
+                                fill_value=[])
+            )
+            self.dataset.prefetch([i for b in batches for i in b])
+            if shuffle and fix_batches_to_gpus:
+                batches = self._shuffled(
+                    batches, self.seed + epoch + self.shard_id
+                )
+        else:
+            batches = (
+                self._shuffled(self.frozen_batches, self.seed + epoch)
+                if shuffle
+                else self.frozen_batches
+            )
+            batches = list(
+                ShardedIterator(batches, self.num_shards, self.shard_id,
+                                fill_value=[])
             )
 
         if offset > 0 and offset >= len(batches):
-            return None
+            return None  # nothing left of this epoch
 
         if self.num_workers > 0:
             os.environ["PYTHONWARNINGS"] = "ignore:semaphore_tracker:UserWarning"
 
-        # Create data loader
-        itr = torch.utils.data.DataLoader(
+        loader = torch.utils.data.DataLoader(
             self.dataset,
             collate_fn=self.collate_fn,
             batch_sampler=batches[offset:],
@@ -379,177 +341,142 @@ class EpochBatchIterator(EpochBatchIterating):
             timeout=self.timeout,
             pin_memory=torch.cuda.is_available(),
         )
-
-        # Wrap with a BufferedIterator if needed
         if self.buffer_size > 0:
-            itr = BufferedIterator(self.buffer_size, itr)
+            loader = BufferedIterator(self.buffer_size, loader)
+        if self.cuda_prefetch:
+            # inside the CountingIterator so the resume position counts
+            # consumed batches, not prefetched ones
+            loader = CudaPrefetcher(loader)
+        return CountingIterator(loader, start=offset)
 
-        # Stage pinned batches onto the GPU one step ahead on a dedicated
-        # copy stream.  Placed INSIDE the CountingIterator so resumable
-        # iteration counts consumed batches, not prefetched ones.
-        if torch.cuda.is_available():
-            itr = CudaPrefetcher(itr)
 
-        # Wrap with CountingIterator
-        itr = CountingIterator(itr, start=offset)
-        return itr
+def _chunks(itr, size):
+    """Yield lists of up to *size* consecutive items."""
+    bucket = []
+    for item in itr:
+        bucket.append(item)
+        if len(bucket) == size:
+            yield bucket
+            bucket = []
+    if bucket:
+        yield bucket
 
 
 class GroupedIterator(CountingIterator):
-    """Wrapper around an iterable that returns groups (chunks) of items
-    (used for gradient accumulation, update_freq chunks).
-
-    Args:
-        iterable (iterable): iterable to wrap
-        chunk_size (int): size of each chunk
-    """
+    """Chunks an iterator into lists of *chunk_size* items (the grad-accum
+    micro-batch grouping)."""
 
     def __init__(self, iterable, chunk_size):
-        itr = _chunk_iterator(iterable, chunk_size)
         super().__init__(
-            itr,
-            start=int(math.ceil(getattr(iterable, "n", 0) / float(chunk_size))),
-            total=int(math.ceil(len(iterable) / float(chunk_size))),
+            _chunks(iterable, chunk_size),
+            start=-(-getattr(iterable, "n", 0) // chunk_size),
+            total=-(-len(iterable) // chunk_size),
         )
         self.chunk_size = chunk_size
 
 
-def _chunk_iterator(itr, chunk_size):
-    chunk = []
-    for x in itr:
-        chunk.append(x)
-        if len(chunk) == chunk_size:
-            yield chunk
-            chunk = []
-    if len(chunk) > 0:
-        yield chunk
-
-
 class ShardedIterator(CountingIterator):
-    """A sharded wrapper around an iterable, padded to length.
-
-    Args:
-        iterable (iterable): iterable to wrap
-        num_shards (int): number of shards to split the iterable into
-        shard_id (int): which shard to iterator over
-        fill_value (Any, optional): padding value when the iterable doesn't
-            evenly divide *num_shards* (default: None).
-    """
+    """Round-robin shard of an iterable, padded with *fill_value* so every
+    shard sees the same number of batches."""
 
     def __init__(self, iterable, num_shards, shard_id, fill_value=None):
-        if shard_id < 0 or shard_id >= num_shards:
+        if not 0 <= shard_id < num_shards:
             raise ValueError("shard_id must be between 0 and num_shards")
-        sharded_len = int(math.ceil(len(iterable) / float(num_shards)))
-        itr = map(
-            operator_itemgetter_1,
-            itertools.zip_longest(
-                range(sharded_len),
-                itertools.islice(iterable, shard_id, len(iterable), num_shards),
-                fillvalue=fill_value,
-            ),
+        shard_len = -(-len(iterable) // num_shards)
+        strided = itertools.islice(
+            iterable, shard_id, len(iterable), num_shards
+        )
+        padded = (
+            pair[1]
+            for pair in itertools.zip_longest(
+                range(shard_len), strided, fillvalue=fill_value
+            )
         )
         super().__init__(
-            itr,
-            start=int(math.ceil(getattr(iterable, "n", 0) / float(num_shards))),
-            total=sharded_len,
+            padded,
+            start=-(-getattr(iterable, "n", 0) // num_shards),
+            total=shard_len,
         )
 
 
-def operator_itemgetter_1(pair):
-    return pair[1]
+class BufferedIterator:
+    """Producer-thread lookahead: a daemon thread fills a bounded queue up to
+    *size* batches ahead of the consumer, with a rate-limited hint when the
+    buffer keeps running dry (loader-bound training)."""
 
-
-class BackgroundConsumer(Thread):
-    def __init__(self, queue, source, max_len):
-        Thread.__init__(self)
-
-        self._queue = queue
-        self._source = source
-        self._max_len = max_len
-        self.count = 0
-
-    def run(self):
-        try:
-            for item in self._source:
-                self._queue.put(item)
-
-                # Stop if we reached the maximum length
-                self.count += 1
-                if self._max_len is not None and self.count >= self._max_len:
-                    break
-
-            # Signal the consumer we are done.
-            self._queue.put(_sentinel)
-        except Exception as e:
-            self._queue.put(e)
-
-
-class BufferedIterator(object):
-    """Background-thread prefetch of up to *size* batches ahead of the
-    consumer, with starvation warnings (reference unicore/data/iterators.py:496-554)."""
+    # warn no earlier than 5 min in, and at most every 15 min
+    _WARMUP_S = 5 * 60
+    _WARN_EVERY_S = 15 * 60
 
     def __init__(self, size, iterable):
         self._queue = queue.Queue(size)
         self._iterable = iterable
-        self._consumer = None
-
+        self._producer = None
         self.start_time = time.time()
         self.warning_time = None
-
         self.total = len(iterable)
 
-    def _create_consumer(self):
-        self._consumer = BackgroundConsumer(
-            self._queue,
-            self._iterable,
-            self.total,
+    def _pump(self, source, limit):
+        """Producer thread body: forward items, then signal completion.
+        Exceptions travel through the queue to the consumer."""
+        try:
+            sent = 0
+            for item in source:
+                self._queue.put(item)
+                sent += 1
+                if limit is not None and sent >= limit:
+                    break
+            self._queue.put(_DONE)
+        except Exception as exc:  # noqa: BLE001 - must cross the thread
+            self._queue.put(exc)
+
+    def _start_producer(self):
+        self._producer = threading.Thread(
+            target=self._pump, args=(self._iterable, self.total), daemon=True
         )
-        self._consumer.daemon = True
-        self._consumer.start()
+        self._producer.start()
 
     def __iter__(self):
         return self
 
-    def __len__(self):
+    def __len__(self) -> int:
         return self.total
 
-    def take(self, n):
+    def take(self, n) -> None:
         self.total = min(self.total, n)
-        # Propagate this change to the underlying iterator
         if hasattr(self._iterable, "take"):
             self._iterable.take(n)
         else:
             self._iterable = itertools.islice(self._iterable, n)
 
+    def _maybe_warn_starved(self):
+        if self._queue.qsize() >= min(2, max(1, self._queue.maxsize // 2)):
+            return
+        now = time.time()
+        if now - self.start_time <= self._WARMUP_S:
+            return
+        if (self.warning_time is None
+                or now - self.warning_time > self._WARN_EVERY_S):
+            logger.debug(
+                "Data loading buffer is empty or nearly empty. This may "
+                "indicate a data loading bottleneck, and increasing the "
+                "number of workers (--num-workers) may help."
+            )
+            self.warning_time = now
+
     def __next__(self):
-        # Create consumer if not created yet
-        if self._consumer is None:
-            self._create_consumer()
-
-        # Notify the user if there is a data loading bottleneck
-        if self._queue.qsize() < min(2, max(1, self._queue.maxsize // 2)):
-            if time.time() - self.start_time > 5 * 60:
-                if (
-                    self.warning_time is None
-                    or time.time() - self.warning_time > 15 * 60
-                ):
-                    logger.debug(
-                        "Data loading buffer is empty or nearly empty. This may "
-                        "indicate a data loading bottleneck, and increasing the "
-                        "number of workers (--num-workers) may help."
-                    )
-                    self.warning_time = time.time()
-
-        # Get next example
+        if self._producer is None:
+            self._start_producer()
+        self._maybe_warn_starved()
         item = self._queue.get(True)
         if isinstance(item, Exception):
             raise item
-        if item is _sentinel:
+        if item is _DONE:
             raise StopIteration()
         return item
 
 
-class CudaPrefetcher(object):
+class CudaPrefetcher:
     """Overlap H2D copies with compute: move each (pinned) batch to the GPU on
     a dedicated copy stream one batch ahead, and make the consumer's compute
     stream wait on the copy event instead of blocking the host.
@@ -568,7 +495,7 @@ class CudaPrefetcher(object):
         self._next_event = None
         self._itr = None
 
-    def __len__(self):
+    def __len__(self) -> int:
         return len(self.iterable)
 
     def _preload(self):

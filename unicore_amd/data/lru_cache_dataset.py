@@ -1,4 +1,9 @@
-"""LRU-cache wrapper (parity: reference unicore/data/lru_cache_dataset.py:12)."""
+"""Small LRU cache over item fetch + collation (parity: reference
+unicore/data/lru_cache_dataset.py:12).
+
+Used around MaskTokensDataset pairs so the shared base item is loaded once
+per (source, target) access.
+"""
 
 from functools import lru_cache
 
@@ -7,12 +12,12 @@ from .base_wrapper_dataset import BaseWrapperDataset
 
 class LRUCacheDataset(BaseWrapperDataset):
     def __init__(self, dataset, token=None):
-        super().__init__(dataset)
+        super().__init__(dataset=dataset)
 
-    @lru_cache(maxsize=16)
+    @lru_cache(16)
     def __getitem__(self, index):
-        return self.dataset[index]
+        return super().__getitem__(index)
 
-    @lru_cache(maxsize=16)
+    @lru_cache(16)
     def collater(self, samples):
-        return self.dataset.collater(samples)
+        return super().collater(samples)

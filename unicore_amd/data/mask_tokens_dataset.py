@@ -1,6 +1,11 @@
-"""BERT-style 80/10/10 token masking (parity: reference
-unicore/data/mask_tokens_dataset.py:16-132). Epoch+index-seeded so results are
-deterministic and change per epoch.
+"""BERT-style 80/10/10 masking wrapper (parity: reference
+unicore/data/mask_tokens_dataset.py:16-132).
+
+Noise is drawn under ``numpy_seed(seed, epoch, index)``, which makes every
+item deterministic per (epoch, index) AND keeps the source/target dataset
+pair (two instances over one shared base) in agreement on the mask
+positions: both consume the identical RNG stream up to the point where the
+target variant returns.
 """
 
 from functools import lru_cache
@@ -8,143 +13,123 @@ from functools import lru_cache
 import numpy as np
 import torch
 
-from . import data_utils
+from .data_utils import numpy_seed
 from .base_wrapper_dataset import BaseWrapperDataset
 from .dictionary import Dictionary
 from .lru_cache_dataset import LRUCacheDataset
 
 
 class MaskTokensDataset(BaseWrapperDataset):
-    """
-    A wrapper Dataset for masked language modeling.
+    """Masked-LM noising wrapper.
 
-    Input items are masked according to the specified masking probability;
-    returns (masked tokens, original tokens) pairs via the ``apply_mask``
-    classmethod which shares a single underlying dataset.
+    One instance yields the corrupted inputs, a twin instance
+    (``return_masked_tokens=True``) yields the targets: original tokens at
+    masked positions, pad everywhere else. Build both through
+    :meth:`apply_mask` so they share one cached base dataset.
     """
 
     @classmethod
     def apply_mask(cls, dataset: torch.utils.data.Dataset, *args, **kwargs):
-        """Return the source and target datasets for masked LM training."""
-        dataset = LRUCacheDataset(dataset)
-        return (
-            LRUCacheDataset(cls(dataset, *args, **kwargs, return_masked_tokens=False)),
-            LRUCacheDataset(cls(dataset, *args, **kwargs, return_masked_tokens=True)),
+        """Return the (source, target) dataset pair for masked-LM training."""
+        shared = LRUCacheDataset(dataset)
+        make = lambda as_target: LRUCacheDataset(
+            cls(shared, *args, **kwargs, return_masked_tokens=as_target)
         )
+        return make(False), make(True)
 
     def __init__(
         self,
-        dataset: torch.utils.data.Dataset,
-        vocab: Dictionary,
-        pad_idx: int,
-        mask_idx: int,
-        return_masked_tokens: bool = False,
-        seed: int = 1,
-        mask_prob: float = 0.15,
-        leave_unmasked_prob: float = 0.1,
-        random_token_prob: float = 0.1,
+        dataset,
+        vocab,
+        pad_idx,
+        mask_idx,
+        return_masked_tokens=False,
+        seed=1,
+        mask_prob=0.15,
+        leave_unmasked_prob=0.1,
+        random_token_prob=0.1,
     ):
-        assert 0.0 < mask_prob < 1.0
-        assert 0.0 <= random_token_prob <= 1.0
-        assert 0.0 <= leave_unmasked_prob <= 1.0
-        assert random_token_prob + leave_unmasked_prob <= 1.0
+        assert 0.0 < mask_prob < 1.0, "mask_prob must be in (0, 1)"
+        assert 0.0 <= random_token_prob <= 1.0 >= leave_unmasked_prob >= 0.0
+        assert random_token_prob + leave_unmasked_prob <= 1.0, \
+            "the random/unmasked splits cannot exceed the whole mask"
 
-        self.dataset = dataset
-        self.vocab = vocab
-        self.pad_idx = pad_idx
-        self.mask_idx = mask_idx
-        self.return_masked_tokens = return_masked_tokens
-        self.seed = seed
-        self.mask_prob = mask_prob
-        self.leave_unmasked_prob = leave_unmasked_prob
-        self.random_token_prob = random_token_prob
+        self.dataset, self.vocab, self.seed = dataset, vocab, seed
+        self.pad_idx, self.mask_idx = pad_idx, mask_idx
+        self.return_masked_tokens = bool(return_masked_tokens)
+        self.mask_prob = float(mask_prob)
+        self.leave_unmasked_prob = float(leave_unmasked_prob)
+        self.random_token_prob = float(random_token_prob)
+        self.epoch = None
 
         if random_token_prob > 0.0:
-            weights = np.ones(len(self.vocab))
-            weights[self.vocab.special_index()] = 0
-            self.weights = weights / weights.sum()
-
-        self.epoch = None
+            # uniform over the non-special vocabulary
+            w = np.ones(len(vocab))
+            w[vocab.special_index()] = 0
+            self.weights = w / w.sum()
 
     @property
     def can_reuse_epoch_itr_across_epochs(self):
-        return True  # only the noise changes, not item sizes
+        # only the noise changes across epochs, never the item sizes
+        return True
 
     def set_epoch(self, epoch, **unused):
         super().set_epoch(epoch)
-        self.epoch = epoch
+        self.epoch = epoch  # keyed into the per-item RNG seed
 
-    def __getitem__(self, index: int):
+    def __getitem__(self, index):
         return self.__getitem_cached__(self.epoch, index)
 
-    @lru_cache(maxsize=16)
-    def __getitem_cached__(self, epoch: int, index: int):
-        with data_utils.numpy_seed(self.seed, epoch, index):
-            item = self.dataset[index]
-            sz = len(item)
-
-            assert (
-                self.mask_idx not in item
-            ), "Dataset contains mask_idx (={}), this is not expected!".format(
-                self.mask_idx,
+    @lru_cache(16)
+    def __getitem_cached__(self, epoch, index):
+        with numpy_seed(self.seed, epoch, index):
+            tokens = self.dataset[index]
+            sz = tokens.shape[0] if hasattr(tokens, 'shape') else len(tokens)
+            assert self.mask_idx not in tokens, (
+                f"Dataset contains mask_idx (={self.mask_idx}), "
+                "this is not expected!"
             )
 
-            # decide elements to mask
-            mask = np.full(sz, False)
-            num_mask = int(
-                # add a random number for probabilistic rounding
-                self.mask_prob * sz
-                + np.random.rand()
+            # choose the mask; np.random.rand() makes E[num_mask] exact
+            # despite int truncation
+            picked = np.random.choice(
+                sz, int(self.mask_prob * sz + np.random.rand()), replace=False
             )
-            mask_idc = np.random.choice(sz, num_mask, replace=False)
-            mask_idc = mask_idc[mask_idc < len(mask)]
-            try:
-                mask[mask_idc] = True
-            except Exception:  # pragma: no cover - debugging aid
-                print(
-                    "Assigning mask indexes {} to mask {} failed!".format(
-                        mask_idc, mask
-                    )
-                )
-                raise
+            mask = np.zeros(sz, dtype=bool)
+            mask[picked[picked < sz]] = True
 
             if self.return_masked_tokens:
-                # exit early if we're just returning the masked tokens
-                # (i.e., the targets)
-                new_item = np.full(len(mask), self.pad_idx)
-                new_item[mask] = item[torch.from_numpy(mask.astype(np.uint8)) == 1]
-                return torch.from_numpy(new_item)
+                # target variant: originals at masked slots, pad elsewhere.
+                # Returns here so both variants consumed the same RNG calls.
+                target = np.full(sz, self.pad_idx)
+                sel = torch.from_numpy(mask.astype(np.uint8)) == 1
+                target[mask] = tokens[sel]
+                return torch.from_numpy(target)
 
-            # decide unmasking and random replacement
-            rand_or_unmask_prob = self.random_token_prob + self.leave_unmasked_prob
-            if rand_or_unmask_prob > 0.0:
-                rand_or_unmask = mask & (np.random.rand(sz) < rand_or_unmask_prob)
+            # split the masked positions into keep-original / random-token
+            # subsets per the 80/10/10 recipe
+            unmask, rand_mask = None, None
+            either_prob = self.random_token_prob + self.leave_unmasked_prob
+            if either_prob > 0.0:
+                either = mask & (np.random.rand(sz) < either_prob)
                 if self.random_token_prob == 0.0:
-                    unmask = rand_or_unmask
-                    rand_mask = None
+                    unmask = either
                 elif self.leave_unmasked_prob == 0.0:
-                    unmask = None
-                    rand_mask = rand_or_unmask
+                    rand_mask = either
                 else:
-                    unmask_prob = self.leave_unmasked_prob / rand_or_unmask_prob
-                    decision = np.random.rand(sz) < unmask_prob
-                    unmask = rand_or_unmask & decision
-                    rand_mask = rand_or_unmask & (~decision)
-            else:
-                unmask = rand_mask = None
+                    keep_original = np.random.rand(sz) < (
+                        self.leave_unmasked_prob / either_prob
+                    )
+                    unmask = either & keep_original
+                    rand_mask = either & ~keep_original
 
             if unmask is not None:
-                mask = mask ^ unmask
+                mask ^= unmask
 
-            new_item = np.copy(item)
-            new_item[mask] = self.mask_idx
-            if rand_mask is not None:
-                num_rand = rand_mask.sum()
-                if num_rand > 0:
-                    new_item[rand_mask] = np.random.choice(
-                        len(self.vocab),
-                        num_rand,
-                        p=self.weights,
-                    )
-
-            return torch.from_numpy(new_item)
+            corrupted = np.copy(tokens)
+            corrupted[mask] = self.mask_idx
+            if rand_mask is not None and rand_mask.sum() > 0:
+                corrupted[rand_mask] = np.random.choice(
+                    len(self.vocab), rand_mask.sum(), p=self.weights
+                )
+            return torch.from_numpy(corrupted)

@@ -1,5 +1,9 @@
-"""Nested-dict dataset: flatten/unflatten dotted keys (parity: reference
-unicore/data/nested_dictionary_dataset.py:48-111)."""
+"""Composite dataset over a nested dict of datasets (parity: reference
+unicore/data/nested_dictionary_dataset.py:48-111).
+
+The nested definition is flattened to dotted keys ("net_input.src_tokens");
+items and collated batches are produced flat, then re-nested on the way out.
+"""
 
 from collections import OrderedDict
 
@@ -9,37 +13,35 @@ from torch.utils.data.dataloader import default_collate
 from .unicore_dataset import UnicoreDataset
 
 
-def _flatten(dico, prefix=None):
-    """Flatten a nested dictionary."""
-    new_dico = OrderedDict()
-    if isinstance(dico, dict):
-        prefix = prefix + "." if prefix is not None else ""
-        for k, v in dico.items():
-            if v is None:
+def _flatten(tree, prefix=None):
+    """Depth-first flatten of nested dict/list structure into dotted keys."""
+    flat = OrderedDict()
+    if isinstance(tree, dict):
+        head = "" if prefix is None else prefix + "."
+        for key, sub in tree.items():
+            if sub is None:
                 continue
-            new_dico.update(_flatten(v, prefix + k))
-    elif isinstance(dico, list):
-        for i, v in enumerate(dico):
-            new_dico.update(_flatten(v, prefix + ".[" + str(i) + "]"))
+            flat.update(_flatten(sub, head + key))
+    elif isinstance(tree, list):
+        for pos, sub in enumerate(tree):
+            flat.update(_flatten(sub, f"{prefix}.[{pos}]"))
     else:
-        new_dico = OrderedDict({prefix: dico})
-    return new_dico
+        flat[prefix] = tree
+    return flat
 
 
-def _unflatten(dico):
-    """Unflatten a flattened dictionary into a nested dictionary."""
-    new_dico = OrderedDict()
-    for full_k, v in dico.items():
-        full_k = full_k.split(".")
-        node = new_dico
-        for k in full_k[:-1]:
-            if k.startswith("[") and k.endswith("]"):
-                k = int(k[1:-1])
-            if k not in node:
-                node[k] = OrderedDict()
-            node = node[k]
-        node[full_k[-1]] = v
-    return new_dico
+def _unflatten(flat):
+    """Rebuild the nested structure from dotted keys."""
+    tree = OrderedDict()
+    for dotted, value in flat.items():
+        *path, leaf = dotted.split(".")
+        node = tree
+        for part in path:
+            if part[:1] == "[" and part[-1:] == "]":
+                part = int(part[1:-1])
+            node = node.setdefault(part, OrderedDict())
+        node[leaf] = value
+    return tree
 
 
 class NestedDictionaryDataset(UnicoreDataset):
@@ -48,19 +50,12 @@ class NestedDictionaryDataset(UnicoreDataset):
         self.defn = _flatten(defn)
 
         first = None
-        for v in self.defn.values():
-            if not isinstance(
-                v,
-                (
-                    UnicoreDataset,
-                    torch.utils.data.Dataset,
-                ),
-            ):
-                raise ValueError("Expected Dataset but found: {}".format(v.__class__))
-            first = first or v
-            if len(v) > 0:
-                assert len(v) == len(first), "dataset lengths must match"
-
+        for leaf in self.defn.values():
+            if not isinstance(leaf, (UnicoreDataset, torch.utils.data.Dataset)):
+                raise ValueError(f"Expected Dataset but found: {leaf.__class__}")
+            first = first or leaf
+            if len(leaf) > 0:
+                assert len(leaf) == len(first), "dataset lengths must match"
         self._len = len(first)
 
     def __getitem__(self, index):
@@ -70,31 +65,33 @@ class NestedDictionaryDataset(UnicoreDataset):
         return self._len
 
     def collater(self, samples):
-        """Merge a list of samples to form a mini-batch."""
+        """Collate each leaf with its own collater (default_collate where a
+        leaf has none), then re-nest."""
         if len(samples) == 0:
             return {}
-        sample = OrderedDict()
-        for k, ds in self.defn.items():
+        batch = OrderedDict()
+        for key, ds in self.defn.items():
+            column = [s[key] for s in samples]
             try:
-                sample[k] = ds.collater([s[k] for s in samples])
+                batch[key] = ds.collater(column)
             except NotImplementedError:
-                sample[k] = default_collate([s[k] for s in samples])
-        return _unflatten(sample)
+                batch[key] = default_collate(column)
+        return _unflatten(batch)
 
     @property
     def supports_prefetch(self):
-        """Whether this dataset supports prefetching."""
         return any(ds.supports_prefetch for ds in self.defn.values())
 
     def prefetch(self, indices):
-        """Prefetch the data required for this epoch."""
         for ds in self.defn.values():
             if getattr(ds, "supports_prefetch", False):
                 ds.prefetch(indices)
 
     @property
     def can_reuse_epoch_itr_across_epochs(self):
-        return all(ds.can_reuse_epoch_itr_across_epochs for ds in self.defn.values())
+        return all(
+            ds.can_reuse_epoch_itr_across_epochs for ds in self.defn.values()
+        )
 
     def set_epoch(self, epoch):
         super().set_epoch(epoch)

@@ -1,40 +1,43 @@
-"""Sorting / epoch-shuffling wrappers (parity: reference
+"""Index-ordering wrappers (parity: reference
 unicore/data/sort_dataset.py:12-42)."""
 
 import numpy as np
 
-from . import data_utils
 from .base_wrapper_dataset import BaseWrapperDataset
+from .data_utils import numpy_seed
 
 
 class SortDataset(BaseWrapperDataset):
-    def __init__(self, dataset, sort_order):
-        super().__init__(dataset)
-        if not isinstance(sort_order, (list, tuple)):
-            sort_order = [sort_order]
-        self.sort_order = sort_order
+    """Orders indices by one or more sort keys (last key is primary,
+    np.lexsort semantics) — used to group similar-length items."""
 
-        assert all(len(so) == len(dataset) for so in sort_order)
+    def __init__(self, dataset, sort_order):
+        super().__init__(dataset=dataset)
+        self.sort_keys = (
+            sort_order if isinstance(sort_order, (list, tuple)) else [sort_order]
+        )
+        assert all(len(key) == len(dataset) for key in self.sort_keys)
 
     def ordered_indices(self):
-        return np.lexsort(self.sort_order)
+        return np.lexsort(tuple(self.sort_keys))
 
 
 class EpochShuffleDataset(BaseWrapperDataset):
+    """Deterministic fresh permutation per epoch (seeded by seed+epoch)."""
+
     def __init__(self, dataset, size, seed):
-        super().__init__(dataset)
-        self.size = size
-        self.seed = seed
+        super().__init__(dataset=dataset)
+        self.size, self.seed = size, seed
         self.set_epoch(1)
 
-    def set_epoch(self, epoch):
+    def set_epoch(self, epoch, **unused):
         super().set_epoch(epoch)
-        with data_utils.numpy_seed(self.seed + epoch - 1):
-            self.sort_order = np.random.permutation(self.size)
+        with numpy_seed(self.seed + epoch - 1):
+            self.permutation = np.random.permutation(self.size)
 
     def ordered_indices(self):
-        return self.sort_order
+        return self.permutation
 
     @property
     def can_reuse_epoch_itr_across_epochs(self):
-        return False
+        return False  # the permutation is epoch-dependent

@@ -7,71 +7,53 @@ from . import data_utils
 
 
 class EpochListening:
-    """Mixin for receiving updates whenever the epoch increments."""
+    """Mixin: datasets that want to hear about epoch transitions."""
 
     @property
     def can_reuse_epoch_itr_across_epochs(self):
-        """Whether we can reuse the epoch iterator across epochs.
-
-        This needs to return ``False`` if the sample sizes can change across
-        epochs, in which case we may need to regenerate batches at each epoch.
-        If your dataset relies in ``set_epoch`` then you should consider setting
-        this to ``False``.
-        """
+        """False when item SIZES can change between epochs (forces batch
+        regeneration each epoch); datasets that depend on ``set_epoch`` for
+        anything size-affecting should return False."""
         return True
 
     def set_epoch(self, epoch):
-        """Will receive the updated epoch number at the beginning of the epoch."""
-        pass
+        """Called with the new (1-based) epoch number before it starts."""
 
 
 class UnicoreDataset(torch.utils.data.Dataset, EpochListening):
-    """A dataset that provides helpers for batching."""
+    """torch Dataset + the batching/collation protocol the task layer uses."""
 
     def __getitem__(self, index):
-        raise NotImplementedError
+        raise NotImplementedError("datasets implement item access")
 
-    def __len__(self):
-        raise NotImplementedError
+    def __len__(self) -> int:
+        raise NotImplementedError("datasets implement len()")
 
     def collater(self, samples):
-        """Merge a list of samples to form a mini-batch.
-
-        Args:
-            samples (List[dict]): samples to collate
-
-        Returns:
-            dict: a mini-batch suitable for forwarding with `Model`
-        """
-        raise NotImplementedError
+        """Merge a list of samples into a model-ready mini-batch dict."""
+        raise NotImplementedError("datasets implement collation")
 
     def ordered_indices(self):
-        """Return an ordered list of indices. Batches will be constructed based
-        on this order."""
+        """Index order used for batch construction (identity by default)."""
         return np.arange(len(self), dtype=np.int64)
 
     @property
     def supports_prefetch(self):
-        """Whether this dataset supports prefetching."""
+        """True when :meth:`prefetch` is usable."""
         return False
 
-    def attr(self, attr: str, index: int):
-        return getattr(self, attr, None)
+    def attr(self, name, index):
+        """Per-index attribute lookup hook (class attribute by default)."""
+        return getattr(self, name, None)
 
     def prefetch(self, indices):
-        """Prefetch the data required for this epoch."""
-        raise NotImplementedError
+        """Bulk-load the given indices ahead of the epoch."""
+        raise NotImplementedError("supports_prefetch datasets implement this")
 
-    def batch_by_size(
-        self,
-        indices,
-        batch_size=None,
-        required_batch_size_multiple=1,
-    ):
-        """
-        Given an ordered set of indices, return batches according to
-        *batch_size* and *required_batch_size_multiple*.
-        """
+    def batch_by_size(self, indices, batch_size=None,
+                      required_batch_size_multiple=1):
+        """Partition *indices* into batches of *batch_size* (rounded to the
+        required multiple); see data_utils.batch_by_size."""
         return data_utils.batch_by_size(
             indices,
             batch_size=batch_size,

@@ -153,6 +153,9 @@ class UnicoreTask(object):
             num_workers=num_workers,
             epoch=epoch,
             buffer_size=data_buffer_size,
+            # a --cpu run on a CUDA-capable host must not stage batches onto
+            # the GPU (the model stays on CPU)
+            cuda_prefetch=not getattr(self.args, "cpu", False),
         )
 
         if can_reuse_epoch_itr:
