@@ -14,38 +14,24 @@ class Adadelta(UnicoreOptimizer):
 
     @classmethod
     def add_args(cls, parser):
-        """Add optimizer-specific arguments to the parser."""
-        parser.add_argument(
-            "--adadelta-rho",
-            type=float,
-            default=0.9,
-            metavar="RHO",
-            help="coefficient used for computing a running average of squared gradients",
-        )
-        parser.add_argument(
-            "--adadelta-eps",
-            type=float,
-            default=1e-6,
-            metavar="EPS",
-            help="term added to the denominator to improve numerical stability",
-        )
-        parser.add_argument(
-            "--weight-decay",
-            "--wd",
-            default=0.0,
-            type=float,
-            metavar="WD",
-            help="weight decay",
-        )
+        parser.add_argument("--adadelta-rho", type=float, default=0.9,
+                            metavar="RHO",
+                            help="running-average coefficient for squared grads")
+        parser.add_argument("--adadelta-eps", type=float, default=1e-6,
+                            metavar="EPS",
+                            help="denominator stability epsilon")
+        parser.add_argument("--weight-decay", "--wd", default=0.0, type=float,
+                            metavar="WD", help="weight decay")
 
     @property
     def optimizer_config(self):
-        return {
-            "lr": self.args.lr[0] if isinstance(self.args.lr, list) else self.args.lr,
-            "rho": self.args.adadelta_rho,
-            "eps": self.args.adadelta_eps,
-            "weight_decay": self.args.weight_decay,
-        }
+        lr = self.args.lr
+        return dict(
+            lr=lr[0] if isinstance(lr, list) else lr,
+            rho=self.args.adadelta_rho,
+            eps=self.args.adadelta_eps,
+            weight_decay=self.args.weight_decay,
+        )
 
     @property
     def supports_flat_params(self):

@@ -1,12 +1,14 @@
-"""Dynamic loss scaler (parity: reference unicore/optim/dynamic_loss_scaler.py:8-71).
+"""Dynamic loss scaling for fp16 training (parity: reference
+unicore/optim/dynamic_loss_scaler.py:8-71).
 
-Semantics preserved exactly: x2 growth every *scale_window* non-overflow
-updates, /2 backoff on overflow with a tolerance percentage, and a
-FloatingPointError abort when the scale would fall below *threshold*.
+Semantics preserved exactly: x ``scale_factor`` growth every
+``scale_window`` clean updates, backoff on overflow subject to a tolerated
+overflow percentage, and a FloatingPointError abort once the scale cannot
+drop further (``min_loss_scale``).
 """
 
 
-class DynamicLossScaler(object):
+class DynamicLossScaler:
     def __init__(
         self,
         init_scale=2.0**15,
@@ -17,56 +19,57 @@ class DynamicLossScaler(object):
         min_loss_scale=1e-4,
     ):
         self.loss_scale = init_scale
-        self.scale_factor = scale_factor
-        self.scale_window = scale_window
-        self.tolerance = tolerance
-        self.threshold = threshold
-        self._iter = 0
-        self._last_overflow_iter = -1
-        self._last_rescale_iter = -1
-        self._overflows_since_rescale = 0
+        self.scale_factor, self.scale_window = scale_factor, scale_window
+        self.tolerance, self.threshold = tolerance, threshold
         self.min_loss_scale = min_loss_scale
+        # update counter plus the positions of the last overflow / rescale,
+        # used to rate the overflow percentage inside a window
+        self._updates = 0
+        self._overflow_at = -1
+        self._rescale_at = -1
+        self._overflow_count = 0
 
     def scale(self, outputs):
-        return self.loss_scale * outputs
+        """Multiply the loss (or anything) by the live scale."""
+        return outputs * self.loss_scale
 
-    def update(self):
-        if (self._iter - self._last_overflow_iter) % self.scale_window == 0:
-            self.loss_scale *= self.scale_factor
-            self._last_rescale_iter = self._iter
-        self._iter += 1
+    def update(self) -> None:
+        """Called after every applied update; grows the scale after a full
+        clean window."""
+        if (self._updates - self._overflow_at) % self.scale_window == 0:
+            self.loss_scale = self.loss_scale * self.scale_factor
+            self._rescale_at = self._updates
+        self._updates += 1
 
-    def _decrease_loss_scale(self):
-        self.loss_scale /= self.scale_factor
-        if self.threshold is not None:
-            self.loss_scale = max(self.loss_scale, self.threshold)
+    def _backoff(self) -> None:
+        floor = self.threshold
+        self.loss_scale = self.loss_scale / self.scale_factor
+        if floor is not None:
+            self.loss_scale = max(self.loss_scale, floor)
 
-    def check_overflow(self, grad_norm):
-        # detect inf and nan
-        if grad_norm == float("inf") or grad_norm != grad_norm:
-            # overflow has occurred
-            prev_scale = self.loss_scale
-            iter_since_rescale = self._iter - self._last_rescale_iter
+    def check_overflow(self, grad_norm) -> None:
+        """Raise OverflowError (skip this update) when the grad norm is
+        inf/nan; FloatingPointError when the scale has already bottomed out."""
+        if grad_norm != float("inf") and grad_norm == grad_norm:
+            return  # finite: nothing to do
+        before = self.loss_scale
+        window = self._updates - self._rescale_at
 
-            self._last_overflow_iter = self._iter
-            self._overflows_since_rescale += 1
-            pct_overflow = self._overflows_since_rescale / float(iter_since_rescale)
-            if pct_overflow >= self.tolerance:
-                self._decrease_loss_scale()
-                self._last_rescale_iter = self._iter
-                self._overflows_since_rescale = 0
+        self._overflow_at = self._updates
+        self._overflow_count += 1
+        if self._overflow_count / float(window) >= self.tolerance:
+            self._backoff()
+            self._rescale_at = self._updates
+            self._overflow_count = 0
 
-            if self.loss_scale <= self.min_loss_scale:
-                # Use FloatingPointError as an uncommon error that parent
-                # functions can safely catch to stop training.
-                self.loss_scale = prev_scale
-                raise FloatingPointError(
-                    (
-                        "Minimum loss scale reached ({}). Your loss is probably exploding. "
-                        "Try lowering the learning rate, using gradient clipping or "
-                        "increasing the batch size."
-                    ).format(self.min_loss_scale)
-                )
+        if self.min_loss_scale >= self.loss_scale:
+            # an uncommon error type the trainer can catch to stop cleanly
+            self.loss_scale = before
+            raise FloatingPointError(
+                f"Minimum loss scale reached ({self.min_loss_scale}). Your "
+                "loss is probably exploding. Try lowering the learning "
+                "rate, using gradient clipping or increasing the batch size."
+            )
 
-            self._iter += 1
-            raise OverflowError("setting loss scale to: " + str(self.loss_scale))
+        self._updates += 1
+        raise OverflowError(f"setting loss scale to: {self.loss_scale}")

@@ -20,26 +20,14 @@ class FusedAdam(torch.optim.Optimizer):
     sweep over memory.
     """
 
-    def __init__(
-        self,
-        params,
-        lr=1e-3,
-        bias_correction=True,
-        betas=(0.9, 0.999),
-        eps=1e-8,
-        weight_decay=0.0,
-        amsgrad=False,
-    ):
+    def __init__(self, params, lr=1e-3, bias_correction=True,
+                 betas=(0.9, 0.999), eps=1e-8, weight_decay=0.0,
+                 amsgrad=False):
         if amsgrad:
             raise RuntimeError("FusedAdam does not support the AMSGrad variant.")
-        defaults = {
-            "lr": lr,
-            "bias_correction": bias_correction,
-            "betas": betas,
-            "eps": eps,
-            "weight_decay": weight_decay,
-        }
-        super().__init__(params, defaults)
+        super().__init__(params, dict(lr=lr, bias_correction=bias_correction,
+                                      betas=betas, eps=eps,
+                                      weight_decay=weight_decay))
 
     @property
     def supports_memory_efficient_fp16(self):
@@ -54,54 +42,31 @@ class FusedAdam(torch.optim.Optimizer):
         return True
 
     def step(self, closure=None, scale=1.0):
-        """Performs a single optimization step.
-
-        Args:
-            closure (callable, optional): A closure that reevaluates the model
-                and returns the loss.
-            scale (float, optional): factor to divide gradient tensor values
-                by before applying to weights (fused into the kernel).
-        """
+        """One optimization step; ``scale`` divides the gradients inside
+        the kernel (no separate unscale sweep over memory)."""
         from unicore_amd import ops
 
-        loss = None
-        if closure is not None:
-            loss = closure()
-
+        loss = closure() if closure is not None else None
         for group in self.param_groups:
             bias_correction = group.get("bias_correction", True)
-
+            beta1, beta2 = group["betas"]
             for p in group["params"]:
                 if p.grad is None:
                     continue
-                grad = p.grad.data
-                if grad.is_sparse:
+                if p.grad.data.is_sparse:
                     raise RuntimeError("FusedAdam does not support sparse gradients")
-
                 state = self.state[p]
                 if len(state) == 0:
-                    state["step"] = 0
-                    # moments in fp32 regardless of param dtype
-                    state["exp_avg"] = torch.zeros_like(p.data, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(
-                        p.data, dtype=torch.float32
+                    # moments live in fp32 regardless of param dtype
+                    state = self.state[p] = dict(
+                        step=0,
+                        exp_avg=torch.zeros_like(p.data, dtype=torch.float32),
+                        exp_avg_sq=torch.zeros_like(p.data, dtype=torch.float32),
                     )
-
                 state["step"] += 1
-                beta1, beta2 = group["betas"]
                 ops.fused_adam(
-                    p.data,
-                    state["exp_avg"],
-                    state["exp_avg_sq"],
-                    grad,
-                    group["lr"],
-                    beta1,
-                    beta2,
-                    group["eps"],
-                    scale,
-                    state["step"],
-                    bias_correction,
-                    group["weight_decay"],
+                    p.data, state["exp_avg"], state["exp_avg_sq"], p.grad.data,
+                    group["lr"], beta1, beta2, group["eps"], scale,
+                    state["step"], bias_correction, group["weight_decay"],
                 )
-
         return loss

@@ -14,26 +14,19 @@ class SGD(UnicoreOptimizer):
 
     @classmethod
     def add_args(cls, parser):
-        """Add optimizer-specific arguments to the parser."""
-        parser.add_argument(
-            "--momentum", default=0.0, type=float, metavar="M", help="momentum factor"
-        )
-        parser.add_argument(
-            "--weight-decay",
-            "--wd",
-            default=0.0,
-            type=float,
-            metavar="WD",
-            help="weight decay",
-        )
+        parser.add_argument("--momentum", default=0.0, type=float, metavar="M",
+                            help="momentum factor")
+        parser.add_argument("--weight-decay", "--wd", default=0.0, type=float,
+                            metavar="WD", help="weight decay")
 
     @property
     def optimizer_config(self):
-        return {
-            "lr": self.args.lr[0] if isinstance(self.args.lr, list) else self.args.lr,
-            "momentum": self.args.momentum,
-            "weight_decay": self.args.weight_decay,
-        }
+        lr = self.args.lr
+        return dict(
+            lr=lr[0] if isinstance(lr, list) else lr,
+            momentum=self.args.momentum,
+            weight_decay=self.args.weight_decay,
+        )
 
     @property
     def supports_flat_params(self):

@@ -1,54 +1,56 @@
-"""Optimizer base class (parity: reference unicore/optim/unicore_optimizer.py:10-189)."""
+"""Optimizer wrapper base (parity: reference
+unicore/optim/unicore_optimizer.py:10-189).
+
+Wraps a ``torch.optim.Optimizer`` and adds the trainer-facing surface:
+grad scaling/clipping, the all-reduce hook, capability flags (queried off
+the inner optimizer through one ``_inner_flag`` helper here), and
+state-dict handling with override support.
+"""
 
 import torch
 
 from unicore_amd import utils
 
 
-class UnicoreOptimizer(object):
+class UnicoreOptimizer:
     def __init__(self, args):
         super().__init__()
         self.args = args
 
     @classmethod
     def add_args(cls, parser):
-        """Add optimizer-specific arguments to the parser."""
-        pass
+        """Hook for optimizer-specific CLI arguments."""
 
     @property
     def optimizer(self):
-        """Return a torch.optim.optimizer.Optimizer instance."""
-        if not hasattr(self, "_optimizer"):
-            raise NotImplementedError
-        if not isinstance(self._optimizer, torch.optim.Optimizer):
-            raise ValueError("_optimizer must be an instance of torch.optim.Optimizer")
+        """The wrapped torch.optim.Optimizer."""
+        self._check_inner()
         return self._optimizer
 
     @optimizer.setter
     def optimizer(self, optimizer):
-        """Reset optimizer instance."""
-        if not hasattr(self, "_optimizer"):
-            raise NotImplementedError
-        if not isinstance(self._optimizer, torch.optim.Optimizer):
-            raise ValueError("_optimizer must be an instance of torch.optim.Optimizer")
+        """Swap the wrapped optimizer instance."""
+        self._check_inner()
         self._optimizer = optimizer
+
+    def _check_inner(self):
+        inner = getattr(self, "_optimizer", None)
+        if inner is None:
+            raise NotImplementedError("subclass must create self._optimizer")
+        if not isinstance(inner, torch.optim.Optimizer):
+            raise ValueError("_optimizer must wrap a torch.optim.Optimizer")
 
     @property
     def optimizer_config(self):
-        """
-        Return a kwarg dictionary that will be used to override optimizer
-        args stored in checkpoints. This allows us to load a checkpoint and
-        resume training using a different set of optimizer args, e.g., with a
-        different learning rate.
-        """
-        raise NotImplementedError
+        """kwargs that override optimizer state restored from checkpoints
+        (so resumes can pick up e.g. a new learning rate)."""
+        raise NotImplementedError("subclass must define optimizer_config")
 
     @property
     def params(self):
-        """Return an iterable of the parameters held by the optimizer."""
-        for param_group in self.param_groups:
-            for p in param_group["params"]:
-                yield p
+        """All parameters across all parameter groups."""
+        for group in self.param_groups:
+            yield from group["params"]
 
     @property
     def param_groups(self):
@@ -58,54 +60,45 @@ class UnicoreOptimizer(object):
         return self._optimizer.__getstate__()
 
     def get_lr(self):
-        """Return the current learning rate."""
+        """Learning rate of the first parameter group."""
         return self.param_groups[0]["lr"]
 
-    def set_lr(self, lr):
-        """Set the learning rate."""
-        for param_group in self.param_groups:
-            param_group["lr"] = lr
+    def set_lr(self, lr) -> None:
+        """Apply one learning rate to every parameter group."""
+        for group in self.param_groups:
+            group["lr"] = lr
 
     def state_dict(self):
-        """Return the optimizer's state dict."""
         return self.optimizer.state_dict()
 
-    def load_state_dict(self, state_dict, optimizer_overrides=None):
-        """Load an optimizer state dict.
-
-        In general we should prefer the configuration of the existing optimizer
-        instance (e.g., learning rate) over that found in the state_dict. This
-        allows us to resume training from a checkpoint using a new set of
-        optimizer args.
-        """
+    def load_state_dict(self, state_dict, optimizer_overrides=None) -> None:
+        """Restore optimizer state; *optimizer_overrides* (e.g. the lr from
+        the current run's config) wins over the checkpointed values."""
         self.optimizer.load_state_dict(state_dict)
-
-        if optimizer_overrides is not None and len(optimizer_overrides) > 0:
-            # override learning rate, momentum, etc. with latest values
+        if optimizer_overrides:
             for group in self.param_groups:
                 group.update(optimizer_overrides)
 
-    def backward(self, loss):
-        """Computes the sum of gradients of the given tensor w.r.t. graph leaves."""
+    def backward(self, loss) -> None:
+        """Backprop entry point (the fp16 wrapper scales here)."""
         loss.backward()
 
-    def all_reduce_grads(self, module):
-        """Manually all-reduce gradients (if required)."""
-        if hasattr(module, "all_reduce_grads"):
-            module.all_reduce_grads()
+    def all_reduce_grads(self, module) -> None:
+        """Trigger the module's manual grad all-reduce, when it has one."""
+        hook = getattr(module, "all_reduce_grads", None)
+        if hook is not None:
+            hook()
 
-    def multiply_grads(self, c):
-        """Multiplies grads by a constant *c*."""
-        grads = [
-            p.grad.data for p in self.params if p.grad is not None
-        ]
-        if len(grads) > 0:
-            torch._foreach_mul_(
-                grads, c.to(grads[0].device) if torch.is_tensor(c) else c
-            )
+    def multiply_grads(self, c) -> None:
+        """Scale all existing gradients by *c* in a single foreach launch."""
+        grads = [p.grad.data for p in self.params if p.grad is not None]
+        if grads:
+            if torch.is_tensor(c):
+                c = c.to(grads[0].device)
+            torch._foreach_mul_(grads, c)
 
     def clip_grad_norm(self, max_norm, aggregate_norm_fn=None):
-        """Clips gradient norm."""
+        """Global L2-norm gradient clipping (multi-tensor kernel path)."""
         return utils.clip_grad_norm_(self.params, max_norm, aggregate_norm_fn)
 
     def per_sample_clip_grad_norm(self, max_norm, aggregate_norm_fn=None):
@@ -114,61 +107,51 @@ class UnicoreOptimizer(object):
             "(--fp16 or --bf16 with --per-sample-clip-norm)"
         )
 
-    def step(self, closure=None, scale=1.0, groups=None):
-        """Performs a single optimization step."""
+    def step(self, closure=None, scale=1.0, groups=None) -> None:
+        """One optimization step; routes *scale* into the kernel when the
+        inner optimizer fuses unscaling, otherwise pre-divides the grads."""
         if self.supports_step_with_scale:
             if self.supports_groups:
                 self.optimizer.step(closure, scale=scale, groups=groups)
             else:
                 self.optimizer.step(closure, scale=scale)
+            return
+        if scale != 1.0:
+            self.multiply_grads(1.0 / scale)
+        if self.supports_groups:
+            self.optimizer.step(closure, groups=groups)
         else:
-            if scale != 1.0:
-                self.multiply_grads(1.0 / scale)
-            if self.supports_groups:
-                self.optimizer.step(closure, groups=groups)
-            else:
-                self.optimizer.step(closure)
+            self.optimizer.step(closure)
 
-    def zero_grad(self):
-        """Clears the gradients of all optimized parameters."""
-        for p in self.params:
-            p.grad = None
+    def zero_grad(self) -> None:
+        for param in self.params:
+            param.grad = None
         self.optimizer.zero_grad()
+
+    # capability flags, forwarded from the wrapped optimizer ---------------
+
+    def _inner_flag(self, name):
+        return getattr(self.optimizer, name, False)
 
     @property
     def supports_memory_efficient_fp16(self):
-        if hasattr(self.optimizer, "supports_memory_efficient_fp16"):
-            return self.optimizer.supports_memory_efficient_fp16
-        return False
+        return self._inner_flag("supports_memory_efficient_fp16")
 
     @property
     def supports_step_with_scale(self):
-        if hasattr(self.optimizer, "supports_step_with_scale"):
-            return self.optimizer.supports_step_with_scale
-        return False
+        return self._inner_flag("supports_step_with_scale")
 
     @property
     def supports_groups(self):
-        if hasattr(self.optimizer, "supports_groups"):
-            return self.optimizer.supports_groups
-        return False
+        return self._inner_flag("supports_groups")
 
     @property
     def supports_flat_params(self):
-        """
-        Whether the optimizer supports collapsing of the model
-        parameters/gradients into a single contiguous Tensor.
-        """
-        if hasattr(self.optimizer, "supports_flat_params"):
-            return self.optimizer.supports_flat_params
-        return False
+        """True when params/grads may be collapsed into one contiguous
+        tensor per group."""
+        return self._inner_flag("supports_flat_params")
 
     def broadcast_global_state_dict(self, state_dict):
-        """
-        Broadcasts a global state dict to all ranks.
-        Useful for optimizers that shard state between ranks.
-        """
-        if hasattr(self.optimizer, "broadcast_global_state_dict"):
-            return self.optimizer.broadcast_global_state_dict(state_dict)
-        else:
-            return state_dict
+        """For optimizers that shard state across ranks."""
+        fn = getattr(self.optimizer, "broadcast_global_state_dict", None)
+        return fn(state_dict) if fn is not None else state_dict
