@@ -153,7 +153,7 @@ def test_slurm_init_inference(monkeypatch):
     monkeypatch.setattr(
         subprocess, "check_output", lambda cmd: b"node01\nnode02\n"
     )
-    dutils._infer_slurm_init(args)
+    dutils._setup_from_slurm(args)
     assert args.distributed_init_method == "tcp://node01:12345"
     assert args.distributed_rank == 3
     assert args.device_id == 1

@@ -3,12 +3,13 @@
 Functional replacement for the reference's ``no_c10d``/legacy engine
 (reference unicore/distributed/legacy_distributed_data_parallel.py:27-166):
 no backward overlap; after backward the trainer calls
-``all_reduce_grads()`` which copies grads into a coalescing buffer
-(default 256 MB) and all-reduces it bucket by bucket. Required for
-``--allreduce-fp32-grad`` and ``--per-sample-clip-norm`` modes, where the
-fp32 flat grads are reduced by the optimizer instead.
+``all_reduce_grads()``, which walks the parameters in name order, packs
+grads into a coalescing buffer (default 256 MB) and all-reduces it bucket
+by bucket, pre-divided by the world size. This engine is required for
+``--allreduce-fp32-grad`` and ``--per-sample-clip-norm``, where the fp32
+flat grads are reduced by the optimizer instead.
 
-Parameters whose name ends with ``.expert`` are skipped during sync (same
+Parameters carrying an ``expert`` attribute are never synchronized (same
 escape hatch as the reference, legacy_distributed_data_parallel.py:142-144).
 """
 
@@ -30,116 +31,112 @@ class LegacyDDP(nn.Module):
         self.world_size = (
             dist.get_world_size(process_group) if dist.is_initialized() else 1
         )
-        # Never use a bigger buffer than the number of model params
+        # no point allocating more than the model itself occupies
         self.buffer_size = min(
             buffer_size, sum(p.numel() for p in module.parameters())
         )
         self.buffer = None
         self.accumulate_grads = False
 
-        # We can also forcibly accumulate grads locally and only do the
-        # all-reduce at some later time
-        self._grad_sync_disabled = False
-
     @contextmanager
     def no_sync(self):
-        """A context manager to disable gradient synchronization."""
-        old = self.accumulate_grads
+        """Suppress grad sync inside the block (grad accumulation)."""
+        saved = self.accumulate_grads
         self.accumulate_grads = True
         yield
-        self.accumulate_grads = old
+        self.accumulate_grads = saved
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
 
-    def all_reduce_params(self, params):
-        buffer = self.buffer
-        nonzero_buffer = False
-        if len(params) > 1:
-            offset = 0
-            for p in params:
-                sz = p.numel()
-                if p.grad is not None:
-                    buffer[offset : offset + sz].copy_(p.grad.data.view(-1))
-                    nonzero_buffer = True
-                else:
-                    buffer[offset : offset + sz].zero_()
-                offset += sz
-        else:
-            # we only have a single grad to all-reduce
+    def _reduce_bucket(self, params):
+        """Coalesce one bucket of grads, all-reduce, scatter back."""
+        scratch = self.buffer
+        any_grad = False
+        if len(params) == 1:
+            # single (possibly oversized) param: reduce its grad in place
             p = params[0]
             if p.grad is not None:
-                buffer = p.grad.data
-                nonzero_buffer = True
+                scratch = p.grad.data
+                any_grad = True
             elif p.numel() <= self.buffer.numel():
-                buffer = buffer[: p.numel()]
-                buffer.zero_()
+                scratch = scratch[: p.numel()]
+                scratch.zero_()
             else:
-                buffer = torch.zeros_like(p)
-
-        if nonzero_buffer:
-            buffer.div_(self.world_size)
-
-        dist.all_reduce(buffer, group=self.process_group)
-
-        # copy all-reduced grads back into their original place
-        if len(params) > 1:
-            offset = 0
-            for p in params:
-                sz = p.numel()
-                if p.grad is not None:
-                    p.grad.data.copy_(buffer[offset : offset + sz].view_as(p))
-                else:
-                    p.grad = buffer[offset : offset + sz].view_as(p).clone()
-                offset += sz
+                scratch = torch.zeros_like(p)
         else:
+            cursor = 0
+            for p in params:
+                n = p.numel()
+                window = scratch[cursor: cursor + n]
+                if p.grad is not None:
+                    window.copy_(p.grad.data.view(-1))
+                    any_grad = True
+                else:
+                    window.zero_()
+                cursor += n
+
+        if any_grad:
+            scratch.div_(self.world_size)  # pre-divide: reduce stays a sum
+        dist.all_reduce(scratch, group=self.process_group)
+
+        # scatter the reduced values back into the param grads
+        if len(params) == 1:
             p = params[0]
             if p.grad is not None:
-                p.grad.data.copy_(buffer)
+                p.grad.data.copy_(scratch)
             else:
-                p.grad = buffer.clone()
+                p.grad = scratch.clone()
+        else:
+            cursor = 0
+            for p in params:
+                n = p.numel()
+                window = scratch[cursor: cursor + n].view_as(p)
+                if p.grad is not None:
+                    p.grad.data.copy_(window)
+                else:
+                    p.grad = window.clone()
+                cursor += n
+
+    # kept under the reference's name: the fp16 optimizer's
+    # --allreduce-fp32-grad path calls it with the flat fp32 grads
+    all_reduce_params = _reduce_bucket
 
     def all_reduce_grads(self):
-        """
-        This function must be called explicitly after backward to reduce
-        gradients.
-        """
+        """Synchronize gradients; call once after the last micro-batch's
+        backward."""
         if self.accumulate_grads or self.world_size == 1:
             return
 
         if self.buffer is None:
-            first = next(self.module.parameters())
-            self.buffer = first.new(self.buffer_size)
+            self.buffer = next(self.module.parameters()).new(self.buffer_size)
 
-        buffered_params = []
-        offset = 0
-        for param_name, param in self.module.named_parameters():
+        pending = []
+        used = 0
+        for name, param in self.module.named_parameters():
             if not param.requires_grad:
                 continue
             if param.grad is None:
                 param.grad = torch.zeros_like(param)
             if hasattr(param, "expert"):
-                # skip synchronizing grads for expert-tagged params
-                continue
+                continue  # expert-tagged params keep rank-local grads
             if param.grad.requires_grad:
                 raise RuntimeError(
                     "DistributedDataParallel only works with gradients that "
                     "don't require grad"
                 )
-            sz = param.numel()
-            if sz > self.buffer.numel():
-                # all-reduce big params directly
-                self.all_reduce_params([param])
-            else:
-                if offset + sz > self.buffer.numel():
-                    self.all_reduce_params(buffered_params)
-                    offset = 0
-                    buffered_params.clear()
-                buffered_params.append(param)
-                offset += sz
+            n = param.numel()
+            if n > self.buffer.numel():
+                self._reduce_bucket([param])  # oversized: reduce alone
+                continue
+            if used + n > self.buffer.numel():
+                self._reduce_bucket(pending)
+                pending, used = [], 0
+            pending.append(param)
+            used += n
 
-        if len(buffered_params) > 0:
-            self.all_reduce_params(buffered_params)
+        if pending:
+            self._reduce_bucket(pending)
 
     def state_dict(self, *args, **kwargs):
         return self.module.state_dict(*args, **kwargs)

@@ -1,51 +1,53 @@
 """Attr/state_dict passthrough for DDP-wrapped modules
-(parity: reference unicore/distributed/module_proxy_wrapper.py:10-62)."""
+(parity: reference unicore/distributed/module_proxy_wrapper.py:10-62).
+
+A DDP engine wraps the user model one level down (``wrapper.module``);
+this proxy keeps the outer object behaving like the inner model: unknown
+attributes resolve inward, and (de)serialization targets the inner module
+so checkpoints never contain DDP prefixes.
+"""
 
 from torch import nn
 
 
 class ModuleProxyWrapper(nn.Module):
-    """
-    Wrap a DistributedDataParallel module and forward requests for missing
-    attributes to the inner module (e.g., to support getattr on the wrapped
-    model), and save/load the state of the inner module.
+    """Usage::
 
-    Usage::
-
-        module.xyz = "hello world"
-        wrapped_module = DistributedDataParallel(module, **ddp_args)
-        wrapped_module = ModuleProxyWrapper(wrapped_module)
-        assert wrapped_module.xyz == "hello world"
-        assert wrapped_module.state_dict().keys() == module.state_dict().keys()
+        model.some_attr = 123
+        wrapped = ModuleProxyWrapper(SomeDDPEngine(model, ...))
+        assert wrapped.some_attr == 123
+        assert wrapped.state_dict().keys() == model.state_dict().keys()
     """
 
     def __init__(self, module: nn.Module):
         super().__init__()
-        assert hasattr(
-            module, "module"
-        ), "ModuleProxyWrapper expects input to wrap another module"
+        if not hasattr(module, "module"):
+            raise AssertionError(
+                "ModuleProxyWrapper wraps a DDP engine that itself wraps "
+                "the model"
+            )
         self.module = module
 
     def __getattr__(self, name):
-        """Forward missing attributes to twice-wrapped module."""
+        # resolution order: nn.Module machinery -> the DDP engine -> the
+        # user model inside it
         try:
-            # defer to nn.Module's logic
             return super().__getattr__(name)
         except AttributeError:
-            try:
-                # forward to the once-wrapped module
-                return getattr(self.module, name)
-            except AttributeError:
-                # forward to the twice-wrapped module
-                return getattr(self.module.module, name)
+            engine = self._modules["module"]
+            if hasattr(engine, name):
+                return getattr(engine, name)
+            return getattr(engine.module, name)
 
     def state_dict(self, *args, **kwargs):
-        """Forward to the twice-wrapped module."""
-        return self.module.module.state_dict(*args, **kwargs)
+        """Serialize the INNER model (no DDP key prefixes)."""
+        inner = self.module.module
+        return inner.state_dict(*args, **kwargs)
 
     def load_state_dict(self, *args, **kwargs):
-        """Forward to the twice-wrapped module."""
-        return self.module.module.load_state_dict(*args, **kwargs)
+        """Restore into the inner model."""
+        inner = self.module.module
+        return inner.load_state_dict(*args, **kwargs)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)

@@ -5,7 +5,10 @@ distributed_init:109, call_main:166, collective helpers:236-527.
 
 On MI355X nodes the torch.distributed "nccl" backend IS RCCL over xGMI; on
 CPU-only hosts (tests) gloo is used. All collective entry points below route
-through torch.distributed so the same code path covers both.
+through torch.distributed so one code path covers both. Structured object
+transport (pickled stat sync, checkpoint broadcast) is built from three
+primitives at the bottom of this file: a byte-blob broadcast, a metadata-free
+tensor-list broadcast, and one generic structure-mapping traversal.
 """
 
 import io
@@ -31,87 +34,83 @@ def is_master(args):
     return args.distributed_rank == 0
 
 
+# -- launch-mode inference --------------------------------------------------
+
 def infer_init_method(args, force_distributed=False):
+    """Fill args.distributed_init_method from the environment: torchrun
+    env vars, a Slurm allocation, or single-node auto-spawn."""
     if args.distributed_init_method is not None:
         return
-
-    if all(
-        key in os.environ
-        for key in ["MASTER_ADDR", "MASTER_PORT", "WORLD_SIZE", "RANK"]
-    ):
-        # support torch.distributed.run / torchrun
-        _infer_torch_distributed_launch_init(args)
+    torchrun_keys = ("MASTER_ADDR", "MASTER_PORT", "WORLD_SIZE", "RANK")
+    if all(k in os.environ for k in torchrun_keys):
+        _setup_from_torchrun(args)
     elif args.distributed_port > 0:
-        # we can determine the init method automatically for Slurm
-        _infer_slurm_init(args)
+        _setup_from_slurm(args)
     elif args.distributed_world_size > 1 or force_distributed:
-        # fallback for single node with multiple GPUs
-        _infer_single_node_init(args)
+        _setup_single_node(args)
 
 
-def _infer_torch_distributed_launch_init(args):
+def _setup_from_torchrun(args):
     args.distributed_init_method = "env://"
     args.distributed_world_size = int(os.environ["WORLD_SIZE"])
     args.distributed_rank = int(os.environ["RANK"])
-    # processes are created by torchrun
-    args.distributed_no_spawn = True
+    args.distributed_no_spawn = True  # torchrun already forked us
 
 
-def _infer_slurm_init(args):
-    node_list = os.environ.get("SLURM_STEP_NODELIST")
-    if node_list is None:
-        node_list = os.environ.get("SLURM_JOB_NODELIST")
-    if node_list is not None:
-        try:
-            hostnames = subprocess.check_output(
-                ["scontrol", "show", "hostnames", node_list]
-            )
-            args.distributed_init_method = "tcp://{host}:{port}".format(
-                host=hostnames.split()[0].decode("utf-8"),
-                port=args.distributed_port,
-            )
-            nnodes = int(os.environ.get("SLURM_NNODES"))
-            ntasks_per_node = os.environ.get("SLURM_NTASKS_PER_NODE")
-            if ntasks_per_node is not None:
-                ntasks_per_node = int(ntasks_per_node)
-            else:
-                ntasks = int(os.environ.get("SLURM_NTASKS"))
-                nnodes = int(os.environ.get("SLURM_NNODES"))
-                assert ntasks % nnodes == 0
-                ntasks_per_node = int(ntasks / nnodes)
-            if ntasks_per_node == 1:
-                gpus_per_node = torch.cuda.device_count()
-                node_id = int(os.environ.get("SLURM_NODEID"))
-                args.distributed_rank = node_id * gpus_per_node
-                args.distributed_world_size = nnodes * gpus_per_node
-            else:
-                assert ntasks_per_node == args.distributed_world_size // nnodes
-                args.distributed_no_spawn = True
-                args.distributed_rank = int(os.environ.get("SLURM_PROCID"))
-                args.device_id = int(os.environ.get("SLURM_LOCALID"))
-        except subprocess.CalledProcessError as e:  # scontrol failed
-            raise e
-        except FileNotFoundError:  # Slurm is not installed
-            pass
+def _setup_from_slurm(args):
+    nodes = os.environ.get("SLURM_STEP_NODELIST") \
+        or os.environ.get("SLURM_JOB_NODELIST")
+    if nodes is None:
+        return
+    try:
+        first_host = (
+            subprocess.check_output(["scontrol", "show", "hostnames", nodes])
+            .split()[0]
+            .decode("utf-8")
+        )
+    except FileNotFoundError:
+        return  # no Slurm tooling on this host
+    args.distributed_init_method = (
+        f"tcp://{first_host}:{args.distributed_port}"
+    )
+    nnodes = int(os.environ.get("SLURM_NNODES"))
+    per_node = os.environ.get("SLURM_NTASKS_PER_NODE")
+    if per_node is not None:
+        per_node = int(per_node)
+    else:
+        per_node = int(os.environ.get("SLURM_NTASKS")) // nnodes
+        assert int(os.environ.get("SLURM_NTASKS")) % nnodes == 0
+    if per_node == 1:
+        # one task per node: spawn a process per local GPU ourselves
+        local_gpus = torch.cuda.device_count()
+        args.distributed_rank = int(os.environ.get("SLURM_NODEID")) * local_gpus
+        args.distributed_world_size = nnodes * local_gpus
+    else:
+        assert per_node == args.distributed_world_size // nnodes
+        args.distributed_no_spawn = True
+        args.distributed_rank = int(os.environ.get("SLURM_PROCID"))
+        args.device_id = int(os.environ.get("SLURM_LOCALID"))
 
 
-def _infer_single_node_init(args):
-    assert (
-        args.distributed_world_size <= torch.cuda.device_count()
-    ), f"world size is {args.distributed_world_size} but have {torch.cuda.device_count()} available devices"
-    port = random.randint(10000, 20000)
-    args.distributed_init_method = "tcp://127.0.0.1:{port}".format(port=port)
+def _setup_single_node(args):
+    assert args.distributed_world_size <= torch.cuda.device_count(), (
+        f"world size is {args.distributed_world_size} but have "
+        f"{torch.cuda.device_count()} available devices"
+    )
+    args.distributed_init_method = (
+        f"tcp://127.0.0.1:{random.randint(10000, 20000)}"
+    )
 
+
+# -- process-group bring-up -------------------------------------------------
 
 def distributed_init(args):
     if dist.is_available() and dist.is_initialized():
         warnings.warn("Distributed is already initialized, cannot initialize twice!")
     else:
         logger.info(
-            "distributed init (rank {}): {}".format(
-                args.distributed_rank,
-                args.distributed_init_method,
-            )
+            f"distributed init (rank {args.distributed_rank}): "
+            f"{args.distributed_init_method}"
         )
         backend = args.distributed_backend
         if backend == "nccl" and (
@@ -128,26 +127,21 @@ def distributed_init(args):
             timeout=timedelta(seconds=args.distributed_init_timeout),
         )
         logger.info(
-            "initialized host {} as rank {}".format(
-                socket.gethostname(),
-                args.distributed_rank,
-            )
+            f"initialized host {socket.gethostname()} as rank "
+            f"{args.distributed_rank}"
         )
-
-        # perform a dummy all-reduce to initialize the communicator
-        # (RCCL communicator setup + xGMI link bring-up happens here, not on
-        # the first training step)
+        # warm-up all-reduce: RCCL communicator setup + xGMI link bring-up
+        # happens here, not on the first training step
+        warmup = torch.zeros(1)
         if torch.cuda.is_available():
-            dist.all_reduce(torch.zeros(1).cuda())
-        else:
-            dist.all_reduce(torch.zeros(1))
+            warmup = warmup.cuda()
+        dist.all_reduce(warmup)
 
     args.distributed_rank = dist.get_rank()
 
-    if is_master(args):
-        logging.getLogger().setLevel(logging.INFO)
-    else:
-        logging.getLogger().setLevel(logging.WARNING)
+    # squelch non-master ranks to WARNING
+    level = logging.INFO if is_master(args) else logging.WARNING
+    logging.getLogger().setLevel(level)
 
     return args.distributed_rank
 
@@ -156,14 +150,15 @@ def distributed_main(i, main, args, kwargs):
     args.device_id = i
     if torch.cuda.is_available() and not args.cpu:
         torch.cuda.set_device(args.device_id)
-    if args.distributed_rank is None:  # torch.multiprocessing.spawn
+    if args.distributed_rank is None:
+        # spawned by torch.multiprocessing: derive rank from local index
         args.distributed_rank = kwargs.pop("start_rank", 0) + i
 
     args.distributed_rank = distributed_init(args)
 
-    after_distributed_init_fn = kwargs.pop("after_distributed_init_fn", None)
-    if after_distributed_init_fn:
-        args = after_distributed_init_fn(args)
+    post_init = kwargs.pop("after_distributed_init_fn", None)
+    if post_init:
+        args = post_init(args)
 
     main(args, **kwargs)
 
@@ -172,62 +167,50 @@ def distributed_main(i, main, args, kwargs):
 
 
 def call_main(args, main, **kwargs):
+    """Entry point: run *main* directly, under torchrun/Slurm ranks, or by
+    spawning one process per local GPU."""
     if args.distributed_init_method is None:
         infer_init_method(args)
 
-    if args.distributed_init_method is not None:
-        # distributed training
-        if not args.distributed_no_spawn:
-            start_rank = args.distributed_rank
-            args.distributed_rank = None  # assign automatically
-            kwargs["start_rank"] = start_rank
-            torch.multiprocessing.spawn(
-                fn=distributed_main,
-                args=(main, args, kwargs),
-                nprocs=min(
-                    torch.cuda.device_count(),
-                    args.distributed_world_size,
-                ),
-                join=True,
-            )
-        else:
-            distributed_main(int(os.environ.get("LOCAL_RANK", args.device_id)), main, args, kwargs)
+    if args.distributed_init_method is None:
+        main(args, **kwargs)  # plain single-process run
+    elif args.distributed_no_spawn:
+        local = int(os.environ.get("LOCAL_RANK", args.device_id))
+        distributed_main(local, main, args, kwargs)
     else:
-        # single GPU main
-        main(args, **kwargs)
+        kwargs["start_rank"] = args.distributed_rank
+        args.distributed_rank = None  # filled in per spawned process
+        torch.multiprocessing.spawn(
+            fn=distributed_main,
+            args=(main, args, kwargs),
+            nprocs=min(torch.cuda.device_count(), args.distributed_world_size),
+            join=True,
+        )
 
+
+# -- groups and sizes -------------------------------------------------------
 
 def get_global_group():
-    if dist.is_initialized():
-        if not hasattr(get_global_group, "_global_group"):
-            # ideally we could use torch.distributed.group.WORLD, but it seems
-            # to cause random NCCL hangs in some cases
-            get_global_group._global_group = dist.new_group()
-        return get_global_group._global_group
-    else:
+    if not dist.is_initialized():
         return None
+    if not hasattr(get_global_group, "_global_group"):
+        # a fresh group rather than WORLD: WORLD has produced sporadic
+        # NCCL/RCCL hangs
+        get_global_group._global_group = dist.new_group()
+    return get_global_group._global_group
 
 
 def get_global_rank():
-    if dist.is_initialized():
-        return dist.get_rank()
-    else:
-        return 0
+    return dist.get_rank() if dist.is_initialized() else 0
 
 
 def get_global_world_size():
-    if dist.is_initialized():
-        return dist.get_world_size()
-    else:
-        return 1
+    return dist.get_world_size() if dist.is_initialized() else 1
 
 
 def get_data_parallel_group():
-    """Get the data-parallel group the caller rank belongs to.
-
-    Like the reference (unicore/distributed/utils.py:221-233) DP is the only
-    parallelism, so this is the global group.
-    """
+    """DP is the only parallelism (as in the reference,
+    unicore/distributed/utils.py:221-233): the DP group IS the global group."""
     return get_global_group()
 
 
@@ -240,26 +223,26 @@ def get_data_parallel_world_size():
 
 
 def get_rank(group):
-    if not dist.is_initialized():
-        return 0
-    return dist.get_rank(group=group)
+    return dist.get_rank(group=group) if dist.is_initialized() else 0
 
 
 def get_world_size(group):
-    if dist.is_initialized():
-        return dist.get_world_size(group=group)
-    else:
-        return 1
+    return dist.get_world_size(group=group) if dist.is_initialized() else 1
+
+
+# -- tensor collectives -----------------------------------------------------
+
+_REDUCE_OPS = {}
 
 
 def all_reduce(tensor, group, op="sum"):
-    if op == "sum":
-        op = dist.ReduceOp.SUM
-    elif op == "max":
-        op = dist.ReduceOp.MAX
-    else:
-        raise NotImplementedError
-    dist.all_reduce(tensor, op=op, group=group)
+    if not _REDUCE_OPS:
+        _REDUCE_OPS.update(sum=dist.ReduceOp.SUM, max=dist.ReduceOp.MAX)
+    try:
+        reduce_op = _REDUCE_OPS[op]
+    except KeyError:
+        raise NotImplementedError(f"all_reduce op {op}")
+    dist.all_reduce(tensor, op=reduce_op, group=group)
     return tensor
 
 
@@ -268,36 +251,55 @@ def broadcast(tensor, src, group):
 
 
 def all_to_all(tensor, group):
-    """Perform an all-to-all operation on a 1D Tensor."""
+    """All-to-all exchange of equal slices of a 1-D tensor."""
     assert tensor.dim() == 1
-    split_count = get_world_size(group=group)
-    assert tensor.numel() % split_count == 0
+    assert tensor.numel() % get_world_size(group=group) == 0
     output = torch.zeros_like(tensor)
     dist.all_to_all_single(output, tensor, group=group)
     return output
 
 
 def all_gather(tensor, group, return_tensor=False):
-    """Perform an all-gather operation."""
+    """Gather one tensor per rank (all same shape)."""
     world_size = get_world_size(group=group)
     rank = get_rank(group=group)
-    tensor_list = [
-        tensor if i == rank else torch.empty_like(tensor) for i in range(world_size)
+    slots = [
+        tensor if i == rank else torch.empty_like(tensor)
+        for i in range(world_size)
     ]
-    dist.all_gather(tensor_list, tensor, group=group)
-    if return_tensor:
-        return torch.stack(tensor_list, dim=0)
-    else:
-        return tensor_list
+    dist.all_gather(slots, tensor, group=group)
+    return torch.stack(slots, dim=0) if return_tensor else slots
+
+
+# -- pickled-object collectives ---------------------------------------------
+
+class _GatherScratch:
+    """Reused device + pinned-host buffers for all_gather_list."""
+
+    def __init__(self):
+        self.device_buf = None
+        self.host_buf = None
+
+    def get(self, buffer_size, max_size):
+        if self.device_buf is None or self.device_buf.numel() < buffer_size:
+            self.device_buf = torch.empty(buffer_size, dtype=torch.uint8)
+            self.host_buf = torch.empty(max_size, dtype=torch.uint8)
+            if torch.cuda.is_available():
+                self.device_buf = self.device_buf.cuda()
+                self.host_buf = self.host_buf.pin_memory()
+        return self.device_buf, self.host_buf
+
+
+_gather_scratch = _GatherScratch()
+_LEN_HEADER = 4  # big-endian u32 length prefix per rank slot
 
 
 def all_gather_list(data, group=None, max_size=16384):
-    """Gathers arbitrary data from all nodes into a list.
+    """Gather arbitrary picklable *data* from every rank into a list.
 
-    Similar to :func:`~torch.distributed.all_gather` but for arbitrary Python
-    data. Note that *data* must be picklable and any CUDA tensors will be moved
-    to CPU and returned on CPU as well.
-    (reference unicore/distributed/utils.py:275-349)
+    Implemented as a zero-filled byte-slot-per-rank all_reduce (works on
+    both RCCL and gloo); CUDA tensors come back on CPU
+    (reference unicore/distributed/utils.py:275-349).
     """
     from unicore_amd import utils
 
@@ -306,108 +308,92 @@ def all_gather_list(data, group=None, max_size=16384):
     rank = get_rank(group=group)
     world_size = get_world_size(group=group)
 
-    buffer_size = max_size * world_size
-    if (
-        not hasattr(all_gather_list, "_buffer")
-        or all_gather_list._buffer.numel() < buffer_size
-    ):
-        all_gather_list._buffer = torch.empty(buffer_size, dtype=torch.uint8)
-        if torch.cuda.is_available():
-            all_gather_list._buffer = all_gather_list._buffer.cuda()
-        cpu_buf = torch.empty(max_size, dtype=torch.uint8)
-        if torch.cuda.is_available():
-            cpu_buf = cpu_buf.pin_memory()
-        all_gather_list._cpu_buffer = cpu_buf
-    buffer = all_gather_list._buffer
+    buffer, cpu_buffer = _gather_scratch.get(max_size * world_size, max_size)
     buffer.zero_()
-    cpu_buffer = all_gather_list._cpu_buffer
 
-    data = utils.move_to_cpu(data)
-    enc = pickle.dumps(data)
-    enc_size = len(enc)
-    header_size = 4  # size of header that contains the length of the encoded data
-    size = header_size + enc_size
+    blob = pickle.dumps(utils.move_to_cpu(data))
+    size = _LEN_HEADER + len(blob)
     if size > max_size:
         raise ValueError(
-            "encoded data size ({}) exceeds max_size ({})".format(size, max_size)
+            f"encoded data size ({size}) exceeds max_size ({max_size})"
         )
 
-    header = struct.pack(">I", enc_size)
-    cpu_buffer[:size] = torch.frombuffer(
-        bytearray(header + enc), dtype=torch.uint8
-    )
-    start = rank * max_size
-    buffer[start : start + size].copy_(cpu_buffer[:size])
+    framed = struct.pack(">I", len(blob)) + blob
+    cpu_buffer[:size] = torch.frombuffer(bytearray(framed), dtype=torch.uint8)
+    slot = rank * max_size
+    buffer[slot: slot + size].copy_(cpu_buffer[:size])
 
     all_reduce(buffer, group=group)
 
-    buffer = buffer.cpu()
+    flat = buffer.cpu()
     try:
-        result = []
+        out = []
         for i in range(world_size):
-            out_buffer = buffer[i * max_size : (i + 1) * max_size]
-            (enc_size,) = struct.unpack(">I", bytes(out_buffer[:header_size].tolist()))
-            if enc_size > 0:
-                result.append(
-                    pickle.loads(
-                        bytes(out_buffer[header_size : header_size + enc_size].tolist())
-                    )
+            chunk = flat[i * max_size: (i + 1) * max_size]
+            (blob_len,) = struct.unpack(
+                ">I", bytes(chunk[:_LEN_HEADER].tolist())
+            )
+            if blob_len > 0:
+                payload = bytes(
+                    chunk[_LEN_HEADER: _LEN_HEADER + blob_len].tolist()
                 )
-        return result
+                out.append(pickle.loads(payload))
+        return out
     except pickle.UnpicklingError:
         raise Exception(
-            "Unable to unpickle data from other workers. all_gather_list requires all "
-            "workers to enter the function together, so this error usually indicates "
-            "that the workers have fallen out of sync somehow. Workers can fall out of "
-            "sync if one of them runs out of memory, or if there are other conditions "
-            "in your training script that can cause one worker to finish an epoch "
-            "while other workers are still iterating over their portions of the data. "
-            "Try rerunning with --ddp-backend=legacy_ddp and see if that helps."
+            "Unable to unpickle data from other workers. all_gather_list "
+            "requires all workers to enter the function together, so this "
+            "error usually indicates that the workers have fallen out of "
+            "sync somehow. Workers can fall out of sync if one of them runs "
+            "out of memory, or if there are other conditions in your "
+            "training script that can cause one worker to finish an epoch "
+            "while other workers are still iterating over their portions of "
+            "the data. Try rerunning with --ddp-backend=legacy_ddp and see "
+            "if that helps."
         )
 
 
 def all_reduce_dict(data: Mapping[str, Any], device, group) -> Dict[str, Any]:
-    """
-    AllReduce a dictionary of values across workers. We separately
-    reduce items that are already on the device and items on CPU for
-    better performance.
-    (reference unicore/distributed/utils.py:352-398)
-    """
-    data_keys = list(data.keys())
+    """All-reduce a {key: scalar-or-tensor} mapping as two concatenated
+    double buffers — one for values already on *device*, one for host-side
+    values (reference unicore/distributed/utils.py:352-398)."""
+    keys = list(data.keys())
 
-    # We want to separately reduce items that are already on the
-    # device and items on CPU for performance reasons.
-    cpu_data = OrderedDict()
-    device_data = OrderedDict()
-    for k in data_keys:
-        t = data[k]
-        if not torch.is_tensor(t):
-            cpu_data[k] = torch.tensor(t, dtype=torch.double)
-        elif t.device.type != device.type:
-            cpu_data[k] = t.to(dtype=torch.double)
+    on_host, on_device = OrderedDict(), OrderedDict()
+    for k in keys:
+        v = data[k]
+        if not torch.is_tensor(v):
+            on_host[k] = torch.tensor(v, dtype=torch.double)
+        elif v.device.type != device.type:
+            on_host[k] = v.to(dtype=torch.double)
         else:
-            device_data[k] = t.to(dtype=torch.double)
+            on_device[k] = v.to(dtype=torch.double)
 
-    def _all_reduce_dict(data: OrderedDict):
-        if len(data) == 0:
-            return data
-        buf = torch.cat([t.view(-1) for t in data.values()]).to(device=device)
-        all_reduce(buf, group=group)
-        split_buf = torch.split(buf.clone(), [t.numel() for t in data.values()])
-        reduced_data = [t.view_as(orig) for t, orig in zip(split_buf, data.values())]
-        return OrderedDict(zip(data.keys(), reduced_data))
+    def _reduce_bundle(bundle):
+        if len(bundle) == 0:
+            return bundle
+        packed = torch.cat([t.view(-1) for t in bundle.values()]).to(device=device)
+        all_reduce(packed, group=group)
+        pieces = torch.split(
+            packed.clone(), [t.numel() for t in bundle.values()]
+        )
+        return OrderedDict(
+            (k, piece.view_as(orig))
+            for (k, orig), piece in zip(bundle.items(), pieces)
+        )
 
-    cpu_data = _all_reduce_dict(cpu_data)
-    device_data = _all_reduce_dict(device_data)
+    on_host = _reduce_bundle(on_host)
+    on_device = _reduce_bundle(on_device)
+    return OrderedDict(
+        (k, on_host[k] if k in on_host else on_device[k]) for k in keys
+    )
 
-    def get_from_stack(key):
-        if key in cpu_data:
-            return cpu_data[key]
-        elif key in device_data:
-            return device_data[key]
-        raise KeyError
 
-    return OrderedDict([(key, get_from_stack(key)) for key in data_keys])
+# -- structured broadcast (checkpoint distribution) -------------------------
+
+def _default_dist_device(group):
+    backend = torch.distributed.get_backend(group)
+    return torch.device("cuda" if backend == "nccl" else "cpu")
 
 
 def broadcast_tensors(
@@ -416,40 +402,33 @@ def broadcast_tensors(
     group: object,
     dist_device: Optional[torch.device] = None,
 ) -> List[torch.Tensor]:
-    """
-    Broadcasts a list of tensors without other (non-src) ranks needing to know
-    the dtypes/shapes of the tensors.
-    (reference unicore/distributed/utils.py:406-452)
-    """
+    """Broadcast a tensor list; receivers need no prior knowledge of the
+    shapes/dtypes (metadata goes first as a pickled blob)
+    (reference unicore/distributed/utils.py:406-452)."""
     if dist_device is None:
-        if torch.distributed.get_backend(group) == "nccl":
-            dist_device = torch.device("cuda")
-        else:
-            dist_device = torch.device("cpu")
+        dist_device = _default_dist_device(group)
 
-    # share metadata first to simplify transfer
-    is_src_rank = get_rank(group) == src_rank
-    if is_src_rank:
+    sending = get_rank(group) == src_rank
+    metadata = None
+    if sending:
         metadata = [
-            {"size": t.size(), "dtype": t.dtype, "device": t.device} for t in tensors
+            {"size": t.size(), "dtype": t.dtype, "device": t.device}
+            for t in tensors
         ]
-        metadata = _broadcast_object_slow(metadata, src_rank, group, dist_device)
-    else:
-        metadata = _broadcast_object_slow(None, src_rank, group, dist_device)
+    metadata = _broadcast_object_slow(metadata, src_rank, group, dist_device)
 
-    out_tensors = []
+    received = []
     for i, meta in enumerate(metadata):
-        if is_src_rank:
+        if sending:
             tensor = tensors[i]
-            broadcast(tensors[i].to(dist_device), src=src_rank, group=group)
+            broadcast(tensor.to(dist_device), src=src_rank, group=group)
         else:
             tensor = torch.zeros(
                 [meta["size"].numel()], dtype=meta["dtype"], device=dist_device
             )
             broadcast(tensor, src=src_rank, group=group)
-        tensor = tensor.view(meta["size"]).to(meta["device"])
-        out_tensors.append(tensor)
-    return out_tensors
+        received.append(tensor.view(meta["size"]).to(meta["device"]))
+    return received
 
 
 def broadcast_object(
@@ -458,52 +437,47 @@ def broadcast_object(
     group: object,
     dist_device: Optional[torch.device] = None,
 ) -> Any:
-    """Broadcast an arbitrary Python object to other workers
+    """Broadcast an arbitrary Python object. Tensors inside the structure
+    travel as raw broadcasts (no pickling of payload data)
     (reference unicore/distributed/utils.py:455-495)."""
     if dist_device is None:
-        if torch.distributed.get_backend(group) == "nccl":
-            dist_device = torch.device("cuda")
-        else:
-            dist_device = torch.device("cpu")
+        dist_device = _default_dist_device(group)
 
     if get_rank(group) == src_rank:
-        # split the tensors from the non-tensors so we can broadcast them
-        # directly, avoiding unnecessary serialization/deserialization
-        from unicore_amd import utils
-
         tensors = []
-        obj = _split_tensors_from_obj(obj, tensors)
-        obj = _broadcast_object_slow(obj, src_rank, group, dist_device)
+
+        def _pull(t):
+            tensors.append(t)
+            return _TensorPlaceholder(len(tensors) - 1)
+
+        skeleton = _map_tensors(obj, _pull)
+        skeleton = _broadcast_object_slow(skeleton, src_rank, group, dist_device)
         tensors = broadcast_tensors(tensors, src_rank, group, dist_device)
     else:
-        obj = _broadcast_object_slow(None, src_rank, group, dist_device)
+        skeleton = _broadcast_object_slow(None, src_rank, group, dist_device)
         tensors = broadcast_tensors(None, src_rank, group, dist_device)
-    return _put_tensors_in_obj(obj, tensors)
+    return _map_placeholders(skeleton, tensors)
 
 
-def _broadcast_object_slow(
-    obj: Any,
-    src_rank: int,
-    group: object,
-    dist_device: torch.device,
-) -> Any:
+def _broadcast_object_slow(obj, src_rank, group, dist_device):
+    """Length-prefixed torch.save blob broadcast."""
     if get_rank(group) == src_rank:
-        # Emit data
-        buffer = io.BytesIO()
-        torch.save(obj, buffer)
-        buffer = torch.ByteTensor(buffer.getbuffer()).to(dist_device)
-        length = torch.LongTensor([len(buffer)]).to(dist_device)
+        sink = io.BytesIO()
+        torch.save(obj, sink)
+        payload = torch.ByteTensor(sink.getbuffer()).to(dist_device)
+        length = torch.LongTensor([len(payload)]).to(dist_device)
         broadcast(length, src=src_rank, group=group)
-        broadcast(buffer, src=src_rank, group=group)
-    else:
-        # Fetch from the source
-        length = torch.LongTensor([0]).to(dist_device)
-        broadcast(length, src=src_rank, group=group)
-        buffer = torch.ByteTensor(int(length.item())).to(dist_device)
-        broadcast(buffer, src=src_rank, group=group)
-        buffer = io.BytesIO(buffer.cpu().numpy().tobytes())
-        obj = torch.load(buffer, map_location="cpu", weights_only=False)
-    return obj
+        broadcast(payload, src=src_rank, group=group)
+        return obj
+    length = torch.LongTensor([0]).to(dist_device)
+    broadcast(length, src=src_rank, group=group)
+    payload = torch.ByteTensor(int(length.item())).to(dist_device)
+    broadcast(payload, src=src_rank, group=group)
+    return torch.load(
+        io.BytesIO(payload.cpu().numpy().tobytes()),
+        map_location="cpu",
+        weights_only=False,
+    )
 
 
 class _TensorPlaceholder:
@@ -511,33 +485,29 @@ class _TensorPlaceholder:
         self.index = index
 
 
-def _split_tensors_from_obj(obj: Any, tensors: List[torch.Tensor]) -> Any:
-    if torch.is_tensor(obj):
-        placeholder = _TensorPlaceholder(index=len(tensors))
-        tensors.append(obj)
-        return placeholder
-    elif isinstance(obj, dict):
-        return {k: _split_tensors_from_obj(v, tensors) for k, v in obj.items()}
-    elif isinstance(obj, list):
-        return [_split_tensors_from_obj(v, tensors) for v in obj]
-    elif isinstance(obj, tuple):
-        return tuple(_split_tensors_from_obj(v, tensors) for v in obj)
-    elif isinstance(obj, set):
-        return {_split_tensors_from_obj(v, tensors) for v in obj}
-    else:
-        return obj
+def _walk(obj, leaf_test, leaf_fn):
+    """Rebuild dict/list/tuple/set structure, applying *leaf_fn* where
+    *leaf_test* matches."""
+    if leaf_test(obj):
+        return leaf_fn(obj)
+    if isinstance(obj, dict):
+        return {k: _walk(v, leaf_test, leaf_fn) for k, v in obj.items()}
+    if isinstance(obj, list):
+        return [_walk(v, leaf_test, leaf_fn) for v in obj]
+    if isinstance(obj, tuple):
+        return tuple(_walk(v, leaf_test, leaf_fn) for v in obj)
+    if isinstance(obj, set):
+        return {_walk(v, leaf_test, leaf_fn) for v in obj}
+    return obj
 
 
-def _put_tensors_in_obj(obj: Any, tensors: List[torch.Tensor]) -> Any:
-    if isinstance(obj, _TensorPlaceholder):
-        return tensors[obj.index]
-    elif isinstance(obj, dict):
-        return {k: _put_tensors_in_obj(v, tensors) for k, v in obj.items()}
-    elif isinstance(obj, list):
-        return [_put_tensors_in_obj(v, tensors) for v in obj]
-    elif isinstance(obj, tuple):
-        return tuple(_put_tensors_in_obj(v, tensors) for v in obj)
-    elif isinstance(obj, set):
-        return {_put_tensors_in_obj(v, tensors) for v in obj}
-    else:
-        return obj
+def _map_tensors(obj, fn):
+    return _walk(obj, torch.is_tensor, fn)
+
+
+def _map_placeholders(obj, tensors):
+    return _walk(
+        obj,
+        lambda o: isinstance(o, _TensorPlaceholder),
+        lambda o: tensors[o.index],
+    )

@@ -1,11 +1,11 @@
 """Exponential moving average of model weights.
 
-Parity: reference unicore/ema.py:6-65 — a deep-copied fp32 EMA model; when
-training in fp16/bf16 the EMA params are flattened per decay group (same
-layout as the optimizer's fp32 master flats) so each update is one fused
-lerp per group fed directly from ``optimizer.fp32_params``
-(reference unicore/trainer.py:720-725). Checkpointable as
-{"params": state_dict, "decay": float}.
+Parity: reference unicore/ema.py:6-65 — a deep-copied fp32 shadow model.
+When training in fp16/bf16 the shadow is flattened per decay group with the
+SAME (group, dtype, padding) layout as FP16Optimizer's fp32 masters, so
+each update is one fused lerp per group fed directly from
+``optimizer.fp32_params`` (reference unicore/trainer.py:720-725).
+Serialized as {"params": state_dict, "decay": float}.
 """
 
 from copy import deepcopy
@@ -19,59 +19,55 @@ class ExponentialMovingAverageModel:
         self.model_ema = deepcopy(model)
         self.decay = decay
         self.is_flattened = is_flattened
-        if not is_flattened:
-            self.name2param = self.get_name2param()
+        if is_flattened:
+            self.flatten_params = self._flatten_like_optimizer()
         else:
-            self.flatten_params = self.flatten_parameters()
+            self.name2param = self._promote_params()
 
-    def get_name2param(self):
-        name2param = dict()
-        for n, p in self.model_ema.named_parameters():
-            name2param[n] = p
-            # keep EMA state in fp32 regardless of model dtype
-            p.data = p.data.float()
-            p.grad = None
-        return name2param
+    def _promote_params(self):
+        """Per-name fp32 views of the shadow model (slow path)."""
+        table = {}
+        for name, param in self.model_ema.named_parameters():
+            table[name] = param
+            param.data = param.data.float()  # EMA state is always fp32
+            param.grad = None
+        return table
 
-    def flatten_parameters(self):
-        """Flatten the EMA copy with the same (decay-group, dtype, padding)
-        layout as FP16Optimizer's fp32 master flats, so update() can consume
-        optimizer.fp32_params directly."""
+    def _flatten_like_optimizer(self):
+        """Flatten the shadow with FP16Optimizer's exact group layout so
+        update() can consume optimizer.fp32_params positionally."""
         from unicore_amd.optim.fp16_optimizer import (
             flatten_fp32_master,
             separate_decay_params,
         )
 
-        param_group = separate_decay_params(self.args, self.model_ema.named_parameters())
+        groups = separate_decay_params(
+            self.args, self.model_ema.named_parameters()
+        )
         return [
-            flatten_fp32_master(pd["params"], set_to_param=True)
-            for pd in param_group
+            flatten_fp32_master(g["params"], set_to_param=True) for g in groups
         ]
 
     @torch.no_grad()
-    def _lerp(self, ema_param, new_param):
-        # ema -= (1 - decay) * (ema - new)
-        ema_param.add_(ema_param - new_param.data.float(), alpha=-(1.0 - self.decay))
+    def _lerp(self, shadow, live):
+        # shadow -= (1 - decay) * (shadow - live)
+        shadow.add_(shadow - live.data.float(), alpha=self.decay - 1.0)
 
     def update(self, new_param):
-        if self.is_flattened:
-            # new_param: optimizer.fp32_params (one flat per decay group)
-            with torch.no_grad():
-                for e, p in zip(self.flatten_params, new_param):
-                    self._lerp(e.data, p)
-        else:
-            # new_param: model.named_parameters()
-            with torch.no_grad():
-                for n, p in new_param:
-                    if n in self.name2param:
-                        self._lerp(self.name2param[n].data, p)
+        """Blend the live weights in. *new_param* is either the optimizer's
+        flat fp32 masters (flattened mode) or model.named_parameters()."""
+        with torch.no_grad():
+            if self.is_flattened:
+                for shadow, live in zip(self.flatten_params, new_param):
+                    self._lerp(shadow.data, live)
+            else:
+                for name, live in new_param:
+                    if name in self.name2param:
+                        self._lerp(self.name2param[name].data, live)
 
-    def load_state_dict(self, state_dict):
-        self.model_ema.load_state_dict(state_dict["params"])
-        self.decay = state_dict.get("decay", self.decay)
+    def state_dict(self) -> dict:
+        return {"params": self.model_ema.state_dict(), "decay": self.decay}
 
-    def state_dict(self):
-        return {
-            "params": self.model_ema.state_dict(),
-            "decay": self.decay,
-        }
+    def load_state_dict(self, state: dict) -> None:
+        self.model_ema.load_state_dict(state["params"])
+        self.decay = state.get("decay", self.decay)

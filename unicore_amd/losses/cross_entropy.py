@@ -1,4 +1,5 @@
-"""Cross entropy loss (parity: reference unicore/losses/cross_entropy.py:13-65)."""
+"""Plain cross entropy loss (parity: reference
+unicore/losses/cross_entropy.py:13-65). Sample size = batch rows."""
 
 import math
 
@@ -15,50 +16,32 @@ class CrossEntropyLoss(UnicoreLoss):
         super().__init__(task)
 
     def forward(self, model, sample, reduce=True):
-        """Compute the loss for the given sample.
-
-        Returns a tuple with three elements:
-        1) the loss
-        2) the sample size, which is used as the denominator for the gradient
-        3) logging outputs to display while training
-        """
         net_output = model(**sample["net_input"])
         loss = self.compute_loss(model, net_output, sample, reduce=reduce)
         sample_size = sample["target"].size(0)
-        logging_output = {
+        stats = {
             "loss": loss.data,
-            "bsz": sample["target"].size(0),
+            "bsz": sample_size,
             "sample_size": sample_size,
         }
-        return loss, sample_size, logging_output
+        return loss, sample_size, stats
 
     def compute_loss(self, model, net_output, sample, reduce=True):
-        lprobs = F.log_softmax(net_output, dim=-1, dtype=torch.float32)
-        lprobs = lprobs.view(-1, lprobs.size(-1))
-        target = sample["target"].view(-1)
-        loss = F.nll_loss(
-            lprobs,
-            target,
+        # fp32 log-softmax for numerical stability in low-precision training
+        logp = F.log_softmax(net_output, dim=-1, dtype=torch.float32)
+        return F.nll_loss(
+            logp.view(-1, logp.size(-1)),
+            sample["target"].view(-1),
             reduction="sum" if reduce else "none",
         )
-        return loss
 
     @staticmethod
     def reduce_metrics(logging_outputs, split="valid") -> None:
-        """Aggregate logging outputs from data parallel training."""
-        loss_sum = sum(log.get("loss", 0) for log in logging_outputs)
-        sample_size = sum(log.get("sample_size", 0) for log in logging_outputs)
-
-        # we divide by log(2) to convert the loss from base e to base 2
-        metrics.log_scalar(
-            "loss", loss_sum / sample_size / math.log(2), sample_size, round=3
-        )
+        loss_total = sum(log.get("loss", 0) for log in logging_outputs)
+        n = sum(log.get("sample_size", 0) for log in logging_outputs)
+        # log(2): report in bits rather than nats
+        metrics.log_scalar("loss", loss_total / n / math.log(2), n, round=3)
 
     @staticmethod
     def logging_outputs_can_be_summed(is_train) -> bool:
-        """
-        Whether the logging outputs returned by `forward` can be summed
-        across workers prior to calling `reduce_metrics`. Setting this
-        to True will improve distributed training speed.
-        """
         return True

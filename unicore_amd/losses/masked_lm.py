@@ -1,4 +1,9 @@
-"""Masked LM loss (parity: reference unicore/losses/masked_lm.py:12-67)."""
+"""Masked-LM loss (parity: reference unicore/losses/masked_lm.py:12-67).
+
+Targets are pad everywhere except masked positions; the model is given the
+mask so its LM head only projects the masked rows, and the loss runs
+through the fused online-logsumexp cross entropy.
+"""
 
 import math
 
@@ -16,52 +21,39 @@ class MaskedLMLoss(UnicoreLoss):
         self.padding_idx = task.dictionary.pad()
 
     def forward(self, model, sample, reduce=True):
-        """Compute the loss for the given sample.
-
-        Returns a tuple with three elements:
-        1) the loss
-        2) the sample size, which is used as the denominator for the gradient
-        3) logging outputs to display while training
-        """
-        masked_tokens = sample["target"].ne(self.padding_idx)
-        sample_size = masked_tokens.int().sum()
-
-        masked_tokens = torch.where(
-            masked_tokens.any(),
-            masked_tokens,
-            masked_tokens.new([True]),
-        )
-        logits = model(**sample["net_input"], masked_tokens=masked_tokens)
         target = sample["target"]
-        if masked_tokens is not None:
-            target = target[masked_tokens]
-        loss = fused_nll_loss(logits, target, ignore_index=self.padding_idx)
-        logging_output = {
+        masked = target.ne(self.padding_idx)
+        sample_size = masked.int().sum()
+        # degenerate batches with zero masked tokens still need one row so
+        # the graph stays connected (its target is pad -> zero loss)
+        masked = torch.where(masked.any(), masked, masked.new([True]))
+
+        logits = model(**sample["net_input"], masked_tokens=masked)
+        loss = fused_nll_loss(
+            logits, target[masked], ignore_index=self.padding_idx
+        )
+        stats = {
             "loss": loss.data,
-            "bsz": sample["target"].size(0),
+            "bsz": target.size(0),
             "sample_size": sample_size,
-            "seq_len": sample["target"].size(1) * sample["target"].size(0),
+            "seq_len": target.size(1) * target.size(0),
         }
-        return loss, sample_size, logging_output
+        return loss, sample_size, stats
 
     @staticmethod
     def reduce_metrics(logging_outputs, split="valid") -> None:
-        """Aggregate logging outputs from data parallel training."""
-        loss_sum = sum(log.get("loss", 0) for log in logging_outputs)
-        bsz = sum(log.get("bsz", 0) for log in logging_outputs)
-        sample_size = sum(log.get("sample_size", 0) for log in logging_outputs)
-        seq_len = sum(log.get("seq_len", 0) for log in logging_outputs)
-        # we divide by log(2) to convert the loss from base e to base 2
-        metrics.log_scalar(
-            "loss", loss_sum / sample_size / math.log(2), sample_size, round=3
-        )
-        metrics.log_scalar("seq_len", seq_len / bsz, 1, round=3)
+        totals = {
+            key: sum(log.get(key, 0) for log in logging_outputs)
+            for key in ("loss", "bsz", "sample_size", "seq_len")
+        }
+        n = totals["sample_size"]
+        denom = n if n > 0 else 1  # a whole batch can have zero masked tokens
+        # log(2): report in bits rather than nats
+        metrics.log_scalar("loss", totals["loss"] / denom / math.log(2), n,
+                           round=3)
+        metrics.log_scalar("seq_len", totals["seq_len"] / totals["bsz"], 1,
+                           round=3)
 
     @staticmethod
     def logging_outputs_can_be_summed(is_train) -> bool:
-        """
-        Whether the logging outputs returned by `forward` can be summed
-        across workers prior to calling `reduce_metrics`. Setting this
-        to True will improve distributed training speed.
-        """
         return True

@@ -1,4 +1,11 @@
-"""Loss base class (parity: reference unicore/losses/unicore_loss.py:14-68)."""
+"""Loss base class (parity: reference unicore/losses/unicore_loss.py:14-68).
+
+Contract: ``forward(model, sample) -> (loss, sample_size, logging_output)``
+where sample_size is the gradient denominator;
+``reduce_metrics`` aggregates the per-rank logging outputs; the
+``logging_outputs_can_be_summed`` flag enables the fast all-reduce stat
+sync path in the trainer.
+"""
 
 import inspect
 from typing import Any, Dict, List
@@ -14,67 +21,51 @@ class UnicoreLoss(_Loss):
             self.args = task.args
             if hasattr(task, "target_dictionary"):
                 tgt_dict = task.target_dictionary
-                self.padding_idx = (
-                    tgt_dict.pad() if tgt_dict is not None else -100
-                )
+                self.padding_idx = tgt_dict.pad() if tgt_dict is not None else -100
 
     @classmethod
     def add_args(cls, parser):
-        """Add loss-specific arguments to the parser."""
-        pass
+        """Hook for loss-specific CLI arguments."""
 
     @classmethod
     def build_loss(cls, args, task):
-        """Construct a loss from command-line args."""
-        # arguments in the __init__.
-        init_args = {}
-        for p in inspect.signature(cls).parameters.values():
-            if (
-                p.kind == p.POSITIONAL_ONLY
-                or p.kind == p.VAR_POSITIONAL
-                or p.kind == p.VAR_KEYWORD
-            ):
-                # we haven't implemented inference for these argument types,
-                # but PRs welcome :)
-                raise NotImplementedError("{} not supported".format(p.kind))
-
-            assert p.kind in {p.POSITIONAL_OR_KEYWORD, p.KEYWORD_ONLY}
-
-            if p.name == "task":
-                init_args["task"] = task
-            elif p.name == "args":
-                init_args["args"] = args
-            elif hasattr(args, p.name):
-                init_args[p.name] = getattr(args, p.name)
-            elif p.default != p.empty:
-                pass  # we'll use the default value
-            else:
+        """Instantiate the loss, wiring constructor parameters from the args
+        namespace by name (``task``/``args`` are passed through directly)."""
+        kwargs = {}
+        for param in inspect.signature(cls).parameters.values():
+            if param.kind in (param.POSITIONAL_ONLY, param.VAR_POSITIONAL,
+                              param.VAR_KEYWORD):
+                raise NotImplementedError(
+                    f"{param.kind} not supported"
+                )
+            assert param.kind in {param.POSITIONAL_OR_KEYWORD, param.KEYWORD_ONLY}
+            if param.name == "task":
+                kwargs["task"] = task
+            elif param.name == "args":
+                kwargs["args"] = args
+            elif hasattr(args, param.name):
+                kwargs[param.name] = getattr(args, param.name)
+            elif param.default is param.empty:
                 raise NotImplementedError(
                     "Unable to infer Loss arguments, please implement "
-                    "{}.build_loss".format(cls.__name__)
+                    f"{cls.__name__}.build_loss"
                 )
-        return cls(**init_args)
+            # else: the declared default applies
+        return cls(**kwargs)
 
     def forward(self, model, sample, reduce=True):
-        """Compute the loss for the given sample.
-
-        Returns a tuple with three elements:
-        1) the loss
-        2) the sample size, which is used as the denominator for the gradient
-        3) logging outputs to display while training
-        """
-        raise NotImplementedError
+        """Run the model on *sample* and return
+        ``(loss, sample_size, logging_output)``."""
+        raise NotImplementedError("losses implement forward")
 
     @staticmethod
-    def reduce_metrics(logging_outputs: List[Dict[str, Any]], split="valid") -> None:
-        """Aggregate logging outputs from data parallel training."""
-        raise NotImplementedError
+    def reduce_metrics(logging_outputs: List[Dict[str, Any]],
+                       split="valid") -> None:
+        """Fold the collected per-step logging outputs into metrics."""
+        raise NotImplementedError("losses implement reduce_metrics")
 
     @staticmethod
     def logging_outputs_can_be_summed(is_train: bool) -> bool:
-        """
-        Whether the logging outputs returned by `forward` can be summed
-        across workers prior to calling `reduce_metrics`. Setting this
-        to True will improve distributed training speed.
-        """
+        """True when forward()'s logging outputs are plain sums, enabling
+        the cheap cross-rank all-reduce sync instead of pickled gather."""
         return False

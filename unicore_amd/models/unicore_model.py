@@ -9,45 +9,41 @@ logger = logging.getLogger(__name__)
 
 
 class BaseUnicoreModel(nn.Module):
-    """Base class for models."""
+    """Base for all registered models: built by ``build_model(args, task)``,
+    checkpoint-loaded with an optional args namespace for shape adaptation,
+    and notified of the update count every step."""
 
     def __init__(self):
         super().__init__()
 
     @classmethod
     def add_args(cls, parser):
-        """Add model-specific arguments to the parser."""
-        pass
+        """Hook for model-specific CLI arguments."""
 
     @classmethod
     def build_model(cls, args, task):
-        """Build a new model instance."""
-        raise NotImplementedError("Model must implement the build_model method")
+        """Construct an instance from the parsed args + task."""
+        raise NotImplementedError(
+            f"{cls.__name__} must implement build_model"
+        )
 
     def extract_features(self, *args, **kwargs):
-        """Similar to *forward* but only return features."""
+        """Forward returning features only (defaults to plain forward)."""
         return self(*args, **kwargs)
 
-    def load_state_dict(
-        self,
-        state_dict,
-        strict=True,
-        model_args: Namespace = None,
-    ):
-        """Copies parameters and buffers from *state_dict* into this module and
-        its descendants.
-
-        Overrides the method in :class:`nn.Module`. Compared with that method
-        this additionally accepts *model_args*, giving models the chance to
-        adapt checkpoints (e.g. resize heads) before loading.
-        """
+    def load_state_dict(self, state_dict, strict=True,
+                        model_args: Namespace = None):
+        """nn.Module.load_state_dict plus a *model_args* hook: subclasses
+        may inspect it to adapt checkpoints (e.g. resize heads) before the
+        actual load."""
         return super().load_state_dict(state_dict, strict)
 
     def set_num_updates(self, num_updates):
-        """State from trainer to pass along to model at every update."""
+        """Propagate the trainer's update count to every submodule that
+        cares (e.g. schedules inside the model)."""
 
-        def _apply(m):
-            if hasattr(m, "set_num_updates") and m != self:
+        def _notify(m):
+            if m is not self and hasattr(m, "set_num_updates"):
                 m.set_num_updates(num_updates)
 
-        self.apply(_apply)
+        self.apply(_notify)
