@@ -1,8 +1,9 @@
 """NaN/Inf forensics: module fwd/bwd hooks that report the first bad tensor.
 
-Parity: reference unicore/nan_detector.py:15-109 — registered around a
-re-run of the failing batch after a FloatingPointError
-(reference unicore/trainer.py:727-748).
+Parity: reference unicore/nan_detector.py:15-109 — the trainer wraps a
+re-run of the failing batch in this context after a FloatingPointError
+(reference unicore/trainer.py:727-748); on exit it also dumps every
+non-finite grad norm.
 """
 
 import logging
@@ -13,97 +14,106 @@ logger = logging.getLogger(__name__)
 
 
 class NanDetector:
-    """
-    Detects the first NaN or Inf in forward and/or backward pass and logs,
-    together with the module name.
-    """
+    """Context manager that hooks every submodule and warns (once per
+    direction) on the first NaN/Inf seen in a forward output or backward
+    grad, naming the module that produced it."""
 
     def __init__(self, model, forward=True, backward=True):
-        self.bhooks = []
         self.fhooks = []
+        self.bhooks = []
         self.forward = forward
         self.backward = backward
         self.named_parameters = list(model.named_parameters())
         self.reset()
-
         for name, mod in model.named_modules():
             mod.__module_name = name
             self.add_hooks(mod)
-
-    def __enter__(self):
-        return self
-
-    def __exit__(self, exc_type, exc_value, exc_traceback):
-        # Dump out all model gnorms to enable better debugging
-        norm = {}
-        gradients = {}
-        for name, param in self.named_parameters:
-            if param.grad is not None:
-                grad_norm = torch.norm(param.grad.data.float(), p=2)
-                norm[name] = grad_norm.item()
-                if torch.isnan(grad_norm).any() or torch.isinf(grad_norm).any():
-                    gradients[name] = param.grad.data
-        if len(gradients) > 0:
-            logger.info("Detected nan/inf grad norm, dumping norms...")
-            logger.info(f"norms: {norm}")
-            logger.info(f"gradients: {gradients}")
-
-        self.close()
 
     def add_hooks(self, module):
         if self.forward:
             self.fhooks.append(module.register_forward_hook(self.fhook_fn))
         if self.backward:
-            self.bhooks.append(module.register_full_backward_hook(self.bhook_fn))
+            self.bhooks.append(
+                module.register_full_backward_hook(self.bhook_fn)
+            )
 
     def reset(self):
+        # one report per direction is enough; the first hit is the culprit
         self.has_printed_f = False
         self.has_printed_b = False
 
-    def _detect(self, tensor, name, backward):
-        err = None
-        if (
-            torch.is_floating_point(tensor)
-            # single value tensors (like the loss) will not provide much info
-            and tensor.numel() >= 2
-        ):
-            with torch.no_grad():
-                if torch.isnan(tensor).any():
-                    err = "NaN"
-                elif torch.isinf(tensor).any():
-                    err = "Inf"
-        if err is not None:
-            err = f"{err} detected in output of {name}, shape: {tuple(tensor.shape)}, {'backward' if backward else 'forward'}"
-        return err
+    def __enter__(self):
+        return self
 
-    def _apply(self, module, inp, x, backward):
-        if torch.is_tensor(x):
+    def __exit__(self, exc_type, exc_value, exc_traceback):
+        self._dump_grad_norms()
+        self.close()
+
+    def _dump_grad_norms(self):
+        """Log every param grad norm when any of them went non-finite."""
+        norms = {}
+        bad_grads = {}
+        for name, param in self.named_parameters:
+            if param.grad is None:
+                continue
+            gnorm = torch.norm(param.grad.data.float(), p=2)
+            norms[name] = gnorm.item()
+            if not torch.isfinite(gnorm).all():
+                bad_grads[name] = param.grad.data
+        if bad_grads:
+            logger.info("Detected nan/inf grad norm, dumping norms...")
+            logger.info(f"norms: {norms}")
+            logger.info(f"gradients: {bad_grads}")
+
+    def _describe(self, tensor, name, backward):
+        """A message naming the first non-finite value, or None if clean.
+        Scalar tensors (the loss) are skipped — no localization value."""
+        if not torch.is_floating_point(tensor) or tensor.numel() < 2:
+            return None
+        with torch.no_grad():
+            if torch.isnan(tensor).any():
+                kind = "NaN"
+            elif torch.isinf(tensor).any():
+                kind = "Inf"
+            else:
+                return None
+        direction = "backward" if backward else "forward"
+        return (
+            f"{kind} detected in output of {name}, "
+            f"shape: {tuple(tensor.shape)}, {direction}"
+        )
+
+    def _scan(self, module, inp, value, backward):
+        if torch.is_tensor(value):
             if isinstance(inp, tuple) and len(inp) > 0:
                 inp = inp[0]
-            err = self._detect(x, module.__module_name, backward)
-            if err is not None:
-                if torch.is_tensor(inp) and not backward:
-                    err += (
-                        f" input max: {inp.max().item()}, input min: {inp.min().item()}"
-                    )
-
-                has_printed_attr = "has_printed_b" if backward else "has_printed_f"
-                logger.warning(err)
-                setattr(self, has_printed_attr, True)
-        elif isinstance(x, dict):
-            for v in x.values():
-                self._apply(module, inp, v, backward)
-        elif isinstance(x, (list, tuple)):
-            for v in x:
-                self._apply(module, inp, v, backward)
+            msg = self._describe(value, module.__module_name, backward)
+            if msg is None:
+                return
+            if torch.is_tensor(inp) and not backward:
+                msg += (
+                    f" input max: {inp.max().item()},"
+                    f" input min: {inp.min().item()}"
+                )
+            logger.warning(msg)
+            if backward:
+                self.has_printed_b = True
+            else:
+                self.has_printed_f = True
+        elif isinstance(value, dict):
+            for v in value.values():
+                self._scan(module, inp, v, backward)
+        elif isinstance(value, (list, tuple)):
+            for v in value:
+                self._scan(module, inp, v, backward)
 
     def fhook_fn(self, module, inp, output):
         if not self.has_printed_f:
-            self._apply(module, inp, output, backward=False)
+            self._scan(module, inp, output, backward=False)
 
     def bhook_fn(self, module, inp, output):
         if not self.has_printed_b:
-            self._apply(module, inp, output, backward=True)
+            self._scan(module, inp, output, backward=True)
 
     def close(self):
         for hook in self.fhooks + self.bhooks:

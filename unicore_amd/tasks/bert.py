@@ -1,8 +1,9 @@
 """BERT masked-LM task over LMDB text data.
 
-Functional parity with reference examples/bert/task.py:31-124 (built-in here
-instead of requiring --user-dir): LMDB -> WordPiece tokenize -> 80/10/10 mask
--> right-pad -> nested dict -> epoch shuffle.
+Functional parity with reference examples/bert/task.py:31-124, built in here
+instead of requiring --user-dir. Pipeline per split:
+LMDB -> WordPiece tokenize -> 80/10/10 mask -> right-pad -> nested dict ->
+seeded shuffle order.
 """
 
 import logging
@@ -27,61 +28,42 @@ logger = logging.getLogger(__name__)
 
 @register_task("bert")
 class BertTask(UnicoreTask):
-    """Task for training masked language models (e.g., BERT)."""
+    """Masked-language-model pretraining (BERT-style)."""
 
     @staticmethod
     def add_args(parser):
-        """Add task-specific arguments to the parser."""
-        parser.add_argument(
-            "data",
-            help="colon separated path to data directories list, "
-            "will be iterated upon during epochs in round-robin manner",
-        )
-        parser.add_argument(
-            "--mask-prob",
-            default=0.15,
-            type=float,
-            help="probability of replacing a token with mask",
-        )
-        parser.add_argument(
-            "--leave-unmasked-prob",
-            default=0.1,
-            type=float,
-            help="probability that a masked token is unmasked",
-        )
-        parser.add_argument(
-            "--random-token-prob",
-            default=0.1,
-            type=float,
-            help="probability of replacing a token with a random token",
-        )
+        parser.add_argument("data",
+                            help="directory holding {split}.lmdb + dict.txt")
+        parser.add_argument("--mask-prob", default=0.15, type=float,
+                            help="fraction of tokens selected for masking")
+        parser.add_argument("--leave-unmasked-prob", default=0.1, type=float,
+                            help="share of selected tokens left unchanged")
+        parser.add_argument("--random-token-prob", default=0.1, type=float,
+                            help="share of selected tokens replaced randomly")
 
     def __init__(self, args, dictionary):
         super().__init__(args)
         self.dictionary = dictionary
         self.seed = args.seed
-
-        # add mask token
         self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
 
     @classmethod
     def setup_task(cls, args, **kwargs):
         dictionary = Dictionary.load(os.path.join(args.data, "dict.txt"))
-        logger.info("dictionary: {} types".format(len(dictionary)))
+        logger.info(f"dictionary: {len(dictionary)} types")
         return cls(args, dictionary)
 
     def load_dataset(self, split, combine=False, **kwargs):
-        """Load a given dataset split (e.g., train, valid, test)."""
-        split_path = os.path.join(self.args.data, split + ".lmdb")
-        dict_path = os.path.join(self.args.data, "dict.txt")
-
-        dataset = LMDBDataset(split_path)
-        dataset = BertTokenizeDataset(
-            dataset, dict_path, max_seq_len=self.args.max_seq_len
+        """Build the {split}.lmdb pipeline and register it under *split*."""
+        data_dir = self.args.data
+        tokens = BertTokenizeDataset(
+            LMDBDataset(os.path.join(data_dir, f"{split}.lmdb")),
+            os.path.join(data_dir, "dict.txt"),
+            max_seq_len=self.args.max_seq_len,
         )
 
-        src_dataset, tgt_dataset = MaskTokensDataset.apply_mask(
-            dataset,
+        src, tgt = MaskTokensDataset.apply_mask(
+            tokens,
             self.dictionary,
             pad_idx=self.dictionary.pad(),
             mask_idx=self.mask_idx,
@@ -92,22 +74,13 @@ class BertTask(UnicoreTask):
         )
 
         with data_utils.numpy_seed(self.args.seed):
-            shuffle = np.random.permutation(len(src_dataset))
+            order = np.random.permutation(len(src))
 
-        self.datasets[split] = SortDataset(
-            NestedDictionaryDataset(
-                {
-                    "net_input": {
-                        "src_tokens": RightPadDataset(
-                            src_dataset,
-                            pad_idx=self.dictionary.pad(),
-                        )
-                    },
-                    "target": RightPadDataset(
-                        tgt_dataset,
-                        pad_idx=self.dictionary.pad(),
-                    ),
-                },
-            ),
-            sort_order=[shuffle],
-        )
+        pad = self.dictionary.pad()
+        nested = NestedDictionaryDataset({
+            "net_input": {
+                "src_tokens": RightPadDataset(src, pad_idx=pad),
+            },
+            "target": RightPadDataset(tgt, pad_idx=pad),
+        })
+        self.datasets[split] = SortDataset(nested, sort_order=[order])
