@@ -19,19 +19,19 @@ class NanDetector:
     grad, naming the module that produced it."""
 
     def __init__(self, model, forward=True, backward=True):
-        self.fhooks = []
-        self.bhooks = []
-        self.forward = forward
-        self.backward = backward
-        self.named_parameters = list(model.named_parameters())
+        self.fhooks, self.bhooks = [], []
+        self.forward, self.backward = forward, backward
+        self.named_parameters = [*model.named_parameters()]
         self.reset()
-        for name, mod in model.named_modules():
-            mod.__module_name = name
-            self.add_hooks(mod)
+        for name, submodule in model.named_modules():
+            submodule.__module_name = name
+            self.add_hooks(submodule)
 
     def add_hooks(self, module):
         if self.forward:
-            self.fhooks.append(module.register_forward_hook(self.fhook_fn))
+            self.fhooks.append(
+                module.register_forward_hook(self.fhook_fn)
+            )
         if self.backward:
             self.bhooks.append(
                 module.register_full_backward_hook(self.bhook_fn)
@@ -39,8 +39,7 @@ class NanDetector:
 
     def reset(self):
         # one report per direction is enough; the first hit is the culprit
-        self.has_printed_f = False
-        self.has_printed_b = False
+        self.has_printed_f = self.has_printed_b = False
 
     def __enter__(self):
         return self
@@ -51,9 +50,8 @@ class NanDetector:
 
     def _dump_grad_norms(self):
         """Log every param grad norm when any of them went non-finite."""
-        norms = {}
-        bad_grads = {}
-        for name, param in self.named_parameters:
+        norms, bad_grads = {}, {}
+        for name, param in iter(self.named_parameters):
             if param.grad is None:
                 continue
             gnorm = torch.norm(param.grad.data.float(), p=2)
@@ -61,7 +59,7 @@ class NanDetector:
             if not torch.isfinite(gnorm).all():
                 bad_grads[name] = param.grad.data
         if bad_grads:
-            logger.info("Detected nan/inf grad norm, dumping norms...")
+            logger.info("Detected nan/inf grad norm, dumping norms ...")
             logger.info(f"norms: {norms}")
             logger.info(f"gradients: {bad_grads}")
 
@@ -71,9 +69,9 @@ class NanDetector:
         if not torch.is_floating_point(tensor) or tensor.numel() < 2:
             return None
         with torch.no_grad():
-            if torch.isnan(tensor).any():
+            if bool(torch.isnan(tensor).any()):
                 kind = "NaN"
-            elif torch.isinf(tensor).any():
+            elif bool(torch.isinf(tensor).any()):
                 kind = "Inf"
             else:
                 return None
@@ -85,12 +83,12 @@ class NanDetector:
 
     def _scan(self, module, inp, value, backward):
         if torch.is_tensor(value):
-            if isinstance(inp, tuple) and len(inp) > 0:
+            if isinstance(inp, tuple) and inp:
                 inp = inp[0]
             msg = self._describe(value, module.__module_name, backward)
             if msg is None:
                 return
-            if torch.is_tensor(inp) and not backward:
+            if not backward and torch.is_tensor(inp):
                 msg += (
                     f" input max: {inp.max().item()},"
                     f" input min: {inp.min().item()}"
@@ -108,13 +106,15 @@ class NanDetector:
                 self._scan(module, inp, v, backward)
 
     def fhook_fn(self, module, inp, output):
-        if not self.has_printed_f:
-            self._scan(module, inp, output, backward=False)
+        if self.has_printed_f:
+            return
+        self._scan(module, inp, output, backward=False)
 
     def bhook_fn(self, module, inp, output):
-        if not self.has_printed_b:
-            self._scan(module, inp, output, backward=True)
+        if self.has_printed_b:
+            return
+        self._scan(module, inp, output, backward=True)
 
     def close(self):
-        for hook in self.fhooks + self.bhooks:
+        for hook in [*self.fhooks, *self.bhooks]:
             hook.remove()

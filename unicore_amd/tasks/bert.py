@@ -43,15 +43,16 @@ class BertTask(UnicoreTask):
 
     def __init__(self, args, dictionary):
         super().__init__(args)
-        self.dictionary = dictionary
-        self.seed = args.seed
-        self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
+        self.dictionary, self.seed = dictionary, args.seed
+        self.mask_idx = dictionary.add_symbol(
+            "[MASK]", is_special=True
+        )
 
     @classmethod
     def setup_task(cls, args, **kwargs):
-        dictionary = Dictionary.load(os.path.join(args.data, "dict.txt"))
-        logger.info(f"dictionary: {len(dictionary)} types")
-        return cls(args, dictionary)
+        vocab = Dictionary.load(os.path.join(args.data, "dict.txt"))
+        logger.info(f"dictionary: {len(vocab)} types")
+        return cls(args, vocab)
 
     def load_dataset(self, split, combine=False, **kwargs):
         """Build the {split}.lmdb pipeline and register it under *split*."""
@@ -62,18 +63,16 @@ class BertTask(UnicoreTask):
             max_seq_len=self.args.max_seq_len,
         )
 
+        cfg = self.args
         src, tgt = MaskTokensDataset.apply_mask(
-            tokens,
-            self.dictionary,
-            pad_idx=self.dictionary.pad(),
-            mask_idx=self.mask_idx,
-            seed=self.args.seed,
-            mask_prob=self.args.mask_prob,
-            leave_unmasked_prob=self.args.leave_unmasked_prob,
-            random_token_prob=self.args.random_token_prob,
+            tokens, self.dictionary,
+            pad_idx=self.dictionary.pad(), mask_idx=self.mask_idx,
+            seed=cfg.seed, mask_prob=cfg.mask_prob,
+            leave_unmasked_prob=cfg.leave_unmasked_prob,
+            random_token_prob=cfg.random_token_prob,
         )
 
-        with data_utils.numpy_seed(self.args.seed):
+        with data_utils.numpy_seed(cfg.seed):
             order = np.random.permutation(len(src))
 
         pad = self.dictionary.pad()
