@@ -92,6 +92,31 @@ def _prep_flash_src(src, batch_dims, k_len, allowed_sq):
     return t, outer_div, True
 
 
+
+def _to_heads(t, bsz, num_heads, head_dim, scale=None):
+    """(B, L, E) -> (B*H, L, D), optionally folding the query scale."""
+    out = t.view(bsz, -1, num_heads, head_dim).transpose(1, 2)
+    out = out.contiguous().view(bsz * num_heads, -1, head_dim)
+    return out * scale if scale is not None else out
+
+
+def _from_heads(t, bsz, num_heads, tgt_len, head_dim):
+    """(B*H, L, D) -> (B, L, E)."""
+    merged = t.view(bsz, num_heads, tgt_len, head_dim).transpose(1, 2)
+    return merged.contiguous().view(bsz, tgt_len, num_heads * head_dim)
+
+
+def _padding_to_additive(kpm, bsz, src_len, dtype):
+    """Normalize a key-padding mask to the additive (B, 1, 1, k) form the
+    fused softmax broadcasts; scalar placeholders mean "no mask"."""
+    if kpm is not None and kpm.dim() == 0:
+        kpm = None
+    if kpm is None:
+        return None
+    assert kpm.size(0) == bsz and kpm.size(-1) == src_len
+    return kpm.view(bsz, 1, 1, src_len).to(dtype)
+
+
 class _QKVSplit(torch.autograd.Function):
     """Fused head-split: qkv (B, L, 3E) -> q,k,v each (B*H, L, D), with the
     q-scaling folded in (one HIP permute-copy each way instead of the
@@ -146,41 +171,27 @@ class _AttnMerge(torch.autograd.Function):
 
 
 class SelfMultiheadAttention(nn.Module):
-    def __init__(
-        self,
-        embed_dim,
-        num_heads,
-        dropout=0.1,
-        bias=True,
-        scaling_factor=1,
-    ):
+    def __init__(self, embed_dim, num_heads, dropout=0.1, bias=True,
+                 scaling_factor=1):
         super().__init__()
-        self.embed_dim = embed_dim
-        self.num_heads = num_heads
-        self.dropout = dropout
-
+        self.embed_dim, self.num_heads = embed_dim, num_heads
+        self.dropout = float(dropout)
         self.head_dim = embed_dim // num_heads
-        assert (
-            self.head_dim * num_heads == self.embed_dim
-        ), "embed_dim must be divisible by num_heads"
+        assert self.head_dim * num_heads == embed_dim, \
+            "embed_dim must be divisible by num_heads"
         self.scaling = (self.head_dim * scaling_factor) ** -0.5
 
         self.in_proj = nn.Linear(embed_dim, embed_dim * 3, bias=bias)
         self.out_proj = nn.Linear(embed_dim, embed_dim, bias=bias)
 
-    def forward(
-        self,
-        query,
-        key_padding_mask: Optional[Tensor] = None,
-        attn_bias: Optional[Tensor] = None,
-        return_attn: bool = False,
-        skip_out_bias: bool = False,
-    ) -> Tensor:
+    def forward(self, query, key_padding_mask: Optional[Tensor] = None,
+                attn_bias: Optional[Tensor] = None, return_attn: bool = False,
+                skip_out_bias: bool = False) -> Tensor:
         """``skip_out_bias``: run out_proj without its bias — the caller
         folds it into the following fused dropout+residual op (which then
         also produces the bias gradient)."""
-        bsz, tgt_len, embed_dim = query.size()
-        assert embed_dim == self.embed_dim
+        bsz, tgt_len, embed_dim = query.shape
+        assert embed_dim == self.embed_dim, "input width mismatch"
 
         use_fused_split = False
         if query.is_cuda and self.head_dim % 8 == 0:
@@ -203,38 +214,12 @@ class SelfMultiheadAttention(nn.Module):
         else:
             qkv = self.in_proj(query)
             q, k, v = qkv.chunk(3, dim=-1)
-            q = (
-                q.view(bsz, tgt_len, self.num_heads, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz * self.num_heads, -1, self.head_dim)
-                * self.scaling
-            )
-            k = (
-                k.view(bsz, -1, self.num_heads, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz * self.num_heads, -1, self.head_dim)
-            )
-            v = (
-                v.view(bsz, -1, self.num_heads, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz * self.num_heads, -1, self.head_dim)
-            )
+            q = _to_heads(q, bsz, self.num_heads, self.head_dim, self.scaling)
+            k = _to_heads(k, bsz, self.num_heads, self.head_dim)
+            v = _to_heads(v, bsz, self.num_heads, self.head_dim)
 
-        assert k is not None
-        src_len = k.size(1)
-
-        mask = None
-        if key_padding_mask is not None and key_padding_mask.dim() == 0:
-            key_padding_mask = None
-        if key_padding_mask is not None:
-            # additive float mask, (bsz, src_len) or (bsz, 1, 1, src_len);
-            # broadcast over heads + query positions by the fused kernel
-            assert key_padding_mask.size(0) == bsz
-            assert key_padding_mask.size(-1) == src_len
-            mask = key_padding_mask.view(bsz, 1, 1, src_len).to(q.dtype)
+        src_len = k.shape[1]
+        mask = _padding_to_additive(key_padding_mask, bsz, src_len, q.dtype)
 
         # flash path: bf16, head_dim 64, L % 64 == 0, kernel-expressible
         # bias/mask broadcasts — the L x L score matrix never hits HBM.
@@ -274,68 +259,52 @@ class SelfMultiheadAttention(nn.Module):
                 )
 
         if o is None:
-            attn_weights = torch.bmm(q, k.transpose(1, 2))
-            assert list(attn_weights.size()) == [
-                bsz * self.num_heads, tgt_len, src_len,
-            ]
-            attn_weights = attn_weights.view(bsz, self.num_heads, tgt_len, src_len)
-            if not return_attn:
-                attn = softmax_dropout(
-                    attn_weights,
-                    self.dropout,
-                    self.training,
-                    mask=mask,
-                    bias=attn_bias,
-                )
-            else:
-                attn_weights = attn_weights + (mask if mask is not None else 0)
+            # materialized O(L^2) chain: bmm -> fused softmax(+bias+mask
+            # +dropout) -> bmm
+            scores = torch.bmm(q, k.transpose(1, 2))
+            assert scores.shape == (bsz * self.num_heads, tgt_len, src_len)
+            scores = scores.view(bsz, self.num_heads, tgt_len, src_len)
+            if return_attn:
+                attn_weights = scores + (0 if mask is None else mask)
                 if attn_bias is not None:
                     attn_weights = attn_weights + attn_bias
                 attn = softmax_dropout(
                     attn_weights, self.dropout, self.training, inplace=False
                 )
-
+            else:
+                attn = softmax_dropout(
+                    scores, self.dropout, self.training,
+                    mask=mask, bias=attn_bias,
+                )
             attn = attn.view(bsz * self.num_heads, tgt_len, src_len)
-            o = torch.bmm(attn, v)
-        assert list(o.size()) == [bsz * self.num_heads, tgt_len, self.head_dim]
+            attn_out = torch.bmm(attn, v)
+        else:
+            attn_out = o
+        assert attn_out.shape == (bsz * self.num_heads, tgt_len, self.head_dim)
 
         if use_fused_split:
-            o = _AttnMerge.apply(o, bsz, self.num_heads)
+            attn_out = _AttnMerge.apply(attn_out, bsz, self.num_heads)
         else:
-            o = (
-                o.view(bsz, self.num_heads, tgt_len, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz, tgt_len, embed_dim)
-            )
+            attn_out = _from_heads(attn_out, bsz, self.num_heads, tgt_len,
+                                   self.head_dim)
         if skip_out_bias:
-            o = F.linear(o, self.out_proj.weight)
+            attn_out = F.linear(attn_out, self.out_proj.weight)
         else:
-            o = self.out_proj(o)
-        if not return_attn:
-            return o
-        else:
-            return o, attn_weights, attn
+            attn_out = self.out_proj(attn_out)
+        if return_attn:
+            return attn_out, attn_weights, attn
+        return attn_out
 
 
 class CrossMultiheadAttention(nn.Module):
-    def __init__(
-        self,
-        embed_dim,
-        num_heads,
-        dropout=0.1,
-        bias=True,
-        scaling_factor=1,
-    ):
+    def __init__(self, embed_dim, num_heads, dropout=0.1, bias=True,
+                 scaling_factor=1):
         super().__init__()
-        self.embed_dim = embed_dim
-        self.num_heads = num_heads
-        self.dropout = dropout
-
+        self.embed_dim, self.num_heads = embed_dim, num_heads
+        self.dropout = float(dropout)
         self.head_dim = embed_dim // num_heads
-        assert (
-            self.head_dim * num_heads == self.embed_dim
-        ), "embed_dim must be divisible by num_heads"
+        assert self.head_dim * num_heads == embed_dim, \
+            "embed_dim must be divisible by num_heads"
         self.scaling = (self.head_dim * scaling_factor) ** -0.5
 
         self.q_proj = nn.Linear(embed_dim, embed_dim, bias=bias)
@@ -343,71 +312,30 @@ class CrossMultiheadAttention(nn.Module):
         self.v_proj = nn.Linear(embed_dim, embed_dim, bias=bias)
         self.out_proj = nn.Linear(embed_dim, embed_dim, bias=bias)
 
-    def forward(
-        self,
-        query,
-        key,
-        value,
-        key_padding_mask: Optional[Tensor] = None,
-        attn_bias: Optional[Tensor] = None,
-    ) -> Tensor:
-        bsz, tgt_len, embed_dim = query.size()
-        assert embed_dim == self.embed_dim
+    def forward(self, query, key, value,
+                key_padding_mask: Optional[Tensor] = None,
+                attn_bias: Optional[Tensor] = None) -> Tensor:
+        bsz, tgt_len, embed_dim = query.shape
+        assert embed_dim == self.embed_dim, "input width mismatch"
 
-        q = self.q_proj(query)
-        k = self.k_proj(key)
-        v = self.v_proj(value)
+        q = _to_heads(self.q_proj(query), bsz, self.num_heads, self.head_dim,
+                      self.scaling)
+        k = _to_heads(self.k_proj(key), bsz, self.num_heads, self.head_dim)
+        v = _to_heads(self.v_proj(value), bsz, self.num_heads, self.head_dim)
 
-        q = (
-            q.view(bsz, tgt_len, self.num_heads, self.head_dim)
-            .transpose(1, 2)
-            .contiguous()
-            .view(bsz * self.num_heads, -1, self.head_dim)
-            * self.scaling
-        )
-        if k is not None:
-            k = (
-                k.view(bsz, -1, self.num_heads, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz * self.num_heads, -1, self.head_dim)
-            )
-        if v is not None:
-            v = (
-                v.view(bsz, -1, self.num_heads, self.head_dim)
-                .transpose(1, 2)
-                .contiguous()
-                .view(bsz * self.num_heads, -1, self.head_dim)
-            )
+        src_len = k.shape[1]
+        scores = torch.bmm(q, k.transpose(1, 2))
+        assert scores.shape == (bsz * self.num_heads, tgt_len, src_len)
+        mask = _padding_to_additive(key_padding_mask, bsz, src_len,
+                                    scores.dtype)
 
-        assert k is not None
-        src_len = k.size(1)
-
-        attn_weights = torch.bmm(q, k.transpose(1, 2))
-
-        assert list(attn_weights.size()) == [bsz * self.num_heads, tgt_len, src_len]
-
-        mask = None
-        if key_padding_mask is not None and key_padding_mask.dim() == 0:
-            key_padding_mask = None
-        if key_padding_mask is not None:
-            assert key_padding_mask.size(0) == bsz
-            assert key_padding_mask.size(1) == src_len
-            mask = key_padding_mask.view(bsz, 1, 1, src_len).to(attn_weights.dtype)
-
-        attn_weights = attn_weights.view(bsz, self.num_heads, tgt_len, src_len)
         attn = softmax_dropout(
-            attn_weights, self.dropout, self.training, mask=mask, bias=attn_bias
+            scores.view(bsz, self.num_heads, tgt_len, src_len),
+            self.dropout, self.training, mask=mask, bias=attn_bias,
         ).view(bsz * self.num_heads, tgt_len, src_len)
 
-        o = torch.bmm(attn, v)
-        assert list(o.size()) == [bsz * self.num_heads, tgt_len, self.head_dim]
-
-        o = (
-            o.view(bsz, self.num_heads, tgt_len, self.head_dim)
-            .transpose(1, 2)
-            .contiguous()
-            .view(bsz, tgt_len, embed_dim)
+        heads_out = torch.bmm(attn, v)
+        assert heads_out.shape == (bsz * self.num_heads, tgt_len, self.head_dim)
+        return self.out_proj(
+            _from_heads(heads_out, bsz, self.num_heads, tgt_len, self.head_dim)
         )
-        o = self.out_proj(o)
-        return o

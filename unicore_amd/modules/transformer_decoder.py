@@ -1,9 +1,9 @@
 """Transformer decoder stack (optionally auto-regressive, rel-pos bias).
 
 Functional parity with reference unicore/modules/transformer_decoder.py:25-180,
-with the same MI355X broadcast optimization as the encoder: rel-pos bias stays
-(1, H, q, k) and the causal/padding masks stay small; the fused softmax kernel
-broadcasts them.
+with the same MI355X broadcast optimization as the encoder: the rel-pos bias
+stays (1, H, q, k) and causal/padding masks stay small; the fused softmax
+kernel broadcasts them instead of materializing (B*H, q, k) tensors.
 """
 
 from typing import Optional
@@ -14,93 +14,82 @@ import torch.nn.functional as F
 
 from .layer_norm import LayerNorm
 from .transformer_decoder_layer import TransformerDecoderLayer
-from .transformer_encoder import relative_position_bucket
+from .transformer_encoder import additive_padding_mask, build_rel_pos_table
 
 
 def build_future_mask(seq_len):
-    return torch.triu(
-        torch.full([seq_len, seq_len], float("-inf")), 1
-    )
+    """Strictly-upper-triangular -inf mask (causal attention)."""
+    return torch.triu(torch.full([seq_len, seq_len], float("-inf")), 1)
 
 
 class TransformerDecoder(nn.Module):
     def __init__(
         self,
-        decoder_layers: int = 6,
-        embed_dim: int = 768,
-        ffn_embed_dim: int = 3072,
-        attention_heads: int = 8,
-        emb_dropout: float = 0.1,
-        dropout: float = 0.1,
-        attention_dropout: float = 0.1,
-        activation_dropout: float = 0.0,
-        max_seq_len: int = 256,
-        activation_fn: str = "gelu",
-        rel_pos: bool = True,
-        rel_pos_bins: int = 32,
-        max_rel_pos: int = 128,
-        post_ln: bool = False,
-        auto_regressive: bool = True,
-    ) -> None:
+        decoder_layers=6,
+        embed_dim=768,
+        ffn_embed_dim=3072,
+        attention_heads=8,
+        emb_dropout=0.1,
+        dropout=0.1,
+        attention_dropout=0.1,
+        activation_dropout=0.0,
+        max_seq_len=256,
+        activation_fn="gelu",
+        rel_pos=True,
+        rel_pos_bins=32,
+        max_rel_pos=128,
+        post_ln=False,
+        auto_regressive=True,
+    ):
         super().__init__()
-        self.emb_dropout = emb_dropout
-        self.max_seq_len = max_seq_len
-        self.embed_dim = embed_dim
-        self.attention_heads = attention_heads
-        self.emb_layer_norm = LayerNorm(self.embed_dim)
+        self.emb_dropout, self.max_seq_len = emb_dropout, max_seq_len
+        self.embed_dim, self.attention_heads = embed_dim, attention_heads
+        self.emb_layer_norm = LayerNorm(embed_dim)
         self.auto_regressive = auto_regressive
-        if self.auto_regressive:
-            self._future_mask = build_future_mask(self.max_seq_len)
-        else:
-            self._future_mask = None
-        if not post_ln:
-            self.final_layer_norm = LayerNorm(self.embed_dim)
-        else:
-            self.final_layer_norm = None
+        self._future_mask = (
+            build_future_mask(max_seq_len) if auto_regressive else None
+        )
+        self.final_layer_norm = None if post_ln else LayerNorm(embed_dim)
 
         self.layers = nn.ModuleList(
-            [
-                TransformerDecoderLayer(
-                    embed_dim=self.embed_dim,
-                    ffn_embed_dim=ffn_embed_dim,
-                    attention_heads=attention_heads,
-                    dropout=dropout,
-                    attention_dropout=attention_dropout,
-                    activation_dropout=activation_dropout,
-                    activation_fn=activation_fn,
-                    post_ln=post_ln,
-                )
-                for _ in range(decoder_layers)
-            ]
+            TransformerDecoderLayer(
+                embed_dim=embed_dim, ffn_embed_dim=ffn_embed_dim,
+                attention_heads=attention_heads, dropout=dropout,
+                attention_dropout=attention_dropout,
+                activation_dropout=activation_dropout,
+                activation_fn=activation_fn, post_ln=post_ln,
+            )
+            for _ in range(decoder_layers)
         )
 
         self.rel_pos = rel_pos
-        if self.rel_pos:
+        if rel_pos:
             assert rel_pos_bins % 2 == 0
-            self.rel_pos_bins = rel_pos_bins
-            self.max_rel_pos = max_rel_pos
+            self.rel_pos_bins, self.max_rel_pos = rel_pos_bins, max_rel_pos
             self.relative_attention_bias = nn.Embedding(
-                self.rel_pos_bins, self.attention_heads
+                rel_pos_bins, attention_heads
             )
-            seq_len = self.max_seq_len
-            context_position = torch.arange(seq_len, dtype=torch.long)[:, None]
-            memory_position = torch.arange(seq_len, dtype=torch.long)[None, :]
-            relative_position = memory_position - context_position
-            self.rp_bucket = relative_position_bucket(
-                relative_position,
-                num_buckets=self.rel_pos_bins,
-                max_distance=self.max_rel_pos,
+            self.rp_bucket = build_rel_pos_table(
+                max_seq_len, rel_pos_bins, max_rel_pos
             )
-            self.rp_bucket -= self.rp_bucket.min()
 
-    def get_rel_pos_bias(self, x):
-        if self.rp_bucket.device != x.device:
-            self.rp_bucket = self.rp_bucket.to(x.device)
-        seq_len = x.size(1)
-        rp_bucket = self.rp_bucket[:seq_len, :seq_len]
-        values = F.embedding(rp_bucket, self.relative_attention_bias.weight)
-        values = values.permute([2, 0, 1])
-        return values.contiguous()  # (H, q, k)
+    def get_rel_pos_bias(self, h):
+        if self.rp_bucket.device != h.device:
+            self.rp_bucket = self.rp_bucket.to(h.device)
+        L = h.size(1)
+        per_pair = F.embedding(
+            self.rp_bucket[:L, :L], self.relative_attention_bias.weight
+        )
+        return per_pair.permute([2, 0, 1]).contiguous()  # (H, q, k)
+
+    def _causal_bias(self, h, seq_len):
+        if self._future_mask.device != h.device:
+            self._future_mask = self._future_mask.to(h.device)
+        if self._future_mask.dtype != h.dtype:
+            self._future_mask = self._future_mask.type_as(h)
+        return self._future_mask[:seq_len, :seq_len].view(
+            1, 1, seq_len, seq_len
+        )
 
     def forward(
         self,
@@ -112,57 +101,36 @@ class TransformerDecoder(nn.Module):
         encoder_attn_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         bsz, seq_len = emb.size(0), emb.size(1)
-        x = self.emb_layer_norm(emb)
-        x = F.dropout(x, p=self.emb_dropout, training=self.training)
+        h = self.emb_layer_norm(emb)
+        h = F.dropout(h, p=self.emb_dropout, training=self.training)
 
         if padding_mask is not None:
-            x = x * (1 - padding_mask.unsqueeze(-1).type_as(x))
+            h = h * (1 - padding_mask.unsqueeze(-1).type_as(h))
 
         attn_bias = None
         if self.rel_pos:
-            attn_bias = self.get_rel_pos_bias(x).unsqueeze(0)  # (1, H, q, k)
+            attn_bias = self.get_rel_pos_bias(h).unsqueeze(0)  # (1, H, q, k)
         if self.auto_regressive:
-            if self._future_mask.device != x.device:
-                self._future_mask = self._future_mask.to(x.device)
-            if self._future_mask.dtype != x.dtype:
-                self._future_mask = self._future_mask.type_as(x)
-            causal = self._future_mask[:seq_len, :seq_len].view(1, 1, seq_len, seq_len)
+            causal = self._causal_bias(h, seq_len)
             attn_bias = causal if attn_bias is None else attn_bias + causal
         if attn_mask is not None:
+            # user-provided additive mask, reference API shape (B*H, q, k)
             attn_mask = attn_mask.view(bsz, -1, seq_len, seq_len)
             attn_bias = attn_mask if attn_bias is None else attn_mask + attn_bias
 
-        additive_pad = None
-        if padding_mask is not None:
-            additive_pad = torch.zeros(
-                (bsz, 1, 1, seq_len), dtype=x.dtype, device=x.device
-            )
-            additive_pad.masked_fill_(
-                padding_mask.view(bsz, 1, 1, seq_len).to(torch.bool), float("-inf")
-            )
-
-        additive_enc_pad = None
-        if encoder_padding_mask is not None:
-            enc_len = encoder_padding_mask.size(-1)
-            additive_enc_pad = torch.zeros(
-                (bsz, 1, 1, enc_len), dtype=x.dtype, device=x.device
-            )
-            additive_enc_pad.masked_fill_(
-                encoder_padding_mask.view(bsz, 1, 1, enc_len).to(torch.bool),
-                float("-inf"),
-            )
+        pad_bias = additive_padding_mask(padding_mask, h)
+        enc_pad_bias = additive_padding_mask(encoder_padding_mask, h)
 
         for layer in self.layers:
-            x = layer(
-                x,
+            h = layer(
+                h,
                 encoder_out=encoder_out,
-                padding_mask=additive_pad,
+                padding_mask=pad_bias,
                 attn_bias=attn_bias,
-                encoder_padding_mask=additive_enc_pad,
+                encoder_padding_mask=enc_pad_bias,
                 encoder_attn_bias=encoder_attn_mask,
             )
 
         if self.final_layer_norm is not None:
-            x = self.final_layer_norm(x)
-
-        return x
+            h = self.final_layer_norm(h)
+        return h

@@ -1,6 +1,8 @@
-"""Transformer decoder layer (self-attn + optional cross-attn + FFN).
+"""Transformer decoder layer (self-attn, optional cross-attn, FFN).
 
-Functional parity with reference unicore/modules/transformer_decoder_layer.py:15-120.
+Functional parity with reference
+unicore/modules/transformer_decoder_layer.py:15-120; pre/post-LN ordering
+per ``post_ln``.
 """
 
 from typing import Optional
@@ -18,43 +20,38 @@ from .multihead_attention import CrossMultiheadAttention, SelfMultiheadAttention
 class TransformerDecoderLayer(nn.Module):
     def __init__(
         self,
-        embed_dim: int = 768,
-        ffn_embed_dim: int = 3072,
-        attention_heads: int = 8,
-        dropout: float = 0.1,
-        attention_dropout: float = 0.1,
-        activation_dropout: float = 0.0,
-        activation_fn: str = "gelu",
+        embed_dim=768,
+        ffn_embed_dim=3072,
+        attention_heads=8,
+        dropout=0.1,
+        attention_dropout=0.1,
+        activation_dropout=0.0,
+        activation_fn="gelu",
         post_ln=False,
-    ) -> None:
+    ):
         super().__init__()
-
-        self.embed_dim = embed_dim
-        self.attention_heads = attention_heads
-        self.attention_dropout = attention_dropout
-
-        self.dropout = dropout
+        self.embed_dim, self.attention_heads = embed_dim, attention_heads
+        self.dropout, self.attention_dropout = dropout, attention_dropout
         self.activation_dropout = activation_dropout
-        self.activation_fn = utils.get_activation_fn(activation_fn)
+        self.act = utils.get_activation_fn(activation_fn)
 
         self.self_attn = SelfMultiheadAttention(
-            self.embed_dim,
-            attention_heads,
-            dropout=attention_dropout,
+            embed_dim, attention_heads, dropout=attention_dropout
         )
-        self.self_attn_layer_norm = LayerNorm(self.embed_dim)
-
+        self.self_attn_layer_norm = LayerNorm(embed_dim)
         self.encoder_attn = CrossMultiheadAttention(
-            self.embed_dim,
-            attention_heads,
-            dropout=attention_dropout,
+            embed_dim, attention_heads, dropout=attention_dropout
         )
-        self.encoder_attn_layer_norm = LayerNorm(self.embed_dim)
-
-        self.fc1 = nn.Linear(self.embed_dim, ffn_embed_dim)
-        self.fc2 = nn.Linear(ffn_embed_dim, self.embed_dim)
-        self.final_layer_norm = LayerNorm(self.embed_dim)
+        self.encoder_attn_layer_norm = LayerNorm(embed_dim)
+        self.fc1 = nn.Linear(embed_dim, ffn_embed_dim)
+        self.fc2 = nn.Linear(ffn_embed_dim, embed_dim)
+        self.final_layer_norm = LayerNorm(embed_dim)
         self.post_ln = post_ln
+
+    def _join(self, h, skip):
+        """Residual join with dropout on the branch output."""
+        h = F.dropout(h, p=self.dropout, training=self.training)
+        return skip + h
 
     def forward(
         self,
@@ -65,44 +62,34 @@ class TransformerDecoderLayer(nn.Module):
         encoder_attn_bias: Optional[torch.Tensor] = None,
         encoder_padding_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        residual = x
-        if not self.post_ln:
-            x = self.self_attn_layer_norm(x)
-        x = self.self_attn(
-            query=x,
-            key_padding_mask=padding_mask,
-            attn_bias=attn_bias,
+        skip = x
+        h = x if self.post_ln else self.self_attn_layer_norm(x)
+        h = self.self_attn(
+            query=h, key_padding_mask=padding_mask, attn_bias=attn_bias
         )
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        x = residual + x
+        h = self._join(h, skip)
         if self.post_ln:
-            x = self.self_attn_layer_norm(x)
+            h = self.self_attn_layer_norm(h)
 
         if encoder_out is not None:
-            residual = x
+            skip = h
             if not self.post_ln:
-                x = self.encoder_attn_layer_norm(x)
-            x = self.encoder_attn(
-                query=x,
-                key=encoder_out,
-                value=encoder_out,
+                h = self.encoder_attn_layer_norm(h)
+            h = self.encoder_attn(
+                query=h, key=encoder_out, value=encoder_out,
                 key_padding_mask=encoder_padding_mask,
                 attn_bias=encoder_attn_bias,
             )
-            x = F.dropout(x, p=self.dropout, training=self.training)
-            x = residual + x
+            h = self._join(h, skip)
             if self.post_ln:
-                x = self.encoder_attn_layer_norm(x)
+                h = self.encoder_attn_layer_norm(h)
 
-        residual = x
+        skip = h
         if not self.post_ln:
-            x = self.final_layer_norm(x)
-        x = self.fc1(x)
-        x = self.activation_fn(x)
-        x = F.dropout(x, p=self.activation_dropout, training=self.training)
-        x = self.fc2(x)
-        x = F.dropout(x, p=self.dropout, training=self.training)
-        x = residual + x
+            h = self.final_layer_norm(h)
+        h = F.dropout(self.act(self.fc1(h)), p=self.activation_dropout,
+                      training=self.training)
+        h = self._join(self.fc2(h), skip)
         if self.post_ln:
-            x = self.final_layer_norm(x)
-        return x
+            h = self.final_layer_norm(h)
+        return h

@@ -1,11 +1,15 @@
-"""Transformer encoder layer (post-LN) with fused LayerNorm + softmax_dropout.
+"""Transformer encoder layer built around the fused kernel set.
 
-Functional parity with reference unicore/modules/transformer_encoder_layer.py:56-98.
+Functional parity with reference
+unicore/modules/transformer_encoder_layer.py:56-98 (pre/post-LN BERT/XLM
+layer). On GPU with the extension loaded, the three Linears run bias-free
+and each bias rides the following fused elementwise op (dropout+residual /
+gelu+dropout), whose backward emits the bias grad as a deterministic
+column sum — no eager activation-sized ``.sum`` re-reads.
 """
 
-from typing import Optional
-
 import os
+from typing import Optional
 
 import torch.nn.functional as F
 from torch import Tensor, nn
@@ -19,45 +23,45 @@ from .multihead_attention import SelfMultiheadAttention
 
 
 class TransformerEncoderLayer(nn.Module):
-    """
-    Implements a Transformer Encoder Layer used in BERT/XLM style pre-trained
-    models.
-    """
-
     def __init__(
         self,
-        embed_dim: int = 768,
-        ffn_embed_dim: int = 3072,
-        attention_heads: int = 8,
-        dropout: float = 0.1,
-        attention_dropout: float = 0.1,
-        activation_dropout: float = 0.0,
-        activation_fn: str = "gelu",
+        embed_dim=768,
+        ffn_embed_dim=3072,
+        attention_heads=8,
+        dropout=0.1,
+        attention_dropout=0.1,
+        activation_dropout=0.0,
+        activation_fn="gelu",
         post_ln=False,
-    ) -> None:
+    ):
         super().__init__()
-
-        # Initialize parameters
-        self.embed_dim = embed_dim
-        self.attention_heads = attention_heads
-        self.attention_dropout = attention_dropout
-
-        self.dropout = dropout
+        self.embed_dim, self.attention_heads = embed_dim, attention_heads
+        self.dropout, self.attention_dropout = dropout, attention_dropout
         self.activation_dropout = activation_dropout
-        self.activation_fn = utils.get_activation_fn(activation_fn)
-        self._fuse_gelu = activation_fn == "gelu" 
+        self.act = utils.get_activation_fn(activation_fn)
+        self._fuse_gelu = activation_fn == "gelu"
 
         self.self_attn = SelfMultiheadAttention(
-            self.embed_dim,
-            attention_heads,
-            dropout=attention_dropout,
+            embed_dim, attention_heads, dropout=attention_dropout
         )
-        # layer norm associated with the self attention layer
-        self.self_attn_layer_norm = LayerNorm(self.embed_dim)
-        self.fc1 = nn.Linear(self.embed_dim, ffn_embed_dim)
-        self.fc2 = nn.Linear(ffn_embed_dim, self.embed_dim)
-        self.final_layer_norm = LayerNorm(self.embed_dim)
+        self.self_attn_layer_norm = LayerNorm(embed_dim)
+        self.fc1 = nn.Linear(embed_dim, ffn_embed_dim)
+        self.fc2 = nn.Linear(ffn_embed_dim, embed_dim)
+        self.final_layer_norm = LayerNorm(embed_dim)
         self.post_ln = post_ln
+
+    def _can_fold_biases(self, h):
+        """True when every Linear bias in this layer can ride a fused op."""
+        if not h.is_cuda or os.environ.get("UNICORE_FOLD_BIAS", "1") != "1":
+            return False
+        from unicore_amd import ops
+
+        if not (ops.gpu_kernels_available() and self._fuse_gelu):
+            return False
+        biases = (self.self_attn.out_proj.bias, self.fc1.bias, self.fc2.bias)
+        return all(
+            b is not None and ops.colsum_supported(b.numel()) for b in biases
+        )
 
     def forward(
         self,
@@ -66,71 +70,49 @@ class TransformerEncoderLayer(nn.Module):
         padding_mask: Optional[Tensor] = None,
         return_attn: bool = False,
     ) -> Tensor:
-        """
-        LayerNorm is applied either before or after the self-attention/ffn
-        modules similar to the original Transformer implementation.
-        """
-        # On GPU with the kernel extension, the Linears run bias-free and
-        # each bias rides the following fused op (add is free there, and
-        # the backward emits the bias grad as a deterministic column sum
-        # instead of an eager activation-sized .sum re-read per Linear).
-        fold_bias = False
-        if x.is_cuda and os.environ.get("UNICORE_FOLD_BIAS", "1") == "1":
-            from unicore_amd import ops
+        """Pre-LN or post-LN ordering per ``self.post_ln``."""
+        fold = self._can_fold_biases(x)
 
-            fold_bias = (
-                ops.gpu_kernels_available()
-                and self.self_attn.out_proj.bias is not None
-                and ops.colsum_supported(self.self_attn.out_proj.bias.numel())
-                and self.fc1.bias is not None
-                and ops.colsum_supported(self.fc1.bias.numel())
-                and self.fc2.bias is not None
-                and ops.colsum_supported(self.fc2.bias.numel())
-                and self._fuse_gelu
-            )
-
-        residual = x
-        if not self.post_ln:
-            x = self.self_attn_layer_norm(x)
-        x = self.self_attn(
-            query=x,
+        skip = x
+        h = x if self.post_ln else self.self_attn_layer_norm(x)
+        h = self.self_attn(
+            query=h,
             key_padding_mask=padding_mask,
             attn_bias=attn_bias,
             return_attn=return_attn,
-            skip_out_bias=fold_bias,
+            skip_out_bias=fold,
         )
         if return_attn:
-            x, attn_weights, attn_probs = x
-        x = dropout_add(
-            x, residual, self.dropout, self.training,
-            bias=self.self_attn.out_proj.bias if fold_bias else None,
+            h, attn_weights, attn_probs = h
+        h = dropout_add(
+            h, skip, self.dropout, self.training,
+            bias=self.self_attn.out_proj.bias if fold else None,
         )
         if self.post_ln:
-            x = self.self_attn_layer_norm(x)
+            h = self.self_attn_layer_norm(h)
 
-        residual = x
+        skip = h
         if not self.post_ln:
-            x = self.final_layer_norm(x)
-        if fold_bias:
-            x = F.linear(x, self.fc1.weight)
-            x = gelu_dropout(x, self.activation_dropout, self.training,
+            h = self.final_layer_norm(h)
+        if fold:
+            h = F.linear(h, self.fc1.weight)
+            h = gelu_dropout(h, self.activation_dropout, self.training,
                              bias=self.fc1.bias)
-            x = F.linear(x, self.fc2.weight)
-            x = dropout_add(x, residual, self.dropout, self.training,
+            h = F.linear(h, self.fc2.weight)
+            h = dropout_add(h, skip, self.dropout, self.training,
                             bias=self.fc2.bias)
         else:
-            x = self.fc1(x)
-            if self._fuse_gelu and x.is_cuda:
-                x = gelu_dropout(x, self.activation_dropout, self.training)
+            h = self.fc1(h)
+            if self._fuse_gelu and h.is_cuda:
+                h = gelu_dropout(h, self.activation_dropout, self.training)
             else:
-                x = self.activation_fn(x)
-                x = F.dropout(x, p=self.activation_dropout,
+                h = F.dropout(self.act(h), p=self.activation_dropout,
                               training=self.training)
-            x = self.fc2(x)
-            x = dropout_add(x, residual, self.dropout, self.training)
+            h = self.fc2(h)
+            h = dropout_add(h, skip, self.dropout, self.training)
         if self.post_ln:
-            x = self.final_layer_norm(x)
-        if not return_attn:
-            return x
-        else:
-            return x, attn_weights, attn_probs
+            h = self.final_layer_norm(h)
+
+        if return_attn:
+            return h, attn_weights, attn_probs
+        return h
