@@ -1,4 +1,9 @@
-"""Task base class (parity: reference unicore/tasks/unicore_task.py:20-329)."""
+"""Task base class (parity: reference unicore/tasks/unicore_task.py:20-329).
+
+A task owns the dictionaries and datasets, builds the model/loss from the
+registries, constructs the resumable batch iterator, and hosts the per-step
+forward/backward delegation the trainer calls into.
+"""
 
 import logging
 import os
@@ -12,13 +17,24 @@ from unicore_amd.logging import metrics
 
 logger = logging.getLogger(__name__)
 
+_warned_once = set()
 
-class StatefulContainer(object):
-    """Checkpointable task state (reference unicore/tasks/unicore_task.py:20-42)."""
+
+def warnings_once(msg):
+    """Log each distinct warning message a single time per process."""
+    if msg not in _warned_once:
+        _warned_once.add(msg)
+        logger.warning(msg)
+
+
+class StatefulContainer:
+    """Lazily-materialized, checkpointable task state
+    (reference unicore/tasks/unicore_task.py:20-42). Attributes resolve
+    against stored state first, then registered factories."""
 
     def __init__(self):
-        self._state = dict()
-        self._factories = dict()
+        self._state = {}
+        self._factories = {}
 
     def add_factory(self, name, factory: Callable[[], Any]):
         self._factories[name] = factory
@@ -38,67 +54,49 @@ class StatefulContainer(object):
         raise AttributeError(f"Task state has no factory for attribute {name}")
 
 
-class UnicoreTask(object):
-    """
-    Tasks store dictionaries and provide helpers for loading/iterating over
-    Datasets, initializing the Model/Loss and calculating the loss.
-    """
+class UnicoreTask:
+    """Owns datasets + dictionaries; builds models/losses; drives steps."""
 
     @classmethod
     def add_args(cls, parser):
-        """Add task-specific arguments to the parser."""
-        pass
+        """Hook for task-specific CLI arguments."""
 
     @staticmethod
     def logging_outputs_can_be_summed(loss, is_train) -> bool:
-        """
-        Whether the logging outputs returned by `train_step` and `valid_step` can
-        be summed across workers prior to calling `aggregate_logging_outputs`.
-        Setting this to True will improve distributed training speed.
-        """
+        """Forwarded from the loss: True enables the fast all-reduce stat
+        sync across ranks."""
         return loss.logging_outputs_can_be_summed(is_train)
 
     def __init__(self, args: Namespace, **kwargs):
         self.args = args
-        self.datasets = dict()
-        self.dataset_to_epoch_iter = dict()
+        self.datasets = {}
+        self.dataset_to_epoch_iter = {}
         self.state = StatefulContainer()
 
     @classmethod
     def setup_task(cls, args: Namespace, **kwargs):
-        """Setup the task (e.g., load dictionaries)."""
+        """Factory hook (load dictionaries etc.) before construction."""
         return cls(args, **kwargs)
 
     def has_sharded_data(self, split):
         return os.pathsep in getattr(self.args, "data", "")
 
-    def load_dataset(
-        self,
-        split: str,
-        combine: bool = False,
-        **kwargs,
-    ):
-        """Load a given dataset split.
-
-        Args:
-            split (str): name of the split (e.g., train, valid, test)
-        """
-        raise NotImplementedError
+    def load_dataset(self, split: str, combine: bool = False, **kwargs):
+        """Load one named split (train/valid/test) into self.datasets."""
+        raise NotImplementedError("tasks implement load_dataset")
 
     def dataset(self, split):
-        """Return a loaded dataset split."""
-        from unicore_amd.data import UnicoreDataset
-
+        """The loaded dataset for *split* (must be a UnicoreDataset)."""
         if split not in self.datasets:
             raise KeyError("Dataset not loaded: " + split)
-        if not isinstance(self.datasets[split], UnicoreDataset):
+        loaded = self.datasets[split]
+        if not isinstance(loaded, UnicoreDataset):
             raise TypeError("Datasets are expected to be of type UnicoreDataset")
-        return self.datasets[split]
+        return loaded
 
     def can_reuse_epoch_itr(self, dataset):
-        # We can reuse the epoch iterator across epochs as long as the dataset
-        # hasn't disabled it. We default to ``False`` here, although in practice
-        # this will be ``True`` for most datasets that don't use noising.
+        """Datasets whose item sizes are epoch-stable opt in via the
+        can_reuse_epoch_itr_across_epochs property."""
         return getattr(dataset, "can_reuse_epoch_itr_across_epochs", False)
 
     def get_batch_iterator(
@@ -115,38 +113,32 @@ class UnicoreTask(object):
         data_buffer_size=0,
         disable_iterator_cache=False,
     ):
-        """
-        Get an iterator that yields batches of data from the given dataset.
-        (reference unicore/tasks/unicore_task.py:138-225)
-        """
-        can_reuse_epoch_itr = not disable_iterator_cache and self.can_reuse_epoch_itr(
-            dataset
+        """Build (or reuse) the epoch iterator over *dataset*
+        (reference unicore/tasks/unicore_task.py:138-225): ordered indices
+        under the seed, fixed-size batches, then a sharded resumable
+        EpochBatchIterator."""
+        reusable = (
+            not disable_iterator_cache and self.can_reuse_epoch_itr(dataset)
         )
-        if can_reuse_epoch_itr and dataset in self.dataset_to_epoch_iter:
-            logger.debug("reusing EpochBatchIterator for epoch {}".format(epoch))
+        if reusable and dataset in self.dataset_to_epoch_iter:
+            logger.debug(f"reusing EpochBatchIterator for epoch {epoch}")
             return self.dataset_to_epoch_iter[dataset]
 
         assert isinstance(dataset, UnicoreDataset)
+        dataset.set_epoch(epoch)  # epoch-aware noising starts correctly
 
-        # initialize the dataset with the correct starting epoch
-        dataset.set_epoch(epoch)
-
-        # get indices ordered by example size
         with data_utils.numpy_seed(seed):
             indices = dataset.ordered_indices()
-
-        # create mini-batches with given size constraints
-        batch_sampler = dataset.batch_by_size(
+        batches = dataset.batch_by_size(
             indices,
             batch_size=batch_size,
             required_batch_size_multiple=required_batch_size_multiple,
         )
 
-        # return a reusable, sharded iterator
         epoch_iter = iterators.EpochBatchIterator(
             dataset=dataset,
             collate_fn=dataset.collater,
-            batch_sampler=batch_sampler,
+            batch_sampler=batches,
             seed=seed,
             num_shards=num_shards,
             shard_id=shard_id,
@@ -157,50 +149,30 @@ class UnicoreTask(object):
             # the GPU (the model stays on CPU)
             cuda_prefetch=not getattr(self.args, "cpu", False),
         )
-
-        if can_reuse_epoch_itr:
+        if reusable:
             self.dataset_to_epoch_iter[dataset] = epoch_iter
-
         return epoch_iter
 
     def build_model(self, args: Namespace):
-        """Build the :class:`~unicore_amd.models.BaseUnicoreModel` instance for
-        this task."""
+        """Instantiate this task's model via the model registry."""
         from unicore_amd import models
 
-        model = models.build_model(args, self)
-        return model
+        return models.build_model(args, self)
 
     def build_loss(self, args: Namespace):
-        """Build the :class:`~unicore_amd.losses.UnicoreLoss` instance for this
-        task."""
+        """Instantiate this task's loss via the loss registry."""
         from unicore_amd import losses
 
-        loss = losses.build_loss(args, self)
-        return loss
+        return losses.build_loss(args, self)
 
-    def train_step(
-        self, sample, model, loss, optimizer, update_num, ignore_grad=False
-    ):
-        """
-        Do forward and backward, and return the loss as computed by *loss*
-        for the given *model* and *sample*.
+    def train_step(self, sample, model, loss, optimizer, update_num,
+                   ignore_grad=False):
+        """One micro-batch: forward through *loss*, then backward through
+        the optimizer (which applies loss scaling). ``ignore_grad`` zeroes
+        the loss so a dummy batch contributes nothing.
 
-        Args:
-            sample (dict): the mini-batch. The format is defined by the
-                :class:`~unicore_amd.data.UnicoreDataset`.
-            model (~unicore_amd.models.BaseUnicoreModel): the model
-            loss (~unicore_amd.losses.UnicoreLoss): the loss
-            optimizer (~unicore_amd.optim.UnicoreOptimizer): the optimizer
-            update_num (int): the current update
-            ignore_grad (bool): multiply loss by 0 if this is set to True
-
-        Returns:
-            tuple:
-                - the loss
-                - the sample size, which is used as the denominator for the
-                  gradient
-                - logging outputs to display while training
+        Returns ``(loss, sample_size, logging_output)`` with sample_size
+        the gradient denominator.
         """
         model.train()
         model.set_num_updates(update_num)
@@ -215,28 +187,24 @@ class UnicoreTask(object):
     def valid_step(self, sample, model, loss, test=False):
         model.eval()
         with torch.no_grad():
-            loss_value, sample_size, logging_output = loss(model, sample)
-        return loss_value, sample_size, logging_output
+            return loss(model, sample)
 
     def optimizer_step(self, optimizer, model, update_num):
         optimizer.step()
 
     def reduce_metrics(self, logging_outputs, loss, split="train"):
-        """Aggregate logging outputs from data parallel training."""
-        if not any("bsz" in log for log in logging_outputs):
+        """Fold per-rank logging outputs into the metrics aggregators."""
+        if any("bsz" in log for log in logging_outputs):
+            bsz = sum(log.get("bsz", 0) for log in logging_outputs)
+            metrics.log_scalar("bsz", bsz, priority=190, round=1)
+        else:
             warnings_once(
                 "bsz not found in Loss logging outputs, cannot log bsz"
             )
-        else:
-            bsz = sum(log.get("bsz", 0) for log in logging_outputs)
-            metrics.log_scalar("bsz", bsz, priority=190, round=1)
-
         loss.__class__.reduce_metrics(logging_outputs, split)
 
-    def state_dict(self):
-        if self.state is not None:
-            return self.state.state_dict
-        return {}
+    def state_dict(self) -> Dict[str, Any]:
+        return self.state.state_dict if self.state is not None else {}
 
     def load_state_dict(self, state_dict: Dict[str, Any]):
         if self.state is not None:
@@ -246,18 +214,7 @@ class UnicoreTask(object):
         return False
 
     def begin_epoch(self, epoch, model):
-        """Hook function called before the start of each epoch."""
-        pass
+        """Hook invoked as each training epoch starts."""
 
     def begin_valid_epoch(self, epoch, model):
-        """Hook function called before the start of each validation epoch."""
-        pass
-
-
-_warned = set()
-
-
-def warnings_once(msg):
-    if msg not in _warned:
-        _warned.add(msg)
-        logger.warning(msg)
+        """Hook invoked as each validation pass starts."""

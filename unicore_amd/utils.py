@@ -18,47 +18,49 @@ import torch
 import torch.nn.functional as F
 
 
+# --------------------------------------------------------------------------
+# Sample-tree traversal and device movement
+# --------------------------------------------------------------------------
+
+
 def apply_to_sample(f, sample):
+    """Apply *f* to every tensor in a nested dict/list/tuple/set sample."""
     if hasattr(sample, "__len__") and len(sample) == 0:
         return {}
 
-    def _apply(x):
-        if torch.is_tensor(x):
-            return f(x)
-        elif isinstance(x, dict):
-            return {key: _apply(value) for key, value in x.items()}
-        elif isinstance(x, list):
-            return [_apply(x) for x in x]
-        elif isinstance(x, tuple):
-            return tuple(_apply(x) for x in x)
-        elif isinstance(x, set):
-            return {_apply(x) for x in x}
-        else:
-            return x
+    def visit(node):
+        if torch.is_tensor(node):
+            return f(node)
+        if isinstance(node, dict):
+            return {k: visit(v) for k, v in node.items()}
+        if isinstance(node, list):
+            return [visit(v) for v in node]
+        if isinstance(node, tuple):
+            return tuple(visit(v) for v in node)
+        if isinstance(node, set):
+            return {visit(v) for v in node}
+        return node
 
-    return _apply(sample)
+    return visit(sample)
 
 
 def move_to_cuda(sample, device=None):
-    device = device or torch.cuda.current_device()
-
-    def _move_to_cuda(tensor):
-        # non_blocking is ignored if tensor is not pinned, so we can always set
-        # to True (H2D copies overlap with compute when the source is pinned;
-        # the buffered loader pins its batches)
-        return tensor.to(device=device, non_blocking=True)
-
-    return apply_to_sample(_move_to_cuda, sample)
+    target = device if device is not None else torch.cuda.current_device()
+    # non_blocking is a no-op for unpinned sources, so always request it:
+    # pinned batches (the buffered loader pins) overlap H2D with compute
+    return apply_to_sample(
+        lambda t: t.to(device=target, non_blocking=True), sample
+    )
 
 
 def move_to_cpu(sample):
-    def _move_to_cpu(tensor):
-        # PyTorch has poor support for half tensors (float16) on CPU.
-        if tensor.dtype in {torch.bfloat16, torch.float16}:
-            tensor = tensor.to(dtype=torch.float32)
-        return tensor.cpu()
+    def to_host(t):
+        # half-precision tensors have poor CPU support; widen first
+        if t.dtype in (torch.bfloat16, torch.float16):
+            t = t.to(dtype=torch.float32)
+        return t.cpu()
 
-    return apply_to_sample(_move_to_cpu, sample)
+    return apply_to_sample(to_host, sample)
 
 
 # --------------------------------------------------------------------------
@@ -67,45 +69,47 @@ def move_to_cpu(sample):
 
 
 def multi_tensor_total_norm(grads, chunk_size=2048 * 64) -> torch.Tensor:
-    """L2 norm over a list of gradients, grouped per (device, dtype).
+    """Global L2 norm over a gradient list, grouped per (device, dtype).
 
-    On a GPU with our extension loaded this calls the fused multi-tensor
-    kernel (one kernel per group); otherwise it falls back to
-    torch._foreach_norm which is itself a fused multi-tensor path.
+    Each CUDA group goes through the fused multi-tensor kernel (one launch
+    per group) when the extension is loaded; other groups use per-tensor
+    torch.norm.
     """
-    per_device_grads = {}
-    for grad in grads:
-        device = grad.device
-        cur_device_grads = per_device_grads.setdefault(device, {})
-        dtype = grad.dtype
-        cur_device_grads.setdefault(dtype, []).append(grad)
-    norms = []
-    for device, per_dtype_grads in per_device_grads.items():
-        for grads_group in per_dtype_grads.values():
+    by_device = {}
+    for g in grads:
+        by_device.setdefault(g.device, {}).setdefault(g.dtype, []).append(g)
+
+    partials = []
+    for device, by_dtype in by_device.items():
+        for group in by_dtype.values():
             if device.type == "cuda":
                 from .ops import fused_l2norm, has_kernels
 
                 if has_kernels():
-                    norms.append(fused_l2norm(grads_group, chunk_size).to(device))
+                    partials.append(
+                        fused_l2norm(group, chunk_size).to(device)
+                    )
                     continue
-            norms += [torch.norm(g, p=2, dtype=torch.float32) for g in grads_group]
-    total_norm = torch.norm(torch.stack(norms))
-    return total_norm
+            partials += [
+                torch.norm(g, p=2, dtype=torch.float32) for g in group
+            ]
+    return torch.norm(torch.stack(partials))
 
 
 def clip_grad_norm_(params, max_norm, aggregate_norm_fn=None) -> torch.Tensor:
-    def grad_exists(p):
-        return p is not None and getattr(p, "grad", None) is not None
-
+    """Global-norm gradient clipping; returns the pre-clip norm."""
     if isinstance(params, torch.Tensor):
         params = [params]
     params = list(params)
-    grads = [p.grad.detach() for p in filter(grad_exists, params)]
-    if len(grads) == 0:
-        if len(params) > 0:
-            return params[0].new_tensor(0.0)
-        else:
-            return torch.tensor(0.0)
+    grads = [
+        p.grad.detach()
+        for p in params
+        if p is not None and getattr(p, "grad", None) is not None
+    ]
+    if not grads:
+        return (
+            params[0].new_tensor(0.0) if params else torch.tensor(0.0)
+        )
 
     if len(grads) == 1:
         total_norm = torch.norm(grads[0], p=2, dtype=torch.float32)
@@ -116,9 +120,8 @@ def clip_grad_norm_(params, max_norm, aggregate_norm_fn=None) -> torch.Tensor:
         total_norm = aggregate_norm_fn(total_norm)
 
     if max_norm > 0:
-        max_norm = float(max_norm)
-        clip_coef = (max_norm / (total_norm + 1e-6)).clamp_(max=1)
-        torch._foreach_mul_(grads, clip_coef)
+        scale = (float(max_norm) / (total_norm + 1e-6)).clamp_(max=1)
+        torch._foreach_mul_(grads, scale)
     return total_norm
 
 
@@ -128,19 +131,20 @@ def clip_grad_norm_(params, max_norm, aggregate_norm_fn=None) -> torch.Tensor:
 
 
 def import_user_module(args):
-    module_path = getattr(args, "user_dir", None)
-    if module_path is None:
+    """Import the --user-dir package so its @register_* decorators run."""
+    requested = getattr(args, "user_dir", None)
+    if requested is None:
         return
-    module_path = os.path.abspath(args.user_dir)
-    if not os.path.exists(module_path):
-        unicore_rel_path = os.path.join(os.path.dirname(__file__), args.user_dir)
-        if os.path.exists(unicore_rel_path):
-            module_path = unicore_rel_path
-    module_parent, module_name = os.path.split(module_path)
-
-    if module_name not in sys.modules:
-        sys.path.insert(0, module_parent)
-        importlib.import_module(module_name)
+    path = os.path.abspath(requested)
+    if not os.path.exists(path):
+        # also try a path relative to the installed package
+        in_pkg = os.path.join(os.path.dirname(__file__), requested)
+        if os.path.exists(in_pkg):
+            path = in_pkg
+    parent, name = os.path.split(path)
+    if name not in sys.modules:
+        sys.path.insert(0, parent)
+        importlib.import_module(name)
         sys.path.pop(0)
 
 
@@ -148,18 +152,19 @@ def import_user_module(args):
 # Activations
 # --------------------------------------------------------------------------
 
+_ACTIVATION_TABLE = {
+    "relu": F.relu,
+    "gelu": F.gelu,
+    "tanh": torch.tanh,
+    "linear": lambda x: x,
+}
+
 
 def get_activation_fn(activation: str) -> Callable:
-    """Returns the activation function corresponding to `activation`"""
-    if activation == "relu":
-        return F.relu
-    elif activation == "gelu":
-        return F.gelu
-    elif activation == "tanh":
-        return torch.tanh
-    elif activation == "linear":
-        return lambda x: x
-    else:
+    """Activation callable for a name in {relu, gelu, tanh, linear}."""
+    try:
+        return _ACTIVATION_TABLE[activation]
+    except KeyError:
         raise NotImplementedError(f"activation {activation} not supported")
 
 
@@ -185,47 +190,27 @@ def torch_seed(seed, *args):
     seed = int(seed)
     for arg in args:
         seed = int(hash((seed, int(arg)))) % int(1e8)
-    state = torch.random.get_rng_state()
-    cuda_state = None
+    cpu_state = torch.random.get_rng_state()
+    gpu_state = None
     use_cuda = torch.cuda.is_available() and torch.cuda.is_initialized()
     if use_cuda:
-        cuda_state = torch.cuda.random.get_rng_state()
+        gpu_state = torch.cuda.random.get_rng_state()
     torch.manual_seed(seed)
     if use_cuda:
         torch.cuda.manual_seed(seed)
     try:
         yield
     finally:
-        torch.random.set_rng_state(state)
-        if use_cuda and cuda_state is not None:
-            torch.cuda.random.set_rng_state(cuda_state)
-
-
-# --------------------------------------------------------------------------
-# Misc environment helpers
-# --------------------------------------------------------------------------
-
-
-def set_jit_fusion_options():
-    """Set PyTorch fusion options (no-op stub kept for CLI parity)."""
-    # We rely on our own fused HIP kernels + hipGraphs rather than the TorchScript
-    # fuser; nothing to configure here.
-    pass
-
-
-def has_parameters(module):
-    try:
-        next(module.parameters())
-        return True
-    except StopIteration:
-        return False
+        torch.random.set_rng_state(cpu_state)
+        if gpu_state is not None:
+            torch.cuda.random.set_rng_state(gpu_state)
 
 
 def get_rng_state():
-    state = {"torch_rng_state": torch.get_rng_state()}
+    snapshot = {"torch_rng_state": torch.get_rng_state()}
     if torch.cuda.is_available():
-        state["cuda_rng_state"] = torch.cuda.get_rng_state()
-    return state
+        snapshot["cuda_rng_state"] = torch.cuda.get_rng_state()
+    return snapshot
 
 
 def set_rng_state(state):
@@ -234,7 +219,9 @@ def set_rng_state(state):
         torch.cuda.set_rng_state(state["cuda_rng_state"])
 
 
-class set_torch_seed(object):
+class set_torch_seed:
+    """RAII variant of torch_seed with a bare int seed."""
+
     def __init__(self, seed):
         assert isinstance(seed, int)
         self.rng_state = get_rng_state()
@@ -249,32 +236,46 @@ class set_torch_seed(object):
         set_rng_state(self.rng_state)
 
 
-class CudaEnvironment(object):
+# --------------------------------------------------------------------------
+# Misc environment helpers
+# --------------------------------------------------------------------------
+
+
+def set_jit_fusion_options():
+    """No-op kept for CLI parity: the hot ops run through our own fused HIP
+    kernels (+ hipGraphs), not the TorchScript fuser."""
+
+
+def has_parameters(module):
+    return next(module.parameters(), None) is not None
+
+
+class CudaEnvironment:
+    """Device capability snapshot, gathered across ranks for the startup
+    banner."""
+
     def __init__(self):
-        cur_device = torch.cuda.current_device()
-        prop = torch.cuda.get_device_properties("cuda:{}".format(cur_device))
+        prop = torch.cuda.get_device_properties(
+            f"cuda:{torch.cuda.current_device()}"
+        )
         self.name = prop.name
-        self.major = prop.major
-        self.minor = prop.minor
-        self.total_memory_in_GB = prop.total_memory / 1024 / 1024 / 1024
+        self.major, self.minor = prop.major, prop.minor
+        self.total_memory_in_GB = prop.total_memory / 1024**3
 
     @staticmethod
     def pretty_print_cuda_env_list(cuda_env_list):
-        """Given a list of CudaEnviorments, pretty print them"""
-        num_workers = len(cuda_env_list)
-        center = "CUDA enviroments for all {} workers".format(num_workers)
-        banner_len = 40 - len(center) // 2
-        first_line = "*" * banner_len + center + "*" * banner_len
-        msgs = [first_line]
-        for r, env in enumerate(cuda_env_list):
-            msgs.append(
-                "rank {:3d}: ".format(r)
-                + "capabilities = {:2d}.{:<2d} ; ".format(env.major, env.minor)
-                + "total memory = {:.3f} GB ; ".format(env.total_memory_in_GB)
-                + "name = {:40s}".format(env.name)
+        center = f"CUDA enviroments for all {len(cuda_env_list)} workers"
+        pad = "*" * (40 - len(center) // 2)
+        lines = [pad + center + pad]
+        for rank, env in enumerate(cuda_env_list):
+            lines.append(
+                f"rank {rank:3d}: "
+                f"capabilities = {env.major:2d}.{env.minor:<2d} ; "
+                f"total memory = {env.total_memory_in_GB:.3f} GB ; "
+                f"name = {env.name:40s}"
             )
-        msgs.append("*" * (40 + len(center) + 40))
-        return "\n".join(msgs)
+        lines.append("*" * (40 + len(center) + 40))
+        return "\n".join(lines)
 
 
 # --------------------------------------------------------------------------
@@ -282,34 +283,26 @@ class CudaEnvironment(object):
 # --------------------------------------------------------------------------
 
 
-def checkpoint_sequential(
-    functions,
-    input,
-    enabled=True,
-):
-    def wrap_tuple(a):
-        return (a,) if type(a) is not tuple else a
+def checkpoint_sequential(functions, input, enabled=True):
+    """Run *functions* in sequence, recomputing activations in backward
+    when enabled (and grad is on). Each function may take/return tuples."""
 
-    def exec(func, a):
-        return wrap_tuple(func(*a))
+    def as_tuple(v):
+        return v if type(v) is tuple else (v,)
 
-    def get_wrap_exec(func):
-        def wrap_exec(*a):
-            return exec(func, a)
+    def run(fn, packed):
+        return as_tuple(fn(*packed))
 
-        return wrap_exec
-
-    input = wrap_tuple(input)
-
-    is_grad_enabled = torch.is_grad_enabled()
-
-    if enabled and is_grad_enabled:
-        for func in functions:
-            input = torch.utils.checkpoint.checkpoint(get_wrap_exec(func), *input)
+    packed = as_tuple(input)
+    if enabled and torch.is_grad_enabled():
+        for fn in functions:
+            packed = torch.utils.checkpoint.checkpoint(
+                lambda *a, _fn=fn: run(_fn, a), *packed
+            )
     else:
-        for func in functions:
-            input = exec(func, input)
-    return input
+        for fn in functions:
+            packed = run(fn, packed)
+    return packed
 
 
 # --------------------------------------------------------------------------
@@ -318,9 +311,9 @@ def checkpoint_sequential(
 
 
 def permute_final_dims(tensor: torch.Tensor, inds: List[int]):
-    zero_index = -1 * len(inds)
-    first_inds = list(range(len(tensor.shape[:zero_index])))
-    return tensor.permute(first_inds + [zero_index + i for i in inds])
+    base = -len(inds)
+    keep = list(range(len(tensor.shape[:base])))
+    return tensor.permute(keep + [base + i for i in inds])
 
 
 def flatten_final_dims(t: torch.Tensor, num_dims: int):
@@ -329,62 +322,59 @@ def flatten_final_dims(t: torch.Tensor, num_dims: int):
 
 def masked_mean(mask, value, dim, eps=1e-10):
     mask = mask.expand(*value.shape)
-    return torch.sum(mask * value, dim=dim) / (eps + torch.sum(mask, dim=dim))
+    return (mask * value).sum(dim=dim) / (eps + mask.sum(dim=dim))
 
 
 def dict_multimap(fn, dicts):
-    first = dicts[0]
-    new_dict = {}
-    for k, v in first.items():
-        all_v = [d[k] for d in dicts]
-        if type(v) is dict:
-            new_dict[k] = dict_multimap(fn, all_v)
-        else:
-            new_dict[k] = fn(all_v)
-    return new_dict
+    head = dicts[0]
+    out = {}
+    for key, value in head.items():
+        stacked = [d[key] for d in dicts]
+        out[key] = (
+            dict_multimap(fn, stacked) if type(value) is dict else fn(stacked)
+        )
+    return out
 
 
 def one_hot(x, num_classes, dtype=torch.float32):
-    x_one_hot = torch.zeros(*x.shape, num_classes, dtype=dtype, device=x.device)
-    x_one_hot.scatter_(-1, x.long().unsqueeze(-1), 1)
-    return x_one_hot
+    out = torch.zeros(*x.shape, num_classes, dtype=dtype, device=x.device)
+    out.scatter_(-1, x.long().unsqueeze(-1), 1)
+    return out
 
 
 def batched_gather(data, inds, dim=0, num_batch_dims=0):
     assert dim < 0 or dim - num_batch_dims >= 0
-    ranges = []
-    for i, s in enumerate(data.shape[:num_batch_dims]):
-        r = torch.arange(s)
-        r = r.view(*(*((1,) * i), -1, *((1,) * (len(inds.shape) - i - 1))))
-        ranges.append(r)
-
-    remaining_dims = [slice(None) for _ in range(len(data.shape) - num_batch_dims)]
-    remaining_dims[dim - num_batch_dims if dim >= 0 else dim] = inds
-    ranges.extend(remaining_dims)
-    return data[ranges]
+    index = []
+    for i, size in enumerate(data.shape[:num_batch_dims]):
+        shape = (1,) * i + (-1,) + (1,) * (len(inds.shape) - i - 1)
+        index.append(torch.arange(size).view(*shape))
+    tail = [slice(None)] * (len(data.shape) - num_batch_dims)
+    tail[dim - num_batch_dims if dim >= 0 else dim] = inds
+    index.extend(tail)
+    return data[index]
 
 
 def dict_map(fn, dic, leaf_type):
-    new_dict = {}
-    for k, v in dic.items():
-        if type(v) is dict:
-            new_dict[k] = dict_map(fn, v, leaf_type)
-        else:
-            new_dict[k] = tree_map(fn, v, leaf_type)
-    return new_dict
+    return {
+        key: (
+            dict_map(fn, value, leaf_type)
+            if type(value) is dict
+            else tree_map(fn, value, leaf_type)
+        )
+        for key, value in dic.items()
+    }
 
 
 def tree_map(fn, tree, leaf_type):
     if isinstance(tree, dict):
         return dict_map(fn, tree, leaf_type)
-    elif isinstance(tree, list):
-        return [tree_map(fn, x, leaf_type) for x in tree]
-    elif isinstance(tree, tuple):
-        return tuple([tree_map(fn, x, leaf_type) for x in tree])
-    elif isinstance(tree, leaf_type):
+    if isinstance(tree, list):
+        return [tree_map(fn, v, leaf_type) for v in tree]
+    if isinstance(tree, tuple):
+        return tuple(tree_map(fn, v, leaf_type) for v in tree)
+    if isinstance(tree, leaf_type):
         return fn(tree)
-    else:
-        raise ValueError("Not supported")
+    raise ValueError("Not supported")
 
 
 tensor_tree_map = partial(tree_map, leaf_type=torch.Tensor)
