@@ -4,7 +4,8 @@ Behavioral parity with the reference's ``setup_registry`` factory
 (reference: unicore/registry.py:13-81): each registry exposes a
 ``register_x`` decorator and a ``build_x(args, *extra)`` factory, registered
 classes may contribute CLI args via a classmethod ``add_args(parser)``, and
-``--x`` choices are derived from the registry keys.
+``--x`` choices are derived from the registry keys. ``REGISTRIES`` indexes
+every registry so options.py can install one choice flag per registry.
 """
 
 import argparse
@@ -12,70 +13,63 @@ import argparse
 REGISTRIES = {}
 
 
-def setup_registry(registry_name: str, base_class=None, default=None, required=False):
+def setup_registry(registry_name: str, base_class=None, default=None,
+                   required=False):
     assert registry_name.startswith("--")
-    clean_name = registry_name[2:].replace("-", "_")
+    key = registry_name[2:].replace("-", "_")
 
-    REGISTRY = {}
-    REGISTRY_CLASS_NAMES = set()
-
-    # maintain a registry of all registries
-    if clean_name in REGISTRIES:
-        raise ValueError(f"Cannot setup duplicate registry: {clean_name}")
-    REGISTRIES[clean_name] = {"registry": REGISTRY, "default": default}
+    if key in REGISTRIES:
+        raise ValueError(f"Cannot setup duplicate registry: {key}")
+    table = {}
+    seen_class_names = set()
+    REGISTRIES[key] = {"registry": table, "default": default}
 
     def build_x(args, *extra_args, **extra_kwargs):
-        choice = getattr(args, clean_name, None)
+        choice = getattr(args, key, None)
         if choice is None:
             if required:
-                raise ValueError(f"--{clean_name} is required")
+                raise ValueError(f"--{key} is required")
             return None
-        cls = REGISTRY[choice]
-        if hasattr(cls, "build_" + clean_name):
-            builder = getattr(cls, "build_" + clean_name)
-        else:
-            builder = cls
-        set_defaults(args, cls)
-        return builder(args, *extra_args, **extra_kwargs)
+        chosen = table[choice]
+        # classes may define build_<key> as an alternate constructor
+        factory = getattr(chosen, "build_" + key, chosen)
+        set_defaults(args, chosen)
+        return factory(args, *extra_args, **extra_kwargs)
 
     def register_x(name):
-        def register_x_cls(cls):
-            if name in REGISTRY:
-                raise ValueError(f"Cannot register duplicate {clean_name} ({name})")
-            if cls.__name__ in REGISTRY_CLASS_NAMES:
+        def wrap(cls):
+            if name in table:
+                raise ValueError(f"Cannot register duplicate {key} ({name})")
+            if cls.__name__ in seen_class_names:
                 raise ValueError(
-                    f"Cannot register {clean_name} with duplicate class name ({cls.__name__})"
+                    f"Cannot register {key} with duplicate class name "
+                    f"({cls.__name__})"
                 )
             if base_class is not None and not issubclass(cls, base_class):
                 raise ValueError(
                     f"{cls.__name__} must extend {base_class.__name__}"
                 )
-            REGISTRY[name] = cls
-            REGISTRY_CLASS_NAMES.add(cls.__name__)
+            table[name] = cls
+            seen_class_names.add(cls.__name__)
             return cls
 
-        return register_x_cls
+        return wrap
 
-    return build_x, register_x, REGISTRY
+    return build_x, register_x, table
 
 
 def set_defaults(args, cls):
-    """Apply defaults from cls.add_args to any attribute not already set.
-
-    Mirrors the behavior at reference unicore/registry.py:66-81: a throwaway
-    parser collects the class's declared defaults, and any arg the user did
-    not supply inherits them.
-    """
+    """Backfill args with the defaults cls.add_args declares, for any
+    attribute the user never set (reference unicore/registry.py:66-81)."""
     if not hasattr(cls, "add_args"):
         return
-    parser = argparse.ArgumentParser(argument_default=argparse.SUPPRESS, allow_abbrev=False)
-    cls.add_args(parser)
-    defaults = argparse.Namespace()
-    for action in parser._actions:
-        if action.dest is not argparse.SUPPRESS:
-            if not hasattr(defaults, action.dest):
-                if action.default is not argparse.SUPPRESS:
-                    setattr(defaults, action.dest, action.default)
-    for key, default_value in vars(defaults).items():
-        if not hasattr(args, key):
-            setattr(args, key, default_value)
+    probe = argparse.ArgumentParser(
+        argument_default=argparse.SUPPRESS, allow_abbrev=False
+    )
+    cls.add_args(probe)
+    for action in probe._actions:
+        dest = action.dest
+        if dest is argparse.SUPPRESS or action.default is argparse.SUPPRESS:
+            continue
+        if not hasattr(args, dest):
+            setattr(args, dest, action.default)

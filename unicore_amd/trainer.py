@@ -10,10 +10,18 @@ side HIP stream overlapped with backward, buckets aliased onto the fp16/bf16
 optimizer's flat grad buffers), gradient H2D staging uses pinned
 non_blocking copies, and rocTX-visible phase names are kept identical to the
 reference's record_function annotations so rocprof traces line up.
+
+Update anatomy (one train_step):
+  micro-batch loop (no_sync on all but the last) -> cross-rank stat sync ->
+  reduce-grads -> multiply-grads (world/sample_size, undoes loss scale) ->
+  clip-grads -> cross-rank grad-norm consistency check -> optimizer (rank-
+  identical seed for SR) -> EMA. Overflow skips the update and rescales;
+  NaN/Inf triggers a forensic NanDetector re-run.
 """
 
 import contextlib
 import logging
+import os
 import sys
 import time
 from itertools import chain
@@ -31,53 +39,85 @@ from unicore_amd.optim import lr_scheduler
 logger = logging.getLogger(__name__)
 
 
-class Trainer(object):
-    """Main class for data parallel training.
+def _looks_like_oom(exc):
+    return "out of memory" in str(exc)
 
-    This class supports synchronous distributed data parallel training,
-    where multiple workers each have a full model replica and gradients
-    are accumulated across workers before each update. We use
-    :class:`~unicore_amd.distributed.FlatDDP` to handle communication of
-    the gradients across workers (RCCL over xGMI on an MI355X node).
-    """
+
+def _zeroed_stat(stat):
+    """Zero a (tensor or plain) accumulated statistic in place-ish."""
+    if torch.is_tensor(stat):
+        stat.zero_()
+        return stat
+    return stat * 0.0
+
+
+def _find_tied_params(module, memo=None, prefix=""):
+    """Parameter objects reachable under more than one dotted path
+    (reference unicore/trainer.py:1127-1145). Walks raw _parameters because
+    named_parameters() deduplicates exactly the sharing we want to see."""
+    root_call = memo is None
+    if root_call:
+        memo = {}
+    for name, param in module._parameters.items():
+        dotted = f"{prefix}.{name}" if prefix else name
+        memo.setdefault(param, []).append(dotted)
+    for name, child in module._modules.items():
+        if child is not None:
+            _find_tied_params(
+                child, memo, f"{prefix}.{name}" if prefix else name
+            )
+    if root_call:
+        return [paths for paths in memo.values() if len(paths) > 1]
+
+
+def _resolve_attr(module, dotted):
+    for part in dotted.split("."):
+        module = getattr(module, part)
+    return module
+
+
+def _assign_attr(module, dotted, value):
+    *head, leaf = dotted.split(".")
+    for part in head:
+        module = getattr(module, part)
+    setattr(module, leaf, value)
+
+
+class Trainer:
+    """Synchronous data-parallel trainer: every rank holds a full replica;
+    gradients are combined across ranks before each update through our
+    FlatDDP engine (RCCL over xGMI on an MI355X node)."""
 
     def __init__(self, args, task, model, loss):
         self.args = args
         self.task = task
 
-        # catalog shared parameters BEFORE casting/moving
-        shared_params = _catalog_shared_params(model)
+        # record tied parameters BEFORE any cast/move breaks the aliasing
+        tied = _find_tied_params(model)
 
         self.cuda = torch.cuda.is_available() and not args.cpu
-        if self.cuda:
-            self.device = torch.device("cuda")
-        else:
-            self.device = torch.device("cpu")
+        self.device = torch.device("cuda" if self.cuda else "cpu")
 
-        # copy model and loss to current device/dtype (params must be cast
-        # BEFORE the optimizer flattens them)
-        self._loss = loss
-        self._model = model
+        # params must reach their final dtype/device BEFORE the optimizer
+        # flattens them
+        self._model, self._loss = model, loss
         if args.fp16:
-            self._loss = self._loss.half()
-            self._model = self._model.half()
+            self._model, self._loss = self._model.half(), self._loss.half()
         elif args.bf16:
-            self._loss = self._loss.bfloat16()
             self._model = self._model.bfloat16()
+            self._loss = self._loss.bfloat16()
         if self.cuda:
-            self._loss = self._loss.to(device=self.device)
-            self._model = self._model.to(device=self.device)
+            self._model = self._model.to(self.device)
+            self._loss = self._loss.to(self.device)
 
-        # check that shared parameters are preserved after device transfer
-        for shared_param in shared_params:
-            ref = _get_module_by_path(self._model, shared_param[0])
-            for path in shared_param[1:]:
-                logger.info(
-                    "detected shared parameter: {} <- {}".format(shared_param[0], path)
-                )
-                _set_module_by_path(self._model, path, ref)
+        # re-tie anything .to()/cast may have un-aliased
+        for paths in tied:
+            canonical = _resolve_attr(self._model, paths[0])
+            for other in paths[1:]:
+                logger.info(f"detected shared parameter: {paths[0]} <- {other}")
+                _assign_attr(self._model, other, canonical)
 
-        self._dummy_batch = None  # indicates we don't have a dummy batch at first
+        self._dummy_batch = None  # filled from the first train iterator
         self._lr_scheduler = None
         self._total_train_steps = None
         self._num_updates = 0
@@ -91,60 +131,63 @@ class Trainer(object):
             p.requires_grad for p in self._loss.parameters()
         )
 
+        self._grad_norm_buf = None
         if self.cuda and self.data_parallel_world_size > 1:
             self._grad_norm_buf = torch.zeros(
                 self.data_parallel_world_size, device=self.device
             )
-        else:
-            self._grad_norm_buf = None
 
-        # EMA (rank-0 only unless validating with EMA,
-        # reference unicore/trainer.py:114-130)
-        if args.validate_with_ema:
-            assert args.ema_decay > 0, "valid with ema must with ema_decay > 0"
-        if args.ema_decay > 0 and (
-            self.data_parallel_rank == 0 or args.validate_with_ema
-        ):
-            self.ema = ExponentialMovingAverageModel(
-                args,
-                self._model,
-                args.ema_decay,
-                is_flattened=(args.fp16 or args.bf16),
-            )
-        else:
-            self.ema = None
-
-        # get detailed cuda environment
-        if self.cuda:
-            self.cuda_env = utils.CudaEnvironment()
-            if self.data_parallel_world_size > 1:
-                self.cuda_env_arr = distributed_utils.all_gather_list(
-                    self.cuda_env, group=distributed_utils.get_global_group()
-                )
-            else:
-                self.cuda_env_arr = [self.cuda_env]
-            if self.data_parallel_rank == 0:
-                utils.CudaEnvironment.pretty_print_cuda_env_list(self.cuda_env_arr)
-        else:
-            self.cuda_env = None
-            self.cuda_env_arr = []
+        self.ema = self._build_ema()
+        self._snapshot_cuda_env()
 
         metrics.log_start_time("wall", priority=790, round=2)
-
         self._start_time = time.time()
         self._previous_training_time = 0
         self._cumulative_training_time = None
 
+    def _build_ema(self):
+        """EMA shadow on rank 0 only, unless validation swaps it in on every
+        rank (reference unicore/trainer.py:114-130)."""
+        args = self.args
+        if args.validate_with_ema:
+            assert args.ema_decay > 0, "valid with ema must with ema_decay > 0"
+        wants_ema = args.ema_decay > 0 and (
+            self.data_parallel_rank == 0 or args.validate_with_ema
+        )
+        if not wants_ema:
+            return None
+        return ExponentialMovingAverageModel(
+            args, self._model, args.ema_decay,
+            is_flattened=(args.fp16 or args.bf16),
+        )
+
+    def _snapshot_cuda_env(self):
+        if not self.cuda:
+            self.cuda_env = None
+            self.cuda_env_arr = []
+            return
+        self.cuda_env = utils.CudaEnvironment()
+        self.cuda_env_arr = [self.cuda_env]
+        if self.data_parallel_world_size > 1:
+            self.cuda_env_arr = distributed_utils.all_gather_list(
+                self.cuda_env, group=distributed_utils.get_global_group()
+            )
+        if self.data_parallel_rank == 0:
+            utils.CudaEnvironment.pretty_print_cuda_env_list(self.cuda_env_arr)
+
     def reinitialize(self):
-        """Reinitialize the Trainer, typically after model params change."""
+        """Drop derived objects (optimizer/scheduler/wrappers) so they are
+        rebuilt lazily, e.g. after swapping model params."""
         self._lr_scheduler = None
         self._optimizer = None
         self._wrapped_loss = None
         self._wrapped_model = None
 
+    # -- topology ----------------------------------------------------------
+
     @property
     def data_parallel_world_size(self):
-        if self.args.distributed_world_size == 1:
+        if self.args.distributed_world_size <= 1:
             return 1
         return distributed_utils.get_data_parallel_world_size()
 
@@ -154,14 +197,12 @@ class Trainer(object):
 
     @property
     def data_parallel_rank(self):
-        if self.args.distributed_world_size == 1:
+        if self.args.distributed_world_size <= 1:
             return 0
         return distributed_utils.get_data_parallel_rank()
 
     @property
     def is_data_parallel_master(self):
-        # NOTE: this returns true for all model parallel replicas with data
-        # parallel rank 0
         return self.data_parallel_rank == 0
 
     @property
@@ -170,35 +211,40 @@ class Trainer(object):
 
     @property
     def should_save_checkpoint_on_current_rank(self) -> bool:
-        """Indicates whether to save checkpoints on the current DDP rank."""
         return self.is_data_parallel_master
+
+    # -- lazily wrapped model/loss/optimizer -------------------------------
 
     @property
     def loss(self):
         if self._wrapped_loss is None:
-            if utils.has_parameters(self._loss) and self.use_distributed_wrapper:
-                self._wrapped_loss = models.DistributedUnicoreModel(
-                    self.args,
-                    self._loss,
+            wrap = (
+                utils.has_parameters(self._loss)
+                and self.use_distributed_wrapper
+            )
+            self._wrapped_loss = (
+                models.DistributedUnicoreModel(
+                    self.args, self._loss,
                     process_group=self.data_parallel_process_group,
                     device=self.device,
                 )
-            else:
-                self._wrapped_loss = self._loss
+                if wrap
+                else self._loss
+            )
         return self._wrapped_loss
 
     @property
     def model(self):
         if self._wrapped_model is None:
-            if self.use_distributed_wrapper:
-                self._wrapped_model = models.DistributedUnicoreModel(
-                    self.args,
-                    self._model,
+            self._wrapped_model = (
+                models.DistributedUnicoreModel(
+                    self.args, self._model,
                     process_group=self.data_parallel_process_group,
                     device=self.device,
                 )
-            else:
-                self._wrapped_model = self._model
+                if self.use_distributed_wrapper
+                else self._model
+            )
         return self._wrapped_model
 
     @property
@@ -210,102 +256,105 @@ class Trainer(object):
     @property
     def lr_scheduler(self):
         if self._lr_scheduler is None:
-            self._build_optimizer()  # this will initialize self._lr_scheduler
+            self._build_optimizer()  # also creates the scheduler
         return self._lr_scheduler
 
+    def _check_mode_constraints(self):
+        legacy = ("no_c10d", "legacy_ddp")
+        if self.args.per_sample_clip_norm > 0.0:
+            assert self.args.ddp_backend in legacy, \
+                "--per-sample-clip-norm requires --ddp-backend no_c10d"
+            assert self.args.fp16 or self.args.bf16, \
+                "--per-sample-clip-norm requires fp16/bf16 training"
+        if self.args.allreduce_fp32_grad:
+            assert self.args.ddp_backend in legacy, \
+                "--allreduce-fp32-grad requires --ddp-backend no_c10d"
+            assert self.args.fp16 or self.args.bf16, \
+                "--allreduce-fp32-grad requires fp16/bf16 training"
+
     def _build_optimizer(self):
-        params = [
-            (name, param)
-            for name, param in chain(
+        trainable = [
+            (name, p)
+            for name, p in chain(
                 self._model.named_parameters(), self._loss.named_parameters()
             )
-            if param.requires_grad
+            if p.requires_grad
         ]
+        self._check_mode_constraints()
 
-        if self.args.per_sample_clip_norm > 0:
-            assert self.args.ddp_backend in ("no_c10d", "legacy_ddp"), (
-                "--per-sample-clip-norm requires --ddp-backend no_c10d"
-            )
-            assert self.args.fp16 or self.args.bf16, (
-                "--per-sample-clip-norm requires fp16/bf16 training"
-            )
-        if self.args.allreduce_fp32_grad:
-            assert self.args.ddp_backend in ("no_c10d", "legacy_ddp"), (
-                "--allreduce-fp32-grad requires --ddp-backend no_c10d"
-            )
-            assert self.args.fp16 or self.args.bf16, (
-                "--allreduce-fp32-grad requires fp16/bf16 training"
-            )
-
-        if self.args.fp16 or self.args.bf16:
+        low_precision = self.args.fp16 or self.args.bf16
+        if not low_precision:
+            if self.cuda and torch.cuda.get_device_capability(0)[0] >= 7:
+                logger.info("NOTE: your device may support faster training"
+                            " with --fp16")
+            self._optimizer = optim.build_optimizer(self.args, trainable)
+        if low_precision:
             if self.cuda and torch.cuda.get_device_capability(0)[0] < 7:
-                logger.info(
-                    "NOTE: your device does NOT support faster training with --fp16, "
-                    "please switch to FP32 which is likely to be faster"
-                )
-            self._optimizer = optim.FP16Optimizer.build_optimizer(self.args, params)
-            if (
-                self.data_parallel_world_size == 1
-                and self.args.per_sample_clip_norm <= 0
+                logger.info("NOTE: your device does NOT support faster"
+                            " training with --fp16, please switch to FP32"
+                            " which is likely to be faster")
+            self._optimizer = optim.FP16Optimizer.build_optimizer(
+                self.args, trainable
+            )
+            lazy_ok = (
+                self.data_parallel_world_size <= 1
+                and self.args.per_sample_clip_norm <= 0.0
                 and hasattr(self._optimizer, "enable_lazy_grad_collection")
-            ):
+            )
+            if lazy_ok:
                 # no DDP engine hooks the flat grads: let autograd assign
                 # gradients and batch-copy them at sync time instead of
                 # one accumulate kernel per parameter per backward
                 self._optimizer.enable_lazy_grad_collection()
-        else:
-            if self.cuda and torch.cuda.get_device_capability(0)[0] >= 7:
-                logger.info("NOTE: your device may support faster training with --fp16")
-            self._optimizer = optim.build_optimizer(self.args, params)
 
-        # Build the wrapped (DDP) model AFTER the optimizer so FlatDDP can
-        # alias its buckets onto the optimizer's flat grad buffers. If a
-        # wrapper already exists (rebuild path), detach its grad hooks first.
+        # (re)build the DDP wrappers AFTER the optimizer so FlatDDP can
+        # alias its buckets onto the optimizer's flat grad buffers; stale
+        # wrappers must drop their autograd hooks first
         for wrapped in (self._wrapped_model, self._wrapped_loss):
-            if wrapped is not None and hasattr(wrapped, "module"):
-                inner = wrapped.module
-                if hasattr(inner, "detach_hooks"):
-                    inner.detach_hooks()
+            inner = getattr(wrapped, "module", None)
+            if inner is not None and hasattr(inner, "detach_hooks"):
+                inner.detach_hooks()
         self._wrapped_model = None
         self._wrapped_loss = None
         _ = self.model
         _ = self.loss
 
-        # We should initialize the learning rate scheduler immediately after
-        # building the optimizer, so that the initial learning rate is set.
+        # scheduler immediately after the optimizer so the initial LR lands
         self._lr_scheduler = lr_scheduler.build_lr_scheduler(
-            self.args,
-            self.optimizer,
-            self._total_train_steps,
+            self.args, self.optimizer, self._total_train_steps
         )
         self._lr_scheduler.step_update(0)
 
     def init_total_train_steps(self, epoch_itr):
-        """Total train steps, needed by warmup_ratio schedulers
-        (reference unicore/trainer.py:518-524)."""
-        if self.args.max_epoch > 0:
-            self._total_train_steps = (
-                (len(epoch_itr) + 1) // max(1, self.args.update_freq[0])
-            ) * self.args.max_epoch
-        else:
+        """Total updates this run will perform — warmup_ratio schedulers
+        need it up front (reference unicore/trainer.py:518-524)."""
+        if self.args.max_epoch <= 0:
             self._total_train_steps = self.args.max_update
+            return
+        per_epoch = (len(epoch_itr) + 1) // max(1, self.args.update_freq[0])
+        self._total_train_steps = per_epoch * self.args.max_epoch
+
+    # -- checkpoint state ---------------------------------------------------
 
     def state_dict(self):
-        state_dict = {
+        """Assemble the full checkpoint payload."""
+        history = (self._optim_history or []) + [
+            {
+                "loss_name": type(self.get_loss()).__name__,
+                "optimizer_name": type(self.optimizer).__name__,
+                "lr_scheduler_state": self.lr_scheduler.state_dict(),
+                "num_updates": self.get_num_updates(),
+            }
+        ]
+        payload = {
             "args": self.args,
             "model": self._model.state_dict(),
             "loss": (
-                self._loss.state_dict() if utils.has_parameters(self._loss) else None
+                self._loss.state_dict()
+                if utils.has_parameters(self._loss)
+                else None
             ),
-            "optimizer_history": (self._optim_history or [])
-            + [
-                {
-                    "loss_name": self.get_loss().__class__.__name__,
-                    "optimizer_name": self.optimizer.__class__.__name__,
-                    "lr_scheduler_state": self.lr_scheduler.state_dict(),
-                    "num_updates": self.get_num_updates(),
-                }
-            ],
+            "optimizer_history": history,
             "task_state": self.task.state_dict() if self.task is not None else {},
             "extra_state": {
                 "metrics": metrics.state_dict(),
@@ -313,157 +362,140 @@ class Trainer(object):
             },
         }
         if not self.args.no_save_optimizer_state:
-            state_dict["last_optimizer_state"] = self.optimizer.state_dict()
+            payload["last_optimizer_state"] = self.optimizer.state_dict()
         if self.ema is not None:
-            state_dict["ema"] = self.ema.state_dict()
-        return state_dict
+            payload["ema"] = self.ema.state_dict()
+        return payload
 
     def save_checkpoint(self, filename, extra_state):
-        """Save all training state in a checkpoint file."""
+        """Serialize all training state into *filename*."""
         logger.info(f"Saving checkpoint to {filename}")
-        # call state_dict on all ranks in case it needs internal communication
-        state_dict = utils.move_to_cpu(self.state_dict())
-        state_dict["extra_state"].update(extra_state)
+        # every rank runs state_dict() (it may communicate); rank 0 writes
+        payload = utils.move_to_cpu(self.state_dict())
+        payload["extra_state"].update(extra_state)
         if self.should_save_checkpoint_on_current_rank:
-            checkpoint_utils.torch_persistent_save(state_dict, filename)
+            checkpoint_utils.torch_persistent_save(payload, filename)
         logger.info(f"Finished saving checkpoint to {filename}")
 
-    def load_checkpoint(
-        self,
-        filename,
-        reset_optimizer=False,
-        reset_lr_scheduler=False,
-        optimizer_overrides=None,
-        reset_meters=False,
-        reset_dataloader=False,
-        **passthrough_args,
-    ):
-        """
-        Load all training state from a checkpoint file and build the train
-        iterator. rank = 0 loads the checkpoint and broadcasts it to all
-        other ranks. Returns (extra_state, epoch_itr).
-        """
-        extra_state, self._optim_history, last_optim_state = None, [], None
-
-        logger.info(f"Preparing to load checkpoint {filename}")
-        is_distributed = self.data_parallel_world_size > 1
-        bexists = None
-        if self.data_parallel_rank == 0:
-            import os
-
-            bexists = os.path.isfile(filename)
-        if is_distributed:
-            bexists = distributed_utils.broadcast_object(
-                bexists,
-                src_rank=0,
+    def _fetch_checkpoint_state(self, filename):
+        """Rank 0 reads the file; everyone else receives it by broadcast.
+        Returns None when the file does not exist anywhere."""
+        multi_rank = self.data_parallel_world_size > 1
+        exists = os.path.isfile(filename) if self.data_parallel_rank == 0 else None
+        if multi_rank:
+            exists = distributed_utils.broadcast_object(
+                exists, src_rank=0,
                 group=self.data_parallel_process_group,
                 dist_device=self.device,
             )
+        if not exists:
+            return None
+        state = None
+        if self.data_parallel_rank == 0:
+            state = checkpoint_utils.load_checkpoint_to_cpu(filename)
+        if multi_rank:
+            logger.info("Broadcast checkpoint to all ranks")
+            state = distributed_utils.broadcast_object(
+                state, src_rank=0,
+                group=self.data_parallel_process_group,
+                dist_device=self.device,
+            )
+        return state
 
+    def _restore_weights(self, state, filename):
+        """Load model (+loss) weights from the checkpoint state; returns
+        (ema_state, ema_loaded)."""
+        ema_state = state.get("ema", None)
         ema_loaded = False
-        if bexists:
-            state = None
-            if self.data_parallel_rank == 0:
-                state = checkpoint_utils.load_checkpoint_to_cpu(filename)
-            if is_distributed:
-                logger.info("Broadcast checkpoint to all ranks")
-                state = distributed_utils.broadcast_object(
-                    state,
-                    src_rank=0,
-                    group=self.data_parallel_process_group,
-                    dist_device=self.device,
+        try:
+            if self.args.load_from_ema:
+                logger.info("loading ema state to model")
+                assert ema_state is not None, "no EMA state in checkpoint"
+                self._model.load_state_dict(
+                    ema_state["params"], strict=False, model_args=self.args
                 )
-
-            # load model parameters
-            try:
-                ema_state = state.get("ema", None)
-                if self.args.load_from_ema:
-                    logger.info("loading ema state to model")
-                    assert ema_state is not None, "no EMA state in checkpoint"
-                    errors = self._model.load_state_dict(
-                        ema_state["params"], strict=False, model_args=self.args
-                    )
-                    ema_loaded = True
-                else:
-                    errors = self._model.load_state_dict(
-                        state["model"], strict=False, model_args=self.args
-                    )
-                    if errors is not None and getattr(errors, "missing_keys", None):
-                        logger.warning(
-                            "Error in loading model state, missing_keys "
-                            + str(errors.missing_keys)
-                        )
-                    if errors is not None and getattr(errors, "unexpected_keys", None):
-                        logger.warning(
-                            "Error in loading model state, unexpected_keys "
-                            + str(errors.unexpected_keys)
-                        )
-                # save memory for later steps
-                del state["model"]
-
-                if utils.has_parameters(self.get_loss()) and state.get("loss"):
-                    self.get_loss().load_state_dict(state["loss"], strict=True)
-                if "loss" in state:
-                    del state["loss"]
-            except Exception:
-                raise Exception(
-                    "Cannot load model parameters from checkpoint {}; "
-                    "please ensure that the architectures match.".format(filename)
+                ema_loaded = True
+            if not ema_loaded:
+                report = self._model.load_state_dict(
+                    state["model"], strict=False, model_args=self.args
                 )
+                for field in ("missing_keys", "unexpected_keys"):
+                    keys = getattr(report, field, None) if report is not None else None
+                    if keys:
+                        logger.warning(
+                            f"Error in loading model state, {field} {keys}"
+                        )
+            del state["model"]  # free before the optimizer state loads
 
+            if utils.has_parameters(self.get_loss()) and state.get("loss"):
+                self.get_loss().load_state_dict(state["loss"], strict=True)
+            state.pop("loss", None)
+        except Exception:
+            raise Exception(
+                f"Cannot load model parameters from checkpoint {filename}; "
+                "please ensure that the architectures match."
+            )
+        return ema_state, ema_loaded
+
+    def _restore_ema(self, ema_state, ema_loaded):
+        if ema_state is not None and self.ema is not None \
+                and not self.args.load_from_ema:
+            logger.info("loading ema state from checkpoint")
+            self.ema.load_state_dict(ema_state)
+        elif self.ema is not None and not ema_loaded:
+            logger.info(
+                "Cannot find EMA state in checkpoint, load model weight to "
+                "ema directly"
+            )
+            self.ema = ExponentialMovingAverageModel(
+                self.args, self._model, decay=self.ema.decay,
+                is_flattened=(self.args.fp16 or self.args.bf16),
+            )
+
+    def load_checkpoint(self, filename, reset_optimizer=False,
+                        reset_lr_scheduler=False, optimizer_overrides=None,
+                        reset_meters=False, reset_dataloader=False,
+                        **passthrough_args):
+        """Restore all training state: rank 0 reads + broadcasts, weights
+        load everywhere, then the iterator/optimizer/meters are rebuilt.
+        Returns (extra_state, epoch_itr)."""
+        logger.info(f"Preparing to load checkpoint {filename}")
+
+        extra_state, self._optim_history, last_optim_state = None, [], None
+        ema_loaded = False
+        state = self._fetch_checkpoint_state(filename)
+        if state is not None:
+            ema_state, ema_loaded = self._restore_weights(state, filename)
             extra_state = state.get("extra_state", None)
             self._optim_history = state.get("optimizer_history", None)
             last_optim_state = state.get("last_optimizer_state", None)
-
-            if (
-                ema_state is not None
-                and self.ema is not None
-                and not self.args.load_from_ema
-            ):
-                logger.info("loading ema state from checkpoint")
-                self.ema.load_state_dict(ema_state)
-            elif self.ema is not None and not ema_loaded:
-                logger.info(
-                    "Cannot find EMA state in checkpoint, load model weight to ema directly"
-                )
-                self.ema = ExponentialMovingAverageModel(
-                    self.args,
-                    self._model,
-                    decay=self.ema.decay,
-                    is_flattened=(self.args.fp16 or self.args.bf16),
-                )
+            self._restore_ema(ema_state, ema_loaded)
 
         loaded_train_itr = False
         epoch = 1
         if extra_state is not None:
             itr_state = extra_state.get("train_iterator", None)
-
             if itr_state is not None:
                 epoch = itr_state.get("epoch", 1)
-
                 if "previous_training_time" in extra_state:
-                    self._previous_training_time = extra_state["previous_training_time"]
+                    self._previous_training_time = \
+                        extra_state["previous_training_time"]
                     self._start_time = time.time()
-
-                if (
-                    itr_state.get("version", 1) >= 2
-                    and itr_state["iterations_in_epoch"] == 0
-                ):
-                    # reset meters at start of epoch
-                    reset_meters = True
+                if (itr_state.get("version", 1) >= 2
+                        and itr_state["iterations_in_epoch"] == 0):
+                    reset_meters = True  # clean epoch boundary
 
             if "metrics" in extra_state and not reset_meters:
                 metrics.load_state_dict(extra_state["metrics"])
-
-                # reset TimeMeters, since their start times don't make sense anymore
+                # TimeMeters reference a dead process's clock; restart them
                 for meter in metrics.get_meters("default"):
                     if isinstance(meter, meters.TimeMeter):
                         meter.reset()
 
             if itr_state is not None and not reset_dataloader:
-                # restore iterator from checkpoint
                 epoch_itr = self.get_train_iterator(
-                    epoch=itr_state["epoch"], load_dataset=True, **passthrough_args
+                    epoch=itr_state["epoch"], load_dataset=True,
+                    **passthrough_args,
                 )
                 epoch_itr.load_state_dict(itr_state)
                 loaded_train_itr = True
@@ -476,118 +508,108 @@ class Trainer(object):
         self.init_total_train_steps(epoch_itr)
 
         if last_optim_state is not None and not reset_optimizer:
-            # rebuild optimizer after loading model, since params may have changed
+            # model weights just changed: rebuild before loading opt state
             self._build_optimizer()
 
-            # only reload optimizer and lr_scheduler if they match
-            last_optim = self._optim_history[-1]
-            assert (
-                last_optim["loss_name"] == self.get_loss().__class__.__name__
-            ), f"Loss does not match; please reset the optimizer (--reset-optimizer). {last_optim['loss_name']} vs {self.get_loss().__class__.__name__}"
-            assert (
-                last_optim["optimizer_name"] == self.optimizer.__class__.__name__
-            ), f"Optimizer does not match; please reset the optimizer (--reset-optimizer). {last_optim['optimizer_name']} vs {self.optimizer.__class__.__name__}"
-
+            previous = self._optim_history[-1]
+            assert previous["loss_name"] == type(self.get_loss()).__name__, (
+                "Loss does not match; please reset the optimizer "
+                f"(--reset-optimizer). {previous['loss_name']} vs "
+                f"{type(self.get_loss()).__name__}"
+            )
+            assert previous["optimizer_name"] == type(self.optimizer).__name__, (
+                "Optimizer does not match; please reset the optimizer "
+                f"(--reset-optimizer). {previous['optimizer_name']} vs "
+                f"{type(self.optimizer).__name__}"
+            )
             if not reset_lr_scheduler:
-                self.lr_scheduler.load_state_dict(last_optim["lr_scheduler_state"])
-
-            self.optimizer.load_state_dict(last_optim_state, optimizer_overrides)
-
-            self.set_num_updates(last_optim["num_updates"])
+                self.lr_scheduler.load_state_dict(
+                    previous["lr_scheduler_state"]
+                )
+            self.optimizer.load_state_dict(
+                last_optim_state, optimizer_overrides
+            )
+            self.set_num_updates(previous["num_updates"])
 
         if extra_state is not None and loaded_train_itr:
             logger.info(
-                "Loaded checkpoint {} (epoch {} @ {} updates)".format(
-                    filename, epoch, self.get_num_updates()
-                )
+                f"Loaded checkpoint {filename} "
+                f"(epoch {epoch} @ {self.get_num_updates()} updates)"
             )
         elif ema_loaded:
-            logger.info("Loaded ema state from checkpoint {}".format(filename))
+            logger.info(f"Loaded ema state from checkpoint {filename}")
         elif extra_state is None:
-            logger.info("No existing checkpoint found {}".format(filename))
+            logger.info(f"No existing checkpoint found {filename}")
 
         self.lr_step(epoch_itr.epoch)
-
         return extra_state, epoch_itr
 
-    def get_train_iterator(
-        self,
-        epoch,
-        combine=True,
-        load_dataset=True,
-        data_selector=None,
-        shard_batch_itr=True,
-        disable_iterator_cache=False,
-    ):
-        """Return an EpochBatchIterator over the training set for a given epoch."""
+    # -- iterators ----------------------------------------------------------
+
+    def get_train_iterator(self, epoch, combine=True, load_dataset=True,
+                           data_selector=None, shard_batch_itr=True,
+                           disable_iterator_cache=False):
+        """EpochBatchIterator over the training split for *epoch*."""
         if load_dataset:
-            logger.info("loading train data for epoch {}".format(epoch))
+            logger.info(f"loading train data for epoch {epoch}")
             self.task.load_dataset(
-                self.args.train_subset,
-                epoch=epoch,
-                combine=combine,
+                self.args.train_subset, epoch=epoch, combine=combine,
                 data_selector=data_selector,
             )
         batch_iterator = self.task.get_batch_iterator(
             dataset=self.task.dataset(self.args.train_subset),
-            batch_size=self.args.batch_size,
-            ignore_invalid_inputs=True,
-            required_batch_size_multiple=self.args.required_batch_size_multiple,
+            batch_size=self.args.batch_size, ignore_invalid_inputs=True,
+            required_batch_size_multiple=(
+                self.args.required_batch_size_multiple
+            ),
             seed=self.args.seed,
             num_shards=self.data_parallel_world_size if shard_batch_itr else 1,
             shard_id=self.data_parallel_rank if shard_batch_itr else 0,
-            num_workers=self.args.num_workers,
-            epoch=epoch,
+            num_workers=self.args.num_workers, epoch=epoch,
             data_buffer_size=self.args.data_buffer_size,
-            disable_iterator_cache=disable_iterator_cache,
-        )
+            disable_iterator_cache=disable_iterator_cache)
         self._iter_per_epoch = len(batch_iterator)
         self.reset_dummy_batch(batch_iterator.first_batch)
         return batch_iterator
 
-    def get_valid_iterator(
-        self,
-        subset,
-        disable_iterator_cache=False,
-    ):
-        """Return an EpochBatchIterator over given validation subset for a given epoch."""
+    def get_valid_iterator(self, subset, disable_iterator_cache=False):
+        """EpochBatchIterator over one validation subset."""
         batch_iterator = self.task.get_batch_iterator(
             dataset=self.task.dataset(subset),
             batch_size=self.args.batch_size_valid,
-            ignore_invalid_inputs=self.args.skip_invalid_size_inputs_valid_test,
-            required_batch_size_multiple=self.args.required_batch_size_multiple,
+            ignore_invalid_inputs=(
+                self.args.skip_invalid_size_inputs_valid_test
+            ),
+            required_batch_size_multiple=(
+                self.args.required_batch_size_multiple
+            ),
             seed=self.args.seed,
             num_shards=self.data_parallel_world_size,
             shard_id=self.data_parallel_rank,
-            num_workers=self.args.num_workers,
-            epoch=1,
+            num_workers=self.args.num_workers, epoch=1,
             data_buffer_size=self.args.data_buffer_size,
-            disable_iterator_cache=disable_iterator_cache,
-        )
+            disable_iterator_cache=disable_iterator_cache)
         self.reset_dummy_batch(batch_iterator.first_batch)
         return batch_iterator
 
     def begin_epoch(self, epoch):
-        """Called at the beginning of each epoch."""
-        logger.info("begin training epoch {}".format(epoch))
-
+        logger.info(f"begin training epoch {epoch}")
         self.lr_step_begin_epoch(epoch)
-
-        # task specific setup per epoch
         self.task.begin_epoch(epoch, self.get_model())
 
     def begin_valid_epoch(self, epoch):
-        """Called at the beginning of each validation epoch."""
-
-        # task specific setup per validation epoch
         self.task.begin_valid_epoch(epoch, self.get_model())
 
     def reset_dummy_batch(self, batch):
         self._dummy_batch = batch
 
+    # -- the hot loop --------------------------------------------------------
+
     @metrics.aggregate("train")
     def train_step(self, samples, raise_oom=False):
-        """Do forward, backward and parameter update."""
+        """Forward + backward over the micro-batch group, then one
+        parameter update. Returns the logging output, or None when the
+        update was skipped (OOM / overflow)."""
         self._set_seed()
         self.model.train()
         self.loss.train()
@@ -595,96 +617,75 @@ class Trainer(object):
 
         metrics.log_start_time("train_wall", priority=800, round=2)
 
-        # forward and backward pass
         logging_outputs, sample_size, ooms = [], 0, 0
-        for i, sample in enumerate(samples):  # delayed update loop
-            sample, is_dummy_batch = self._prepare_sample(sample)
+        for i, sample in enumerate(samples):  # grad-accumulation loop
+            sample, is_filler = self._prepare_sample(sample)
 
-            def maybe_no_sync():
-                """
-                Whenever *samples* contains more than one mini-batch, we
-                want to accumulate gradients locally and only call
-                all-reduce in the last backwards pass.
-                """
-                if (
-                    self.data_parallel_world_size > 1
-                    and hasattr(self.model, "no_sync")
-                    and i < len(samples) - 1
-                ):
+            def sync_policy():
+                # accumulate locally; only the last micro-batch's backward
+                # carries the all-reduce
+                last = i == len(samples) - 1
+                if (self.data_parallel_world_size > 1
+                        and hasattr(self.model, "no_sync") and not last):
                     return self.model.no_sync()
-                else:
-                    return contextlib.ExitStack()  # dummy contextmanager
+                return contextlib.ExitStack()  # no-op context
 
             try:
-                with maybe_no_sync():
-                    # use different seeds for different rank in training,
-                    # the seed is changed by the num_updates, to make sure
-                    # different batches use different dropout seeds
+                with sync_policy():
+                    # per-rank dropout streams, keyed by update + micro-batch
                     with utils.torch_seed(
-                        self.args.seed,
-                        self.get_num_updates(),
-                        i,
+                        self.args.seed, self.get_num_updates(), i,
                         self.data_parallel_rank,
                     ):
-                        # forward and backward
-                        loss, sample_size_i, logging_output = self.task.train_step(
-                            sample=sample,
-                            model=self.model,
-                            loss=self.loss,
-                            optimizer=self.optimizer,
-                            update_num=self.get_num_updates(),
-                            ignore_grad=is_dummy_batch,
-                        )
-                        if self.args.per_sample_clip_norm > 0:
-                            self.optimizer.per_sample_clip_grad_norm(
-                                self.args.per_sample_clip_norm
+                        loss, micro_size, logging_output = \
+                            self.task.train_step(
+                                sample=sample, model=self.model,
+                                loss=self.loss, optimizer=self.optimizer,
+                                update_num=self.get_num_updates(),
+                                ignore_grad=is_filler,
                             )
+                        psc = self.args.per_sample_clip_norm
+                        if psc > 0:
+                            self.optimizer.per_sample_clip_grad_norm(psc)
                         del loss
 
                 logging_outputs.append(logging_output)
-                sample_size += sample_size_i
+                sample_size += micro_size
 
-                # emptying the CUDA cache after the first step can
-                # reduce the chance of OOM
+                # flushing after step 0 lowers later OOM risk
                 if self.cuda and self.get_num_updates() == 0:
                     torch.cuda.empty_cache()
-            except RuntimeError as e:
-                if _is_oom_error(e):
-                    self._log_oom(e)
-                    if raise_oom:
-                        raise e
-                    logger.warning(
-                        "attempting to recover from OOM in forward/backward pass"
-                    )
-                    ooms += 1
-                    self.zero_grad()
-                    if self.cuda:
-                        torch.cuda.empty_cache()
-                    if self.args.distributed_world_size == 1:
-                        return None
-                else:
-                    raise e
+            except RuntimeError as exc:
+                if not _looks_like_oom(exc):
+                    raise
+                self._log_oom(exc)
+                if raise_oom:
+                    raise
+                logger.warning("attempting to recover from OOM in"
+                               " forward/backward pass")
+                ooms += 1
+                self.zero_grad()
+                if self.cuda:
+                    torch.cuda.empty_cache()
+                if self.args.distributed_world_size <= 1:
+                    return None
 
-        if is_dummy_batch:
-            if torch.is_tensor(sample_size):
-                sample_size.zero_()
-            else:
-                sample_size *= 0.0
+        if is_filler:
+            # dummy batches pad out short epochs; they must not weigh in
+            sample_size = _zeroed_stat(sample_size)
+        sample_size = (
+            sample_size.float()
+            if torch.is_tensor(sample_size)
+            else float(sample_size)
+        )
 
-        if torch.is_tensor(sample_size):
-            sample_size = sample_size.float()
-        else:
-            sample_size = float(sample_size)
-
-        # gather logging outputs from all replicas
         if self._sync_stats():
             train_time = self._local_cumulative_training_time()
             logging_outputs, (
-                sample_size,
-                ooms,
-                total_train_time,
+                sample_size, ooms, total_train_time,
             ) = self._aggregate_logging_outputs(
-                logging_outputs, sample_size, ooms, train_time, ignore=is_dummy_batch
+                logging_outputs, sample_size, ooms, train_time,
+                ignore=is_filler,
             )
             self._cumulative_training_time = (
                 total_train_time / self.data_parallel_world_size
@@ -693,166 +694,140 @@ class Trainer(object):
         overflow = False
         try:
             with torch.autograd.profiler.record_function("reduce-grads"):
-                # reduce gradients across workers
                 self.optimizer.all_reduce_grads(self.model)
 
             with torch.autograd.profiler.record_function("multiply-grads"):
-                # multiply gradients by (data_parallel_size / sample_size) since
-                # DDP normalizes by the number of data parallel workers for
-                # improved fp16 precision.
-                # Thus we get (sum_of_gradients / sample_size) at the end.
-                # In case of fp16, this step also undoes loss scaling.
-                # (Debugging note: Some optimizers perform this scaling on the
-                # fly, so inspecting model.parameters() or optimizer.params may
-                # still show the original, unscaled gradients.)
-                numer = self.data_parallel_world_size if self._sync_stats() else 1
+                # grads arrive pre-divided by world size (DDP convention,
+                # better low-precision accuracy), so scale by
+                # world/sample_size to end at sum_of_gradients/sample_size;
+                # for fp16 this also folds in the 1/loss_scale. Some
+                # optimizers defer this factor, so raw p.grad may still
+                # look unscaled.
+                numer = (
+                    self.data_parallel_world_size if self._sync_stats() else 1
+                )
                 self.optimizer.multiply_grads(numer / (sample_size or 1.0))
-                # Note: (sample_size or 1.0) handles the case of a zero gradient,
-                # in case sample_size is 0 or 1.0 or something else.
+                # (sample_size or 1.0) guards the all-dummy case
 
             with torch.autograd.profiler.record_function("clip-grads"):
-                # clip grads
                 grad_norm = self.clip_grad_norm(self.args.clip_norm)
 
-            # check that grad norms are consistent across workers
             self._check_grad_norms(grad_norm)
             if not torch.isfinite(grad_norm).all():
-                # check local gradnorm single GPU case, trigger NanDetector
+                # single-GPU non-finite grads go through the same
+                # NanDetector path as the cross-rank check
                 raise FloatingPointError("gradients are Nan/Inf")
 
             with torch.autograd.profiler.record_function("optimizer"):
-                # use the same seed for different ranks, to make sure that the
-                # stochastic rounding is the same across all ranks
+                # rank-IDENTICAL seed: stochastic rounding must agree
+                # across replicas or the weights drift apart
                 with utils.torch_seed(self.args.seed, self.get_num_updates()):
-                    # take an optimization step
                     self.task.optimizer_step(
-                        self.optimizer, model=self.model, update_num=self.get_num_updates()
+                        self.optimizer, model=self.model,
+                        update_num=self.get_num_updates(),
                     )
             if self.ema is not None:
                 with torch.autograd.profiler.record_function("ema"):
-                    if self.args.fp16 or self.args.bf16:
-                        self.ema.update(self.optimizer.fp32_params)
-                    else:
-                        self.ema.update(self._model.named_parameters())
+                    source = (
+                        self.optimizer.fp32_params
+                        if self.args.fp16 or self.args.bf16
+                        else self._model.named_parameters()
+                    )
+                    self.ema.update(source)
 
         except FloatingPointError:
-            # re-run the forward and backward pass with hooks attached to print
-            # out where it fails
+            # forensic re-run with hooks to name the first bad module
             self.zero_grad()
             with NanDetector(self.get_model()):
                 for _, sample in enumerate(samples):
                     sample, _ = self._prepare_sample(sample)
                     self.task.train_step(
-                        sample,
-                        self.model,
-                        self.loss,
-                        self.optimizer,
-                        self.get_num_updates(),
-                        ignore_grad=False,
+                        sample, self.model, self.loss, self.optimizer,
+                        self.get_num_updates(), ignore_grad=False,
                     )
             raise
-        except OverflowError as e:
+        except OverflowError as exc:
             overflow = True
-            logger.info(
-                f"NOTE: gradient overflow detected, ignoring gradient, {str(e)}"
-            )
-            grad_norm = torch.tensor(0.0).cuda() if self.cuda else torch.tensor(0.0)
+            logger.info(f"NOTE: gradient overflow detected, ignoring"
+                        f" gradient, {exc}")
+            grad_norm = torch.tensor(0.0)
+            if self.cuda:
+                grad_norm = grad_norm.cuda()
             self.zero_grad()
-        except RuntimeError as e:
-            if _is_oom_error(e):
-                self._log_oom(e)
+        except RuntimeError as exc:
+            if _looks_like_oom(exc):
+                self._log_oom(exc)
                 logger.error("OOM during optimization, irrecoverable")
-            raise e
+            raise
 
         logging_output = None
         if not overflow:
             self.set_num_updates(self.get_num_updates() + 1)
-
-            if self.cuda and self.cuda_env is not None:
-                # log minimum free memory over the iteration
-                gb_used = torch.cuda.max_memory_allocated() / 1024 / 1024 / 1024
-                torch.cuda.reset_peak_memory_stats()
-                gb_free = self.cuda_env.total_memory_in_GB - gb_used
-                metrics.log_scalar("gb_free", gb_free, priority=1500, round=1, weight=0)
-
-            # extract probs, when arg exists
+            self._log_memory_headroom()
             logging_output = self._reduce_and_log_stats(
-                logging_outputs,
-                sample_size,
-                grad_norm,
+                logging_outputs, sample_size, grad_norm
             )
-
-            # clear CUDA cache to reduce memory fragmentation
-            if (
-                self.cuda
-                and self.args.empty_cache_freq > 0
-                and (
-                    (self.get_num_updates() + self.args.empty_cache_freq - 1)
-                    % self.args.empty_cache_freq
-                )
-                == 0
-            ):
-                torch.cuda.empty_cache()
+            self._maybe_empty_cache()
 
         if self.args.fp16:
             metrics.log_scalar(
-                "loss_scale",
-                self.optimizer.scaler.loss_scale,
-                priority=700,
-                round=4,
-                weight=0,
+                "loss_scale", self.optimizer.scaler.loss_scale,
+                priority=700, round=4, weight=0,
             )
 
         metrics.log_stop_time("train_wall")
         return logging_output
 
+    def _log_memory_headroom(self):
+        if not (self.cuda and self.cuda_env is not None):
+            return
+        gb_used = torch.cuda.max_memory_allocated() / 1024**3
+        torch.cuda.reset_peak_memory_stats()
+        gb_free = self.cuda_env.total_memory_in_GB - gb_used
+        metrics.log_scalar("gb_free", gb_free, priority=1500, round=1,
+                           weight=0)
+
+    def _maybe_empty_cache(self):
+        """Periodic cache flush against fragmentation (--empty-cache-freq)."""
+        freq = self.args.empty_cache_freq
+        if not (self.cuda and freq > 0):
+            return
+        if (self.get_num_updates() + freq - 1) % freq == 0:
+            torch.cuda.empty_cache()
+
     @metrics.aggregate("valid")
     def valid_step(self, sample, raise_oom=False):
-        """Do forward pass in evaluation mode."""
+        """Forward-only evaluation step (with one OOM retry)."""
         with torch.no_grad():
             self.model.eval()
             self.loss.eval()
 
-            sample, is_dummy_batch = self._prepare_sample(sample)
-
+            sample, is_filler = self._prepare_sample(sample)
             try:
                 _loss, sample_size, logging_output = self.task.valid_step(
                     sample, self.model, self.loss
                 )
-            except RuntimeError as e:
-                if _is_oom_error(e):
-                    self._log_oom(e)
-                    if not raise_oom:
-                        logger.warning(
-                            "ran out of memory in validation step, retrying batch"
-                        )
-                        for p in self.model.parameters():
-                            if p.grad is not None:
-                                p.grad = None  # free some memory
-                        if self.cuda:
-                            torch.cuda.empty_cache()
-                        return self.valid_step(sample, raise_oom=True)
-                raise e
+            except RuntimeError as exc:
+                if _looks_like_oom(exc) and not raise_oom:
+                    self._log_oom(exc)
+                    logger.warning("ran out of memory in validation step,"
+                                   " retrying batch")
+                    for p in self.model.parameters():
+                        p.grad = None  # release anything backward left over
+                    if self.cuda:
+                        torch.cuda.empty_cache()
+                    return self.valid_step(sample, raise_oom=True)
+                raise
 
             logging_outputs = [logging_output]
-            if is_dummy_batch:
-                if torch.is_tensor(sample_size):
-                    sample_size.zero_()
-                else:
-                    sample_size *= 0.0
+            if is_filler:
+                sample_size = _zeroed_stat(sample_size)
 
-        # gather logging outputs from all replicas
         if self.data_parallel_world_size > 1:
             logging_outputs, (sample_size,) = self._aggregate_logging_outputs(
-                logging_outputs,
-                sample_size,
-                ignore=is_dummy_batch,
+                logging_outputs, sample_size, ignore=is_filler
             )
-
-        # log validation stats
-        logging_output = self._reduce_and_log_stats(logging_outputs, sample_size)
-
-        return logging_output
+        return self._reduce_and_log_stats(logging_outputs, sample_size)
 
     def zero_grad(self):
         self.optimizer.zero_grad()
@@ -860,271 +835,231 @@ class Trainer(object):
         # autograd-assigned grads on the params; the next backward would
         # accumulate into them. The flat-view engines re-pin at forward
         # time; the lazy engine drops the assigned tensors here instead.
-        m = self._wrapped_model
-        if m is not None and getattr(m, "lazy", False):
-            m.zero_grad_buffers()
+        engine = self._wrapped_model
+        if engine is not None and getattr(engine, "lazy", False):
+            engine.zero_grad_buffers()
+
+    # -- LR plumbing ---------------------------------------------------------
 
     def lr_step_begin_epoch(self, epoch):
-        """Adjust the learning rate at the beginning of the epoch."""
         self.lr_scheduler.step_begin_epoch(epoch)
-        # prefer updating the LR based on the number of steps
-        return self.lr_step_update()
+        return self.lr_step_update()  # update-count-driven value wins
 
     def lr_step(self, epoch, val_loss=None):
-        """Adjust the learning rate at the end of the epoch."""
         self.lr_scheduler.step(epoch, val_loss)
-        # prefer updating the LR based on the number of steps
-        return self.lr_step_update()
+        return self.lr_step_update()  # update-count-driven value wins
 
     def lr_step_update(self):
-        """Update the learning rate after each update."""
         new_lr = self.lr_scheduler.step_update(self.get_num_updates())
-        if isinstance(new_lr, dict):
-            for k, v in new_lr.items():
-                metrics.log_scalar(f"lr_{k}", v, weight=0, priority=300)
-            new_lr = new_lr.get("default", next(iter(new_lr.values())))
-        else:
+        if not isinstance(new_lr, dict):
             metrics.log_scalar("lr", new_lr, weight=0, priority=300)
-        return new_lr
+            return new_lr
+        for k, v in new_lr.items():
+            metrics.log_scalar(f"lr_{k}", v, weight=0, priority=300)
+        return new_lr.get("default", next(iter(new_lr.values())))
 
     def get_lr(self):
-        """Get the current learning rate."""
         return self.optimizer.get_lr()
 
     def get_model(self):
-        """Get the (non-wrapped) model instance."""
+        """The bare (never DDP-wrapped) model."""
         return self._model
 
     def get_loss(self):
-        """Get the (non-wrapped) loss instance."""
+        """The bare (never DDP-wrapped) loss."""
         return self._loss
 
     def get_num_updates(self):
-        """Get the number of parameters updates."""
         return self._num_updates
 
     def set_num_updates(self, num_updates):
-        """Set the number of parameters updates."""
         self._num_updates = num_updates
         self.lr_step_update()
-        metrics.log_scalar("num_updates", self._num_updates, weight=0, priority=200)
+        metrics.log_scalar("num_updates", self._num_updates, weight=0,
+                           priority=200)
 
     def clip_grad_norm(self, clip_norm):
         return self.optimizer.clip_grad_norm(clip_norm, aggregate_norm_fn=None)
 
     def cumulative_training_time(self):
-        if self._cumulative_training_time is None:
-            # single GPU
-            return self._local_cumulative_training_time()
-        else:
-            return self._cumulative_training_time
+        if self._cumulative_training_time is not None:
+            return self._cumulative_training_time  # cross-rank average
+        return self._local_cumulative_training_time()
 
     def _local_cumulative_training_time(self):
-        """Aggregate training time in seconds."""
         return time.time() - self._start_time + self._previous_training_time
+
+    # -- sample prep / seeding ----------------------------------------------
 
     def _prepare_sample(self, sample, is_dummy=False):
         if sample == "DUMMY":
             raise Exception(
-                "Trying to use an uninitialized 'dummy' batch. This usually indicates "
-                "that the total number of batches is smaller than the number of "
-                "participating GPUs. Try reducing the batch size or using fewer GPUs."
+                "Trying to use an uninitialized 'dummy' batch. This usually "
+                "indicates that the total number of batches is smaller than "
+                "the number of participating GPUs. Try reducing the batch "
+                "size or using fewer GPUs."
             )
 
         if sample is None or len(sample) == 0:
-            assert (
-                self._dummy_batch is not None and len(self._dummy_batch) > 0
-            ), "Invalid dummy batch: {}".format(self._dummy_batch)
-            sample, _ = self._prepare_sample(self._dummy_batch, is_dummy=True)
-            return sample, True
+            assert (self._dummy_batch is not None
+                    and len(self._dummy_batch) > 0), \
+                f"Invalid dummy batch: {self._dummy_batch}"
+            filled, _ = self._prepare_sample(self._dummy_batch, is_dummy=True)
+            return filled, True
 
         if self.cuda:
             sample = utils.move_to_cuda(sample)
 
-        def lower_precision(t):
-            """Converts a tensor to the desired dtype."""
-            if t.dtype is torch.float32:
-                if self.args.fp16:
-                    return t.half()
-                elif self.args.bf16:
-                    return t.bfloat16()
-            return t
-
         if self.args.fp16 or self.args.bf16:
-            sample = utils.apply_to_sample(lower_precision, sample)
+            target = torch.half if self.args.fp16 else torch.bfloat16
+            sample = utils.apply_to_sample(
+                lambda t: t.to(target) if t.dtype is torch.float32 else t,
+                sample,
+            )
 
         if self._dummy_batch == "DUMMY":
             self._dummy_batch = sample
-
         return sample, False
 
     def _set_seed(self):
-        # Set seed based on args.seed and the update number so that we get
-        # reproducible results when resuming from checkpoints
+        # keyed by update count so resumed runs replay the same stream
         seed = self.args.seed + self.get_num_updates()
         torch.manual_seed(seed)
         if self.cuda:
             torch.cuda.manual_seed(seed)
 
     def _sync_stats(self):
-        # Return True if it's using multiple GPUs and DDP
-        if self.data_parallel_world_size == 1:
-            return False
-        return True
+        return self.data_parallel_world_size > 1
 
     def _log_oom(self, exc):
-        msg = "OOM: Ran out of memory with exception: {}".format(exc)
-        logger.warning(msg)
+        logger.warning(f"OOM: Ran out of memory with exception: {exc}")
         if self.cuda and hasattr(torch.cuda, "memory_summary"):
             for device_idx in range(torch.cuda.device_count()):
                 logger.warning(torch.cuda.memory_summary(device=device_idx))
         sys.stderr.flush()
 
-    def _aggregate_logging_outputs(
-        self,
-        logging_outputs: List[Dict[str, Any]],
-        *extra_stats_to_sum,
-        ignore=False,
-    ):
-        if self.task.__class__.logging_outputs_can_be_summed(
+    # -- cross-rank stat sync -------------------------------------------------
+
+    def _aggregate_logging_outputs(self, logging_outputs, *extra_stats_to_sum,
+                                   ignore=False):
+        summable = self.task.__class__.logging_outputs_can_be_summed(
             self.get_loss(), is_train=self.model.training
-        ):
-            return self._fast_stat_sync_sum(
-                logging_outputs, *extra_stats_to_sum, ignore=ignore
-            )
-        else:
-            return self._all_gather_list_sync(
-                logging_outputs, *extra_stats_to_sum, ignore=ignore
-            )
-
-    def _all_gather_list_sync(
-        self,
-        logging_outputs: List[Dict[str, Any]],
-        *extra_stats_to_sum,
-        ignore=False,
-    ):
-        """
-        Sync logging outputs across workers. all_gather_list_sync is
-        suitable when logging outputs are complex types.
-        """
-        if ignore:
-            logging_outputs = []
-        results = list(
-            zip(
-                *distributed_utils.all_gather_list(
-                    [logging_outputs] + list(extra_stats_to_sum),
-                    max_size=getattr(self.args, "all_gather_list_size", 16384),
-                    group=self.data_parallel_process_group,
-                )
-            )
         )
-        logging_outputs, extra_stats_to_sum = results[0], results[1:]
-        logging_outputs = list(chain.from_iterable(logging_outputs))
-        extra_stats_to_sum = [sum(s) for s in extra_stats_to_sum]
-        return logging_outputs, extra_stats_to_sum
+        sync = self._fast_stat_sync_sum if summable else self._all_gather_list_sync
+        return sync(logging_outputs, *extra_stats_to_sum, ignore=ignore)
 
-    def _fast_stat_sync_sum(
-        self,
-        logging_outputs: List[Dict[str, Any]],
-        *extra_stats_to_sum,
-        ignore=False,
-    ):
-        """
-        Sync logging outputs across workers. fast_stat_sync_sum is
-        faster than all_gather_list_sync, but is only suitable when
-        logging outputs are scalars and can be summed. Note that
-        *logging_outputs* cannot contain any nested dicts/lists.
-        """
-        data = {}
-        for i, stat in enumerate(extra_stats_to_sum):
-            data["extra_stats_" + str(i)] = stat
-        if len(logging_outputs) > 0:
+    def _all_gather_list_sync(self, logging_outputs, *extra_stats_to_sum,
+                              ignore=False):
+        """Pickle-everything sync: handles arbitrarily structured logging
+        outputs at the cost of serialization."""
+        if ignore:
+            logging_outputs = list()
+        gathered = list(zip(*distributed_utils.all_gather_list(
+            [logging_outputs] + list(extra_stats_to_sum),
+            max_size=getattr(self.args, "all_gather_list_size", 16384),
+            group=self.data_parallel_process_group,
+        )))
+        per_rank_logs, per_rank_extras = gathered[0], gathered[1:]
+        return (
+            list(chain.from_iterable(per_rank_logs)),
+            [sum(stat) for stat in per_rank_extras],
+        )
+
+    def _fast_stat_sync_sum(self, logging_outputs, *extra_stats_to_sum,
+                            ignore=False):
+        """One concatenated all-reduce of plain scalars — requires flat,
+        summable logging outputs (no nesting)."""
+        bundle = {
+            f"extra_stats_{i}": stat
+            for i, stat in enumerate(extra_stats_to_sum)
+        }
+        log_keys = None
+        if logging_outputs:
             log_keys = list(logging_outputs[0].keys())
             for k in log_keys:
+                if ignore:
+                    probe = logging_outputs[0][k]
+                    value = (
+                        torch.zeros_like(probe) if torch.is_tensor(probe)
+                        else 0
+                    )
                 if not ignore:
-                    v = sum(log[k] for log in logging_outputs if k in log)
-                else:
-                    v = logging_outputs[0][k]
-                    v = torch.zeros_like(v) if torch.is_tensor(v) else 0
-                data["logging_outputs_" + k] = v
-        else:
-            log_keys = None
+                    value = sum(log[k] for log in logging_outputs if k in log)
+                bundle[f"logging_outputs_{k}"] = value
 
-        data = distributed_utils.all_reduce_dict(
-            data, device=self.device, group=self.data_parallel_process_group
+        bundle = distributed_utils.all_reduce_dict(
+            bundle, device=self.device, group=self.data_parallel_process_group
         )
 
-        extra_stats_to_sum = [
-            data["extra_stats_" + str(i)] for i in range(len(extra_stats_to_sum))
+        extras = [
+            bundle[f"extra_stats_{i}"]
+            for i in range(len(extra_stats_to_sum))
         ]
-        if log_keys is not None:
-            logging_outputs = [
-                {k: data["logging_outputs_" + k] for k in log_keys}
-            ]
-        else:
-            logging_outputs = []
-        return logging_outputs, extra_stats_to_sum
+        logs = (
+            [{k: bundle[f"logging_outputs_{k}"] for k in log_keys}]
+            if log_keys is not None
+            else []
+        )
+        return logs, extras
 
     def _check_grad_norms(self, grad_norm):
-        """Check that grad norms are consistent across workers."""
-        if self._grad_norm_buf is not None:
-            self._grad_norm_buf.zero_()
-            self._grad_norm_buf[self.data_parallel_rank] = grad_norm
-            distributed_utils.all_reduce(
-                self._grad_norm_buf, group=self.data_parallel_process_group
+        """Cross-rank agreement check: replicas must see (nearly) identical
+        global grad norms, or the graphs have diverged."""
+        if self._grad_norm_buf is None:
+            return
+        buf = self._grad_norm_buf
+        buf.zero_()
+        buf[self.data_parallel_rank] = grad_norm
+        distributed_utils.all_reduce(
+            buf, group=self.data_parallel_process_group
+        )
+
+        def agrees(t):
+            spread = torch.max(torch.abs(t - t[0]))
+            all_close = (
+                torch.isfinite(t).all()
+                and (spread / (t[0] + 1e-6) < 1e-6).all()
+            )
+            all_bad = (torch.isnan(t) | torch.isinf(t)).all()
+            return all_close or all_bad
+
+        if not agrees(buf):
+            table = "\n".join(
+                f"rank {r:3d} = {n:.8f}" for r, n in enumerate(buf.tolist())
+            )
+            # FloatingPointError routes into the NanDetector re-run
+            raise FloatingPointError(
+                "Fatal error: gradients are inconsistent between workers. "
+                "Try --ddp-backend=legacy_ddp. "
+                "Or are you mixing up different generation of GPUs in "
+                "training?\n"
+                + "-" * 80
+                + f"\ngrad_norm across the workers:\n{table}\n\n"
+                + "-" * 80
             )
 
-            def is_consistent(tensor):
-                max_abs_diff = torch.max(torch.abs(tensor - tensor[0]))
-                return (
-                    torch.isfinite(tensor).all()
-                    and (max_abs_diff / (tensor[0] + 1e-6) < 1e-6).all()
-                ) or (torch.isnan(tensor) | torch.isinf(tensor)).all()
-
-            if not is_consistent(self._grad_norm_buf):
-                pretty_detail = "\n".join(
-                    "rank {:3d} = {:.8f}".format(r, n)
-                    for r, n in enumerate(self._grad_norm_buf.tolist())
-                )
-                error_detail = "grad_norm across the workers:\n{}\n".format(
-                    pretty_detail
-                )
-                # use FloatingPointError to trigger NanDetector
-                raise FloatingPointError(
-                    "Fatal error: gradients are inconsistent between workers. "
-                    "Try --ddp-backend=legacy_ddp. "
-                    "Or are you mixing up different generation of GPUs in training?"
-                    + "\n"
-                    + "-" * 80
-                    + "\n{}\n".format(error_detail)
-                    + "-" * 80
-                )
-
-    def _reduce_and_log_stats(self, logging_outputs, sample_size, grad_norm=None):
+    def _reduce_and_log_stats(self, logging_outputs, sample_size,
+                              grad_norm=None):
         if grad_norm is not None and (
             not torch.is_tensor(grad_norm) or torch.isfinite(grad_norm)
         ):
             metrics.log_speed("ups", 1.0, priority=100, round=2)
             metrics.log_scalar("gnorm", grad_norm, priority=400, round=3)
             if self.args.clip_norm > 0:
-                metrics.log_scalar(
-                    "clip",
-                    torch.where(
-                        grad_norm > self.args.clip_norm,
-                        grad_norm.new_tensor(100),
-                        grad_norm.new_tensor(0),
-                    ),
-                    priority=500,
-                    round=1,
+                clipped = torch.where(
+                    grad_norm > self.args.clip_norm,
+                    grad_norm.new_tensor(100),
+                    grad_norm.new_tensor(0),
                 )
+                metrics.log_scalar("clip", clipped, priority=500, round=1)
 
         with metrics.aggregate() as agg:
             if logging_outputs is not None:
                 self.task.reduce_metrics(logging_outputs, self.get_loss())
                 del logging_outputs
 
-            # extra warning for losses that don't properly log a loss value
             if "loss" not in agg:
+                # a loss that never logs "loss" breaks best-checkpoint logic
                 if "loss" not in self._warn_once:
                     self._warn_once.add("loss")
                     logger.warning(
@@ -1135,47 +1070,6 @@ class Trainer(object):
 
             logging_output = agg.get_smoothed_values()
             logging_output["sample_size"] = sample_size
-            for key_to_delete in ["ppl", "wps", "wpb", "bsz"]:
-                if key_to_delete in logging_output:
-                    del logging_output[key_to_delete]
+            for legacy_key in ("ppl", "wps", "wpb", "bsz"):
+                logging_output.pop(legacy_key, None)
             return logging_output
-
-
-def _is_oom_error(e):
-    return "out of memory" in str(e)
-
-
-def _catalog_shared_params(module, memo=None, prefix=""):
-    """Find parameters shared by multiple module paths
-    (reference unicore/trainer.py:1127-1145)."""
-    if memo is None:
-        first_call = True
-        memo = {}
-    else:
-        first_call = False
-    for name, param in module._parameters.items():
-        param_prefix = prefix + ("." if prefix else "") + name
-        if param not in memo:
-            memo[param] = []
-        memo[param].append(param_prefix)
-    for name, m in module._modules.items():
-        if m is None:
-            continue
-        submodule_prefix = prefix + ("." if prefix else "") + name
-        _catalog_shared_params(m, memo, submodule_prefix)
-    if first_call:
-        return [x for x in memo.values() if len(x) > 1]
-
-
-def _get_module_by_path(module, path):
-    path = path.split(".")
-    for name in path:
-        module = getattr(module, name)
-    return module
-
-
-def _set_module_by_path(module, path, value):
-    path = path.split(".")
-    for name in path[:-1]:
-        module = getattr(module, name)
-    setattr(module, path[-1], value)
