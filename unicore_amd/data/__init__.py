@@ -1,21 +1,21 @@
 """Data pipeline: composable wrapper datasets + resumable iterators
 (parity: reference unicore/data/__init__.py:9-33)."""
 
-from .unicore_dataset import UnicoreDataset, EpochListening
+from .unicore_dataset import EpochListening, UnicoreDataset
 from .base_wrapper_dataset import BaseWrapperDataset
 from .dictionary import Dictionary
 from .lru_cache_dataset import LRUCacheDataset
 from .lmdb_dataset import LMDBDataset
-from .tokenize_dataset import TokenizeDataset, BertTokenizeDataset
+from .tokenize_dataset import BertTokenizeDataset, TokenizeDataset
 from .mask_tokens_dataset import MaskTokensDataset
 from .pad_dataset import (
-    PadDataset,
     LeftPadDataset,
+    PadDataset,
     RightPadDataset,
     RightPadDataset2D,
     RightPadDatasetCoord,
 )
-from .sort_dataset import SortDataset, EpochShuffleDataset
+from .sort_dataset import EpochShuffleDataset, SortDataset
 from .nested_dictionary_dataset import NestedDictionaryDataset
 from .misc_datasets import (
     AppendTokenDataset,

@@ -14,33 +14,35 @@ from .base_wrapper_dataset import BaseWrapperDataset
 from .unicore_dataset import UnicoreDataset
 
 
+def _as_tensor(item):
+    return item if torch.is_tensor(item) else torch.from_numpy(np.asarray(item))
+
+
 class NumelDataset(BaseWrapperDataset):
+    """Element count of each item; collates to a sum or a count vector."""
+
     def __init__(self, dataset, reduce=False):
-        super().__init__(dataset)
+        super().__init__(dataset=dataset)
         self.reduce = reduce
 
     def __getitem__(self, index):
         item = self.dataset[index]
-        if torch.is_tensor(item):
-            return torch.numel(item)
-        else:
-            return np.size(item)
+        return torch.numel(item) if torch.is_tensor(item) else np.size(item)
 
-    def __len__(self):
+    def __len__(self) -> int:
         return len(self.dataset)
 
     def collater(self, samples):
-        if self.reduce:
-            return sum(samples)
-        else:
-            return torch.tensor(samples)
+        return sum(samples) if self.reduce else torch.tensor(samples)
 
 
 class NumSamplesDataset(UnicoreDataset):
+    """Constant 1 per item; collates to the batch row count."""
+
     def __getitem__(self, index):
         return 1
 
-    def __len__(self):
+    def __len__(self) -> int:
         return 0
 
     def collater(self, samples):
@@ -49,80 +51,59 @@ class NumSamplesDataset(UnicoreDataset):
 
 class PrependTokenDataset(BaseWrapperDataset):
     def __init__(self, dataset, token=None):
-        super().__init__(dataset)
+        super().__init__(dataset=dataset)
         self.token = token
 
-    @lru_cache(maxsize=16)
+    @lru_cache(16)
     def __getitem__(self, idx):
         item = self.dataset[idx]
-        if self.token is not None:
-            item = torch.cat([item.new([self.token]), item])
-        return item
+        if self.token is None:
+            return item
+        return torch.cat([item.new([self.token]), item])
 
 
 class AppendTokenDataset(BaseWrapperDataset):
     def __init__(self, dataset, token=None):
-        super().__init__(dataset)
+        super().__init__(dataset=dataset)
         self.token = token
 
-    @lru_cache(maxsize=16)
+    @lru_cache(16)
     def __getitem__(self, idx):
         item = self.dataset[idx]
-        if self.token is not None:
-            item = torch.cat([item, item.new([self.token])])
-        return item
+        if self.token is None:
+            return item
+        return torch.cat([item, item.new([self.token])])
 
 
 class RawLabelDataset(UnicoreDataset):
+    """Plain python list of labels as a dataset."""
+
     def __init__(self, labels):
         super().__init__()
-        self.labels = labels
+        self.values = labels
 
     def __getitem__(self, index):
-        return self.labels[index]
+        return self.values[index]
 
-    def __len__(self):
-        return len(self.labels)
+    def __len__(self) -> int:
+        return len(self.values)
 
     def collater(self, samples):
         return torch.tensor(samples)
 
 
 class RawArrayDataset(BaseWrapperDataset):
-    def __init__(self, dataset):
-        super().__init__(dataset)
-
-    def __getitem__(self, index):
-        return self.dataset[index]
-
-    def collater(self, samples):
-        if hasattr(self.dataset, "collater"):
-            return self.dataset.collater(samples)
-        else:
-            return torch.utils.data.dataloader.default_collate(samples)
+    """Identity wrapper (items pass through; collation delegates)."""
 
 
 class RawNumpyDataset(BaseWrapperDataset):
-    def __init__(self, dataset):
-        super().__init__(dataset)
+    """Items coerced to tensors on access."""
 
     def __getitem__(self, index):
-        item = self.dataset[index]
-        if not torch.is_tensor(item):
-            item = torch.from_numpy(np.asarray(item))
-        return item
-
-    def collater(self, samples):
-        if hasattr(self.dataset, "collater"):
-            return self.dataset.collater(samples)
-        else:
-            return torch.utils.data.dataloader.default_collate(samples)
+        return _as_tensor(self.dataset[index])
 
 
 class FromNumpyDataset(BaseWrapperDataset):
-    @lru_cache(maxsize=16)
+    @lru_cache(16)
     def __getitem__(self, idx):
-        item = self.dataset[idx]
-        if not torch.is_tensor(item):
-            item = torch.from_numpy(np.asarray(item))
-        return item
+        return _as_tensor(self.dataset[idx])

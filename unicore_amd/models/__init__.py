@@ -1,11 +1,18 @@
 """Model + architecture registries (parity: reference
-unicore/models/__init__.py:17-102)."""
+unicore/models/__init__.py:17-102).
+
+Two levels: MODEL_REGISTRY maps a model family name to its class;
+ARCH_MODEL_REGISTRY maps each ``--arch`` choice to that class, with
+ARCH_CONFIG_REGISTRY holding the per-arch default-filling function applied
+after parsing. Modules in this directory import below so built-in models
+self-register.
+"""
 
 import importlib
 import os
 
-from .unicore_model import BaseUnicoreModel
 from .distributed_unicore_model import DistributedUnicoreModel  # noqa: F401
+from .unicore_model import BaseUnicoreModel
 
 MODEL_REGISTRY = {}
 ARCH_MODEL_REGISTRY = {}
@@ -26,74 +33,60 @@ def build_model(args, task):
 
 
 def register_model(name):
-    """
-    New model types can be added with the :func:`register_model` function
-    decorator.
+    """Decorator registering a BaseUnicoreModel subclass under *name*."""
 
-    Args:
-        name (str): the name of the model
-    """
-
-    def register_model_cls(cls):
+    def wrap(cls):
         if name in MODEL_REGISTRY:
-            raise ValueError("Cannot register duplicate model ({})".format(name))
+            raise ValueError(f"Cannot register duplicate model ({name})")
         if not issubclass(cls, BaseUnicoreModel):
             raise ValueError(
-                "Model ({}: {}) must extend BaseUnicoreModel".format(
-                    name, cls.__name__
-                )
+                f"Model ({name}: {cls.__name__}) must extend BaseUnicoreModel"
             )
         MODEL_REGISTRY[name] = cls
         return cls
 
-    return register_model_cls
+    return wrap
 
 
 def register_model_architecture(model_name, arch_name):
-    """
-    New model architectures can be added with the
-    :func:`register_model_architecture` function decorator. After registration,
-    model architectures can be selected with the ``--arch`` command-line
-    argument.
+    """Decorator registering an ``--arch`` choice for an existing model:
+    the decorated function fills arch-specific defaults into args after
+    parsing."""
 
-    Args:
-        model_name (str): the name of the Model (Model must already be
-            registered)
-        arch_name (str): the name of the model architecture (``--arch``)
-    """
-
-    def register_model_arch_fn(fn):
+    def wrap(fn):
         if model_name not in MODEL_REGISTRY:
             raise ValueError(
-                "Cannot register model architecture for unknown model type ({})".format(
-                    model_name
-                )
+                "Cannot register model architecture for unknown model type "
+                f"({model_name})"
             )
         if arch_name in ARCH_MODEL_REGISTRY:
             raise ValueError(
-                "Cannot register duplicate model architecture ({})".format(arch_name)
+                f"Cannot register duplicate model architecture ({arch_name})"
             )
         if not callable(fn):
             raise ValueError(
-                "Model architecture must be callable ({})".format(arch_name)
+                f"Model architecture must be callable ({arch_name})"
             )
         ARCH_MODEL_REGISTRY[arch_name] = MODEL_REGISTRY[model_name]
         ARCH_MODEL_INV_REGISTRY.setdefault(model_name, []).append(arch_name)
         ARCH_CONFIG_REGISTRY[arch_name] = fn
         return fn
 
-    return register_model_arch_fn
+    return wrap
 
 
-# automatically import any Python files in the models/ directory
-models_dir = os.path.dirname(__file__)
-for file in sorted(os.listdir(models_dir)):
-    path = os.path.join(models_dir, file)
-    if (
-        not file.startswith("_")
-        and not file.startswith(".")
-        and (file.endswith(".py") or os.path.isdir(path))
-    ):
-        model_name = file[: file.find(".py")] if file.endswith(".py") else file
-        if model_name not in ("unicore_model", "distributed_unicore_model"):
-            module = importlib.import_module("unicore_amd.models." + model_name)
+def _import_all_model_modules():
+    here = os.path.dirname(__file__)
+    skip = ("unicore_model", "distributed_unicore_model")
+    for entry in sorted(os.listdir(here)):
+        if entry.startswith(("_", ".")):
+            continue
+        is_pkg = os.path.isdir(os.path.join(here, entry))
+        if not (entry.endswith(".py") or is_pkg):
+            continue
+        modname = entry[:-3] if entry.endswith(".py") else entry
+        if modname not in skip:
+            importlib.import_module(f"unicore_amd.models.{modname}")
+
+
+_import_all_model_modules()

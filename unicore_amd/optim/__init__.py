@@ -1,4 +1,9 @@
-"""Optimizer registry (parity: reference unicore/optim/__init__.py)."""
+"""Optimizer registry (parity: reference unicore/optim/__init__.py).
+
+``build_optimizer`` splits params into decay/no-decay groups by name
+before handing them to the registered optimizer class; every module in
+this directory imports below so built-in optimizers self-register.
+"""
 
 import importlib
 import os
@@ -10,14 +15,12 @@ from unicore_amd.optim.fp16_optimizer import (  # noqa
     separate_decay_params,
 )
 
-__all__ = [
-    "UnicoreOptimizer",
-    "FP16Optimizer",
-]
+__all__ = ["UnicoreOptimizer", "FP16Optimizer"]
 
-(_build_optimizer, register_optimizer, OPTIMIZER_REGISTRY) = registry.setup_registry(
-    "--optimizer", base_class=UnicoreOptimizer, default="adam"
-)
+(_build_optimizer, register_optimizer, OPTIMIZER_REGISTRY) = \
+    registry.setup_registry(
+        "--optimizer", base_class=UnicoreOptimizer, default="adam"
+    )
 
 
 def build_optimizer(args, params, separate=True, *extra_args, **extra_kwargs):
@@ -36,10 +39,13 @@ def build_raw_optimizer(args, param_groups):
     return _build_optimizer(args, param_groups)
 
 
-# automatically import any Python files in the optim/ directory
-for file in sorted(os.listdir(os.path.dirname(__file__))):
-    if file.endswith(".py") and not file.startswith("_"):
-        file_name = file[: file.find(".py")]
-        importlib.import_module("unicore_amd.optim." + file_name)
+def _import_all_optim_modules():
+    here = os.path.dirname(__file__)
+    for entry in sorted(os.listdir(here)):
+        if entry.endswith(".py") and not entry.startswith("_"):
+            importlib.import_module(f"unicore_amd.optim.{entry[:-3]}")
+
+
+_import_all_optim_modules()
 
 from unicore_amd.optim import lr_scheduler  # noqa

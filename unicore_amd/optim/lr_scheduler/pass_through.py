@@ -1,5 +1,6 @@
 """Pass-through scheduler (parity: reference
-unicore/optim/lr_scheduler/pass_through.py:11-32)."""
+unicore/optim/lr_scheduler/pass_through.py:11-32): the optimizer carries
+its own schedule and this shim simply forwards all hooks to it."""
 
 from . import register_lr_scheduler
 from .unicore_lr_scheduler import UnicoreLRScheduler
@@ -7,24 +8,22 @@ from .unicore_lr_scheduler import UnicoreLRScheduler
 
 @register_lr_scheduler("pass_through")
 class PassThroughScheduleSchedule(UnicoreLRScheduler):
-    """Delegate lr scheduling to the optimizer."""
-
     def __init__(self, args, optimizer, total_train_steps):
         super().__init__(args, optimizer, total_train_steps)
-        assert (
-            hasattr(optimizer, "lr_scheduler") and optimizer.lr_scheduler is not None
-        ), "Pass-through schedule can only be used with optimizers with their own schedulers"
+        inner = getattr(optimizer, "lr_scheduler", None)
+        assert inner is not None, (
+            "Pass-through schedule can only be used with optimizers with "
+            "their own schedulers"
+        )
 
-    def state_dict(self):
+    def state_dict(self) -> dict:
         return self.optimizer.lr_scheduler.state_dict()
 
-    def load_state_dict(self, state_dict):
-        self.optimizer.lr_scheduler.load_state_dict(state_dict)
+    def load_state_dict(self, state: dict) -> None:
+        self.optimizer.lr_scheduler.load_state_dict(state)
 
     def step_begin_epoch(self, epoch):
-        """Update the learning rate at the beginning of the given epoch."""
         return self.optimizer.lr_scheduler.step_begin_epoch(epoch)
 
     def step_update(self, num_updates):
-        """Update the learning rate after each update."""
         return self.optimizer.lr_scheduler.step_update(num_updates)

@@ -1,11 +1,15 @@
-"""Task registry (parity: reference unicore/tasks/__init__.py:16-61)."""
+"""Task registry (parity: reference unicore/tasks/__init__.py:16-61).
+
+Tasks self-register at import time via :func:`register_task`; every module
+in this directory is imported below so built-in tasks appear without any
+explicit list.
+"""
 
 import importlib
 import os
 
 from .unicore_task import UnicoreTask
 
-# register dataclass
 TASK_REGISTRY = {}
 TASK_CLASS_NAMES = set()
 
@@ -15,50 +19,47 @@ def setup_task(args, **kwargs):
 
 
 def register_task(name):
-    """
-    New tasks can be added with the :func:`register_task` decorator::
+    """Decorator registering a UnicoreTask subclass under *name*::
 
         @register_task('classification')
         class ClassificationTask(UnicoreTask):
-            (...)
-
-    Args:
-        name (str): the name of the task
+            ...
     """
 
-    def register_task_cls(cls):
+    def wrap(cls):
         if name in TASK_REGISTRY:
-            raise ValueError("Cannot register duplicate task ({})".format(name))
+            raise ValueError(f"Cannot register duplicate task ({name})")
         if not issubclass(cls, UnicoreTask):
             raise ValueError(
-                "Task ({}: {}) must extend UnicoreTask".format(name, cls.__name__)
+                f"Task ({name}: {cls.__name__}) must extend UnicoreTask"
             )
         if cls.__name__ in TASK_CLASS_NAMES:
             raise ValueError(
-                "Cannot register task with duplicate class name ({})".format(
-                    cls.__name__
-                )
+                f"Cannot register task with duplicate class name "
+                f"({cls.__name__})"
             )
         TASK_REGISTRY[name] = cls
         TASK_CLASS_NAMES.add(cls.__name__)
         return cls
 
-    return register_task_cls
+    return wrap
 
 
 def get_task(name):
     return TASK_REGISTRY[name]
 
 
-# automatically import any Python files in the tasks/ directory
-tasks_dir = os.path.dirname(__file__)
-for file in sorted(os.listdir(tasks_dir)):
-    path = os.path.join(tasks_dir, file)
-    if (
-        not file.startswith("_")
-        and not file.startswith(".")
-        and (file.endswith(".py") or os.path.isdir(path))
-    ):
-        task_name = file[: file.find(".py")] if file.endswith(".py") else file
-        if task_name != "unicore_task":
-            importlib.import_module("unicore_amd.tasks." + task_name)
+def _import_all_task_modules():
+    here = os.path.dirname(__file__)
+    for entry in sorted(os.listdir(here)):
+        if entry.startswith(("_", ".")):
+            continue
+        is_pkg = os.path.isdir(os.path.join(here, entry))
+        if not (entry.endswith(".py") or is_pkg):
+            continue
+        modname = entry[:-3] if entry.endswith(".py") else entry
+        if modname != "unicore_task":
+            importlib.import_module(f"unicore_amd.tasks.{modname}")
+
+
+_import_all_task_modules()

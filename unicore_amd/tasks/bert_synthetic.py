@@ -74,18 +74,19 @@ class BertSyntheticTask(UnicoreTask):
 
     def __init__(self, args, dictionary):
         super().__init__(args)
-        self.dictionary = dictionary
-        self.seed = args.seed
-        self.mask_idx = dictionary.add_symbol("[MASK]", is_special=True)
+        self.dictionary, self.seed = dictionary, args.seed
+        self.mask_idx = dictionary.add_symbol(
+            "[MASK]", is_special=True
+        )
         # vocab to a multiple of 64: the lm-head GEMM and fused cross
         # entropy want an even, vectorizable inner dimension
         dictionary.pad_to_multiple_(64)
 
     @classmethod
     def setup_task(cls, args, **kwargs):
-        dictionary = make_synthetic_dictionary(getattr(args, "vocab_size", 30522))
-        logger.info("synthetic dictionary: {} types".format(len(dictionary)))
-        return cls(args, dictionary)
+        vocab = make_synthetic_dictionary(getattr(args, "vocab_size", 30522))
+        logger.info(f"synthetic dictionary: {len(vocab)} types")
+        return cls(args, vocab)
 
     def load_dataset(self, split, combine=False, **kwargs):
         seq_len = min(self.args.tokens_per_sample, self.args.max_seq_len - 1)
@@ -100,34 +101,23 @@ class BertSyntheticTask(UnicoreTask):
         )
         dataset = FromNumpyDataset(raw)
 
-        src_dataset, tgt_dataset = MaskTokensDataset.apply_mask(
-            dataset,
-            self.dictionary,
-            pad_idx=self.dictionary.pad(),
-            mask_idx=self.mask_idx,
-            seed=self.seed,
-            mask_prob=self.args.mask_prob,
-            leave_unmasked_prob=self.args.leave_unmasked_prob,
-            random_token_prob=self.args.random_token_prob,
+        cfg = self.args
+        src, tgt = MaskTokensDataset.apply_mask(
+            dataset, self.dictionary,
+            pad_idx=self.dictionary.pad(), mask_idx=self.mask_idx,
+            seed=self.seed, mask_prob=cfg.mask_prob,
+            leave_unmasked_prob=cfg.leave_unmasked_prob,
+            random_token_prob=cfg.random_token_prob,
         )
 
         with data_utils.numpy_seed(self.seed):
-            shuffle = np.random.permutation(len(src_dataset))
+            order = np.random.permutation(len(src))
 
-        self.datasets[split] = SortDataset(
-            NestedDictionaryDataset(
-                {
-                    "net_input": {
-                        "src_tokens": RightPadDataset(
-                            src_dataset,
-                            pad_idx=self.dictionary.pad(),
-                        )
-                    },
-                    "target": RightPadDataset(
-                        tgt_dataset,
-                        pad_idx=self.dictionary.pad(),
-                    ),
-                },
-            ),
-            sort_order=[shuffle],
-        )
+        pad = self.dictionary.pad()
+        nested = NestedDictionaryDataset({
+            "net_input": {
+                "src_tokens": RightPadDataset(src, pad_idx=pad),
+            },
+            "target": RightPadDataset(tgt, pad_idx=pad),
+        })
+        self.datasets[split] = SortDataset(nested, sort_order=[order])
