@@ -1,0 +1,196 @@
+"""REAL multi-rank RCCL on a single MI355X.
+
+Two processes build a genuine world_size=2 RCCL group on one GPU and run
+FlatDDP's actual comm-stream machinery under live concurrent all-reduces —
+the path the faked-world_size tests could not cover: bucket ordering under
+real RCCL, side-stream/hipEvent synchronization, no_sync grad accumulation,
+and a full bf16 trainer step with cross-rank grad-norm consistency.
+
+Each child writes its results to a file; the parent asserts. Children run
+via torch.multiprocessing.spawn with MASTER_ADDR=127.0.0.1.
+"""
+
+import json
+import os
+import sys
+import tempfile
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(
+    not torch.cuda.is_available(), reason="needs an MI355X"
+)
+
+WORLD = 2
+
+
+def _init_rccl(rank, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    # both ranks share cuda:0 on a 1-GPU box
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=rank, world_size=WORLD)
+    warm = torch.ones(1, device="cuda")
+    dist.all_reduce(warm)
+    torch.cuda.synchronize()
+    return dist
+
+
+def _flat_ddp_worker(rank, port, out_dir):
+    """Grad parity: each rank backprops its own batch through FlatDDP; the
+    synced grads must equal the average of both per-rank reference grads
+    (every rank can recompute both references deterministically)."""
+    dist = _init_rccl(rank, port)
+    try:
+        from unicore_amd.distributed import FlatDDP
+
+        torch.manual_seed(7)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(64, 128), torch.nn.GELU(),
+            torch.nn.Linear(128, 64),
+        ).cuda()
+        ddp = FlatDDP(model, process_group=dist.group.WORLD,
+                      bucket_cap_mb=0.0005)  # tiny buckets: several reduces
+
+        # per-rank batches are derived from the rank so both ranks can
+        # rebuild both references locally
+        def batch_for(r):
+            g = torch.Generator(device="cpu").manual_seed(100 + r)
+            return torch.randn(8, 64, generator=g).cuda()
+
+        reference = []
+        for r in range(WORLD):
+            clone = torch.nn.Sequential(
+                torch.nn.Linear(64, 128), torch.nn.GELU(),
+                torch.nn.Linear(128, 64),
+            ).cuda()
+            clone.load_state_dict(model.state_dict())
+            clone(batch_for(r)).pow(2).mean().backward()
+            reference.append([p.grad.clone() for p in clone.parameters()])
+
+        # pass 1: plain synced backward
+        ddp(batch_for(rank)).pow(2).mean().backward()
+        ddp.finish_grad_sync()
+        torch.cuda.synchronize()
+        max_err = 0.0
+        for i, p in enumerate(model.parameters()):
+            want = (reference[0][i] + reference[1][i]) / WORLD
+            max_err = max(max_err, (p.grad - want).abs().max().item())
+
+        # pass 2: grad accumulation — no_sync for the first micro-batch,
+        # synced on the second; result = mean over ranks of (sum of two
+        # local micro-grads)
+        ddp.zero_grad_buffers()
+        for p in model.parameters():
+            p.grad = None
+        with ddp.no_sync():
+            ddp(batch_for(rank)).pow(2).mean().backward()
+        ddp(batch_for(rank)).pow(2).mean().backward()
+        ddp.finish_grad_sync()
+        torch.cuda.synchronize()
+        max_err2 = 0.0
+        for i, p in enumerate(model.parameters()):
+            want = (reference[0][i] + reference[1][i]) * 2 / WORLD
+            max_err2 = max(max_err2, (p.grad - want).abs().max().item())
+
+        with open(os.path.join(out_dir, f"rank{rank}.json"), "w") as f:
+            json.dump({"max_err": max_err, "max_err_accum": max_err2}, f)
+    finally:
+        dist.destroy_process_group()
+
+
+def _trainer_worker(rank, port, out_dir):
+    """Full bf16 trainer integration at real world_size=2: three updates of
+    BERT-tiny through FlatDDP + FP16Optimizer; records per-step losses
+    (must agree across ranks thanks to the stat sync) and the grad-norm
+    consistency check must pass."""
+    dist = _init_rccl(rank, port)
+    try:
+        from unicore_amd import options, tasks
+        from unicore_amd.trainer import Trainer
+
+        argv = [
+            "--task", "bert_synthetic",
+            "--arch", "bert_base",
+            "--loss", "masked_lm",
+            "--optimizer", "adam",
+            "--lr-scheduler", "fixed",
+            "--lr", "1e-4",
+            "--batch-size", "4",
+            "--dataset-size", "32",
+            "--tokens-per-sample", "64",
+            "--max-seq-len", "66",
+            "--vocab-size", "512",
+            "--encoder-layers", "2",
+            "--encoder-embed-dim", "128",
+            "--encoder-ffn-embed-dim", "256",
+            "--encoder-attention-heads", "2",
+            "--log-format", "none",
+            "--num-workers", "0",
+            "--seed", "3",
+            "--bf16",
+            "--clip-norm", "1.0",
+            "--ddp-backend", "c10d",
+        ]
+        parser = options.get_training_parser()
+        args = options.parse_args_and_arch(parser, input_args=argv)
+        args.distributed_world_size = WORLD
+        args.distributed_rank = rank
+        args.device_id = 0
+        args.distributed_no_spawn = True
+
+        torch.manual_seed(args.seed)
+        task = tasks.setup_task(args)
+        task.load_dataset("train")
+        model = task.build_model(args)
+        loss = task.build_loss(args)
+        trainer = Trainer(args, task, model, loss)
+        epoch_itr = trainer.get_train_iterator(epoch=1)
+        trainer.init_total_train_steps(epoch_itr)
+        itr = epoch_itr.next_epoch_itr(shuffle=False)
+
+        losses = []
+        for step in range(3):
+            out = trainer.train_step([next(itr)])
+            assert out is not None, "update was skipped"
+            losses.append(float(out["loss"]))
+        torch.cuda.synchronize()
+
+        with open(os.path.join(out_dir, f"trainer_rank{rank}.json"), "w") as f:
+            json.dump({"losses": losses}, f)
+    finally:
+        dist.destroy_process_group()
+
+
+def _spawn(target, port, out_dir):
+    import torch.multiprocessing as mp
+
+    mp.spawn(target, args=(port, out_dir), nprocs=WORLD, join=True)
+
+
+@requires_gpu
+def test_flat_ddp_two_ranks_one_gpu(tmp_path):
+    _spawn(_flat_ddp_worker, 29511, str(tmp_path))
+    for rank in range(WORLD):
+        with open(tmp_path / f"rank{rank}.json") as f:
+            res = json.load(f)
+        assert res["max_err"] < 1e-5, res
+        assert res["max_err_accum"] < 1e-5, res
+
+
+@requires_gpu
+def test_trainer_two_ranks_one_gpu(tmp_path):
+    _spawn(_trainer_worker, 29513, str(tmp_path))
+    records = []
+    for rank in range(WORLD):
+        with open(tmp_path / f"trainer_rank{rank}.json") as f:
+            records.append(json.load(f)["losses"])
+    # the cross-rank stat sync must make the reported losses identical
+    assert records[0] == pytest.approx(records[1], rel=1e-6)
+    # and the trajectory must be finite and moving
+    assert all(l == l and l != float("inf") for l in records[0])
