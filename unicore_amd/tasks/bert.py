@@ -54,11 +54,27 @@ class BertTask(UnicoreTask):
         logger.info(f"dictionary: {len(vocab)} types")
         return cls(args, vocab)
 
+    def _open_split(self, data_dir, split):
+        """{split}.lmdb (reference format) or {split}.kv (our lmdb-free
+        single-file store) — whichever exists."""
+        lmdb_path = os.path.join(data_dir, f"{split}.lmdb")
+        if os.path.isfile(lmdb_path):
+            return LMDBDataset(lmdb_path)
+        kv_path = os.path.join(data_dir, f"{split}.kv")
+        if os.path.isfile(kv_path):
+            from unicore_amd.data.kv_dataset import KVDataset
+
+            return KVDataset(kv_path)
+        raise FileNotFoundError(
+            f"no {split}.lmdb or {split}.kv under {data_dir} "
+            "(run examples/bert/prepare_corpus.py)"
+        )
+
     def load_dataset(self, split, combine=False, **kwargs):
-        """Build the {split}.lmdb pipeline and register it under *split*."""
+        """Build the {split} pipeline and register it under *split*."""
         data_dir = self.args.data
         tokens = BertTokenizeDataset(
-            LMDBDataset(os.path.join(data_dir, f"{split}.lmdb")),
+            self._open_split(data_dir, split),
             os.path.join(data_dir, "dict.txt"),
             max_seq_len=self.args.max_seq_len,
         )
