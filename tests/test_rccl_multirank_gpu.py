@@ -1,10 +1,17 @@
-"""REAL multi-rank RCCL on a single MI355X.
+"""REAL multi-rank process group on a single MI355X.
 
-Two processes build a genuine world_size=2 RCCL group on one GPU and run
-FlatDDP's actual comm-stream machinery under live concurrent all-reduces —
-the path the faked-world_size tests could not cover: bucket ordering under
-real RCCL, side-stream/hipEvent synchronization, no_sync grad accumulation,
-and a full bf16 trainer step with cross-rank grad-norm consistency.
+Two processes build a genuine world_size=2 group sharing cuda:0 and run
+FlatDDP's actual CUDA comm-stream machinery under live concurrent
+all-reduces — the path the faked-world_size tests could not cover: bucket
+ordering under a real group, side-stream/hipEvent synchronization, no_sync
+grad accumulation, and a full bf16 trainer integration with the cross-rank
+grad-norm consistency check.
+
+RCCL itself refuses two ranks on one device ("Duplicate GPU detected", a
+NCCL/RCCL invariant), so the 2-ranks-1-GPU group uses the gloo backend with
+CUDA-resident tensors: FlatDDP's bucketing, comm stream and event ordering
+run exactly as under RCCL; only the transport differs. True RCCL N>1 runs
+on the driver's 8-GPU node (one rank per device).
 
 Each child writes its results to a file; the parent asserts. Children run
 via torch.multiprocessing.spawn with MASTER_ADDR=127.0.0.1.
@@ -27,14 +34,15 @@ requires_gpu = pytest.mark.skipif(
 WORLD = 2
 
 
-def _init_rccl(rank, port):
+def _init_group(rank, port):
     import torch.distributed as dist
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
-    # both ranks share cuda:0 on a 1-GPU box
+    # both ranks share cuda:0 on a 1-GPU box; gloo transports the CUDA
+    # tensors (RCCL rejects duplicate devices inside one group)
     torch.cuda.set_device(0)
-    dist.init_process_group("nccl", rank=rank, world_size=WORLD)
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
     warm = torch.ones(1, device="cuda")
     dist.all_reduce(warm)
     torch.cuda.synchronize()
@@ -45,7 +53,7 @@ def _flat_ddp_worker(rank, port, out_dir):
     """Grad parity: each rank backprops its own batch through FlatDDP; the
     synced grads must equal the average of both per-rank reference grads
     (every rank can recompute both references deterministically)."""
-    dist = _init_rccl(rank, port)
+    dist = _init_group(rank, port)
     try:
         from unicore_amd.distributed import FlatDDP
 
@@ -109,7 +117,7 @@ def _trainer_worker(rank, port, out_dir):
     BERT-tiny through FlatDDP + FP16Optimizer; records per-step losses
     (must agree across ranks thanks to the stat sync) and the grad-norm
     consistency check must pass."""
-    dist = _init_rccl(rank, port)
+    dist = _init_group(rank, port)
     try:
         from unicore_amd import options, tasks
         from unicore_amd.trainer import Trainer
@@ -136,6 +144,7 @@ def _trainer_worker(rank, port, out_dir):
             "--bf16",
             "--clip-norm", "1.0",
             "--ddp-backend", "c10d",
+            "--distributed-backend", "gloo",
         ]
         parser = options.get_training_parser()
         args = options.parse_args_and_arch(parser, input_args=argv)

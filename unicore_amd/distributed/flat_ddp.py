@@ -83,6 +83,14 @@ class FlatDDP(nn.Module):
         self._comm_stream = (
             torch.cuda.Stream() if self._use_cuda and self.world_size > 1 else None
         )
+        # RCCL collectives are ordered on the stream they were launched on;
+        # gloo (CUDA tensors staged through the host) is NOT — its work
+        # handle must be waited in addition to the comm-stream event
+        try:
+            backend = dist.get_backend(process_group) if dist.is_initialized() else ""
+        except Exception:
+            backend = ""
+        self._stream_ordered_comm = str(backend) == "nccl"
 
         self._buckets: List[_Bucket] = []
         self._param_to_bucket = {}
@@ -275,7 +283,12 @@ class FlatDDP(nn.Module):
             with torch.cuda.stream(self._comm_stream):
                 ev.wait(self._comm_stream)
                 bucket.flat.div_(self.world_size)
-                dist.all_reduce(bucket.flat, group=self.process_group)
+                work = dist.all_reduce(
+                    bucket.flat, group=self.process_group, async_op=True
+                )
+                if not self._stream_ordered_comm:
+                    # host-staged backend: completion is not stream-visible
+                    bucket.work = work
                 bucket.event = torch.cuda.Event()
                 bucket.event.record(self._comm_stream)
         else:
@@ -313,10 +326,10 @@ class FlatDDP(nn.Module):
             if not b.launched:
                 self._launch_reduce(b)
         for b in self._buckets:
-            if b.event is not None:
-                b.event.wait(torch.cuda.current_stream())
             if b.work is not None:
                 b.work.wait()
+            if b.event is not None:
+                b.event.wait(torch.cuda.current_stream())
 
     def zero_grad_buffers(self):
         """Reset grad state for a fresh step.  Lazy mode only drops the
