@@ -182,6 +182,14 @@ def main():
 
     if use_cuda:
         torch.cuda.set_device(local_rank)
+        # offline-tuned GEMM algorithm table (tools/tunableop_pershape.sh);
+        # opt out with UNICORE_NO_TUNED_GEMM=1
+        if os.environ.get("UNICORE_NO_TUNED_GEMM", "0") != "1":
+            from unicore_amd.utils import load_gemm_tunings
+
+            tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                 "tools", "tuned_gemm_bert.csv")
+            load_gemm_tunings(tuned)
     if world_size > 1:
         import torch.distributed as dist
 

@@ -587,6 +587,7 @@ __global__ void flash_dot_do_o_kernel(float* __restrict__ di,
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    float* __restrict__ dbias_acc,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
     const float* __restrict__ lse, const float* __restrict__ di,
@@ -705,12 +706,24 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
       keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
                        kv0 + ks2 * 32 + lg * 8, pthresh, keep);
       bf16x8 dsa;
+      float dsf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const float pv = bf16_bits_to_f32((uint16_t)(unsigned short)pa[j]);
         float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
         if (DROP) dpv = keep[j] ? dpv * pinv : 0.f;
-        dsa[j] = (short)f32_to_bf16_bits(pv * (dpv - di_row));
+        dsf[j] = pv * (dpv - di_row);
+        dsa[j] = (short)f32_to_bf16_bits(dsf[j]);
+      }
+      if (HAS_BIAS && dbias_acc != nullptr) {
+        // fused dbias: accumulate dS straight into the (nb, L, L) fp32
+        // bias-grad buffer (pre-rounding values) instead of materializing
+        // the (BH, L, L) dS tensor for a torch reduction
+        const int64_t brow =
+            ((bh / bias_od) % bias_nb) * (int64_t)bias_q + ((q0 + lr) % bias_q);
+        float* bdst = dbias_acc + brow * (int64_t)L + kv0 + ks2 * 32 + lg * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) atomicAdd(bdst + j, dsf[j]);
       }
       if (ds_out != nullptr) {
         union {
@@ -748,6 +761,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
 __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    float* __restrict__ dbias_acc,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
     const float* __restrict__ lse, const float* __restrict__ di,
@@ -899,13 +913,24 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
             load_frag(&lds_t[wid][mtile][lr][ks2 * 32 + lg * 8]);
         keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + mtile * 16 + lr),
                          kv0 + ks2 * 32 + lg * 8, pthresh, keep[mtile]);
+        float dsf[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float pv =
               bf16_bits_to_f32((uint16_t)(unsigned short)pa[mtile][ks2][j]);
           float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
           if (DROP) dpv = keep[mtile][j] ? dpv * pinv : 0.f;
-          dsa[mtile][j] = (short)f32_to_bf16_bits(pv * (dpv - di_row[mtile]));
+          dsf[j] = pv * (dpv - di_row[mtile]);
+          dsa[mtile][j] = (short)f32_to_bf16_bits(dsf[j]);
+        }
+        if (HAS_BIAS && dbias_acc != nullptr) {
+          // fused dbias accumulation (see kres variant)
+          const int64_t brow = ((bh / bias_od) % bias_nb) * (int64_t)bias_q +
+                               ((q0 + mtile * 16 + lr) % bias_q);
+          float* bdst =
+              dbias_acc + brow * (int64_t)L + kv0 + ks2 * 32 + lg * 8;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) atomicAdd(bdst + j, dsf[j]);
         }
         if (ds_out != nullptr) {
           union {
@@ -948,6 +973,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
+    float* __restrict__ dbias_acc,
     const uint16_t* __restrict__ dop,
     const uint16_t* __restrict__ qp, const uint16_t* __restrict__ kp,
     const uint16_t* __restrict__ vp, const float* __restrict__ lse,
@@ -1066,15 +1092,25 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
       keep_bits8<DROP>(seed, (uint64_t)(bh * L + q0 + lr),
                        kv0 + ks2 * 32 + lg * 8, pthresh, keep);
       bf16x8 dsa;
+      float dsf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const float pv = bf16_bits_to_f32((uint16_t)(unsigned short)pa[j]);
         float dpv = bf16_bits_to_f32((uint16_t)(unsigned short)dpa[j]);
         if (DROP) dpv = keep[j] ? dpv * pinv : 0.f;
-        dsa[j] = (short)f32_to_bf16_bits(pv * (dpv - di_row));
+        dsf[j] = pv * (dpv - di_row);
+        dsa[j] = (short)f32_to_bf16_bits(dsf[j]);
+      }
+      if (HAS_BIAS && dbias_acc != nullptr) {
+        // fused dbias accumulation (see kres variant)
+        const int64_t brow =
+            ((bh / bias_od) % bias_nb) * (int64_t)bias_q + ((q0 + lr) % bias_q);
+        float* bdst = dbias_acc + brow * (int64_t)L + kv0 + ks2 * 32 + lg * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) atomicAdd(bdst + j, dsf[j]);
       }
       if (ds_out != nullptr) {
-        // materialize dS for the bias gradient (contiguous 16 B store;
+        // materialize dS for the bias gradient (deterministic fallback;
         // the broadcast-batch sum happens as one torch reduction)
         union {
           bf16x8 v;
@@ -1680,11 +1716,21 @@ std::vector<at::Tensor> flash_attn_backward(
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  at::Tensor ds;
+  // bias gradient: default is the FUSED path — dS accumulates straight into
+  // an (nb, L, L) fp32 buffer via atomics inside the dq kernel, skipping the
+  // (BH, L, L) dS materialization + torch reduction entirely. Atomic order
+  // is non-deterministic, so torch.use_deterministic_algorithms() selects
+  // the materialized-dS fallback.
+  at::Tensor ds, dbias;
   if (bd.ptr && bias_needs_grad) {
     TORCH_CHECK(bd.q == L, "flash_attn: bias grad requires bias_q == L");
-    ds = at::empty({BH, (int64_t)L, (int64_t)L}, q.options());
+    if (at::globalContext().deterministicAlgorithms())
+      ds = at::empty({BH, (int64_t)L, (int64_t)L}, q.options());
+    else
+      dbias = at::zeros({bd.nb, (int64_t)bd.q, (int64_t)L},
+                        q.options().dtype(at::kFloat));
   }
+  float* dbias_ptr = dbias.defined() ? dbias.data_ptr<float>() : nullptr;
   const dim3 grid(L / BM, BH);
 
   auto launch_all = [&](auto hb, auto hm, auto dr) {
@@ -1697,6 +1743,7 @@ std::vector<at::Tensor> flash_attn_backward(
               reinterpret_cast<uint16_t*>(dq.data_ptr()),
               ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr())
                            : nullptr,
+              dbias_ptr,
               reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
               reinterpret_cast<const uint16_t*>(q.data_ptr()),
               reinterpret_cast<const uint16_t*>(k.data_ptr()),
@@ -1709,6 +1756,7 @@ std::vector<at::Tensor> flash_attn_backward(
       flash_bwd_dq_kres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
           reinterpret_cast<uint16_t*>(dq.data_ptr()),
           ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
+          dbias_ptr,
           reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
           reinterpret_cast<const uint16_t*>(q.data_ptr()),
           reinterpret_cast<const uint16_t*>(k.data_ptr()),
@@ -1720,6 +1768,7 @@ std::vector<at::Tensor> flash_attn_backward(
       flash_bwd_dq_kernel<HB, HM, DR><<<grid, 256, 0, stream>>>(
         reinterpret_cast<uint16_t*>(dq.data_ptr()),
         ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
+        dbias_ptr,
         reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
         reinterpret_cast<const uint16_t*>(q.data_ptr()),
         reinterpret_cast<const uint16_t*>(k.data_ptr()),
@@ -1766,6 +1815,14 @@ std::vector<at::Tensor> flash_attn_backward(
     pick(std::false_type{}, std::false_type{});
 
   C10_CUDA_KERNEL_LAUNCH_CHECK();
-  if (ds.defined()) return {dq, dk, dv, ds};
+  if (dbias.defined()) return {dq, dk, dv, dbias};
+  if (ds.defined()) {
+    // deterministic fallback: reduce the materialized dS over the
+    // broadcast axes here (BH rows decompose as (outer, nb, od))
+    const int64_t outer = BH / (bd.nb * bd.od);
+    auto db = ds.view({outer, bd.nb, bd.od, (int64_t)L, (int64_t)L})
+                  .sum(at::IntArrayRef{0, 2}, /*keepdim=*/false, at::kFloat);
+    return {dq, dk, dv, db};
+  }
   return {dq, dk, dv};
 }

@@ -93,9 +93,10 @@ def _flat_ddp_worker(rank, port, out_dir):
         # pass 2: grad accumulation — no_sync for the first micro-batch,
         # synced on the second; result = mean over ranks of (sum of two
         # local micro-grads)
+        # zero_grad_buffers zeroes the flats and re-pins p.grad onto the
+        # bucket views — the accumulation contract (manually Nulling p.grad
+        # here would detach the first micro-batch's grads from the buckets)
         ddp.zero_grad_buffers()
-        for p in model.parameters():
-            p.grad = None
         with ddp.no_sync():
             ddp(batch_for(rank)).pow(2).mean().backward()
         ddp(batch_for(rank)).pow(2).mean().backward()

@@ -60,13 +60,10 @@ class _FlashAttn(torch.autograd.Function):
         dq, dk, dv = grads[0], grads[1], grads[2]
         dbias = None
         if len(grads) > 3:
-            # grads[3] is materialized dS (BH, L, L); batch rows decompose
-            # as (outer, src_nb, outer_div) per the broadcast descriptor, so
-            # the bias gradient sums over both the outer and inner-repeat axes
-            ds = grads[3]
-            nb = bias_t.shape[0]
-            od = ctx.bias_od
-            dbias = ds.view(-1, nb, od, ds.shape[1], ds.shape[2]).sum((0, 2))
+            # grads[3] is the finished (nb, L, L) fp32 bias gradient: the
+            # kernel fuses the broadcast-batch reduction into dq's dS pass
+            # (deterministic mode reduces a materialized dS host-side)
+            dbias = grads[3].to(bias_t.dtype)
         return dq, dk, dv, dbias, None, None, None, None, None
 
 

@@ -102,6 +102,9 @@ _COMMON_FLAGS = [
     ("--profile", dict(action="store_true",
                        help="emit autograd profiler ranges (rocTX under "
                             "rocprof)")),
+    ("--gemm-tuning-file", dict(default=None, metavar="CSV",
+                                help="offline TunableOp results to load "
+                                     "(see tools/tunableop_pershape.sh)")),
     ("--ema-decay", dict(default=-1.0, type=float,
                          help="EMA decay for model params (<0 disables)")),
     ("--validate-with-ema", dict(action="store_true")),

@@ -246,6 +246,23 @@ def set_jit_fusion_options():
     kernels (+ hipGraphs), not the TorchScript fuser."""
 
 
+def load_gemm_tunings(path) -> bool:
+    """Arm TunableOp with an offline-tuned hipBLASLt/rocBLAS algorithm
+    table (tuning itself stays OFF — results were produced by
+    tools/tunableop_pershape.sh). Returns True when the file was loaded.
+
+    Safe to call on any stack: entries whose validators (ROCm/hipBLASLt
+    versions) do not match are ignored by TunableOp."""
+    if not path or not os.path.exists(path) or not torch.cuda.is_available():
+        return False
+    import torch.cuda.tunable as tunable
+
+    tunable.enable(True)
+    tunable.tuning_enable(False)
+    tunable.read_file(path)
+    return True
+
+
 def has_parameters(module):
     return next(module.parameters(), None) is not None
 

@@ -55,6 +55,9 @@ def _count_params(model, trainable_only=False):
 def main(args) -> None:
     utils.import_user_module(args)
     utils.set_jit_fusion_options()
+    if getattr(args, "gemm_tuning_file", None):
+        if utils.load_gemm_tunings(args.gemm_tuning_file):
+            logger.info(f"loaded GEMM tunings from {args.gemm_tuning_file}")
 
     assert args.batch_size is not None, \
         "Must specify batch size either with --batch-size"
