@@ -102,11 +102,26 @@ def test_flash_bwd_parity_no_dropout(L):
 
     o, lse, seed = ops.flash_attn_fwd(q, k, v, bias, 1, mask, H, 0.0, True)
     d_out = torch.randn_like(o) * 0.3
-    dq, dk, dv, ds = ops.flash_attn_bwd(
+    # fused path: grads[3] is the finished (H, L, L) fp32 bias gradient
+    dq, dk, dv, dbias = ops.flash_attn_bwd(
         d_out, q, k, v, o, lse, bias, 1, True, mask, H, 0.0, False, int(seed)
     )
-    # bias grad = dS summed over the broadcast batch
-    dbias = ds.view(B, H, L, L).sum(0)
+    assert dbias.shape == (H, L, L) and dbias.dtype == torch.float32
+
+    # the deterministic fallback (materialized dS + host reduction) must
+    # produce the same gradient
+    torch.use_deterministic_algorithms(True, warn_only=True)
+    try:
+        _, _, _, dbias_det = ops.flash_attn_bwd(
+            d_out, q, k, v, o, lse, bias, 1, True, mask, H, 0.0, False,
+            int(seed)
+        )
+    finally:
+        torch.use_deterministic_algorithms(False)
+    # fused accumulates pre-rounding fp32; fallback sums bf16-rounded dS
+    assert (dbias - dbias_det).abs().max().item() < 0.05 * (
+        dbias.abs().max().item() + 1e-6
+    )
 
     # fp32 reference
     qr = q.float().view(B, H, L, 64).requires_grad_(True)
