@@ -1716,19 +1716,27 @@ std::vector<at::Tensor> flash_attn_backward(
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  // bias gradient: default is the FUSED path — dS accumulates straight into
-  // an (nb, L, L) fp32 buffer via atomics inside the dq kernel, skipping the
-  // (BH, L, L) dS materialization + torch reduction entirely. Atomic order
-  // is non-deterministic, so torch.use_deterministic_algorithms() selects
-  // the materialized-dS fallback.
+  // bias gradient, two strategies:
+  //  * short L: materialize dS (BH, L, L) and reduce it over the broadcast
+  //    batches — deterministic, and faster than atomics when the broadcast
+  //    factor is large (measured: fp32 atomicAdd contention at L=512/B=96
+  //    serializes the dq kernel ~4x);
+  //  * long L (>= 2048): accumulate dS straight into the (nb, L, L) fp32
+  //    buffer with atomics inside the dq kernel — the dS materialization
+  //    stops fitting (BH*L*L*2 bytes) while the broadcast factor (and so
+  //    the atomic contention) is small. This is what makes a TRAINABLE
+  //    pair bias affordable at Uni-Fold-scale sequence lengths.
+  // torch.use_deterministic_algorithms() forces the materialized path.
   at::Tensor ds, dbias;
   if (bd.ptr && bias_needs_grad) {
     TORCH_CHECK(bd.q == L, "flash_attn: bias grad requires bias_q == L");
-    if (at::globalContext().deterministicAlgorithms())
-      ds = at::empty({BH, (int64_t)L, (int64_t)L}, q.options());
-    else
+    const bool fuse = L >= 2048 &&
+                      !at::globalContext().deterministicAlgorithms();
+    if (fuse)
       dbias = at::zeros({bd.nb, (int64_t)bd.q, (int64_t)L},
                         q.options().dtype(at::kFloat));
+    else
+      ds = at::empty({BH, (int64_t)L, (int64_t)L}, q.options());
   }
   float* dbias_ptr = dbias.defined() ? dbias.data_ptr<float>() : nullptr;
   const dim3 grid(L / BM, BH);
