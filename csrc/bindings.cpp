@@ -54,6 +54,10 @@ std::vector<at::Tensor> gelu_dropout_backward(at::Tensor grad, at::Tensor x,
                                               std::optional<at::Tensor> bias,
                                               at::Tensor dmask, double p);
 at::Tensor mfma_gemm_16x16x32(at::Tensor A, at::Tensor B);
+std::vector<at::Tensor> dropout_add_ln_forward(
+    at::Tensor x, at::Tensor res, std::optional<at::Tensor> bias,
+    at::Tensor gamma, at::Tensor beta, double p, bool is_training, double eps);
+
 std::vector<at::Tensor> dropout_add_forward(at::Tensor x, at::Tensor res,
                                             std::optional<at::Tensor> bias,
                                             double p, bool is_training);
@@ -141,6 +145,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "flash attention forward (bf16, D=64) -> (o, lse, seed)");
   m.def("flash_attn_backward", &flash_attn_backward,
         "flash attention backward -> (dq, dk, dv[, dS])");
+  m.def("dropout_add_ln_forward", &dropout_add_ln_forward,
+        "fused dropout+residual+LayerNorm forward");
   m.def("dropout_add_forward", &dropout_add_forward,
         "fused dropout + residual add forward");
   m.def("dropout_add_backward", &dropout_add_backward,

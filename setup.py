@@ -31,6 +31,7 @@ HIP_SOURCES = [
     "csrc/mfma_probe.hip",
     "csrc/flash_attn.hip",
     "csrc/dropout_add.hip",
+    "csrc/dropout_add_ln.hip",
     "csrc/embedding.hip",
     "csrc/cross_entropy.hip",
     "csrc/gaussian.hip",

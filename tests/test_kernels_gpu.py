@@ -1000,3 +1000,76 @@ def test_msa_arrange_parity(col, dtype):
     x.grad = None
     mref.backward(gm)
     assert torch.equal(gf, x.grad)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_dropout_add_ln_fused_parity(dtype):
+    """Fused dropout+residual+LN vs the eager chain: exact composition at
+    p=0 (forward AND all input grads), mask-consistent at p>0."""
+    from unicore_amd.modules.dropout_add_ln import _DropoutAddLN
+    from unicore_amd.modules.layer_norm import LayerNorm
+
+    torch.manual_seed(5)
+    N, C = 96, 768
+    ln = LayerNorm(C).cuda().to(dtype)
+    with torch.no_grad():
+        ln.weight.uniform_(0.5, 1.5)
+        ln.bias.uniform_(-0.5, 0.5)
+    x = (torch.randn(N, C, device="cuda", dtype=dtype) * 0.5).requires_grad_(True)
+    res = torch.randn(N, C, device="cuda", dtype=dtype).requires_grad_(True)
+    bias = (torch.randn(C, device="cuda", dtype=dtype) * 0.1).requires_grad_(True)
+
+    out = _DropoutAddLN.apply(x, res, bias, ln.weight, ln.bias, 0.0, True,
+                              ln.eps)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    x2 = x.detach().float().requires_grad_(True)
+    res2 = res.detach().float().requires_grad_(True)
+    bias2 = bias.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(
+        res2 + x2 + bias2, (C,), ln.weight.float(), ln.bias.float(), ln.eps
+    )
+    ref.backward(g.float())
+
+    tol = 2e-2 if dtype == torch.bfloat16 else 2e-5
+    for got, want, name in [
+        (out, ref, "out"), (x.grad, x2.grad, "dx"),
+        (res.grad, res2.grad, "dres"), (bias.grad, bias2.grad, "dbias"),
+    ]:
+        d = (got.float() - want).abs().max().item()
+        s = want.abs().max().item() + 1e-6
+        assert d / s < tol, f"{name}: {d} vs {s}"
+
+
+@requires_gpu
+def test_dropout_add_ln_dropout_mask_consistent():
+    """At p>0 the fused output must equal the eager composition built from
+    the kernel's own keep-mask."""
+    from unicore_amd import ops
+
+    torch.manual_seed(6)
+    N, C, p = 64, 512, 0.3
+    x = torch.randn(N, C, device="cuda", dtype=torch.bfloat16)
+    res = torch.randn_like(x)
+    gamma = torch.rand(C, device="cuda", dtype=torch.bfloat16) + 0.5
+    beta = torch.randn(C, device="cuda", dtype=torch.bfloat16) * 0.3
+
+    normed, summed, dmask, mean, invvar = ops.dropout_add_ln_fwd(
+        x, res, None, gamma, beta, p, True, 1e-5
+    )
+    bits = dmask.view(-1, 1) >> torch.arange(8, device="cuda",
+                                             dtype=torch.uint8)
+    keep = (bits & 1).bool().view(N, C)
+    want_sum = res.float() + torch.where(
+        keep, x.float() / (1 - p), torch.zeros_like(x.float())
+    )
+    assert (summed.float() - want_sum).abs().max().item() < 2e-2
+    want_norm = torch.nn.functional.layer_norm(
+        want_sum, (C,), gamma.float(), beta.float(), 1e-5
+    )
+    assert (normed.float() - want_norm).abs().max().item() < 5e-2
+    # keep rate sane
+    rate = keep.float().mean().item()
+    assert abs(rate - (1 - p)) < 0.05

@@ -17,6 +17,7 @@ from torch import Tensor, nn
 from unicore_amd import utils
 
 from .dropout_add import dropout_add
+from .dropout_add_ln import dropout_add_ln
 from .gelu_dropout import gelu_dropout
 from .layer_norm import LayerNorm
 from .multihead_attention import SelfMultiheadAttention
@@ -84,12 +85,18 @@ class TransformerEncoderLayer(nn.Module):
         )
         if return_attn:
             h, attn_weights, attn_probs = h
-        h = dropout_add(
-            h, skip, self.dropout, self.training,
-            bias=self.self_attn.out_proj.bias if fold else None,
-        )
+        attn_bias_fold = self.self_attn.out_proj.bias if fold else None
         if self.post_ln:
-            h = self.self_attn_layer_norm(h)
+            # single kernel: dropout + residual + LN (the sum never makes
+            # an extra HBM round-trip)
+            h = dropout_add_ln(
+                h, skip, self.self_attn_layer_norm, self.dropout,
+                self.training, bias=attn_bias_fold,
+            )
+        else:
+            h = dropout_add(
+                h, skip, self.dropout, self.training, bias=attn_bias_fold,
+            )
 
         skip = h
         if not self.post_ln:
@@ -99,8 +106,7 @@ class TransformerEncoderLayer(nn.Module):
             h = gelu_dropout(h, self.activation_dropout, self.training,
                              bias=self.fc1.bias)
             h = F.linear(h, self.fc2.weight)
-            h = dropout_add(h, skip, self.dropout, self.training,
-                            bias=self.fc2.bias)
+            fc2_bias_fold = self.fc2.bias
         else:
             h = self.fc1(h)
             if self._fuse_gelu and h.is_cuda:
@@ -109,9 +115,15 @@ class TransformerEncoderLayer(nn.Module):
                 h = F.dropout(self.act(h), p=self.activation_dropout,
                               training=self.training)
             h = self.fc2(h)
-            h = dropout_add(h, skip, self.dropout, self.training)
+            fc2_bias_fold = None
         if self.post_ln:
-            h = self.final_layer_norm(h)
+            h = dropout_add_ln(
+                h, skip, self.final_layer_norm, self.dropout, self.training,
+                bias=fc2_bias_fold,
+            )
+        else:
+            h = dropout_add(h, skip, self.dropout, self.training,
+                            bias=fc2_bias_fold)
 
         if return_attn:
             return h, attn_weights, attn_probs

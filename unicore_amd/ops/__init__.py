@@ -178,6 +178,13 @@ def flash_attn_bwd(d_out, q, k, v, o, lse, bias, bias_outer_div, bias_needs_grad
     )
 
 
+def dropout_add_ln_fwd(x, res, bias, gamma, beta, p, is_training, eps):
+    require_kernels()
+    return _kernels.dropout_add_ln_forward(
+        x, res, bias, gamma, beta, float(p), bool(is_training), float(eps)
+    )
+
+
 def dropout_add_fwd(x, res, p, is_training, bias=None):
     require_kernels()
     return _kernels.dropout_add_forward(x, res, bias, float(p), bool(is_training))
