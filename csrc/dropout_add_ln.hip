@@ -100,6 +100,27 @@ __global__ void dropout_add_ln_fwd_kernel(
   }
 }
 
+#define DISPATCH_FTYPES(st, NAME, ...)                               \
+  switch (st) {                                                      \
+    case at::ScalarType::Float: {                                    \
+      using scalar_t = float;                                        \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::Half: {                                     \
+      using scalar_t = __half;                                       \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    case at::ScalarType::BFloat16: {                                 \
+      using scalar_t = __hip_bfloat16;                               \
+      __VA_ARGS__;                                                   \
+      break;                                                         \
+    }                                                                \
+    default:                                                         \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", st);          \
+  }
+
 }  // namespace
 
 // returns {normed, summed, dmask, mean, invvar}
