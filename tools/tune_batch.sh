@@ -15,7 +15,7 @@ timeout 300 python bench.py --steps 3 --warmup 2 --batch-size ${BS} \
 
 # stage 2: per-shape isolated tuning of the stable hot shapes
 ROWS=$((BS * 512))
-grep -E "${ROWS}|B_$((BS * 12))" gpurun_out/untuned_bs${BS}.csv* \
+grep -hE "${ROWS}|B_$((BS * 12))" gpurun_out/untuned_bs${BS}*.csv \
   | sed 's/^[^:]*://' | sort -u > gpurun_out/stable_bs${BS}.csv
 wc -l gpurun_out/stable_bs${BS}.csv
 i=0

@@ -11,6 +11,8 @@ through the keep-mask and emits the folded-bias column sum; the residual
 gradient IS d_sum. Eager fallback composes F.dropout + add + F.layer_norm.
 """
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -58,6 +60,7 @@ def dropout_add_ln(x, residual, ln, p, is_training, bias=None):
         and x.shape == residual.shape
         and x.shape[-1] % 8 == 0
         and x.shape[-1] <= 2048
+        and os.environ.get("UNICORE_FUSED_LN_JOIN", "1") == "1"
     ):
         from unicore_amd import ops
 
