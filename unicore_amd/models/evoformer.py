@@ -267,6 +267,14 @@ class EvoformerModel(BaseUnicoreModel):
         parser.add_argument("--activation-checkpoint", action="store_true",
                             help="recompute each Evoformer block in "
                                  "backward (utils.checkpoint_sequential)")
+        parser.add_argument("--hip-graph-blocks", action="store_true",
+                            help="capture each Evoformer block as a hipGraph "
+                                 "(fwd+bwd) and replay it — removes the "
+                                 "per-step launch overhead of this "
+                                 "launch-bound stack. Requires --dropout 0 "
+                                 "(dropout kernels take their philox seed "
+                                 "as a kernel argument, which a captured "
+                                 "graph would freeze) and static shapes.")
 
     def __init__(self, args, dictionary):
         super().__init__()
@@ -289,10 +297,35 @@ class EvoformerModel(BaseUnicoreModel):
         self.lm_head = nn.Linear(args.msa_dim, len(dictionary))
         self.recycle_iters = args.recycle_iters
         self.activation_checkpoint = args.activation_checkpoint
+        self.hip_graph_blocks = getattr(args, "hip_graph_blocks", False)
+        self._graphed_blocks = None
 
     @classmethod
     def build_model(cls, args, task):
         return cls(args, task.dictionary)
+
+    def _active_blocks(self, msa, pair):
+        """The block list, hipGraph-captured on first use when requested."""
+        if not (self.hip_graph_blocks and msa.is_cuda):
+            return self.blocks
+        if self._graphed_blocks is None:
+            assert self.args.dropout == 0, (
+                "--hip-graph-blocks requires --dropout 0: the dropout "
+                "kernels key philox off a kernel argument, which graph "
+                "replay would freeze"
+            )
+            assert not self.activation_checkpoint, (
+                "--hip-graph-blocks and --activation-checkpoint are "
+                "mutually exclusive"
+            )
+            sample = (
+                msa.detach().clone().requires_grad_(True),
+                pair.detach().clone().requires_grad_(True),
+            )
+            self._graphed_blocks = torch.cuda.make_graphed_callables(
+                tuple(self.blocks), tuple(sample for _ in self.blocks)
+            )
+        return self._graphed_blocks
 
     def _trunk(self, msa, pair):
         if self.activation_checkpoint and self.training:
@@ -303,7 +336,7 @@ class EvoformerModel(BaseUnicoreModel):
             ]
             msa, pair = utils.checkpoint_sequential(fns, (msa, pair))
         else:
-            for blk in self.blocks:
+            for blk in self._active_blocks(msa, pair):
                 msa, pair = blk(msa, pair)
         return msa, pair
 
@@ -335,3 +368,4 @@ def evoformer_base_architecture(args):
     args.max_rel_pos = getattr(args, "max_rel_pos", 32)
     args.recycle_iters = getattr(args, "recycle_iters", None) or 0
     args.activation_checkpoint = getattr(args, "activation_checkpoint", False)
+    args.hip_graph_blocks = getattr(args, "hip_graph_blocks", False)
