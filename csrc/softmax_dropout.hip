@@ -18,6 +18,8 @@
 // source row = ((b / outer_div) % src_nb) * src_q + (q % src_q).
 #include "common.h"
 
+#include <cstdlib>
+
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <ATen/cuda/CUDAGeneratorImpl.h>
@@ -491,8 +493,12 @@ std::vector<at::Tensor> softmax_dropout_forward(
     // bias-major iteration order applies when a bias broadcasts over an
     // outer batch (od == 1, e.g. the (1, H, L, L) rel-pos bias): the
     // per-(h, q) bias row then stays cache-hot across all `outer` batches
+    static const bool bias_major_enabled = []() {
+      const char* e = std::getenv("UNICORE_SM_BIASMAJOR");
+      return e == nullptr || e[0] != '0';
+    }();
     int64_t bias_major_outer = 0;
-    if (bsrc.ptr && bsrc.od == 1 && bsrc.q == q_len &&
+    if (bias_major_enabled && bsrc.ptr && bsrc.od == 1 && bsrc.q == q_len &&
         n_rows % (bsrc.nb * (int64_t)q_len) == 0) {
       const int64_t outer = n_rows / (bsrc.nb * (int64_t)q_len);
       // only worth permuting when the broadcast actually repeats, and only
