@@ -27,11 +27,11 @@ evorun () {
     --task evoformer_synthetic --arch evoformer --loss masked_msa \
     --optimizer adam --adam-betas '(0.9, 0.99)' --adam-eps 1e-6 --clip-norm 0.1 \
     --lr-scheduler polynomial_decay --lr 1e-3 --warmup-updates 1000 \
-    --total-num-update 20000 --max-update 24 --dataset-size 256 \
+    --total-num-update 20000 --max-update 64 --dataset-size 256 \
     --batch-size 1 --update-freq 8 --msa-depth 128 --residues 256 \
     --dropout 0.0 $1 \
     --bf16 --bf16-sr --ddp-backend c10d \
-    --log-interval 8 --log-format simple --no-save \
+    --log-interval 32 --log-format simple --no-save \
     --save-dir /tmp/ck_evo 2>&1 | grep -E "train_inner" | tail -1
 }
 echo "=== EVO plain (dropout 0) ==="; evorun ""
