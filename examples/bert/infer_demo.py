@@ -31,6 +31,10 @@ def main():
     p.add_argument("--iters", type=int, default=10)
     p.add_argument("--bf16", action="store_true")
     p.add_argument("--cpu", action="store_true")
+    p.add_argument("--hip-graph", action="store_true",
+                   help="capture the whole forward as a hipGraph and replay "
+                        "it per batch (static shapes; removes all launch "
+                        "overhead from the serving loop)")
     a = p.parse_args()
 
     state = checkpoint_utils.load_checkpoint_to_cpu(a.checkpoint)
@@ -53,14 +57,38 @@ def main():
     if use_cuda:
         toks = toks.cuda()
 
+    def fwd(tokens):
+        out = model(tokens)
+        return out[0] if isinstance(out, tuple) else out
+
     with torch.no_grad():
-        logits = model(toks)[0] if isinstance(model(toks), tuple) else model(toks)
+        if a.hip_graph and use_cuda:
+            # warm up twice on a side stream, then capture one replayable
+            # graph over static in/out buffers
+            static_in = toks.clone()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    fwd(static_in)
+            torch.cuda.current_stream().wait_stream(s)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_out = fwd(static_in)
+
+            def run_batch(tokens):
+                static_in.copy_(tokens)
+                graph.replay()
+                return static_out
+        else:
+            run_batch = fwd
+
+        logits = run_batch(toks)
         if use_cuda:
             torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(a.iters):
-            out = model(toks)
-            logits = out[0] if isinstance(out, tuple) else out
+            logits = run_batch(toks)
         if use_cuda:
             torch.cuda.synchronize()
         dt = time.perf_counter() - t0
