@@ -304,10 +304,22 @@ class EvoformerModel(BaseUnicoreModel):
     def build_model(cls, args, task):
         return cls(args, task.dictionary)
 
-    def _active_blocks(self, msa, pair):
-        """The block list, hipGraph-captured on first use when requested."""
-        if not (self.hip_graph_blocks and msa.is_cuda):
-            return self.blocks
+    class _TrunkRunner(nn.Module):
+        """The whole block loop as one callable, so hipGraph capture wraps
+        ONE graph around every block (a graph per block pays a static-buffer
+        copy of the 16 MB msa/pair tensors per block — measured slower)."""
+
+        def __init__(self, blocks):
+            super().__init__()
+            self.blocks = blocks
+
+        def forward(self, msa, pair):
+            for blk in self.blocks:
+                msa, pair = blk(msa, pair)
+            return msa, pair
+
+    def _graphed_trunk(self, msa, pair):
+        """One fwd+bwd hipGraph over the whole trunk, captured on first use."""
         if self._graphed_blocks is None:
             assert self.args.dropout == 0, (
                 "--hip-graph-blocks requires --dropout 0: the dropout "
@@ -323,11 +335,13 @@ class EvoformerModel(BaseUnicoreModel):
                 pair.detach().clone().requires_grad_(True),
             )
             self._graphed_blocks = torch.cuda.make_graphed_callables(
-                tuple(self.blocks), tuple(sample for _ in self.blocks)
+                self._TrunkRunner(self.blocks), sample
             )
-        return self._graphed_blocks
+        return self._graphed_blocks(msa, pair)
 
     def _trunk(self, msa, pair):
+        if self.hip_graph_blocks and msa.is_cuda:
+            return self._graphed_trunk(msa, pair)
         if self.activation_checkpoint and self.training:
             # recompute each block in backward; the msa/pair pair threads
             # through as the tuple state (utils.checkpoint_sequential)
@@ -336,7 +350,7 @@ class EvoformerModel(BaseUnicoreModel):
             ]
             msa, pair = utils.checkpoint_sequential(fns, (msa, pair))
         else:
-            for blk in self._active_blocks(msa, pair):
+            for blk in self.blocks:
                 msa, pair = blk(msa, pair)
         return msa, pair
 
