@@ -2,6 +2,9 @@
 """Round-2 micro A/Bs: softmax bias-major on/off, embedding backward vs
 torch, on the BERT bench shapes. Run on a GPU box."""
 import os
+import sys
+
+sys.path.insert(0, __file__.rsplit("/tools/", 1)[0] if "/tools/" in __file__ else ".")
 import time
 
 import torch
