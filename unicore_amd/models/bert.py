@@ -158,8 +158,15 @@ class BertLMHead(nn.Module):
     def forward(self, features, masked_tokens=None, **kwargs):
         if masked_tokens is not None:
             # only the masked rows reach the vocab GEMM — big memory and
-            # compute savings at 15% mask rate
-            features = features[masked_tokens, :]
+            # compute savings at 15% mask rate. A bool mask selects rows
+            # directly; a flat long index (the pad-to-bucket path: row
+            # count rounded up so the GEMM shape is batch-invariant and
+            # tunable) gathers from the flattened tokens.
+            if masked_tokens.dtype == torch.bool:
+                features = features[masked_tokens, :]
+            else:
+                flat = features.reshape(-1, features.size(-1))
+                features = flat.index_select(0, masked_tokens)
         h = self.layer_norm(self.act(self.dense(features)))
         return F.linear(h, self.weight) + self.bias
 
