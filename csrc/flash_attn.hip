@@ -814,6 +814,26 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
 
   f32x4 dq_acc[2][4] = {};
   const int n_tiles = L / BN;
+  // T14 prefetch: the next tile's K/V B-fragments are issued right after
+  // the current tile's MFMAs consume the registers, so their ~500-cycle
+  // HBM latency hides under the exp/LDS/dS/dq phases. Essential here
+  // because 264+ VGPRs leave ONE wave per SIMD — no other wave exists to
+  // hide the latency (PMC: ~60%% of cycles were memory waits).
+  bf16x8 bk_r[4][2], bvt_r[4][2];
+  auto issue_kv = [&](int tile) {
+    const int kvb = tile * BN;
+#pragma unroll
+    for (int cb = 0; cb < 4; ++cb)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bk_r[cb][ks] = load_frag(
+            kp + (bh * L + kvb + cb * 16 + lr) * (int64_t)HD + ks * 32 +
+            lg * 8);
+        bvt_r[cb][ks] = load_frag(
+            vp + (bh * L + kvb + cb * 16 + lr) * (int64_t)HD + ks * 32 +
+            lg * 8);
+      }
+  };
   // ping-pong K^T staging: stage tile t+1 into the other buffer while
   // computing tile t, one barrier per iteration
   auto stage_kt = [&](int tile) {
@@ -835,6 +855,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     }
   };
   stage_kt(0);
+  issue_kv(0);
   __syncthreads();
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
@@ -846,10 +867,8 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       f32x4 acc0 = {}, acc1 = {}, accd0 = {}, accd1 = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bk = load_frag(
-            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
-        const bf16x8 bvt = load_frag(
-            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        const bf16x8 bk = bk_r[cb][ks];
+        const bf16x8 bvt = bvt_r[cb][ks];
         acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][ks], bk, acc0, 0, 0, 0);
         acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][ks], bk, acc1, 0, 0, 0);
         accd0 =
@@ -863,6 +882,9 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       dp[1][cb] = accd1;
     }
     __builtin_amdgcn_s_setprio(0);
+    // re-issue the SAME registers for tile t+1 — the loads fly under the
+    // exp/LDS/dS/dq phases below (T14 write-late / issue-early)
+    if (t + 1 < n_tiles) issue_kv(t + 1);
 #pragma unroll
     for (int mtile = 0; mtile < 2; ++mtile)
 #pragma unroll
