@@ -1376,7 +1376,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_q32_kernel(
 }
 
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
-__global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
+__global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
@@ -1422,6 +1422,24 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
 
   f32x4 dk_acc[4] = {}, dv_acc[4] = {};
   const int n_tiles = L / BM;
+  // T14 prefetch of the next q-tile's Q/dO B-fragments (same idea as the
+  // dq kernel): +64 VGPRs drops this kernel from 3 to 2 blocks/CU, but
+  // the loads then hide under the exp/LDS/dS/dkv phases
+  bf16x8 bq_r[4][2], bdo_r[4][2];
+  auto issue_q = [&](int tile) {
+    const int qb = tile * BM;
+#pragma unroll
+    for (int cq = 0; cq < 4; ++cq)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bq_r[cq][ks] = load_frag(
+            qp + (bh * L + qb + cq * 16 + lr) * (int64_t)HD + ks * 32 +
+            lg * 8);
+        bdo_r[cq][ks] = load_frag(
+            dop + (bh * L + qb + cq * 16 + lr) * (int64_t)HD + ks * 32 +
+            lg * 8);
+      }
+  };
   // ping-pong staging of Q^T/dO^T: tile tq+1 stages into the other
   // buffer while tile tq computes, one barrier per iteration
   auto stage_q = [&](int tile) {
@@ -1445,6 +1463,7 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
     }
   };
   stage_q(0);
+  issue_q(0);
   __syncthreads();
   for (int tq = 0; tq < n_tiles; ++tq) {
     const int q0 = tq * BM;
@@ -1452,20 +1471,20 @@ __global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
-      const int qcol = q0 + cq * 16 + lr;
       f32x4 acc = {}, accd = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bq =
-            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq, acc, 0, 0, 0);
-        const bf16x8 bdo =
-            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
-        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo, accd, 0, 0, 0);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq_r[cq][ks],
+                                                      acc, 0, 0, 0);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo_r[cq][ks],
+                                                       accd, 0, 0, 0);
       }
       st[cq] = acc;
       dpt[cq] = accd;
     }
+    // re-issue the same registers for tile tq+1 (loads fly under the
+    // remaining phases)
+    if (tq + 1 < n_tiles) issue_q(tq + 1);
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
       const int qcol = q0 + cq * 16 + lr;
