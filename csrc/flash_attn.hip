@@ -882,9 +882,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       dp[1][cb] = accd1;
     }
     __builtin_amdgcn_s_setprio(0);
-    // re-issue the SAME registers for tile t+1 — the loads fly under the
-    // exp/LDS/dS/dq phases below (T14 write-late / issue-early)
-    if (t + 1 < n_tiles) issue_kv(t + 1);
 #pragma unroll
     for (int mtile = 0; mtile < 2; ++mtile)
 #pragma unroll
@@ -902,6 +899,12 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
           s[mtile][cb][r] = __expf(sv - lse_r[mtile][r]);
         }
       }
+    // re-issue the SAME registers for tile t+1. MUST come after the bias
+    // loads above: the vm queue is FIFO, so any short load issued behind
+    // these 16 prefetches would both drain them and stall itself. From
+    // here to the next MFMA phase only LDS traffic (lgkm counter) and
+    // stores remain, so the prefetch latency hides fully (T14).
+    if (t + 1 < n_tiles) issue_kv(t + 1);
     // redistribute P (both M-tiles), read A-fragments, then reuse for dS
 #pragma unroll
     for (int mtile = 0; mtile < 2; ++mtile)
@@ -1482,9 +1485,6 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
       st[cq] = acc;
       dpt[cq] = accd;
     }
-    // re-issue the same registers for tile tq+1 (loads fly under the
-    // remaining phases)
-    if (tq + 1 < n_tiles) issue_q(tq + 1);
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
       const int qcol = q0 + cq * 16 + lr;
@@ -1517,6 +1517,11 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
         dpt[cq][r] = pv * (dpv - di_c);
       }
     }
+    // re-issue the same registers for tile tq+1 AFTER the scalar
+    // lse/di/bias loads above (FIFO vm queue: a short load issued behind
+    // the prefetch would drain it and stall itself); from here to the
+    // next MFMA phase only LDS traffic remains (T14)
+    if (tq + 1 < n_tiles) issue_q(tq + 1);
     // redistribute P^T first (shared buffer, wave-local ordering), read
     // both A-fragments into registers, then reuse the buffer for dS^T
 #pragma unroll
