@@ -814,26 +814,12 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
 
   f32x4 dq_acc[2][4] = {};
   const int n_tiles = L / BN;
-  // T14 prefetch: the next tile's K/V B-fragments are issued right after
-  // the current tile's MFMAs consume the registers, so their ~500-cycle
-  // HBM latency hides under the exp/LDS/dS/dq phases. Essential here
-  // because 264+ VGPRs leave ONE wave per SIMD — no other wave exists to
-  // hide the latency (PMC: ~60%% of cycles were memory waits).
-  bf16x8 bk_r[4][2], bvt_r[4][2];
-  auto issue_kv = [&](int tile) {
-    const int kvb = tile * BN;
-#pragma unroll
-    for (int cb = 0; cb < 4; ++cb)
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        bk_r[cb][ks] = load_frag(
-            kp + (bh * L + kvb + cb * 16 + lr) * (int64_t)HD + ks * 32 +
-            lg * 8);
-        bvt_r[cb][ks] = load_frag(
-            vp + (bh * L + kvb + cb * 16 + lr) * (int64_t)HD + ks * 32 +
-            lg * 8);
-      }
-  };
+  // NOTE (measured): a T14 register prefetch of the next tile's K/V
+  // fragments was tried here (issue after the bias loads, single register
+  // set) — dq went 630 -> 832-1099 us. At 264+ VGPR / 1 wave/SIMD the
+  // extra 96 registers hurt scheduling more than the ~500-cycle load
+  // latency costs; the FIFO vm-queue also forces later short loads to
+  // drain any outstanding prefetch. Reverted; kept as a record.
   // ping-pong K^T staging: stage tile t+1 into the other buffer while
   // computing tile t, one barrier per iteration
   auto stage_kt = [&](int tile) {
@@ -855,7 +841,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     }
   };
   stage_kt(0);
-  issue_kv(0);
   __syncthreads();
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * BN;
@@ -867,8 +852,10 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
       f32x4 acc0 = {}, acc1 = {}, accd0 = {}, accd1 = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const bf16x8 bk = bk_r[cb][ks];
-        const bf16x8 bvt = bvt_r[cb][ks];
+        const bf16x8 bk = load_frag(
+            kp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
+        const bf16x8 bvt = load_frag(
+            vp + (bh * L + kv0 + cb * 16 + lr) * (int64_t)HD + ks * 32 + lg * 8);
         acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][ks], bk, acc0, 0, 0, 0);
         acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][ks], bk, acc1, 0, 0, 0);
         accd0 =
@@ -899,12 +886,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
           s[mtile][cb][r] = __expf(sv - lse_r[mtile][r]);
         }
       }
-    // re-issue the SAME registers for tile t+1. MUST come after the bias
-    // loads above: the vm queue is FIFO, so any short load issued behind
-    // these 16 prefetches would both drain them and stall itself. From
-    // here to the next MFMA phase only LDS traffic (lgkm counter) and
-    // stores remain, so the prefetch latency hides fully (T14).
-    if (t + 1 < n_tiles) issue_kv(t + 1);
     // redistribute P (both M-tiles), read A-fragments, then reuse for dS
 #pragma unroll
     for (int mtile = 0; mtile < 2; ++mtile)
@@ -1379,7 +1360,7 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_q32_kernel(
 }
 
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
-__global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
+__global__ __launch_bounds__(256, 3) void flash_bwd_dkv_qres_kernel(
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
@@ -1425,24 +1406,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
 
   f32x4 dk_acc[4] = {}, dv_acc[4] = {};
   const int n_tiles = L / BM;
-  // T14 prefetch of the next q-tile's Q/dO B-fragments (same idea as the
-  // dq kernel): +64 VGPRs drops this kernel from 3 to 2 blocks/CU, but
-  // the loads then hide under the exp/LDS/dS/dkv phases
-  bf16x8 bq_r[4][2], bdo_r[4][2];
-  auto issue_q = [&](int tile) {
-    const int qb = tile * BM;
-#pragma unroll
-    for (int cq = 0; cq < 4; ++cq)
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        bq_r[cq][ks] = load_frag(
-            qp + (bh * L + qb + cq * 16 + lr) * (int64_t)HD + ks * 32 +
-            lg * 8);
-        bdo_r[cq][ks] = load_frag(
-            dop + (bh * L + qb + cq * 16 + lr) * (int64_t)HD + ks * 32 +
-            lg * 8);
-      }
-  };
+  // (T14 prefetch tried and reverted here too: 845 -> 918 us at the
+  // 3->2 blocks/CU occupancy cost; see the dq kernel's note.)
   // ping-pong staging of Q^T/dO^T: tile tq+1 stages into the other
   // buffer while tile tq computes, one barrier per iteration
   auto stage_q = [&](int tile) {
@@ -1466,7 +1431,6 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
     }
   };
   stage_q(0);
-  issue_q(0);
   __syncthreads();
   for (int tq = 0; tq < n_tiles; ++tq) {
     const int q0 = tq * BM;
@@ -1474,13 +1438,16 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int cq = 0; cq < 4; ++cq) {
+      const int qcol = q0 + cq * 16 + lr;
       f32x4 acc = {}, accd = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq_r[cq][ks],
-                                                      acc, 0, 0, 0);
-        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo_r[cq][ks],
-                                                       accd, 0, 0, 0);
+        const bf16x8 bq =
+            load_frag(qp + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq, acc, 0, 0, 0);
+        const bf16x8 bdo =
+            load_frag(dop + (bh * L + qcol) * (int64_t)HD + ks * 32 + lg * 8);
+        accd = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[ks], bdo, accd, 0, 0, 0);
       }
       st[cq] = acc;
       dpt[cq] = accd;
@@ -1517,11 +1484,6 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_dkv_qres_kernel(
         dpt[cq][r] = pv * (dpv - di_c);
       }
     }
-    // re-issue the same registers for tile tq+1 AFTER the scalar
-    // lse/di/bias loads above (FIFO vm queue: a short load issued behind
-    // the prefetch would drain it and stall itself); from here to the
-    // next MFMA phase only LDS traffic remains (T14)
-    if (tq + 1 < n_tiles) issue_q(tq + 1);
     // redistribute P^T first (shared buffer, wave-local ordering), read
     // both A-fragments into registers, then reuse the buffer for dS^T
 #pragma unroll
