@@ -587,7 +587,6 @@ __global__ void flash_dot_do_o_kernel(float* __restrict__ di,
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
 __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
-    float* __restrict__ dbias_acc,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
     const float* __restrict__ lse, const float* __restrict__ di,
@@ -715,16 +714,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
         dsf[j] = pv * (dpv - di_row);
         dsa[j] = (short)f32_to_bf16_bits(dsf[j]);
       }
-      if (HAS_BIAS && dbias_acc != nullptr) {
-        // fused dbias: accumulate dS straight into the (nb, L, L) fp32
-        // bias-grad buffer (pre-rounding values) instead of materializing
-        // the (BH, L, L) dS tensor for a torch reduction
-        const int64_t brow =
-            ((bh / bias_od) % bias_nb) * (int64_t)bias_q + ((q0 + lr) % bias_q);
-        float* bdst = dbias_acc + brow * (int64_t)L + kv0 + ks2 * 32 + lg * 8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) atomicAdd(bdst + j, dsf[j]);
-      }
       if (ds_out != nullptr) {
         union {
           bf16x8 v;
@@ -761,7 +750,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kres_kernel(
 template <bool HAS_BIAS, bool HAS_MASK, bool DROP, int LMAX>
 __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
     uint16_t* __restrict__ dq, uint16_t* __restrict__ ds_out,
-    float* __restrict__ dbias_acc,
     const uint16_t* __restrict__ dop, const uint16_t* __restrict__ qp,
     const uint16_t* __restrict__ kp, const uint16_t* __restrict__ vp,
     const float* __restrict__ lse, const float* __restrict__ di,
@@ -928,15 +916,6 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_k32_kernel(
           if (DROP) dpv = keep[mtile][j] ? dpv * pinv : 0.f;
           dsf[j] = pv * (dpv - di_row[mtile]);
           dsa[mtile][j] = (short)f32_to_bf16_bits(dsf[j]);
-        }
-        if (HAS_BIAS && dbias_acc != nullptr) {
-          // fused dbias accumulation (see kres variant)
-          const int64_t brow = ((bh / bias_od) % bias_nb) * (int64_t)bias_q +
-                               ((q0 + mtile * 16 + lr) % bias_q);
-          float* bdst =
-              dbias_acc + brow * (int64_t)L + kv0 + ks2 * 32 + lg * 8;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) atomicAdd(bdst + j, dsf[j]);
         }
         if (ds_out != nullptr) {
           union {
@@ -1759,7 +1738,6 @@ std::vector<at::Tensor> flash_attn_backward(
               reinterpret_cast<uint16_t*>(dq.data_ptr()),
               ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr())
                            : nullptr,
-              dbias_ptr,
               reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
               reinterpret_cast<const uint16_t*>(q.data_ptr()),
               reinterpret_cast<const uint16_t*>(k.data_ptr()),
@@ -1772,7 +1750,6 @@ std::vector<at::Tensor> flash_attn_backward(
       flash_bwd_dq_kres_kernel<HB, HM, DR, 512><<<grid, 256, 0, stream>>>(
           reinterpret_cast<uint16_t*>(dq.data_ptr()),
           ds.defined() ? reinterpret_cast<uint16_t*>(ds.data_ptr()) : nullptr,
-          dbias_ptr,
           reinterpret_cast<const uint16_t*>(d_out.data_ptr()),
           reinterpret_cast<const uint16_t*>(q.data_ptr()),
           reinterpret_cast<const uint16_t*>(k.data_ptr()),
