@@ -101,7 +101,9 @@ class BertModel(BaseUnicoreModel):
         features_only = features_only or classification_head_name is not None
 
         pad_mask = src_tokens.eq(self.padding_idx)
-        if not pad_mask.any():
+        if pad_mask.is_cuda and torch.cuda.is_current_stream_capturing():
+            pass  # .any() is a host sync — illegal inside hipGraph capture
+        elif not pad_mask.any():
             pad_mask = None
 
         seq_len = src_tokens.size(1)
