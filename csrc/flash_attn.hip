@@ -541,8 +541,12 @@ std::vector<at::Tensor> flash_attn_forward(at::Tensor q, at::Tensor k, at::Tenso
   else
     pick(std::false_type{}, std::false_type{});
   C10_CUDA_KERNEL_LAUNCH_CHECK();
+  // seed is host-side state: keep it on CPU so `int(seed)` in the autograd
+  // wrapper is free (a CUDA scalar here cost a D2H sync per forward and
+  // broke hipGraph capture).
   return {o, lse_t,
-          at::scalar_tensor((int64_t)seed, q.options().dtype(at::kLong))};
+          at::scalar_tensor((int64_t)seed,
+                            at::TensorOptions().dtype(at::kLong))};
 }
 
 // ===========================================================================
