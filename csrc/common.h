@@ -69,6 +69,35 @@ __device__ __forceinline__ void load8(const T* p, float (&f)[8]) {
   }
 }
 
+// NV-element variant (NV in {2,4,8}); same alignment contract as load8
+// scaled down: the byte width of the load (NV * sizeof(T)) must divide the
+// pointer's offset from a 16-byte-aligned base.
+template <typename T, int NV>
+__device__ __forceinline__ void loadN(const T* p, float (&f)[NV]) {
+  static_assert(NV == 2 || NV == 4 || NV == 8, "loadN: NV must be 2/4/8");
+  if constexpr (NV == 8) {
+    load8(p, f);
+  } else if constexpr (sizeof(T) == 2) {
+    union {
+      uint2 u2;
+      unsigned u1;
+      T t[NV];
+    } U;
+    if constexpr (NV == 4)
+      U.u2 = *reinterpret_cast<const uint2*>(p);
+    else
+      U.u1 = *reinterpret_cast<const unsigned*>(p);
+#pragma unroll
+    for (int j = 0; j < NV; ++j) f[j] = Cvt<T>::to_f(U.t[j]);
+  } else if constexpr (NV == 4) {
+    const float4 a = *reinterpret_cast<const float4*>(p);
+    f[0] = a.x; f[1] = a.y; f[2] = a.z; f[3] = a.w;
+  } else {
+    const float2 a = *reinterpret_cast<const float2*>(p);
+    f[0] = a.x; f[1] = a.y;
+  }
+}
+
 template <typename T>
 __device__ __forceinline__ void store8(T* p, const float (&f)[8]) {
   if constexpr (sizeof(T) == 2) {

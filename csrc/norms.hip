@@ -304,6 +304,54 @@ __global__ void norm_bwd_gb_partial_kernel(const T* __restrict__ dy,
   }
 }
 
+// stage 1, vectorized (n2 % (64*NV) == 0): lane tx owns NV consecutive
+// columns at (bx*64 + tx)*NV, loaded as one 2*NV-byte access — the scalar
+// variant's per-element 2-byte loads held this kernel to ~0.85 TB/s fabric
+// (PMC, profiles/bert_r2_pmc.txt) while its elementwise peers ran 3.2-3.9.
+template <typename T, bool RMS, int NV>
+__global__ void norm_bwd_gb_partial_vec_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ invvar,
+    int64_t n1, int n2, int rows_per_cta, float* __restrict__ part_g,
+    float* __restrict__ part_b) {
+  __shared__ float lds[4][64][2 * NV];
+  const int tx = threadIdx.x;
+  const int ty = threadIdx.y;
+  const int c0 = (blockIdx.x * 64 + tx) * NV;
+  const int64_t row_start = (int64_t)blockIdx.y * rows_per_cta;
+  const int64_t row_end = min(row_start + (int64_t)rows_per_cta, n1);
+  float acc_g[NV] = {}, acc_b[NV] = {};
+  for (int64_t r = row_start + ty; r < row_end; r += 4) {
+    const float iv = invvar[r];
+    const float mu = RMS ? 0.f : mean[r];
+    float d[NV], xv[NV];
+    loadN<T, NV>(dy + r * (int64_t)n2 + c0, d);
+    loadN<T, NV>(x + r * (int64_t)n2 + c0, xv);
+#pragma unroll
+    for (int j = 0; j < NV; ++j) {
+      acc_g[j] += d[j] * (xv[j] - mu) * iv;
+      acc_b[j] += d[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < NV; ++j) {
+    lds[ty][tx][j] = acc_g[j];
+    lds[ty][tx][NV + j] = acc_b[j];
+  }
+  __syncthreads();
+  if (ty == 0) {
+    const int64_t o = (int64_t)blockIdx.y * n2 + c0;
+#pragma unroll
+    for (int j = 0; j < NV; ++j) {
+      part_g[o + j] =
+          lds[0][tx][j] + lds[1][tx][j] + lds[2][tx][j] + lds[3][tx][j];
+      if constexpr (!RMS)
+        part_b[o + j] = lds[0][tx][NV + j] + lds[1][tx][NV + j] +
+                        lds[2][tx][NV + j] + lds[3][tx][NV + j];
+    }
+  }
+}
+
 // fold [rb][n2] partials down to [gridDim.y][n2] (keeps the final reduce
 // wide; without this the last stage ran on ceil(n2/256) blocks serially over
 // hundreds of partial rows — measured 12% of a BERT-base step)
@@ -466,10 +514,33 @@ std::vector<at::Tensor> norm_backward_impl(const at::Tensor& grad_out,
     }
 
     const dim3 gb_block(64, 4);
-    const dim3 gb_grid(gx, rb);
-    norm_bwd_gb_partial_kernel<scalar_t, RMS><<<gb_grid, gb_block, 0, stream>>>(
-        dyp, xp, mp, ivp, s.n1, s.n2, rows_per_cta, part_g.data_ptr<float>(),
-        RMS ? nullptr : part_b.data_ptr<float>());
+    // vector stage 1 when a whole number of 64-lane x NV-column blocks
+    // tiles the row exactly; scalar fallback otherwise
+    const int gb_nv = (s.n2 % 512 == 0) ? 8
+                      : (s.n2 % 256 == 0) ? 4
+                      : (s.n2 % 128 == 0) ? 2 : 0;
+    if (gb_nv > 0) {
+      const dim3 vec_grid(s.n2 / (64 * gb_nv), rb);
+      auto launch_gb = [&](auto nv_tag) {
+        constexpr int NV = decltype(nv_tag)::value;
+        norm_bwd_gb_partial_vec_kernel<scalar_t, RMS, NV>
+            <<<vec_grid, gb_block, 0, stream>>>(
+                dyp, xp, mp, ivp, s.n1, s.n2, rows_per_cta,
+                part_g.data_ptr<float>(),
+                RMS ? nullptr : part_b.data_ptr<float>());
+      };
+      if (gb_nv == 8)
+        launch_gb(std::integral_constant<int, 8>{});
+      else if (gb_nv == 4)
+        launch_gb(std::integral_constant<int, 4>{});
+      else
+        launch_gb(std::integral_constant<int, 2>{});
+    } else {
+      const dim3 gb_grid(gx, rb);
+      norm_bwd_gb_partial_kernel<scalar_t, RMS><<<gb_grid, gb_block, 0, stream>>>(
+          dyp, xp, mp, ivp, s.n1, s.n2, rows_per_cta, part_g.data_ptr<float>(),
+          RMS ? nullptr : part_b.data_ptr<float>());
+    }
     const float* red_g = part_g.data_ptr<float>();
     const float* red_b = RMS ? nullptr : part_b.data_ptr<float>();
     int red_rb = rb;
