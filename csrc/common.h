@@ -274,6 +274,27 @@ __device__ __forceinline__ void colsum_block_fold(const float (&acc)[8], int c0,
     partials_row[c] = s_col[c];
 }
 
+// stage A of the two-stage column fold (see fold.h): coalesced chunk sums
+// (nb, C) -> (gridDim.y, C). Lanes own consecutive columns, so every wave
+// load is one contiguous 256 B segment — the one-block-per-column final
+// kernel below reads columns as strided 4 B gathers instead, which held the
+// single-stage fold to 1.5 TB/s on 12k+ partial rows (PMC evidence).
+static __global__ void unicore_col_fold_stage_kernel(
+    const float* __restrict__ in, float* __restrict__ out, int nb, int chunk,
+    int C) {
+  const int c = blockIdx.x * 256 + (int)threadIdx.x;
+  if (c >= C) return;
+  const int r0 = blockIdx.y * chunk;
+  const int r1 = min(r0 + chunk, nb);
+  float s = 0.f;
+  int r = r0;
+  for (; r + 4 <= r1; r += 4)
+    s += (in[(int64_t)r * C + c] + in[(int64_t)(r + 1) * C + c]) +
+         (in[(int64_t)(r + 2) * C + c] + in[(int64_t)(r + 3) * C + c]);
+  for (; r < r1; ++r) s += in[(int64_t)r * C + c];
+  out[(int64_t)blockIdx.y * C + c] = s;
+}
+
 // partials (nb, C) -> out (C); one block per column, deterministic.
 static __global__ void unicore_col_fold_kernel(const float* __restrict__ partials,
                                                float* __restrict__ out, int nb,

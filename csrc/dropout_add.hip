@@ -5,6 +5,7 @@
 // Backward: dx = g * keep * pinv (one pass); d_residual = g (pass-through,
 // no kernel).
 #include "common.h"
+#include "fold.h"
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
@@ -217,8 +218,8 @@ std::vector<at::Tensor> dropout_add_backward(at::Tensor grad, at::Tensor dmask,
     }
   });
   if (bgrad) {
-    unicore_col_fold_kernel<<<C, 256, 0, stream>>>(
-        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, C);
+    unicore_fold_columns(partials.data_ptr<float>(), dbias.data_ptr<float>(),
+                         grid, C, partials.options(), stream);
   }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {dx, dbias};

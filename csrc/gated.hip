@@ -7,6 +7,7 @@
 // Linears run bias-free) and the backward emits both bias grads as
 // deterministic column sums via the common.h colsum machinery.
 #include "common.h"
+#include "fold.h"
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
@@ -219,8 +220,8 @@ std::vector<at::Tensor> gated_mul_backward(at::Tensor grad, at::Tensor x,
           nullptr, C, n8);
   });
   if (bgrad) {
-    unicore_col_fold_kernel<<<2 * C, 256, 0, stream>>>(
-        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, 2 * C);
+    unicore_fold_columns(partials.data_ptr<float>(), dbias.data_ptr<float>(),
+                         grid, 2 * C, partials.options(), stream);
   }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {dx, dg, dbias};

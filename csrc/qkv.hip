@@ -8,6 +8,7 @@
 // Same HBM traffic as one pass (16 B/lane vectors); kills the separate
 // q-scale sweep, the torch cat kernel and 5 extra launches per layer.
 #include "common.h"
+#include "fold.h"
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
@@ -210,8 +211,8 @@ std::vector<at::Tensor> qkv_split_backward(at::Tensor dq, at::Tensor dk,
           D / 8, (float)scale);
   });
   if (bgrad) {
-    unicore_col_fold_kernel<<<(int)C, 256, 0, stream>>>(
-        partials.data_ptr<float>(), dbias.data_ptr<float>(), grid, (int)C);
+    unicore_fold_columns(partials.data_ptr<float>(), dbias.data_ptr<float>(),
+                         grid, (int)C, partials.options(), stream);
   }
   C10_CUDA_KERNEL_LAUNCH_CHECK();
   return {dqkv, dbias};
