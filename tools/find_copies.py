@@ -56,16 +56,20 @@ def main():
         for i in range(2):
             trainer.train_step([batches[i % len(batches)]])
         torch.cuda.synchronize()
-    evs = prof.key_averages(group_by_stack_n=16)
+    evs = prof.key_averages(group_by_input_shape=True)
     rows = [e for e in evs
-            if ("copy" in e.key.lower() or "Memcpy" in e.key)
+            if ("copy" in e.key.lower() or "contiguous" in e.key
+                or "clone" in e.key or "Memcpy" in e.key)
             and e.device_time_total > 0]
     rows.sort(key=lambda e: -e.device_time_total)
-    for e in rows[:10]:
-        print(f"== {e.key}  n={e.count}  cuda_total={e.device_time_total/1000:.2f}ms")
-        for line in (e.stack or [])[:16]:
-            print("   ", line)
-        print()
+    for e in rows[:14]:
+        print(f"== {e.key[:70]}  n={e.count}  "
+              f"cuda_total={e.device_time_total/1000:.2f}ms  "
+              f"shapes={e.input_shapes}")
+    # and the aggregate top ops for orientation
+    print("\n--- top 20 ops by CUDA time ---")
+    for e in sorted(prof.key_averages(), key=lambda x: -x.device_time_total)[:20]:
+        print(f"{e.device_time_total/1000:9.2f}ms n={e.count:5d}  {e.key[:80]}")
 
 
 if __name__ == "__main__":
