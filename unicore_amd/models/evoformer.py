@@ -23,6 +23,7 @@ from unicore_amd.models import (
 )
 from unicore_amd import utils
 from unicore_amd.modules.msa_arrange import msa_arrange, msa_merge
+from unicore_amd.modules.multihead_attention import qk_scores
 from unicore_amd.modules import (
     LayerNorm,
     dropout_add,
@@ -73,7 +74,7 @@ class MSARowAttentionWithPairBias(nn.Module):
         q = arrange(q) * self.scaling
         k = arrange(k)
         v = arrange(v)
-        scores = torch.bmm(q, k.transpose(1, 2)).view(B, H, S, L, L)
+        scores = qk_scores(q, k).view(B, H, S, L, L)
         bias = self.pair_bias(self.pair_norm(pair))  # (B, L, L, H)
         bias = bias.permute(0, 3, 1, 2).unsqueeze(2)  # (B, H, 1, L, L)
         attn = softmax_dropout(scores, self.dropout, self.training, bias=bias)
@@ -114,7 +115,7 @@ class MSAColumnAttention(nn.Module):
         q = arrange(q) * self.scaling
         k = arrange(k)
         v = arrange(v)
-        scores = torch.bmm(q, k.transpose(1, 2)).view(B * L * H, S, S)
+        scores = qk_scores(q, k).view(B * L * H, S, S)
         attn = softmax_dropout(scores, self.dropout, self.training)
         o = torch.bmm(attn, v)
         o = msa_merge(o, B, S, L, H, col=True)
