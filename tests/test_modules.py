@@ -241,3 +241,29 @@ def test_broadcast_descr_matches_torch_semantics():
                     src_batch, batch_dims, b, idx)
             checked += 1
     assert checked >= 10  # the contiguous-block patterns all verified
+
+
+def test_qk_scores_matches_autograd_bmm():
+    """qk_scores computes dk = dS^T @ q directly (contiguous layout) instead
+    of differentiating through the k-transpose view; gradients must match
+    plain bmm to fp64 precision, and dk must come out contiguous."""
+    from unicore_amd.modules.multihead_attention import qk_scores
+
+    torch.manual_seed(3)
+    q = torch.randn(6, 10, 8, dtype=torch.float64, requires_grad=True)
+    k = torch.randn(6, 10, 8, dtype=torch.float64, requires_grad=True)
+    s = qk_scores(q, k)
+    ds = torch.randn_like(s)
+    s.backward(ds)
+    dq_got, dk_got = q.grad.clone(), k.grad.clone()
+    assert dk_got.is_contiguous()
+
+    q2 = q.detach().clone().requires_grad_(True)
+    k2 = k.detach().clone().requires_grad_(True)
+    torch.bmm(q2, k2.transpose(1, 2)).backward(ds)
+    assert torch.allclose(dq_got, q2.grad, atol=1e-12)
+    assert torch.allclose(dk_got, k2.grad, atol=1e-12)
+
+    # no-grad path returns the same values without the Function
+    with torch.no_grad():
+        assert torch.allclose(qk_scores(q.detach(), k.detach()), s.detach())
