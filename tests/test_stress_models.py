@@ -184,3 +184,36 @@ def test_evoformer_hip_graph_blocks_parity():
     plain = run(False)
     graphed = run(True)
     assert graphed == pytest.approx(plain, rel=2e-2), (plain, graphed)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_evoformer_fused_preln_chain_parity(monkeypatch):
+    """The EvoformerBlock's fused pre-LN residual chain (dropout_add_ln_pre
+    joins emitting summed + normed) must match the eager composition of the
+    same math (UNICORE_FUSED_LN_JOIN=0 falls back inside the wrapper) for
+    outputs AND parameter gradients."""
+    from unicore_amd.models.evoformer import EvoformerBlock
+
+    def run(fused):
+        monkeypatch.setenv("UNICORE_FUSED_LN_JOIN", "1" if fused else "0")
+        torch.manual_seed(5)
+        blk = EvoformerBlock(d_msa=64, d_pair=32, heads=4, dropout=0.0)
+        blk = blk.cuda().bfloat16().train()
+        torch.manual_seed(9)
+        msa = torch.randn(1, 8, 16, 64, device="cuda", dtype=torch.bfloat16)
+        pair = torch.randn(1, 16, 16, 32, device="cuda", dtype=torch.bfloat16)
+        m, p = blk(msa, pair)
+        (m.float().square().mean() + p.float().square().mean()).backward()
+        grads = {k: v.grad.float().clone() for k, v in
+                 blk.named_parameters() if v.grad is not None}
+        return m.float(), p.float(), grads
+
+    m1, p1, g1 = run(True)
+    m0, p0, g0 = run(False)
+    assert torch.allclose(m1, m0, atol=3e-2, rtol=3e-2), (m1 - m0).abs().max()
+    assert torch.allclose(p1, p0, atol=3e-2, rtol=3e-2), (p1 - p0).abs().max()
+    assert set(g1) == set(g0)
+    for k in g1:
+        assert torch.allclose(g1[k], g0[k], atol=5e-2, rtol=5e-2), \
+            (k, (g1[k] - g0[k]).abs().max())

@@ -31,6 +31,7 @@ from unicore_amd.modules import (
     gelu_dropout,
     softmax_dropout,
 )
+from unicore_amd.modules.dropout_add_ln import dropout_add_ln_pre
 from unicore_amd.modules.embedding import Embedding
 
 
@@ -60,10 +61,10 @@ class MSARowAttentionWithPairBias(nn.Module):
         self.dropout = dropout
         self.scaling = self.head_dim**-0.5
 
-    def forward(self, msa, pair, skip_out_bias=False):
+    def forward(self, msa, pair, skip_out_bias=False, x_normed=None):
         B, S, L, D = msa.shape
         H, Dh = self.heads, self.head_dim
-        x = self.norm(msa)
+        x = self.norm(msa) if x_normed is None else x_normed
         q, k, v = self.qkv(x).chunk(3, dim=-1)
         # (B, S, L, H, Dh) -> (B, H, S, L, Dh): heads OUTSIDE the msa-row dim
         # so the (B, H, 1, L, L) pair bias is a contiguous-block broadcast
@@ -102,10 +103,10 @@ class MSAColumnAttention(nn.Module):
         self.dropout = dropout
         self.scaling = self.head_dim**-0.5
 
-    def forward(self, msa, skip_out_bias=False):
+    def forward(self, msa, skip_out_bias=False, x_normed=None):
         B, S, L, D = msa.shape
         H, Dh = self.heads, self.head_dim
-        x = self.norm(msa)
+        x = self.norm(msa) if x_normed is None else x_normed
         q, k, v = self.qkv(x).chunk(3, dim=-1)
 
         # attention over the S dimension for each column l
@@ -148,6 +149,14 @@ class Transition(nn.Module):
         y = self.fc2(F.gelu(self.fc1(y)))
         return y if residual is None else residual + y
 
+    def tail(self, y):
+        """fc1 -> GELU -> fc2 WITHOUT the trailing fc2 bias: the caller's
+        fused residual join folds it in (caller guarantees CUDA +
+        _fold_ok(fc1.bias, fc2.bias))."""
+        y = F.linear(y, self.fc1.weight)
+        y = gelu_dropout(y, 0.0, self.training, bias=self.fc1.bias)
+        return F.linear(y, self.fc2.weight)
+
 
 class OuterProductMean(nn.Module):
     def __init__(self, d_msa, d_pair, c=32):
@@ -158,8 +167,8 @@ class OuterProductMean(nn.Module):
         self.out = nn.Linear(c * c, d_pair)
         self.c = c
 
-    def forward(self, msa, skip_out_bias=False):
-        x = self.norm(msa)
+    def forward(self, msa, skip_out_bias=False, x_normed=None):
+        x = self.norm(msa) if x_normed is None else x_normed
         a = self.a(x)  # (B, S, L, c)
         b = self.b(x)
         o = torch.einsum("bsic,bsjd->bijcd", a.float(), b.float()) / msa.shape[1]
@@ -182,8 +191,8 @@ class TriangleMultiplication(nn.Module):
         self.gate = nn.Linear(d_pair, d_pair)
         self.outgoing = outgoing
 
-    def forward(self, pair):
-        p = self.norm(pair)
+    def forward(self, pair, p_normed=None):
+        p = self.norm(pair) if p_normed is None else p_normed
         if _fold_ok(self.a_proj.bias, self.a_gate.bias, self.b_proj.bias,
                     self.b_gate.bias, self.out.bias, self.gate.bias):
             a = gated_mul(F.linear(p, self.a_proj.weight),
@@ -219,32 +228,44 @@ class EvoformerBlock(nn.Module):
         self.pair_transition = Transition(d_pair)
 
     def forward(self, msa, pair):
-        # residual joins go through the fused dropout_add (p=0) with the
-        # producing Linear's bias folded in where the module tail allows
-        # (the triangle-mult tails are gated, so their bias cannot move)
         fold = msa.is_cuda and _fold_ok(
-            self.row_attn.out.bias, self.col_attn.out.bias, self.opm.out.bias
+            self.row_attn.out.bias, self.col_attn.out.bias,
+            self.msa_transition.fc1.bias, self.msa_transition.fc2.bias,
+            self.opm.out.bias,
+            self.pair_transition.fc1.bias, self.pair_transition.fc2.bias,
         )
         if fold:
-            msa = dropout_add(
-                self.row_attn(msa, pair, skip_out_bias=True), msa, 0.0,
-                self.training, bias=self.row_attn.out.bias,
-            )
-            msa = dropout_add(
-                self.col_attn(msa, skip_out_bias=True), msa, 0.0,
-                self.training, bias=self.col_attn.out.bias,
-            )
-        else:
-            msa = msa + self.row_attn(msa, pair)
-            msa = msa + self.col_attn(msa)
+            # pre-LN residual chain: each join is ONE fused kernel emitting
+            # both the summed stream and the next sub-module's normed input
+            # (the 6 standalone LayerNorms + 6 residual adds per block
+            # collapse into the joins; only the block's first norm remains)
+            t = self.training
+            n = self.row_attn.norm(msa)
+            y = self.row_attn(msa, pair, skip_out_bias=True, x_normed=n)
+            msa, n = dropout_add_ln_pre(y, msa, self.col_attn.norm, 0.0, t,
+                                        bias=self.row_attn.out.bias)
+            y = self.col_attn(msa, skip_out_bias=True, x_normed=n)
+            msa, n = dropout_add_ln_pre(y, msa, self.msa_transition.norm, 0.0,
+                                        t, bias=self.col_attn.out.bias)
+            y = self.msa_transition.tail(n)
+            msa, n = dropout_add_ln_pre(y, msa, self.opm.norm, 0.0, t,
+                                        bias=self.msa_transition.fc2.bias)
+            y = self.opm(msa, skip_out_bias=True, x_normed=n)
+            pair, pn = dropout_add_ln_pre(y, pair, self.tri_out.norm, 0.0, t,
+                                          bias=self.opm.out.bias)
+            y = self.tri_out(pair, p_normed=pn)
+            pair, pn = dropout_add_ln_pre(y, pair, self.tri_in.norm, 0.0, t)
+            y = self.tri_in(pair, p_normed=pn)
+            pair, pn = dropout_add_ln_pre(y, pair, self.pair_transition.norm,
+                                          0.0, t)
+            y = self.pair_transition.tail(pn)
+            pair = dropout_add(y, pair, 0.0, t,
+                               bias=self.pair_transition.fc2.bias)
+            return msa, pair
+        msa = msa + self.row_attn(msa, pair)
+        msa = msa + self.col_attn(msa)
         msa = self.msa_transition(msa, residual=msa)
-        if fold:
-            pair = dropout_add(
-                self.opm(msa, skip_out_bias=True), pair, 0.0,
-                self.training, bias=self.opm.out.bias,
-            )
-        else:
-            pair = pair + self.opm(msa)
+        pair = pair + self.opm(msa)
         pair = pair + self.tri_out(pair)
         pair = pair + self.tri_in(pair)
         pair = self.pair_transition(pair, residual=pair)
