@@ -267,3 +267,32 @@ def test_qk_scores_matches_autograd_bmm():
     # no-grad path returns the same values without the Function
     with torch.no_grad():
         assert torch.allclose(qk_scores(q.detach(), k.detach()), s.detach())
+
+
+def test_dropout_add_ln_pre_eager_fallback():
+    """CPU path of dropout_add_ln_pre: (summed, normed) must equal the
+    manual compose, with gradients flowing through both outputs."""
+    from unicore_amd.modules.dropout_add_ln import dropout_add_ln_pre
+
+    torch.manual_seed(11)
+    ln = torch.nn.LayerNorm(16)
+    x = torch.randn(4, 6, 16, requires_grad=True)
+    res = torch.randn(4, 6, 16, requires_grad=True)
+    bias = torch.randn(16, requires_grad=True)
+
+    s, n = dropout_add_ln_pre(x, res, ln, 0.0, True, bias=bias)
+    (s.square().mean() + n.square().mean()).backward()
+    got = (s.detach().clone(), n.detach().clone(),
+           x.grad.clone(), res.grad.clone(), bias.grad.clone())
+
+    for t in (x, res, bias):
+        t.grad = None
+    ln.weight.grad = ln.bias.grad = None
+    s2 = res + x + bias
+    n2 = ln(s2)
+    (s2.square().mean() + n2.square().mean()).backward()
+    assert torch.allclose(got[0], s2)
+    assert torch.allclose(got[1], n2)
+    assert torch.allclose(got[2], x.grad, atol=1e-6)
+    assert torch.allclose(got[3], res.grad, atol=1e-6)
+    assert torch.allclose(got[4], bias.grad, atol=1e-6)
