@@ -3,6 +3,11 @@
 # grad-accum 8 — BASELINE.json stress config 5.
 set -e
 n_gpu=${1:-8}
+# offline-tuned hipBLASLt/rocBLAS algorithms for this config's batched
+# einsum GEMMs (+2.3% measured; see tools/tune_evoformer.sh)
+TUNED="$(dirname "$0")/../../tools/tuned_gemm_evoformer.csv"
+EXTRA=()
+[ -f "$TUNED" ] && EXTRA=(--gemm-tuning-file "$TUNED")
 exec python -m torch.distributed.run --nnodes=1 --nproc-per-node "$n_gpu" \
   --master-addr 127.0.0.1 --master-port 29502 \
   -m unicore_cli.train \
@@ -13,4 +18,4 @@ exec python -m torch.distributed.run --nnodes=1 --nproc-per-node "$n_gpu" \
   --batch-size 1 --update-freq 8 --msa-depth 128 --residues 256 \
   --bf16 --bf16-sr --ddp-backend c10d \
   --log-interval 10 --log-format simple \
-  --save-dir ./checkpoints_evo "${@:2}"
+  --save-dir ./checkpoints_evo "${EXTRA[@]}" "${@:2}"
