@@ -13,7 +13,7 @@ NGPU=${3:-1}
 
 python -m torch.distributed.run --nnodes=1 --nproc-per-node "$NGPU" \
     --master-addr 127.0.0.1 --master-port 29600 \
-    $(python -c 'import unicore_cli.train as m; print(m.__file__)') \
+    -m unicore_cli.train \
     "$DATA" \
     --task bert --arch bert_base --loss masked_lm \
     --optimizer adam --adam-betas '(0.9, 0.98)' --adam-eps 1e-6 \
