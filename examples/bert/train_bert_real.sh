@@ -21,4 +21,5 @@ python -m torch.distributed.run --nnodes=1 --nproc-per-node "$NGPU" \
     --warmup-updates 1000 --clip-norm 1.0 \
     --batch-size 32 --max-seq-len 512 \
     --bf16 --max-update 100000 --save-interval-updates 1000 \
-    --save-dir ./bert_real_ckpt --log-format simple --log-interval 50
+    --save-dir ./bert_real_ckpt --log-format simple --log-interval 50 \
+    "${@:4}"
