@@ -75,9 +75,15 @@ class _DropoutAddLNPre(torch.autograd.Function):
         from unicore_amd import ops
 
         summed, dmask, mean, invvar, gamma = ctx.saved_tensors
-        ds, dgamma, dbeta = ops.layernorm_bwd(
-            d_norm.contiguous(), summed, mean, invvar, gamma
-        )
+        if d_norm is None:
+            # normed output unused downstream: only the stream grad flows
+            ds = torch.zeros_like(summed)
+            dgamma = torch.zeros_like(gamma)
+            dbeta = torch.zeros_like(gamma)
+        else:
+            ds, dgamma, dbeta = ops.layernorm_bwd(
+                d_norm.contiguous(), summed, mean, invvar, gamma
+            )
         if d_sum is not None:
             ds = ds + d_sum
         dbias = None
