@@ -1076,3 +1076,26 @@ def test_dropout_add_ln_dropout_mask_consistent():
     # keep rate sane
     rate = keep.float().mean().item()
     assert abs(rate - (1 - p)) < 0.05
+
+
+@requires_gpu
+def test_dropout_add_ln_pre_stream_only_grad():
+    """If only the summed stream of dropout_add_ln_pre is used downstream,
+    autograd passes d_norm=None — the backward must treat the normed
+    branch's grad as zero and still flow the stream grad through."""
+    import torch.nn as nn
+
+    from unicore_amd.modules.dropout_add_ln import dropout_add_ln_pre
+
+    torch.manual_seed(2)
+    ln = nn.LayerNorm(64).cuda().bfloat16()
+    x = torch.randn(8, 16, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    res = torch.randn_like(x).requires_grad_(True)
+    s, _n = dropout_add_ln_pre(x, res, ln, 0.0, True)
+    s.float().square().mean().backward()
+    ref = (x.detach() + res.detach()).float()
+    dref = 2 * ref / ref.numel()
+    assert torch.allclose(x.grad.float(), dref, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(res.grad.float(), dref, atol=1e-2, rtol=1e-2)
+    assert ln.weight.grad is None or ln.weight.grad.abs().max() == 0
