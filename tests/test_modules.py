@@ -296,3 +296,38 @@ def test_dropout_add_ln_pre_eager_fallback():
     assert torch.allclose(got[2], x.grad, atol=1e-6)
     assert torch.allclose(got[3], res.grad, atol=1e-6)
     assert torch.allclose(got[4], bias.grad, atol=1e-6)
+
+
+def test_model_base_contract():
+    """BaseUnicoreModel: set_num_updates must reach every submodule that
+    defines it (without recursing into itself), and load_state_dict must
+    accept the model_args kwarg (reference unicore/models/unicore_model.py
+    :18-58 behavior contract)."""
+    from unicore_amd.models.unicore_model import BaseUnicoreModel
+
+    seen = []
+
+    class Inner(torch.nn.Module):
+        def set_num_updates(self, n):
+            seen.append(n)
+
+    class M(BaseUnicoreModel):
+        def __init__(self):
+            super().__init__()
+            self.a = Inner()
+            self.b = torch.nn.Sequential(Inner(), torch.nn.Linear(2, 2))
+
+        def forward(self, x):
+            return self.b[1](x)
+
+    m = M()
+    m.set_num_updates(7)
+    assert seen == [7, 7]
+
+    sd = m.state_dict()
+    m2 = M()
+    m2.load_state_dict(sd, strict=True, model_args=object())
+    assert torch.equal(m2.b[1].weight, m.b[1].weight)
+    # extract_features defaults to forward
+    x = torch.randn(3, 2)
+    assert torch.allclose(m.extract_features(x), m(x))
